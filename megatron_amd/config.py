@@ -1,0 +1,208 @@
+"""Configuration dataclasses.
+
+Capability analog of the reference's TransformerConfig /
+ModelParallelConfig / OptimizerConfig / DistributedDataParallelConfig
+(megatron/core/transformer/transformer_config.py,
+ megatron/core/model_parallel_config.py,
+ megatron/core/optimizer/optimizer_config.py,
+ megatron/core/distributed/distributed_data_parallel_config.py) —
+re-designed as a lean, MI355X-first surface: only fields that change
+behavior on a single 8-GPU xGMI node are first-class.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from dataclasses import dataclass, field
+from typing import Callable, Optional
+
+import torch
+
+
+@dataclass
+class ParallelConfig:
+    """Process-grid shape.  World is factored rank = tp-fastest order
+    ``tp -> cp -> dp -> pp`` (reference: parallel_state.py:601
+    ``initialize_model_parallel`` default order 'tp-cp-ep-dp-pp')."""
+
+    tensor_parallel_size: int = 1
+    pipeline_parallel_size: int = 1
+    context_parallel_size: int = 1
+    expert_parallel_size: int = 1
+    virtual_pipeline_parallel_size: Optional[int] = None
+    sequence_parallel: bool = False
+    # expert tensor parallel size (defaults to tensor_parallel_size)
+    expert_tensor_parallel_size: Optional[int] = None
+
+
+@dataclass
+class TransformerConfig(ParallelConfig):
+    """Model architecture + numerics config."""
+
+    # ---- architecture ----
+    num_layers: int = 2
+    hidden_size: int = 64
+    ffn_hidden_size: Optional[int] = None
+    num_attention_heads: int = 4
+    num_query_groups: Optional[int] = None  # GQA; None -> MHA
+    kv_channels: Optional[int] = None
+    vocab_size: int = 128
+    max_position_embeddings: int = 4096
+    # 'rmsnorm' (llama) or 'layernorm'
+    normalization: str = "rmsnorm"
+    layernorm_epsilon: float = 1e-5
+    # 'swiglu' (llama), 'gelu', 'squared_relu'
+    activation: str = "swiglu"
+    add_linear_bias: bool = False  # llama-style: no bias anywhere
+    untie_embeddings_and_output_weights: bool = True
+    position_embedding_type: str = "rope"  # 'rope' | 'learned' | 'none'
+    rotary_base: float = 500000.0  # llama-3 default
+    rotary_percent: float = 1.0
+    attention_dropout: float = 0.0
+    hidden_dropout: float = 0.0
+    # sliding-window attention: None or window size (causal look-back)
+    window_size: Optional[int] = None
+    qk_layernorm: bool = False
+    softmax_scale: Optional[float] = None
+
+    # ---- MoE ----
+    num_experts: Optional[int] = None
+    moe_router_topk: int = 2
+    moe_ffn_hidden_size: Optional[int] = None
+    moe_aux_loss_coeff: float = 0.0
+    moe_z_loss_coeff: float = 0.0
+    moe_router_score_function: str = "softmax"  # 'softmax' | 'sigmoid'
+    moe_router_pre_softmax: bool = False
+    moe_shared_expert_intermediate_size: Optional[int] = None
+    moe_grouped_gemm: bool = True
+    moe_token_dispatcher_type: str = "alltoall"  # 'alltoall' | 'allgather'
+    moe_expert_capacity_factor: Optional[float] = None  # None -> dropless
+    moe_router_dtype: str = "fp32"
+    # layer frequency: 1 = every layer is MoE, k = every k-th layer
+    moe_layer_freq: int = 1
+
+    # ---- multi-token prediction ----
+    mtp_num_layers: int = 0
+    mtp_loss_scaling_factor: float = 0.1
+
+    # ---- numerics ----
+    params_dtype: torch.dtype = torch.float32
+    bf16: bool = False
+    fp16: bool = False
+    fp8: bool = False  # fp8 GEMM path (CDNA4 e4m3fn) — later phase
+    attention_softmax_in_fp32: bool = True
+    init_method_std: float = 0.02
+    # activation recompute: None | 'full' | 'selective'
+    recompute_granularity: Optional[str] = None
+    recompute_num_layers: Optional[int] = None
+    # distribute saved activations over TP group when recomputing
+    distribute_saved_activations: bool = False
+
+    # ---- execution ----
+    # overlap TP dgrad all-reduce with wgrad GEMM (reference layers.py:622)
+    async_tensor_model_parallel_allreduce: bool = True
+    gradient_accumulation_fusion: bool = True
+    persist_layer_norm: bool = True
+    deterministic_mode: bool = False
+    cross_entropy_fusion: bool = True
+    # hook points used by pipeline schedules (reference
+    # model_parallel_config.py:211-223 three-hook contract)
+    no_sync_func: Optional[Callable] = None
+    grad_sync_func: Optional[Callable] = None
+    param_sync_func: Optional[Callable] = None
+    finalize_model_grads_func: Optional[Callable] = None
+    # pipeline schedule knobs
+    microbatch_group_size_per_vp_stage: Optional[int] = None
+    overlap_p2p_comm: bool = True
+    batch_p2p_comm: bool = False
+    variable_seq_lengths: bool = False
+    pipeline_dtype: Optional[torch.dtype] = None
+    grad_scale_func: Optional[Callable] = None
+    enable_autocast: bool = False
+    autocast_dtype: Optional[torch.dtype] = None
+    num_microbatches_with_partial_activation_checkpoints: Optional[int] = None
+    deallocate_pipeline_outputs: bool = True
+    defer_embedding_wgrad_compute: bool = False
+    calculate_per_token_loss: bool = False
+
+    timers: Optional[object] = None
+
+    def __post_init__(self):
+        if self.ffn_hidden_size is None:
+            self.ffn_hidden_size = 4 * self.hidden_size
+        if self.kv_channels is None:
+            assert self.hidden_size % self.num_attention_heads == 0
+            self.kv_channels = self.hidden_size // self.num_attention_heads
+        if self.num_query_groups is None:
+            self.num_query_groups = self.num_attention_heads
+        if self.expert_tensor_parallel_size is None:
+            self.expert_tensor_parallel_size = self.tensor_parallel_size
+        if self.moe_ffn_hidden_size is None:
+            self.moe_ffn_hidden_size = self.ffn_hidden_size
+        if self.bf16:
+            self.params_dtype = torch.bfloat16
+        if self.fp16:
+            self.params_dtype = torch.float16
+        if self.pipeline_dtype is None:
+            self.pipeline_dtype = self.params_dtype
+        assert self.num_attention_heads % self.num_query_groups == 0
+        if self.tensor_parallel_size > 1:
+            assert self.num_attention_heads % self.tensor_parallel_size == 0
+            assert (
+                self.num_query_groups % self.tensor_parallel_size == 0
+                or self.tensor_parallel_size % self.num_query_groups == 0
+            )
+        if self.sequence_parallel:
+            assert self.tensor_parallel_size > 1 or True  # allowed; no-op at tp=1
+        if self.num_experts is not None:
+            assert self.num_experts % self.expert_parallel_size == 0
+
+    def replace(self, **kw) -> "TransformerConfig":
+        return dataclasses.replace(self, **kw)
+
+
+@dataclass
+class OptimizerConfig:
+    """Analog of the reference OptimizerConfig (optimizer/optimizer_config.py)."""
+
+    optimizer: str = "adam"
+    lr: float = 1e-4
+    min_lr: float = 0.0
+    weight_decay: float = 0.1
+    adam_beta1: float = 0.9
+    adam_beta2: float = 0.95
+    adam_eps: float = 1e-8
+    clip_grad: float = 1.0
+    # mixed precision
+    bf16: bool = False
+    fp16: bool = False
+    params_dtype: torch.dtype = torch.float32
+    loss_scale: Optional[float] = None  # static; None -> dynamic when fp16
+    initial_loss_scale: float = 2**32
+    min_loss_scale: float = 1.0
+    loss_scale_window: int = 1000
+    hysteresis: int = 2
+    # ZeRO-1 distributed optimizer
+    use_distributed_optimizer: bool = False
+    overlap_param_gather: bool = False
+    # lr schedule
+    lr_decay_style: str = "cosine"  # 'constant' | 'linear' | 'cosine' | 'wsd'
+    lr_warmup_iters: int = 0
+    lr_decay_iters: Optional[int] = None
+    lr_wsd_decay_iters: Optional[int] = None
+    start_weight_decay: Optional[float] = None
+    end_weight_decay: Optional[float] = None
+    weight_decay_incr_style: str = "constant"
+
+
+@dataclass
+class DDPConfig:
+    """Analog of DistributedDataParallelConfig."""
+
+    grad_reduce_in_fp32: bool = False
+    overlap_grad_reduce: bool = True
+    use_distributed_optimizer: bool = False
+    bucket_size: Optional[int] = 40_000_000  # elements per bucket target
+    average_in_collective: bool = True
+    check_for_nan_in_grad: bool = False
+    align_param_gather: bool = False

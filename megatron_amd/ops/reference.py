@@ -1,0 +1,125 @@
+"""Plain-PyTorch reference implementations of every fused HIP op.
+
+These are the numerics oracles (tests compare the HIP kernels against these
+in fp32) and the CPU execution path.  They are NOT used on GPU unless the
+native extension is deliberately disabled — on a GPU box a missing extension
+raises (see ops/__init__.py).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    dtype = x.dtype
+    xf = x.float()
+    var = xf.pow(2).mean(dim=-1, keepdim=True)
+    out = xf * torch.rsqrt(var + eps)
+    return (out * weight.float()).to(dtype)
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor], eps: float) -> torch.Tensor:
+    dtype = x.dtype
+    out = F.layer_norm(x.float(), (x.size(-1),), weight.float(), bias.float() if bias is not None else None, eps)
+    return out.to(dtype)
+
+
+def swiglu(x: torch.Tensor) -> torch.Tensor:
+    """x: [..., 2*ffn]; llama convention: silu(x1) * x2 with x1 = first half."""
+    x1, x2 = x.chunk(2, dim=-1)
+    return (F.silu(x1.float()) * x2.float()).to(x.dtype)
+
+
+def geglu(x: torch.Tensor) -> torch.Tensor:
+    x1, x2 = x.chunk(2, dim=-1)
+    return (F.gelu(x1.float(), approximate="tanh") * x2.float()).to(x.dtype)
+
+
+def squared_relu(x: torch.Tensor) -> torch.Tensor:
+    return torch.pow(F.relu(x.float()), 2).to(x.dtype)
+
+
+def rope_freqs(
+    seq_len: int,
+    dim: int,
+    base: float = 10000.0,
+    device=None,
+    dtype=torch.float32,
+    rotary_percent: float = 1.0,
+) -> torch.Tensor:
+    """Precomputed rotation angles [seq, dim_rot/2] (host-side table — G13/App-B:
+    on-device trig turns RoPE memory-bound into VALU-bound)."""
+    rot_dim = int(dim * rotary_percent)
+    inv_freq = 1.0 / (base ** (torch.arange(0, rot_dim, 2, device=device, dtype=torch.float32) / rot_dim))
+    t = torch.arange(seq_len, device=device, dtype=torch.float32)
+    freqs = torch.outer(t, inv_freq)  # [seq, rot_dim/2]
+    return freqs.to(dtype)
+
+
+def rope_apply(x: torch.Tensor, freqs: torch.Tensor) -> torch.Tensor:
+    """Rotate-half RoPE (GPT-NeoX / llama convention).
+
+    x: [s, b, h, d]; freqs: [s, d_rot/2] (fp32).  First d_rot channels rotated.
+    """
+    s, b, h, d = x.shape
+    d_rot = freqs.size(1) * 2
+    xf = x.float()
+    x_rot, x_pass = xf[..., :d_rot], xf[..., d_rot:]
+    cos = torch.cos(freqs).view(s, 1, 1, -1)
+    sin = torch.sin(freqs).view(s, 1, 1, -1)
+    x1, x2 = x_rot[..., : d_rot // 2], x_rot[..., d_rot // 2 :]
+    out_rot = torch.cat((x1 * cos - x2 * sin, x2 * cos + x1 * sin), dim=-1)
+    if d_rot < d:
+        out_rot = torch.cat((out_rot, x_pass), dim=-1)
+    return out_rot.to(x.dtype)
+
+
+def bias_dropout_add(
+    x: torch.Tensor, bias: Optional[torch.Tensor], residual: torch.Tensor, p: float, training: bool
+) -> torch.Tensor:
+    if bias is not None:
+        x = x + bias
+    if p > 0.0:
+        x = F.dropout(x, p=p, training=training)
+    return residual + x
+
+
+def attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = True,
+    scale: Optional[float] = None,
+    window: Optional[int] = None,
+) -> torch.Tensor:
+    """Unfused reference attention.
+
+    q: [s, b, hq, d], k/v: [s, b, hkv, d] -> out [s, b, hq, d].
+    GQA: hq is a multiple of hkv.  fp32 softmax.
+    """
+    s, b, hq, d = q.shape
+    hkv = k.shape[2]
+    rep = hq // hkv
+    if scale is None:
+        scale = 1.0 / math.sqrt(d)
+    qf = q.permute(1, 2, 0, 3).float()  # [b, hq, s, d]
+    kf = k.permute(1, 2, 0, 3).float()
+    vf = v.permute(1, 2, 0, 3).float()
+    if rep > 1:
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b, hq, s, s]
+    sk = k.shape[0]
+    if causal:
+        mask = torch.ones(s, sk, dtype=torch.bool, device=q.device).tril_()
+        if window is not None:
+            mask &= torch.ones(s, sk, dtype=torch.bool, device=q.device).triu_(-window + 1)
+        scores = scores.masked_fill(~mask, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    out = torch.matmul(probs, vf)  # [b, hq, s, d]
+    return out.permute(2, 0, 1, 3).to(q.dtype)

@@ -1,0 +1,390 @@
+"""Process-group topology ("the grid").
+
+MI355X-first re-think of the reference's ``parallel_state.py`` (2,692 LoC of
+~40 cached global groups; reference parallel_state.py:601
+``initialize_model_parallel``, :269 ``generate_masked_orthogonal_rank_groups``).
+
+Design: one :class:`ParallelGrid` object owns an N-D factoring of the world
+with named axes ordered fastest-varying-first ``(tp, cp, dp, pp)``; expert
+groups (ep/etp) are carved out of the (cp, dp) span.  Module-level singleton
+accessors mirror the reference's API shape so the rest of the framework reads
+naturally, but all state lives on the grid object (testable without globals).
+
+Backend notes: on ROCm, torch.distributed backend "nccl" IS RCCL over xGMI.
+One process per GPU.  Gloo twin groups are created for CPU-side exchanges
+(checkpoint shard exchange) when gloo is available.
+"""
+
+from __future__ import annotations
+
+import itertools
+import os
+from datetime import timedelta
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+# ---------------------------------------------------------------------------
+# rank-group math
+# ---------------------------------------------------------------------------
+
+
+def decompose(index: int, shape: List[int]) -> List[int]:
+    """index -> coords with shape[0] the fastest-varying axis."""
+    coords = []
+    for s in shape:
+        coords.append(index % s)
+        index //= s
+    return coords
+
+
+def compose(coords: List[int], shape: List[int]) -> int:
+    index, stride = 0, 1
+    for c, s in zip(coords, shape):
+        index += c * stride
+        stride *= s
+    return index
+
+
+def orthogonal_rank_groups(world_size: int, shape: List[int], mask: List[bool]) -> List[List[int]]:
+    """Enumerate rank groups for an orthogonal axis subset.
+
+    ``shape`` factors ``world_size`` with axis 0 fastest-varying.  Axes with
+    ``mask[i] == True`` vary *within* a group; the rest enumerate groups.
+    Equivalent in behavior to the reference's
+    generate_masked_orthogonal_rank_groups (parallel_state.py:269), written
+    from the coordinate decomposition directly.
+    """
+    assert len(shape) == len(mask)
+    total = 1
+    for s in shape:
+        total *= s
+    assert total == world_size, f"shape {shape} does not factor world {world_size}"
+
+    in_axes = [i for i, m in enumerate(mask) if m]
+    out_axes = [i for i, m in enumerate(mask) if not m]
+    groups = []
+    for out_coords in itertools.product(*[range(shape[i]) for i in reversed(out_axes)]):
+        out_coords = list(reversed(out_coords))
+        group = []
+        for in_coords in itertools.product(*[range(shape[i]) for i in reversed(in_axes)]):
+            in_coords = list(reversed(in_coords))
+            coords = [0] * len(shape)
+            for axis, c in zip(out_axes, out_coords):
+                coords[axis] = c
+            for axis, c in zip(in_axes, in_coords):
+                coords[axis] = c
+            group.append(compose(coords, shape))
+        groups.append(sorted(group))
+    return groups
+
+
+# ---------------------------------------------------------------------------
+# the grid
+# ---------------------------------------------------------------------------
+
+
+class _GroupHandle:
+    """A process group plus its rank list (and optional gloo twin)."""
+
+    __slots__ = ("group", "ranks", "gloo")
+
+    def __init__(self, group, ranks, gloo=None):
+        self.group = group
+        self.ranks = ranks
+        self.gloo = gloo
+
+
+class ParallelGrid:
+    """Owns every process group for one training job."""
+
+    AXES = ("tp", "cp", "dp", "pp")  # fastest-varying first
+
+    def __init__(
+        self,
+        tensor_parallel_size: int = 1,
+        pipeline_parallel_size: int = 1,
+        context_parallel_size: int = 1,
+        expert_parallel_size: int = 1,
+        expert_tensor_parallel_size: Optional[int] = None,
+        virtual_pipeline_parallel_size: Optional[int] = None,
+        world_size: Optional[int] = None,
+        rank: Optional[int] = None,
+        backend: Optional[str] = None,
+        create_gloo_groups: bool = True,
+    ):
+        if world_size is None:
+            world_size = dist.get_world_size()
+        if rank is None:
+            rank = dist.get_rank()
+        tp, cp, pp = tensor_parallel_size, context_parallel_size, pipeline_parallel_size
+        assert world_size % (tp * cp * pp) == 0, (
+            f"world {world_size} not divisible by tp*cp*pp = {tp}*{cp}*{pp}"
+        )
+        dp = world_size // (tp * cp * pp)
+        ep = expert_parallel_size
+        etp = expert_tensor_parallel_size if expert_tensor_parallel_size is not None else tp
+        # expert grid replaces (tp, cp, dp) span with (etp, ep, edp)
+        assert (tp * cp * dp) % (etp * ep) == 0, (
+            f"tp*cp*dp = {tp*cp*dp} not divisible by etp*ep = {etp}*{ep}"
+        )
+        self.tp, self.cp, self.dp, self.pp, self.ep, self.etp = tp, cp, dp, pp, ep, etp
+        self.edp = (tp * cp * dp) // (etp * ep)
+        self.vpp = virtual_pipeline_parallel_size
+        self.world_size, self.rank = world_size, rank
+        self.shape = [tp, cp, dp, pp]
+        self.coords = decompose(rank, self.shape)  # [tp, cp, dp, pp]
+
+        self._groups: Dict[str, _GroupHandle] = {}
+        self._vpp_rank: Optional[int] = 0 if virtual_pipeline_parallel_size else None
+
+        initialized = dist.is_initialized()
+        if backend is None:
+            backend = dist.get_backend() if initialized else "gloo"
+        self._backend = str(backend)
+        self._gloo_ok = create_gloo_groups and dist.is_gloo_available() and initialized
+
+        def make(name: str, mask: List[bool], shape: Optional[List[int]] = None, gloo=False):
+            groups = orthogonal_rank_groups(world_size, shape or self.shape, mask)
+            self._register(name, groups, gloo=gloo)
+
+        # core orthogonal groups over (tp, cp, dp, pp)
+        make("tp", [True, False, False, False])
+        make("cp", [False, True, False, False])
+        make("dp", [False, False, True, False], gloo=True)
+        make("pp", [False, False, False, True])
+        make("tp_cp", [True, True, False, False])
+        make("dp_cp", [False, True, True, False], gloo=True)
+        make("mp", [True, True, False, True])  # model-parallel: tp x cp x pp
+        make("tp_dp_cp", [True, True, True, False])
+
+        # expert groups: factor the (tp*cp*dp) span as (etp, ep, edp), pp slowest
+        espan_shape = [etp, ep, self.edp, pp]
+        make("etp", [True, False, False, False], espan_shape)
+        make("ep", [False, True, False, False], espan_shape)
+        make("expert_dp", [False, False, True, False], espan_shape, gloo=True)
+        make("etp_ep", [True, True, False, False], espan_shape)
+
+        # embedding group: first and last pp stage within each (dp, cp, tp) column
+        emb_groups = []
+        for g in orthogonal_rank_groups(world_size, self.shape, [False, False, False, True]):
+            if len(g) == 1:
+                emb_groups.append(g)
+            else:
+                emb_groups.append([g[0], g[-1]])
+        self._register("embd", emb_groups)
+        pos_groups = [[g[0]] for g in orthogonal_rank_groups(world_size, self.shape, [False, False, False, True])]
+        self._register("pos_embd", pos_groups)
+
+    # -- group creation ----------------------------------------------------
+
+    def _register(self, name: str, groups: List[List[int]], gloo: bool = False):
+        my_group, my_ranks, my_gloo = None, None, None
+        use_dist = dist.is_initialized() and self.world_size > 1
+        for ranks in groups:
+            pg = dist.new_group(ranks=ranks, backend=self._backend) if use_dist else None
+            pg_gloo = (
+                dist.new_group(ranks=ranks, backend="gloo", timeout=timedelta(minutes=30))
+                if (use_dist and gloo and self._gloo_ok and self._backend != "gloo")
+                else None
+            )
+            if self.rank in ranks:
+                my_group, my_ranks, my_gloo = pg, ranks, pg_gloo
+        if my_ranks is None:
+            my_ranks = [self.rank]
+        self._groups[name] = _GroupHandle(my_group, my_ranks, my_gloo)
+
+    # -- accessors -----------------------------------------------------------
+
+    def group(self, name: str):
+        return self._groups[name].group
+
+    def ranks(self, name: str) -> List[int]:
+        return self._groups[name].ranks
+
+    def gloo_group(self, name: str):
+        h = self._groups[name]
+        return h.gloo if h.gloo is not None else h.group
+
+    def size(self, name: str) -> int:
+        return len(self._groups[name].ranks)
+
+    def rank_in(self, name: str) -> int:
+        return self._groups[name].ranks.index(self.rank)
+
+    # convenience properties mirroring the reference accessor names
+    @property
+    def tp_rank(self) -> int:
+        return self.coords[0]
+
+    @property
+    def cp_rank(self) -> int:
+        return self.coords[1]
+
+    @property
+    def dp_rank(self) -> int:
+        return self.coords[2]
+
+    @property
+    def pp_rank(self) -> int:
+        return self.coords[3]
+
+    @property
+    def vpp_rank(self) -> Optional[int]:
+        return self._vpp_rank
+
+    def set_vpp_rank(self, r: Optional[int]):
+        self._vpp_rank = r
+
+    def is_pipeline_first_stage(self, ignore_virtual: bool = False) -> bool:
+        if not ignore_virtual and self.vpp is not None and self._vpp_rank is not None:
+            if self._vpp_rank != 0:
+                return False
+        return self.pp_rank == 0
+
+    def is_pipeline_last_stage(self, ignore_virtual: bool = False) -> bool:
+        if not ignore_virtual and self.vpp is not None and self._vpp_rank is not None:
+            if self._vpp_rank != self.vpp - 1:
+                return False
+        return self.pp_rank == self.pp - 1
+
+    def pipeline_prev_rank(self) -> int:
+        ranks = self.ranks("pp")
+        return ranks[(self.pp_rank - 1) % self.pp]
+
+    def pipeline_next_rank(self) -> int:
+        ranks = self.ranks("pp")
+        return ranks[(self.pp_rank + 1) % self.pp]
+
+    def pipeline_first_rank(self) -> int:
+        return self.ranks("pp")[0]
+
+    def pipeline_last_rank(self) -> int:
+        return self.ranks("pp")[-1]
+
+    def destroy(self):
+        self._groups.clear()
+
+
+# ---------------------------------------------------------------------------
+# module-level singleton (the reference's "mpu" role)
+# ---------------------------------------------------------------------------
+
+_GRID: Optional[ParallelGrid] = None
+
+
+def initialize_model_parallel(
+    tensor_parallel_size: int = 1,
+    pipeline_parallel_size: int = 1,
+    context_parallel_size: int = 1,
+    expert_parallel_size: int = 1,
+    expert_tensor_parallel_size: Optional[int] = None,
+    virtual_pipeline_parallel_size: Optional[int] = None,
+    **kw,
+) -> ParallelGrid:
+    """Build and install the global grid (reference parallel_state.py:601)."""
+    global _GRID
+    _GRID = ParallelGrid(
+        tensor_parallel_size=tensor_parallel_size,
+        pipeline_parallel_size=pipeline_parallel_size,
+        context_parallel_size=context_parallel_size,
+        expert_parallel_size=expert_parallel_size,
+        expert_tensor_parallel_size=expert_tensor_parallel_size,
+        virtual_pipeline_parallel_size=virtual_pipeline_parallel_size,
+        **kw,
+    )
+    return _GRID
+
+
+def get_grid() -> ParallelGrid:
+    assert _GRID is not None, "call initialize_model_parallel() first"
+    return _GRID
+
+
+def grid_initialized() -> bool:
+    return _GRID is not None
+
+
+def destroy_model_parallel():
+    global _GRID
+    if _GRID is not None:
+        _GRID.destroy()
+    _GRID = None
+
+
+def init_distributed(backend: Optional[str] = None, timeout_minutes: int = 30):
+    """Initialize torch.distributed from torchrun env vars; pick device."""
+    if dist.is_initialized():
+        return
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29511")
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if backend == "nccl" and torch.cuda.is_available():
+        local_rank = int(os.environ.get("LOCAL_RANK", rank % max(torch.cuda.device_count(), 1)))
+        torch.cuda.set_device(local_rank)
+    dist.init_process_group(backend=backend, rank=rank, world_size=world, timeout=timedelta(minutes=timeout_minutes))
+
+
+# thin functional accessors (used widely; mirror reference get_*_group)
+def get_tensor_model_parallel_group():
+    return get_grid().group("tp")
+
+
+def get_tensor_model_parallel_world_size() -> int:
+    return get_grid().size("tp") if grid_initialized() else 1
+
+
+def get_tensor_model_parallel_rank() -> int:
+    return get_grid().rank_in("tp") if grid_initialized() else 0
+
+
+def get_data_parallel_group(with_context_parallel: bool = False):
+    return get_grid().group("dp_cp" if with_context_parallel else "dp")
+
+
+def get_data_parallel_world_size(with_context_parallel: bool = False) -> int:
+    return get_grid().size("dp_cp" if with_context_parallel else "dp")
+
+
+def get_data_parallel_rank(with_context_parallel: bool = False) -> int:
+    return get_grid().rank_in("dp_cp" if with_context_parallel else "dp")
+
+
+def get_pipeline_model_parallel_group():
+    return get_grid().group("pp")
+
+
+def get_pipeline_model_parallel_world_size() -> int:
+    return get_grid().size("pp") if grid_initialized() else 1
+
+
+def get_pipeline_model_parallel_rank() -> int:
+    return get_grid().rank_in("pp") if grid_initialized() else 0
+
+
+def get_context_parallel_group():
+    return get_grid().group("cp")
+
+
+def get_context_parallel_world_size() -> int:
+    return get_grid().size("cp") if grid_initialized() else 1
+
+
+def get_context_parallel_rank() -> int:
+    return get_grid().rank_in("cp") if grid_initialized() else 0
+
+
+def get_expert_model_parallel_group():
+    return get_grid().group("ep")
+
+
+def get_expert_model_parallel_world_size() -> int:
+    return get_grid().size("ep") if grid_initialized() else 1
+
+
+def get_expert_model_parallel_rank() -> int:
+    return get_grid().rank_in("ep") if grid_initialized() else 0

@@ -1,0 +1,89 @@
+"""Self-attention with GQA, RoPE, TP/SP sharding.
+
+Capability analog of reference megatron/core/transformer/attention.py
+(Attention :289, SelfAttention :1645): fused QKV column-parallel linear ->
+GQA split -> RoPE -> fused (flash-style) core attention -> row-parallel proj.
+The core attention is the hand-written CDNA4 MFMA kernel (ops.flash_attention,
+K1 in SURVEY.md §2.3); the torch reference path runs on CPU.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from megatron_amd import ops
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.layers import ColumnParallelLinear, RowParallelLinear
+
+
+class SelfAttention(nn.Module):
+    def __init__(self, config, layer_number: int = 0):
+        super().__init__()
+        self.config = config
+        self.layer_number = layer_number
+        tp = G.get_tensor_model_parallel_world_size()
+        self.hidden_size = config.hidden_size
+        self.kv_channels = config.kv_channels
+        self.num_heads = config.num_attention_heads
+        self.num_query_groups = config.num_query_groups
+        assert self.num_heads % tp == 0
+        self.num_heads_per_partition = self.num_heads // tp
+        if self.num_query_groups >= tp:
+            assert self.num_query_groups % tp == 0
+            self.num_query_groups_per_partition = self.num_query_groups // tp
+        else:
+            raise NotImplementedError("kv replication for tp > num_query_groups not yet supported")
+        q_size = self.num_heads * self.kv_channels
+        kv_size = self.num_query_groups * self.kv_channels
+        self.qkv_size = q_size + 2 * kv_size
+        self.linear_qkv = ColumnParallelLinear(
+            self.hidden_size, self.qkv_size, config=config, bias=config.add_linear_bias
+        )
+        self.linear_proj = RowParallelLinear(
+            q_size, self.hidden_size, config=config, bias=config.add_linear_bias
+        )
+        self.softmax_scale = config.softmax_scale or (1.0 / (self.kv_channels**0.5))
+        self.window = config.window_size
+        if config.qk_layernorm:
+            self.q_layernorm = nn.Parameter(torch.ones(self.kv_channels, dtype=config.params_dtype))
+            self.k_layernorm = nn.Parameter(torch.ones(self.kv_channels, dtype=config.params_dtype))
+        else:
+            self.q_layernorm = None
+            self.k_layernorm = None
+
+    def forward(self, hidden_states: torch.Tensor, rotary_freqs: Optional[torch.Tensor] = None,
+                attention_mask=None, inference_context=None) -> torch.Tensor:
+        # hidden_states: [s(/tp if SP), b, h]
+        qkv, _ = self.linear_qkv(hidden_states)  # [s, b, qkv_size/tp]
+        s, b = qkv.shape[0], qkv.shape[1]
+        ng = self.num_query_groups_per_partition
+        rep = self.num_heads_per_partition // ng
+        d = self.kv_channels
+        # fused-QKV layout: per query group [rep*d q | d k | d v] (reference attention.py GQA split)
+        qkv = qkv.view(s, b, ng, (rep + 2) * d)
+        q, k, v = torch.split(qkv, [rep * d, d, d], dim=3)
+        q = q.reshape(s, b, ng * rep, d)
+        k = k.reshape(s, b, ng, d)
+        v = v.reshape(s, b, ng, d)
+
+        if self.q_layernorm is not None:
+            q = ops.rms_norm(q, self.q_layernorm, self.config.layernorm_epsilon)
+            k = ops.rms_norm(k, self.k_layernorm, self.config.layernorm_epsilon)
+
+        if rotary_freqs is not None:
+            q = ops.rope_apply(q, rotary_freqs)
+            k = ops.rope_apply(k, rotary_freqs)
+
+        if inference_context is not None:
+            k, v = inference_context.append_kv(self.layer_number, k, v)
+            core_out = ops.flash_attention(
+                q, k, v, causal=(q.shape[0] == k.shape[0]), scale=self.softmax_scale, window=self.window
+            )
+        else:
+            core_out = ops.flash_attention(q, k, v, causal=True, scale=self.softmax_scale, window=self.window)
+        core_out = core_out.reshape(s, b, ng * rep * d)
+        out, _ = self.linear_proj(core_out)
+        return out

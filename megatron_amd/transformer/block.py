@@ -1,0 +1,141 @@
+"""TransformerLayer + TransformerBlock.
+
+Capability analog of reference megatron/core/transformer/transformer_layer.py
+(:302) and transformer_block.py (:486): pre-norm residual layers
+(input_layernorm -> attention -> residual -> pre_mlp_layernorm -> MLP/MoE ->
+residual), per-pipeline-stage layer construction, final norm, activation
+recompute dispatch.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from megatron_amd import ops
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.random import checkpoint as rng_checkpoint
+from megatron_amd.transformer.attention import SelfAttention
+from megatron_amd.transformer.mlp import MLP
+
+
+class Norm(nn.Module):
+    """RMSNorm (default, llama) or LayerNorm via the fused HIP kernel (K3)."""
+
+    def __init__(self, config, hidden_size: Optional[int] = None):
+        super().__init__()
+        self.config = config
+        h = hidden_size or config.hidden_size
+        self.kind = config.normalization
+        self.eps = config.layernorm_epsilon
+        self.weight = nn.Parameter(torch.ones(h, dtype=config.params_dtype))
+        self.weight.sequence_parallel_dup = True  # replicated over TP; grads all-reduced
+        if self.kind == "layernorm":
+            self.bias = nn.Parameter(torch.zeros(h, dtype=config.params_dtype))
+        else:
+            self.register_parameter("bias", None)
+
+    def forward(self, x):
+        if self.kind == "rmsnorm":
+            return ops.rms_norm(x, self.weight, self.eps)
+        return ops.reference.layer_norm(x, self.weight, self.bias, self.eps)
+
+
+def _make_mixer(config, layer_number: int):
+    if config.num_experts is not None and (layer_number % config.moe_layer_freq == config.moe_layer_freq - 1):
+        from megatron_amd.moe.moe_layer import MoELayer
+
+        return MoELayer(config, layer_number=layer_number)
+    return MLP(config)
+
+
+class TransformerLayer(nn.Module):
+    def __init__(self, config, layer_number: int = 0):
+        super().__init__()
+        self.config = config
+        self.layer_number = layer_number
+        self.input_layernorm = Norm(config)
+        self.self_attention = SelfAttention(config, layer_number=layer_number)
+        self.pre_mlp_layernorm = Norm(config)
+        self.mlp = _make_mixer(config, layer_number)
+        self.hidden_dropout = config.hidden_dropout
+
+    def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None):
+        residual = hidden_states
+        x = self.input_layernorm(hidden_states)
+        x = self.self_attention(x, rotary_freqs=rotary_freqs, attention_mask=attention_mask,
+                                inference_context=inference_context)
+        x = ops.bias_dropout_add(x, None, residual, self.hidden_dropout, self.training)
+        residual = x
+        y = self.pre_mlp_layernorm(x)
+        y = self.mlp(y)
+        y = ops.bias_dropout_add(y, None, residual, self.hidden_dropout, self.training)
+        return y
+
+
+def get_num_layers_to_build(config) -> int:
+    """Per-pipeline-stage layer count (reference transformer_block.py:71)."""
+    pp = config.pipeline_parallel_size
+    vpp = config.virtual_pipeline_parallel_size
+    chunks = pp * (vpp or 1)
+    assert config.num_layers % chunks == 0, (
+        f"num_layers {config.num_layers} must divide pp*vpp = {chunks}"
+    )
+    return config.num_layers // chunks
+
+
+def get_layer_offset(config, vp_stage: Optional[int] = None) -> int:
+    """Global index of this stage's first layer (reference
+    transformer_layer.py:57 get_transformer_layer_offset)."""
+    pp_rank = G.get_pipeline_model_parallel_rank() if G.grid_initialized() else 0
+    pp = config.pipeline_parallel_size
+    vpp = config.virtual_pipeline_parallel_size
+    per_chunk = get_num_layers_to_build(config)
+    if vpp is not None and vp_stage is not None:
+        # interleaved: chunk c on pp rank r holds layers [ (c*pp + r) * per_chunk , ... )
+        return (vp_stage * pp + pp_rank) * per_chunk
+    return pp_rank * per_chunk
+
+
+class TransformerBlock(nn.Module):
+    def __init__(self, config, pre_process: bool = True, post_process: bool = True,
+                 vp_stage: Optional[int] = None):
+        super().__init__()
+        self.config = config
+        self.pre_process = pre_process
+        self.post_process = post_process
+        num_layers = get_num_layers_to_build(config)
+        offset = get_layer_offset(config, vp_stage)
+        self.layers = nn.ModuleList(
+            [TransformerLayer(config, layer_number=offset + i) for i in range(num_layers)]
+        )
+        self.final_layernorm = Norm(config) if post_process else None
+
+    def _checkpointed(self, layer, *args):
+        def run(*a):
+            return layer(*a)
+
+        return rng_checkpoint(run, self.config.distribute_saved_activations, *args)
+
+    def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None):
+        recompute = (
+            self.config.recompute_granularity == "full" and self.training and inference_context is None
+        )
+        num_ckpt = (
+            self.config.recompute_num_layers
+            if self.config.recompute_num_layers is not None
+            else len(self.layers)
+        )
+        for i, layer in enumerate(self.layers):
+            if recompute and i < num_ckpt:
+                hidden_states = self._checkpointed(layer, hidden_states, rotary_freqs)
+            else:
+                hidden_states = layer(
+                    hidden_states, rotary_freqs=rotary_freqs, attention_mask=attention_mask,
+                    inference_context=inference_context,
+                )
+        if self.final_layernorm is not None:
+            hidden_states = self.final_layernorm(hidden_states)
+        return hidden_states

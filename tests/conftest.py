@@ -1,0 +1,17 @@
+import pytest
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: test requires an MI355X GPU (run via gpurun)")
+
+
+@pytest.fixture(autouse=True)
+def _reset_parallel_state():
+    yield
+    # tests that initialize the single-process grid must not leak it
+    from megatron_amd.parallel import grid as G
+
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        G.destroy_model_parallel()

@@ -1,0 +1,64 @@
+"""Rank-group math (no torch.distributed needed)."""
+
+import pytest
+
+from megatron_amd.parallel.grid import ParallelGrid, compose, decompose, orthogonal_rank_groups
+
+
+def test_decompose_compose_roundtrip():
+    shape = [2, 3, 4]
+    for i in range(24):
+        assert compose(decompose(i, shape), shape) == i
+
+
+def test_orthogonal_groups_tp_dp():
+    # world 8 = tp2 * cp1 * dp2 * pp2
+    shape = [2, 1, 2, 2]
+    tp_groups = orthogonal_rank_groups(8, shape, [True, False, False, False])
+    assert sorted(map(tuple, tp_groups)) == [(0, 1), (2, 3), (4, 5), (6, 7)]
+    dp_groups = orthogonal_rank_groups(8, shape, [False, False, True, False])
+    assert sorted(map(tuple, dp_groups)) == [(0, 2), (1, 3), (4, 6), (5, 7)]
+    pp_groups = orthogonal_rank_groups(8, shape, [False, False, False, True])
+    assert sorted(map(tuple, pp_groups)) == [(0, 4), (1, 5), (2, 6), (3, 7)]
+
+
+def test_every_rank_in_exactly_one_group():
+    shape = [2, 2, 2, 2]
+    for mask in ([True, False, True, False], [False, True, False, True], [True, True, False, False]):
+        groups = orthogonal_rank_groups(16, shape, mask)
+        seen = sorted(r for g in groups for r in g)
+        assert seen == list(range(16))
+
+
+@pytest.mark.parametrize("tp,cp,pp,world", [(2, 1, 2, 8), (4, 1, 2, 8), (2, 2, 2, 8), (1, 1, 1, 1)])
+def test_grid_shapes(tp, cp, pp, world):
+    for rank in range(world):
+        g = ParallelGrid(
+            tensor_parallel_size=tp, pipeline_parallel_size=pp, context_parallel_size=cp,
+            world_size=world, rank=rank,
+        )
+        assert g.tp * g.cp * g.dp * g.pp == world
+        assert g.rank in g.ranks("tp")
+        assert g.rank in g.ranks("dp")
+        assert g.rank in g.ranks("pp")
+        assert len(g.ranks("tp")) == tp
+        assert len(g.ranks("pp")) == pp
+        # coords consistency
+        assert g.ranks("tp")[g.tp_rank] == g.rank
+        assert g.ranks("pp")[g.pp_rank] == g.rank
+
+
+def test_embedding_group_first_last_stage():
+    g = ParallelGrid(tensor_parallel_size=2, pipeline_parallel_size=2, world_size=8, rank=0)
+    # embd group for rank 0: first and last pp stage of its pipeline
+    assert g.ranks("embd") == [0, 4]
+
+
+def test_expert_groups():
+    # world 8: tp2 dp4; ep2 -> etp2, ep2, edp2
+    g = ParallelGrid(tensor_parallel_size=2, expert_parallel_size=2, world_size=8, rank=0)
+    assert g.ep == 2 and g.edp == 2
+    assert len(g.ranks("ep")) == 2
+    assert len(g.ranks("expert_dp")) == 2
+    seen = set(g.ranks("ep")) | set(g.ranks("expert_dp")) | set(g.ranks("etp"))
+    assert 0 in seen

@@ -1,0 +1,166 @@
+"""Tiny-GPT model-level tests (CPU)."""
+
+import math
+
+import torch
+
+from megatron_amd.config import TransformerConfig
+from megatron_amd.models.gpt import GPTModel
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.random import model_parallel_seed
+
+from tests.utils import assert_close, init_single, spawn_dist
+
+
+def _tiny_cfg(**kw):
+    base = dict(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        vocab_size=96, ffn_hidden_size=128, gradient_accumulation_fusion=False,
+    )
+    base.update(kw)
+    return TransformerConfig(**base)
+
+
+def test_forward_backward_and_init_loss():
+    init_single()
+    torch.manual_seed(5)
+    m = GPTModel(_tiny_cfg())
+    ids = torch.randint(0, 96, (2, 16))
+    labels = torch.randint(0, 96, (2, 16))
+    loss = m(ids, labels=labels)
+    assert loss.shape == (16, 2)
+    # random init -> loss close to ln(V)
+    assert abs(loss.mean().item() - math.log(96)) < 0.5
+    loss.mean().backward()
+    for n, p in m.named_parameters():
+        assert p.grad is not None, n
+
+
+def test_logits_path():
+    init_single()
+    m = GPTModel(_tiny_cfg())
+    ids = torch.randint(0, 96, (2, 16))
+    logits = m(ids)
+    assert logits.shape == (16, 2, 96)
+
+
+def test_recompute_full_matches_no_recompute():
+    init_single(seed=99)
+    torch.manual_seed(11)
+    cfg = _tiny_cfg()
+    m = GPTModel(cfg)
+    ids = torch.randint(0, 96, (2, 16))
+    labels = torch.randint(0, 96, (2, 16))
+    loss1 = m(ids, labels=labels).mean()
+    loss1.backward()
+    g1 = {n: p.grad.clone() for n, p in m.named_parameters()}
+    m.zero_grad()
+    m.decoder.config = cfg.replace(recompute_granularity="full")
+    loss2 = m(ids, labels=labels).mean()
+    loss2.backward()
+    assert_close(loss1, loss2, rtol=1e-6, atol=1e-6)
+    for n, p in m.named_parameters():
+        assert_close(g1[n], p.grad, rtol=1e-5, atol=1e-5, msg=n)
+
+
+def test_tied_embeddings():
+    init_single()
+    cfg = _tiny_cfg(untie_embeddings_and_output_weights=False)
+    m = GPTModel(cfg)
+    assert m.output_layer.weight is m.embedding.weight
+
+
+def _tp2_model_case(rank, world):
+    """TP=2 run must match TP=1 run given identically-sharded weights."""
+    G.initialize_model_parallel(tensor_parallel_size=world)
+    model_parallel_seed(1234)
+    cfg = _tiny_cfg(tensor_parallel_size=world)
+    torch.manual_seed(3)
+    # single-rank reference model built with tp=1 grid in a throwaway namespace
+    # is hard inside the same process; instead check TP model self-consistency:
+    # loss must be identical across the two TP ranks.
+    m = GPTModel(cfg)
+    ids = torch.randint(0, 96, (2, 16))
+    labels = torch.randint(0, 96, (2, 16))
+    loss = m(ids, labels=labels).mean()
+    import torch.distributed as dist
+
+    other = loss.detach().clone()
+    dist.broadcast(other, src=0)
+    assert torch.allclose(loss, other, rtol=1e-5, atol=1e-5)
+    loss.backward()
+
+
+def test_tp2_loss_identical_across_ranks():
+    spawn_dist(_tp2_model_case, 2)
+
+
+def _tp_vs_single_case(rank, world):
+    """TP=2 forward must numerically match the same full weights run densely."""
+    import torch.distributed as dist
+    import torch.nn.functional as F
+    from megatron_amd.ops import reference as ref
+
+    G.initialize_model_parallel(tensor_parallel_size=world)
+    model_parallel_seed(1234)
+    cfg = _tiny_cfg(tensor_parallel_size=world)
+    torch.manual_seed(3)
+    m = GPTModel(cfg)
+    ids = torch.randint(0, 96, (1, 8))
+
+    # gather full weights from the sharded model and run a dense equivalent
+    def gather_w(w, dim):
+        full = [torch.empty_like(w) for _ in range(world)]
+        dist.all_gather(full, w.detach().contiguous())
+        return torch.cat(full, dim=dim)
+
+    logits = m(ids)  # [s, b, V/tp]
+    full_logits = [torch.empty_like(logits) for _ in range(world)]
+    dist.all_gather(full_logits, logits.detach().contiguous())
+    full_logits = torch.cat(full_logits, dim=-1)
+
+    # dense recompute
+    emb_w = gather_w(m.embedding.weight, 0)
+    h = F.embedding(ids, emb_w).transpose(0, 1)
+    for layer in m.decoder.layers:
+        attn, mlp = layer.self_attention, layer.mlp
+        x = ref.rms_norm(h, layer.input_layernorm.weight.detach(), cfg.layernorm_epsilon)
+        qkv_w = gather_w(attn.linear_qkv.weight, 0)
+        qkv = torch.matmul(x, qkv_w.t())
+        s, b = qkv.shape[:2]
+        ng_full = cfg.num_query_groups
+        rep = cfg.num_attention_heads // ng_full
+        d = cfg.kv_channels
+        # per-rank [ng/tp, (rep+2)d] chunks concatenated: reorder to dense split
+        qkv = qkv.view(s, b, ng_full, (rep + 2) * d)
+        q, k, v = torch.split(qkv, [rep * d, d, d], dim=3)
+        q = q.reshape(s, b, ng_full * rep, d)
+        k = k.reshape(s, b, ng_full, d)
+        v = v.reshape(s, b, ng_full, d)
+        freqs = ref.rope_freqs(s, d, base=cfg.rotary_base)
+        q, k = ref.rope_apply(q, freqs), ref.rope_apply(k, freqs)
+        a = ref.attention(q, k, v, causal=True).reshape(s, b, -1)
+        proj_w = gather_w(attn.linear_proj.weight, 1)
+        h = h + torch.matmul(a, proj_w.t())
+        x = ref.rms_norm(h, layer.pre_mlp_layernorm.weight.detach(), cfg.layernorm_epsilon)
+        fc1_w = gather_w(mlp.linear_fc1.weight, 0)
+        up = torch.matmul(x, fc1_w.t())
+        # per-rank gated layout: [gate_shard; up_shard] per rank chunk
+        chunks = up.chunk(world, dim=-1)
+        gates = torch.cat([c.chunk(2, dim=-1)[0] for c in chunks], dim=-1)
+        ups = torch.cat([c.chunk(2, dim=-1)[1] for c in chunks], dim=-1)
+        act = F.silu(gates) * ups
+        fc2_w_parts = [torch.empty_like(mlp.linear_fc2.weight) for _ in range(world)]
+        dist.all_gather(fc2_w_parts, mlp.linear_fc2.weight.detach().contiguous())
+        y = sum(torch.matmul(act.chunk(world, -1)[r], fc2_w_parts[r].t()) for r in range(world))
+        h = h + y
+    h = ref.rms_norm(h, m.decoder.final_layernorm.weight.detach(), cfg.layernorm_epsilon)
+    out_w = gather_w(m.output_layer.weight, 0)
+    dense_logits = torch.matmul(h, out_w.t())
+    assert torch.allclose(full_logits, dense_logits, rtol=1e-3, atol=1e-3), (
+        (full_logits - dense_logits).abs().max()
+    )
+
+
+def test_tp2_matches_dense_recompute():
+    spawn_dist(_tp_vs_single_case, 2)

@@ -1,0 +1,66 @@
+"""Reference-op sanity (the same oracles the GPU numerics tests compare against)."""
+
+import math
+
+import torch
+import torch.nn.functional as F
+
+from megatron_amd.ops import reference as ref
+from tests.utils import assert_close
+
+
+def test_rms_norm_matches_manual():
+    x = torch.randn(4, 8, 64)
+    w = torch.randn(64)
+    out = ref.rms_norm(x, w, 1e-5)
+    manual = x / torch.sqrt(x.pow(2).mean(-1, keepdim=True) + 1e-5) * w
+    assert_close(out, manual, rtol=1e-5, atol=1e-5)
+
+
+def test_swiglu():
+    x = torch.randn(10, 32)
+    out = ref.swiglu(x)
+    x1, x2 = x.chunk(2, -1)
+    assert_close(out, F.silu(x1) * x2, rtol=1e-5, atol=1e-5)
+
+
+def test_rope_preserves_norm_and_zero_rotation():
+    s, b, h, d = 12, 2, 4, 16
+    x = torch.randn(s, b, h, d)
+    freqs = ref.rope_freqs(s, d, base=10000.0)
+    out = ref.rope_apply(x, freqs)
+    assert out.shape == x.shape
+    # norms preserved per (pair) rotation
+    assert_close(out.norm(dim=-1), x.norm(dim=-1), rtol=1e-4, atol=1e-4)
+    # position 0 has zero angle -> identity
+    assert_close(out[0], x[0], rtol=1e-6, atol=1e-6)
+
+
+def test_attention_vs_sdpa_causal():
+    torch.manual_seed(0)
+    s, b, hq, hkv, d = 16, 2, 4, 2, 8
+    q = torch.randn(s, b, hq, d)
+    k = torch.randn(s, b, hkv, d)
+    v = torch.randn(s, b, hkv, d)
+    out = ref.attention(q, k, v, causal=True)
+    qs = q.permute(1, 2, 0, 3)
+    ks = k.permute(1, 2, 0, 3).repeat_interleave(2, dim=1)
+    vs = v.permute(1, 2, 0, 3).repeat_interleave(2, dim=1)
+    sdpa = F.scaled_dot_product_attention(qs, ks, vs, is_causal=True)
+    assert_close(out, sdpa.permute(2, 0, 1, 3), rtol=1e-4, atol=1e-4)
+
+
+def test_attention_sliding_window():
+    torch.manual_seed(0)
+    s, b, h, d = 16, 1, 2, 8
+    q = torch.randn(s, b, h, d)
+    k = torch.randn(s, b, h, d)
+    v = torch.randn(s, b, h, d)
+    out = ref.attention(q, k, v, causal=True, window=4)
+    # manual
+    scale = 1 / math.sqrt(d)
+    scores = torch.einsum("sbhd,tbhd->bhst", q, k) * scale
+    mask = torch.ones(s, s, dtype=torch.bool).tril_() & torch.ones(s, s, dtype=torch.bool).triu_(-3)
+    scores = scores.masked_fill(~mask, float("-inf"))
+    manual = torch.einsum("bhst,tbhd->sbhd", torch.softmax(scores, -1), v)
+    assert_close(out, manual, rtol=1e-4, atol=1e-4)
