@@ -1,0 +1,308 @@
+"""Data-parallel gradient management: contiguous buffers, bucketed async
+reduce, backward overlap.
+
+Capability analog of reference megatron/core/distributed/
+distributed_data_parallel.py:22 + param_and_grad_buffer.py
+(_ParamAndGradBucket :92, _ParamAndGradBucketGroup :179, _ParamAndGradBuffer
+:1005), redesigned for one 8-GPU xGMI node:
+
+  * grads accumulate into ONE contiguous per-dtype ``grad_data`` buffer;
+    each param's ``main_grad`` is a view into it (288 GB HBM3E -> big
+    resident buffers, no per-step allocation).
+  * buffer split into ~bucket_size-element buckets, padded so each bucket
+    divides evenly by the DP size (reduce-scatter shards stay aligned).
+  * bucket-ready triggers an async all-reduce (plain DDP) or reduce-scatter
+    (distributed optimizer) on the DP group, overlapped with backward.
+    xGMI is fully-connected single-hop: RCCL reduce-scatter engages all 7
+    links, so fewer/larger buckets beat many small ones.
+  * readiness signaling: non-fused params use
+    ``register_post_accumulate_grad_hook``; params whose wgrad is fused
+    straight into main_grad (tensor_parallel/layers.py here) call the
+    ``_ddp_grad_ready_cb`` attribute from the autograd thread — no dummy
+    grad tensors (the reference's workaround, layers.py:691-720).
+"""
+
+from __future__ import annotations
+
+import math
+from contextlib import contextmanager
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from megatron_amd.config import DDPConfig
+from megatron_amd.parallel import grid as G
+
+
+def _pad_to(x: int, align: int) -> int:
+    return int(math.ceil(x / align) * align) if align > 1 else x
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter], start: int, end: int, index: int):
+        self.params = params
+        self.start = start
+        self.end = end
+        self.index = index
+        self.params_with_grad = set()
+        self.comm_handle = None
+        self.grad_view: Optional[torch.Tensor] = None   # [end-start] of grad_data
+        self.param_view: Optional[torch.Tensor] = None  # [end-start] of param_data
+        self.local_grad_shard: Optional[torch.Tensor] = None  # this rank's RS shard
+
+
+class ParamAndGradBuffer:
+    """Contiguous param/grad storage + bucket bookkeeping for one dtype group."""
+
+    def __init__(
+        self,
+        params: List[torch.nn.Parameter],
+        ddp_config: DDPConfig,
+        dp_group,
+        param_dtype: torch.dtype,
+        grad_dtype: torch.dtype,
+        device: torch.device,
+    ):
+        self.ddp_config = ddp_config
+        self.dp_group = dp_group
+        self.dp_size = dist.get_world_size(group=dp_group) if (dp_group is not None and dist.is_initialized()) else 1
+        self.grad_dtype = grad_dtype
+        self.param_dtype = param_dtype
+        self.device = device
+        self.is_last_microbatch = True
+
+        # --- assign offsets (params in reverse registration order: backward
+        #     finishes roughly in that order, so early buckets fill first) ---
+        align = 64 * self.dp_size  # keep shards 256B-aligned per rank
+        bucket_target = ddp_config.bucket_size or 40_000_000
+        self.buckets: List[_Bucket] = []
+        self.param_index: Dict[torch.nn.Parameter, tuple] = {}  # param -> (start, end, bucket_idx)
+
+        offset = 0
+        cur_params: List[torch.nn.Parameter] = []
+        cur_start = 0
+        for p in params:
+            n = p.data.nelement()
+            self.param_index[p] = (offset, offset + n, len(self.buckets))
+            cur_params.append(p)
+            offset += n
+            if offset - cur_start >= bucket_target:
+                offset = _pad_to(offset, align)
+                self.buckets.append(_Bucket(cur_params, cur_start, offset, len(self.buckets)))
+                cur_params, cur_start = [], offset
+        if cur_params:
+            offset = _pad_to(offset, align)
+            self.buckets.append(_Bucket(cur_params, cur_start, offset, len(self.buckets)))
+        self.total_elements = offset
+
+        # --- allocate storage ---
+        self.grad_data = torch.zeros(self.total_elements, dtype=grad_dtype, device=device)
+        self.param_data = None
+        if ddp_config.use_distributed_optimizer:
+            self.param_data = torch.empty(self.total_elements, dtype=param_dtype, device=device)
+
+        for p in params:
+            start, end, bidx = self.param_index[p]
+            p.main_grad = self.grad_data[start:end].view(p.data.shape)
+            if self.param_data is not None:
+                with torch.no_grad():
+                    self.param_data[start:end].view(p.data.shape).copy_(p.data)
+                    new_data = self.param_data[start:end].view(p.data.shape)
+                p.data = new_data
+
+        for b in self.buckets:
+            b.grad_view = self.grad_data[b.start : b.end]
+            if self.param_data is not None:
+                b.param_view = self.param_data[b.start : b.end]
+                shard = (b.end - b.start) // self.dp_size
+                rank = dist.get_rank(group=dp_group) if self.dp_size > 1 else 0
+                b.local_grad_shard = self.grad_data[b.start + rank * shard : b.start + (rank + 1) * shard]
+
+    # -- per-bucket collectives ---------------------------------------------
+
+    def _launch_grad_reduce(self, bucket: _Bucket, async_op: bool):
+        if self.dp_size == 1:
+            return
+        # AVG happens in-collective on RCCL; gloo has no AVG -> pre-scale + SUM
+        if self.ddp_config.average_in_collective and dist.get_backend(self.dp_group) == "nccl":
+            op = dist.ReduceOp.AVG
+        else:
+            bucket.grad_view.mul_(1.0 / self.dp_size)
+            op = dist.ReduceOp.SUM
+        if self.ddp_config.use_distributed_optimizer:
+            bucket.comm_handle = dist.reduce_scatter_tensor(
+                bucket.local_grad_shard, bucket.grad_view, op=op, group=self.dp_group, async_op=async_op
+            )
+        else:
+            bucket.comm_handle = dist.all_reduce(bucket.grad_view, op=op, group=self.dp_group, async_op=async_op)
+
+    def start_grad_sync(self):
+        for b in self.buckets:
+            if b.comm_handle is None:
+                self._launch_grad_reduce(b, async_op=True)
+
+    def finish_grad_sync(self):
+        self.start_grad_sync()
+        for b in self.buckets:
+            if b.comm_handle is not None and not isinstance(b.comm_handle, bool):
+                b.comm_handle.wait()
+            b.comm_handle = None
+            b.params_with_grad = set()
+
+    def start_param_sync(self, async_op: bool = False):
+        """all-gather updated params (dist-opt) bucket by bucket."""
+        if self.param_data is None or self.dp_size == 1:
+            return
+        self._param_handles = []
+        for b in self.buckets:
+            shard = (b.end - b.start) // self.dp_size
+            rank = dist.get_rank(group=self.dp_group)
+            local = b.param_view[rank * shard : (rank + 1) * shard]
+            h = dist.all_gather_into_tensor(b.param_view, local.contiguous(), group=self.dp_group, async_op=async_op)
+            if async_op:
+                self._param_handles.append(h)
+
+    def finish_param_sync(self):
+        for h in getattr(self, "_param_handles", []):
+            h.wait()
+        self._param_handles = []
+
+    def zero_grad(self):
+        self.grad_data.zero_()
+        for b in self.buckets:
+            b.comm_handle = None
+            b.params_with_grad = set()
+
+    # -- readiness ------------------------------------------------------------
+
+    def mark_ready(self, param: torch.nn.Parameter):
+        if self.is_last_microbatch and self.ddp_config.overlap_grad_reduce:
+            _, _, bidx = self.param_index[param]
+            b = self.buckets[bidx]
+            b.params_with_grad.add(param)
+            if len(b.params_with_grad) == len(b.params) and b.comm_handle is None:
+                self._launch_grad_reduce(b, async_op=True)
+
+
+class DistributedDataParallel(nn.Module):
+    """Wraps one model chunk (reference distributed_data_parallel.py:22)."""
+
+    def __init__(self, config, ddp_config: DDPConfig, module: nn.Module):
+        super().__init__()
+        self.module = module
+        self.config = config
+        self.ddp_config = ddp_config
+        self.dp_group = (
+            G.get_data_parallel_group(with_context_parallel=True) if G.grid_initialized() else None
+        )
+
+        dense_params, expert_params = [], []
+        for p in self.module.parameters():
+            if not p.requires_grad:
+                continue
+            (expert_params if getattr(p, "is_expert_parallel", False) else dense_params).append(p)
+
+        device = dense_params[0].device if dense_params else torch.device("cpu")
+        grad_dtype = torch.float32 if ddp_config.grad_reduce_in_fp32 else (
+            dense_params[0].dtype if dense_params else torch.float32
+        )
+        self.buffers: List[ParamAndGradBuffer] = []
+        self.param_to_buffer: Dict[torch.nn.Parameter, ParamAndGradBuffer] = {}
+        if dense_params:
+            buf = ParamAndGradBuffer(
+                list(reversed(dense_params)), ddp_config, self.dp_group,
+                dense_params[0].dtype, grad_dtype, device,
+            )
+            self.buffers.append(buf)
+            for p in buf.param_index:
+                self.param_to_buffer[p] = buf
+        if expert_params:
+            edp_group = G.get_grid().group("expert_dp") if G.grid_initialized() else None
+            ebuf = ParamAndGradBuffer(
+                list(reversed(expert_params)), ddp_config, edp_group,
+                expert_params[0].dtype, grad_dtype, device,
+            )
+            self.buffers.append(ebuf)
+            for p in ebuf.param_index:
+                self.param_to_buffer[p] = ebuf
+
+        self._grad_hook_handles = []
+        for p in self.param_to_buffer:
+            p.grad_added_to_main_grad = False
+            p._ddp_grad_ready_cb = self._make_ready_cb(p)  # fused wgrad path
+            h = p.register_post_accumulate_grad_hook(self._make_post_acc_hook(p))
+            self._grad_hook_handles.append(h)
+
+    def _make_post_acc_hook(self, param):
+        buffer = self.param_to_buffer[param]
+
+        def hook(p):
+            if p.grad is not None:
+                param.main_grad.add_(p.grad.to(param.main_grad.dtype))
+                p.grad = None
+            buffer.mark_ready(p)
+
+        return hook
+
+    def _make_ready_cb(self, param):
+        buffer = self.param_to_buffer[param]
+
+        def cb():
+            buffer.mark_ready(param)
+
+        return cb
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    @contextmanager
+    def no_sync(self):
+        for buf in self.buffers:
+            buf.is_last_microbatch = False
+        try:
+            yield
+        finally:
+            for buf in self.buffers:
+                buf.is_last_microbatch = True
+
+    def start_grad_sync(self):
+        for buf in self.buffers:
+            buf.start_grad_sync()
+
+    def finish_grad_sync(self):
+        for buf in self.buffers:
+            buf.finish_grad_sync()
+
+    def start_param_sync(self, async_op: bool = False, force_sync: bool = False):
+        for buf in self.buffers:
+            buf.start_param_sync(async_op=async_op)
+
+    def finish_param_sync(self):
+        for buf in self.buffers:
+            buf.finish_param_sync()
+
+    def zero_grad_buffer(self):
+        for p in self.param_to_buffer:
+            p.grad_added_to_main_grad = False
+        for buf in self.buffers:
+            buf.zero_grad()
+
+    def broadcast_params(self):
+        """Sync initial params across DP (rank0 -> all)."""
+        if self.dp_group is None or dist.get_world_size(group=self.dp_group) == 1:
+            return
+        src = dist.get_process_group_ranks(self.dp_group)[0]
+        for buf in self.buffers:
+            if buf.param_data is not None:
+                dist.broadcast(buf.param_data, src=src, group=self.dp_group)
+            else:
+                for p in buf.param_index:
+                    dist.broadcast(p.data, src=src, group=self.dp_group)
+
+    def state_dict(self, *args, **kwargs):
+        return self.module.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, *args, **kwargs):
+        return self.module.load_state_dict(*args, **kwargs)

@@ -1,0 +1,68 @@
+"""End-of-step cross-group gradient reductions.
+
+Capability analog of reference megatron/core/distributed/finalize_model_grads.py
+(:164 word-embedding grads over the embd group, :416 non-TP-replicated
+(sequence-parallel-duplicated) grads over TP, :560 entry point).
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+import torch.distributed as dist
+
+from megatron_amd.parallel import grid as G
+
+
+def _grad_of(param):
+    return getattr(param, "main_grad", None) if getattr(param, "main_grad", None) is not None else param.grad
+
+
+def _allreduce_layernorm_grads(models: List[torch.nn.Module], config):
+    """SP shards the sequence over TP: norm/bias grads (computed from a seq
+    slice) must be summed over the TP group."""
+    if not (config.sequence_parallel and config.tensor_parallel_size > 1):
+        return
+    grid = G.get_grid()
+    group = grid.group("tp")
+    grads = []
+    for model in models:
+        for p in model.parameters():
+            if getattr(p, "sequence_parallel_dup", False) and p.requires_grad:
+                g = _grad_of(p)
+                if g is not None:
+                    grads.append(g.data)
+    if grads and group is not None:
+        flat = torch._utils._flatten_dense_tensors(grads)
+        dist.all_reduce(flat, group=group)
+        for g, synced in zip(grads, torch._utils._unflatten_dense_tensors(flat, grads)):
+            g.copy_(synced)
+
+
+def _allreduce_word_embedding_grads(models: List[torch.nn.Module], config):
+    """Tied input/output embeddings live on first and last PP stage: their
+    grads are summed over the embd group (reference finalize_model_grads:164)."""
+    grid = G.get_grid()
+    if grid.pp == 1:
+        return
+    if not (grid.is_pipeline_first_stage(ignore_virtual=True) or grid.is_pipeline_last_stage(ignore_virtual=True)):
+        return
+    for model in models:
+        core = model.module if hasattr(model, "module") else model
+        if getattr(core, "share_embeddings_and_output_weights", False):
+            w = core.shared_embedding_or_output_weight()
+            if w is not None and w.requires_grad:
+                g = _grad_of(w)
+                if g is not None and grid.group("embd") is not None and len(grid.ranks("embd")) > 1:
+                    dist.all_reduce(g.data, group=grid.group("embd"))
+
+
+def finalize_model_grads(models: List[torch.nn.Module], config=None):
+    if config is None:
+        core = models[0].module if hasattr(models[0], "module") else models[0]
+        config = core.config
+    if not G.grid_initialized():
+        return
+    _allreduce_word_embedding_grads(models, config)
+    _allreduce_layernorm_grads(models, config)

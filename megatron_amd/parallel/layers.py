@@ -128,8 +128,10 @@ class _ParallelLinearFn(torch.autograd.Function):
         if ctx.grad_accum_fusion and hasattr(weight, "main_grad"):
             _wgrad_accum(weight.main_grad, go2, in2)
             grad_weight = None
-            if not getattr(weight, "grad_added_to_main_grad", False):
-                weight.grad_added_to_main_grad = True
+            weight.grad_added_to_main_grad = True
+            cb = getattr(weight, "_ddp_grad_ready_cb", None)
+            if cb is not None:
+                cb()  # bucket-readiness signal for overlapped grad reduce
         else:
             grad_weight = torch.matmul(go2.t(), in2)
         grad_bias = go2.sum(dim=0) if ctx.use_bias else None

@@ -1,0 +1,100 @@
+"""Training step + setup helpers.
+
+Capability analog of reference megatron/training/training.py (train_step
+:2372, setup_model_and_optimizer :2042, get_model :1732) — lean version;
+the full pretrain() CLI app lives in pretrain_gpt.py at the repo root.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
+from megatron_amd.distributed import DistributedDataParallel, finalize_model_grads
+from megatron_amd.optimizer import ChainedOptimizer, get_optimizer
+from megatron_amd.parallel import grid as G
+from megatron_amd.pipeline.schedules import get_forward_backward_func
+
+
+def setup_model_and_optimizer(
+    model_provider: Callable[..., torch.nn.Module],
+    config: TransformerConfig,
+    opt_config: OptimizerConfig,
+    ddp_config: Optional[DDPConfig] = None,
+    device: Optional[torch.device] = None,
+):
+    """Build (virtual-chunked) model list, wrap in DDP, build optimizer."""
+    if ddp_config is None:
+        ddp_config = DDPConfig(
+            use_distributed_optimizer=opt_config.use_distributed_optimizer,
+            grad_reduce_in_fp32=True,
+        )
+    vpp = config.virtual_pipeline_parallel_size
+    chunks = []
+    n_chunks = vpp if (vpp is not None and G.get_pipeline_model_parallel_world_size() > 1) else 1
+    grid = G.get_grid() if G.grid_initialized() else None
+    for vp in range(n_chunks):
+        if grid is not None:
+            grid.set_vpp_rank(vp if vpp else None)
+        pre = G.grid_initialized() and grid.is_pipeline_first_stage() or not G.grid_initialized()
+        post = G.grid_initialized() and grid.is_pipeline_last_stage() or not G.grid_initialized()
+        m = model_provider(config, pre_process=pre, post_process=post, vp_stage=vp if vpp else None)
+        if device is not None:
+            m = m.to(device)
+        chunks.append(m)
+    if grid is not None:
+        grid.set_vpp_rank(0 if vpp else None)
+
+    ddp_chunks = [DistributedDataParallel(config, ddp_config, m) for m in chunks]
+    for c in ddp_chunks:
+        c.broadcast_params()
+    optimizer = get_optimizer(opt_config, ddp_chunks)
+
+    # wire the schedule hooks (reference three-hook contract)
+    config.finalize_model_grads_func = finalize_model_grads
+    return ddp_chunks, optimizer
+
+
+def train_step(
+    forward_step_func,
+    data_iterator,
+    model_chunks: List,
+    optimizer: ChainedOptimizer,
+    config: TransformerConfig,
+    num_microbatches: int,
+    seq_length: int,
+    micro_batch_size: int,
+):
+    """One optimizer step (reference training.py:2372)."""
+    for chunk in model_chunks:
+        chunk.zero_grad_buffer()
+    optimizer.zero_grad() if hasattr(optimizer, "zero_grad") else None
+
+    fb_func = get_forward_backward_func()
+    losses, num_tokens = fb_func(
+        forward_step_func=forward_step_func,
+        data_iterator=data_iterator if isinstance(data_iterator, list) else [data_iterator],
+        model=model_chunks,
+        num_microbatches=num_microbatches,
+        seq_length=seq_length,
+        micro_batch_size=micro_batch_size,
+    )
+
+    ok, grad_norm, _ = optimizer.step()
+
+    # aggregate loss over DP(xCP) and microbatches: token-weighted average
+    loss_sum = torch.zeros((), dtype=torch.float32, device=num_tokens.device)
+    for m in losses:
+        if "loss_sum" in m:
+            loss_sum += m["loss_sum"]
+    total_tokens = num_tokens.clone()
+    if G.grid_initialized() and dist.is_initialized():
+        group = G.get_grid().group("dp_cp")
+        if group is not None and dist.get_world_size(group=group) > 1:
+            dist.all_reduce(loss_sum, group=group)
+            dist.all_reduce(total_tokens, group=group)
+    mean_loss = loss_sum / total_tokens.clamp(min=1)
+    return {"lm_loss": mean_loss.item(), "grad_norm": None if grad_norm is None else float(grad_norm), "skipped": not ok}
