@@ -1,0 +1,206 @@
+"""Flagship training benchmark (driver contract).
+
+  python bench.py --gpus N --steps K --warmup W [--model llama3-8b]
+
+Measures the BASELINE.json metric: whole-node tokens/s (+ MFU) for
+Llama-3-8B / 70B on synthetic data of the benchmark shape with random-init
+weights, bf16, seq 4096.  For N>1 the driver launches this under
+torch.distributed.run with one rank per GPU over RCCL; per-GPU work is fixed
+(weak scaling).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
+from megatron_amd.datasets.mock import MockGPTDataIterator
+from megatron_amd.models.gpt import GPTModel
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.random import model_parallel_seed
+from megatron_amd.training.flops import MI355X_BF16_DENSE_PEAK_TFLOPS, num_floating_point_operations
+from megatron_amd.training.training import setup_model_and_optimizer, train_step
+
+MODELS = {
+    # llama-3 8B (BASELINE.json "Llama-3 8B DP=8 bf16")
+    "llama3-8b": dict(
+        num_layers=32, hidden_size=4096, num_attention_heads=32, num_query_groups=8,
+        ffn_hidden_size=14336, vocab_size=128256, rotary_base=500000.0,
+    ),
+    # llama-3 70B (BASELINE.json "Llama-3 70B TP=8 over xGMI")
+    "llama3-70b": dict(
+        num_layers=80, hidden_size=8192, num_attention_heads=64, num_query_groups=8,
+        ffn_hidden_size=28672, vocab_size=128256, rotary_base=500000.0,
+    ),
+    # tiny shape for CI plumbing
+    "tiny": dict(
+        num_layers=2, hidden_size=256, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=512, vocab_size=1024,
+    ),
+}
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--seq-len", type=int, default=4096)
+    p.add_argument("--micro-batch-size", type=int, default=1)
+    p.add_argument("--grad-accum", type=int, default=8, help="microbatches per step per DP rank")
+    p.add_argument("--tp", type=int, default=None, help="tensor parallel size (default: model-dependent)")
+    p.add_argument("--pp", type=int, default=1)
+    p.add_argument("--vpp", type=int, default=None)
+    p.add_argument("--recompute", action="store_true")
+    p.add_argument("--no-dist-opt", action="store_true")
+    p.add_argument("--seed", type=int, default=1234)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    use_dist = world > 1 or "RANK" in os.environ
+
+    if torch.cuda.is_available():
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    if use_dist and not torch.distributed.is_initialized():
+        G.init_distributed()
+
+    tp = args.tp
+    if tp is None:
+        tp = min(world, 8) if args.model == "llama3-70b" else 1
+    if args.model == "llama3-70b" and world < 2 and args.tp is None:
+        tp = 1  # single-GPU measurement of the 70B shape is impossible; caller sets --tp
+    sp = tp > 1
+
+    if use_dist:
+        G.initialize_model_parallel(tensor_parallel_size=tp, pipeline_parallel_size=args.pp,
+                                    virtual_pipeline_parallel_size=args.vpp)
+    else:
+        G.initialize_model_parallel(tensor_parallel_size=1, world_size=1, rank=0)
+    model_parallel_seed(args.seed)
+    grid = G.get_grid()
+    dp = grid.dp
+
+    mdl = dict(MODELS[args.model])
+    # 70B at 1 GPU cannot fit; scale layer count for sub-node smoke unless full node
+    cfg = TransformerConfig(
+        **mdl,
+        bf16=True if device.type == "cuda" else False,
+        max_position_embeddings=args.seq_len,
+        tensor_parallel_size=tp,
+        pipeline_parallel_size=args.pp,
+        virtual_pipeline_parallel_size=args.vpp,
+        sequence_parallel=sp,
+        recompute_granularity="full" if args.recompute else None,
+        gradient_accumulation_fusion=device.type == "cuda",
+    )
+    opt_cfg = OptimizerConfig(
+        lr=3e-4, weight_decay=0.1, clip_grad=1.0, bf16=cfg.bf16,
+        use_distributed_optimizer=not args.no_dist_opt,
+    )
+    ddp_cfg = DDPConfig(
+        grad_reduce_in_fp32=True, overlap_grad_reduce=True,
+        use_distributed_optimizer=not args.no_dist_opt, bucket_size=40_000_000,
+    )
+
+    def provider(config, pre_process=True, post_process=True, vp_stage=None):
+        return GPTModel(config, pre_process=pre_process, post_process=post_process, vp_stage=vp_stage)
+
+    chunks, optimizer = setup_model_and_optimizer(provider, cfg, opt_cfg, ddp_cfg, device=device)
+
+    data = MockGPTDataIterator(args.micro_batch_size, args.seq_len, cfg.vocab_size,
+                               seed=args.seed, device=str(device), dp_rank=grid.rank_in("dp_cp"))
+    data_it = iter(data)
+
+    def forward_step(it, model):
+        batch = next(data_it)
+
+        def loss_func(loss_sb):
+            s = loss_sb.sum()
+            return s, torch.tensor(loss_sb.numel(), device=loss_sb.device), {"loss_sum": s.detach()}
+
+        out = model(batch["tokens"], labels=batch["labels"])
+        return out, loss_func
+
+    def one_step():
+        return train_step(forward_step, None, chunks, optimizer, cfg,
+                          args.grad_accum, args.seq_len, args.micro_batch_size)
+
+    def barrier_sync():
+        if torch.distributed.is_initialized():
+            torch.distributed.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        one_step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    last = None
+    for _ in range(args.steps):
+        last = one_step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if torch.distributed.is_initialized():
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if device.type == "cuda":
+            t = t.to(device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    global_batch = args.micro_batch_size * args.grad_accum * dp
+    tokens_per_step = global_batch * args.seq_len
+    tokens_per_s = tokens_per_step / (elapsed / args.steps)
+    n_gpus = world if device.type == "cuda" else world
+    flops_per_step = num_floating_point_operations(cfg, global_batch, args.seq_len)
+    tflops_per_gpu = flops_per_step / (elapsed / args.steps) / max(n_gpus, 1) / 1e12
+    mfu = tflops_per_gpu / MI355X_BF16_DENSE_PEAK_TFLOPS
+
+    if rank == 0:
+        par = f"tp{tp}" + (f"pp{args.pp}" if args.pp > 1 else "") + f"dp{dp}"
+        print(json.dumps({
+            "metric": "tokens/s",
+            "value": round(tokens_per_s, 1),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if cfg.bf16 else "fp32",
+            "data": "synthetic",
+            "tflops_per_gpu": round(tflops_per_gpu, 1),
+            "mfu": round(mfu, 4),
+            "loss": None if last is None else round(last["lm_loss"], 4),
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": args.seq_len,
+                "parallelism": par,
+                "micro_batch_size": args.micro_batch_size,
+                "distributed_optimizer": not args.no_dist_opt,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

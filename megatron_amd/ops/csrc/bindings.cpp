@@ -1,0 +1,34 @@
+// Python bindings for the megatron_amd CDNA4 HIP kernels.
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w, torch::Tensor rstd);
+torch::Tensor swiglu_fwd(torch::Tensor x);
+torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor x);
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cost, torch::Tensor sint);
+torch::Tensor rope_bwd(torch::Tensor dy, torch::Tensor cost, torch::Tensor sint);
+void multi_tensor_adamw(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
+                        std::vector<torch::Tensor> exp_avgs, std::vector<torch::Tensor> exp_avg_sqs,
+                        std::vector<torch::Tensor> model_params,
+                        double lr, double beta1, double beta2, double eps, double wd, long step);
+torch::Tensor multi_tensor_l2norm(std::vector<torch::Tensor> tensors);
+void wgrad_gemm_accum(torch::Tensor main_grad, torch::Tensor grad_output, torch::Tensor input);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    bool causal, double scale, long window);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor out, torch::Tensor lse,
+                                    bool causal, double scale, long window);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("rope_bwd", &rope_bwd);
+  m.def("multi_tensor_adamw", &multi_tensor_adamw);
+  m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
+  m.def("wgrad_gemm_accum", &wgrad_gemm_accum);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+}

@@ -1,0 +1,41 @@
+"""In-tree build of the megatron_amd CDNA4 HIP extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+Builds megatron_amd/ops/_hip_ops.so next to the package sources so the
+gpurun snapshot carries the binary (no site-packages install, no JIT cache).
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+SRC = [
+    "megatron_amd/ops/csrc/bindings.cpp",
+    "megatron_amd/ops/csrc/rmsnorm.hip",
+    "megatron_amd/ops/csrc/swiglu.hip",
+    "megatron_amd/ops/csrc/rope.hip",
+    "megatron_amd/ops/csrc/adamw.hip",
+    "megatron_amd/ops/csrc/wgrad.hip",
+    "megatron_amd/ops/csrc/attention_fwd.hip",
+    "megatron_amd/ops/csrc/attention_bwd.hip",
+]
+
+setup(
+    name="megatron_amd_hip_ops",
+    ext_modules=[
+        CUDAExtension(
+            name="megatron_amd.ops._hip_ops",
+            sources=SRC,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+            libraries=["rocblas"],
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)
