@@ -115,10 +115,11 @@ def attention(
         vf = vf.repeat_interleave(rep, dim=1)
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b, hq, s, s]
     sk = k.shape[0]
+    off = sk - s  # decode: query row i attends kv <= i + off
     if causal:
-        mask = torch.ones(s, sk, dtype=torch.bool, device=q.device).tril_()
+        mask = torch.ones(s, sk, dtype=torch.bool, device=q.device).tril_(off)
         if window is not None:
-            mask &= torch.ones(s, sk, dtype=torch.bool, device=q.device).triu_(-window + 1)
+            mask &= torch.ones(s, sk, dtype=torch.bool, device=q.device).triu_(off - window + 1)
         scores = scores.masked_fill(~mask, float("-inf"))
     probs = torch.softmax(scores, dim=-1)
     out = torch.matmul(probs, vf)  # [b, hq, s, d]

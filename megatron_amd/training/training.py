@@ -41,9 +41,12 @@ def setup_model_and_optimizer(
             grid.set_vpp_rank(vp if vpp else None)
         pre = G.grid_initialized() and grid.is_pipeline_first_stage() or not G.grid_initialized()
         post = G.grid_initialized() and grid.is_pipeline_last_stage() or not G.grid_initialized()
-        m = model_provider(config, pre_process=pre, post_process=post, vp_stage=vp if vpp else None)
         if device is not None:
+            with torch.device(device):  # init weights directly on the GPU
+                m = model_provider(config, pre_process=pre, post_process=post, vp_stage=vp if vpp else None)
             m = m.to(device)
+        else:
+            m = model_provider(config, pre_process=pre, post_process=post, vp_stage=vp if vpp else None)
         chunks.append(m)
     if grid is not None:
         grid.set_vpp_rank(0 if vpp else None)
