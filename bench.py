@@ -190,6 +190,7 @@ def main():
             "data": "synthetic",
             "tflops_per_gpu": round(tflops_per_gpu, 1),
             "mfu": round(mfu, 4),
+            "mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 1) if device.type == "cuda" else None,
             "loss": None if last is None else round(last["lm_loss"], 4),
             "config": {
                 "model": args.model,
