@@ -115,6 +115,14 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         }
         *reinterpret_cast<short8*>(&K_lds[r][c8 * 8]) = kv8;
         *reinterpret_cast<short8*>(&V_lds[r][c8 * 8]) = vv8;
+      }
+      // transposed K: r-fast mapping avoids the 8-row-stride bank aliasing
+      for (int idx = threadIdx.x; idx < G; idx += 256) {
+        int r = idx & (KVBLK - 1), c8 = idx / KVBLK;
+        int krow = k0 + r;
+        short8 kv8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (krow < skv)
+          kv8 = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + c8 * 8);
 #pragma unroll
         for (int j = 0; j < 8; ++j) Kt_lds[c8 * 8 + j][r] = kv8[j];
       }
@@ -152,6 +160,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     __syncthreads();
 
     // dq += dS K   (A = dS row-major from dS_lds, B = K^T from Kt_lds)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < KVBLK / 32; ++ks) {
       short8 at = *reinterpret_cast<const short8*>(&dS_lds[wid][cl][ks * 32 + rg * 8]);
@@ -162,6 +171,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         dq_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bK, dq_acc[n], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -179,7 +189,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 // dk/dv kernel
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(512) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ drow,
     bf16* __restrict__ dk, bf16* __restrict__ dv,
@@ -191,10 +201,10 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
 
   __shared__ short BufB[QBLK][RPAD];   // row-major Q (phase 1) then dO (phase 3)
   __shared__ short BufA[D][TPAD];      // transposed dO (phase 2) then Q^T (phase 4)
-  __shared__ short P_lds[4][16][TPAD]; // per-wave PT / dST round-trips
+  __shared__ short P_lds[8][16][TPAD]; // per-wave PT / dST round-trips
 
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-  const int k0 = blockIdx.x * 64;
+  const int k0 = blockIdx.x * 128;  // 8 waves x 16 kv rows
   const int batch = blockIdx.y / hkv, kv_head = blockIdx.y % hkv;
   const int rep = hq / hkv;
   const long q_srow = (long)b * hq * D, k_srow = (long)b * hkv * D;
@@ -231,7 +241,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   int tq_end = (sq - 1) / QBLK;
   if (window > 0) {
     // kv row k attends from q rows >= k - off ... < k - off + window
-    tq_end = min(tq_end, (k0 + 63 - off + window - 1) / QBLK);
+    tq_end = min(tq_end, (k0 + 127 - off + window - 1) / QBLK);
   }
 
   for (int hg = 0; hg < rep; ++hg) {
@@ -249,12 +259,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
         for (int idx = threadIdx.x; idx < G; idx += 256) {
           int r = idx / (D / 8), c8 = idx % (D / 8);
           int qrow = qt0 + r;
-          short8 q8 = short8{0, 0, 0, 0, 0, 0, 0, 0}, do8 = q8;
-          if (qrow < sq) {
+          short8 q8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+          if (qrow < sq)
             q8 = *reinterpret_cast<const short8*>(qbase + (long)qrow * q_srow + c8 * 8);
-            do8 = *reinterpret_cast<const short8*>(dobase + (long)qrow * q_srow + c8 * 8);
-          }
           *reinterpret_cast<short8*>(&BufB[r][c8 * 8]) = q8;
+        }
+        for (int idx = threadIdx.x; idx < G; idx += 512) {
+          int r = idx & (QBLK - 1), c8 = idx / QBLK;
+          int qrow = qt0 + r;
+          short8 do8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+          if (qrow < sq)
+            do8 = *reinterpret_cast<const short8*>(dobase + (long)qrow * q_srow + c8 * 8);
 #pragma unroll
           for (int j = 0; j < 8; ++j) BufA[c8 * 8 + j][r] = do8[j];
         }
@@ -288,6 +303,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
       __syncthreads();
 
       // dV[kv, d] += PT[kv, q] x dO[q, d]  (A from P_lds, B from BufA=dOt)
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks = 0; ks < QBLK / 32; ++ks) {
         short8 at = *reinterpret_cast<const short8*>(&P_lds[wid][cl][ks * 32 + rg * 8]);
@@ -298,6 +314,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
           dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, dv_acc[n], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
       __syncthreads();
 
       // ---- phase 3: restage BufB <- dO row-major ; BufA <- Q^T ----
@@ -306,12 +323,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
         for (int idx = threadIdx.x; idx < G; idx += 256) {
           int r = idx / (D / 8), c8 = idx % (D / 8);
           int qrow = qt0 + r;
-          short8 q8 = short8{0, 0, 0, 0, 0, 0, 0, 0}, do8 = q8;
-          if (qrow < sq) {
-            q8 = *reinterpret_cast<const short8*>(qbase + (long)qrow * q_srow + c8 * 8);
+          short8 do8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+          if (qrow < sq)
             do8 = *reinterpret_cast<const short8*>(dobase + (long)qrow * q_srow + c8 * 8);
-          }
           *reinterpret_cast<short8*>(&BufB[r][c8 * 8]) = do8;
+        }
+        for (int idx = threadIdx.x; idx < G; idx += 512) {
+          int r = idx & (QBLK - 1), c8 = idx / QBLK;
+          int qrow = qt0 + r;
+          short8 q8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+          if (qrow < sq)
+            q8 = *reinterpret_cast<const short8*>(qbase + (long)qrow * q_srow + c8 * 8);
 #pragma unroll
           for (int j = 0; j < 8; ++j) BufA[c8 * 8 + j][r] = q8[j];
         }
@@ -337,6 +359,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
       __syncthreads();
 
       // dK[kv, d] += dST[kv, q] x Q[q, d]  (A from P_lds, B from BufA=Qt)
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks = 0; ks < QBLK / 32; ++ks) {
         short8 at = *reinterpret_cast<const short8*>(&P_lds[wid][cl][ks * 32 + rg * 8]);
@@ -347,6 +370,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
           dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bq, dk_acc[n], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
       __syncthreads();
     }
   }
@@ -388,13 +412,13 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                      drow.data_ptr<float>(), rows, b * hq, sq, d8);
 
   dim3 grid_dq((sq + 63) / 64, b * hq);
-  dim3 grid_dkv((skv + 63) / 64, b * hkv);
+  dim3 grid_dkv((skv + 127) / 128, b * hkv);
   if (d == 128) {
     hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid_dq, dim3(256), 0, stream,
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
                        (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
                        (bf16*)dq.data_ptr(), sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), grid_dkv, dim3(256), 0, stream,
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), grid_dkv, dim3(512), 0, stream,
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
                        (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
                        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
@@ -404,7 +428,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
                        (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
                        (bf16*)dq.data_ptr(), sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64>), grid_dkv, dim3(256), 0, stream,
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64>), grid_dkv, dim3(512), 0, stream,
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
                        (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
                        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),

@@ -29,7 +29,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     bf16* __restrict__ o, float* __restrict__ lse,
     int sq, int skv, int b, int hq, int hkv, float scale,
     int causal, int window) {
-  constexpr int QBLK = 64, KVBLK = 64;
+  constexpr int QBLK = 64, KVBLK = 64;  // 4 waves x 16 q rows
   constexpr int KPAD = D + 8;   // K_lds row stride (bf16)
   constexpr int VPAD = KVBLK + 8;
   constexpr int ND = D / 16;    // d-tiles (4 or 8)
@@ -85,22 +85,27 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   for (int t = t_start; t <= t_end; ++t) {
     const int k0 = t * KVBLK;
-    // ---- stage K (row-major, padded) and V (transposed) ----
+    // ---- stage K (row-major, coalesced) and V (transposed, r-fast) ----
     {
-      // 64 rows x D cols, short8 granules: (KVBLK * D / 8) loads over 256 threads
       constexpr int G = KVBLK * D / 8;
       for (int idx = threadIdx.x; idx < G; idx += 256) {
+        // K: coalesced loads, row-major vector stores
         int r = idx / (D / 8), c8 = idx % (D / 8);
         int krow = k0 + r;
-        short8 kv8, vv8;
-        if (krow < skv) {
+        short8 kv8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (krow < skv)
           kv8 = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + c8 * 8);
-          vv8 = *reinterpret_cast<const short8*>(vbase + (long)krow * k_srow + c8 * 8);
-        } else {
-          kv8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
-          vv8 = kv8;
-        }
         *reinterpret_cast<short8*>(&K_lds[r][c8 * 8]) = kv8;
+      }
+      for (int idx = threadIdx.x; idx < G; idx += 256) {
+        // V: r-fast mapping -> transposed scalar stores hit ~all banks
+        // (row-major mapping would put all 16 same-j lanes in one bank:
+        //  8-row x 16B-aligned stride aliases mod 32 banks)
+        int r = idx & (KVBLK - 1), c8 = idx / KVBLK;
+        int krow = k0 + r;
+        short8 vv8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (krow < skv)
+          vv8 = *reinterpret_cast<const short8*>(vbase + (long)krow * k_srow + c8 * 8);
 #pragma unroll
         for (int j = 0; j < 8; ++j) Vt_lds[c8 * 8 + j][r] = vv8[j];
       }
@@ -109,6 +114,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
     // ---- S = scale * Q K^T  (4 col-tiles of 16) ----
     f32x4 s[KVBLK / 16];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int n = 0; n < KVBLK / 16; ++n) {
       s[n] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -118,11 +124,28 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[kd], bk, s[n], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + online softmax ----
+    // interior tiles need no masking (wave rows all >= row0):
+    const int off = skv - sq;
+    bool tile_full = (k0 + KVBLK <= skv) && (row0 + 15 < sq);
+    if (causal) tile_full &= (k0 + KVBLK - 1 <= row0 + off);
+    if (window > 0) tile_full &= (k0 >= row0 + 15 + off - window + 1);
     float pmax[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) pmax[r] = -1e30f;
+    if (tile_full) {
+#pragma unroll
+      for (int n = 0; n < KVBLK / 16; ++n) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float val = s[n][r] * scale;
+          s[n][r] = val;
+          pmax[r] = fmaxf(pmax[r], val);
+        }
+      }
+    } else {
 #pragma unroll
     for (int n = 0; n < KVBLK / 16; ++n) {
       int col = k0 + n * 16 + cl;
@@ -130,12 +153,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int r = 0; r < 4; ++r) {
         int row = row0 + rg * 4 + r;
         bool ok = (col < skv) && (row < sq);
-        if (causal) ok &= (col <= row + (skv - sq));
-        if (window > 0) ok &= (col > row + (skv - sq) - window);
+        if (causal) ok &= (col <= row + off);
+        if (window > 0) ok &= (col > row + off - window);
         float val = ok ? s[n][r] * scale : -1e30f;
         s[n][r] = val;
         pmax[r] = fmaxf(pmax[r], val);
       }
+    }
     }
     // row-reduce max over the 16 lanes of each group
 #pragma unroll
@@ -151,6 +175,17 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       m_run[r] = m_new;
       psum[r] = 0.f;
     }
+    if (tile_full) {
+#pragma unroll
+      for (int n = 0; n < KVBLK / 16; ++n) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = exp2f((s[n][r] - m_run[r]) * LOG2E);
+          s[n][r] = p;
+          psum[r] += p;
+        }
+      }
+    } else {
 #pragma unroll
     for (int n = 0; n < KVBLK / 16; ++n) {
 #pragma unroll
@@ -160,6 +195,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         s[n][r] = p;
         psum[r] += p;
       }
+    }
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -176,9 +212,12 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) P_lds[wid][rg * 4 + r][n * 16 + cl] = f2sbf(s[n][r]);
     }
-    __syncthreads();
+    // P_lds is wave-private: same-wave LDS RAW needs only the compiler's
+    // lgkmcnt (same-array dependency), not a block-wide barrier.
+    __builtin_amdgcn_s_waitcnt(0 /* vmcnt0 lgkmcnt0 expcnt0 */);
 
     // ---- O += P V ----
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < KVBLK / 32; ++ks) {
       short8 ptmp = *reinterpret_cast<const short8*>(&P_lds[wid][cl][ks * 32 + rg * 8]);
@@ -189,6 +228,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         oacc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, oacc[n], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
