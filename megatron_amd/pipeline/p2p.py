@@ -1,0 +1,110 @@
+"""Point-to-point pipeline communication.
+
+Capability analog of reference megatron/core/pipeline_parallel/
+p2p_communication.py (P2PCommunicator :140, _p2p_ops :55, _batched_p2p_ops
+:17).  All four directions can be fused into one ``batch_isend_irecv`` so
+concurrent sends/recvs match without ordering deadlocks — on one xGMI node
+every PP hop is a single direct link.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from megatron_amd.parallel import grid as G
+
+
+class P2PCommunicator:
+    def __init__(self, config, seq_length: int, micro_batch_size: int):
+        self.config = config
+        grid = G.get_grid()
+        self.grid = grid
+        self.prev_rank = grid.pipeline_prev_rank()
+        self.next_rank = grid.pipeline_next_rank()
+        self.group = grid.group("pp")
+        s = seq_length
+        if config.sequence_parallel:
+            s //= max(1, config.tensor_parallel_size)
+        s //= max(1, config.context_parallel_size)
+        self.shape = (s, micro_batch_size, config.hidden_size)
+        self.dtype = config.pipeline_dtype
+
+    def _buf(self, device) -> torch.Tensor:
+        return torch.empty(self.shape, dtype=self.dtype, device=device, requires_grad=True)
+
+    def _device(self):
+        return torch.device("cuda", torch.cuda.current_device()) if torch.cuda.is_available() else torch.device("cpu")
+
+    def communicate(
+        self,
+        tensor_send_next: Optional[torch.Tensor] = None,
+        tensor_send_prev: Optional[torch.Tensor] = None,
+        recv_next: bool = False,
+        recv_prev: bool = False,
+    ):
+        """Fused bidirectional exchange; returns (tensor_recv_prev, tensor_recv_next)."""
+        dev = self._device()
+        tensor_recv_prev = self._buf(dev) if recv_prev else None
+        tensor_recv_next = self._buf(dev) if recv_next else None
+        ops = []
+        # rank-parity ordering keeps pairwise matching deterministic even if
+        # the backend serializes (reference p2p_communication.py:67)
+        even = self.grid.pp_rank % 2 == 0
+        def add_sends():
+            if tensor_send_prev is not None:
+                ops.append(dist.P2POp(dist.isend, tensor_send_prev.contiguous(), self.prev_rank, group=self.group))
+            if tensor_send_next is not None:
+                ops.append(dist.P2POp(dist.isend, tensor_send_next.contiguous(), self.next_rank, group=self.group))
+        def add_recvs():
+            if tensor_recv_prev is not None:
+                ops.append(dist.P2POp(dist.irecv, tensor_recv_prev, self.prev_rank, group=self.group))
+            if tensor_recv_next is not None:
+                ops.append(dist.P2POp(dist.irecv, tensor_recv_next, self.next_rank, group=self.group))
+        if even:
+            add_sends()
+            add_recvs()
+        else:
+            add_recvs()
+            add_sends()
+        if ops:
+            reqs = dist.batch_isend_irecv(ops)
+            for r in reqs:
+                r.wait()
+        return tensor_recv_prev, tensor_recv_next
+
+    # -- convenience wrappers (reference API shape) --------------------------
+
+    def recv_forward(self, is_first_stage: bool):
+        if is_first_stage:
+            return None
+        t, _ = self.communicate(recv_prev=True)
+        return t
+
+    def recv_backward(self, is_last_stage: bool):
+        if is_last_stage:
+            return None
+        _, t = self.communicate(recv_next=True)
+        return t
+
+    def send_forward(self, output, is_last_stage: bool):
+        if not is_last_stage:
+            self.communicate(tensor_send_next=output)
+
+    def send_backward(self, input_grad, is_first_stage: bool):
+        if not is_first_stage:
+            self.communicate(tensor_send_prev=input_grad)
+
+    def send_forward_recv_backward(self, output, is_last_stage: bool):
+        if is_last_stage:
+            return None
+        _, grad = self.communicate(tensor_send_next=output, recv_next=True)
+        return grad
+
+    def send_backward_recv_forward(self, input_grad, is_first_stage: bool):
+        if is_first_stage:
+            return None
+        t, _ = self.communicate(tensor_send_prev=input_grad, recv_prev=True)
+        return t

@@ -95,9 +95,14 @@ def train_step(
             loss_sum += m["loss_sum"]
     total_tokens = num_tokens.clone()
     if G.grid_initialized() and dist.is_initialized():
-        group = G.get_grid().group("dp_cp")
+        grid = G.get_grid()
+        group = grid.group("dp_cp")
         if group is not None and dist.get_world_size(group=group) > 1:
             dist.all_reduce(loss_sum, group=group)
             dist.all_reduce(total_tokens, group=group)
+        # loss lives on the last PP stage: share it down the pipe for logging
+        if grid.pp > 1 and grid.group("pp") is not None:
+            dist.broadcast(loss_sum, src=grid.pipeline_last_rank(), group=grid.group("pp"))
+            dist.broadcast(total_tokens, src=grid.pipeline_last_rank(), group=grid.group("pp"))
     mean_loss = loss_sum / total_tokens.clamp(min=1)
     return {"lm_loss": mean_loss.item(), "grad_norm": None if grad_norm is None else float(grad_norm), "skipped": not ok}
