@@ -37,6 +37,13 @@ MODELS = {
         num_layers=80, hidden_size=8192, num_attention_heads=64, num_query_groups=8,
         ffn_hidden_size=28672, vocab_size=128256, rotary_base=500000.0,
     ),
+    # Mixtral 8x7B (BASELINE.json "Mixtral 8x7B expert-parallel (all-to-all over xGMI)")
+    "mixtral-8x7b": dict(
+        num_layers=32, hidden_size=4096, num_attention_heads=32, num_query_groups=8,
+        ffn_hidden_size=14336, vocab_size=32000, rotary_base=1000000.0,
+        num_experts=8, moe_router_topk=2, moe_ffn_hidden_size=14336,
+        moe_aux_loss_coeff=0.01,
+    ),
     # tiny shape for CI plumbing
     "tiny": dict(
         num_layers=2, hidden_size=256, num_attention_heads=4, num_query_groups=2,
@@ -56,6 +63,7 @@ def parse_args():
     p.add_argument("--grad-accum", type=int, default=8, help="microbatches per step per DP rank")
     p.add_argument("--tp", type=int, default=None, help="tensor parallel size (default: model-dependent)")
     p.add_argument("--pp", type=int, default=1)
+    p.add_argument("--ep", type=int, default=None, help="expert parallel size (default: world for MoE)")
     p.add_argument("--vpp", type=int, default=None)
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--no-dist-opt", action="store_true")
@@ -86,11 +94,15 @@ def main():
         tp = 1  # single-GPU measurement of the 70B shape is impossible; caller sets --tp
     sp = tp > 1
 
+    is_moe = "num_experts" in MODELS[args.model]
+    ep = args.ep if args.ep is not None else (min(world, 8) if is_moe else 1)
     if use_dist:
         G.initialize_model_parallel(tensor_parallel_size=tp, pipeline_parallel_size=args.pp,
-                                    virtual_pipeline_parallel_size=args.vpp)
+                                    virtual_pipeline_parallel_size=args.vpp,
+                                    expert_parallel_size=ep)
     else:
         G.initialize_model_parallel(tensor_parallel_size=1, world_size=1, rank=0)
+        ep = 1
     model_parallel_seed(args.seed)
     grid = G.get_grid()
     dp = grid.dp
@@ -105,6 +117,7 @@ def main():
         pipeline_parallel_size=args.pp,
         virtual_pipeline_parallel_size=args.vpp,
         sequence_parallel=sp,
+        expert_parallel_size=ep,
         recompute_granularity="full" if args.recompute else None,
         gradient_accumulation_fusion=device.type == "cuda",
     )
@@ -174,7 +187,7 @@ def main():
     mfu = tflops_per_gpu / MI355X_BF16_DENSE_PEAK_TFLOPS
 
     if rank == 0:
-        par = f"tp{tp}" + (f"pp{args.pp}" if args.pp > 1 else "") + f"dp{dp}"
+        par = f"tp{tp}" + (f"pp{args.pp}" if args.pp > 1 else "") + f"dp{dp}" + (f"ep{ep}" if ep > 1 else "")
         print(json.dumps({
             "metric": "tokens/s",
             "value": round(tokens_per_s, 1),

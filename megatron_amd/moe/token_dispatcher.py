@@ -1,0 +1,130 @@
+"""MoE token dispatchers.
+
+Capability analog of reference megatron/core/transformer/moe/
+token_dispatcher.py (MoEAlltoAllTokenDispatcher :372, MoEAllGatherTokenDispatcher
+:230) + the A2A metadata computation.
+
+xGMI note: the node is fully connected, so all_to_all_single is single-hop —
+the reference's DeepEP/NVSHMEM fused path exists because NVLink-domain
+crossing is expensive; here plain RCCL A2A (C12) is already topology-optimal.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.mappings import all_to_all
+
+
+def permute(tokens: torch.Tensor, top_idx: torch.Tensor, num_experts: int):
+    """Replicate+sort tokens by target expert.
+
+    tokens [T, h], top_idx [T, k] -> (permuted [T*k, h], sort_order [T*k],
+    tokens_per_expert [E]).  HIP gather kernel (K12) replaces the
+    index_select later; torch path is already a single gather.
+    """
+    T, k = top_idx.shape
+    flat_experts = top_idx.reshape(-1)  # [T*k] expert id per (token, slot)
+    sort_order = torch.argsort(flat_experts, stable=True)
+    src_token = sort_order // k  # originating token row
+    permuted = tokens.index_select(0, src_token)
+    tokens_per_expert = torch.bincount(flat_experts, minlength=num_experts)
+    return permuted, sort_order, tokens_per_expert
+
+
+def unpermute(permuted: torch.Tensor, sort_order: torch.Tensor, probs: torch.Tensor, T: int):
+    """Inverse of permute with prob weighting: out[t] = sum_k prob[t,k] * x[(t,k)]."""
+    k = probs.shape[1]
+    inv = torch.argsort(sort_order)
+    out_flat = permuted.index_select(0, inv)
+    out_flat = out_flat.view(T, k, -1) * probs.unsqueeze(-1).to(permuted.dtype)
+    return out_flat.sum(dim=1)
+
+
+class MoEAlltoAllTokenDispatcher:
+    """permute -> A2A(EP) -> sort-by-local-expert -> experts -> A2A -> unpermute."""
+
+    def __init__(self, config):
+        self.config = config
+        self.num_experts = config.num_experts
+        self.ep = G.get_expert_model_parallel_world_size() if G.grid_initialized() else 1
+        self.num_local_experts = self.num_experts // max(self.ep, 1)
+        self.group = G.get_grid().group("ep") if G.grid_initialized() else None
+
+    def dispatch(self, tokens: torch.Tensor, probs: torch.Tensor, top_idx: torch.Tensor):
+        T = tokens.shape[0]
+        permuted, sort_order, tokens_per_expert = permute(tokens, top_idx, self.num_experts)
+        self._sort_order = sort_order
+        self._T = T
+        self._probs = probs
+        if self.ep == 1:
+            self._restore = None
+            return permuted, tokens_per_expert
+
+        # per-EP-peer split sizes (each peer owns num_local_experts experts)
+        input_splits = tokens_per_expert.view(self.ep, self.num_local_experts).sum(dim=1)
+        counts_matrix = torch.empty(self.ep * self.num_experts, dtype=tokens_per_expert.dtype,
+                                    device=tokens_per_expert.device)
+        torch.distributed.all_gather_into_tensor(
+            counts_matrix, tokens_per_expert.contiguous(), group=self.group
+        )
+        counts_matrix = counts_matrix.view(self.ep, self.num_experts)  # [src_rank, expert]
+        rank = torch.distributed.get_rank(group=self.group)
+        my_slice = counts_matrix[:, rank * self.num_local_experts : (rank + 1) * self.num_local_experts]
+        output_splits = my_slice.sum(dim=1)  # tokens arriving from each peer
+        self._input_splits = input_splits.tolist()
+        self._output_splits = output_splits.tolist()
+
+        recv = all_to_all(self.group, permuted, self._output_splits, self._input_splits)
+
+        # received tokens are grouped by (src_rank, local_expert); resort to
+        # (local_expert, src_rank) so each expert's tokens are contiguous
+        # (reference sort_chunks_by_idxs moe_utils.py:628)
+        chunk_sizes = my_slice.reshape(-1)  # [ep * n_local] in (rank, expert) order
+        idx = (
+            torch.arange(self.ep * self.num_local_experts, device=tokens.device)
+            .view(self.ep, self.num_local_experts).t().reshape(-1)
+        )
+        self._chunk_sizes = chunk_sizes.tolist()
+        self._chunk_perm = idx.tolist()
+        chunks = torch.split(recv, self._chunk_sizes)
+        reordered = torch.cat([chunks[i] for i in self._chunk_perm], dim=0) if len(chunks) > 1 else recv
+        tokens_per_local_expert = my_slice.sum(dim=0)
+        self._restore = True
+        return reordered, tokens_per_local_expert
+
+    def combine(self, expert_out: torch.Tensor) -> torch.Tensor:
+        if self.ep > 1:
+            # invert the (expert, rank) reorder, then A2A back
+            sizes = [self._chunk_sizes[i] for i in self._chunk_perm]
+            chunks = torch.split(expert_out, sizes)
+            inv = [0] * len(self._chunk_perm)
+            for pos, orig in enumerate(self._chunk_perm):
+                inv[orig] = pos
+            expert_out = torch.cat([chunks[inv[i]] for i in range(len(inv))], dim=0) if len(chunks) > 1 else expert_out
+            expert_out = all_to_all(self.group, expert_out, self._input_splits, self._output_splits)
+        return unpermute(expert_out, self._sort_order, self._probs, self._T)
+
+
+class MoEAllGatherTokenDispatcher:
+    """Simpler dispatcher: every rank computes its local experts on the full
+    (DP-local) token set using the routing map; no A2A.  Used as the
+    equivalence oracle for the alltoall dispatcher (reference :230)."""
+
+    def __init__(self, config):
+        self.config = config
+        self.num_experts = config.num_experts
+        self.ep = G.get_expert_model_parallel_world_size() if G.grid_initialized() else 1
+        self.num_local_experts = self.num_experts // max(self.ep, 1)
+        assert self.ep == 1, "allgather dispatcher here supports ep=1 (oracle use)"
+
+    def dispatch(self, tokens, probs, top_idx):
+        permuted, sort_order, tokens_per_expert = permute(tokens, top_idx, self.num_experts)
+        self._sort_order, self._T, self._probs = sort_order, tokens.shape[0], probs
+        return permuted, tokens_per_expert
+
+    def combine(self, expert_out):
+        return unpermute(expert_out, self._sort_order, self._probs, self._T)

@@ -31,6 +31,9 @@ def _fwd(forward_step_func, data_iterator, model, input_tensor, losses, num_toke
         losses.append(metrics)
         num_tokens_acc.add_(num_tokens)
         scale = 1.0 / (max(int(num_tokens), 1) * num_microbatches)
+        from megatron_amd.moe.router import AuxLossScaler
+
+        AuxLossScaler.main_loss_backward_scale = scale
         out = loss * scale
         if config.grad_scale_func is not None:
             out = config.grad_scale_func(out)
