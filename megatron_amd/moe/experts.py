@@ -38,6 +38,8 @@ class GroupedMLP(nn.Module):
         self.weight2 = nn.Parameter(
             torch.empty(self.num_local_experts, config.hidden_size, self.ffn_per_partition, dtype=config.params_dtype)
         )
+        if self.gated:
+            self.weight1.is_gated_fc1 = True
         for w in (self.weight1, self.weight2):
             w.is_expert_parallel = True
             w.tensor_parallel = tp > 1

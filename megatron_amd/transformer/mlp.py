@@ -27,6 +27,10 @@ class MLP(nn.Module):
         self.linear_fc2 = RowParallelLinear(
             ffn, config.hidden_size, config=config, bias=config.add_linear_bias, is_expert=is_expert
         )
+        if self.gated:
+            # per-TP-rank layout is [gate_shard ; up_shard]: checkpointing must
+            # split this weight into two global tensors to be reshard-safe
+            self.linear_fc1.weight.is_gated_fc1 = True
         self.activation = config.activation
 
     def _act(self, x: torch.Tensor) -> torch.Tensor:
