@@ -59,8 +59,8 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="llama3-8b")
     p.add_argument("--seq-len", type=int, default=4096)
-    p.add_argument("--micro-batch-size", type=int, default=1)
-    p.add_argument("--grad-accum", type=int, default=8, help="microbatches per step per DP rank")
+    p.add_argument("--micro-batch-size", type=int, default=4)
+    p.add_argument("--grad-accum", type=int, default=2, help="microbatches per step per DP rank")
     p.add_argument("--tp", type=int, default=None, help="tensor parallel size (default: model-dependent)")
     p.add_argument("--pp", type=int, default=1)
     p.add_argument("--ep", type=int, default=None, help="expert parallel size (default: world for MoE)")
