@@ -1,0 +1,227 @@
+"""CLI argument surface.
+
+Capability analog of reference megatron/training/arguments.py (:45
+add_megatron_arguments, 36 groups / ~1,200 flags): the flags that change
+behavior on an MI355X node, grouped the same way, with parse-time
+cross-validation and a translation into the config dataclasses.
+"""
+
+from __future__ import annotations
+
+import argparse
+
+import torch
+
+from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
+
+
+def build_arg_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(description="megatron_amd pretraining", allow_abbrev=False)
+
+    g = p.add_argument_group("model")
+    g.add_argument("--num-layers", type=int, required=False, default=2)
+    g.add_argument("--hidden-size", type=int, default=64)
+    g.add_argument("--ffn-hidden-size", type=int, default=None)
+    g.add_argument("--num-attention-heads", type=int, default=4)
+    g.add_argument("--num-query-groups", type=int, default=None)
+    g.add_argument("--kv-channels", type=int, default=None)
+    g.add_argument("--vocab-size", type=int, default=128256)
+    g.add_argument("--max-position-embeddings", type=int, default=4096)
+    g.add_argument("--normalization", choices=["rmsnorm", "layernorm"], default="rmsnorm")
+    g.add_argument("--norm-epsilon", type=float, default=1e-5)
+    g.add_argument("--swiglu", action="store_true", default=True)
+    g.add_argument("--activation", choices=["swiglu", "geglu", "gelu", "squared_relu"], default="swiglu")
+    g.add_argument("--add-linear-bias", action="store_true")
+    g.add_argument("--untie-embeddings-and-output-weights", action="store_true", default=True)
+    g.add_argument("--position-embedding-type", choices=["rope", "learned", "none"], default="rope")
+    g.add_argument("--rotary-base", type=float, default=500000.0)
+    g.add_argument("--rotary-percent", type=float, default=1.0)
+    g.add_argument("--attention-dropout", type=float, default=0.0)
+    g.add_argument("--hidden-dropout", type=float, default=0.0)
+    g.add_argument("--window-size", type=int, default=None)
+    g.add_argument("--qk-layernorm", action="store_true")
+    g.add_argument("--init-method-std", type=float, default=0.02)
+
+    g = p.add_argument_group("moe")
+    g.add_argument("--num-experts", type=int, default=None)
+    g.add_argument("--moe-router-topk", type=int, default=2)
+    g.add_argument("--moe-ffn-hidden-size", type=int, default=None)
+    g.add_argument("--moe-aux-loss-coeff", type=float, default=0.0)
+    g.add_argument("--moe-z-loss-coeff", type=float, default=0.0)
+    g.add_argument("--moe-router-score-function", choices=["softmax", "sigmoid"], default="softmax")
+    g.add_argument("--moe-shared-expert-intermediate-size", type=int, default=None)
+    g.add_argument("--moe-token-dispatcher-type", choices=["alltoall", "allgather"], default="alltoall")
+    g.add_argument("--moe-layer-freq", type=int, default=1)
+
+    g = p.add_argument_group("parallelism")
+    g.add_argument("--tensor-model-parallel-size", "--tp", type=int, default=1)
+    g.add_argument("--pipeline-model-parallel-size", "--pp", type=int, default=1)
+    g.add_argument("--virtual-pipeline-model-parallel-size", "--vpp", type=int, default=None)
+    g.add_argument("--context-parallel-size", "--cp", type=int, default=1)
+    g.add_argument("--expert-model-parallel-size", "--ep", type=int, default=1)
+    g.add_argument("--expert-tensor-parallel-size", type=int, default=None)
+    g.add_argument("--sequence-parallel", action="store_true")
+
+    g = p.add_argument_group("training")
+    g.add_argument("--micro-batch-size", type=int, default=1)
+    g.add_argument("--global-batch-size", type=int, default=None)
+    g.add_argument("--seq-length", type=int, default=4096)
+    g.add_argument("--train-iters", type=int, default=10)
+    g.add_argument("--eval-interval", type=int, default=0)
+    g.add_argument("--eval-iters", type=int, default=2)
+    g.add_argument("--exit-interval", type=int, default=None)
+    g.add_argument("--recompute-granularity", choices=["full", "selective"], default=None)
+    g.add_argument("--recompute-num-layers", type=int, default=None)
+    g.add_argument("--bf16", action="store_true")
+    g.add_argument("--fp16", action="store_true")
+    g.add_argument("--seed", type=int, default=1234)
+    g.add_argument("--deterministic-mode", action="store_true")
+
+    g = p.add_argument_group("optimizer")
+    g.add_argument("--lr", type=float, default=3e-4)
+    g.add_argument("--min-lr", type=float, default=0.0)
+    g.add_argument("--lr-decay-style", choices=["constant", "linear", "cosine", "wsd"], default="cosine")
+    g.add_argument("--lr-warmup-iters", type=int, default=0)
+    g.add_argument("--lr-decay-iters", type=int, default=None)
+    g.add_argument("--weight-decay", type=float, default=0.1)
+    g.add_argument("--adam-beta1", type=float, default=0.9)
+    g.add_argument("--adam-beta2", type=float, default=0.95)
+    g.add_argument("--adam-eps", type=float, default=1e-8)
+    g.add_argument("--clip-grad", type=float, default=1.0)
+    g.add_argument("--loss-scale", type=float, default=None)
+    g.add_argument("--use-distributed-optimizer", action="store_true")
+    g.add_argument("--overlap-param-gather", action="store_true")
+
+    g = p.add_argument_group("ddp")
+    g.add_argument("--overlap-grad-reduce", action="store_true", default=True)
+    g.add_argument("--no-overlap-grad-reduce", dest="overlap_grad_reduce", action="store_false")
+    g.add_argument("--accumulate-allreduce-grads-in-fp32", dest="grad_reduce_in_fp32",
+                   action="store_true", default=True)
+    g.add_argument("--bucket-size", type=int, default=40_000_000)
+
+    g = p.add_argument_group("checkpointing")
+    g.add_argument("--save", type=str, default=None)
+    g.add_argument("--load", type=str, default=None)
+    g.add_argument("--save-interval", type=int, default=None)
+    g.add_argument("--async-save", action="store_true")
+    g.add_argument("--no-load-rng", action="store_true")
+
+    g = p.add_argument_group("data")
+    g.add_argument("--data-path", type=str, nargs="*", default=None)
+    g.add_argument("--mock-data", action="store_true")
+    g.add_argument("--tokenizer-type", type=str, default="NullTokenizer")
+    g.add_argument("--tokenizer-model", type=str, default=None)
+    g.add_argument("--split", type=str, default="969,30,1")
+    g.add_argument("--num-workers", type=int, default=0)
+
+    g = p.add_argument_group("logging")
+    g.add_argument("--log-interval", type=int, default=1)
+    g.add_argument("--log-throughput", action="store_true", default=True)
+    g.add_argument("--log-timers", action="store_true")
+    g.add_argument("--tensorboard-dir", type=str, default=None)
+    g.add_argument("--log-memory", action="store_true")
+
+    g = p.add_argument_group("profiling")
+    g.add_argument("--profile", action="store_true")
+    g.add_argument("--profile-step-start", type=int, default=3)
+    g.add_argument("--profile-step-end", type=int, default=5)
+    g.add_argument("--profile-dir", type=str, default="./torchprof")
+
+    return p
+
+
+def validate_args(args) -> None:
+    world = args.world_size
+    mp = args.tensor_model_parallel_size * args.pipeline_model_parallel_size * args.context_parallel_size
+    assert world % mp == 0, f"world {world} not divisible by tp*pp*cp={mp}"
+    dp = world // mp
+    if args.global_batch_size is None:
+        args.global_batch_size = args.micro_batch_size * dp
+    gbs_div = args.micro_batch_size * dp
+    assert args.global_batch_size % gbs_div == 0, (
+        f"global batch {args.global_batch_size} must divide by micro_batch*dp={gbs_div}"
+    )
+    args.num_microbatches = args.global_batch_size // gbs_div
+    args.data_parallel_size = dp
+    if args.fp16 and args.bf16:
+        raise ValueError("choose one of --fp16 / --bf16")
+    if args.sequence_parallel and args.tensor_model_parallel_size == 1:
+        args.sequence_parallel = False
+    if args.num_experts is not None:
+        assert args.num_experts % args.expert_model_parallel_size == 0
+
+
+def configs_from_args(args):
+    cfg = TransformerConfig(
+        num_layers=args.num_layers,
+        hidden_size=args.hidden_size,
+        ffn_hidden_size=args.ffn_hidden_size,
+        num_attention_heads=args.num_attention_heads,
+        num_query_groups=args.num_query_groups,
+        kv_channels=args.kv_channels,
+        vocab_size=args.vocab_size,
+        max_position_embeddings=args.max_position_embeddings,
+        normalization=args.normalization,
+        layernorm_epsilon=args.norm_epsilon,
+        activation=args.activation,
+        add_linear_bias=args.add_linear_bias,
+        untie_embeddings_and_output_weights=args.untie_embeddings_and_output_weights,
+        position_embedding_type=args.position_embedding_type,
+        rotary_base=args.rotary_base,
+        rotary_percent=args.rotary_percent,
+        attention_dropout=args.attention_dropout,
+        hidden_dropout=args.hidden_dropout,
+        window_size=args.window_size,
+        qk_layernorm=args.qk_layernorm,
+        init_method_std=args.init_method_std,
+        num_experts=args.num_experts,
+        moe_router_topk=args.moe_router_topk,
+        moe_ffn_hidden_size=args.moe_ffn_hidden_size,
+        moe_aux_loss_coeff=args.moe_aux_loss_coeff,
+        moe_z_loss_coeff=args.moe_z_loss_coeff,
+        moe_router_score_function=args.moe_router_score_function,
+        moe_shared_expert_intermediate_size=args.moe_shared_expert_intermediate_size,
+        moe_token_dispatcher_type=args.moe_token_dispatcher_type,
+        moe_layer_freq=args.moe_layer_freq,
+        tensor_parallel_size=args.tensor_model_parallel_size,
+        pipeline_parallel_size=args.pipeline_model_parallel_size,
+        virtual_pipeline_parallel_size=args.virtual_pipeline_model_parallel_size,
+        context_parallel_size=args.context_parallel_size,
+        expert_parallel_size=args.expert_model_parallel_size,
+        expert_tensor_parallel_size=args.expert_tensor_parallel_size,
+        sequence_parallel=args.sequence_parallel,
+        bf16=args.bf16,
+        fp16=args.fp16,
+        recompute_granularity=args.recompute_granularity,
+        recompute_num_layers=args.recompute_num_layers,
+        deterministic_mode=args.deterministic_mode,
+        gradient_accumulation_fusion=torch.cuda.is_available(),
+    )
+    opt_cfg = OptimizerConfig(
+        lr=args.lr, min_lr=args.min_lr, weight_decay=args.weight_decay,
+        adam_beta1=args.adam_beta1, adam_beta2=args.adam_beta2, adam_eps=args.adam_eps,
+        clip_grad=args.clip_grad, bf16=args.bf16, fp16=args.fp16,
+        loss_scale=args.loss_scale,
+        use_distributed_optimizer=args.use_distributed_optimizer,
+        overlap_param_gather=args.overlap_param_gather,
+        lr_decay_style=args.lr_decay_style, lr_warmup_iters=args.lr_warmup_iters,
+        lr_decay_iters=args.lr_decay_iters,
+    )
+    ddp_cfg = DDPConfig(
+        grad_reduce_in_fp32=args.grad_reduce_in_fp32,
+        overlap_grad_reduce=args.overlap_grad_reduce,
+        use_distributed_optimizer=args.use_distributed_optimizer,
+        bucket_size=args.bucket_size,
+    )
+    return cfg, opt_cfg, ddp_cfg
+
+
+def parse_and_validate_args(argv=None):
+    import os
+
+    parser = build_arg_parser()
+    args = parser.parse_args(argv)
+    args.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    args.rank = int(os.environ.get("RANK", "0"))
+    validate_args(args)
+    return args

@@ -1,0 +1,46 @@
+"""Training application tests (reference: megatron/training/training.py pretrain loop,
+arguments.py parse_and_validate_args) — CPU, single process, mock data."""
+
+import os
+
+import pytest
+import torch
+
+from megatron_amd.training.arguments import parse_and_validate_args
+from megatron_amd.training.pretrain import pretrain
+
+TINY = [
+    "--num-layers", "2", "--hidden-size", "64", "--num-attention-heads", "4",
+    "--num-query-groups", "2", "--ffn-hidden-size", "128", "--seq-length", "32",
+    "--micro-batch-size", "2", "--global-batch-size", "4", "--vocab-size", "128",
+    "--mock-data", "--log-interval", "0",
+]
+
+
+def model_provider(config, pre_process=True, post_process=True, vp_stage=None):
+    from megatron_amd.models.gpt import GPTModel
+
+    return GPTModel(config, pre_process=pre_process, post_process=post_process, vp_stage=vp_stage)
+
+
+def test_parse_and_validate():
+    args = parse_and_validate_args(TINY + ["--train-iters", "5"])
+    assert args.num_microbatches == 2
+    assert args.data_parallel_size == 1
+    assert args.global_batch_size == 4
+
+
+def test_parse_rejects_bad_gbs():
+    with pytest.raises(AssertionError):
+        parse_and_validate_args(TINY + ["--global-batch-size", "3"])
+
+
+def test_pretrain_runs_and_resumes(tmp_path):
+    ckpt = str(tmp_path / "ckpt")
+    it = pretrain(model_provider, TINY + ["--train-iters", "3", "--save", ckpt, "--seed", "7"])
+    assert it == 3
+    assert os.path.exists(os.path.join(ckpt, "latest_checkpointed_iteration.txt"))
+    # resume continues from saved iteration and finishes the remaining steps
+    it2 = pretrain(model_provider, TINY + ["--train-iters", "5", "--save", ckpt,
+                                           "--load", ckpt, "--seed", "7"])
+    assert it2 == 5
