@@ -11,7 +11,7 @@ import os
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from setuptools import setup
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension
 
 SRC = [
     "megatron_amd/ops/csrc/bindings.cpp",
@@ -35,7 +35,12 @@ setup(
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
             },
             libraries=["rocblas"],
-        )
+        ),
+        CppExtension(
+            name="megatron_amd.datasets._data_helpers",
+            sources=["megatron_amd/datasets/csrc/data_helpers.cpp"],
+            extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
+        ),
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )
