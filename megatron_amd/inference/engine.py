@@ -1,0 +1,255 @@
+"""Inference engines.
+
+Capability analog of reference megatron/core/inference/engines/
+(StaticInferenceEngine static_engine.py; DynamicInferenceEngine
+dynamic_engine.py — continuous batching, paged KV, chunked prefill).
+
+Static engine: fixed batch, contiguous KV. Prefill runs the whole padded
+batch in one flash-attention pass (causal masking makes pad positions inert;
+their KV slots are overwritten during decode). Dynamic engine: request queue,
+paged KV blocks, one prefill or one batched decode step per engine step.
+"""
+
+from __future__ import annotations
+
+import itertools
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence
+
+import torch
+
+from megatron_amd.inference.contexts import DynamicInferenceContext, StaticInferenceContext
+from megatron_amd.inference.sampling import SamplingParams, log_prob_of, sample
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.mappings import gather_from_tensor_model_parallel_region
+
+
+@dataclass
+class GenerationResult:
+    request_id: int
+    prompt_tokens: List[int]
+    output_tokens: List[int] = field(default_factory=list)
+    log_probs: List[float] = field(default_factory=list)
+    text: Optional[str] = None
+    finished: bool = False
+
+
+def _full_logits(logits_tp: torch.Tensor) -> torch.Tensor:
+    """[.., V/tp] -> [.., V] (all-gather over TP; identity at TP=1)."""
+    if G.get_tensor_model_parallel_world_size() == 1:
+        return logits_tp
+    return gather_from_tensor_model_parallel_region(logits_tp)
+
+
+class StaticInferenceEngine:
+    def __init__(self, model, tokenizer=None, max_batch: int = 8, max_seq: int = 2048,
+                 device=None):
+        self.model = model.eval()
+        self.tokenizer = tokenizer
+        cfg = model.config
+        device = device or next(model.parameters()).device
+        self.device = device
+        self.eod = tokenizer.eod if tokenizer is not None else cfg.vocab_size - 1
+        tp = G.get_tensor_model_parallel_world_size()
+        self.context = StaticInferenceContext(
+            cfg.num_layers, max_batch, max_seq,
+            (cfg.num_query_groups or cfg.num_attention_heads) // tp,
+            cfg.kv_channels, dtype=cfg.params_dtype, device=device)
+        self.max_seq = max_seq
+
+    @torch.no_grad()
+    def generate(self, prompts: Sequence, params: SamplingParams = SamplingParams()) -> List[GenerationResult]:
+        if self.tokenizer is not None and isinstance(prompts[0], str):
+            prompts = [self.tokenizer.tokenize(p) for p in prompts]
+        prompts = [list(p) for p in prompts]
+        b = len(prompts)
+        assert b <= self.context.max_batch
+        lens = [len(p) for p in prompts]
+        Lmax = max(lens)
+        assert Lmax + params.max_tokens <= self.max_seq
+        gen = None
+        if params.seed is not None:
+            gen = torch.Generator(device="cpu").manual_seed(params.seed)
+
+        tokens = torch.full((b, Lmax), self.eod, dtype=torch.long, device=self.device)
+        for i, p in enumerate(prompts):
+            tokens[i, : len(p)] = torch.as_tensor(p, device=self.device)
+
+        self.context.reset(b)
+        logits_tp = self.model(tokens, inference_context=self.context)  # [Lmax, b, V/tp]
+        self.context.set_prompt_lens(lens)
+        last_pos = torch.as_tensor(lens, device=self.device) - 1
+        last = logits_tp[last_pos, torch.arange(b, device=self.device)]  # [b, V/tp]
+        logits = _full_logits(last).float()
+
+        results = [GenerationResult(i, p) for i, p in enumerate(prompts)]
+        finished = torch.zeros(b, dtype=torch.bool, device=self.device)
+        for _ in range(params.max_tokens):
+            next_tok = sample(logits.cpu() if gen is not None else logits, params, gen).to(self.device)
+            if params.return_log_probs:
+                lp = log_prob_of(logits, next_tok)
+            for i in range(b):
+                if not bool(finished[i]):
+                    results[i].output_tokens.append(int(next_tok[i]))
+                    if params.return_log_probs:
+                        results[i].log_probs.append(float(lp[i]))
+            if params.stop_on_eod:
+                finished |= next_tok == self.eod
+            if bool(finished.all()):
+                break
+            logits_tp = self.model(next_tok.view(b, 1), inference_context=self.context)
+            self.context.advance(1)
+            logits = _full_logits(logits_tp[0]).float()
+
+        for r in results:
+            if params.stop_on_eod and r.output_tokens and r.output_tokens[-1] == self.eod:
+                r.output_tokens = r.output_tokens[:-1]
+            r.finished = True
+            if self.tokenizer is not None:
+                r.text = self.tokenizer.detokenize(r.output_tokens)
+        return results
+
+
+@dataclass
+class _Request:
+    rid: int
+    prompt: List[int]
+    params: SamplingParams
+    result: GenerationResult
+    block_table: List[int] = field(default_factory=list)
+    cached: int = 0          # prompt tokens already in the KV cache
+    next_input: Optional[int] = None  # token to feed at the next decode step
+    gen: Optional[torch.Generator] = None
+
+
+class DynamicInferenceEngine:
+    """Continuous batching over a paged KV cache."""
+
+    def __init__(self, model, tokenizer=None, num_blocks: int = 512, block_size: int = 256,
+                 max_batch: int = 64, max_prefill_tokens: int = 8192, device=None):
+        self.model = model.eval()
+        self.tokenizer = tokenizer
+        cfg = model.config
+        device = device or next(model.parameters()).device
+        self.device = device
+        self.eod = tokenizer.eod if tokenizer is not None else cfg.vocab_size - 1
+        tp = G.get_tensor_model_parallel_world_size()
+        self.context = DynamicInferenceContext(
+            cfg.num_layers,
+            (cfg.num_query_groups or cfg.num_attention_heads) // tp,
+            cfg.kv_channels, num_blocks=num_blocks, block_size=block_size,
+            dtype=cfg.params_dtype, device=device)
+        self.max_batch = max_batch
+        self.max_prefill_tokens = max_prefill_tokens
+        self._ids = itertools.count()
+        self.waiting: List[_Request] = []
+        self.active: List[_Request] = []
+        self.finished: Dict[int, GenerationResult] = {}
+
+    def add_request(self, prompt, params: SamplingParams = SamplingParams()) -> int:
+        if self.tokenizer is not None and isinstance(prompt, str):
+            prompt = self.tokenizer.tokenize(prompt)
+        rid = next(self._ids)
+        gen = torch.Generator(device="cpu").manual_seed(params.seed) if params.seed is not None else None
+        req = _Request(rid, list(prompt), params, GenerationResult(rid, list(prompt)), gen=gen)
+        self.waiting.append(req)
+        return rid
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.active)
+
+    def _blocks_for(self, n_tokens: int) -> int:
+        bs = self.context.block_size
+        return (n_tokens + bs - 1) // bs
+
+    def _ensure_blocks(self, req: _Request, total_tokens: int) -> bool:
+        need = self._blocks_for(total_tokens) - len(req.block_table)
+        if need <= 0:
+            return True
+        if need > self.context.allocator.num_free:
+            return False
+        req.block_table.extend(self.context.allocator.allocate(need))
+        return True
+
+    def _finish(self, req: _Request):
+        self.context.allocator.free(req.block_table)
+        req.block_table = []
+        req.result.finished = True
+        if self.tokenizer is not None:
+            req.result.text = self.tokenizer.detokenize(req.result.output_tokens)
+        self.finished[req.rid] = req.result
+
+    def _sample_row(self, logits_row: torch.Tensor, req: _Request) -> int:
+        logits_row = logits_row.unsqueeze(0)
+        if req.gen is not None:
+            logits_row = logits_row.cpu()
+        tok = int(sample(logits_row, req.params, req.gen)[0])
+        if req.params.return_log_probs:
+            req.result.log_probs.append(float(log_prob_of(logits_row, torch.tensor([tok]))[0]))
+        return tok
+
+    @torch.no_grad()
+    def step(self) -> None:
+        """One engine step: chunked prefill of the next waiting request, or one
+        batched decode step over all active requests."""
+        if self.waiting and len(self.active) < self.max_batch:
+            req = self.waiting[0]
+            chunk = min(self.max_prefill_tokens, len(req.prompt) - req.cached)
+            if not self._ensure_blocks(req, req.cached + chunk):
+                if not self.active:
+                    raise RuntimeError("KV pool too small for a single prompt chunk")
+            else:
+                self.context.begin_prefill(req.block_table, req.cached)
+                toks = torch.as_tensor(req.prompt[req.cached:req.cached + chunk],
+                                       device=self.device).view(1, -1)
+                logits_tp = self.model(toks, inference_context=self.context)  # [chunk, 1, V/tp]
+                req.cached += chunk
+                if req.cached == len(req.prompt):
+                    logits = _full_logits(logits_tp[-1, 0]).float()
+                    tok = self._sample_row(logits, req)
+                    req.result.output_tokens.append(tok)
+                    self.waiting.pop(0)
+                    if req.params.stop_on_eod and tok == self.eod:
+                        req.result.output_tokens.pop()
+                        self._finish(req)
+                    elif len(req.result.output_tokens) >= req.params.max_tokens:
+                        self._finish(req)
+                    else:
+                        req.next_input = tok
+                        self.active.append(req)
+                return
+
+        if not self.active:
+            return
+        # batched decode
+        batch = self.active
+        for req in batch:
+            total = len(req.prompt) + len(req.result.output_tokens)
+            ok = self._ensure_blocks(req, total + 1)
+            assert ok, "KV pool exhausted during decode"  # TODO: preemption
+        tables = [r.block_table for r in batch]
+        lens = [len(r.prompt) + len(r.result.output_tokens) - 1 for r in batch]  # cached so far
+        self.context.begin_decode(tables, lens)
+        toks = torch.as_tensor([r.next_input for r in batch], device=self.device).view(-1, 1)
+        logits_tp = self.model(toks, inference_context=self.context)  # [1, b, V/tp]
+        logits = _full_logits(logits_tp[0]).float()
+        still = []
+        for i, req in enumerate(batch):
+            tok = self._sample_row(logits[i], req)
+            req.result.output_tokens.append(tok)
+            if req.params.stop_on_eod and tok == self.eod:
+                req.result.output_tokens.pop()
+                self._finish(req)
+            elif len(req.result.output_tokens) >= req.params.max_tokens:
+                self._finish(req)
+            else:
+                req.next_input = tok
+                still.append(req)
+        self.active = still
+
+    @torch.no_grad()
+    def generate(self, prompts: Sequence, params: SamplingParams = SamplingParams()) -> List[GenerationResult]:
+        ids = [self.add_request(p, params) for p in prompts]
+        while self.has_work():
+            self.step()
+        return [self.finished.pop(i) for i in ids]

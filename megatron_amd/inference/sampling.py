@@ -1,0 +1,54 @@
+"""Token sampling (capability analog of reference inference text-generation
+controller sampling: greedy / temperature / top-k / top-p, log-probs)."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+
+@dataclass
+class SamplingParams:
+    max_tokens: int = 64
+    temperature: float = 1.0
+    top_k: int = 0          # 0 = off
+    top_p: float = 0.0      # 0 = off
+    greedy: bool = False
+    stop_on_eod: bool = True
+    return_log_probs: bool = False
+    seed: Optional[int] = None
+
+
+def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0) -> torch.Tensor:
+    """Mask logits outside top-k / nucleus top-p to -inf. logits: [b, V]."""
+    if top_k > 0:
+        kth = torch.topk(logits, min(top_k, logits.size(-1)), dim=-1).values[..., -1, None]
+        logits = logits.masked_fill(logits < kth, float("-inf"))
+    if top_p > 0.0:
+        sorted_logits, sorted_idx = torch.sort(logits, descending=True, dim=-1)
+        probs = torch.softmax(sorted_logits, dim=-1)
+        cum = torch.cumsum(probs, dim=-1)
+        # keep the smallest prefix with cumulative prob >= top_p (always >= 1 token)
+        drop = cum - probs >= top_p
+        sorted_logits = sorted_logits.masked_fill(drop, float("-inf"))
+        logits = torch.full_like(logits, float("-inf")).scatter_(-1, sorted_idx, sorted_logits)
+    return logits
+
+
+def sample(logits: torch.Tensor, params: SamplingParams,
+           generator: Optional[torch.Generator] = None) -> torch.Tensor:
+    """logits: [b, V] (full vocab, fp32) -> next token ids [b]."""
+    if params.greedy or params.temperature == 0.0:
+        return logits.argmax(dim=-1)
+    logits = logits / max(params.temperature, 1e-6)
+    logits = filter_logits(logits, params.top_k, params.top_p)
+    probs = torch.softmax(logits.float(), dim=-1)
+    return torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+
+
+def log_prob_of(logits: torch.Tensor, tokens: torch.Tensor) -> torch.Tensor:
+    """log p(token) under logits. logits [b, V], tokens [b] -> [b]."""
+    logp = torch.log_softmax(logits.float(), dim=-1)
+    return logp.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)

@@ -1,0 +1,58 @@
+"""REST text-generation server.
+
+Capability analog of reference megatron/core/inference/text_generation_server/
+(Flask MegatronServer + tools/run_text_generation_server.py), built on FastAPI/
+uvicorn (in-image). POST /api/generate {"prompts": [...], "max_tokens": ...,
+"temperature"/"top_k"/"top_p"/"greedy"/"logprobs"} -> generations. The engine
+runs under a lock; continuous batching batches concurrent requests naturally
+when driven through the dynamic engine's queue.
+"""
+
+import threading
+
+from fastapi import FastAPI, Request
+
+from megatron_amd.inference.sampling import SamplingParams
+
+
+def create_app(engine, tokenizer=None):
+
+    app = FastAPI(title="megatron_amd text generation")
+    lock = threading.Lock()
+
+    @app.get("/health")
+    def health():
+        return {"status": "ok"}
+
+    @app.post("/api/generate")
+    async def generate(request: Request):
+        req = await request.json()
+        prompts = req["prompts"]
+        logprobs = bool(req.get("logprobs", False))
+        params = SamplingParams(
+            max_tokens=int(req.get("max_tokens", 64)),
+            temperature=float(req.get("temperature", 1.0)),
+            top_k=int(req.get("top_k", 0)), top_p=float(req.get("top_p", 0.0)),
+            greedy=bool(req.get("greedy", False)), return_log_probs=logprobs,
+            seed=req.get("seed"))
+        with lock:
+            results = engine.generate(prompts, params)
+        return {
+            "generations": [
+                {
+                    "request_id": r.request_id,
+                    "tokens": r.output_tokens,
+                    "text": r.text,
+                    "logprobs": r.log_probs if logprobs else None,
+                }
+                for r in results
+            ]
+        }
+
+    return app
+
+
+def run_server(engine, tokenizer=None, host: str = "127.0.0.1", port: int = 5000):
+    import uvicorn
+
+    uvicorn.run(create_app(engine, tokenizer), host=host, port=port, log_level="warning")

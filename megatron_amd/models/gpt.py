@@ -116,7 +116,14 @@ class GPTModel(nn.Module):
                 G.get_tensor_model_parallel_world_size() if self.config.sequence_parallel else 1
             )
 
-        rotary = self._rotary_freqs(seq_len, hidden.device)
+        if inference_context is not None and self.config.position_embedding_type == "rope":
+            # positions come from the KV context (cache offset / per-request
+            # lengths under continuous batching): index the full freq table
+            table = self._rotary_freqs(self.config.max_position_embeddings, hidden.device)
+            pos = inference_context.rope_positions(seq_len)  # [s] or [s, b]
+            rotary = table[pos]
+        else:
+            rotary = self._rotary_freqs(seq_len, hidden.device)
         hidden = self.decoder(hidden, rotary_freqs=rotary, attention_mask=attention_mask,
                               inference_context=inference_context)
 

@@ -139,7 +139,12 @@ def rope_cos_sin(freqs: torch.Tensor):
 
 
 def rope_apply(x: torch.Tensor, freqs: torch.Tensor) -> torch.Tensor:
-    """x [s,b,h,d], freqs [s, d_rot/2] fp32 -> rotate-half RoPE."""
+    """x [s,b,h,d], freqs [s, d_rot/2] fp32 -> rotate-half RoPE.
+
+    freqs may also be [s, b, d_rot/2] (per-row positions, dynamic-batching
+    decode) — that path uses the torch implementation."""
+    if freqs.dim() == 3:
+        return ref.rope_apply_per_row(x, freqs)
     if _use_native(x):
         cos, sin = rope_cos_sin(freqs)
         return _RopeFn.apply(x.contiguous(), cos, sin)

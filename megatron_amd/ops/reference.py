@@ -79,6 +79,24 @@ def rope_apply(x: torch.Tensor, freqs: torch.Tensor) -> torch.Tensor:
     return out_rot.to(x.dtype)
 
 
+def rope_apply_per_row(x: torch.Tensor, freqs: torch.Tensor) -> torch.Tensor:
+    """Rotate-half RoPE with per-batch-row angles.
+
+    x: [s, b, h, d]; freqs: [s, b, d_rot/2] (fp32) — each sequence in the
+    batch sits at its own position (continuous-batching decode)."""
+    s, b, h, d = x.shape
+    d_rot = freqs.size(-1) * 2
+    xf = x.float()
+    x_rot, x_pass = xf[..., :d_rot], xf[..., d_rot:]
+    cos = torch.cos(freqs).view(s, b, 1, -1)
+    sin = torch.sin(freqs).view(s, b, 1, -1)
+    x1, x2 = x_rot[..., : d_rot // 2], x_rot[..., d_rot // 2 :]
+    out_rot = torch.cat((x1 * cos - x2 * sin, x2 * cos + x1 * sin), dim=-1)
+    if d_rot < d:
+        out_rot = torch.cat((out_rot, x_pass), dim=-1)
+    return out_rot.to(x.dtype)
+
+
 def bias_dropout_add(
     x: torch.Tensor, bias: Optional[torch.Tensor], residual: torch.Tensor, p: float, training: bool
 ) -> torch.Tensor:

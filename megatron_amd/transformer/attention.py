@@ -78,9 +78,10 @@ class SelfAttention(nn.Module):
             k = ops.rope_apply(k, rotary_freqs)
 
         if inference_context is not None:
-            k, v = inference_context.append_kv(self.layer_number, k, v)
-            core_out = ops.flash_attention(
-                q, k, v, causal=(q.shape[0] == k.shape[0]), scale=self.softmax_scale, window=self.window
+            # the context owns the KV cache and the attention kernel choice
+            # (contiguous flash for prefill, paged masked decode for dynamic)
+            core_out = inference_context.attend(
+                self.layer_number, q, k, v, self.softmax_scale, self.window
             )
         else:
             core_out = ops.flash_attention(q, k, v, causal=True, scale=self.softmax_scale, window=self.window)
