@@ -4,16 +4,23 @@
 // causal (with kv-offset for decode) or full; GQA (hq multiple of hkv);
 // optional sliding window.  Outputs: o [s, b, hq, d] bf16, lse [b, hq, s] f32.
 //
-// Structure (v1, correctness-first; see guide §B for the technique ladder):
-//   block = 4 waves, Q-tile 64 rows (16/wave), KV-tile 64.
-//   Q in registers; K staged in LDS row-major (+8 bf16 pad kills the
-//   32-way bank conflict); V staged TRANSPOSED (Vt[d][kv]) so the PV
-//   B-fragment reads are contiguous 16B ds_reads.
-//   mfma_f32_16x16x32_bf16 fragments:
-//     A: row = lane&15, k = (lane>>4)*8+j   (8 contiguous bf16)
-//     B: col = lane&15, k = (lane>>4)*8+j
-//     D: col = lane&15, row = (lane>>4)*4+reg   [guide §3, m89/m91-verified]
-//   Online softmax in fp32 with per-row m/l tracked per lane-group.
+// v2 structure (technique stack per the CDNA4 guide §5.5 catalog):
+//   * 8 waves, Q-block 256 rows (32/wave as 2x16 sub-tiles): K/V staging is
+//     amortized over 4x the MFMA work of the 64-row v1.
+//   * swapped QK^T - mfma(A=K, B=Q) gives S^T[kv][q] with q = lane&15, so the
+//     softmax reduction is 2 shuffles (xor 16/32) and Q stays in registers
+//     for the whole kernel (T12 structure).
+//   * reg-staged issue-early/write-late K+V prefetch (T14): global loads for
+//     tile t+1 issue before tile t's compute, drain (vmcnt 0) + LDS write +
+//     one barrier at tile end. Double-buffered LDS.
+//   * XOR chunk swizzle on K/Vt/P LDS tiles (T2): row-bit xor into the 16B
+//     chunk index keeps ds_read_b128 bank-uniform with unpadded power-of-2
+//     strides (16B alignment preserved).
+//   * defer-max rescale (T13, THR=8): skip the O/l rescale pass while the
+//     running max grows by < THR (exp headroom is harmless in f32 accum).
+//   * per-wave causal tile skip + block-level tile range; XCD-bijective
+//     workgroup remap so all q-chunks of one (batch,head) share one XCD's L2.
+//   * s_setprio around MFMA clusters (T5).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
@@ -22,18 +29,360 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define LOG2E 1.44269504088896340736f
+#define WAIT_VM0 0x0F70    // vmcnt=0, expcnt/lgkmcnt = no-wait
+#define RESCALE_THR 8.0f
+
+// 16B-chunk XOR swizzle: row-major [R][C] bf16 tile, C a multiple of 8.
+__device__ __forceinline__ int swz8(int row, int chunk) { return chunk ^ (row & 7); }
 
 template <int D>  // head dim: 64 or 128
+__global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    bf16* __restrict__ o, float* __restrict__ lse,
+    int sq, int skv, int b, int hq, int hkv, float scale,
+    int causal, int window, int nqc) {
+  constexpr int QBLK = 256, KVBLK = 64;
+  constexpr int NKD = D / 32;      // QK k-steps over d
+  constexpr int ND = D / 16;       // d sub-tiles
+  constexpr int KCH = D / 8;       // 16B chunks per K row
+  constexpr int SREG = KVBLK * D / 8 / 512;  // staged short8 per lane (2 at D=128)
+
+  __shared__ __align__(16) short K_lds[2][KVBLK][D];   // [kv][d], chunk-swizzled
+  __shared__ __align__(16) short Vt_lds[2][D][KVBLK];   // [d][kv], chunk-swizzled
+  __shared__ __align__(16) short P_lds[8][32][KVBLK];   // per-wave [q][kv], chunk-swizzled
+
+  // ---- XCD-bijective workgroup remap (T1; bijective per m204) ----
+  const int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {
+    int q8 = nwg >> 3, r8 = nwg & 7;
+    int xcd = bid & 7, idx = bid >> 3;
+    bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
+  }
+  const int bh = bid / nqc;       // batch * hq + head (contiguous per XCD)
+  const int q0 = (bid % nqc) * QBLK;
+  const int batch = bh / hq;
+  const int head = bh % hq;
+  const int kv_head = head / (hq / hkv);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int g = lane >> 4;              // 16-lane group (0..3)
+  const int c = lane & 15;
+
+  const long q_srow = (long)b * hq * D;
+  const long k_srow = (long)b * hkv * D;
+  const bf16* qbase = q + ((long)batch * hq + head) * D;
+  const bf16* kbase = k + ((long)batch * hkv + kv_head) * D;
+  const bf16* vbase = v + ((long)batch * hkv + kv_head) * D;
+  bf16* obase = o + ((long)batch * hq + head) * D;
+
+  const int q0w = q0 + wid * 32;        // this wave's first q row
+
+  // ---- Q tile -> B fragments (kept in registers for the whole kernel) ----
+  // B[col=q][k=d]: lane (g,c) holds Q[q0w + qt*16 + c][ks*32 + g*8 ..]
+  bf16x8 qb[2][NKD];
+#pragma unroll
+  for (int qt = 0; qt < 2; ++qt) {
+    int qrow = min(q0w + qt * 16 + c, sq - 1);
+    const bf16* qr = qbase + (long)qrow * q_srow;
+#pragma unroll
+    for (int ks = 0; ks < NKD; ++ks) {
+      short8 tmp = *reinterpret_cast<const short8*>(qr + ks * 32 + g * 8);
+      qb[qt][ks] = *reinterpret_cast<bf16x8*>(&tmp);
+    }
+  }
+
+  f32x4 oacc[2][ND];
+#pragma unroll
+  for (int qt = 0; qt < 2; ++qt)
+#pragma unroll
+    for (int n = 0; n < ND; ++n) oacc[qt][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float m_run[2] = {-1e30f, -1e30f}, l_run[2] = {0.f, 0.f};
+
+  const int off = skv - sq;
+  int t_end = causal ? min((q0 + QBLK - 1 + off) / KVBLK, (skv - 1) / KVBLK)
+                     : (skv - 1) / KVBLK;
+  int t_start = 0;
+  if (window > 0) t_start = max(0, (q0 + off - window + 1) / KVBLK);
+
+  // staging lane->element maps. K: row-linear (coalesced global loads,
+  // vector LDS stores). V: kv-fast (global loads scatter across rows -- the
+  // tile still covers every byte, L2 absorbs the ordering -- but the
+  // transposed scalar LDS stores then spread over all 32 banks; the
+  // row-linear map would leave only 4 distinct banks per store: 16-way
+  // conflict, measured 3.2e8 SQ_LDS_BANK_CONFLICT vs 1.6e8 SQ_BUSY).
+  int st_row[SREG], st_c8[SREG], sv_row[SREG], sv_c8[SREG];
+#pragma unroll
+  for (int i = 0; i < SREG; ++i) {
+    int idx = (int)threadIdx.x + i * 512;
+    st_row[i] = idx / KCH;
+    st_c8[i] = idx % KCH;
+    sv_row[i] = idx & (KVBLK - 1);
+    sv_c8[i] = idx / KVBLK;
+  }
+
+  short8 kreg[SREG], vreg[SREG];
+  // ---- prologue: stage tile t_start ----
+  {
+    const int k0 = t_start * KVBLK;
+#pragma unroll
+    for (int i = 0; i < SREG; ++i) {
+      int krow = k0 + st_row[i];
+      kreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (krow < skv)
+        kreg[i] = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + st_c8[i] * 8);
+      int vrow = k0 + sv_row[i];
+      if (vrow < skv)
+        vreg[i] = *reinterpret_cast<const short8*>(vbase + (long)vrow * k_srow + sv_c8[i] * 8);
+    }
+#pragma unroll
+    for (int i = 0; i < SREG; ++i) {
+      *reinterpret_cast<short8*>(&K_lds[0][st_row[i]][swz8(st_row[i], st_c8[i]) * 8]) = kreg[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = sv_c8[i] * 8 + j;
+        Vt_lds[0][d][swz8(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = vreg[i][j];
+      }
+    }
+    __syncthreads();
+  }
+
+  int cur = 0;
+  for (int t = t_start; t <= t_end; ++t) {
+    const int k0 = t * KVBLK;
+    // ---- issue next tile's global loads early (T14) ----
+    const bool have_next = (t + 1 <= t_end);
+    if (have_next) {
+      const int k0n = k0 + KVBLK;
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+        int krow = k0n + st_row[i];
+        kreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        vreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (krow < skv)
+          kreg[i] = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + st_c8[i] * 8);
+        int vrow = k0n + sv_row[i];
+        if (vrow < skv)
+          vreg[i] = *reinterpret_cast<const short8*>(vbase + (long)vrow * k_srow + sv_c8[i] * 8);
+      }
+    }
+
+    // per-wave skip: tile entirely masked for this wave's 32 q rows
+    bool wave_skip = false;
+    if (causal && k0 > q0w + 31 + off) wave_skip = true;
+    if (window > 0 && k0 + KVBLK - 1 < q0w + off - window + 1) wave_skip = true;
+
+    if (!wave_skip) {
+      // ---- S^T = scale * K Q^T : D_S[row=kv][col=q] ----
+      f32x4 st[2][KVBLK / 16];
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kvt = 0; kvt < KVBLK / 16; ++kvt) {
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) st[qt][kvt] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < NKD; ++ks) {
+          int krow = kvt * 16 + c;
+          short8 tmp = *reinterpret_cast<const short8*>(
+              &K_lds[cur][krow][swz8(krow, ks * 4 + g) * 8]);
+          bf16x8 ak = *reinterpret_cast<bf16x8*>(&tmp);
+#pragma unroll
+          for (int qt = 0; qt < 2; ++qt)
+            st[qt][kvt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak, qb[qt][ks], st[qt][kvt], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- mask + online softmax (per lane: q = q0w + qt*16 + c) ----
+      bool tile_full = (k0 + KVBLK <= skv) && (q0w + 31 < sq);
+      if (causal) tile_full &= (k0 + KVBLK - 1 <= q0w + off);
+      if (window > 0) tile_full &= (k0 >= q0w + 31 + off - window + 1);
+
+      float pmax[2] = {-1e30f, -1e30f};
+      if (tile_full) {
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt)
+#pragma unroll
+          for (int kvt = 0; kvt < KVBLK / 16; ++kvt)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              float val = st[qt][kvt][r] * scale;
+              st[qt][kvt][r] = val;
+              pmax[qt] = fmaxf(pmax[qt], val);
+            }
+      } else {
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) {
+          int qrow = q0w + qt * 16 + c;
+#pragma unroll
+          for (int kvt = 0; kvt < KVBLK / 16; ++kvt)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              int kvcol = k0 + kvt * 16 + g * 4 + r;
+              bool ok = (kvcol < skv) && (qrow < sq);
+              if (causal) ok &= (kvcol <= qrow + off);
+              if (window > 0) ok &= (kvcol > qrow + off - window);
+              float val = ok ? st[qt][kvt][r] * scale : -1e30f;
+              st[qt][kvt][r] = val;
+              pmax[qt] = fmaxf(pmax[qt], val);
+            }
+        }
+      }
+      // reduce max over the 4 lanes holding this q column (xor 16, 32)
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt) {
+        pmax[qt] = fmaxf(pmax[qt], __shfl_xor(pmax[qt], 16, 64));
+        pmax[qt] = fmaxf(pmax[qt], __shfl_xor(pmax[qt], 32, 64));
+      }
+
+      // defer-max (T13): only rescale when the max grew by > THR
+      bool need = (pmax[0] > m_run[0] + RESCALE_THR) || (pmax[1] > m_run[1] + RESCALE_THR) ||
+                  (m_run[0] == -1e30f);
+      if (__any(need)) {
+        float alpha[2];
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) {
+          float m_new = fmaxf(m_run[qt], pmax[qt]);
+          alpha[qt] = __expf(m_run[qt] - m_new);
+          m_run[qt] = m_new;
+          l_run[qt] *= alpha[qt];
+        }
+        // broadcast alpha from q=c layout to O's q=g*4+r layout
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float a = __shfl(alpha[qt], (lane & 48) + g * 4 + r, 64);
+#pragma unroll
+            for (int n = 0; n < ND; ++n) oacc[qt][n][r] *= a;
+          }
+        }
+      }
+
+      // exp + row-sum (l stays lane-partial; reduced in the epilogue)
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt) {
+        float ps = 0.f;
+#pragma unroll
+        for (int kvt = 0; kvt < KVBLK / 16; ++kvt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float sv = st[qt][kvt][r];
+            float p = exp2f((sv - m_run[qt]) * LOG2E);
+            if (sv <= -1e29f) p = 0.f;
+            st[qt][kvt][r] = p;
+            ps += p;
+          }
+        l_run[qt] += ps;
+      }
+
+      // ---- P -> LDS (pack pairs, b32 stores into swizzled chunks) ----
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt) {
+        int qrow = qt * 16 + c;
+        int* prow = reinterpret_cast<int*>(&P_lds[wid][0][0]) + qrow * (KVBLK / 2);
+#pragma unroll
+        for (int kvt = 0; kvt < KVBLK / 16; ++kvt) {
+#pragma unroll
+          for (int p = 0; p < 2; ++p) {
+            unsigned lo = (unsigned short)f2sbf(st[qt][kvt][2 * p]);
+            unsigned hi = (unsigned short)f2sbf(st[qt][kvt][2 * p + 1]);
+            int pi = kvt * 8 + g * 2 + p;           // dword index in the row
+            prow[swz8(qrow, pi >> 2) * 4 + (pi & 3)] = (int)(lo | (hi << 16));
+          }
+        }
+      }
+      // P_lds is wave-private: only this wave's LDS ops need to drain
+      __builtin_amdgcn_s_waitcnt(0xC07F /* lgkmcnt 0 only */);
+      __builtin_amdgcn_sched_barrier(0);
+
+      // ---- O += P V : D_O[row=q][col=d], A=P, B=Vt ----
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks2 = 0; ks2 < KVBLK / 32; ++ks2) {
+        bf16x8 pa[2];
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) {
+          int qrow = qt * 16 + c;
+          short8 tmp = *reinterpret_cast<const short8*>(
+              &P_lds[wid][qrow][swz8(qrow, ks2 * 4 + g) * 8]);
+          pa[qt] = *reinterpret_cast<bf16x8*>(&tmp);
+        }
+#pragma unroll
+        for (int n = 0; n < ND; ++n) {
+          int vrow = n * 16 + c;
+          short8 tmp = *reinterpret_cast<const short8*>(
+              &Vt_lds[cur][vrow][swz8(vrow, ks2 * 4 + g) * 8]);
+          bf16x8 bv = *reinterpret_cast<bf16x8*>(&tmp);
+#pragma unroll
+          for (int qt = 0; qt < 2; ++qt)
+            oacc[qt][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa[qt], bv, oacc[qt][n], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+
+    // ---- write staged tile t+1 into the other LDS buffer, flip ----
+    if (have_next) {
+      __builtin_amdgcn_s_waitcnt(WAIT_VM0);
+      __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+        *reinterpret_cast<short8*>(&K_lds[cur ^ 1][st_row[i]][swz8(st_row[i], st_c8[i]) * 8]) = kreg[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = sv_c8[i] * 8 + j;
+          Vt_lds[cur ^ 1][d][swz8(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = vreg[i][j];
+        }
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue ----
+  // finish l: reduce the lane-partials over the 4 lanes per q column
+  float linv_c[2];
+#pragma unroll
+  for (int qt = 0; qt < 2; ++qt) {
+    l_run[qt] += __shfl_xor(l_run[qt], 16, 64);
+    l_run[qt] += __shfl_xor(l_run[qt], 32, 64);
+    linv_c[qt] = (l_run[qt] > 0.f) ? 1.f / l_run[qt] : 0.f;
+    int qrow = q0w + qt * 16 + c;
+    if (g == 0 && qrow < sq)
+      lse[((long)batch * hq + head) * sq + qrow] = m_run[qt] + __logf(fmaxf(l_run[qt], 1e-30f));
+  }
+#pragma unroll
+  for (int qt = 0; qt < 2; ++qt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float inv = __shfl(linv_c[qt], (lane & 48) + g * 4 + r, 64);
+      int qrow = q0w + qt * 16 + g * 4 + r;
+      if (qrow >= sq) continue;
+      bf16* orow = obase + (long)qrow * q_srow;
+#pragma unroll
+      for (int n = 0; n < ND; ++n) orow[n * 16 + c] = f2bf(oacc[qt][n][r] * inv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// v1 kernel (64-row blocks) kept for short sequences (sq < 256).
+// ---------------------------------------------------------------------------
+
+template <int D>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     bf16* __restrict__ o, float* __restrict__ lse,
     int sq, int skv, int b, int hq, int hkv, float scale,
     int causal, int window) {
-  constexpr int QBLK = 64, KVBLK = 64;  // 4 waves x 16 q rows
-  constexpr int KPAD = D + 8;   // K_lds row stride (bf16)
+  constexpr int QBLK = 64, KVBLK = 64;
+  constexpr int KPAD = D + 8;
   constexpr int VPAD = KVBLK + 8;
-  constexpr int ND = D / 16;    // d-tiles (4 or 8)
-  constexpr int NKD = D / 32;   // k-steps over d
+  constexpr int ND = D / 16;
+  constexpr int NKD = D / 32;
 
   __shared__ short K_lds[KVBLK][KPAD];
   __shared__ short Vt_lds[D][VPAD];
@@ -42,26 +391,25 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int q0 = blockIdx.x * QBLK;
-  const int bh = blockIdx.y;            // batch * hq + head
+  const int bh = blockIdx.y;
   const int batch = bh / hq;
   const int head = bh % hq;
   const int kv_head = head / (hq / hkv);
 
-  const long q_srow = (long)b * hq * D;   // q stride along s
+  const long q_srow = (long)b * hq * D;
   const long k_srow = (long)b * hkv * D;
   const bf16* qbase = q + ((long)batch * hq + head) * D;
   const bf16* kbase = k + ((long)batch * hkv + kv_head) * D;
   const bf16* vbase = v + ((long)batch * hkv + kv_head) * D;
   bf16* obase = o + ((long)batch * hq + head) * D;
 
-  const int row0 = q0 + wid * 16;       // this wave's first q row
-  const int rg = lane >> 4;             // 16-lane group (0..3)
-  const int cl = lane & 15;             // col-in-tile / row-in-A
+  const int row0 = q0 + wid * 16;
+  const int rg = lane >> 4;
+  const int cl = lane & 15;
 
-  // ---- load Q tile into registers (A fragments per 32-wide k step) ----
   bf16x8 qa[NKD];
   {
-    int qrow = row0 + cl;  // A-frag row
+    int qrow = row0 + cl;
     const bf16* qr = qbase + (long)min(qrow, sq - 1) * q_srow;
 #pragma unroll
     for (int kd = 0; kd < NKD; ++kd) {
@@ -77,7 +425,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
 
-  // causal: query row i attends to kv <= i + (skv - sq)
   int t_end = causal ? min((q0 + QBLK - 1 + (skv - sq)) / KVBLK, (skv - 1) / KVBLK)
                      : (skv - 1) / KVBLK;
   int t_start = 0;
@@ -85,11 +432,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   for (int t = t_start; t <= t_end; ++t) {
     const int k0 = t * KVBLK;
-    // ---- stage K (row-major, coalesced) and V (transposed, r-fast) ----
     {
       constexpr int G = KVBLK * D / 8;
       for (int idx = threadIdx.x; idx < G; idx += 256) {
-        // K: coalesced loads, row-major vector stores
         int r = idx / (D / 8), c8 = idx % (D / 8);
         int krow = k0 + r;
         short8 kv8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
@@ -98,9 +443,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         *reinterpret_cast<short8*>(&K_lds[r][c8 * 8]) = kv8;
       }
       for (int idx = threadIdx.x; idx < G; idx += 256) {
-        // V: r-fast mapping -> transposed scalar stores hit ~all banks
-        // (row-major mapping would put all 16 same-j lanes in one bank:
-        //  8-row x 16B-aligned stride aliases mod 32 banks)
         int r = idx & (KVBLK - 1), c8 = idx / KVBLK;
         int krow = k0 + r;
         short8 vv8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
@@ -112,7 +454,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- S = scale * Q K^T  (4 col-tiles of 16) ----
     f32x4 s[KVBLK / 16];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -126,8 +467,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- mask + online softmax ----
-    // interior tiles need no masking (wave rows all >= row0):
     const int off = skv - sq;
     bool tile_full = (k0 + KVBLK <= skv) && (row0 + 15 < sq);
     if (causal) tile_full &= (k0 + KVBLK - 1 <= row0 + off);
@@ -161,17 +500,16 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       }
     }
     }
-    // row-reduce max over the 16 lanes of each group
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
 #pragma unroll
-      for (int off = 1; off < 16; off <<= 1) pmax[r] = fmaxf(pmax[r], __shfl_xor(pmax[r], off, 64));
+      for (int off2 = 1; off2 < 16; off2 <<= 1) pmax[r] = fmaxf(pmax[r], __shfl_xor(pmax[r], off2, 64));
     }
     float alpha[4], psum[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float m_new = fmaxf(m_run[r], pmax[r]);
-      alpha[r] = __expf(m_run[r] - m_new);   // exp(-inf - -inf)=exp(0-0) safe: m_run>=-1e30
+      alpha[r] = __expf(m_run[r] - m_new);
       m_run[r] = m_new;
       psum[r] = 0.f;
     }
@@ -191,7 +529,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float p = exp2f((s[n][r] - m_run[r]) * LOG2E);
-        if (s[n][r] <= -1e29f) p = 0.f;  // fully-masked guard
+        if (s[n][r] <= -1e29f) p = 0.f;
         s[n][r] = p;
         psum[r] += p;
       }
@@ -200,23 +538,19 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
 #pragma unroll
-      for (int off = 1; off < 16; off <<= 1) psum[r] += __shfl_xor(psum[r], off, 64);
+      for (int off2 = 1; off2 < 16; off2 <<= 1) psum[r] += __shfl_xor(psum[r], off2, 64);
       l_run[r] = l_run[r] * alpha[r] + psum[r];
 #pragma unroll
       for (int n = 0; n < ND; ++n) oacc[n][r] *= alpha[r];
     }
 
-    // ---- P to LDS (D-layout -> A-layout round trip) ----
 #pragma unroll
     for (int n = 0; n < KVBLK / 16; ++n) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) P_lds[wid][rg * 4 + r][n * 16 + cl] = f2sbf(s[n][r]);
     }
-    // P_lds is wave-private: same-wave LDS RAW needs only the compiler's
-    // lgkmcnt (same-array dependency), not a block-wide barrier.
-    __builtin_amdgcn_s_waitcnt(0 /* vmcnt0 lgkmcnt0 expcnt0 */);
+    __builtin_amdgcn_s_waitcnt(0);
 
-    // ---- O += P V ----
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < KVBLK / 32; ++ks) {
@@ -232,7 +566,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue ----
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int row = row0 + rg * 4 + r;
@@ -245,9 +578,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   }
 }
 
-// PV B-fragment note: B[k][col] with col = d index, k = kv index ->
-// V[k0+k][dcol] = Vt_lds[dcol][k]; read above as Vt_lds[n*16+cl][ks*32+rg*8+j]. OK.
-
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     bool causal, double scale, long window) {
   TORCH_CHECK(q.dim() == 4 && q.dtype() == torch::kBFloat16);
@@ -258,8 +588,24 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
   auto out = torch::empty_like(qc);
   auto lse = torch::empty({b, hq, sq}, q.options().dtype(torch::kFloat32));
-  dim3 grid((sq + 63) / 64, b * hq);
   auto stream = at::cuda::getCurrentHIPStream();
+  static const bool force_v1 = getenv("MEGATRON_AMD_ATTN_V1") != nullptr;
+  if (sq >= 256 && !force_v1) {
+    int nqc = (sq + 255) / 256;
+    dim3 grid(nqc * b * hq);
+    if (d == 128)
+      hipLaunchKernelGGL((attn_fwd_kernel_v2<128>), grid, dim3(512), 0, stream,
+                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+                         (bf16*)out.data_ptr(), lse.data_ptr<float>(),
+                         sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window, nqc);
+    else
+      hipLaunchKernelGGL((attn_fwd_kernel_v2<64>), grid, dim3(512), 0, stream,
+                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+                         (bf16*)out.data_ptr(), lse.data_ptr<float>(),
+                         sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window, nqc);
+    return {out, lse};
+  }
+  dim3 grid((sq + 63) / 64, b * hq);
   if (d == 128)
     hipLaunchKernelGGL((attn_fwd_kernel<128>), grid, dim3(256), 0, stream,
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
