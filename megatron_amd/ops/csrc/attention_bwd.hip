@@ -1,14 +1,19 @@
 // Fused attention BACKWARD for CDNA4 (K1).  Atomic-free two-kernel split
 // (the FA2-style trade: P/dS recomputed in both kernels, no dq atomics):
-//   bwd_dq :  block per (q-tile, b*hq) — dq = dS K
-//   bwd_dkv:  block per (kv-tile, b*hkv), q-heads of the GQA group looped
+//   bwd_dq :  block per (128 q rows, b*hq)  — dq = dS K
+//   bwd_dkv:  block per (128 kv rows, b*hkv), q-heads of the GQA group looped
 //             inside — dv = P^T dO ; dk = dS^T Q
 // plus a rowsum preprocess  Drow = sum_d(dO * O).
 //
-// Swapped-operand trick throughout (guide §B): computing the TRANSPOSED
-// score tile ST = mfma(K_a, Q_b) lets both operands come from row-major
-// loads, since the B-fragment of X^T reads the same per-lane elements as
-// the A-fragment of X.
+// v2 structure (same technique stack as attention_fwd v2):
+//   * swapped-operand MFMAs so every LDS operand is consumed as an A-fragment
+//     of a row-major tile; the P/dS D-fragment -> A-fragment relayout is two
+//     register-pair permlane swaps (dfrag_to_at, common.h) - no LDS round-trip.
+//   * single-staged swizzled LDS tiles per iteration tile, reg-staged
+//     issue-early/write-late prefetch (T14), two barriers per tile.
+//   * separate staging lane maps: row-linear (coalesced) for row-major
+//     buffers, kv-fast for transposed buffers (bank-spread scalar stores).
+//   * per-wave causal/window tile skip; scale folded into one fma per exp.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
@@ -17,6 +22,9 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define LOG2E 1.44269504088896340736f
+#define WAIT_VM0 0x0F70
+
+__device__ __forceinline__ int swzb(int row, int chunk) { return chunk ^ (row & 7); }
 
 // ---------------------------------------------------------------------------
 // Drow[b, h, s] = sum_d dO[s,b,h,:] * O[s,b,h,:]
@@ -37,33 +45,36 @@ __global__ void attn_bwd_rowsum_kernel(const short8* __restrict__ dout, const sh
     }
     acc = wave_reduce_sum(acc);
     if (lane == 0) {
-      long s = row / bh, rem = row % bh;  // row = s*bh + (b*h... actually b_then_h)
-      drow[rem * sq + s] = acc;           // out layout [b*h, s]
+      long s = row / bh, rem = row % bh;
+      drow[rem * sq + s] = acc;
     }
   }
 }
 
 // ---------------------------------------------------------------------------
-// dq kernel
+// dq kernel: 8 waves, 128 q rows per block (16 per wave), kv tiles of 64.
+//   ST  = mfma(A=K_lds,  B=Q_regs)   D[kv][q]
+//   dPT = mfma(A=V_lds,  B=dO_regs)  D[kv][q]
+//   dq += mfma(A=dS_frag(permlane),  B=Kt_lds)
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ drow,
     bf16* __restrict__ dq,
     int sq, int skv, int b, int hq, int hkv, float scale, int causal, int window) {
   constexpr int KVBLK = 64;
-  constexpr int RPAD = D + 8;      // row-major row stride
-  constexpr int TPAD = KVBLK + 8;  // transposed row stride
   constexpr int ND = D / 16, NKD = D / 32;
+  constexpr int KCH = D / 8;
+  constexpr int SREG = KVBLK * D / 8 / 512;
 
-  __shared__ short K_lds[KVBLK][RPAD];   // row-major (ST A-operand)
-  __shared__ short Kt_lds[D][TPAD];      // transposed (dq B-operand)
-  __shared__ short V_lds[KVBLK][RPAD];   // row-major (dPT A-operand)
-  __shared__ short dS_lds[4][16][TPAD];  // per-wave dS^T->dS round-trip
+  __shared__ __align__(16) short K_lds[KVBLK][D];   // row-major, swizzled
+  __shared__ __align__(16) short V_lds[KVBLK][D];   // row-major, swizzled
+  __shared__ __align__(16) short Kt_lds[D][KVBLK];  // transposed, swizzled
 
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-  const int q0 = blockIdx.x * 64;
+  const int g = lane >> 4, c = lane & 15;
+  const int q0 = blockIdx.x * 128;
   const int batch = blockIdx.y / hq, head = blockIdx.y % hq;
   const int kv_head = head / (hq / hkv);
   const long q_srow = (long)b * hq * D, k_srow = (long)b * hkv * D;
@@ -75,118 +86,167 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const float* lse_row = lse + ((long)batch * hq + head) * sq;
   const float* dr_row = drow + ((long)batch * hq + head) * sq;
 
-  const int row0 = q0 + wid * 16, rg = lane >> 4, cl = lane & 15;
+  const int row0 = q0 + wid * 16;
   const int off = skv - sq;
 
   bf16x8 qa[NKD], doa[NKD];
   {
-    int qrow = min(row0 + cl, sq - 1);
+    int qrow = min(row0 + c, sq - 1);
     const bf16* qr = qbase + (long)qrow * q_srow;
     const bf16* dor = dobase + (long)qrow * q_srow;
 #pragma unroll
     for (int kd = 0; kd < NKD; ++kd) {
-      short8 t1 = *reinterpret_cast<const short8*>(qr + kd * 32 + rg * 8);
+      short8 t1 = *reinterpret_cast<const short8*>(qr + kd * 32 + g * 8);
       qa[kd] = *reinterpret_cast<bf16x8*>(&t1);
-      short8 t2 = *reinterpret_cast<const short8*>(dor + kd * 32 + rg * 8);
+      short8 t2 = *reinterpret_cast<const short8*>(dor + kd * 32 + g * 8);
       doa[kd] = *reinterpret_cast<bf16x8*>(&t2);
     }
   }
-  float my_lse = lse_row[min(row0 + cl, sq - 1)];
-  float my_dr = dr_row[min(row0 + cl, sq - 1)];
+  const float c1 = scale * LOG2E;
+  const float my_lse_l2 = lse_row[min(row0 + c, sq - 1)] * LOG2E;
+  const float my_dr_s = dr_row[min(row0 + c, sq - 1)] * scale;
 
   f32x4 dq_acc[ND];
 #pragma unroll
   for (int n = 0; n < ND; ++n) dq_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  int t_end = causal ? min((q0 + 63 + off) / KVBLK, (skv - 1) / KVBLK) : (skv - 1) / KVBLK;
+  int t_end = causal ? min((q0 + 127 + off) / KVBLK, (skv - 1) / KVBLK) : (skv - 1) / KVBLK;
   int t_start = (window > 0) ? max(0, (q0 + off - window + 1) / KVBLK) : 0;
+
+  int st_row[SREG], st_c8[SREG], sv_row[SREG], sv_c8[SREG];
+#pragma unroll
+  for (int i = 0; i < SREG; ++i) {
+    int idx = (int)threadIdx.x + i * 512;
+    st_row[i] = idx / KCH;
+    st_c8[i] = idx % KCH;
+    sv_row[i] = idx & (KVBLK - 1);
+    sv_c8[i] = idx / KVBLK;
+  }
+
+  short8 kreg[SREG], vreg[SREG], ktreg[SREG];
+  auto stage_loads = [&](int k0) {
+#pragma unroll
+    for (int i = 0; i < SREG; ++i) {
+      int krow = k0 + st_row[i];
+      kreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      ktreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (krow < skv) {
+        kreg[i] = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + st_c8[i] * 8);
+        vreg[i] = *reinterpret_cast<const short8*>(vbase + (long)krow * k_srow + st_c8[i] * 8);
+      }
+      int trow = k0 + sv_row[i];
+      if (trow < skv)
+        ktreg[i] = *reinterpret_cast<const short8*>(kbase + (long)trow * k_srow + sv_c8[i] * 8);
+    }
+  };
+  auto stage_writes = [&]() {
+#pragma unroll
+    for (int i = 0; i < SREG; ++i) {
+      *reinterpret_cast<short8*>(&K_lds[st_row[i]][swzb(st_row[i], st_c8[i]) * 8]) = kreg[i];
+      *reinterpret_cast<short8*>(&V_lds[st_row[i]][swzb(st_row[i], st_c8[i]) * 8]) = vreg[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = sv_c8[i] * 8 + j;
+        Kt_lds[d][swzb(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = ktreg[i][j];
+      }
+    }
+  };
+
+  stage_loads(t_start * KVBLK);
+  stage_writes();
+  __syncthreads();
 
   for (int t = t_start; t <= t_end; ++t) {
     const int k0 = t * KVBLK;
-    {
-      constexpr int G = KVBLK * D / 8;
-      for (int idx = threadIdx.x; idx < G; idx += 256) {
-        int r = idx / (D / 8), c8 = idx % (D / 8);
-        int krow = k0 + r;
-        short8 kv8 = short8{0, 0, 0, 0, 0, 0, 0, 0}, vv8 = kv8;
-        if (krow < skv) {
-          kv8 = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + c8 * 8);
-          vv8 = *reinterpret_cast<const short8*>(vbase + (long)krow * k_srow + c8 * 8);
+    const bool have_next = (t + 1 <= t_end);
+    if (have_next) stage_loads(k0 + KVBLK);
+
+    bool wave_skip = (causal && k0 > row0 + 15 + off) ||
+                     (window > 0 && k0 + KVBLK - 1 < row0 + off - window + 1);
+    if (!wave_skip) {
+      f32x4 st[KVBLK / 16], dpt[KVBLK / 16];
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int n = 0; n < KVBLK / 16; ++n) {
+        st[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+        dpt[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+        int krow = n * 16 + c;
+#pragma unroll
+        for (int kd = 0; kd < NKD; ++kd) {
+          short8 t1 = *reinterpret_cast<const short8*>(&K_lds[krow][swzb(krow, kd * 4 + g) * 8]);
+          bf16x8 ak = *reinterpret_cast<bf16x8*>(&t1);
+          short8 t2 = *reinterpret_cast<const short8*>(&V_lds[krow][swzb(krow, kd * 4 + g) * 8]);
+          bf16x8 av = *reinterpret_cast<bf16x8*>(&t2);
+          st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak, qa[kd], st[n], 0, 0, 0);
+          dpt[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, doa[kd], dpt[n], 0, 0, 0);
         }
-        *reinterpret_cast<short8*>(&K_lds[r][c8 * 8]) = kv8;
-        *reinterpret_cast<short8*>(&V_lds[r][c8 * 8]) = vv8;
       }
-      // transposed K: r-fast mapping avoids the 8-row-stride bank aliasing
-      for (int idx = threadIdx.x; idx < G; idx += 256) {
-        int r = idx & (KVBLK - 1), c8 = idx / KVBLK;
-        int krow = k0 + r;
-        short8 kv8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
-        if (krow < skv)
-          kv8 = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + c8 * 8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) Kt_lds[c8 * 8 + j][r] = kv8[j];
-      }
-    }
-    __syncthreads();
+      __builtin_amdgcn_s_setprio(0);
 
-    // per kv-subtile n: ST = K Q^T ; PT ; dPT = V dO^T ; dST -> dS_lds
+      // P, dS in D-layout [kv = g*4+r][q = c]; pack dS pairs along kv
+      int pk[KVBLK / 16][2];
+      const int qrow = row0 + c;
 #pragma unroll
-    for (int n = 0; n < KVBLK / 16; ++n) {
-      f32x4 st = f32x4{0.f, 0.f, 0.f, 0.f};
-      f32x4 dpt = f32x4{0.f, 0.f, 0.f, 0.f};
+      for (int n = 0; n < KVBLK / 16; ++n) {
+        float ds[4];
 #pragma unroll
-      for (int kd = 0; kd < NKD; ++kd) {
-        bf16x8 ak = *reinterpret_cast<const bf16x8*>(&K_lds[n * 16 + cl][kd * 32 + rg * 8]);
-        bf16x8 av = *reinterpret_cast<const bf16x8*>(&V_lds[n * 16 + cl][kd * 32 + rg * 8]);
-        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak, qa[kd], st, 0, 0, 0);
-        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, doa[kd], dpt, 0, 0, 0);
+        for (int r = 0; r < 4; ++r) {
+          int kvrow = k0 + n * 16 + g * 4 + r;
+          bool ok = (kvrow < skv) && (qrow < sq);
+          if (causal) ok &= (kvrow <= qrow + off);
+          if (window > 0) ok &= (kvrow > qrow + off - window);
+          float pt = ok ? exp2f(__builtin_fmaf(st[n][r], c1, -my_lse_l2)) : 0.f;
+          ds[r] = pt * __builtin_fmaf(dpt[n][r], scale, -my_dr_s);
+        }
+        pk[n][0] = pack_bf16x2(ds[0], ds[1]);
+        pk[n][1] = pack_bf16x2(ds[2], ds[3]);
       }
-      // D-layout: row = kv (rg*4+r), col = q (cl); this wave's q cols are
-      // row0+cl, matching the my_lse/my_dr loads above.
-      int qrow = row0 + cl;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int kvrow = k0 + n * 16 + rg * 4 + r;
-        bool ok = (kvrow < skv) && (qrow < sq);
-        if (causal) ok &= (kvrow <= qrow + off);
-        if (window > 0) ok &= (kvrow > qrow + off - window);
-        // lse/drow are per q-row: q index here is cl, but my_lse was loaded
-        // with index row0+cl on THIS lane -> matches col ✓
-        float pt = ok ? exp2f((st[r] * scale - my_lse) * LOG2E) : 0.f;
-        float dst = pt * (dpt[r] - my_dr) * scale;
-        dS_lds[wid][cl][n * 16 + rg * 4 + r] = f2sbf(dst);
-      }
-    }
-    __syncthreads();
 
-    // dq += dS K   (A = dS row-major from dS_lds, B = K^T from Kt_lds)
-    __builtin_amdgcn_s_setprio(1);
+      // dq += dS K : A = dS[q][kv] (permlane relayout), B = Kt
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int ks = 0; ks < KVBLK / 32; ++ks) {
-      short8 at = *reinterpret_cast<const short8*>(&dS_lds[wid][cl][ks * 32 + rg * 8]);
-      bf16x8 a = *reinterpret_cast<bf16x8*>(&at);
+      for (int ks2 = 0; ks2 < KVBLK / 32; ++ks2) {
+        int av4[4];
+        dfrag_to_at(pk[2 * ks2], pk[2 * ks2 + 1], av4);
+        int4 av = make_int4(av4[0], av4[1], av4[2], av4[3]);
+        bf16x8 a = *reinterpret_cast<bf16x8*>(&av);
 #pragma unroll
-      for (int n = 0; n < ND; ++n) {
-        bf16x8 bK = *reinterpret_cast<const bf16x8*>(&Kt_lds[n * 16 + cl][ks * 32 + rg * 8]);
-        dq_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bK, dq_acc[n], 0, 0, 0);
+        for (int n = 0; n < ND; ++n) {
+          int vrow = n * 16 + c;
+          short8 t3 = *reinterpret_cast<const short8*>(&Kt_lds[vrow][swzb(vrow, ks2 * 4 + g) * 8]);
+          bf16x8 bK = *reinterpret_cast<bf16x8*>(&t3);
+          dq_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bK, dq_acc[n], 0, 0, 0);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
-    __syncthreads();
+
+    if (have_next) {
+      __syncthreads();
+      __builtin_amdgcn_s_waitcnt(WAIT_VM0);
+      __builtin_amdgcn_sched_barrier(0);
+      stage_writes();
+      __syncthreads();
+    }
   }
 
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    int row = row0 + rg * 4 + r;
+    int row = row0 + g * 4 + r;
     if (row >= sq) continue;
     bf16* dqr = dqbase + (long)row * q_srow;
 #pragma unroll
-    for (int n = 0; n < ND; ++n) dqr[n * 16 + cl] = f2bf(dq_acc[n][r]);
+    for (int n = 0; n < ND; ++n) dqr[n * 16 + c] = f2bf(dq_acc[n][r]);
   }
 }
 
 // ---------------------------------------------------------------------------
-// dk/dv kernel
+// dk/dv kernel: 8 waves, 128 kv rows per block (16 per wave), q tiles of 64.
+//   S  = mfma(A=Q_lds,  B=K_regs)   D[q][kv]
+//   dP = mfma(A=dO_lds, B=V_regs)   D[q][kv]
+//   dV += mfma(A=P^T_frag(permlane),  B=dOt_lds)
+//   dK += mfma(A=dS^T_frag(permlane), B=Qt_lds)
 // ---------------------------------------------------------------------------
 template <int D>
 __global__ __launch_bounds__(512) void attn_bwd_dkv_kernel(
@@ -195,15 +255,19 @@ __global__ __launch_bounds__(512) void attn_bwd_dkv_kernel(
     bf16* __restrict__ dk, bf16* __restrict__ dv,
     int sq, int skv, int b, int hq, int hkv, float scale, int causal, int window) {
   constexpr int QBLK = 64;
-  constexpr int RPAD = D + 8;
-  constexpr int TPAD = QBLK + 8;
   constexpr int ND = D / 16, NKD = D / 32;
+  constexpr int KCH = D / 8;
+  constexpr int SREG = QBLK * D / 8 / 512;
 
-  __shared__ short BufB[QBLK][RPAD];   // row-major Q (phase 1) then dO (phase 3)
-  __shared__ short BufA[D][TPAD];      // transposed dO (phase 2) then Q^T (phase 4)
-  __shared__ short P_lds[8][16][TPAD]; // per-wave PT / dST round-trips
+  __shared__ __align__(16) short Q_lds[QBLK][D];
+  __shared__ __align__(16) short dO_lds[QBLK][D];
+  __shared__ __align__(16) short Qt_lds[D][QBLK];
+  __shared__ __align__(16) short dOt_lds[D][QBLK];
+  __shared__ float lse_s[QBLK];   // lse * LOG2E for the staged q tile
+  __shared__ float dr_s[QBLK];    // drow * scale
 
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int g = lane >> 4, c = lane & 15;
   const int k0 = blockIdx.x * 128;  // 8 waves x 16 kv rows
   const int batch = blockIdx.y / hkv, kv_head = blockIdx.y % hkv;
   const int rep = hq / hkv;
@@ -213,19 +277,21 @@ __global__ __launch_bounds__(512) void attn_bwd_dkv_kernel(
   bf16* dkbase = dk + ((long)batch * hkv + kv_head) * D;
   bf16* dvbase = dv + ((long)batch * hkv + kv_head) * D;
 
-  const int krow0 = k0 + wid * 16, rg = lane >> 4, cl = lane & 15;
+  const int krow0 = k0 + wid * 16;
   const int off = skv - sq;
+  const float c1 = scale * LOG2E;
 
+  // K/V B-fragments: lane (g,c) holds row krow0 + c (= B col c)
   bf16x8 ka[NKD], va[NKD];
   {
-    int krow = min(krow0 + cl, skv - 1);
+    int krow = min(krow0 + c, skv - 1);
     const bf16* kr = kbase + (long)krow * k_srow;
     const bf16* vr = vbase + (long)krow * k_srow;
 #pragma unroll
     for (int kd = 0; kd < NKD; ++kd) {
-      short8 t1 = *reinterpret_cast<const short8*>(kr + kd * 32 + rg * 8);
+      short8 t1 = *reinterpret_cast<const short8*>(kr + kd * 32 + g * 8);
       ka[kd] = *reinterpret_cast<bf16x8*>(&t1);
-      short8 t2 = *reinterpret_cast<const short8*>(vr + kd * 32 + rg * 8);
+      short8 t2 = *reinterpret_cast<const short8*>(vr + kd * 32 + g * 8);
       va[kd] = *reinterpret_cast<bf16x8*>(&t2);
     }
   }
@@ -239,9 +305,16 @@ __global__ __launch_bounds__(512) void attn_bwd_dkv_kernel(
 
   int tq_start = causal ? max(0, (k0 - off) / QBLK) : 0;
   int tq_end = (sq - 1) / QBLK;
-  if (window > 0) {
-    // kv row k attends from q rows >= k - off ... < k - off + window
-    tq_end = min(tq_end, (k0 + 127 - off + window - 1) / QBLK);
+  if (window > 0) tq_end = min(tq_end, (k0 + 127 - off + window - 1) / QBLK);
+
+  int st_row[SREG], st_c8[SREG], sv_row[SREG], sv_c8[SREG];
+#pragma unroll
+  for (int i = 0; i < SREG; ++i) {
+    int idx = (int)threadIdx.x + i * 512;
+    st_row[i] = idx / KCH;
+    st_c8[i] = idx % KCH;
+    sv_row[i] = idx & (QBLK - 1);
+    sv_c8[i] = idx / QBLK;
   }
 
   for (int hg = 0; hg < rep; ++hg) {
@@ -251,140 +324,146 @@ __global__ __launch_bounds__(512) void attn_bwd_dkv_kernel(
     const float* lse_row = lse + ((long)batch * hq + head) * sq;
     const float* dr_row = drow + ((long)batch * hq + head) * sq;
 
+    short8 qrm[SREG], dorm[SREG], qtr[SREG], dotr[SREG];
+    auto stage_loads = [&](int qt0) {
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+        int r1 = qt0 + st_row[i];
+        qrm[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        dorm[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        qtr[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        dotr[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (r1 < sq) {
+          qrm[i] = *reinterpret_cast<const short8*>(qbase + (long)r1 * q_srow + st_c8[i] * 8);
+          dorm[i] = *reinterpret_cast<const short8*>(dobase + (long)r1 * q_srow + st_c8[i] * 8);
+        }
+        int r2 = qt0 + sv_row[i];
+        if (r2 < sq) {
+          qtr[i] = *reinterpret_cast<const short8*>(qbase + (long)r2 * q_srow + sv_c8[i] * 8);
+          dotr[i] = *reinterpret_cast<const short8*>(dobase + (long)r2 * q_srow + sv_c8[i] * 8);
+        }
+      }
+    };
+    auto stage_writes = [&](int qt0) {
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+        *reinterpret_cast<short8*>(&Q_lds[st_row[i]][swzb(st_row[i], st_c8[i]) * 8]) = qrm[i];
+        *reinterpret_cast<short8*>(&dO_lds[st_row[i]][swzb(st_row[i], st_c8[i]) * 8]) = dorm[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = sv_c8[i] * 8 + j;
+          Qt_lds[d][swzb(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = qtr[i][j];
+          dOt_lds[d][swzb(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = dotr[i][j];
+        }
+      }
+      if (threadIdx.x < QBLK) {
+        int qi = min(qt0 + (int)threadIdx.x, sq - 1);
+        lse_s[threadIdx.x] = lse_row[qi] * LOG2E;
+        dr_s[threadIdx.x] = dr_row[qi] * scale;
+      }
+    };
+
+    stage_loads(tq_start * QBLK);
+    stage_writes(tq_start * QBLK);
+    __syncthreads();
+
     for (int t = tq_start; t <= tq_end; ++t) {
       const int qt0 = t * QBLK;
-      // ---- phase 1: stage Q row-major + dO transposed ----
-      {
-        constexpr int G = QBLK * D / 8;
-        for (int idx = threadIdx.x; idx < G; idx += 256) {
-          int r = idx / (D / 8), c8 = idx % (D / 8);
-          int qrow = qt0 + r;
-          short8 q8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
-          if (qrow < sq)
-            q8 = *reinterpret_cast<const short8*>(qbase + (long)qrow * q_srow + c8 * 8);
-          *reinterpret_cast<short8*>(&BufB[r][c8 * 8]) = q8;
-        }
-        for (int idx = threadIdx.x; idx < G; idx += 512) {
-          int r = idx & (QBLK - 1), c8 = idx / QBLK;
-          int qrow = qt0 + r;
-          short8 do8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
-          if (qrow < sq)
-            do8 = *reinterpret_cast<const short8*>(dobase + (long)qrow * q_srow + c8 * 8);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) BufA[c8 * 8 + j][r] = do8[j];
-        }
-      }
-      __syncthreads();
+      const bool have_next = (t + 1 <= tq_end);
+      if (have_next) stage_loads(qt0 + QBLK);
 
-      // ---- phase 2: ST = K Q^T ; PT ; dV += PT dO (via BufA=dOt) ----
-      f32x4 pt[QBLK / 16];
+      // this wave's kv rows are krow0..+15; skip fully-masked q tiles
+      bool wave_skip = (causal && qt0 + QBLK - 1 < krow0 - off) ||
+                       (window > 0 && qt0 >= krow0 + 15 - off + window);
+      if (!wave_skip) {
+        f32x4 st[QBLK / 16], dpt[QBLK / 16];
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int n = 0; n < QBLK / 16; ++n) {
-        f32x4 st = f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int n = 0; n < QBLK / 16; ++n) {
+          st[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+          dpt[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+          int qrow = n * 16 + c;
 #pragma unroll
-        for (int kd = 0; kd < NKD; ++kd) {
-          bf16x8 bq = *reinterpret_cast<const bf16x8*>(&BufB[n * 16 + cl][kd * 32 + rg * 8]);
-          st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kd], bq, st, 0, 0, 0);
+          for (int kd = 0; kd < NKD; ++kd) {
+            short8 t1 = *reinterpret_cast<const short8*>(&Q_lds[qrow][swzb(qrow, kd * 4 + g) * 8]);
+            bf16x8 aq = *reinterpret_cast<bf16x8*>(&t1);
+            short8 t2 = *reinterpret_cast<const short8*>(&dO_lds[qrow][swzb(qrow, kd * 4 + g) * 8]);
+            bf16x8 ado = *reinterpret_cast<bf16x8*>(&t2);
+            st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq, ka[kd], st[n], 0, 0, 0);
+            dpt[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado, va[kd], dpt[n], 0, 0, 0);
+          }
         }
-        // D-layout: row = kv (krow0 + rg*4+r), col = q (qt0 + n*16 + cl)
-        // BUT my_lse was loaded at q index qt0+cl — per-n q col is qt0+n*16+cl!
-        float lse_n = lse_row[min(qt0 + n * 16 + cl, sq - 1)];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int kvrow = krow0 + rg * 4 + r;
-          int qrow = qt0 + n * 16 + cl;
-          bool ok = (kvrow < skv) && (qrow < sq);
-          if (causal) ok &= (kvrow <= qrow + off);
-          if (window > 0) ok &= (kvrow > qrow + off - window);
-          pt[n][r] = ok ? exp2f((st[r] * scale - lse_n) * LOG2E) : 0.f;
-          P_lds[wid][rg * 4 + r][n * 16 + cl] = f2sbf(pt[n][r]);
-        }
-      }
-      __syncthreads();
+        __builtin_amdgcn_s_setprio(0);
 
-      // dV[kv, d] += PT[kv, q] x dO[q, d]  (A from P_lds, B from BufA=dOt)
-      __builtin_amdgcn_s_setprio(1);
+        // P, dS in D-layout [q = g*4+r][kv = c]; kv col = krow0 + c
+        int pkP[QBLK / 16][2], pkD[QBLK / 16][2];
+        const int kvcol = krow0 + c;
 #pragma unroll
-      for (int ks = 0; ks < QBLK / 32; ++ks) {
-        short8 at = *reinterpret_cast<const short8*>(&P_lds[wid][cl][ks * 32 + rg * 8]);
-        bf16x8 a = *reinterpret_cast<bf16x8*>(&at);
+        for (int n = 0; n < QBLK / 16; ++n) {
+          float pv[4], dsv[4];
 #pragma unroll
-        for (int n = 0; n < ND; ++n) {
-          bf16x8 bdo = *reinterpret_cast<const bf16x8*>(&BufA[n * 16 + cl][ks * 32 + rg * 8]);
-          dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, dv_acc[n], 0, 0, 0);
+          for (int r = 0; r < 4; ++r) {
+            int qi = n * 16 + g * 4 + r;
+            int qrow = qt0 + qi;
+            bool ok = (kvcol < skv) && (qrow < sq);
+            if (causal) ok &= (kvcol <= qrow + off);
+            if (window > 0) ok &= (kvcol > qrow + off - window);
+            float pt = ok ? exp2f(__builtin_fmaf(st[n][r], c1, -lse_s[qi])) : 0.f;
+            pv[r] = pt;
+            dsv[r] = pt * __builtin_fmaf(dpt[n][r], scale, -dr_s[qi]);
+          }
+          pkP[n][0] = pack_bf16x2(pv[0], pv[1]);
+          pkP[n][1] = pack_bf16x2(pv[2], pv[3]);
+          pkD[n][0] = pack_bf16x2(dsv[0], dsv[1]);
+          pkD[n][1] = pack_bf16x2(dsv[2], dsv[3]);
         }
-      }
-      __builtin_amdgcn_s_setprio(0);
-      __syncthreads();
 
-      // ---- phase 3: restage BufB <- dO row-major ; BufA <- Q^T ----
-      {
-        constexpr int G = QBLK * D / 8;
-        for (int idx = threadIdx.x; idx < G; idx += 256) {
-          int r = idx / (D / 8), c8 = idx % (D / 8);
-          int qrow = qt0 + r;
-          short8 do8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
-          if (qrow < sq)
-            do8 = *reinterpret_cast<const short8*>(dobase + (long)qrow * q_srow + c8 * 8);
-          *reinterpret_cast<short8*>(&BufB[r][c8 * 8]) = do8;
-        }
-        for (int idx = threadIdx.x; idx < G; idx += 512) {
-          int r = idx & (QBLK - 1), c8 = idx / QBLK;
-          int qrow = qt0 + r;
-          short8 q8 = short8{0, 0, 0, 0, 0, 0, 0, 0};
-          if (qrow < sq)
-            q8 = *reinterpret_cast<const short8*>(qbase + (long)qrow * q_srow + c8 * 8);
+        // dV += P^T dO ; dK += dS^T Q   (A via permlane, B from transposed LDS)
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) BufA[c8 * 8 + j][r] = q8[j];
+        for (int ks2 = 0; ks2 < QBLK / 32; ++ks2) {
+          int avP[4], avD[4];
+          dfrag_to_at(pkP[2 * ks2], pkP[2 * ks2 + 1], avP);
+          dfrag_to_at(pkD[2 * ks2], pkD[2 * ks2 + 1], avD);
+          int4 aP4 = make_int4(avP[0], avP[1], avP[2], avP[3]);
+          int4 aD4 = make_int4(avD[0], avD[1], avD[2], avD[3]);
+          bf16x8 aP = *reinterpret_cast<bf16x8*>(&aP4);
+          bf16x8 aD = *reinterpret_cast<bf16x8*>(&aD4);
+#pragma unroll
+          for (int n = 0; n < ND; ++n) {
+            int drow_ = n * 16 + c;
+            short8 t3 = *reinterpret_cast<const short8*>(&dOt_lds[drow_][swzb(drow_, ks2 * 4 + g) * 8]);
+            bf16x8 bdo = *reinterpret_cast<bf16x8*>(&t3);
+            short8 t4 = *reinterpret_cast<const short8*>(&Qt_lds[drow_][swzb(drow_, ks2 * 4 + g) * 8]);
+            bf16x8 bq = *reinterpret_cast<bf16x8*>(&t4);
+            dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aP, bdo, dv_acc[n], 0, 0, 0);
+            dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aD, bq, dk_acc[n], 0, 0, 0);
+          }
         }
+        __builtin_amdgcn_s_setprio(0);
       }
-      __syncthreads();
 
-      // ---- phase 4: dPT = V dO^T (B from BufB=dO row-major); dST; dK += dST Q ----
-#pragma unroll
-      for (int n = 0; n < QBLK / 16; ++n) {
-        f32x4 dpt = f32x4{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int kd = 0; kd < NKD; ++kd) {
-          bf16x8 bdo = *reinterpret_cast<const bf16x8*>(&BufB[n * 16 + cl][kd * 32 + rg * 8]);
-          dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kd], bdo, dpt, 0, 0, 0);
-        }
-        float dr_n = dr_row[min(qt0 + n * 16 + cl, sq - 1)];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float dst = pt[n][r] * (dpt[r] - dr_n) * scale;
-          P_lds[wid][rg * 4 + r][n * 16 + cl] = f2sbf(dst);
-        }
+      if (have_next) {
+        __syncthreads();
+        __builtin_amdgcn_s_waitcnt(WAIT_VM0);
+        __builtin_amdgcn_sched_barrier(0);
+        stage_writes(qt0 + QBLK);
+        __syncthreads();
       }
-      __syncthreads();
-
-      // dK[kv, d] += dST[kv, q] x Q[q, d]  (A from P_lds, B from BufA=Qt)
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int ks = 0; ks < QBLK / 32; ++ks) {
-        short8 at = *reinterpret_cast<const short8*>(&P_lds[wid][cl][ks * 32 + rg * 8]);
-        bf16x8 a = *reinterpret_cast<bf16x8*>(&at);
-#pragma unroll
-        for (int n = 0; n < ND; ++n) {
-          bf16x8 bq = *reinterpret_cast<const bf16x8*>(&BufA[n * 16 + cl][ks * 32 + rg * 8]);
-          dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bq, dk_acc[n], 0, 0, 0);
-        }
-      }
-      __builtin_amdgcn_s_setprio(0);
-      __syncthreads();
     }
+    __syncthreads();  // hg boundary: next head restages from the prologue
   }
 
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    int row = krow0 + rg * 4 + r;
+    int row = krow0 + g * 4 + r;
     if (row >= skv) continue;
     bf16* dkr = dkbase + (long)row * k_srow;
     bf16* dvr = dvbase + (long)row * k_srow;
 #pragma unroll
     for (int n = 0; n < ND; ++n) {
-      dkr[n * 16 + cl] = f2bf(dk_acc[n][r]);
-      dvr[n * 16 + cl] = f2bf(dv_acc[n][r]);
+      dkr[n * 16 + c] = f2bf(dk_acc[n][r]);
+      dvr[n * 16 + c] = f2bf(dv_acc[n][r]);
     }
   }
 }
@@ -411,10 +490,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                      (const short8*)doc.data_ptr(), (const short8*)oc.data_ptr(),
                      drow.data_ptr<float>(), rows, b * hq, sq, d8);
 
-  dim3 grid_dq((sq + 63) / 64, b * hq);
+  dim3 grid_dq((sq + 127) / 128, b * hq);
   dim3 grid_dkv((skv + 127) / 128, b * hkv);
   if (d == 128) {
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid_dq, dim3(256), 0, stream,
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid_dq, dim3(512), 0, stream,
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
                        (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
                        (bf16*)dq.data_ptr(), sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
@@ -424,7 +503,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
                        sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
   } else {
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), grid_dq, dim3(256), 0, stream,
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), grid_dq, dim3(512), 0, stream,
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
                        (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
                        (bf16*)dq.data_ptr(), sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);

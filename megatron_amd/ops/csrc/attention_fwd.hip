@@ -35,7 +35,9 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 // 16B-chunk XOR swizzle: row-major [R][C] bf16 tile, C a multiple of 8.
 __device__ __forceinline__ int swz8(int row, int chunk) { return chunk ^ (row & 7); }
 
-template <int D>  // head dim: 64 or 128
+// ABL: 0=full, 1=no-softmax, 2=no-PV, 3=no-staging (ablation probes; wrong
+// numerics for 1-3, perf triage only)
+template <int D, int ABL = 0>
 __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     bf16* __restrict__ o, float* __restrict__ lse,
@@ -47,9 +49,10 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
   constexpr int KCH = D / 8;       // 16B chunks per K row
   constexpr int SREG = KVBLK * D / 8 / 512;  // staged short8 per lane (2 at D=128)
 
-  __shared__ __align__(16) short K_lds[2][KVBLK][D];   // [kv][d], chunk-swizzled
-  __shared__ __align__(16) short Vt_lds[2][D][KVBLK];   // [d][kv], chunk-swizzled
-  __shared__ __align__(16) short P_lds[8][32][KVBLK];   // per-wave [q][kv], chunk-swizzled
+  // single-buffered: the T14 staging registers are the second buffer; the
+  // extra barrier costs less than the halved occupancy of 2x LDS buffers
+  __shared__ __align__(16) short K_lds[KVBLK][D];      // [kv][d], chunk-swizzled
+  __shared__ __align__(16) short Vt_lds[D][KVBLK];     // [d][kv], chunk-swizzled
 
   // ---- XCD-bijective workgroup remap (T1; bijective per m204) ----
   const int nwg = gridDim.x;
@@ -139,21 +142,20 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
     }
 #pragma unroll
     for (int i = 0; i < SREG; ++i) {
-      *reinterpret_cast<short8*>(&K_lds[0][st_row[i]][swz8(st_row[i], st_c8[i]) * 8]) = kreg[i];
+      *reinterpret_cast<short8*>(&K_lds[st_row[i]][swz8(st_row[i], st_c8[i]) * 8]) = kreg[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int d = sv_c8[i] * 8 + j;
-        Vt_lds[0][d][swz8(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = vreg[i][j];
+        Vt_lds[d][swz8(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = vreg[i][j];
       }
     }
     __syncthreads();
   }
 
-  int cur = 0;
   for (int t = t_start; t <= t_end; ++t) {
     const int k0 = t * KVBLK;
     // ---- issue next tile's global loads early (T14) ----
-    const bool have_next = (t + 1 <= t_end);
+    const bool have_next = (ABL != 3) && (t + 1 <= t_end);
     if (have_next) {
       const int k0n = k0 + KVBLK;
 #pragma unroll
@@ -186,7 +188,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
         for (int ks = 0; ks < NKD; ++ks) {
           int krow = kvt * 16 + c;
           short8 tmp = *reinterpret_cast<const short8*>(
-              &K_lds[cur][krow][swz8(krow, ks * 4 + g) * 8]);
+              &K_lds[krow][swz8(krow, ks * 4 + g) * 8]);
           bf16x8 ak = *reinterpret_cast<bf16x8*>(&tmp);
 #pragma unroll
           for (int qt = 0; qt < 2; ++qt)
@@ -196,10 +198,12 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
       __builtin_amdgcn_s_setprio(0);
 
       // ---- mask + online softmax (per lane: q = q0w + qt*16 + c) ----
+      if (ABL != 1) {
       bool tile_full = (k0 + KVBLK <= skv) && (q0w + 31 < sq);
       if (causal) tile_full &= (k0 + KVBLK - 1 <= q0w + off);
       if (window > 0) tile_full &= (k0 >= q0w + 31 + off - window + 1);
 
+      // m/p tracked in RAW score domain (scale folded into the exp fma)
       float pmax[2] = {-1e30f, -1e30f};
       if (tile_full) {
 #pragma unroll
@@ -207,11 +211,8 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
 #pragma unroll
           for (int kvt = 0; kvt < KVBLK / 16; ++kvt)
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              float val = st[qt][kvt][r] * scale;
-              st[qt][kvt][r] = val;
-              pmax[qt] = fmaxf(pmax[qt], val);
-            }
+            for (int r = 0; r < 4; ++r)
+              pmax[qt] = fmaxf(pmax[qt], st[qt][kvt][r]);
       } else {
 #pragma unroll
         for (int qt = 0; qt < 2; ++qt) {
@@ -224,9 +225,8 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
               bool ok = (kvcol < skv) && (qrow < sq);
               if (causal) ok &= (kvcol <= qrow + off);
               if (window > 0) ok &= (kvcol > qrow + off - window);
-              float val = ok ? st[qt][kvt][r] * scale : -1e30f;
-              st[qt][kvt][r] = val;
-              pmax[qt] = fmaxf(pmax[qt], val);
+              if (!ok) st[qt][kvt][r] = -1e30f;
+              pmax[qt] = fmaxf(pmax[qt], st[qt][kvt][r]);
             }
         }
       }
@@ -237,15 +237,16 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
         pmax[qt] = fmaxf(pmax[qt], __shfl_xor(pmax[qt], 32, 64));
       }
 
-      // defer-max (T13): only rescale when the max grew by > THR
-      bool need = (pmax[0] > m_run[0] + RESCALE_THR) || (pmax[1] > m_run[1] + RESCALE_THR) ||
+      // defer-max (T13): only rescale when the max grew by > THR (raw units)
+      const float thr_raw = RESCALE_THR / scale;
+      bool need = (pmax[0] > m_run[0] + thr_raw) || (pmax[1] > m_run[1] + thr_raw) ||
                   (m_run[0] == -1e30f);
       if (__any(need)) {
         float alpha[2];
 #pragma unroll
         for (int qt = 0; qt < 2; ++qt) {
           float m_new = fmaxf(m_run[qt], pmax[qt]);
-          alpha[qt] = __expf(m_run[qt] - m_new);
+          alpha[qt] = __expf((m_run[qt] - m_new) * scale);
           m_run[qt] = m_new;
           l_run[qt] *= alpha[qt];
         }
@@ -261,16 +262,18 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
         }
       }
 
-      // exp + row-sum (l stays lane-partial; reduced in the epilogue)
+      // exp + row-sum: p = exp2(sv*c1 - m*c1), one fma + one exp each
+      const float c1 = scale * LOG2E;
 #pragma unroll
       for (int qt = 0; qt < 2; ++qt) {
         float ps = 0.f;
+        const float mc = m_run[qt] * c1;
 #pragma unroll
         for (int kvt = 0; kvt < KVBLK / 16; ++kvt)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             float sv = st[qt][kvt][r];
-            float p = exp2f((sv - m_run[qt]) * LOG2E);
+            float p = exp2f(__builtin_fmaf(sv, c1, -mc));
             if (sv <= -1e29f) p = 0.f;
             st[qt][kvt][r] = p;
             ps += p;
@@ -278,43 +281,59 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
         l_run[qt] += ps;
       }
 
-      // ---- P -> LDS (pack pairs, b32 stores into swizzled chunks) ----
+      }  // ABL != 1
+      if (ABL == 2) {  // keep QK live, skip PV
 #pragma unroll
-      for (int qt = 0; qt < 2; ++qt) {
-        int qrow = qt * 16 + c;
-        int* prow = reinterpret_cast<int*>(&P_lds[wid][0][0]) + qrow * (KVBLK / 2);
+        for (int qt = 0; qt < 2; ++qt)
 #pragma unroll
-        for (int kvt = 0; kvt < KVBLK / 16; ++kvt) {
+          for (int kvt = 0; kvt < KVBLK / 16; ++kvt)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) oacc[qt][0][r & 3] += st[qt][kvt][r];
+      } else {
+      // ---- pack P to bf16 pairs in registers ----
+      // pk[qt][kvt][p] at lane (g,c): kv pair (16*kvt + 4g + 2p, +1) of q col c
+      int pk[2][KVBLK / 16][2];
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt)
+#pragma unroll
+        for (int kvt = 0; kvt < KVBLK / 16; ++kvt)
 #pragma unroll
           for (int p = 0; p < 2; ++p) {
             unsigned lo = (unsigned short)f2sbf(st[qt][kvt][2 * p]);
             unsigned hi = (unsigned short)f2sbf(st[qt][kvt][2 * p + 1]);
-            int pi = kvt * 8 + g * 2 + p;           // dword index in the row
-            prow[swz8(qrow, pi >> 2) * 4 + (pi & 3)] = (int)(lo | (hi << 16));
+            pk[qt][kvt][p] = (int)(lo | (hi << 16));
           }
-        }
-      }
-      // P_lds is wave-private: only this wave's LDS ops need to drain
-      __builtin_amdgcn_s_waitcnt(0xC07F /* lgkmcnt 0 only */);
-      __builtin_amdgcn_sched_barrier(0);
 
-      // ---- O += P V : D_O[row=q][col=d], A=P, B=Vt ----
+      // ---- O += P V : D_O[row=q][col=d], A=P (built in-register), B=Vt ----
+      // The D-fragment -> A-fragment relayout is exactly two register-pair
+      // swaps (T12 family, pure VALU, no LDS):
+      //   (U, W)   = permlane32_swap(pk[2ks2][p], pk[2ks2+1][p])
+      //   (A_p, A_{p+2}) = permlane16_swap(U, W)
+      // giving A-frag pair u at lane (g,c) = kv pair 16*ks2*2 + 4g' ... =
+      // pair index 16*ks2 + 4g + u as the A layout requires.
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks2 = 0; ks2 < KVBLK / 32; ++ks2) {
         bf16x8 pa[2];
 #pragma unroll
         for (int qt = 0; qt < 2; ++qt) {
-          int qrow = qt * 16 + c;
-          short8 tmp = *reinterpret_cast<const short8*>(
-              &P_lds[wid][qrow][swz8(qrow, ks2 * 4 + g) * 8]);
-          pa[qt] = *reinterpret_cast<bf16x8*>(&tmp);
+          int au[4];
+#pragma unroll
+          for (int p = 0; p < 2; ++p) {
+            auto uw = __builtin_amdgcn_permlane32_swap(
+                pk[qt][2 * ks2][p], pk[qt][2 * ks2 + 1][p], false, false);
+            auto aa = __builtin_amdgcn_permlane16_swap(uw[0], uw[1], false, false);
+            au[p] = aa[0];
+            au[p + 2] = aa[1];
+          }
+          int4 av = make_int4(au[0], au[1], au[2], au[3]);
+          pa[qt] = *reinterpret_cast<bf16x8*>(&av);
         }
 #pragma unroll
         for (int n = 0; n < ND; ++n) {
           int vrow = n * 16 + c;
           short8 tmp = *reinterpret_cast<const short8*>(
-              &Vt_lds[cur][vrow][swz8(vrow, ks2 * 4 + g) * 8]);
+              &Vt_lds[vrow][swz8(vrow, ks2 * 4 + g) * 8]);
           bf16x8 bv = *reinterpret_cast<bf16x8*>(&tmp);
 #pragma unroll
           for (int qt = 0; qt < 2; ++qt)
@@ -322,24 +341,25 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
         }
       }
       __builtin_amdgcn_s_setprio(0);
+      }  // ABL != 2
     }
 
-    // ---- write staged tile t+1 into the other LDS buffer, flip ----
+    // ---- overwrite the LDS tile with staged tile t+1 ----
     if (have_next) {
+      __syncthreads();  // everyone done reading tile t
       __builtin_amdgcn_s_waitcnt(WAIT_VM0);
       __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
       for (int i = 0; i < SREG; ++i) {
-        *reinterpret_cast<short8*>(&K_lds[cur ^ 1][st_row[i]][swz8(st_row[i], st_c8[i]) * 8]) = kreg[i];
+        *reinterpret_cast<short8*>(&K_lds[st_row[i]][swz8(st_row[i], st_c8[i]) * 8]) = kreg[i];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int d = sv_c8[i] * 8 + j;
-          Vt_lds[cur ^ 1][d][swz8(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = vreg[i][j];
+          Vt_lds[d][swz8(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = vreg[i][j];
         }
       }
+      __syncthreads();  // tile t+1 visible
     }
-    __syncthreads();
-    cur ^= 1;
   }
 
   // ---- epilogue ----
@@ -352,7 +372,8 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel_v2(
     linv_c[qt] = (l_run[qt] > 0.f) ? 1.f / l_run[qt] : 0.f;
     int qrow = q0w + qt * 16 + c;
     if (g == 0 && qrow < sq)
-      lse[((long)batch * hq + head) * sq + qrow] = m_run[qt] + __logf(fmaxf(l_run[qt], 1e-30f));
+      lse[((long)batch * hq + head) * sq + qrow] =
+          m_run[qt] * scale + __logf(fmaxf(l_run[qt], 1e-30f));
   }
 #pragma unroll
   for (int qt = 0; qt < 2; ++qt) {
@@ -590,19 +611,25 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto lse = torch::empty({b, hq, sq}, q.options().dtype(torch::kFloat32));
   auto stream = at::cuda::getCurrentHIPStream();
   static const bool force_v1 = getenv("MEGATRON_AMD_ATTN_V1") != nullptr;
+  static const char* abl_env = getenv("MEGA_ATTN_ABL");
+  static const int abl = abl_env ? atoi(abl_env) : 0;
   if (sq >= 256 && !force_v1) {
     int nqc = (sq + 255) / 256;
     dim3 grid(nqc * b * hq);
-    if (d == 128)
-      hipLaunchKernelGGL((attn_fwd_kernel_v2<128>), grid, dim3(512), 0, stream,
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, grid, dim3(512), 0, stream,
                          (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
                          (bf16*)out.data_ptr(), lse.data_ptr<float>(),
                          sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window, nqc);
-    else
-      hipLaunchKernelGGL((attn_fwd_kernel_v2<64>), grid, dim3(512), 0, stream,
-                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
-                         (bf16*)out.data_ptr(), lse.data_ptr<float>(),
-                         sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window, nqc);
+    };
+    if (d == 128) {
+      if (abl == 1) launch(attn_fwd_kernel_v2<128, 1>);
+      else if (abl == 2) launch(attn_fwd_kernel_v2<128, 2>);
+      else if (abl == 3) launch(attn_fwd_kernel_v2<128, 3>);
+      else launch(attn_fwd_kernel_v2<128, 0>);
+    } else {
+      launch(attn_fwd_kernel_v2<64, 0>);
+    }
     return {out, lse};
   }
   dim3 grid((sq + 63) / 64, b * hq);

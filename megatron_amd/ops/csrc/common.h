@@ -41,6 +41,36 @@ __device__ __forceinline__ short f2sbf(float f) {
   return (short)(cvt.u >> 16);
 }
 
+// D-fragment -> transposed-A-fragment relayout (CDNA4, pure VALU).
+//
+// Input: a 16x64 MFMA D-tile of M held as 4 sub-tiles of f32x4 packed to
+// bf16 pairs: pk[t][p] at lane (g = lane>>4, c = lane&15) holds the pair
+// (M[16t + 4g + 2p][c], M[16t + 4g + 2p + 1][c]).
+// Output: the A-fragment (bf16x8 as int4) of M^T for k-step `ks` (k = 32
+// wide): lane (g,c) gets M^T[row=c][k = 32ks + 8g .. +7].
+// Exactly two register-pair swaps per packed stream:
+//   (U, W) = permlane32_swap(pk[2ks][p], pk[2ks+1][p])
+//   (A_p, A_{p+2}) = permlane16_swap(U, W)
+__device__ __forceinline__ void dfrag_to_at(const int pk0[2], const int pk1[2], int* av4) {
+  auto uw0 = __builtin_amdgcn_permlane32_swap(pk0[0], pk1[0], false, false);
+  auto aa0 = __builtin_amdgcn_permlane16_swap(uw0[0], uw0[1], false, false);
+  auto uw1 = __builtin_amdgcn_permlane32_swap(pk0[1], pk1[1], false, false);
+  auto aa1 = __builtin_amdgcn_permlane16_swap(uw1[0], uw1[1], false, false);
+  av4[0] = aa0[0];
+  av4[1] = aa1[0];
+  av4[2] = aa0[1];
+  av4[3] = aa1[1];
+}
+
+// pack two f32 into one dword of 2 bf16 (lo = first)
+__device__ __forceinline__ int pack_bf16x2(float lo, float hi) {
+  union { float f; unsigned u; } a, b;
+  a.f = lo; b.f = hi;
+  unsigned la = (a.u + (0x7fff + ((a.u >> 16) & 1))) >> 16;
+  unsigned lb = (b.u + (0x7fff + ((b.u >> 16) & 1))) >> 16;
+  return (int)((la & 0xffffu) | (lb << 16));
+}
+
 // full-wave reduction (64 lanes)
 __device__ __forceinline__ float wave_reduce_sum(float v) {
 #pragma unroll
