@@ -28,6 +28,7 @@ class ParallelConfig:
     tensor_parallel_size: int = 1
     pipeline_parallel_size: int = 1
     context_parallel_size: int = 1
+    cp_comm_type: str = "p2p"  # ring (p2p) or Ulysses head-scatter (a2a)
     expert_parallel_size: int = 1
     virtual_pipeline_parallel_size: Optional[int] = None
     sequence_parallel: bool = False

@@ -116,7 +116,18 @@ class GPTModel(nn.Module):
                 G.get_tensor_model_parallel_world_size() if self.config.sequence_parallel else 1
             )
 
-        if inference_context is not None and self.config.position_embedding_type == "rope":
+        cp = G.get_context_parallel_world_size()
+        if cp > 1 and self.config.position_embedding_type == "rope":
+            # tokens are CP-sharded: index the full-sequence freq table at this
+            # rank's global positions (reference rope_utils.py:48 analog)
+            from megatron_amd.parallel.context_parallel import cp_rope_positions
+
+            s_global = seq_len * cp
+            table = self._rotary_freqs(s_global, hidden.device)
+            pos = cp_rope_positions(s_global, G.get_context_parallel_rank(), cp,
+                                    hidden.device, mode=self.config.cp_comm_type)
+            rotary = table[pos]
+        elif inference_context is not None and self.config.position_embedding_type == "rope":
             # positions come from the KV context (cache offset / per-request
             # lengths under continuous batching): index the full freq table
             table = self._rotary_freqs(self.config.max_position_embeddings, hidden.device)

@@ -1,0 +1,285 @@
+"""Context parallelism: ring attention over RCCL p2p (xGMI is fully connected,
+so the ring sends are single-hop) and Ulysses-style head-scatter all-to-all.
+
+Capability analog of the reference's CP support (parallel_state.py:124-131 CP
+groups, extensions/transformer_engine.py:2086-2113 cp_comm_type p2p/a2a,
+core/utils.py get_batch_on_this_cp_rank, rope_utils.py:48 freq slicing) — but
+the ring exchange and online-softmax merge are implemented here on top of our
+own flash kernel instead of inside a vendor library.
+
+Load-balanced causal sharding (p2p mode): the sequence is cut into 2*cp
+chunks; rank r holds chunks (r, 2cp-1-r), so every rank does the same amount
+of causal-attention work. Chunk ids order the masking: a q-chunk attends a
+kv-chunk fully if q_id > kv_id, causally if equal, not at all if less.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from megatron_amd import ops
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.mappings import all_to_all
+
+
+# ---------------------------------------------------------------------------
+# batch / position sharding
+# ---------------------------------------------------------------------------
+
+def cp_chunk_ids(cp_rank: int, cp_size: int):
+    return (cp_rank, 2 * cp_size - 1 - cp_rank)
+
+
+def slice_for_cp_rank(x: torch.Tensor, cp_rank: int, cp_size: int, seq_dim: int = 1,
+                      mode: str = "p2p") -> torch.Tensor:
+    """Slice a full-sequence tensor to this CP rank's shard.
+
+    p2p (ring): load-balanced 2-chunk layout; a2a (Ulysses): contiguous."""
+    if cp_size == 1:
+        return x
+    s = x.size(seq_dim)
+    if mode == "a2a":
+        assert s % cp_size == 0
+        return x.narrow(seq_dim, cp_rank * (s // cp_size), s // cp_size)
+    assert s % (2 * cp_size) == 0, f"seq {s} must divide 2*cp={2*cp_size}"
+    L = s // (2 * cp_size)
+    c0, c1 = cp_chunk_ids(cp_rank, cp_size)
+    return torch.cat([x.narrow(seq_dim, c0 * L, L), x.narrow(seq_dim, c1 * L, L)], dim=seq_dim)
+
+
+def cp_rope_positions(seq_len_global: int, cp_rank: int, cp_size: int,
+                      device, mode: str = "p2p") -> torch.Tensor:
+    """Global positions of this rank's tokens (for RoPE table indexing)."""
+    if cp_size == 1:
+        return torch.arange(seq_len_global, device=device)
+    if mode == "a2a":
+        L = seq_len_global // cp_size
+        return torch.arange(cp_rank * L, (cp_rank + 1) * L, device=device)
+    L = seq_len_global // (2 * cp_size)
+    c0, c1 = cp_chunk_ids(cp_rank, cp_size)
+    return torch.cat([torch.arange(c0 * L, (c0 + 1) * L, device=device),
+                      torch.arange(c1 * L, (c1 + 1) * L, device=device)])
+
+
+def get_batch_on_this_cp_rank(batch: dict, mode: str = "p2p") -> dict:
+    cp = G.get_context_parallel_world_size()
+    if cp == 1:
+        return batch
+    r = G.get_context_parallel_rank()
+    return {k: (slice_for_cp_rank(v, r, cp, seq_dim=1, mode=mode)
+                if v.dim() >= 2 else v) for k, v in batch.items()}
+
+
+# ---------------------------------------------------------------------------
+# attention partials (native flash kernel on GPU, fp32 torch on CPU)
+# ---------------------------------------------------------------------------
+
+def _fwd_partial(q, k, v, causal: bool, scale: float):
+    """returns (out [s,b,hq,d] same dtype, lse [b,hq,s] fp32)."""
+    if ops.has_native() and q.is_cuda:
+        return ops._C.attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
+                               causal, scale, 0)
+    s, b, hq, d = q.shape
+    hkv = k.shape[2]
+    rep = hq // hkv
+    qf = q.permute(1, 2, 0, 3).float()
+    kf = k.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
+    vf = v.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b,hq,s,skv]
+    if causal:
+        skv = k.shape[0]
+        mask = torch.ones(s, skv, dtype=torch.bool, device=q.device).tril_(skv - s)
+        scores = scores.masked_fill(~mask, float("-inf"))
+    lse = torch.logsumexp(scores, dim=-1)  # [b,hq,s]
+    out = torch.matmul(torch.softmax(scores, dim=-1), vf)
+    return out.permute(2, 0, 1, 3).to(q.dtype), lse
+
+
+def _bwd_partial(dout, q, k, v, out, lse, causal: bool, scale: float):
+    """FA2-style manual backward of one partial (global out/lse): returns
+    (dq, dk, dv) in fp32."""
+    if ops.has_native() and q.is_cuda:
+        dq, dk, dv = ops._C.attn_bwd(dout.contiguous(), q.contiguous(), k.contiguous(),
+                                     v.contiguous(), out.contiguous(), lse.contiguous(),
+                                     causal, scale, 0)
+        return dq.float(), dk.float(), dv.float()
+    s, b, hq, d = q.shape
+    hkv = k.shape[2]
+    rep = hq // hkv
+    qf = q.permute(1, 2, 0, 3).float()
+    kf = k.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
+    vf = v.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
+    dof = dout.permute(1, 2, 0, 3).float()
+    of = out.permute(1, 2, 0, 3).float()
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    p = torch.exp(scores - lse.unsqueeze(-1))  # uses the GLOBAL lse
+    if causal:
+        skv = k.shape[0]
+        mask = torch.ones(s, skv, dtype=torch.bool, device=q.device).tril_(skv - s)
+        p = p * mask
+    dvf = torch.matmul(p.transpose(-1, -2), dof)
+    dp = torch.matmul(dof, vf.transpose(-1, -2))
+    drow = (dof * of).sum(-1, keepdim=True)
+    ds = p * (dp - drow) * scale
+    dqf = torch.matmul(ds, kf)
+    dkf = torch.matmul(ds.transpose(-1, -2), qf)
+    if rep > 1:
+        dkf = dkf.view(b, hkv, rep, *dkf.shape[2:]).sum(2)
+        dvf = dvf.view(b, hkv, rep, *dvf.shape[2:]).sum(2)
+    perm = lambda t: t.permute(2, 0, 1, 3)
+    return perm(dqf), perm(dkf), perm(dvf)
+
+
+def _merge(o_run, lse_run, o_new, lse_new):
+    """online-softmax merge of two partials (fp32)."""
+    lse_max = torch.maximum(lse_run, lse_new)
+    a = torch.exp(lse_run - lse_max)
+    bexp = torch.exp(lse_new - lse_max)
+    lse_out = lse_max + torch.log(a + bexp)
+    w1 = torch.exp(lse_run - lse_out).permute(2, 0, 1).unsqueeze(-1)  # [s,b,h,1]
+    w2 = torch.exp(lse_new - lse_out).permute(2, 0, 1).unsqueeze(-1)
+    return o_run * w1 + o_new.float() * w2, lse_out
+
+
+def _ring_peers(group):
+    cp = dist.get_world_size(group)
+    ranks = dist.get_process_group_ranks(group)
+    r = ranks.index(dist.get_rank())
+    return cp, r, ranks[(r + 1) % cp], ranks[(r - 1) % cp], ranks
+
+
+def _ring_sendrecv(send: torch.Tensor, group) -> torch.Tensor:
+    """one ring step: send to next, receive from prev (batched p2p)."""
+    cp, r, nxt, prv, _ = _ring_peers(group)
+    recv = torch.empty_like(send)
+    ops_ = [dist.P2POp(dist.isend, send.contiguous(), nxt, group=group),
+            dist.P2POp(dist.irecv, recv, prv, group=group)]
+    for w in dist.batch_isend_irecv(ops_):
+        w.wait()
+    return recv
+
+
+class _RingAttention(torch.autograd.Function):
+    """Causal ring flash attention over the CP group (C14/K2).
+
+    q/k/v: [2L, b, h, d] in the load-balanced 2-chunk layout. KV rotates
+    around the ring; partials merge by LSE. Backward re-rotates KV and
+    accumulates dK/dV in a buffer that travels with them.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale, group):
+        cp, r, *_ = _ring_peers(group)
+        L = q.shape[0] // 2
+        my_chunks = cp_chunk_ids(r, cp)
+        o = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        lse = torch.full((q.shape[1], q.shape[2], q.shape[0]), float("-inf"),
+                         dtype=torch.float32, device=q.device)
+        kv = torch.stack([k, v])
+        for step in range(cp):
+            src = (r - step) % cp
+            src_chunks = cp_chunk_ids(src, cp)
+            # exchange BEFORE compute so the send overlaps the partials
+            kv_next = _ring_sendrecv(kv, group) if step < cp - 1 else None
+            for qi in range(2):
+                for ki in range(2):
+                    if my_chunks[qi] < src_chunks[ki]:
+                        continue
+                    causal = my_chunks[qi] == src_chunks[ki]
+                    qc = q[qi * L:(qi + 1) * L]
+                    kc = kv[0, ki * L:(ki + 1) * L]
+                    vc = kv[1, ki * L:(ki + 1) * L]
+                    oc, lsec = _fwd_partial(qc, kc, vc, causal, scale)
+                    sl = slice(qi * L, (qi + 1) * L)
+                    o_m, lse_m = _merge(o[sl], lse[..., sl], oc, lsec)
+                    o[sl] = o_m
+                    lse[..., sl] = lse_m
+            if kv_next is not None:
+                kv = kv_next
+        out = o.to(q.dtype)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.scale, ctx.group = scale, group
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        scale, group = ctx.scale, ctx.group
+        cp, r, *_ = _ring_peers(group)
+        L = q.shape[0] // 2
+        my_chunks = cp_chunk_ids(r, cp)
+        dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        # kv and its gradient accumulator travel together around the ring;
+        # after cp steps the accumulator is back at the kv owner.
+        kv = torch.stack([k, v])
+        dkv = torch.zeros(kv.shape, dtype=torch.float32, device=q.device)
+        for step in range(cp):
+            src = (r - step) % cp
+            src_chunks = cp_chunk_ids(src, cp)
+            for qi in range(2):
+                for ki in range(2):
+                    if my_chunks[qi] < src_chunks[ki]:
+                        continue
+                    causal = my_chunks[qi] == src_chunks[ki]
+                    sl = slice(qi * L, (qi + 1) * L)
+                    kl = slice(ki * L, (ki + 1) * L)
+                    dqc, dkc, dvc = _bwd_partial(
+                        dout[sl], q[sl], kv[0, kl], kv[1, kl], out[sl], lse[..., sl],
+                        causal, scale)
+                    dq[sl] += dqc
+                    dkv[0, kl] += dkc
+                    dkv[1, kl] += dvc
+            if step < cp - 1:
+                kv = _ring_sendrecv(kv, group)
+                dkv = _ring_sendrecv(dkv, group)
+        # final rotation returns dkv to its owner (cp-1 hops done, one more)
+        if cp > 1:
+            dkv = _ring_sendrecv(dkv, group)
+        return dq.to(q.dtype), dkv[0].to(q.dtype), dkv[1].to(q.dtype), None, None
+
+
+def ring_attention(q, k, v, scale: Optional[float] = None, group=None):
+    """q [2L,b,hq,d], k/v [2L,b,hkv,d] in CP 2-chunk layout -> out [2L,b,hq,d]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    group = group if group is not None else G.get_context_parallel_group()
+    if group is None or dist.get_world_size(group) == 1:
+        return ops.flash_attention(q, k, v, causal=True, scale=scale)
+    return _RingAttention.apply(q, k, v, scale, group)
+
+
+def ulysses_attention(q, k, v, scale: Optional[float] = None, group=None):
+    """Ulysses a2a CP: scatter heads / gather sequence around full attention.
+
+    q [s/cp, b, hq, d] (contiguous slicing) -> out [s/cp, b, hq, d]. Needs
+    hq and hkv divisible by cp. Single-hop on fully-connected xGMI.
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    group = group if group is not None else G.get_context_parallel_group()
+    cp = dist.get_world_size(group) if group is not None else 1
+    if cp == 1:
+        return ops.flash_attention(q, k, v, causal=True, scale=scale)
+    sl, b, hq, d = q.shape
+    hkv = k.shape[2]
+    assert hq % cp == 0 and hkv % cp == 0, "Ulysses needs heads divisible by cp"
+
+    def sp2hp(x):
+        s_, b_, h_, d_ = x.shape
+        xt = x.reshape(s_, b_, cp, h_ // cp, d_).permute(2, 0, 1, 3, 4).contiguous()
+        out = all_to_all(group, xt.view(cp * s_, b_, h_ // cp, d_))
+        return out  # [s_full, b, h/cp, d]
+
+    def hp2sp(x):
+        s_full = x.shape[0]
+        out = all_to_all(group, x.contiguous())
+        s_ = s_full // cp
+        return out.view(cp, s_, b, x.shape[2], d).permute(1, 2, 0, 3, 4).reshape(s_, b, x.shape[2] * cp, d)
+
+    qh, kh, vh = sp2hp(q), sp2hp(k), sp2hp(v)
+    oh = ops.flash_attention(qh, kh, vh, causal=True, scale=scale)
+    return hp2sp(oh)

@@ -58,6 +58,7 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--pipeline-model-parallel-size", "--pp", type=int, default=1)
     g.add_argument("--virtual-pipeline-model-parallel-size", "--vpp", type=int, default=None)
     g.add_argument("--context-parallel-size", "--cp", type=int, default=1)
+    g.add_argument("--cp-comm-type", choices=["p2p", "a2a"], default="p2p")
     g.add_argument("--expert-model-parallel-size", "--ep", type=int, default=1)
     g.add_argument("--expert-tensor-parallel-size", type=int, default=None)
     g.add_argument("--sequence-parallel", action="store_true")
@@ -187,6 +188,7 @@ def configs_from_args(args):
         pipeline_parallel_size=args.pipeline_model_parallel_size,
         virtual_pipeline_parallel_size=args.virtual_pipeline_model_parallel_size,
         context_parallel_size=args.context_parallel_size,
+        cp_comm_type=args.cp_comm_type,
         expert_parallel_size=args.expert_model_parallel_size,
         expert_tensor_parallel_size=args.expert_tensor_parallel_size,
         sequence_parallel=args.sequence_parallel,

@@ -88,6 +88,10 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
 
     def forward_step(data_iterator, model):
         batch = next(data_iterator)
+        if cfg.context_parallel_size > 1:
+            from megatron_amd.parallel.context_parallel import get_batch_on_this_cp_rank
+
+            batch = get_batch_on_this_cp_rank(batch, mode=cfg.cp_comm_type)
 
         def loss_func(loss_sb):
             if "loss_mask" in batch:

@@ -83,6 +83,14 @@ class SelfAttention(nn.Module):
             core_out = inference_context.attend(
                 self.layer_number, q, k, v, self.softmax_scale, self.window
             )
+        elif G.get_context_parallel_world_size() > 1:
+            from megatron_amd.parallel.context_parallel import ring_attention, ulysses_attention
+
+            assert self.window is None, "sliding window not supported under CP yet"
+            if self.config.cp_comm_type == "a2a":
+                core_out = ulysses_attention(q, k, v, scale=self.softmax_scale)
+            else:
+                core_out = ring_attention(q, k, v, scale=self.softmax_scale)
         else:
             core_out = ops.flash_attention(q, k, v, causal=True, scale=self.softmax_scale, window=self.window)
         core_out = core_out.reshape(s, b, ng * rep * d)
