@@ -152,14 +152,24 @@ def _ring_peers(group):
     return cp, r, ranks[(r + 1) % cp], ranks[(r - 1) % cp], ranks
 
 
-def _ring_sendrecv(send: torch.Tensor, group) -> torch.Tensor:
-    """one ring step: send to next, receive from prev (batched p2p)."""
+def _ring_sendrecv_begin(send: torch.Tensor, group):
+    """start one ring step: send to next, receive from prev (batched p2p).
+    Returns (recv_buffer, work_handles) - call _ring_wait before reading."""
     cp, r, nxt, prv, _ = _ring_peers(group)
     recv = torch.empty_like(send)
     ops_ = [dist.P2POp(dist.isend, send.contiguous(), nxt, group=group),
             dist.P2POp(dist.irecv, recv, prv, group=group)]
-    for w in dist.batch_isend_irecv(ops_):
+    return recv, dist.batch_isend_irecv(ops_)
+
+
+def _ring_wait(works):
+    for w in works:
         w.wait()
+
+
+def _ring_sendrecv(send: torch.Tensor, group) -> torch.Tensor:
+    recv, works = _ring_sendrecv_begin(send, group)
+    _ring_wait(works)
     return recv
 
 
@@ -183,8 +193,10 @@ class _RingAttention(torch.autograd.Function):
         for step in range(cp):
             src = (r - step) % cp
             src_chunks = cp_chunk_ids(src, cp)
-            # exchange BEFORE compute so the send overlaps the partials
-            kv_next = _ring_sendrecv(kv, group) if step < cp - 1 else None
+            # start the exchange BEFORE compute so it overlaps the partials
+            works = None
+            if step < cp - 1:
+                kv_next, works = _ring_sendrecv_begin(kv, group)
             for qi in range(2):
                 for ki in range(2):
                     if my_chunks[qi] < src_chunks[ki]:
@@ -198,7 +210,8 @@ class _RingAttention(torch.autograd.Function):
                     o_m, lse_m = _merge(o[sl], lse[..., sl], oc, lsec)
                     o[sl] = o_m
                     lse[..., sl] = lse_m
-            if kv_next is not None:
+            if works is not None:
+                _ring_wait(works)
                 kv = kv_next
         out = o.to(q.dtype)
         ctx.save_for_backward(q, k, v, out, lse)
