@@ -121,6 +121,13 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--log-timers", action="store_true")
     g.add_argument("--tensorboard-dir", type=str, default=None)
     g.add_argument("--log-memory", action="store_true")
+    g.add_argument("--log-straggler", action="store_true")
+    g.add_argument("--straggler-report-interval", type=int, default=10)
+    g.add_argument("--use-wandb", action="store_true")
+    g.add_argument("--wandb-project", type=str, default=None)
+    g.add_argument("--check-weight-hash-across-dp-replicas-interval", type=int, default=None)
+    g.add_argument("--rerun-mode", choices=["disabled", "validate_results"], default="disabled")
+    g.add_argument("--log-energy", action="store_true")
 
     g = p.add_argument_group("profiling")
     g.add_argument("--profile", action="store_true")

@@ -24,6 +24,13 @@ from megatron_amd.parallel.random import model_parallel_seed
 from megatron_amd.training.arguments import configs_from_args, parse_and_validate_args
 from megatron_amd.training.flops import MI355X_BF16_DENSE_PEAK_TFLOPS, num_floating_point_operations
 from megatron_amd.training.training import setup_model_and_optimizer, train_step
+from megatron_amd.utils.metrics import MetricsLogger, append_progress_log
+from megatron_amd.utils.rerun_state_machine import (
+    RerunDataIterator,
+    get_rerun_state_machine,
+    initialize_rerun_state_machine,
+)
+from megatron_amd.utils.straggler import EnergyMonitor, StragglerDetector
 from megatron_amd.utils.timers import Timers
 
 
@@ -73,6 +80,19 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
 
     chunks, optimizer = setup_model_and_optimizer(model_provider, cfg, opt_cfg, ddp_cfg, device=device)
     scheduler = OptimizerParamScheduler(optimizer, opt_cfg, args.train_iters)
+    initialize_rerun_state_machine(args.rerun_mode)
+    straggler = StragglerDetector(enabled=args.log_straggler)
+    energy = EnergyMonitor() if args.log_energy else None
+    metrics = MetricsLogger(args.tensorboard_dir, rank=args.rank,
+                            use_wandb=args.use_wandb, wandb_project=args.wandb_project)
+
+    from megatron_amd.training.theoretical_memory import format_report, report
+
+    cfg.seq_length = args.seq_length
+    _print_rank0(format_report(report(cfg, args.micro_batch_size, args.num_microbatches,
+                                      args.data_parallel_size,
+                                      args.use_distributed_optimizer,
+                                      args.recompute_granularity == "full")))
 
     iteration = 0
     if args.load:
@@ -84,7 +104,7 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             _print_rank0(f"no checkpoint found in {args.load}; starting fresh")
 
     n_chunks = len(chunks)
-    data_iters = [build_data_iterator(args, str(device)) for _ in range(n_chunks)]
+    data_iters = [RerunDataIterator(build_data_iterator(args, str(device))) for _ in range(n_chunks)]
 
     def forward_step(data_iterator, model):
         batch = next(data_iterator)
@@ -101,6 +121,9 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             else:
                 s = loss_sb.sum()
                 ntok = torch.tensor(loss_sb.numel(), device=loss_sb.device)
+            # rerun machine: flag NaN/Inf losses for replay classification
+            get_rerun_state_machine().validate_result(
+                s, lambda t: not bool(torch.isfinite(t).all()), "nan/inf loss")
             return s, ntok, {"loss_sum": s.detach()}
 
         out = model(batch["tokens"], labels=batch["labels"])
@@ -123,11 +146,25 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             )
             prof.__enter__()
         timers("iteration").start()
+        straggler.start()
         result = train_step(forward_step, data_iters, chunks, optimizer, cfg,
                             args.num_microbatches, args.seq_length, args.micro_batch_size)
+        straggler.stop()
         timers("iteration").stop()
+        if "exit_code" in result:
+            if args.save:
+                save_checkpoint(args.save, chunks, optimizer, iteration, scheduler)
+            _print_rank0(f"rerun state machine requested exit (code {result['exit_code']})")
+            metrics.close()
+            sys.exit(result["exit_code"])
         iteration += 1
         scheduler.step()
+        if (args.check_weight_hash_across_dp_replicas_interval
+                and iteration % args.check_weight_hash_across_dp_replicas_interval == 0):
+            from megatron_amd.distributed.checks import check_param_hashes_across_dp_replicas
+
+            if not check_param_hashes_across_dp_replicas(chunks):
+                _print_rank0(f"WARNING: parameter hash mismatch across DP replicas at iteration {iteration}")
         if prof is not None and iteration == args.profile_step_end:
             prof.__exit__(None, None, None)
             prof = None
@@ -144,7 +181,20 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                 msg += f" | grad norm {result['grad_norm']:.3f}"
             if args.log_memory and torch.cuda.is_available():
                 msg += f" | mem {torch.cuda.max_memory_allocated()/2**30:.1f}GB"
+            if energy is not None:
+                e = energy.lap()
+                if e is not None:
+                    msg += f" | energy {e:.0f}J"
             _print_rank0(msg)
+            metrics.log(iteration, lm_loss=result["lm_loss"], lr=optimizer.get_lr(),
+                        iter_time_ms=t * 1000, tokens_per_s=tokens_per_s,
+                        tflops_per_gpu=tflops, grad_norm=result.get("grad_norm"))
+            if args.log_straggler and iteration % args.straggler_report_interval == 0:
+                rep = straggler.report()
+                if rep is not None:
+                    _print_rank0(f"straggler: min rank {rep.min_rank} {rep.min_time_ms:.1f}ms | "
+                                 f"max rank {rep.max_rank} {rep.max_time_ms:.1f}ms | "
+                                 f"mean {rep.mean_time_ms:.1f}ms | power {rep.power_w}W | temp {rep.temp_c}C")
 
         if args.save and args.save_interval and iteration % args.save_interval == 0:
             save_checkpoint(args.save, chunks, optimizer, iteration, scheduler,
@@ -157,4 +207,6 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
     if args.save:
         save_checkpoint(args.save, chunks, optimizer, iteration, scheduler)
         _print_rank0(f"saved final checkpoint at iteration {iteration}")
+        append_progress_log(args.save, args.rank, f"finished at iteration {iteration}")
+    metrics.close()
     return iteration

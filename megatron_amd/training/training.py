@@ -17,6 +17,7 @@ from megatron_amd.distributed import DistributedDataParallel, finalize_model_gra
 from megatron_amd.optimizer import ChainedOptimizer, get_optimizer
 from megatron_amd.parallel import grid as G
 from megatron_amd.pipeline.schedules import get_forward_backward_func
+from megatron_amd.utils.rerun_state_machine import get_rerun_state_machine
 
 
 def setup_model_and_optimizer(
@@ -71,20 +72,29 @@ def train_step(
     seq_length: int,
     micro_batch_size: int,
 ):
-    """One optimizer step (reference training.py:2372)."""
-    for chunk in model_chunks:
-        chunk.zero_grad_buffer()
-    optimizer.zero_grad() if hasattr(optimizer, "zero_grad") else None
-
+    """One optimizer step (reference training.py:2372). The rerun state
+    machine wraps only the forward/backward (optimizer state is untouched on
+    a replay, reference training.py:2395)."""
+    rsm = get_rerun_state_machine()
+    iters = data_iterator if isinstance(data_iterator, list) else [data_iterator]
     fb_func = get_forward_backward_func()
-    losses, num_tokens = fb_func(
-        forward_step_func=forward_step_func,
-        data_iterator=data_iterator if isinstance(data_iterator, list) else [data_iterator],
-        model=model_chunks,
-        num_microbatches=num_microbatches,
-        seq_length=seq_length,
-        micro_batch_size=micro_batch_size,
-    )
+    losses = num_tokens = None
+    while rsm.should_run_forward_backward(iters):
+        for chunk in model_chunks:
+            chunk.zero_grad_buffer()
+        optimizer.zero_grad() if hasattr(optimizer, "zero_grad") else None
+        losses, num_tokens = fb_func(
+            forward_step_func=forward_step_func,
+            data_iterator=iters,
+            model=model_chunks,
+            num_microbatches=num_microbatches,
+            seq_length=seq_length,
+            micro_batch_size=micro_batch_size,
+        )
+    exit_code = rsm.should_checkpoint_and_exit()
+    if exit_code is not None:
+        return {"lm_loss": float("nan"), "grad_norm": None, "skipped": True,
+                "exit_code": exit_code}
 
     ok, grad_norm, _ = optimizer.step()
 

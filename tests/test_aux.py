@@ -1,0 +1,145 @@
+"""Aux subsystem tests: rerun state machine (transient vs deterministic
+classification, data replay), straggler detector, metrics logger, param-hash
+DP check, theoretical memory. Reference analogs: core/rerun_state_machine.py,
+core/utils.py StragglerDetector / check_param_hashes_across_dp_replicas."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from megatron_amd.utils.rerun_state_machine import (
+    EXIT_CODE_FAILED_ON_RESULT_VALIDATION,
+    EXIT_CODE_SUCCESS_ON_RESULT_VALIDATION,
+    RerunDataIterator,
+    RerunMode,
+    RerunStateMachine,
+)
+from tests.utils import init_single, spawn_dist
+
+
+def test_rerun_disabled_runs_once():
+    rsm = RerunStateMachine(RerunMode.DISABLED)
+    it = RerunDataIterator(iter(range(10)))
+    runs = 0
+    while rsm.should_run_forward_backward(it):
+        runs += 1
+        next(it)
+    assert runs == 1
+    assert rsm.should_checkpoint_and_exit() is None
+
+
+def test_rerun_replays_same_data_and_classifies_deterministic():
+    rsm = RerunStateMachine(RerunMode.VALIDATE_RESULTS)
+    it = RerunDataIterator(iter([torch.tensor([1.0]), torch.tensor([2.0]), torch.tensor([3.0])]))
+    seen = []
+    while rsm.should_run_forward_backward(it):
+        b = next(it)
+        seen.append(float(b))
+        # deterministic NaN: same result both runs
+        rsm.validate_result(torch.tensor(float("nan")), lambda t: not torch.isfinite(t).all())
+    assert seen == [1.0, 1.0]  # second run replayed the SAME batch
+    assert rsm.should_checkpoint_and_exit() == EXIT_CODE_FAILED_ON_RESULT_VALIDATION
+    assert rsm.stats["persistent"] == 1
+
+
+def test_rerun_classifies_transient():
+    rsm = RerunStateMachine(RerunMode.VALIDATE_RESULTS)
+    it = RerunDataIterator(iter([0, 0, 0]))
+    vals = iter([float("nan"), 5.0])  # bad first run, clean replay
+    while rsm.should_run_forward_backward(it):
+        next(it)
+        rsm.validate_result(torch.tensor(next(vals)), lambda t: not torch.isfinite(t).all())
+    assert rsm.should_checkpoint_and_exit() == EXIT_CODE_SUCCESS_ON_RESULT_VALIDATION
+    assert rsm.stats["transient"] == 1
+
+
+def test_rerun_clean_iterations_no_rerun():
+    rsm = RerunStateMachine(RerunMode.VALIDATE_RESULTS)
+    it = RerunDataIterator(iter(range(6)))
+    for _ in range(3):
+        runs = 0
+        while rsm.should_run_forward_backward(it):
+            runs += 1
+            next(it)
+            rsm.validate_result(torch.tensor(1.0), lambda t: not torch.isfinite(t).all())
+        assert runs == 1
+        assert rsm.should_checkpoint_and_exit() is None
+    assert rsm.stats["reruns"] == 0
+
+
+def test_straggler_detector_single():
+    from megatron_amd.utils.straggler import StragglerDetector
+
+    d = StragglerDetector(enabled=True)
+    for _ in range(3):
+        d.start()
+        torch.randn(64, 64) @ torch.randn(64, 64)
+        d.stop()
+    rep = d.report()
+    assert rep is not None and rep.max_time_ms >= rep.min_time_ms >= 0.0
+
+
+def test_metrics_logger(tmp_path):
+    from megatron_amd.utils.metrics import MetricsLogger
+
+    m = MetricsLogger(str(tmp_path), rank=0)
+    m.log(1, lm_loss=2.5, lr=1e-4)
+    m.log(2, lm_loss=2.4, lr=1e-4)
+    m.close()
+    lines = open(tmp_path / "metrics.jsonl").read().strip().splitlines()
+    assert len(lines) == 2
+    assert json.loads(lines[0])["lm_loss"] == 2.5
+
+
+def _run_hash_check(rank, world):
+    from megatron_amd.distributed.checks import check_param_hashes_across_dp_replicas
+    from megatron_amd.parallel import grid as G
+
+    G.initialize_model_parallel()
+    torch.manual_seed(7)  # same params everywhere
+    m = torch.nn.Linear(8, 8)
+    assert check_param_hashes_across_dp_replicas([m])
+    if rank == 1:  # desync one replica
+        with torch.no_grad():
+            m.weight += 1.0
+    assert not check_param_hashes_across_dp_replicas([m])
+
+
+def test_param_hash_across_dp():
+    spawn_dist(_run_hash_check, world_size=2)
+
+
+def test_theoretical_memory_llama8b():
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.training.theoretical_memory import num_parameters, report
+
+    init_single()
+    cfg = TransformerConfig(num_layers=32, hidden_size=4096, num_attention_heads=32,
+                            num_query_groups=8, ffn_hidden_size=14336, vocab_size=128256,
+                            max_position_embeddings=4096)
+    n = num_parameters(cfg)
+    assert 7.5e9 < n < 8.5e9  # Llama-3-8B
+    r = report(cfg, micro_batch_size=4, num_microbatches=2)
+    assert 0 < r["total_gb"] < 288
+
+
+def test_pretrain_with_rerun_and_metrics(tmp_path):
+    from megatron_amd.training.pretrain import pretrain
+
+    def provider(config, pre_process=True, post_process=True, vp_stage=None):
+        from megatron_amd.models.gpt import GPTModel
+
+        return GPTModel(config, pre_process=pre_process, post_process=post_process)
+
+    it = pretrain(provider, [
+        "--num-layers", "1", "--hidden-size", "32", "--num-attention-heads", "2",
+        "--num-query-groups", "2", "--ffn-hidden-size", "64", "--seq-length", "32",
+        "--micro-batch-size", "1", "--global-batch-size", "1", "--vocab-size", "64",
+        "--mock-data", "--train-iters", "2", "--log-interval", "1",
+        "--rerun-mode", "validate_results", "--tensorboard-dir", str(tmp_path),
+        "--log-straggler", "--straggler-report-interval", "1",
+    ])
+    assert it == 2
+    assert os.path.exists(tmp_path / "metrics.jsonl")
