@@ -175,12 +175,16 @@ def build_gpt_datasets(data_paths: Sequence, seq_length: int, seed: int,
     return tuple(out)
 
 
-def build_gpt_train_iterator(args, device, dp_rank: int, dp_size: int):
-    """Rank-sharded infinite iterator over the train split (used by pretrain)."""
+def build_gpt_train_iterator(args, device, dp_rank: int, dp_size: int,
+                             start_sample: int = 0, split: str = "train"):
+    """Rank-sharded infinite iterator over one split (used by pretrain).
+    ``start_sample`` resumes mid-epoch after a checkpoint load."""
     train_samples = args.train_iters * args.global_batch_size
-    train, _, _ = build_gpt_datasets(args.data_path, args.seq_length, args.seed,
-                                     train_samples, args.split)
-    sampler = _ShardedSequentialSampler(len(train), args.micro_batch_size, dp_rank, dp_size)
+    ds = build_gpt_datasets(args.data_path, args.seq_length, args.seed,
+                            train_samples, args.split)
+    train = {"train": ds[0], "valid": ds[1] or ds[0], "test": ds[2] or ds[0]}[split]
+    sampler = _ShardedSequentialSampler(len(train), args.micro_batch_size, dp_rank, dp_size,
+                                        start_sample=start_sample)
     loader = torch.utils.data.DataLoader(
         train, batch_sampler=sampler, num_workers=args.num_workers, pin_memory=device != "cpu")
 
@@ -196,13 +200,15 @@ class _ShardedSequentialSampler(torch.utils.data.Sampler):
     """Contiguous global-batch order, strided over DP ranks (matches the
     reference MegatronPretrainingSampler semantics)."""
 
-    def __init__(self, total: int, micro_batch: int, dp_rank: int, dp_size: int):
+    def __init__(self, total: int, micro_batch: int, dp_rank: int, dp_size: int,
+                 start_sample: int = 0):
         self.total, self.mbs, self.dp_rank, self.dp_size = total, micro_batch, dp_rank, dp_size
+        self.start = start_sample
 
     def __iter__(self):
         stride = self.mbs * self.dp_size
         lo = self.dp_rank * self.mbs
-        for start in range(0, self.total - stride + 1, stride):
+        for start in range(self.start, self.total - stride + 1, stride):
             yield list(range(start + lo, start + lo + self.mbs))
 
     def __len__(self):

@@ -71,6 +71,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--eval-interval", type=int, default=0)
     g.add_argument("--eval-iters", type=int, default=2)
     g.add_argument("--exit-interval", type=int, default=None)
+    g.add_argument("--exit-signal-handler", action="store_true")
+    g.add_argument("--exit-duration-in-mins", type=float, default=None)
     g.add_argument("--recompute-granularity", choices=["full", "selective"], default=None)
     g.add_argument("--recompute-num-layers", type=int, default=None)
     g.add_argument("--bf16", action="store_true")

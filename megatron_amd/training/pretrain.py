@@ -9,6 +9,7 @@ over RCCL.
 from __future__ import annotations
 
 import os
+import signal
 import sys
 import time
 from typing import Callable, Optional
@@ -60,15 +61,66 @@ def initialize(args):
     model_parallel_seed(args.seed)
 
 
-def build_data_iterator(args, device):
+def build_data_iterator(args, device, start_sample: int = 0, split: str = "train"):
     grid = G.get_grid()
     if args.mock_data or not args.data_path:
+        # deterministic restart: fold the consumed-sample count into the seed
         it = MockGPTDataIterator(args.micro_batch_size, args.seq_length, args.vocab_size,
-                                 seed=args.seed, device=device, dp_rank=grid.rank_in("dp_cp"))
+                                 seed=args.seed + (0 if split == "train" else 7919) + start_sample,
+                                 device=device, dp_rank=grid.rank_in("dp_cp"))
         return iter(it)
     from megatron_amd.datasets.gpt_dataset import build_gpt_train_iterator
 
-    return build_gpt_train_iterator(args, device, grid.rank_in("dp_cp"), grid.size("dp_cp"))
+    return build_gpt_train_iterator(args, device, grid.rank_in("dp_cp"), grid.size("dp_cp"),
+                                    start_sample=start_sample, split=split)
+
+
+@torch.no_grad()
+def evaluate(forward_step, args, chunks, cfg, device) -> float:
+    """Mean validation loss over --eval-iters batches (reference
+    training.py:4202 evaluate)."""
+    for m in chunks:
+        m.eval()
+    it = build_data_iterator(args, str(device), split="valid")
+    total = torch.zeros((), dtype=torch.float32)
+    ntok = torch.zeros((), dtype=torch.float64)
+    for _ in range(args.eval_iters):
+        batch = next(it)
+        out, loss_func = forward_step(iter([batch]), chunks[-1])
+        loss_sum, n, _ = loss_func(out if out.dim() == 2 else out)
+        total += loss_sum.detach().float().cpu()
+        ntok += float(n)
+    if dist.is_initialized():
+        grid = G.get_grid()
+        group = grid.group("dp_cp")
+        if group is not None and dist.get_world_size(group) > 1:
+            t = torch.stack([total.double(), ntok])
+            dist.all_reduce(t, group=group)
+            total, ntok = t[0].float(), t[1]
+    for m in chunks:
+        m.train()
+    return float(total / max(float(ntok), 1.0))
+
+
+class DistSignalHandler:
+    """SIGTERM-driven save-and-exit (reference training/dist_signal_handler.py):
+    the flag is all-reduced so every rank exits at the same iteration."""
+
+    def __init__(self, sig=signal.SIGTERM):
+        self.signal_received = False
+        try:
+            signal.signal(sig, self._handler)
+        except ValueError:
+            pass  # not the main thread (tests)
+
+    def _handler(self, signum, frame):
+        self.signal_received = True
+
+    def should_exit(self) -> bool:
+        flag = torch.tensor([1.0 if self.signal_received else 0.0])
+        if dist.is_initialized():
+            dist.all_reduce(flag, op=dist.ReduceOp.MAX)
+        return bool(flag.item())
 
 
 def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
@@ -95,6 +147,8 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                                       args.recompute_granularity == "full")))
 
     iteration = 0
+    start_time = time.time()
+    sig_handler = DistSignalHandler() if args.exit_signal_handler else None
     if args.load:
         try:
             iteration = load_checkpoint(args.load, chunks, optimizer, scheduler,
@@ -104,7 +158,9 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             _print_rank0(f"no checkpoint found in {args.load}; starting fresh")
 
     n_chunks = len(chunks)
-    data_iters = [RerunDataIterator(build_data_iterator(args, str(device))) for _ in range(n_chunks)]
+    consumed = iteration * args.global_batch_size  # dataloader resume point
+    data_iters = [RerunDataIterator(build_data_iterator(args, str(device), start_sample=consumed))
+                  for _ in range(n_chunks)]
 
     def forward_step(data_iterator, model):
         batch = next(data_iterator)
@@ -196,12 +252,27 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                                  f"max rank {rep.max_rank} {rep.max_time_ms:.1f}ms | "
                                  f"mean {rep.mean_time_ms:.1f}ms | power {rep.power_w}W | temp {rep.temp_c}C")
 
+        if args.eval_interval and iteration % args.eval_interval == 0 and args.eval_iters:
+            val = evaluate(forward_step, args, chunks, cfg, device)
+            _print_rank0(f"validation loss at iteration {iteration}: {val:.4f}")
+            metrics.log(iteration, valid_loss=val)
+
         if args.save and args.save_interval and iteration % args.save_interval == 0:
             save_checkpoint(args.save, chunks, optimizer, iteration, scheduler,
                             async_save=args.async_save)
             _print_rank0(f"saved checkpoint at iteration {iteration}")
         if args.exit_interval and iteration % args.exit_interval == 0:
             _print_rank0(f"exiting at iteration {iteration} (--exit-interval)")
+            break
+        if sig_handler is not None and sig_handler.should_exit():
+            _print_rank0(f"SIGTERM received: checkpoint and exit at iteration {iteration}")
+            if args.save:
+                save_checkpoint(args.save, chunks, optimizer, iteration, scheduler)
+            break
+        if args.exit_duration_in_mins and (time.time() - start_time) / 60 > args.exit_duration_in_mins:
+            _print_rank0(f"exiting after {args.exit_duration_in_mins} min (--exit-duration-in-mins)")
+            if args.save:
+                save_checkpoint(args.save, chunks, optimizer, iteration, scheduler)
             break
 
     if args.save:
