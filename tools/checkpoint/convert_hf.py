@@ -1,0 +1,150 @@
+"""HuggingFace <-> megatron_amd checkpoint conversion (Llama family).
+
+Capability analog of reference tools/checkpoint/convert.py with the
+loader_llama/saver pairs: maps HF `LlamaForCausalLM` weights to our GPTModel
+layout (fused per-query-group QKV, concatenated gated fc1) and back.
+
+    # HF -> megatron_amd (single full-model checkpoint, TP/PP reshard on load)
+    python tools/checkpoint/convert_hf.py --load-hf <hf_dir> --save <ckpt_dir>
+
+    # megatron_amd -> HF
+    python tools/checkpoint/convert_hf.py --load <ckpt_dir> --save-hf <hf_dir>
+"""
+
+from __future__ import annotations
+
+import argparse
+import sys
+
+sys.path.insert(0, ".")
+
+import torch  # noqa: E402
+
+
+def hf_to_mcore_state_dict(hf_sd: dict, cfg) -> dict:
+    """HF LlamaForCausalLM names -> our GPTModel names, with QKV fusion.
+
+    Fused QKV row layout (attention.py GQA split): for each query group g:
+    [rep*d rows of Q heads g*rep..(g+1)*rep) | d rows of K head g | d rows of
+    V head g]. Gated fc1 = [gate ; up].
+    """
+    d = cfg.kv_channels
+    ng = cfg.num_query_groups or cfg.num_attention_heads
+    rep = cfg.num_attention_heads // ng
+    out = {"embedding.weight": hf_sd["model.embed_tokens.weight"]}
+    for i in range(cfg.num_layers):
+        hf = f"model.layers.{i}."
+        us = f"decoder.layers.{i}."
+        q = hf_sd[hf + "self_attn.q_proj.weight"]
+        k = hf_sd[hf + "self_attn.k_proj.weight"]
+        v = hf_sd[hf + "self_attn.v_proj.weight"]
+        groups = []
+        for g in range(ng):
+            groups.append(q[g * rep * d:(g + 1) * rep * d])
+            groups.append(k[g * d:(g + 1) * d])
+            groups.append(v[g * d:(g + 1) * d])
+        out[us + "self_attention.linear_qkv.weight"] = torch.cat(groups, dim=0)
+        out[us + "self_attention.linear_proj.weight"] = hf_sd[hf + "self_attn.o_proj.weight"]
+        out[us + "mlp.linear_fc1.weight"] = torch.cat(
+            [hf_sd[hf + "mlp.gate_proj.weight"], hf_sd[hf + "mlp.up_proj.weight"]], dim=0)
+        out[us + "mlp.linear_fc2.weight"] = hf_sd[hf + "mlp.down_proj.weight"]
+        out[us + "input_layernorm.weight"] = hf_sd[hf + "input_layernorm.weight"]
+        out[us + "pre_mlp_layernorm.weight"] = hf_sd[hf + "post_attention_layernorm.weight"]
+    out["decoder.final_layernorm.weight"] = hf_sd["model.norm.weight"]
+    if "lm_head.weight" in hf_sd:
+        out["output_layer.weight"] = hf_sd["lm_head.weight"]
+    else:  # tied embeddings
+        out["output_layer.weight"] = hf_sd["model.embed_tokens.weight"]
+    return out
+
+
+def mcore_to_hf_state_dict(sd: dict, cfg) -> dict:
+    d = cfg.kv_channels
+    ng = cfg.num_query_groups or cfg.num_attention_heads
+    rep = cfg.num_attention_heads // ng
+    ffn = cfg.ffn_hidden_size
+    out = {"model.embed_tokens.weight": sd["embedding.weight"]}
+    for i in range(cfg.num_layers):
+        hf = f"model.layers.{i}."
+        us = f"decoder.layers.{i}."
+        qkv = sd[us + "self_attention.linear_qkv.weight"]
+        gsz = (rep + 2) * d
+        qs, ks, vs = [], [], []
+        for g in range(ng):
+            blk = qkv[g * gsz:(g + 1) * gsz]
+            qs.append(blk[: rep * d])
+            ks.append(blk[rep * d: rep * d + d])
+            vs.append(blk[rep * d + d:])
+        out[hf + "self_attn.q_proj.weight"] = torch.cat(qs, 0)
+        out[hf + "self_attn.k_proj.weight"] = torch.cat(ks, 0)
+        out[hf + "self_attn.v_proj.weight"] = torch.cat(vs, 0)
+        out[hf + "self_attn.o_proj.weight"] = sd[us + "self_attention.linear_proj.weight"]
+        fc1 = sd[us + "mlp.linear_fc1.weight"]
+        out[hf + "mlp.gate_proj.weight"] = fc1[:ffn]
+        out[hf + "mlp.up_proj.weight"] = fc1[ffn:]
+        out[hf + "mlp.down_proj.weight"] = sd[us + "mlp.linear_fc2.weight"]
+        out[hf + "input_layernorm.weight"] = sd[us + "input_layernorm.weight"]
+        out[hf + "post_attention_layernorm.weight"] = sd[us + "pre_mlp_layernorm.weight"]
+    out["model.norm.weight"] = sd["decoder.final_layernorm.weight"]
+    out["lm_head.weight"] = sd["output_layer.weight"]
+    return out
+
+
+def config_from_hf(hf_cfg):
+    from megatron_amd.config import TransformerConfig
+
+    return TransformerConfig(
+        num_layers=hf_cfg.num_hidden_layers,
+        hidden_size=hf_cfg.hidden_size,
+        num_attention_heads=hf_cfg.num_attention_heads,
+        num_query_groups=getattr(hf_cfg, "num_key_value_heads", hf_cfg.num_attention_heads),
+        ffn_hidden_size=hf_cfg.intermediate_size,
+        vocab_size=hf_cfg.vocab_size,
+        max_position_embeddings=hf_cfg.max_position_embeddings,
+        rotary_base=getattr(hf_cfg, "rope_theta", 10000.0),
+        layernorm_epsilon=hf_cfg.rms_norm_eps,
+        untie_embeddings_and_output_weights=not getattr(hf_cfg, "tie_word_embeddings", False),
+    )
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser()
+    p.add_argument("--load-hf", type=str, default=None, help="HF model dir")
+    p.add_argument("--save", type=str, default=None, help="megatron_amd checkpoint dir")
+    p.add_argument("--load", type=str, default=None)
+    p.add_argument("--save-hf", type=str, default=None)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    args = p.parse_args(argv)
+
+    from megatron_amd.checkpoint import load_checkpoint, save_checkpoint
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel import grid as G
+
+    G.initialize_model_parallel(world_size=1, rank=0)
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+
+    if args.load_hf:
+        from transformers import AutoConfig, AutoModelForCausalLM
+
+        hf_cfg = AutoConfig.from_pretrained(args.load_hf)
+        hf = AutoModelForCausalLM.from_pretrained(args.load_hf, torch_dtype=dtype)
+        cfg = config_from_hf(hf_cfg)
+        cfg.bf16 = args.dtype == "bf16"
+        sd = hf_to_mcore_state_dict(hf.state_dict(), cfg)
+        model = GPTModel(cfg)
+        missing, unexpected = model.load_state_dict(sd, strict=False)
+        assert not unexpected, f"unexpected keys: {unexpected[:5]}"
+        save_checkpoint(args.save, [model], None, 0)
+        print(f"saved megatron_amd checkpoint to {args.save} "
+              f"({sum(p.numel() for p in model.parameters())/1e6:.1f}M params)")
+    elif args.load:
+        from transformers import AutoConfig, AutoModelForCausalLM
+
+        raise SystemExit("--load -> --save-hf requires a config source; use "
+                         "convert_to_hf() from python with an explicit TransformerConfig")
+    else:
+        p.error("need --load-hf/--save or --load/--save-hf")
+
+
+if __name__ == "__main__":
+    main()
