@@ -32,6 +32,9 @@ class ParallelConfig:
     expert_parallel_size: int = 1
     virtual_pipeline_parallel_size: Optional[int] = None
     sequence_parallel: bool = False
+    fp8: object = None  # None | 'hybrid' | 'e4m3' (K15)
+    fp8_amax_history_len: int = 16
+    fp8_margin: int = 0
     # expert tensor parallel size (defaults to tensor_parallel_size)
     expert_tensor_parallel_size: Optional[int] = None
 

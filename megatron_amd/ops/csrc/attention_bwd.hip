@@ -249,7 +249,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
 //   dK += mfma(A=dS^T_frag(permlane), B=Qt_lds)
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(512) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(512, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ drow,
     bf16* __restrict__ dk, bf16* __restrict__ dv,

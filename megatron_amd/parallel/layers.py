@@ -64,7 +64,8 @@ class _ParallelLinearFn(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, inp, weight, bias, grad_accum_fusion, async_grad_allreduce, sequence_parallel):
+    def forward(ctx, inp, weight, bias, grad_accum_fusion, async_grad_allreduce,
+                sequence_parallel, fp8_recipes=None):
         ctx.use_bias = bias is not None
         ctx.grad_accum_fusion = grad_accum_fusion
         ctx.async_grad_allreduce = async_grad_allreduce
@@ -74,6 +75,25 @@ class _ParallelLinearFn(torch.autograd.Function):
         else:
             total_input = inp
         ctx.save_for_backward(inp, weight)
+        ctx.fp8 = False
+        if fp8_recipes is not None:
+            from megatron_amd.ops import fp8 as F8
+
+            if F8.fp8_eligible(total_input, weight):
+                rx, rw, rg = fp8_recipes
+                x2d = total_input.reshape(-1, total_input.shape[-1])
+                xq, inv_x = F8.quantize_fp8(x2d, rx.scale_for(x2d))
+                wq, inv_w = F8.quantize_fp8(weight, rw.scale_for(weight))
+                output = F8._scaled_mm(xq, wq.t(), inv_x, inv_w, total_input.dtype)
+                output = output.reshape(*total_input.shape[:-1], weight.shape[0])
+                # column-major fp8 weight copy for the dgrad GEMM
+                ctx.wq_cm = wq.t().contiguous().t()
+                ctx.inv_w = inv_w
+                ctx.rg = rg
+                ctx.fp8 = True
+                if bias is not None:
+                    output = output + bias
+                return output
         output = torch.matmul(total_input, weight.t())
         if bias is not None:
             output = output + bias
@@ -99,7 +119,16 @@ class _ParallelLinearFn(torch.autograd.Function):
             total_input = inp
             gather_handle = None
 
-        grad_input = torch.matmul(grad_output, weight)
+        if ctx.fp8:
+            from megatron_amd.ops import fp8 as F8
+
+            go2d = grad_output.reshape(-1, grad_output.shape[-1])
+            dyq, inv_dy = F8.quantize_fp8(go2d, ctx.rg.scale_for(go2d), fmt=torch.float8_e5m2)
+            grad_input = torch._scaled_mm(dyq, ctx.wq_cm, scale_a=inv_dy, scale_b=ctx.inv_w,
+                                          out_dtype=grad_output.dtype)
+            grad_input = grad_input.reshape(*grad_output.shape[:-1], weight.shape[1])
+        else:
+            grad_input = torch.matmul(grad_output, weight)
 
         if gather_handle is not None:
             gather_handle.wait()
@@ -142,13 +171,15 @@ class _ParallelLinearFn(torch.autograd.Function):
         if allreduce_handle is not None:
             allreduce_handle.wait()
 
-        return grad_input, grad_weight, grad_bias, None, None, None
+        return grad_input, grad_weight, grad_bias, None, None, None, None
 
 
 def linear_with_grad_accumulation_and_async_allreduce(
-    inp, weight, bias, grad_accum_fusion, async_grad_allreduce, sequence_parallel
+    inp, weight, bias, grad_accum_fusion, async_grad_allreduce, sequence_parallel,
+    fp8_recipes=None,
 ):
-    return _ParallelLinearFn.apply(inp, weight, bias, grad_accum_fusion, async_grad_allreduce, sequence_parallel)
+    return _ParallelLinearFn.apply(inp, weight, bias, grad_accum_fusion,
+                                   async_grad_allreduce, sequence_parallel, fp8_recipes)
 
 
 def _init_weight(weight: torch.Tensor, init_std: float, generator: Optional[torch.Generator] = None):
@@ -188,6 +219,12 @@ class ColumnParallelLinear(nn.Module):
             config.async_tensor_model_parallel_allreduce and tp > 1 and not self.sequence_parallel
         )
         self.grad_accum_fusion = config.gradient_accumulation_fusion
+        self.fp8_recipes = None
+        if getattr(config, "fp8", None):
+            from megatron_amd.ops.fp8 import make_recipes
+
+            self.fp8_recipes = make_recipes(config.fp8, config.fp8_amax_history_len,
+                                            config.fp8_margin)
         self.weight = nn.Parameter(
             torch.empty(self.output_size_per_partition, input_size, dtype=config.params_dtype)
         )
@@ -213,7 +250,8 @@ class ColumnParallelLinear(nn.Module):
         else:
             inp = x
         out = linear_with_grad_accumulation_and_async_allreduce(
-            inp, self.weight, bias, self.grad_accum_fusion, self.async_tp_allreduce, self.sequence_parallel
+            inp, self.weight, bias, self.grad_accum_fusion, self.async_tp_allreduce,
+            self.sequence_parallel, self.fp8_recipes if self.training else None
         )
         if self.gather_output:
             from megatron_amd.parallel.mappings import gather_from_tensor_model_parallel_region
@@ -251,6 +289,12 @@ class RowParallelLinear(nn.Module):
         self.skip_bias_add = skip_bias_add
         self.sequence_parallel = config.sequence_parallel and tp > 1
         self.grad_accum_fusion = config.gradient_accumulation_fusion
+        self.fp8_recipes = None
+        if getattr(config, "fp8", None):
+            from megatron_amd.ops.fp8 import make_recipes
+
+            self.fp8_recipes = make_recipes(config.fp8, config.fp8_amax_history_len,
+                                            config.fp8_margin)
         self.weight = nn.Parameter(
             torch.empty(output_size, self.input_size_per_partition, dtype=config.params_dtype)
         )
@@ -269,7 +313,8 @@ class RowParallelLinear(nn.Module):
     def forward(self, x):
         assert self.input_is_parallel
         out_parallel = linear_with_grad_accumulation_and_async_allreduce(
-            x, self.weight, None, self.grad_accum_fusion, False, False
+            x, self.weight, None, self.grad_accum_fusion, False, False,
+            self.fp8_recipes if self.training else None
         )
         if self.sequence_parallel:
             out = reduce_scatter_to_sequence_parallel_region(out_parallel)

@@ -76,6 +76,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--recompute-granularity", choices=["full", "selective"], default=None)
     g.add_argument("--recompute-num-layers", type=int, default=None)
     g.add_argument("--bf16", action="store_true")
+    g.add_argument("--fp8-format", choices=["hybrid", "e4m3"], default=None)
+    g.add_argument("--fp8-amax-history-len", type=int, default=16)
+    g.add_argument("--fp8-margin", type=int, default=0)
     g.add_argument("--fp16", action="store_true")
     g.add_argument("--seed", type=int, default=1234)
     g.add_argument("--deterministic-mode", action="store_true")
@@ -203,6 +206,9 @@ def configs_from_args(args):
         sequence_parallel=args.sequence_parallel,
         bf16=args.bf16,
         fp16=args.fp16,
+        fp8=args.fp8_format,
+        fp8_amax_history_len=args.fp8_amax_history_len,
+        fp8_margin=args.fp8_margin,
         recompute_granularity=args.recompute_granularity,
         recompute_num_layers=args.recompute_num_layers,
         deterministic_mode=args.deterministic_mode,
