@@ -112,9 +112,9 @@ class StaticInferenceContext:
 class KVBlockAllocator:
     """Fixed-size KV block free-list (reference kv_block_allocator.py)."""
 
-    def __init__(self, num_blocks: int):
+    def __init__(self, num_blocks: int, first_id: int = 0):
         self.num_blocks = num_blocks
-        self._free = list(range(num_blocks - 1, -1, -1))
+        self._free = list(range(first_id + num_blocks - 1, first_id - 1, -1))
 
     @property
     def num_free(self) -> int:
@@ -142,11 +142,14 @@ class DynamicInferenceContext:
                  dtype=torch.bfloat16, device="cuda"):
         self.block_size = block_size
         self.num_layers = num_layers
-        self.allocator = KVBlockAllocator(num_blocks)
-        self.k_cache = [torch.empty(num_blocks, block_size, num_kv_heads, head_dim,
+        # physical block 0 is a scratch target for padded rows in hipGraph
+        # decode replays; the allocator only hands out ids 1..num_blocks
+        self.allocator = KVBlockAllocator(num_blocks, first_id=1)
+        self.k_cache = [torch.empty(num_blocks + 1, block_size, num_kv_heads, head_dim,
                                     dtype=dtype, device=device) for _ in range(num_layers)]
         self.v_cache = [torch.empty_like(self.k_cache[0]) for _ in range(num_layers)]
         self.device = device
+        self._static = False
         # step state, set by the engine before each forward
         self._mode: str = "decode"  # or "prefill"
         self._block_tables: Optional[torch.Tensor] = None  # [b, max_blocks] int64
@@ -156,11 +159,22 @@ class DynamicInferenceContext:
     # -- engine-facing step setup -------------------------------------------
 
     def begin_prefill(self, block_table: List[int], prior_len: int):
+        self._static = False
         self._mode = "prefill"
         self._prefill_table = torch.as_tensor(block_table, dtype=torch.long, device=self.device)
         self._prior_len = prior_len
 
+    def begin_decode_static(self, tables_buf: torch.Tensor, lens_buf: torch.Tensor):
+        """Graph-capture decode mode: fixed-shape persistent buffers; the
+        gather always spans the full (padded) table width so no host-side
+        data-dependent shapes appear inside the captured region."""
+        self._mode = "decode"
+        self._static = True
+        self._block_tables = tables_buf
+        self._context_lens = lens_buf
+
     def begin_decode(self, block_tables: List[List[int]], context_lens: List[int]):
+        self._static = False
         self._mode = "decode"
         maxb = max(len(t) for t in block_tables)
         bt = torch.zeros(len(block_tables), maxb, dtype=torch.long, device=self.device)
@@ -211,8 +225,11 @@ class DynamicInferenceContext:
         flat_k[slots] = k[0]
         flat_v[slots] = v[0]
         new_lens = lens + 1
-        max_blocks_needed = int(torch.div(new_lens.max() + bs - 1, bs, rounding_mode="floor"))
-        tables = self._block_tables[:, :max_blocks_needed]  # [b, nb]
+        if self._static:
+            tables = self._block_tables  # fixed width under graph capture
+        else:
+            max_blocks_needed = int(torch.div(new_lens.max() + bs - 1, bs, rounding_mode="floor"))
+            tables = self._block_tables[:, :max_blocks_needed]  # [b, nb]
         # gather [b, nb*bs, hkv, d] via block-id expansion
         slot_grid = (tables.unsqueeze(-1) * bs +
                      torch.arange(bs, device=self.device).view(1, 1, bs)).reshape(b, -1)

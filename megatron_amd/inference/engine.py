@@ -24,6 +24,60 @@ from megatron_amd.parallel import grid as G
 from megatron_amd.parallel.mappings import gather_from_tensor_model_parallel_region
 
 
+class _DecodeGraphRunner:
+    """hipGraph-captured decode steps (reference: per-step CUDA-graph decode,
+    dynamic_engine.py). One graph per (batch-bucket, table-width-bucket);
+    inputs are copied into persistent buffers, padded rows write their KV into
+    the reserved scratch block 0. Decode is launch-bound at small batch - the
+    whole multi-layer step replays as one hipGraph."""
+
+    def __init__(self, model, context, device):
+        self.model, self.context, self.device = model, context, device
+        self.graphs = {}
+
+    @staticmethod
+    def _bucket(n: int, lo: int = 8) -> int:
+        b = lo
+        while b < n:
+            b *= 2
+        return b
+
+    def _build(self, bb: int, nbb: int):
+        toks = torch.zeros(bb, 1, dtype=torch.long, device=self.device)
+        tables = torch.zeros(bb, nbb, dtype=torch.long, device=self.device)
+        lens = torch.ones(bb, dtype=torch.long, device=self.device)
+        self.context.begin_decode_static(tables, lens)
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # warmup per graph-capture rules
+                self.model(toks, inference_context=self.context)
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        self.context.begin_decode_static(tables, lens)
+        with torch.cuda.graph(g):
+            out = self.model(toks, inference_context=self.context)
+        return {"graph": g, "toks": toks, "tables": tables, "lens": lens, "out": out}
+
+    def run(self, tok_list, tables_list, lens_list) -> torch.Tensor:
+        b = len(tok_list)
+        nb = max(len(t) for t in tables_list)
+        key = (self._bucket(b), self._bucket(nb, lo=1))
+        entry = self.graphs.get(key)
+        if entry is None:
+            entry = self._build(*key)
+            self.graphs[key] = entry
+        entry["toks"].zero_()
+        entry["tables"].zero_()
+        entry["lens"].fill_(1)
+        entry["toks"][:b, 0] = torch.as_tensor(tok_list, device=self.device)
+        for i, t in enumerate(tables_list):
+            entry["tables"][i, : len(t)] = torch.as_tensor(t, dtype=torch.long, device=self.device)
+        entry["lens"][:b] = torch.as_tensor(lens_list, device=self.device)
+        entry["graph"].replay()
+        return entry["out"][0, :b]
+
+
 @dataclass
 class GenerationResult:
     request_id: int
@@ -126,7 +180,8 @@ class DynamicInferenceEngine:
     """Continuous batching over a paged KV cache."""
 
     def __init__(self, model, tokenizer=None, num_blocks: int = 512, block_size: int = 256,
-                 max_batch: int = 64, max_prefill_tokens: int = 8192, device=None):
+                 max_batch: int = 64, max_prefill_tokens: int = 8192, device=None,
+                 use_hip_graphs: bool = True):
         self.model = model.eval()
         self.tokenizer = tokenizer
         cfg = model.config
@@ -141,6 +196,10 @@ class DynamicInferenceEngine:
             dtype=cfg.params_dtype, device=device)
         self.max_batch = max_batch
         self.max_prefill_tokens = max_prefill_tokens
+        self._graphs = None
+        if (use_hip_graphs and torch.cuda.is_available() and device.type == "cuda"
+                and tp == 1):
+            self._graphs = _DecodeGraphRunner(self.model, self.context, device)
         self._ids = itertools.count()
         self.waiting: List[_Request] = []
         self.active: List[_Request] = []
@@ -229,13 +288,29 @@ class DynamicInferenceEngine:
             assert ok, "KV pool exhausted during decode"  # TODO: preemption
         tables = [r.block_table for r in batch]
         lens = [len(r.prompt) + len(r.result.output_tokens) - 1 for r in batch]  # cached so far
-        self.context.begin_decode(tables, lens)
-        toks = torch.as_tensor([r.next_input for r in batch], device=self.device).view(-1, 1)
-        logits_tp = self.model(toks, inference_context=self.context)  # [1, b, V/tp]
-        logits = _full_logits(logits_tp[0]).float()
+        if self._graphs is not None:
+            logits_row = self._graphs.run([r.next_input for r in batch], tables, lens)
+            logits = _full_logits(logits_row).float()
+        else:
+            self.context.begin_decode(tables, lens)
+            toks = torch.as_tensor([r.next_input for r in batch], device=self.device).view(-1, 1)
+            logits_tp = self.model(toks, inference_context=self.context)  # [1, b, V/tp]
+            logits = _full_logits(logits_tp[0]).float()
+        # batched sampling fast path: all requests share one params object
+        batch_toks = None
+        p0 = batch[0].params
+        if all((r.params is p0) and r.gen is None for r in batch):
+            batch_toks = sample(logits, p0).tolist()
+            if p0.return_log_probs:
+                lps = log_prob_of(logits, torch.as_tensor(batch_toks, device=logits.device)).tolist()
         still = []
         for i, req in enumerate(batch):
-            tok = self._sample_row(logits[i], req)
+            if batch_toks is not None:
+                tok = int(batch_toks[i])
+                if p0.return_log_probs:
+                    req.result.log_probs.append(float(lps[i]))
+            else:
+                tok = self._sample_row(logits[i], req)
             req.result.output_tokens.append(tok)
             if req.params.stop_on_eod and tok == self.eod:
                 req.result.output_tokens.pop()

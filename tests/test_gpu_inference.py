@@ -62,3 +62,14 @@ def test_dynamic_engine_gpu_matches_static(gpu_model):
                                      max_prefill_tokens=16).generate(prompts, params)
     for a, b in zip(static, dynamic):
         assert a.output_tokens == b.output_tokens
+
+
+def test_graph_decode_matches_eager(gpu_model):
+    prompts = [[3, 7, 11, 2, 9], [5, 1], list(range(40, 80))]
+    params = SamplingParams(max_tokens=8, greedy=True, stop_on_eod=False)
+    eager = DynamicInferenceEngine(gpu_model, num_blocks=32, block_size=16,
+                                   use_hip_graphs=False).generate(prompts, params)
+    graphed = DynamicInferenceEngine(gpu_model, num_blocks=32, block_size=16,
+                                     use_hip_graphs=True).generate(prompts, params)
+    for a, b in zip(eager, graphed):
+        assert a.output_tokens == b.output_tokens
