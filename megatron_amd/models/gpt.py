@@ -55,6 +55,14 @@ class GPTModel(nn.Module):
             )
             if self.share_embeddings_and_output_weights and pre_process:
                 self.output_layer.weight = self.embedding.weight
+            self.mtp = None
+            if config.mtp_num_layers:
+                assert pre_process, "MTP v1 needs the embedding on this stage (pp=1)"
+                from megatron_amd.transformer.multi_token_prediction import (
+                    MultiTokenPredictionBlock,
+                )
+
+                self.mtp = MultiTokenPredictionBlock(config)
         self._rope_cache = {}
         # set by pipeline runner between stages
         self.input_tensor: Optional[torch.Tensor] = None
@@ -151,4 +159,8 @@ class GPTModel(nn.Module):
             return logits
         labels_sb = labels.transpose(0, 1).contiguous()  # [s, b]
         loss = vocab_parallel_cross_entropy(logits, labels_sb)
+        if self.mtp is not None and input_ids is not None:
+            loss = loss + self.mtp(
+                hidden, input_ids, labels, self.embedding, self.output_layer,
+                rotary, vocab_parallel_cross_entropy)
         return loss

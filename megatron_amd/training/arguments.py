@@ -41,6 +41,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--window-size", type=int, default=None)
     g.add_argument("--qk-layernorm", action="store_true")
     g.add_argument("--init-method-std", type=float, default=0.02)
+    g.add_argument("--mtp-num-layers", type=int, default=0)
+    g.add_argument("--mtp-loss-scaling-factor", type=float, default=0.1)
 
     g = p.add_argument_group("moe")
     g.add_argument("--num-experts", type=int, default=None)
@@ -187,6 +189,8 @@ def configs_from_args(args):
         window_size=args.window_size,
         qk_layernorm=args.qk_layernorm,
         init_method_std=args.init_method_std,
+        mtp_num_layers=args.mtp_num_layers,
+        mtp_loss_scaling_factor=args.mtp_loss_scaling_factor,
         num_experts=args.num_experts,
         moe_router_topk=args.moe_router_topk,
         moe_ffn_hidden_size=args.moe_ffn_hidden_size,
