@@ -30,6 +30,7 @@ class ParallelConfig:
     context_parallel_size: int = 1
     cp_comm_type: str = "p2p"  # ring (p2p) or Ulysses head-scatter (a2a)
     mtp_num_layers: int = 0  # multi-token prediction depths (0 = off)
+    causal_attention: bool = True  # False = bidirectional encoder (BERT)
     mtp_loss_scaling_factor: float = 0.1
     expert_parallel_size: int = 1
     virtual_pipeline_parallel_size: Optional[int] = None

@@ -92,7 +92,8 @@ class SelfAttention(nn.Module):
             else:
                 core_out = ring_attention(q, k, v, scale=self.softmax_scale)
         else:
-            core_out = ops.flash_attention(q, k, v, causal=True, scale=self.softmax_scale, window=self.window)
+            core_out = ops.flash_attention(q, k, v, causal=self.config.causal_attention,
+                                           scale=self.softmax_scale, window=self.window)
         core_out = core_out.reshape(s, b, ng * rep * d)
         out, _ = self.linear_proj(core_out)
         return out
