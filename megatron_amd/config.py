@@ -31,6 +31,13 @@ class ParallelConfig:
     cp_comm_type: str = "p2p"  # ring (p2p) or Ulysses head-scatter (a2a)
     mtp_num_layers: int = 0  # multi-token prediction depths (0 = off)
     causal_attention: bool = True  # False = bidirectional encoder (BERT)
+    # multi-latent attention (DeepSeek): low-rank q/kv + decoupled rope
+    multi_latent_attention: bool = False
+    q_lora_rank: Optional[int] = None
+    kv_lora_rank: int = 512
+    qk_nope_head_dim: int = 64
+    qk_rope_head_dim: int = 64
+    v_head_dim: int = 128
     mtp_loss_scaling_factor: float = 0.1
     expert_parallel_size: int = 1
     virtual_pipeline_parallel_size: Optional[int] = None

@@ -42,6 +42,12 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--qk-layernorm", action="store_true")
     g.add_argument("--init-method-std", type=float, default=0.02)
     g.add_argument("--mtp-num-layers", type=int, default=0)
+    g.add_argument("--multi-latent-attention", action="store_true")
+    g.add_argument("--q-lora-rank", type=int, default=None)
+    g.add_argument("--kv-lora-rank", type=int, default=512)
+    g.add_argument("--qk-nope-head-dim", type=int, default=64)
+    g.add_argument("--qk-rope-head-dim", type=int, default=64)
+    g.add_argument("--v-head-dim", type=int, default=128)
     g.add_argument("--mtp-loss-scaling-factor", type=float, default=0.1)
 
     g = p.add_argument_group("moe")
@@ -190,6 +196,12 @@ def configs_from_args(args):
         qk_layernorm=args.qk_layernorm,
         init_method_std=args.init_method_std,
         mtp_num_layers=args.mtp_num_layers,
+        multi_latent_attention=args.multi_latent_attention,
+        q_lora_rank=args.q_lora_rank,
+        kv_lora_rank=args.kv_lora_rank,
+        qk_nope_head_dim=args.qk_nope_head_dim,
+        qk_rope_head_dim=args.qk_rope_head_dim,
+        v_head_dim=args.v_head_dim,
         mtp_loss_scaling_factor=args.mtp_loss_scaling_factor,
         num_experts=args.num_experts,
         moe_router_topk=args.moe_router_topk,

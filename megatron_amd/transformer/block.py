@@ -57,7 +57,12 @@ class TransformerLayer(nn.Module):
         self.config = config
         self.layer_number = layer_number
         self.input_layernorm = Norm(config)
-        self.self_attention = SelfAttention(config, layer_number=layer_number)
+        if config.multi_latent_attention:
+            from megatron_amd.transformer.multi_latent_attention import MLASelfAttention
+
+            self.self_attention = MLASelfAttention(config, layer_number=layer_number)
+        else:
+            self.self_attention = SelfAttention(config, layer_number=layer_number)
         self.pre_mlp_layernorm = Norm(config)
         self.mlp = _make_mixer(config, layer_number)
         self.hidden_dropout = config.hidden_dropout
