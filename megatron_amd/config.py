@@ -99,6 +99,18 @@ class TransformerConfig(ParallelConfig):
     mtp_num_layers: int = 0
     mtp_loss_scaling_factor: float = 0.1
 
+    # ---- Mamba / hybrid SSM (reference ssm/mamba_mixer.py:144) ----
+    mamba_state_dim: int = 128
+    mamba_head_dim: int = 64
+    mamba_num_groups: int = 8
+    mamba_d_conv: int = 4
+    mamba_expand: int = 2
+    mamba_chunk_size: int = 128
+    # hybrid layer allocation: explicit pattern ("M"/"*"/"-") wins over ratios
+    hybrid_override_pattern: str = ""
+    hybrid_attention_ratio: float = 0.0
+    hybrid_mlp_ratio: float = 0.0
+
     # ---- numerics ----
     params_dtype: torch.dtype = torch.float32
     bf16: bool = False

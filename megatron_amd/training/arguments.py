@@ -43,6 +43,13 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--init-method-std", type=float, default=0.02)
     g.add_argument("--mtp-num-layers", type=int, default=0)
     g.add_argument("--multi-latent-attention", action="store_true")
+    # Mamba / hybrid SSM (reference ssm/ CLI surface)
+    g.add_argument("--mamba-state-dim", type=int, default=128)
+    g.add_argument("--mamba-head-dim", type=int, default=64)
+    g.add_argument("--mamba-num-groups", type=int, default=8)
+    g.add_argument("--hybrid-override-pattern", type=str, default="")
+    g.add_argument("--hybrid-attention-ratio", type=float, default=0.0)
+    g.add_argument("--hybrid-mlp-ratio", type=float, default=0.0)
     g.add_argument("--q-lora-rank", type=int, default=None)
     g.add_argument("--kv-lora-rank", type=int, default=512)
     g.add_argument("--qk-nope-head-dim", type=int, default=64)
@@ -197,6 +204,12 @@ def configs_from_args(args):
         init_method_std=args.init_method_std,
         mtp_num_layers=args.mtp_num_layers,
         multi_latent_attention=args.multi_latent_attention,
+        mamba_state_dim=args.mamba_state_dim,
+        mamba_head_dim=args.mamba_head_dim,
+        mamba_num_groups=args.mamba_num_groups,
+        hybrid_override_pattern=args.hybrid_override_pattern,
+        hybrid_attention_ratio=args.hybrid_attention_ratio,
+        hybrid_mlp_ratio=args.hybrid_mlp_ratio,
         q_lora_rank=args.q_lora_rank,
         kv_lora_rank=args.kv_lora_rank,
         qk_nope_head_dim=args.qk_nope_head_dim,
