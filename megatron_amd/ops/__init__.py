@@ -188,7 +188,9 @@ class _FlashAttnFn(torch.autograd.Function):
 def flash_attention(q, k, v, causal=True, scale=None, window=None):
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    if _use_native(q):
+    # the MFMA kernel tiles for d in {64, 128}; other (test-size) head dims
+    # take the plain-PyTorch path
+    if _use_native(q) and q.shape[-1] in (64, 128):
         return _FlashAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale, window)
     return ref.attention(q, k, v, causal=causal, scale=scale, window=window)
 

@@ -32,7 +32,7 @@ def test_bert_gpu_bf16():
     torch.manual_seed(0)
     from megatron_amd.models.bert import BertModel
 
-    cfg = TransformerConfig(num_layers=2, hidden_size=128, num_attention_heads=8,
+    cfg = TransformerConfig(num_layers=2, hidden_size=512, num_attention_heads=8,
                             vocab_size=512, max_position_embeddings=128, bf16=True,
                             position_embedding_type="learned")
     model = BertModel(cfg).cuda()
@@ -46,7 +46,7 @@ def test_t5_gpu_bf16():
     torch.manual_seed(1)
     from megatron_amd.models.t5 import T5Model
 
-    cfg = TransformerConfig(num_layers=2, hidden_size=128, num_attention_heads=8,
+    cfg = TransformerConfig(num_layers=2, hidden_size=512, num_attention_heads=8,
                             vocab_size=512, max_position_embeddings=128, bf16=True)
     model = T5Model(cfg).cuda()
     enc = torch.randint(0, 512, (2, 48), device="cuda")
@@ -77,7 +77,7 @@ def test_mamba_gpu_bf16_train_and_decode():
     torch.manual_seed(3)
     from megatron_amd.models.mamba import MambaModel
 
-    cfg = TransformerConfig(num_layers=3, hidden_size=256, num_attention_heads=8,
+    cfg = TransformerConfig(num_layers=3, hidden_size=256, num_attention_heads=4,
                             vocab_size=512, max_position_embeddings=256, bf16=True,
                             mamba_state_dim=64, mamba_head_dim=64, mamba_num_groups=2,
                             mamba_chunk_size=32, hybrid_override_pattern="M*M")
@@ -120,7 +120,7 @@ def test_fsdp_gpu_single_rank():
         from megatron_amd.distributed.fsdp import FullyShardedDataParallel
         from megatron_amd.models.gpt import GPTModel
 
-        cfg = TransformerConfig(num_layers=2, hidden_size=128, num_attention_heads=8,
+        cfg = TransformerConfig(num_layers=2, hidden_size=512, num_attention_heads=8,
                                 vocab_size=256, max_position_embeddings=128, bf16=True)
         torch.manual_seed(4)
         model = GPTModel(cfg).cuda()
