@@ -213,7 +213,11 @@ class FullyShardedDataParallel(nn.Module):
             # rebuild rank-0's flat buffer on every rank
             flat = torch.empty(u.flat_size, dtype=u.dtype, device=u.model_shard.device)
             dist.all_gather_into_tensor(flat, u.model_shard, group=self.dp_group)
-            dist.broadcast(flat, src=dist.get_global_rank(self.dp_group, 0), group=self.dp_group)
+            if self.dp_group is None or self.dp_group is dist.group.WORLD:
+                src = 0
+            else:
+                src = dist.get_global_rank(self.dp_group, 0)
+            dist.broadcast(flat, src=src, group=self.dp_group)
             u.model_shard.copy_(flat[u.dp_rank * u.shard_size : (u.dp_rank + 1) * u.shard_size])
             u.master_shard.data.copy_(u.model_shard.float())
 
