@@ -52,19 +52,44 @@ def _make_mixer(config, layer_number: int):
 
 
 class TransformerLayer(nn.Module):
-    def __init__(self, config, layer_number: int = 0):
+    """One decoder layer.  `submodules` (a TransformerLayerSubmodules spec,
+    reference transformer_layer.py:243) overrides any slot declaratively;
+    unset slots fall back to the config-driven defaults."""
+
+    def __init__(self, config, layer_number: int = 0, submodules=None):
         super().__init__()
+        from megatron_amd.transformer.spec_utils import build_module
+
         self.config = config
         self.layer_number = layer_number
-        self.input_layernorm = Norm(config)
-        if config.multi_latent_attention:
+        sub = submodules
+        if sub is not None and sub.input_layernorm is not None:
+            self.input_layernorm = build_module(sub.input_layernorm, config)
+        else:
+            self.input_layernorm = Norm(config)
+        if sub is not None and sub.self_attention is not None:
+            self.self_attention = build_module(sub.self_attention, config, layer_number=layer_number)
+        elif config.multi_latent_attention:
             from megatron_amd.transformer.multi_latent_attention import MLASelfAttention
 
             self.self_attention = MLASelfAttention(config, layer_number=layer_number)
         else:
             self.self_attention = SelfAttention(config, layer_number=layer_number)
-        self.pre_mlp_layernorm = Norm(config)
-        self.mlp = _make_mixer(config, layer_number)
+        if sub is not None and sub.pre_mlp_layernorm is not None:
+            self.pre_mlp_layernorm = build_module(sub.pre_mlp_layernorm, config)
+        else:
+            self.pre_mlp_layernorm = Norm(config)
+        if sub is not None and sub.mlp is not None:
+            mlp_cls = sub.mlp
+            from megatron_amd.moe.moe_layer import MoELayer
+            from megatron_amd.transformer.spec_utils import get_module
+
+            if get_module(mlp_cls) is MoELayer:
+                self.mlp = build_module(mlp_cls, config, layer_number=layer_number)
+            else:
+                self.mlp = build_module(mlp_cls, config)
+        else:
+            self.mlp = _make_mixer(config, layer_number)
         self.hidden_dropout = config.hidden_dropout
 
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None):
@@ -105,17 +130,30 @@ def get_layer_offset(config, vp_stage: Optional[int] = None) -> int:
 
 
 class TransformerBlock(nn.Module):
+    """Layer stack for this pipeline stage.  `layer_specs` (optional list of
+    ModuleSpec, one per GLOBAL layer — reference get_gpt_decoder_block_spec)
+    switches construction to the declarative path."""
+
     def __init__(self, config, pre_process: bool = True, post_process: bool = True,
-                 vp_stage: Optional[int] = None):
+                 vp_stage: Optional[int] = None, layer_specs=None):
         super().__init__()
         self.config = config
         self.pre_process = pre_process
         self.post_process = post_process
         num_layers = get_num_layers_to_build(config)
         offset = get_layer_offset(config, vp_stage)
-        self.layers = nn.ModuleList(
-            [TransformerLayer(config, layer_number=offset + i) for i in range(num_layers)]
-        )
+        if layer_specs is not None:
+            from megatron_amd.transformer.spec_utils import build_module
+
+            assert len(layer_specs) == config.num_layers
+            self.layers = nn.ModuleList(
+                [build_module(layer_specs[offset + i], config, layer_number=offset + i)
+                 for i in range(num_layers)]
+            )
+        else:
+            self.layers = nn.ModuleList(
+                [TransformerLayer(config, layer_number=offset + i) for i in range(num_layers)]
+            )
         self.final_layernorm = Norm(config) if post_process else None
 
     def _checkpointed(self, layer, *args):
