@@ -92,6 +92,16 @@ class TransformerConfig(ParallelConfig):
     moe_token_dispatcher_type: str = "alltoall"  # 'alltoall' | 'allgather'
     moe_expert_capacity_factor: Optional[float] = None  # None -> dropless
     moe_router_dtype: str = "fp32"
+    # DeepSeek node-limited routing (reference moe_utils.py:673): experts split
+    # into num_groups; each token routes only within its best group_topk groups.
+    moe_router_num_groups: Optional[int] = None
+    moe_router_group_topk: Optional[int] = None
+    moe_input_jitter_eps: Optional[float] = None
+    moe_aux_loss_type: str = "aux"  # 'aux' (switch, batch-level) | 'seq_aux'
+    # aux-loss-free balancing (reference finalize_model_grads.py:334): after
+    # each step, expert_bias += rate * sign(mean_load - expert_load).
+    moe_router_enable_expert_bias: bool = False
+    moe_router_bias_update_rate: float = 1e-3
     # layer frequency: 1 = every layer is MoE, k = every k-th layer
     moe_layer_freq: int = 1
 

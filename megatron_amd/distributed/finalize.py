@@ -58,6 +58,36 @@ def _allreduce_word_embedding_grads(models: List[torch.nn.Module], config):
                     dist.all_reduce(g.data, group=grid.group("embd"))
 
 
+def update_router_expert_bias(models: List[torch.nn.Module], config):
+    """Aux-loss-free load balancing (reference finalize_model_grads.py:334,
+    `_update_router_expert_bias`): sum each router's per-step routed-token
+    counts over all data-parallel replicas, then nudge the routing bias toward
+    the mean load: bias += rate * sign(mean_load - expert_load)."""
+    if not getattr(config, "moe_router_enable_expert_bias", False):
+        return
+    from megatron_amd.moe.router import TopKRouter
+
+    routers, counts = [], []
+    for model in models:
+        for m in model.modules():
+            if isinstance(m, TopKRouter):
+                routers.append(m)
+                counts.append(m.local_tokens_per_expert)
+    if not routers:
+        return
+    stacked = torch.stack(counts)  # [L, E]
+    if G.grid_initialized() and dist.is_initialized():
+        group = G.get_grid().group("dp_cp")
+        if group is not None and dist.get_world_size(group) > 1:
+            dist.all_reduce(stacked, group=group)
+    rate = config.moe_router_bias_update_rate
+    mean_load = stacked.mean(dim=-1, keepdim=True)
+    update = rate * torch.sign(mean_load - stacked)
+    for r, u in zip(routers, update):
+        r.expert_bias += u.to(r.expert_bias.dtype)
+        r.local_tokens_per_expert.zero_()
+
+
 def finalize_model_grads(models: List[torch.nn.Module], config=None):
     if config is None:
         core = models[0].module if hasattr(models[0], "module") else models[0]
@@ -66,3 +96,4 @@ def finalize_model_grads(models: List[torch.nn.Module], config=None):
         return
     _allreduce_word_embedding_grads(models, config)
     _allreduce_layernorm_grads(models, config)
+    update_router_expert_bias(models, config)

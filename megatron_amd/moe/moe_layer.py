@@ -44,6 +44,7 @@ class MoELayer(nn.Module):
             full = hidden_states
         tokens = full.reshape(-1, h)
 
+        self.router.seq_len = full.shape[0]
         probs, top_idx = self.router(tokens)
         dispatched, tokens_per_expert = self.dispatcher.dispatch(tokens, probs, top_idx)
         expert_out = self.experts(dispatched, tokens_per_expert)

@@ -2,8 +2,10 @@
 
 Capability analog of reference megatron/core/transformer/moe/router.py:144
 (TopKRouter): fp32 gating, softmax/sigmoid score functions, pre/post-softmax
-top-k, switch load-balancing aux loss (moe_utils.py:63), z-loss, and the
-grad-injection scaler for aux losses (MoEAuxLossAutoScaler analog).
+top-k, group-limited (node-limited) top-k (moe_utils.py:673), input jitter,
+switch/sequence load-balancing aux losses (moe_utils.py:63), z-loss,
+aux-loss-free expert-bias balancing (finalize_model_grads.py:334 update rule),
+and the grad-injection scaler for aux losses (MoEAuxLossAutoScaler analog).
 """
 
 from __future__ import annotations
@@ -33,6 +35,27 @@ class AuxLossScaler(torch.autograd.Function):
         return grad_output, torch.full_like(aux, scale)
 
 
+def group_limited_topk(
+    scores: torch.Tensor, topk: int, num_groups: int, group_topk: int
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """DeepSeek node-limited routing (reference moe_utils.py:673).
+
+    Experts are partitioned into `num_groups` contiguous groups (nodes); each
+    token may only route to experts inside its best `group_topk` groups, where
+    a group's score is the sum of its top-2 expert scores.  Returns
+    (values, indices) of the final per-token topk over the masked scores.
+    """
+    T, E = scores.shape
+    gsz = E // num_groups
+    grouped = scores.view(T, num_groups, gsz)
+    group_scores = grouped.topk(min(2, gsz), dim=-1).values.sum(dim=-1)  # [T, G]
+    top_groups = group_scores.topk(group_topk, dim=-1).indices  # [T, group_topk]
+    group_mask = torch.zeros_like(group_scores).scatter_(1, top_groups, 1.0)
+    score_mask = group_mask.unsqueeze(-1).expand(T, num_groups, gsz).reshape(T, E)
+    masked = scores.masked_fill(score_mask == 0, float("-inf"))
+    return torch.topk(masked, topk, dim=-1)
+
+
 class TopKRouter(nn.Module):
     def __init__(self, config):
         super().__init__()
@@ -41,44 +64,78 @@ class TopKRouter(nn.Module):
         self.topk = config.moe_router_topk
         self.score_function = config.moe_router_score_function
         self.pre_softmax = config.moe_router_pre_softmax
+        self.num_groups = getattr(config, "moe_router_num_groups", None)
+        self.group_topk = getattr(config, "moe_router_group_topk", None)
+        self.jitter_eps = getattr(config, "moe_input_jitter_eps", None)
         # gating in fp32 (reference router.py gate fp32 option)
         self.weight = nn.Parameter(torch.empty(self.num_experts, config.hidden_size, dtype=torch.float32))
         with torch.no_grad():
             self.weight.normal_(0.0, config.init_method_std)
-        # aux-free balancing bias (updated outside autograd)
+        # aux-free balancing bias (updated outside autograd by
+        # megatron_amd.distributed.finalize.update_router_expert_bias)
         self.register_buffer("expert_bias", torch.zeros(self.num_experts), persistent=True)
+        # per-step local routed-token counts, consumed by the bias update
+        self.register_buffer("local_tokens_per_expert", torch.zeros(self.num_experts), persistent=False)
+        self.seq_len: Optional[int] = None  # set by MoELayer for seq-aux loss
         self.aux_losses = {}
+
+    def _topk(self, scores: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        if self.num_groups is not None and self.num_groups > 1:
+            return group_limited_topk(scores, self.topk, self.num_groups, self.group_topk or 1)
+        return torch.topk(scores, self.topk, dim=-1)
 
     def forward(self, hidden: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
         """hidden [T, h] -> (probs [T, topk], indices [T, topk]); also stashes
         aux losses for injection by the MoE layer."""
+        if self.jitter_eps and self.training:
+            # multiplicative input jitter (reference router.py apply_input_jitter)
+            noise = torch.empty_like(hidden).uniform_(1.0 - self.jitter_eps, 1.0 + self.jitter_eps)
+            hidden = hidden * noise
         logits = F.linear(hidden.float(), self.weight)  # [T, E]
         self.aux_losses = {}
 
         if self.score_function == "sigmoid":
             scores = torch.sigmoid(logits)
-            scores_for_topk = scores + self.expert_bias
-            top_vals, top_idx = torch.topk(scores_for_topk, self.topk, dim=-1)
+            _, top_idx = self._topk(scores + self.expert_bias)
             probs = scores.gather(-1, top_idx)
             probs = probs / probs.sum(dim=-1, keepdim=True).clamp(min=1e-20)
             full_probs = scores / scores.sum(dim=-1, keepdim=True).clamp(min=1e-20)
         elif self.pre_softmax:
             full_probs = torch.softmax(logits, dim=-1)
-            probs, top_idx = torch.topk(full_probs + self.expert_bias, self.topk, dim=-1)
+            _, top_idx = self._topk(full_probs + self.expert_bias)
             probs = full_probs.gather(-1, top_idx)
         else:
-            top_logits, top_idx = torch.topk(logits + self.expert_bias, self.topk, dim=-1)
+            _, top_idx = self._topk(logits + self.expert_bias)
             top_logits = logits.gather(-1, top_idx)
             probs = torch.softmax(top_logits, dim=-1)
             full_probs = torch.softmax(logits, dim=-1)
 
-        # switch load-balancing aux loss (reference moe_utils.py:63)
+        routing_map = torch.zeros_like(logits).scatter_(1, top_idx, 1.0)
+        with torch.no_grad():
+            self.local_tokens_per_expert = routing_map.sum(dim=0)
+
+        # load-balancing aux loss: 'aux' = switch-style over the whole batch
+        # (reference moe_utils.py:63); 'seq_aux' averages the loss per sequence
+        # (DeepSeek-V2 style) using config.seq_length-sized slices.
         if self.config.moe_aux_loss_coeff > 0:
-            T = logits.shape[0]
-            routing_map = torch.zeros_like(logits).scatter_(1, top_idx, 1.0)
-            f = routing_map.mean(dim=0) * self.num_experts / self.topk  # fraction per expert
-            P = full_probs.mean(dim=0)
-            aux = (f * P).sum() * self.num_experts * self.config.moe_aux_loss_coeff
+            if getattr(self.config, "moe_aux_loss_type", "aux") == "seq_aux" and self.seq_len:
+                s = self.seq_len  # set by MoELayer from the activation shape
+                T = logits.shape[0]
+                if T % s == 0 and T >= s:
+                    # tokens were flattened from [s, b, h]: dim 0 is sequence pos
+                    rm = routing_map.view(s, -1, self.num_experts)
+                    fp = full_probs.view(s, -1, self.num_experts)
+                    f = rm.mean(dim=0) * self.num_experts / self.topk  # [B, E]
+                    P = fp.mean(dim=0)
+                    aux = (f * P).sum(dim=-1).mean() * self.num_experts * self.config.moe_aux_loss_coeff
+                else:  # fall back to batch-level when tokens aren't seq-shaped
+                    f = routing_map.mean(dim=0) * self.num_experts / self.topk
+                    P = full_probs.mean(dim=0)
+                    aux = (f * P).sum() * self.num_experts * self.config.moe_aux_loss_coeff
+            else:
+                f = routing_map.mean(dim=0) * self.num_experts / self.topk  # fraction per expert
+                P = full_probs.mean(dim=0)
+                aux = (f * P).sum() * self.num_experts * self.config.moe_aux_loss_coeff
             self.aux_losses["load_balancing_loss"] = aux
         if self.config.moe_z_loss_coeff > 0:
             z = torch.logsumexp(logits, dim=-1)
