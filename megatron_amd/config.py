@@ -201,7 +201,13 @@ class TransformerConfig(ParallelConfig):
 class OptimizerConfig:
     """Analog of the reference OptimizerConfig (optimizer/optimizer_config.py)."""
 
-    optimizer: str = "adam"
+    optimizer: str = "adam"  # 'adam' | 'muon' | 'sgd'
+    # Muon (reference optimizer/muon.py): orthogonalized momentum for 2-D weights
+    muon_momentum: float = 0.95
+    muon_ns_steps: int = 5
+    # keep fp32 main params + Adam moments in pinned host memory
+    # (reference optimizer/cpu_offloading/ HybridDeviceOptimizer)
+    optimizer_cpu_offload: bool = False
     lr: float = 1e-4
     min_lr: float = 0.0
     weight_decay: float = 0.1

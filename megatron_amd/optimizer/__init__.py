@@ -24,7 +24,15 @@ from megatron_amd.optimizer.scheduler import OptimizerParamScheduler  # noqa: F4
 def get_optimizer(opt_config: OptimizerConfig, model_chunks: List) -> "ChainedOptimizer":
     """Build the optimizer for a list of DDP-wrapped model chunks
     (reference optimizer/__init__.py:991 get_megatron_optimizer)."""
-    if opt_config.use_distributed_optimizer:
+    if opt_config.optimizer == "muon":
+        from megatron_amd.optimizer.muon import MuonOptimizer
+
+        opt = MuonOptimizer(opt_config, model_chunks)
+    elif opt_config.optimizer_cpu_offload:
+        from megatron_amd.optimizer.cpu_offload import CPUOffloadOptimizer
+
+        opt = CPUOffloadOptimizer(opt_config, model_chunks)
+    elif opt_config.use_distributed_optimizer:
         opt = DistributedOptimizer(opt_config, model_chunks)
     elif opt_config.bf16 or opt_config.fp16:
         opt = MixedPrecisionOptimizer(opt_config, model_chunks)

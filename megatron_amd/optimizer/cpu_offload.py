@@ -1,0 +1,121 @@
+"""Optimizer-state CPU offloading.
+
+Capability analog of reference megatron/core/optimizer/cpu_offloading/
+(HybridDeviceOptimizer): fp32 main params and Adam moments live in pinned
+host memory; each step streams the device grads to the host on a side HIP
+stream, runs AdamW on the CPU, and streams the updated params back, freeing
+3x fp32 state from the 288 GB HBM3E pool for activations / bigger models.
+On a CPU-only build the copies are no-ops and the math is identical, which
+is what the unit tests check.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+
+from megatron_amd import ops
+from megatron_amd.config import OptimizerConfig
+from megatron_amd.optimizer.clip import (
+    clip_grads_by_total_norm,
+    get_grad_norm,
+    param_is_not_tensor_parallel_duplicate,
+)
+from megatron_amd.optimizer.optimizer import _BaseOptimizer, _model_chunks_params, _wd_group
+
+
+def _host_like(p: torch.Tensor) -> torch.Tensor:
+    t = torch.zeros(p.shape, dtype=torch.float32, device="cpu")
+    if torch.cuda.is_available():
+        t = t.pin_memory()
+    return t
+
+
+class CPUOffloadOptimizer(_BaseOptimizer):
+    """AdamW with all optimizer state in (pinned) host memory."""
+
+    def __init__(self, config: OptimizerConfig, model_chunks: List):
+        super().__init__(config, model_chunks)
+        self.params = _model_chunks_params(model_chunks)
+        self.main_params = []
+        for p in self.params:
+            mp = _host_like(p)
+            mp.copy_(p.detach().float().cpu())
+            self.main_params.append(mp)
+        self.exp_avg = [_host_like(p) for p in self.params]
+        self.exp_avg_sq = [_host_like(p) for p in self.params]
+        self._grad_host = [_host_like(p) for p in self.params]
+        self._d2h_stream = torch.cuda.Stream() if torch.cuda.is_available() else None
+
+    @torch.no_grad()
+    def step(self) -> Tuple[bool, Optional[torch.Tensor], Optional[int]]:
+        self.finish_grad_sync()
+        dev_grads = []
+        for p in self.params:
+            g = getattr(p, "main_grad", None)
+            if g is None:
+                g = p.grad if p.grad is not None else torch.zeros_like(p)
+            dev_grads.append(g.float())
+        norm_grads = [g for p, g in zip(self.params, dev_grads) if param_is_not_tensor_parallel_duplicate(p)]
+        total_norm = get_grad_norm(norm_grads)
+        if self.config.clip_grad > 0:
+            clip_grads_by_total_norm(dev_grads, self.config.clip_grad, total_norm)
+
+        # device -> host grads (async on a side stream when on GPU)
+        if self._d2h_stream is not None:
+            self._d2h_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._d2h_stream):
+                for hg, dg in zip(self._grad_host, dev_grads):
+                    hg.copy_(dg, non_blocking=True)
+            self._d2h_stream.synchronize()
+        else:
+            for hg, dg in zip(self._grad_host, dev_grads):
+                hg.copy_(dg)
+
+        self.step_count += 1
+        decay_mask = [_wd_group(p) for p in self.params]
+        for apply_wd in (True, False):
+            idx = [i for i, m in enumerate(decay_mask) if m == apply_wd]
+            if not idx:
+                continue
+            ops.fused_adamw(
+                [self.main_params[i] for i in idx],
+                [self._grad_host[i] for i in idx],
+                [self.exp_avg[i] for i in idx],
+                [self.exp_avg_sq[i] for i in idx],
+                self._lr,
+                self.config.adam_beta1,
+                self.config.adam_beta2,
+                self.config.adam_eps,
+                self._wd if apply_wd else 0.0,
+                self.step_count,
+            )
+        # host -> device updated params
+        for p, mp in zip(self.params, self.main_params):
+            p.data.copy_(mp.to(p.dtype), non_blocking=True)
+        if torch.cuda.is_available():
+            torch.cuda.current_stream().synchronize()
+        return True, total_norm, None
+
+    def reload_model_params(self):
+        for mp, p in zip(self.main_params, self.params):
+            mp.copy_(p.detach().float().cpu())
+
+    def state_dict(self):
+        return {
+            "step": self.step_count,
+            "main_params": self.main_params,
+            "exp_avg": self.exp_avg,
+            "exp_avg_sq": self.exp_avg_sq,
+        }
+
+    def load_state_dict(self, sd):
+        self.step_count = sd["step"]
+        for dst, src in zip(
+            self.main_params + self.exp_avg + self.exp_avg_sq,
+            sd["main_params"] + sd["exp_avg"] + sd["exp_avg_sq"],
+        ):
+            dst.copy_(src.cpu() if isinstance(src, torch.Tensor) else torch.as_tensor(src))
+        for p, mp in zip(self.params, self.main_params):
+            p.data.copy_(mp.to(p.dtype))

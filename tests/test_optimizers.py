@@ -1,0 +1,126 @@
+"""Muon and CPU-offload optimizers (CPU).
+
+Muon: Newton-Schulz orthogonalization quality, update semantics, and a
+training-progress check.  CPU offload: exact step-for-step equivalence with
+the in-memory FP32 optimizer (on CPU the copies are identity, so the math
+must match bit-for-bit module the clip path).
+"""
+
+import copy
+
+import torch
+import torch.nn as nn
+
+from megatron_amd.config import OptimizerConfig
+from megatron_amd.optimizer import get_optimizer
+from megatron_amd.optimizer.cpu_offload import CPUOffloadOptimizer
+from megatron_amd.optimizer.muon import MuonOptimizer, muon_param, newton_schulz_orthogonalize
+from megatron_amd.optimizer.optimizer import FP32Optimizer
+
+from tests.utils import assert_close, init_single
+
+
+class _Tiny(nn.Module):
+    def __init__(self):
+        super().__init__()
+        torch.manual_seed(7)
+        self.fc1 = nn.Linear(16, 32)
+        self.fc2 = nn.Linear(32, 16)
+        self.norm = nn.LayerNorm(16)
+
+    def forward(self, x):
+        return self.norm(self.fc2(torch.relu(self.fc1(x))))
+
+
+def _loss_and_backward(model, x, y):
+    loss = ((model(x) - y) ** 2).mean()
+    model.zero_grad(set_to_none=True)
+    loss.backward()
+    return loss
+
+
+def test_newton_schulz_orthogonalizes():
+    torch.manual_seed(0)
+    # ill-conditioned input: singular values spread over 3 decades
+    U, _ = torch.linalg.qr(torch.randn(24, 24))
+    V, _ = torch.linalg.qr(torch.randn(48, 24))
+    s = torch.logspace(0, -3, 24)
+    g = U @ torch.diag(s) @ V.T
+    O = newton_schulz_orthogonalize(g, steps=5)
+    # Muon's quintic drives all singular values into a band around 1
+    # (by design ~[0.7, 1.2], not exactly 1) — vs the input's 1000x spread.
+    sv = torch.linalg.svdvals(O)
+    assert float(sv.min()) > 0.3 and float(sv.max()) < 1.5, sv
+    # sign/direction preserved: <O, g> > 0
+    assert float((O * g).sum()) > 0
+
+
+def test_muon_param_selection():
+    m = _Tiny()
+    kinds = {name: muon_param(p) for name, p in m.named_parameters()}
+    assert kinds["fc1.weight"] and kinds["fc2.weight"]
+    assert not kinds["fc1.bias"] and not kinds["norm.weight"]
+    m.fc1.weight.muon_exclude = True
+    assert not muon_param(m.fc1.weight)
+
+
+def test_muon_trains():
+    init_single()
+    model = _Tiny()
+    cfg = OptimizerConfig(optimizer="muon", lr=3e-3, weight_decay=0.0, clip_grad=1.0)
+    opt = get_optimizer(cfg, [model])
+    assert isinstance(opt.chained_optimizers[0], MuonOptimizer)
+    torch.manual_seed(1)
+    x = torch.randn(64, 16)
+    y = torch.tanh(x @ torch.randn(16, 16) * 0.5)  # learnable target
+    first = float(_loss_and_backward(model, x, y).detach())
+    for _ in range(60):
+        _loss_and_backward(model, x, y)
+        ok, norm, _ = opt.step()
+        assert ok and torch.isfinite(norm)
+    last = float(_loss_and_backward(model, x, y).detach())
+    assert last < 0.5 * first, (first, last)
+
+
+def test_muon_state_roundtrip():
+    init_single()
+    model = _Tiny()
+    cfg = OptimizerConfig(optimizer="muon", lr=1e-3)
+    opt = MuonOptimizer(cfg, [model])
+    x, y = torch.randn(8, 16), torch.randn(8, 16)
+    _loss_and_backward(model, x, y)
+    opt.step()
+    sd = copy.deepcopy(opt.state_dict())
+    model2 = _Tiny()
+    opt2 = MuonOptimizer(cfg, [model2])
+    opt2.load_state_dict(sd)
+    for a, b in zip(opt.main_params, opt2.main_params):
+        assert_close(a, b, rtol=0, atol=0)
+    for p, p2 in zip(model.parameters(), model2.parameters()):
+        assert_close(p, p2, rtol=0, atol=0)
+
+
+def test_cpu_offload_matches_fp32():
+    init_single()
+    model_a = _Tiny()
+    model_b = copy.deepcopy(model_a)
+    cfg = OptimizerConfig(lr=1e-3, weight_decay=0.01, clip_grad=1.0)
+    opt_a = FP32Optimizer(cfg, [model_a])
+    opt_b = CPUOffloadOptimizer(cfg, [model_b])
+    torch.manual_seed(2)
+    for step in range(5):
+        x, y = torch.randn(16, 16), torch.randn(16, 16)
+        _loss_and_backward(model_a, x, y)
+        _loss_and_backward(model_b, x, y)
+        opt_a.step()
+        opt_b.step()
+    for (na, pa), (nb, pb) in zip(model_a.named_parameters(), model_b.named_parameters()):
+        assert_close(pa, pb, rtol=1e-6, atol=1e-6, msg=na)
+
+
+def test_cpu_offload_selected_by_config():
+    init_single()
+    model = _Tiny()
+    cfg = OptimizerConfig(lr=1e-3, optimizer_cpu_offload=True)
+    opt = get_optimizer(cfg, [model])
+    assert isinstance(opt.chained_optimizers[0], CPUOffloadOptimizer)
