@@ -171,3 +171,65 @@ def test_training_from_indexed_corpus(tmp_path):
         "--train-iters", "2", "--log-interval", "0", "--data-path", prefix,
     ])
     assert it == 2
+
+
+class _FixedTokens(torch.utils.data.Dataset):
+    def __init__(self, n=4, s=64, vocab=1000, seed=0):
+        g = torch.Generator().manual_seed(seed)
+        self.data = [torch.randint(0, vocab - 200, (s,), generator=g) for _ in range(n)]
+
+    def __len__(self):
+        return len(self.data)
+
+    def __getitem__(self, i):
+        return {"tokens": self.data[i]}
+
+
+def test_t5_span_corruption_reconstruction():
+    from megatron_amd.datasets.t5_dataset import T5SpanCorruptionDataset
+
+    V, bos, eos = 1000, 1, 2
+    base = _FixedTokens(vocab=V)
+    ds = T5SpanCorruptionDataset(base, vocab_size=V, bos_id=bos, eos_id=eos, seed=5)
+    sample = ds[0]
+    enc, dec, labels = sample["encoder_tokens"], sample["decoder_tokens"], sample["labels"]
+    orig = base[0]["tokens"]
+    # decoder input is bos + labels shifted right
+    assert int(dec[0]) == bos
+    assert torch.equal(dec[1:], labels[:-1])
+    assert int(labels[-1]) == eos
+    # splice the target spans back into the encoder input -> original sequence
+    sentinels = set(range(V - 100, V))
+    recon = []
+    # build sentinel -> span map from labels
+    spans, cur = {}, None
+    for t in labels.tolist()[:-1]:
+        if t in sentinels:
+            cur = t
+            spans[cur] = []
+        else:
+            spans[cur].append(t)
+    for t in enc.tolist():
+        if t in sentinels:
+            recon.extend(spans[t])
+        else:
+            recon.append(t)
+    assert recon == orig.tolist()
+    # ~15% of tokens masked
+    n_masked = sum(len(v) for v in spans.values())
+    assert 0.05 * orig.numel() <= n_masked <= 0.3 * orig.numel()
+    # deterministic per (seed, idx)
+    again = ds[0]
+    assert torch.equal(again["encoder_tokens"], enc)
+
+
+def test_t5_pad_batch_shapes():
+    from megatron_amd.datasets.t5_dataset import T5SpanCorruptionDataset, pad_t5_batch
+
+    base = _FixedTokens(n=3, s=48)
+    ds = T5SpanCorruptionDataset(base, vocab_size=1000, bos_id=1, eos_id=2)
+    batch = pad_t5_batch([ds[i] for i in range(3)], enc_len=64, dec_len=32)
+    assert batch["encoder_tokens"].shape == (3, 64)
+    assert batch["decoder_tokens"].shape == (3, 32)
+    assert batch["loss_mask"].sum() > 0
+    assert batch["encoder_mask"][0].sum() == ds[0]["encoder_tokens"].numel()
