@@ -126,3 +126,51 @@ def test_gpt_cp2_matches_single():
     loss_ref, grad_ref = _loss_single(cfg_kwargs, tokens, labels)
     for mode in ("p2p", "a2a"):
         spawn_dist(_run_gpt_cp, 2, cfg_kwargs, tokens, labels, loss_ref, grad_ref, mode)
+
+
+def test_balanced_cp_scheduler_assignments_complete():
+    from megatron_amd.parallel.balanced_cp import BalancedCPScheduler
+
+    sched = BalancedCPScheduler(world_size=8, max_cp=4, chunk_target=1024)
+    seq_lens = [512, 1024, 4096, 2048, 8192, 256, 1024, 3072, 128, 6144]
+    per_rank = sched.schedule(seq_lens)
+    # every sample appears exactly once per member of its CP group, cp_ranks 0..cp-1
+    seen = {}
+    for r, assigns in per_rank.items():
+        for a in assigns:
+            seen.setdefault(a.sample, []).append((r, a.cp_rank, a.cp_size, a.ranks))
+    assert set(seen.keys()) == set(range(len(seq_lens)))
+    for i, entries in seen.items():
+        cp = entries[0][2]
+        assert len(entries) == cp
+        ranks = sorted(e[0] for e in entries)
+        assert tuple(ranks) == entries[0][3]
+        assert ranks[0] % cp == 0  # aligned window
+        assert sorted(e[1] for e in entries) == list(range(cp))
+
+
+def test_balanced_cp_scheduler_balances_quadratic_work():
+    import random
+
+    from megatron_amd.parallel.balanced_cp import BalancedCPScheduler, quadratic_cost
+
+    rng = random.Random(0)
+    seq_lens = [rng.choice([256, 512, 1024, 2048, 4096, 8192]) for _ in range(64)]
+    sched = BalancedCPScheduler(world_size=8, max_cp=8, chunk_target=1024)
+    ratio = sched.balance_ratio(seq_lens)
+    assert ratio < 1.15, ratio
+    # naive fixed-CP=1 round-robin for comparison
+    loads = [0.0] * 8
+    for i, s in enumerate(seq_lens):
+        loads[i % 8] += quadratic_cost(s)
+    naive = max(loads) / (sum(loads) / 8)
+    assert ratio <= naive
+
+
+def test_pick_cp_size_powers_of_two():
+    from megatron_amd.parallel.balanced_cp import pick_cp_size
+
+    assert pick_cp_size(1024, 8, 4096) == 1
+    assert pick_cp_size(8192, 8, 4096) == 2
+    assert pick_cp_size(32768, 8, 4096) == 8
+    assert pick_cp_size(10 ** 6, 4, 4096) == 4  # clamped at max_cp
