@@ -1,0 +1,91 @@
+"""Online weight refit (training -> inference resharding).
+
+The refit path assembles full global tensors from flat-atlas shards (owner
+writes + summed all-reduce) and re-slices them into the destination model's
+sharding.  Checks: single-process round trip, TP=2 round trip with
+different init (gloo), and the assembly itself against local shards.
+"""
+
+import torch
+import torch.distributed as dist
+
+from megatron_amd.config import TransformerConfig
+from megatron_amd.models.gpt import GPTModel
+from megatron_amd.parallel import grid as G
+from megatron_amd.parallel.random import model_parallel_seed
+from megatron_amd.resharding import assemble_global_tensors, refit_model
+from megatron_amd.checkpoint.state_dict import model_sharded_state_dict
+
+from tests.utils import assert_close, init_single, spawn_dist
+
+
+def _cfg(tp=1):
+    return TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        vocab_size=96, ffn_hidden_size=128, tensor_parallel_size=tp,
+        gradient_accumulation_fusion=False,
+    )
+
+
+def test_refit_single_process_round_trip():
+    init_single()
+    cfg = _cfg()
+    torch.manual_seed(10)
+    src = GPTModel(cfg)
+    torch.manual_seed(20)
+    dst = GPTModel(cfg)
+    # different init to start
+    assert not torch.equal(src.embedding.weight, dst.embedding.weight)
+    n = refit_model(src, dst)
+    assert n > 0
+    for (na, pa), (nb, pb) in zip(src.named_parameters(), dst.named_parameters()):
+        assert_close(pa, pb, rtol=0, atol=1e-6, msg=na)
+
+
+def _tp2_case(rank, world):
+    G.initialize_model_parallel(tensor_parallel_size=2)
+    model_parallel_seed(100)
+    src = GPTModel(_cfg(tp=2))
+    model_parallel_seed(200)
+    dst = GPTModel(_cfg(tp=2))
+    refit_model(src, dst)
+    for (na, pa), (nb, pb) in zip(src.named_parameters(), dst.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), f"rank {rank}: {na}"
+    # forward equivalence on a broadcast batch
+    tokens = torch.randint(0, 96, (2, 16))
+    dist.broadcast(tokens, src=0)
+    with torch.no_grad():
+        a = src(tokens, position_ids=None, attention_mask=None)
+        b = dst(tokens, position_ids=None, attention_mask=None)
+    assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_refit_tp2_round_trip():
+    spawn_dist(_tp2_case, 2)
+
+
+def _assemble_case(rank, world):
+    G.initialize_model_parallel(tensor_parallel_size=2)
+    model_parallel_seed(7)
+    model = GPTModel(_cfg(tp=2))
+    shard_map, _ = model_sharded_state_dict(model)
+    # a column-parallel (dim 0) weight: fc2 row-parallel is dim 1; use
+    # the output layer / embedding which shard the vocab over dim 0
+    key = "model.embedding.weight"
+    full = assemble_global_tensors(shard_map, keys=[key])[key]
+    local = model.embedding.weight
+    shard_rows = local.shape[0]
+    assert full.shape[0] == shard_rows * 2
+    expect = full[rank * shard_rows:(rank + 1) * shard_rows]
+    assert torch.allclose(expect, local.float(), atol=1e-6)
+    # replicated param (final norm): full == local on every rank
+    for name, p in model.named_parameters():
+        if "layernorm" in name and p.dim() == 1:
+            key2 = "model." + name
+            full2 = assemble_global_tensors(shard_map, keys=[key2])[key2]
+            assert torch.allclose(full2, p.float(), atol=1e-6)
+            break
+
+
+def test_assemble_global_tensors_tp2():
+    spawn_dist(_assemble_case, 2)
