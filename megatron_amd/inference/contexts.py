@@ -58,9 +58,11 @@ class StaticInferenceContext:
     def __init__(self, num_layers: int, max_batch: int, max_seq: int,
                  num_kv_heads: int, head_dim: int, dtype=torch.bfloat16, device="cuda"):
         self.max_batch, self.max_seq = max_batch, max_seq
-        self.k_cache = [torch.empty(max_seq, max_batch, num_kv_heads, head_dim,
+        # zeros not empty: masked attention still computes 0*v for padded
+        # slots, so inf/NaN garbage in reused memory would poison the output
+        self.k_cache = [torch.zeros(max_seq, max_batch, num_kv_heads, head_dim,
                                     dtype=dtype, device=device) for _ in range(num_layers)]
-        self.v_cache = [torch.empty_like(self.k_cache[0]) for _ in range(num_layers)]
+        self.v_cache = [torch.zeros_like(self.k_cache[0]) for _ in range(num_layers)]
         # per-row filled length; uniform during same-length prefill, ragged after
         self.context_lens = torch.zeros(max_batch, dtype=torch.long, device=device)
         self.batch_size = 0
@@ -145,9 +147,9 @@ class DynamicInferenceContext:
         # physical block 0 is a scratch target for padded rows in hipGraph
         # decode replays; the allocator only hands out ids 1..num_blocks
         self.allocator = KVBlockAllocator(num_blocks, first_id=1)
-        self.k_cache = [torch.empty(num_blocks + 1, block_size, num_kv_heads, head_dim,
+        self.k_cache = [torch.zeros(num_blocks + 1, block_size, num_kv_heads, head_dim,
                                     dtype=dtype, device=device) for _ in range(num_layers)]
-        self.v_cache = [torch.empty_like(self.k_cache[0]) for _ in range(num_layers)]
+        self.v_cache = [torch.zeros_like(self.k_cache[0]) for _ in range(num_layers)]
         self.device = device
         self._static = False
         # step state, set by the engine before each forward

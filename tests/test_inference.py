@@ -143,3 +143,40 @@ def test_rest_server(tiny_model):
     assert all(isinstance(g["tokens"], list) for g in gens)
     assert gens[0]["text"] is not None
     assert len(gens[0]["logprobs"]) == len(gens[0]["tokens"])
+
+
+def test_dp_coordinator_matches_single_engine(tiny_model):
+    from megatron_amd.inference.coordinator import DataParallelCoordinator
+
+    params = SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False)
+    prompts = [[1, 2, 3], [4, 5], [7, 8, 9, 10], [11], [12, 13], [14, 15, 16]]
+    single = StaticInferenceEngine(tiny_model, max_batch=8, max_seq=128)
+    expected = single.generate(prompts, params)
+
+    engines = [StaticInferenceEngine(tiny_model, max_batch=4, max_seq=128) for _ in range(2)]
+    coord = DataParallelCoordinator(engines, max_batch_per_engine=2)
+    try:
+        results = coord.generate(prompts, params)
+        for r, e in zip(results, expected):
+            assert r.output_tokens == e.output_tokens
+        stats = coord.stats()
+        assert stats["replicas"] == 2 and all(v == 0 for v in stats["inflight"])
+    finally:
+        coord.shutdown()
+
+
+def test_dp_coordinator_error_propagates(tiny_model):
+    from megatron_amd.inference.coordinator import DataParallelCoordinator
+
+    class Boom:
+        def generate(self, prompts, params):
+            raise RuntimeError("engine exploded")
+
+    coord = DataParallelCoordinator([Boom()])
+    try:
+        import pytest as _pytest
+
+        with _pytest.raises(RuntimeError, match="exploded"):
+            coord.generate([[1, 2]], SamplingParams(max_tokens=2))
+    finally:
+        coord.shutdown()
