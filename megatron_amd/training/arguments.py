@@ -18,6 +18,9 @@ from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
 def build_arg_parser() -> argparse.ArgumentParser:
     p = argparse.ArgumentParser(description="megatron_amd pretraining", allow_abbrev=False)
 
+    p.add_argument("--yaml-cfg", type=str, default=None,
+                   help="yaml config file overlaid onto defaults (CLI flags win)")
+
     g = p.add_argument_group("model")
     g.add_argument("--num-layers", type=int, required=False, default=2)
     g.add_argument("--hidden-size", type=int, default=64)
@@ -262,11 +265,39 @@ def configs_from_args(args):
     return cfg, opt_cfg, ddp_cfg
 
 
+def apply_yaml_config(args, path: str, parser=None):
+    """Overlay a yaml config file onto parsed args (reference
+    training/yaml_arguments.py).  Keys use either underscores or the CLI's
+    dashes; explicit CLI flags win over yaml values, yaml wins over
+    defaults."""
+    import yaml
+
+    with open(path) as f:
+        doc = yaml.safe_load(f) or {}
+    if parser is None:
+        parser = build_arg_parser()
+    defaults = vars(parser.parse_args([]))
+    unknown = []
+    for key, value in doc.items():
+        attr = key.replace("-", "_")
+        if attr not in defaults:
+            unknown.append(key)
+            continue
+        # only apply when the CLI left the default in place
+        if getattr(args, attr) == defaults[attr]:
+            setattr(args, attr, value)
+    if unknown:
+        raise ValueError(f"unknown yaml config keys: {unknown}")
+    return args
+
+
 def parse_and_validate_args(argv=None):
     import os
 
     parser = build_arg_parser()
     args = parser.parse_args(argv)
+    if getattr(args, "yaml_cfg", None):
+        apply_yaml_config(args, args.yaml_cfg, parser)
     args.world_size = int(os.environ.get("WORLD_SIZE", "1"))
     args.rank = int(os.environ.get("RANK", "0"))
     validate_args(args)

@@ -91,3 +91,43 @@ def test_fp8_gpt_training_step():
         losses.append(float(loss))
     assert losses[-1] < losses[0], f"fp8 training did not reduce loss: {losses}"
     G.destroy_model_parallel()
+
+
+def test_mxfp4_round_trip_exact_values():
+    from megatron_amd.ops.fp4 import FP4_VALUES, dequantize_mxfp4, quantize_mxfp4
+
+    # a block of exactly representable values at scale 1 must round-trip exactly
+    vals = torch.tensor([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0] * 4)
+    q, s = quantize_mxfp4(vals.unsqueeze(0))
+    assert torch.allclose(s, torch.ones_like(s))  # amax=6 -> scale 1
+    back = dequantize_mxfp4(q, s)
+    assert torch.equal(back.squeeze(0), vals)
+
+
+def test_mxfp4_error_bounded():
+    from megatron_amd.ops.fp4 import mxfp4_quantization_error, quantize_mxfp4
+
+    torch.manual_seed(0)
+    t = torch.randn(64, 128) * 3.7
+    err = mxfp4_quantization_error(t)
+    assert err < 0.12, err  # fp4 w/ block scales: ~6-10% RMS on gaussians
+    q, s = quantize_mxfp4(t)
+    # scales are powers of two
+    assert torch.allclose(torch.log2(s), torch.log2(s).round(), atol=0)
+    # every quantized magnitude is on the E2M1 grid
+    from megatron_amd.ops.fp4 import FP4_VALUES
+    flat = q.abs().reshape(-1, 1)
+    on_grid = (flat - FP4_VALUES).abs().min(dim=-1).values
+    assert float(on_grid.max()) == 0.0
+
+
+def test_mxfp4_pack_unpack():
+    from megatron_amd.ops.fp4 import pack_fp4_codes, quantize_mxfp4, unpack_fp4_codes
+
+    torch.manual_seed(1)
+    t = torch.randn(4, 64)
+    q, s = quantize_mxfp4(t)
+    packed = pack_fp4_codes(q)
+    assert packed.dtype == torch.uint8 and packed.numel() == q.numel() // 2
+    back = unpack_fp4_codes(packed, q.numel()).reshape(q.shape)
+    assert torch.equal(back, q)

@@ -44,3 +44,22 @@ def test_pretrain_runs_and_resumes(tmp_path):
     it2 = pretrain(model_provider, TINY + ["--train-iters", "5", "--save", ckpt,
                                            "--load", ckpt, "--seed", "7"])
     assert it2 == 5
+
+
+def test_yaml_config_overlay(tmp_path):
+    from megatron_amd.training.arguments import parse_and_validate_args
+
+    cfg = tmp_path / "run.yaml"
+    cfg.write_text("num-layers: 6\nhidden_size: 256\nlr: 0.0005\n")
+    args = parse_and_validate_args(["--yaml-cfg", str(cfg), "--hidden-size", "128",
+                                    "--num-attention-heads", "4"])
+    assert args.num_layers == 6          # from yaml
+    assert args.hidden_size == 128       # CLI beats yaml
+    assert args.lr == 0.0005             # underscore/dash both accepted
+
+    bad = tmp_path / "bad.yaml"
+    bad.write_text("not-a-flag: 1\n")
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="unknown yaml"):
+        parse_and_validate_args(["--yaml-cfg", str(bad)])
