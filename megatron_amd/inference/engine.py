@@ -204,6 +204,11 @@ class DynamicInferenceEngine:
         self.waiting: List[_Request] = []
         self.active: List[_Request] = []
         self.finished: Dict[int, GenerationResult] = {}
+        # host-offload preemption (LIFO preempt, FIFO restore)
+        from megatron_amd.inference.offload import KVHostOffloader
+
+        self.offloader = KVHostOffloader(self.context)
+        self.preempted: List[tuple] = []  # (req, handle)
 
     def add_request(self, prompt, params: SamplingParams = SamplingParams()) -> int:
         if self.tokenizer is not None and isinstance(prompt, str):
@@ -215,7 +220,29 @@ class DynamicInferenceEngine:
         return rid
 
     def has_work(self) -> bool:
-        return bool(self.waiting or self.active)
+        return bool(self.waiting or self.active or self.preempted)
+
+    def _preempt_one(self, exclude=None) -> bool:
+        """Swap the youngest active request's KV to host, freeing its blocks."""
+        for victim in reversed(self.active):
+            if victim is exclude:
+                continue
+            self.active.remove(victim)
+            handle = self.offloader.swap_out(victim.block_table)
+            victim.block_table = []
+            self.preempted.append((victim, handle))
+            return True
+        return False
+
+    def _restore_preempted(self):
+        while self.preempted and len(self.active) < self.max_batch:
+            req, handle = self.preempted[0]
+            need = self.offloader.num_blocks_of(handle) + 1  # +1 headroom to decode
+            if self.context.allocator.num_free < need:
+                break
+            self.preempted.pop(0)
+            req.block_table = self.offloader.swap_in(handle)
+            self.active.append(req)
 
     def _blocks_for(self, n_tokens: int) -> int:
         bs = self.context.block_size
@@ -251,6 +278,7 @@ class DynamicInferenceEngine:
     def step(self) -> None:
         """One engine step: chunked prefill of the next waiting request, or one
         batched decode step over all active requests."""
+        self._restore_preempted()
         if self.waiting and len(self.active) < self.max_batch:
             req = self.waiting[0]
             chunk = min(self.max_prefill_tokens, len(req.prompt) - req.cached)
@@ -280,12 +308,15 @@ class DynamicInferenceEngine:
 
         if not self.active:
             return
-        # batched decode
-        batch = self.active
-        for req in batch:
+        # batched decode; on pool exhaustion preempt youngest requests to host
+        for req in list(self.active):
+            if req not in self.active:  # already preempted this pass
+                continue
             total = len(req.prompt) + len(req.result.output_tokens)
-            ok = self._ensure_blocks(req, total + 1)
-            assert ok, "KV pool exhausted during decode"  # TODO: preemption
+            while not self._ensure_blocks(req, total + 1):
+                if not self._preempt_one(exclude=req):
+                    raise RuntimeError("KV pool exhausted and nothing left to preempt")
+        batch = self.active
         tables = [r.block_table for r in batch]
         lens = [len(r.prompt) + len(r.result.output_tokens) - 1 for r in batch]  # cached so far
         if self._graphs is not None:
