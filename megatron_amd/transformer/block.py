@@ -175,6 +175,13 @@ class TransformerBlock(nn.Module):
             if recompute and i < num_ckpt:
                 hidden_states = self._checkpointed(layer, hidden_states, rotary_freqs)
             else:
+                if getattr(self, "_graphed_layers", None) is not None and not recompute:
+                    from megatron_amd.transformer.hip_graphs import graphed_layer_or_none
+
+                    g = graphed_layer_or_none(self, i, hidden_states, inference_context)
+                    if g is not None:
+                        hidden_states = g(hidden_states)
+                        continue
                 hidden_states = layer(
                     hidden_states, rotary_freqs=rotary_freqs, attention_mask=attention_mask,
                     inference_context=inference_context,
