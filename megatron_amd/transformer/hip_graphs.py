@@ -4,44 +4,104 @@ Capability analog of reference megatron/core/transformer/cuda_graphs.py
 (`CudaGraphManager`, per-layer capture) and full_cuda_graph.py: short
 launch-bound inner loops (small models, decode-sized microbatches) replay a
 captured hipGraph instead of relaunching every kernel.  torch.cuda.CUDAGraph
-IS hipGraph on ROCm; `torch.cuda.make_graphed_callables` records one
-forward and one backward graph per layer, with shared memory pools.
+IS hipGraph on ROCm.
 
-Scope notes (vs the reference's 3.5k-LoC manager):
-  * capture the plain module stack BEFORE DDP wrapping (bucket grad hooks
-    fire outside the graphed region — same constraint as the reference's
-    `external` grad mode);
-  * static shapes: one (s, b, h) per capture, the wrapper falls back to
-    eager for any other shape;
-  * rotary freqs / masks are closed over as static tensors (they are
-    step-invariant in pretraining).
+Unlike `torch.cuda.make_graphed_callables` (whose multi-graph shared-mempool
+capture segfaults in capture_end on ROCm 7.2 — measured on MI355X), each
+layer here gets its OWN forward and backward graph with default pools:
+  * capture-time: warmup on a side stream, then record fwd with grad enabled
+    into static input/output buffers, then record bwd as autograd.grad of
+    the captured forward's graph (retained) into static grad buffers;
+  * run-time: a custom autograd.Function copies into the static input,
+    replays the fwd graph, and on backward copies the incoming grad in,
+    replays the bwd graph, and hands autograd clones of the static grads
+    (params are explicit Function inputs so their grads accumulate
+    normally, which keeps DDP bucket hooks outside the graphed region —
+    the reference's `external` grad mode).
 
-The dynamic inference engine has its own decode-step graph runner
-(inference/engine.py _DecodeGraphRunner); this module is the training side.
+Scope: static shapes (one (s, b, h) per capture; eager fallback otherwise,
+and always under activation recompute or inference contexts).  Rotary
+freqs / masks are closed over as static tensors (step-invariant in
+pretraining).  The dynamic inference engine has its own decode-step graph
+runner (inference/engine.py _DecodeGraphRunner); this module is the
+training side.
 """
 
 from __future__ import annotations
 
-from typing import Optional
+from typing import List, Optional
 
 import torch
 import torch.nn as nn
 
 
-class _StaticArgLayer(nn.Module):
-    """Close rotary freqs / mask over a layer so its graphed signature is a
-    single positional hidden-states tensor."""
+class _LayerGraphs:
+    """Captured state for one layer at one shape."""
 
-    def __init__(self, layer: nn.Module, rotary_freqs: Optional[torch.Tensor],
-                 attention_mask: Optional[torch.Tensor]):
+    def __init__(self, layer: nn.Module, sample: torch.Tensor,
+                 rotary_freqs: Optional[torch.Tensor],
+                 attention_mask: Optional[torch.Tensor],
+                 num_warmup_iters: int = 3):
+        self.layer = layer
+        self.params: List[torch.nn.Parameter] = [p for p in layer.parameters() if p.requires_grad]
+
+        def fwd(inp):
+            return layer(inp, rotary_freqs=rotary_freqs, attention_mask=attention_mask)
+
+        # warmup must run off the default stream (capture prerequisite)
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(num_warmup_iters):
+                xi = sample.clone().requires_grad_(True)
+                out = fwd(xi)
+                torch.autograd.grad(out, [xi] + self.params, torch.ones_like(out),
+                                    allow_unused=True)
+        torch.cuda.current_stream().wait_stream(side)
+        for p in self.params:
+            p.grad = None
+        torch.cuda.synchronize()
+
+        self.static_input = sample.clone().requires_grad_(True)
+        self.g_fwd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_fwd):
+            self.static_output = fwd(self.static_input)
+
+        self.static_grad_output = torch.empty_like(self.static_output)
+        self.g_bwd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_bwd):
+            grads = torch.autograd.grad(
+                self.static_output, [self.static_input] + self.params,
+                self.static_grad_output, retain_graph=True, allow_unused=True)
+        self.static_grad_input = grads[0]
+        self.static_param_grads = grads[1:]
+
+
+class _GraphedLayerFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, g: _LayerGraphs, x, *params):
+        ctx.g = g
+        g.static_input.copy_(x.detach())
+        g.g_fwd.replay()
+        return g.static_output.detach().clone()
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        g = ctx.g
+        g.static_grad_output.copy_(grad_output)
+        g.g_bwd.replay()
+        dx = g.static_grad_input.clone() if g.static_grad_input is not None else None
+        dps = tuple(p.clone() if p is not None else None for p in g.static_param_grads)
+        return (None, dx) + dps
+
+
+class GraphedLayer(nn.Module):
+    def __init__(self, graphs: _LayerGraphs):
         super().__init__()
-        self.inner = layer  # shared parameters, not a copy
-        self.rotary_freqs = rotary_freqs
-        self.attention_mask = attention_mask
+        self.graphs = graphs
 
-    def forward(self, hidden_states):
-        return self.inner(hidden_states, rotary_freqs=self.rotary_freqs,
-                          attention_mask=self.attention_mask)
+    def forward(self, x):
+        return _GraphedLayerFn.apply(self.graphs, x, *self.graphs.params)
 
 
 def capture_block_hip_graphs(block, sample_hidden: torch.Tensor,
@@ -55,16 +115,21 @@ def capture_block_hip_graphs(block, sample_hidden: torch.Tensor,
     if not torch.cuda.is_available():
         raise RuntimeError("hipGraph capture requires a GPU")
     assert sample_hidden.is_cuda, "sample must live on the device"
-    wrappers = tuple(
-        _StaticArgLayer(layer, rotary_freqs, attention_mask) for layer in block.layers
-    )
-    sample_args = tuple((sample_hidden.clone().requires_grad_(True),) for _ in wrappers)
-    graphed = torch.cuda.make_graphed_callables(
-        wrappers, sample_args, num_warmup_iters=num_warmup_iters
-    )
-    block._graphed_layers = list(graphed)
+    # ROCm 7.2: hipGraph capture_end segfaults if any LIVE autograd graph
+    # exists at capture time (measured on MI355X — a dangling eager forward
+    # is enough).  Capture at a step boundary and drop dead references here.
+    import gc
+
+    gc.collect()
+    torch.cuda.synchronize()
+    graphed = []
+    for layer in block.layers:
+        graphs = _LayerGraphs(layer, sample_hidden, rotary_freqs, attention_mask,
+                              num_warmup_iters=num_warmup_iters)
+        graphed.append(GraphedLayer(graphs))
+    block._graphed_layers = graphed
     block._graph_shape = tuple(sample_hidden.shape)
-    return len(block._graphed_layers)
+    return len(graphed)
 
 
 def graphed_layer_or_none(block, index: int, hidden_states: torch.Tensor,
