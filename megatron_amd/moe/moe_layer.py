@@ -30,6 +30,7 @@ class MoELayer(nn.Module):
             self.dispatcher = MoEAllGatherTokenDispatcher(config)
         self.experts = GroupedMLP(config)
         self.shared_expert = None
+        self._comm_stream = None
         if config.moe_shared_expert_intermediate_size:
             from megatron_amd.transformer.mlp import MLP
 
@@ -46,12 +47,31 @@ class MoELayer(nn.Module):
 
         self.router.seq_len = full.shape[0]
         probs, top_idx = self.router(tokens)
-        dispatched, tokens_per_expert = self.dispatcher.dispatch(tokens, probs, top_idx)
+
+        # shared-expert / dispatch-a2a overlap (reference shared_experts.py +
+        # combined_1f1b.py's a2a-hiding intent): the EP all-to-all runs on a
+        # dedicated HIP stream while the shared expert's GEMMs fill the
+        # compute units; autograd replays each op on its recording stream so
+        # the backward overlaps the same way.  On CPU streams are no-ops.
+        shared_out = None
+        if self.shared_expert is not None and torch.cuda.is_available() and tokens.is_cuda:
+            if self._comm_stream is None:
+                self._comm_stream = torch.cuda.Stream()
+            self._comm_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._comm_stream):
+                dispatched, tokens_per_expert = self.dispatcher.dispatch(tokens, probs, top_idx)
+            shared_out = self.shared_expert(full.reshape(-1, h))
+            torch.cuda.current_stream().wait_stream(self._comm_stream)
+        else:
+            dispatched, tokens_per_expert = self.dispatcher.dispatch(tokens, probs, top_idx)
+            if self.shared_expert is not None:
+                shared_out = self.shared_expert(full.reshape(-1, h))
+
         expert_out = self.experts(dispatched, tokens_per_expert)
         out = self.dispatcher.combine(expert_out)
 
-        if self.shared_expert is not None:
-            out = out + self.shared_expert(full.reshape(-1, h))
+        if shared_out is not None:
+            out = out + shared_out
 
         for aux in self.router.aux_losses.values():
             out = AuxLossScaler.apply(out, aux)

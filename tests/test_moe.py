@@ -202,3 +202,60 @@ def test_expert_bias_update_moves_toward_balance():
         elif counts[e] < mean:
             assert bias[e] > 0
     assert int(layer.router.local_tokens_per_expert.sum()) == 0
+
+
+def _ep_shared_case(rank, world):
+    G.initialize_model_parallel(expert_parallel_size=world)
+    model_parallel_seed(1234)
+    cfg = _cfg(ep=world, moe_shared_expert_intermediate_size=40)
+    layer = MoELayer(cfg)
+
+    def fullfill(shape, key):
+        g = torch.Generator().manual_seed(zlib.crc32(key.encode()) % (2**31))
+        return torch.randn(shape, generator=g) * 0.1
+
+    n_local = cfg.num_experts // world
+    with torch.no_grad():
+        layer.router.weight.copy_(fullfill((4, 32), "router"))
+        w1 = fullfill((4, 96, 32), "w1")
+        w2 = fullfill((4, 32, 48), "w2")
+        layer.experts.weight1.copy_(w1[rank * n_local:(rank + 1) * n_local])
+        layer.experts.weight2.copy_(w2[rank * n_local:(rank + 1) * n_local])
+        for i, p in enumerate(layer.shared_expert.parameters()):
+            p.copy_(fullfill(tuple(p.shape), f"shared{i}"))
+    torch.manual_seed(1)
+    x = torch.randn(6, 2, 32)
+    out = layer(x)
+    if rank == 0:
+        torch.save(out.detach(), os.environ["MOE_TEST_OUT"])
+
+
+def test_ep2_shared_expert_matches_ep1(tmp_path, monkeypatch):
+    """Shared-expert path (overlapped with dispatch on GPU) must stay
+    equivalent to the EP=1 sequential computation."""
+    out_path = tmp_path / "moe_shared.pt"
+    monkeypatch.setenv("MOE_TEST_OUT", str(out_path))
+    init_single()
+    cfg = _cfg(moe_shared_expert_intermediate_size=40)
+    layer = MoELayer(cfg)
+
+    def fullfill(shape, key):
+        g = torch.Generator().manual_seed(zlib.crc32(key.encode()) % (2**31))
+        return torch.randn(shape, generator=g) * 0.1
+
+    with torch.no_grad():
+        layer.router.weight.copy_(fullfill((4, 32), "router"))
+        layer.experts.weight1.copy_(fullfill((4, 96, 32), "w1"))
+        layer.experts.weight2.copy_(fullfill((4, 32, 48), "w2"))
+        for i, p in enumerate(layer.shared_expert.parameters()):
+            p.copy_(fullfill(tuple(p.shape), f"shared{i}"))
+    torch.manual_seed(1)
+    x = torch.randn(6, 2, 32)
+    ref = layer(x)
+    # backward works through the shared path
+    ref.sum().backward()
+    assert next(layer.shared_expert.parameters()).grad is not None
+
+    spawn_dist(_ep_shared_case, 2)
+    out = torch.load(out_path)
+    assert_close(out, ref.detach(), rtol=1e-4, atol=1e-5)
