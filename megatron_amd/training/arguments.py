@@ -70,6 +70,13 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--moe-shared-expert-intermediate-size", type=int, default=None)
     g.add_argument("--moe-token-dispatcher-type", choices=["alltoall", "allgather"], default="alltoall")
     g.add_argument("--moe-layer-freq", type=int, default=1)
+    g.add_argument("--moe-router-pre-softmax", action="store_true", default=False)
+    g.add_argument("--moe-router-num-groups", type=int, default=None)
+    g.add_argument("--moe-router-group-topk", type=int, default=None)
+    g.add_argument("--moe-input-jitter-eps", type=float, default=None)
+    g.add_argument("--moe-aux-loss-type", choices=["aux", "seq_aux"], default="aux")
+    g.add_argument("--moe-router-enable-expert-bias", action="store_true", default=False)
+    g.add_argument("--moe-router-bias-update-rate", type=float, default=1e-3)
 
     g = p.add_argument_group("parallelism")
     g.add_argument("--tensor-model-parallel-size", "--tp", type=int, default=1)
@@ -102,6 +109,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--deterministic-mode", action="store_true")
 
     g = p.add_argument_group("optimizer")
+    g.add_argument("--optimizer", choices=["adam", "muon"], default="adam")
+    g.add_argument("--optimizer-cpu-offload", action="store_true", default=False)
+    g.add_argument("--muon-momentum", type=float, default=0.95)
     g.add_argument("--lr", type=float, default=3e-4)
     g.add_argument("--min-lr", type=float, default=0.0)
     g.add_argument("--lr-decay-style", choices=["constant", "linear", "cosine", "wsd"], default="cosine")
@@ -150,6 +160,10 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--wandb-project", type=str, default=None)
     g.add_argument("--check-weight-hash-across-dp-replicas-interval", type=int, default=None)
     g.add_argument("--rerun-mode", choices=["disabled", "validate_results"], default="disabled")
+    g.add_argument("--fault-injection-type", choices=["crash", "hang", "nan_loss"], default=None)
+    g.add_argument("--fault-injection-iteration", type=int, default=None)
+    g.add_argument("--fault-injection-ranks", type=int, nargs="*", default=[])
+    g.add_argument("--inprocess-restarts", type=int, default=0)
     g.add_argument("--log-energy", action="store_true")
 
     g = p.add_argument_group("profiling")
@@ -228,6 +242,13 @@ def configs_from_args(args):
         moe_shared_expert_intermediate_size=args.moe_shared_expert_intermediate_size,
         moe_token_dispatcher_type=args.moe_token_dispatcher_type,
         moe_layer_freq=args.moe_layer_freq,
+        moe_router_pre_softmax=args.moe_router_pre_softmax,
+        moe_router_num_groups=args.moe_router_num_groups,
+        moe_router_group_topk=args.moe_router_group_topk,
+        moe_input_jitter_eps=args.moe_input_jitter_eps,
+        moe_aux_loss_type=args.moe_aux_loss_type,
+        moe_router_enable_expert_bias=args.moe_router_enable_expert_bias,
+        moe_router_bias_update_rate=args.moe_router_bias_update_rate,
         tensor_parallel_size=args.tensor_model_parallel_size,
         pipeline_parallel_size=args.pipeline_model_parallel_size,
         virtual_pipeline_parallel_size=args.virtual_pipeline_model_parallel_size,
@@ -247,6 +268,9 @@ def configs_from_args(args):
         gradient_accumulation_fusion=torch.cuda.is_available(),
     )
     opt_cfg = OptimizerConfig(
+        optimizer=args.optimizer,
+        optimizer_cpu_offload=args.optimizer_cpu_offload,
+        muon_momentum=args.muon_momentum,
         lr=args.lr, min_lr=args.min_lr, weight_decay=args.weight_decay,
         adam_beta1=args.adam_beta1, adam_beta2=args.adam_beta2, adam_eps=args.adam_eps,
         clip_grad=args.clip_grad, bf16=args.bf16, fp16=args.fp16,

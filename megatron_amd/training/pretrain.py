@@ -133,6 +133,14 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
     chunks, optimizer = setup_model_and_optimizer(model_provider, cfg, opt_cfg, ddp_cfg, device=device)
     scheduler = OptimizerParamScheduler(optimizer, opt_cfg, args.train_iters)
     initialize_rerun_state_machine(args.rerun_mode)
+    fault_injector = None
+    if getattr(args, "fault_injection_type", None):
+        from megatron_amd.utils.fault_injection import FaultInjector, FaultInjectorConfig
+
+        fault_injector = FaultInjector(FaultInjectorConfig(
+            enabled=True, fault_type=args.fault_injection_type,
+            at_iteration=args.fault_injection_iteration,
+            ranks=list(args.fault_injection_ranks)), rank=args.rank)
     straggler = StragglerDetector(enabled=args.log_straggler)
     energy = EnergyMonitor() if args.log_energy else None
     metrics = MetricsLogger(args.tensorboard_dir, rank=args.rank,
@@ -207,6 +215,8 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                             args.num_microbatches, args.seq_length, args.micro_batch_size)
         straggler.stop()
         timers("iteration").stop()
+        if fault_injector is not None:
+            fault_injector.maybe_inject(iteration)
         if "exit_code" in result:
             if args.save:
                 save_checkpoint(args.save, chunks, optimizer, iteration, scheduler)
