@@ -102,6 +102,9 @@ class TransformerConfig(ParallelConfig):
     # each step, expert_bias += rate * sign(mean_load - expert_load).
     moe_router_enable_expert_bias: bool = False
     moe_router_bias_update_rate: float = 1e-3
+    # MuonClip: clip attention-logit growth by rescaling q/k weights after
+    # each step (reference optimizer/qk_clip.py); None disables tracking.
+    qk_clip_threshold: Optional[float] = None
     # layer frequency: 1 = every layer is MoE, k = every k-th layer
     moe_layer_freq: int = 1
 

@@ -77,6 +77,11 @@ class SelfAttention(nn.Module):
             q = ops.rope_apply(q, rotary_freqs)
             k = ops.rope_apply(k, rotary_freqs)
 
+        if getattr(self.config, "qk_clip_threshold", None) and self.training:
+            from megatron_amd.optimizer.qk_clip import max_logits_per_group
+
+            self.last_max_logit = max_logits_per_group(q, k, self.softmax_scale)
+
         if inference_context is not None:
             # the context owns the KV cache and the attention kernel choice
             # (contiguous flash for prefill, paged masked decode for dynamic)
