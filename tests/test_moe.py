@@ -259,3 +259,56 @@ def test_ep2_shared_expert_matches_ep1(tmp_path, monkeypatch):
     spawn_dist(_ep_shared_case, 2)
     out = torch.load(out_path)
     assert_close(out, ref.detach(), rtol=1e-4, atol=1e-5)
+
+
+def test_upcycling_dense_equivalence():
+    """Upcycled MoE with identical experts and softmax(topk) probs (sum=1)
+    must reproduce the dense model's logits exactly (reference
+    upcycling_utils invariant)."""
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.moe.upcycling import upcycle_dense_to_moe
+    from megatron_amd.config import TransformerConfig
+
+    init_single()
+    common = dict(num_layers=2, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+                  vocab_size=64, ffn_hidden_size=48, gradient_accumulation_fusion=False)
+    torch.manual_seed(5)
+    dense = GPTModel(TransformerConfig(**common))
+    torch.manual_seed(99)
+    moe = GPTModel(TransformerConfig(**common, num_experts=4, moe_router_topk=2,
+                                     moe_ffn_hidden_size=48))
+    n = upcycle_dense_to_moe(dense, moe)
+    assert n == 2
+    toks = torch.randint(0, 64, (2, 8))
+    with torch.no_grad():
+        out_d = dense(toks, position_ids=None, attention_mask=None)
+        out_m = moe(toks, position_ids=None, attention_mask=None)
+    assert_close(out_m, out_d, rtol=1e-5, atol=1e-5)
+    # with noise the experts diverge
+    upcycle_dense_to_moe(dense, moe, noise_std=0.02)
+    with torch.no_grad():
+        out_n = moe(toks, position_ids=None, attention_mask=None)
+    assert not torch.allclose(out_n, out_d, atol=1e-4)
+
+
+def test_moe_stats_tracker():
+    from megatron_amd.moe.moe_logging import MoEStatsTracker
+
+    init_single()
+    cfg = _cfg(moe_aux_loss_coeff=0.01)
+    layer = MoELayer(cfg)
+    tracker = MoEStatsTracker(layer)
+    assert len(tracker.routers) == 1
+    x = torch.randn(6, 2, cfg.hidden_size)
+    layer(x)
+    tracker.collect()
+    layer(x)
+    tracker.collect()
+    rep = tracker.report()
+    st = rep["moe_layer_0"]
+    assert st["tokens_routed"] == 2 * 12 * cfg.moe_router_topk
+    assert st["max_violation"] >= 1.0
+    assert 0.0 < st["expert_utilization"] <= 1.0
+    assert "load_balancing_loss" in st
+    # reset happened
+    assert tracker.report()["moe_layer_0"]["tokens_routed"] == 0
