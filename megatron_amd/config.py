@@ -213,6 +213,9 @@ class OptimizerConfig:
     optimizer_cpu_offload: bool = False
     lr: float = 1e-4
     min_lr: float = 0.0
+    # separate lr for input/output embeddings (reference --decoupled-lr)
+    decoupled_lr: Optional[float] = None
+    decoupled_min_lr: Optional[float] = None
     weight_decay: float = 0.1
     adam_beta1: float = 0.9
     adam_beta2: float = 0.95

@@ -41,6 +41,9 @@ class GPTModel(nn.Module):
 
         if pre_process:
             self.embedding = VocabParallelEmbedding(config.vocab_size, config.hidden_size, config=config)
+            # decoupled-lr group + Muon exclusion (reference is_embedding_or_output_parameter)
+            self.embedding.weight.is_embedding_or_output_parameter = True
+            self.embedding.weight.muon_exclude = True
             if config.position_embedding_type == "learned":
                 self.position_embedding = nn.Embedding(
                     config.max_position_embeddings, config.hidden_size, dtype=config.params_dtype
@@ -53,6 +56,8 @@ class GPTModel(nn.Module):
                 config.hidden_size, config.vocab_size, config=config, bias=False, gather_output=False,
                 skip_bias_add=True,
             )
+            self.output_layer.weight.is_embedding_or_output_parameter = True
+            self.output_layer.weight.muon_exclude = True
             if self.share_embeddings_and_output_weights and pre_process:
                 self.output_layer.weight = self.embedding.weight
             self.mtp = None

@@ -21,9 +21,41 @@ from megatron_amd.optimizer.dist_optimizer import DistributedOptimizer  # noqa: 
 from megatron_amd.optimizer.scheduler import OptimizerParamScheduler  # noqa: F401
 
 
+def _is_embedding_or_output(p) -> bool:
+    return getattr(p, "is_embedding_or_output_parameter", False)
+
+
+def _subclass_with(opt_cls, flt, lr_ratio: float):
+    """Specialize an optimizer class with a param filter + lr multiplier
+    (decoupled-lr groups, reference optimizer/__init__.py lr_mult)."""
+
+    class _Grouped(opt_cls):
+        pass
+
+    _Grouped.__name__ = opt_cls.__name__
+    _Grouped.param_filter = staticmethod(flt) if flt is not None else None
+    _Grouped.lr_ratio = lr_ratio
+    return _Grouped
+
+
 def get_optimizer(opt_config: OptimizerConfig, model_chunks: List) -> "ChainedOptimizer":
     """Build the optimizer for a list of DDP-wrapped model chunks
-    (reference optimizer/__init__.py:991 get_megatron_optimizer)."""
+    (reference optimizer/__init__.py:991 get_megatron_optimizer).  With
+    `decoupled_lr` set, embedding/output params go to a second chained
+    sub-optimizer whose lr tracks the scheduler at the decoupled ratio."""
+    if opt_config.decoupled_lr is not None:
+        assert not opt_config.use_distributed_optimizer, \
+            "decoupled-lr groups not supported with the distributed optimizer yet"
+        base_cls = (FP32Optimizer if not (opt_config.bf16 or opt_config.fp16)
+                    else MixedPrecisionOptimizer)
+        if opt_config.optimizer == "muon":
+            from megatron_amd.optimizer.muon import MuonOptimizer
+
+            base_cls = MuonOptimizer
+        ratio = opt_config.decoupled_lr / opt_config.lr
+        main = _subclass_with(base_cls, lambda p: not _is_embedding_or_output(p), 1.0)
+        emb = _subclass_with(base_cls, _is_embedding_or_output, ratio)
+        return ChainedOptimizer([main(opt_config, model_chunks), emb(opt_config, model_chunks)])
     if opt_config.optimizer == "muon":
         from megatron_amd.optimizer.muon import MuonOptimizer
 

@@ -22,11 +22,11 @@ from megatron_amd.optimizer.clip import (
 )
 
 
-def _model_chunks_params(model_chunks) -> List[torch.nn.Parameter]:
+def _model_chunks_params(model_chunks, param_filter=None) -> List[torch.nn.Parameter]:
     params = []
     for chunk in model_chunks:
         for p in chunk.parameters():
-            if p.requires_grad:
+            if p.requires_grad and (param_filter is None or param_filter(p)):
                 params.append(p)
     return params
 
@@ -38,16 +38,21 @@ def _wd_group(param, name_hint: str = "") -> bool:
 
 
 class _BaseOptimizer:
+    # scheduler lr is multiplied by this (decoupled-lr groups, reference
+    # optimizer/__init__.py lr_mult for embedding/output params)
+    lr_ratio: float = 1.0
+    param_filter = None
+
     def __init__(self, config: OptimizerConfig, model_chunks: List):
         self.config = config
         self.model_chunks = model_chunks
         self.step_count = 0
-        self._lr = config.lr
+        self._lr = config.lr * self.lr_ratio
         self._wd = config.weight_decay
 
     # scheduler interface
     def set_lr(self, lr: float):
-        self._lr = lr
+        self._lr = lr * self.lr_ratio
 
     def set_wd(self, wd: float):
         self._wd = wd
@@ -83,7 +88,7 @@ class FP32Optimizer(_BaseOptimizer):
 
     def __init__(self, config: OptimizerConfig, model_chunks: List):
         super().__init__(config, model_chunks)
-        self.params = _model_chunks_params(model_chunks)
+        self.params = _model_chunks_params(model_chunks, self.param_filter)
         self.exp_avg = [torch.zeros_like(p, dtype=torch.float32) for p in self.params]
         self.exp_avg_sq = [torch.zeros_like(p, dtype=torch.float32) for p in self.params]
 
@@ -147,7 +152,7 @@ class MixedPrecisionOptimizer(_BaseOptimizer):
 
     def __init__(self, config: OptimizerConfig, model_chunks: List):
         super().__init__(config, model_chunks)
-        self.params = _model_chunks_params(model_chunks)
+        self.params = _model_chunks_params(model_chunks, self.param_filter)
         self.main_params = [p.detach().clone().float() for p in self.params]
         self.exp_avg = [torch.zeros_like(mp) for mp in self.main_params]
         self.exp_avg_sq = [torch.zeros_like(mp) for mp in self.main_params]

@@ -114,6 +114,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--muon-momentum", type=float, default=0.95)
     g.add_argument("--lr", type=float, default=3e-4)
     g.add_argument("--min-lr", type=float, default=0.0)
+    g.add_argument("--decoupled-lr", type=float, default=None)
+    g.add_argument("--decoupled-min-lr", type=float, default=None)
     g.add_argument("--lr-decay-style", choices=["constant", "linear", "cosine", "wsd"], default="cosine")
     g.add_argument("--lr-warmup-iters", type=int, default=0)
     g.add_argument("--lr-decay-iters", type=int, default=None)
@@ -271,7 +273,8 @@ def configs_from_args(args):
         optimizer=args.optimizer,
         optimizer_cpu_offload=args.optimizer_cpu_offload,
         muon_momentum=args.muon_momentum,
-        lr=args.lr, min_lr=args.min_lr, weight_decay=args.weight_decay,
+        lr=args.lr, min_lr=args.min_lr,
+        decoupled_lr=args.decoupled_lr, decoupled_min_lr=args.decoupled_min_lr, weight_decay=args.weight_decay,
         adam_beta1=args.adam_beta1, adam_beta2=args.adam_beta2, adam_eps=args.adam_eps,
         clip_grad=args.clip_grad, bf16=args.bf16, fp16=args.fp16,
         loss_scale=args.loss_scale,

@@ -124,3 +124,40 @@ def test_cpu_offload_selected_by_config():
     cfg = OptimizerConfig(lr=1e-3, optimizer_cpu_offload=True)
     opt = get_optimizer(cfg, [model])
     assert isinstance(opt.chained_optimizers[0], CPUOffloadOptimizer)
+
+
+def test_decoupled_lr_groups():
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.optimizer.scheduler import OptimizerParamScheduler
+
+    init_single()
+    torch.manual_seed(0)
+    model = GPTModel(TransformerConfig(
+        num_layers=1, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+        vocab_size=64, ffn_hidden_size=48, gradient_accumulation_fusion=False))
+    cfg = OptimizerConfig(lr=1e-3, decoupled_lr=1e-4, lr_decay_style="constant")
+    opt = get_optimizer(cfg, [model])
+    assert len(opt.chained_optimizers) == 2
+    main, emb = opt.chained_optimizers
+    emb_names = {id(p) for p in emb.params}
+    assert id(model.embedding.weight) in emb_names
+    assert id(model.output_layer.weight) in emb_names
+    for p in main.params:
+        assert not getattr(p, "is_embedding_or_output_parameter", False)
+    # every trainable param is in exactly one group
+    assert len(main.params) + len(emb.params) == sum(1 for p in model.parameters() if p.requires_grad)
+    # scheduler drives both lrs at their ratios
+    opt.set_lr(5e-4)
+    assert abs(main.get_lr() - 5e-4) < 1e-12
+    assert abs(emb.get_lr() - 5e-5) < 1e-12
+    # a step trains both groups
+    toks = torch.randint(0, 64, (2, 8))
+    out = model(toks, position_ids=None, attention_mask=None)
+    out.square().mean().backward()
+    w_emb = model.embedding.weight.detach().clone()
+    w_main = model.decoder.layers[0].self_attention.linear_qkv.weight.detach().clone()
+    ok, _, _ = opt.step()
+    assert ok
+    assert not torch.equal(w_emb, model.embedding.weight)
+    assert not torch.equal(w_main, model.decoder.layers[0].self_attention.linear_qkv.weight)
