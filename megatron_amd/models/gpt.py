@@ -111,6 +111,7 @@ class GPTModel(nn.Module):
         labels: Optional[torch.Tensor] = None,
         loss_mask: Optional[torch.Tensor] = None,
         inference_context=None,
+        packed_seq_params=None,
     ):
         """input_ids/labels: [b, s].  Returns loss [s, b] (labels given) or
         logits [s, b, V/tp]."""
@@ -146,10 +147,15 @@ class GPTModel(nn.Module):
             table = self._rotary_freqs(self.config.max_position_embeddings, hidden.device)
             pos = inference_context.rope_positions(seq_len)  # [s] or [s, b]
             rotary = table[pos]
+        elif packed_seq_params is not None and self.config.position_embedding_type == "rope":
+            # THD pack: positions restart at each document boundary
+            table = self._rotary_freqs(packed_seq_params.max_seqlen, hidden.device)
+            rotary = table[packed_seq_params.positions().to(hidden.device)]
         else:
             rotary = self._rotary_freqs(seq_len, hidden.device)
         hidden = self.decoder(hidden, rotary_freqs=rotary, attention_mask=attention_mask,
-                              inference_context=inference_context)
+                              inference_context=inference_context,
+                              packed_seq_params=packed_seq_params)
 
         if not self.post_process:
             return hidden

@@ -107,6 +107,40 @@ def bias_dropout_add(
     return residual + x
 
 
+def attention_varlen(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cu_seqlens: torch.Tensor,
+    causal: bool = True,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Packed (THD) attention: q/k/v [t, 1, h, d] with document boundaries
+    at cu_seqlens; each token attends only within its own document
+    (block-diagonal causal mask).  fp32 softmax, GQA via repeat."""
+    t, b, hq, d = q.shape
+    assert b == 1, "packed sequences use batch 1 (thd layout)"
+    hkv = k.shape[2]
+    rep = hq // hkv
+    if scale is None:
+        scale = 1.0 / math.sqrt(d)
+    qf = q.permute(1, 2, 0, 3).float()  # [1, hq, t, d]
+    kf = k.permute(1, 2, 0, 3).float()
+    vf = v.permute(1, 2, 0, 3).float()
+    if rep > 1:
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [1, hq, t, t]
+    seg = torch.bucketize(torch.arange(t, device=q.device), cu_seqlens[1:-1], right=True)
+    mask = seg.view(t, 1) == seg.view(1, t)
+    if causal:
+        mask = mask & torch.ones(t, t, dtype=torch.bool, device=q.device).tril_()
+    scores = scores.masked_fill(~mask, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    out = torch.matmul(probs, vf)  # [1, hq, t, d]
+    return out.permute(2, 0, 1, 3).to(q.dtype)
+
+
 def attention(
     q: torch.Tensor,
     k: torch.Tensor,

@@ -55,7 +55,7 @@ class SelfAttention(nn.Module):
             self.k_layernorm = None
 
     def forward(self, hidden_states: torch.Tensor, rotary_freqs: Optional[torch.Tensor] = None,
-                attention_mask=None, inference_context=None) -> torch.Tensor:
+                attention_mask=None, inference_context=None, packed_seq_params=None) -> torch.Tensor:
         # hidden_states: [s(/tp if SP), b, h]
         qkv, _ = self.linear_qkv(hidden_states)  # [s, b, qkv_size/tp]
         s, b = qkv.shape[0], qkv.shape[1]
@@ -82,7 +82,13 @@ class SelfAttention(nn.Module):
 
             self.last_max_logit = max_logits_per_group(q, k, self.softmax_scale)
 
-        if inference_context is not None:
+        if packed_seq_params is not None and inference_context is None:
+            from megatron_amd.ops import reference as _ref
+
+            core_out = _ref.attention_varlen(
+                q, k, v, packed_seq_params.cu_seqlens.to(q.device),
+                causal=self.config.causal_attention, scale=self.softmax_scale)
+        elif inference_context is not None:
             # the context owns the KV cache and the attention kernel choice
             # (contiguous flash for prefill, paged masked decode for dynamic)
             core_out = inference_context.attend(
