@@ -170,6 +170,10 @@ def build_arg_parser() -> argparse.ArgumentParser:
 
     g = p.add_argument_group("profiling")
     g.add_argument("--profile", action="store_true")
+    g.add_argument("--gpu-sniff-test", action="store_true",
+                   help="pre-run GEMM health check on every visible GPU")
+    g.add_argument("--profile-ranges", action="store_true",
+                   help="emit rocTX/profiler range annotations")
     g.add_argument("--profile-step-start", type=int, default=3)
     g.add_argument("--profile-step-end", type=int, default=5)
     g.add_argument("--profile-dir", type=str, default="./torchprof")

@@ -132,6 +132,16 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
 
     chunks, optimizer = setup_model_and_optimizer(model_provider, cfg, opt_cfg, ddp_cfg, device=device)
     scheduler = OptimizerParamScheduler(optimizer, opt_cfg, args.train_iters)
+    if getattr(args, "gpu_sniff_test", False):
+        from megatron_amd.utils.gpu_health import gpu_sniff_test
+
+        problems = gpu_sniff_test()
+        if problems:
+            raise RuntimeError("GPU health check failed: " + "; ".join(problems))
+    if getattr(args, "profile_ranges", False):
+        from megatron_amd.utils.annotations import enable_annotations
+
+        enable_annotations(True)
     initialize_rerun_state_machine(args.rerun_mode)
     fault_injector = None
     if getattr(args, "fault_injection_type", None):
