@@ -63,3 +63,25 @@ def test_yaml_config_overlay(tmp_path):
 
     with _pytest.raises(ValueError, match="unknown yaml"):
         parse_and_validate_args(["--yaml-cfg", str(bad)])
+
+
+def test_pretrain_t5_entry(tmp_path):
+    import pretrain_t5
+
+    it = pretrain(pretrain_t5.model_provider,
+                  TINY + ["--train-iters", "2", "--seed", "3"],
+                  forward_step_builder=pretrain_t5.forward_step_builder)
+    assert it == 2
+
+
+def test_pretrain_muon_and_fault_injection(tmp_path):
+    from megatron_amd.utils.fault_injection import InjectedFault
+
+    # muon trains through the CLI path
+    it = pretrain(model_provider, TINY + ["--train-iters", "2", "--optimizer", "muon"])
+    assert it == 2
+    # an injected crash surfaces at the configured iteration
+    with pytest.raises(InjectedFault):
+        pretrain(model_provider, TINY + ["--train-iters", "5",
+                                         "--fault-injection-type", "crash",
+                                         "--fault-injection-iteration", "1"])
