@@ -225,3 +225,20 @@ def test_kv_offloader_round_trip(tiny_model):
     for l in range(2):
         assert torch.equal(ctx.k_cache[l][new_blocks], k_before[l])
         assert torch.equal(ctx.v_cache[l][new_blocks], v_before[l])
+
+
+def test_disaggregated_prefill_decode_matches_single(tiny_model):
+    from megatron_amd.inference.disaggregation import disaggregated_generate
+
+    params = SamplingParams(max_tokens=8, greedy=True, stop_on_eod=False)
+    prompts = [[3, 7, 11, 2, 9], [5, 1], [8, 8, 4, 2, 1, 0, 9]]
+    single = DynamicInferenceEngine(tiny_model, num_blocks=32, block_size=4)
+    expected = single.generate(prompts, params)
+
+    prefill_eng = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=4)
+    decode_eng = DynamicInferenceEngine(tiny_model, num_blocks=32, block_size=4)
+    got = disaggregated_generate(prefill_eng, decode_eng, prompts, params)
+    for a, b in zip(expected, got):
+        assert a.output_tokens == b.output_tokens
+    # prefill pool fully drained back
+    assert prefill_eng.context.allocator.num_free == 16
