@@ -18,6 +18,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor out, torch::Tensor lse,
                                     bool causal, double scale, long window);
+std::vector<int64_t> symm_ipc_handle(torch::Tensor buf);
+int64_t symm_open_handle(std::vector<int64_t> bytes);
+void symm_close_handle(int64_t ptr);
+void symm_allreduce(std::vector<int64_t> peer_ptrs, int64_t payload_bytes,
+                    torch::Tensor local_in, torch::Tensor out, int64_t rank,
+                    int64_t seq);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -31,4 +37,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_gemm_accum", &wgrad_gemm_accum);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("symm_ipc_handle", &symm_ipc_handle);
+  m.def("symm_open_handle", &symm_open_handle);
+  m.def("symm_close_handle", &symm_close_handle);
+  m.def("symm_allreduce", &symm_allreduce);
 }

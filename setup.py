@@ -22,6 +22,7 @@ SRC = [
     "megatron_amd/ops/csrc/wgrad.hip",
     "megatron_amd/ops/csrc/attention_fwd.hip",
     "megatron_amd/ops/csrc/attention_bwd.hip",
+    "megatron_amd/ops/csrc/symm_allreduce.hip",
 ]
 
 setup(

@@ -176,3 +176,17 @@ def test_profile_annotations():
         assert any("marked-region" in n for n in names)
     finally:
         enable_annotations(False)
+
+
+def test_symmetric_allreduce_fallback_cpu():
+    """Without CUDA/IPC the wrapper must transparently fall back (identity
+    at world 1); the kernel itself needs a multi-GPU node (driver tier)."""
+    import torch
+
+    from megatron_amd.parallel.symm_collectives import SymmetricAllReduce
+
+    ar = SymmetricAllReduce()
+    assert not ar.enabled
+    t = torch.randn(8)
+    out = ar.all_reduce(t.clone())
+    assert torch.equal(out, t)
