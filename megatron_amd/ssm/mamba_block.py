@@ -31,7 +31,7 @@ class MambaLayer(nn.Module):
         self.hidden_dropout = config.hidden_dropout
 
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None,
-                inference_context=None, inference_state=None):
+                inference_context=None, inference_state=None, packed_seq_params=None):
         residual = hidden_states
         x = self.norm(hidden_states)
         x = self.mixer(x, inference_state=inference_state)
@@ -48,7 +48,7 @@ class AttentionLayer(nn.Module):
         self.hidden_dropout = config.hidden_dropout
 
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None,
-                inference_context=None, inference_state=None):
+                inference_context=None, inference_state=None, packed_seq_params=None):
         residual = hidden_states
         x = self.norm(hidden_states)
         x = self.self_attention(x, rotary_freqs=rotary_freqs, attention_mask=attention_mask,
@@ -66,7 +66,7 @@ class MLPLayer(nn.Module):
         self.hidden_dropout = config.hidden_dropout
 
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None,
-                inference_context=None, inference_state=None):
+                inference_context=None, inference_state=None, packed_seq_params=None):
         residual = hidden_states
         x = self.norm(hidden_states)
         x = self.mlp(x)

@@ -70,8 +70,9 @@ class MLASelfAttention(nn.Module):
         return self._freq_cache[key]
 
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None,
-                inference_context=None):
+                inference_context=None, packed_seq_params=None):
         assert inference_context is None, "MLA KV-cache (absorbed) path: round 2"
+        assert packed_seq_params is None, "MLA packed (THD) path: round 2"
         s, b = hidden_states.shape[0], hidden_states.shape[1]
         if self.linear_q_down is not None:
             q, _ = self.linear_q_up(self.q_norm(self.linear_q_down(hidden_states)))
