@@ -19,27 +19,61 @@ from megatron_amd.parallel import grid as G
 from megatron_amd.parallel.mappings import all_to_all
 
 
-def permute(tokens: torch.Tensor, top_idx: torch.Tensor, num_experts: int):
+def expert_capacity_mask(top_idx: torch.Tensor, probs: torch.Tensor,
+                         num_experts: int, capacity: int,
+                         drop_policy: str = "probs") -> torch.Tensor:
+    """Keep mask [T*k]: at most `capacity` (token, slot) assignments kept
+    per expert (reference router capacity / moe_token_drop_policy).
+    'probs' keeps the highest-probability assignments; 'position' the
+    earliest tokens."""
+    T, k = top_idx.shape
+    n = T * k
+    flat = top_idx.reshape(-1)
+    if drop_policy == "probs":
+        order = torch.argsort(probs.reshape(-1).float(), descending=True, stable=True)
+    else:
+        order = torch.arange(n, device=flat.device)
+    priority = torch.empty(n, dtype=torch.long, device=flat.device)
+    priority[order] = torch.arange(n, device=flat.device)
+    # rank of each assignment within its expert, by priority
+    grouped = torch.argsort(flat * (n + 1) + priority)
+    counts = torch.bincount(flat, minlength=num_experts)
+    starts = torch.cumsum(counts, 0) - counts
+    ranks = torch.empty(n, dtype=torch.long, device=flat.device)
+    ranks[grouped] = torch.arange(n, device=flat.device) - starts.repeat_interleave(counts)
+    return ranks < capacity
+
+
+def permute(tokens: torch.Tensor, top_idx: torch.Tensor, num_experts: int,
+            keep_mask: torch.Tensor = None):
     """Replicate+sort tokens by target expert.
 
-    tokens [T, h], top_idx [T, k] -> (permuted [T*k, h], sort_order [T*k],
-    tokens_per_expert [E]).  HIP gather kernel (K12) replaces the
-    index_select later; torch path is already a single gather.
-    """
+    tokens [T, h], top_idx [T, k] -> (permuted [n_kept, h], sort_order
+    [n_kept] (global (token,slot) flat ids in expert order),
+    tokens_per_expert [E]).  keep_mask [T*k] drops capacity-overflow
+    assignments (their tokens ride the residual only).  HIP gather kernel
+    (K12) replaces the index_select later; torch path is a single gather."""
     T, k = top_idx.shape
     flat_experts = top_idx.reshape(-1)  # [T*k] expert id per (token, slot)
-    sort_order = torch.argsort(flat_experts, stable=True)
+    if keep_mask is not None:
+        kept = keep_mask.nonzero(as_tuple=True)[0]
+        order_within = torch.argsort(flat_experts[kept], stable=True)
+        sort_order = kept[order_within]
+    else:
+        sort_order = torch.argsort(flat_experts, stable=True)
     src_token = sort_order // k  # originating token row
     permuted = tokens.index_select(0, src_token)
-    tokens_per_expert = torch.bincount(flat_experts, minlength=num_experts)
+    tokens_per_expert = torch.bincount(flat_experts[sort_order], minlength=num_experts)
     return permuted, sort_order, tokens_per_expert
 
 
 def unpermute(permuted: torch.Tensor, sort_order: torch.Tensor, probs: torch.Tensor, T: int):
-    """Inverse of permute with prob weighting: out[t] = sum_k prob[t,k] * x[(t,k)]."""
+    """Inverse of permute with prob weighting: out[t] = sum_k prob[t,k] * x[(t,k)]
+    (dropped (t,k) slots contribute zero)."""
     k = probs.shape[1]
-    inv = torch.argsort(sort_order)
-    out_flat = permuted.index_select(0, inv)
+    out_flat = torch.zeros(T * k, permuted.shape[-1], dtype=permuted.dtype,
+                           device=permuted.device)
+    out_flat = out_flat.index_copy(0, sort_order, permuted)
     out_flat = out_flat.view(T, k, -1) * probs.unsqueeze(-1).to(permuted.dtype)
     return out_flat.sum(dim=1)
 
@@ -54,9 +88,21 @@ class MoEAlltoAllTokenDispatcher:
         self.num_local_experts = self.num_experts // max(self.ep, 1)
         self.group = G.get_grid().group("ep") if G.grid_initialized() else None
 
+    def _keep_mask(self, tokens, probs, top_idx):
+        cf = getattr(self.config, "moe_expert_capacity_factor", None)
+        if not cf:
+            return None
+        import math
+
+        T, k = top_idx.shape
+        cap = max(1, math.ceil(T * k / self.num_experts * cf))
+        return expert_capacity_mask(top_idx, probs, self.num_experts, cap,
+                                    getattr(self.config, "moe_token_drop_policy", "probs"))
+
     def dispatch(self, tokens: torch.Tensor, probs: torch.Tensor, top_idx: torch.Tensor):
         T = tokens.shape[0]
-        permuted, sort_order, tokens_per_expert = permute(tokens, top_idx, self.num_experts)
+        permuted, sort_order, tokens_per_expert = permute(
+            tokens, top_idx, self.num_experts, self._keep_mask(tokens, probs, top_idx))
         self._sort_order = sort_order
         self._T = T
         self._probs = probs
@@ -121,8 +167,20 @@ class MoEAllGatherTokenDispatcher:
         self.num_local_experts = self.num_experts // max(self.ep, 1)
         assert self.ep == 1, "allgather dispatcher here supports ep=1 (oracle use)"
 
+    def _keep_mask(self, tokens, probs, top_idx):
+        cf = getattr(self.config, "moe_expert_capacity_factor", None)
+        if not cf:
+            return None
+        import math
+
+        T, k = top_idx.shape
+        cap = max(1, math.ceil(T * k / self.num_experts * cf))
+        return expert_capacity_mask(top_idx, probs, self.num_experts, cap,
+                                    getattr(self.config, "moe_token_drop_policy", "probs"))
+
     def dispatch(self, tokens, probs, top_idx):
-        permuted, sort_order, tokens_per_expert = permute(tokens, top_idx, self.num_experts)
+        permuted, sort_order, tokens_per_expert = permute(
+            tokens, top_idx, self.num_experts, self._keep_mask(tokens, probs, top_idx))
         self._sort_order, self._T, self._probs = sort_order, tokens.shape[0], probs
         return permuted, tokens_per_expert
 

@@ -312,3 +312,39 @@ def test_moe_stats_tracker():
     assert "load_balancing_loss" in st
     # reset happened
     assert tracker.report()["moe_layer_0"]["tokens_routed"] == 0
+
+
+def test_expert_capacity_mask_policies():
+    from megatron_amd.moe.token_dispatcher import expert_capacity_mask
+
+    top_idx = torch.tensor([[0], [0], [0], [1]])
+    probs = torch.tensor([[0.1], [0.9], [0.5], [1.0]])
+    keep = expert_capacity_mask(top_idx, probs, num_experts=2, capacity=2, drop_policy="probs")
+    # expert 0 has 3 assignments; lowest-prob (0.1) dropped
+    assert keep.tolist() == [False, True, True, True]
+    keep_pos = expert_capacity_mask(top_idx, probs, num_experts=2, capacity=2, drop_policy="position")
+    assert keep_pos.tolist() == [True, True, False, True]
+
+
+def test_capacity_dropping_layer():
+    init_single()
+    cfg = _cfg(num_experts=4, moe_expert_capacity_factor=0.5)
+    layer = MoELayer(cfg)
+    _fill(layer)
+    x = torch.randn(8, 2, cfg.hidden_size)
+    out = layer(x)
+    assert out.shape == x.shape
+    # dropless twin differs (some tokens were dropped)
+    cfg2 = _cfg(num_experts=4)
+    layer2 = MoELayer(cfg2)
+    _fill(layer2)
+    out2 = layer2(x)
+    assert not torch.allclose(out, out2, atol=1e-5)
+    # with a huge capacity factor, dropping is a no-op -> dropless equality
+    cfg3 = _cfg(num_experts=4, moe_expert_capacity_factor=100.0)
+    layer3 = MoELayer(cfg3)
+    _fill(layer3)
+    assert_close(layer3(x), out2, rtol=1e-5, atol=1e-6)
+    # backward flows
+    out.sum().backward()
+    assert layer.experts.weight1.grad is not None
