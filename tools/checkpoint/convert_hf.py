@@ -1,4 +1,4 @@
-"""HuggingFace <-> megatron_amd checkpoint conversion (Llama family).
+"""HuggingFace <-> megatron_amd checkpoint conversion (Llama + Mixtral).
 
 Capability analog of reference tools/checkpoint/convert.py with the
 loader_llama/saver pairs: maps HF `LlamaForCausalLM` weights to our GPTModel
@@ -45,9 +45,22 @@ def hf_to_mcore_state_dict(hf_sd: dict, cfg) -> dict:
             groups.append(v[g * d:(g + 1) * d])
         out[us + "self_attention.linear_qkv.weight"] = torch.cat(groups, dim=0)
         out[us + "self_attention.linear_proj.weight"] = hf_sd[hf + "self_attn.o_proj.weight"]
-        out[us + "mlp.linear_fc1.weight"] = torch.cat(
-            [hf_sd[hf + "mlp.gate_proj.weight"], hf_sd[hf + "mlp.up_proj.weight"]], dim=0)
-        out[us + "mlp.linear_fc2.weight"] = hf_sd[hf + "mlp.down_proj.weight"]
+        if hf + "block_sparse_moe.gate.weight" in hf_sd:
+            # Mixtral MoE block: gate -> router, experts w1/w3 -> fused
+            # [gate;up] stacks, w2 -> weight2 [E, h, ffn]
+            out[us + "mlp.router.weight"] = hf_sd[hf + "block_sparse_moe.gate.weight"].float()
+            E = cfg.num_experts
+            w1s, w2s = [], []
+            for e in range(E):
+                ex = hf + f"block_sparse_moe.experts.{e}."
+                w1s.append(torch.cat([hf_sd[ex + "w1.weight"], hf_sd[ex + "w3.weight"]], dim=0))
+                w2s.append(hf_sd[ex + "w2.weight"])
+            out[us + "mlp.experts.weight1"] = torch.stack(w1s)
+            out[us + "mlp.experts.weight2"] = torch.stack(w2s)
+        else:
+            out[us + "mlp.linear_fc1.weight"] = torch.cat(
+                [hf_sd[hf + "mlp.gate_proj.weight"], hf_sd[hf + "mlp.up_proj.weight"]], dim=0)
+            out[us + "mlp.linear_fc2.weight"] = hf_sd[hf + "mlp.down_proj.weight"]
         out[us + "input_layernorm.weight"] = hf_sd[hf + "input_layernorm.weight"]
         out[us + "pre_mlp_layernorm.weight"] = hf_sd[hf + "post_attention_layernorm.weight"]
     out["decoder.final_layernorm.weight"] = hf_sd["model.norm.weight"]
@@ -79,10 +92,21 @@ def mcore_to_hf_state_dict(sd: dict, cfg) -> dict:
         out[hf + "self_attn.k_proj.weight"] = torch.cat(ks, 0)
         out[hf + "self_attn.v_proj.weight"] = torch.cat(vs, 0)
         out[hf + "self_attn.o_proj.weight"] = sd[us + "self_attention.linear_proj.weight"]
-        fc1 = sd[us + "mlp.linear_fc1.weight"]
-        out[hf + "mlp.gate_proj.weight"] = fc1[:ffn]
-        out[hf + "mlp.up_proj.weight"] = fc1[ffn:]
-        out[hf + "mlp.down_proj.weight"] = sd[us + "mlp.linear_fc2.weight"]
+        if us + "mlp.router.weight" in sd:
+            out[hf + "block_sparse_moe.gate.weight"] = sd[us + "mlp.router.weight"]
+            w1 = sd[us + "mlp.experts.weight1"]  # [E, 2*moe_ffn, h]
+            w2 = sd[us + "mlp.experts.weight2"]  # [E, h, moe_ffn]
+            moe_ffn = w1.shape[1] // 2
+            for e in range(w1.shape[0]):
+                ex = hf + f"block_sparse_moe.experts.{e}."
+                out[ex + "w1.weight"] = w1[e, :moe_ffn]
+                out[ex + "w3.weight"] = w1[e, moe_ffn:]
+                out[ex + "w2.weight"] = w2[e]
+        else:
+            fc1 = sd[us + "mlp.linear_fc1.weight"]
+            out[hf + "mlp.gate_proj.weight"] = fc1[:ffn]
+            out[hf + "mlp.up_proj.weight"] = fc1[ffn:]
+            out[hf + "mlp.down_proj.weight"] = sd[us + "mlp.linear_fc2.weight"]
         out[hf + "input_layernorm.weight"] = sd[us + "input_layernorm.weight"]
         out[hf + "post_attention_layernorm.weight"] = sd[us + "pre_mlp_layernorm.weight"]
     out["model.norm.weight"] = sd["decoder.final_layernorm.weight"]
@@ -104,6 +128,9 @@ def config_from_hf(hf_cfg):
         rotary_base=getattr(hf_cfg, "rope_theta", 10000.0),
         layernorm_epsilon=hf_cfg.rms_norm_eps,
         untie_embeddings_and_output_weights=not getattr(hf_cfg, "tie_word_embeddings", False),
+        num_experts=getattr(hf_cfg, "num_local_experts", None),
+        moe_router_topk=getattr(hf_cfg, "num_experts_per_tok", 2),
+        moe_router_pre_softmax=False,
     )
 
 
