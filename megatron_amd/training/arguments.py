@@ -158,6 +158,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--log-memory", action="store_true")
     g.add_argument("--log-straggler", action="store_true")
     g.add_argument("--straggler-report-interval", type=int, default=10)
+    g.add_argument("--straggler-ctrlr-port", type=int, default=None,
+                   help="TCP port for runtime straggler-detection toggling (curl host:port)")
     g.add_argument("--use-wandb", action="store_true")
     g.add_argument("--wandb-project", type=str, default=None)
     g.add_argument("--check-weight-hash-across-dp-replicas-interval", type=int, default=None)

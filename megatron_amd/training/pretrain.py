@@ -151,7 +151,8 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             enabled=True, fault_type=args.fault_injection_type,
             at_iteration=args.fault_injection_iteration,
             ranks=list(args.fault_injection_ranks)), rank=args.rank)
-    straggler = StragglerDetector(enabled=args.log_straggler)
+    straggler = StragglerDetector(enabled=args.log_straggler,
+                                  control_port=getattr(args, 'straggler_ctrlr_port', None))
     energy = EnergyMonitor() if args.log_energy else None
     metrics = MetricsLogger(args.tensorboard_dir, rank=args.rank,
                             use_wandb=args.use_wandb, wandb_project=args.wandb_project)

@@ -69,13 +69,55 @@ class StragglerReport:
 class StragglerDetector:
     """Times a per-rank section each step; reports min/max ranks on demand."""
 
-    def __init__(self, enabled: bool = False, device_index: int = 0):
+    def __init__(self, enabled: bool = False, device_index: int = 0,
+                 control_port: Optional[int] = None):
         self.enabled = enabled
         self._use_events = torch.cuda.is_available()
         self._start_evt = None
         self._t0 = 0.0
         self._elapsed: List[float] = []
         self.telemetry = AmdGpuTelemetry(device_index)
+        self._server = None
+        if control_port is not None:
+            self._start_control_server(control_port)
+
+    def _start_control_server(self, port: int):
+        """Runtime on/off toggle via `curl host:<port>` (reference
+        StragglerDetector's port-toggled control, core/utils.py:1445)."""
+        import socket
+        import threading
+
+        srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        srv.bind(("127.0.0.1", port))
+        srv.listen(4)
+        self._server = srv
+        self.control_port = srv.getsockname()[1]
+
+        def serve():
+            while True:
+                try:
+                    conn, _ = srv.accept()
+                except OSError:
+                    return  # closed
+                with conn:
+                    try:
+                        conn.recv(1024)
+                        self.enabled = not self.enabled
+                        state = b"on" if self.enabled else b"off"
+                        conn.sendall(b"HTTP/1.0 200 OK\r\n\r\nstraggler detection " + state + b"\r\n")
+                    except OSError:
+                        pass
+
+        threading.Thread(target=serve, daemon=True).start()
+
+    def close(self):
+        if self._server is not None:
+            try:
+                self._server.close()
+            except OSError:
+                pass
+            self._server = None
 
     def start(self):
         if not self.enabled:

@@ -190,3 +190,20 @@ def test_symmetric_allreduce_fallback_cpu():
     t = torch.randn(8)
     out = ar.all_reduce(t.clone())
     assert torch.equal(out, t)
+
+
+def test_straggler_control_port_toggle():
+    import socket
+
+    from megatron_amd.utils.straggler import StragglerDetector
+
+    det = StragglerDetector(enabled=False, control_port=0)  # 0 -> ephemeral
+    try:
+        for expect in (True, False, True):
+            with socket.create_connection(("127.0.0.1", det.control_port), timeout=5) as c:
+                c.sendall(b"GET / HTTP/1.0\r\n\r\n")
+                resp = c.recv(256)
+            assert b"straggler detection" in resp
+            assert det.enabled is expect
+    finally:
+        det.close()
