@@ -1,0 +1,124 @@
+"""Combined 1F1B: layer-granular co-scheduling of forward and backward.
+
+Capability analog of reference megatron/core/pipeline_parallel/
+combined_1f1b.py:35 (+ pipeline_parallel/utils.py `AbstractSchedulePlan` /
+`ScheduleNode`, models/common/model_chunk_schedule_plan.py): in the 1F1B
+steady state, the forward of microbatch i and the backward of microbatch
+i-1 are both resident; instead of running them whole-model-at-a-time, each
+is decomposed into per-layer ScheduleNodes and executed interleaved —
+fwd layer k of microbatch i, then bwd layer N-1-k of microbatch i-1, ...
+
+Why: with MoE layers, a forward node's EP all-to-all (already launched on
+the dedicated comm stream by MoELayer) overlaps the *backward* node's
+GEMMs instead of idling the GPU, and vice versa — the reference needs
+NVSHMEM+DeepEP for this; on one MI355X node the a2a is a single-hop xGMI
+transfer on its own stream, so plain interleaving exposes the overlap.
+
+Mechanics: a ScheduleNode detaches its input (requires_grad), runs the
+layer, and keeps (input, output); node.backward(gout) runs
+torch.autograd.backward on just that segment and hands back input.grad —
+manual chain rule over the layer boundary, numerically identical to one
+whole backward (same ops, same order within a layer).
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional
+
+import torch
+
+
+class ScheduleNode:
+    """One schedulable segment (a transformer layer or embedding/head)."""
+
+    def __init__(self, fn: Callable, name: str = ""):
+        self.fn = fn
+        self.name = name
+        self._inp: Optional[torch.Tensor] = None
+        self._out: Optional[torch.Tensor] = None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self._inp = x.detach().requires_grad_(x.requires_grad or x.is_floating_point())
+        self._out = self.fn(self._inp)
+        return self._out
+
+    def backward(self, grad_output: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+        if self._out is None:
+            raise RuntimeError(f"backward before forward on node {self.name!r}")
+        if grad_output is None:  # loss node: scalar output
+            torch.autograd.backward(self._out)
+        else:
+            torch.autograd.backward(self._out, grad_output)
+        g = self._inp.grad
+        self._inp = self._out = None  # free activations
+        return g
+
+
+class ModelChunkSchedulePlan:
+    """Decomposes a model chunk into nodes: [pre] + one per layer + [post].
+
+    `pre_fn` / `post_fn` wrap embedding / final-norm+head+loss; layer nodes
+    call decoder layers with the step-invariant kwargs closed over."""
+
+    def __init__(self, layers: List[Callable], pre_fn: Optional[Callable] = None,
+                 post_fn: Optional[Callable] = None):
+        self.nodes: List[ScheduleNode] = []
+        if pre_fn is not None:
+            self.nodes.append(ScheduleNode(pre_fn, "pre"))
+        for i, layer in enumerate(layers):
+            self.nodes.append(ScheduleNode(layer, f"layer{i}"))
+        if post_fn is not None:
+            self.nodes.append(ScheduleNode(post_fn, "post"))
+
+    @classmethod
+    def from_gpt(cls, model, rotary_freqs=None, attention_mask=None,
+                 loss_fn: Optional[Callable] = None):
+        """Plan for a full (pp=1) GPT chunk: embedding node + layer nodes +
+        final-norm/head/loss node.  Under pipeline parallelism the pre/post
+        nodes exist only on the first/last stage (reference
+        model_chunk_schedule_plan.py)."""
+        core = model.module if hasattr(model, "module") else model
+
+        layer_fns = [
+            (lambda h, _l=l: _l(h, rotary_freqs=rotary_freqs, attention_mask=attention_mask))
+            for l in core.decoder.layers
+        ]
+
+        def pre(tokens):
+            return core.embedding(tokens)
+
+        def post(h):
+            h = core.decoder.final_layernorm(h)
+            logits, _ = core.output_layer(h)
+            return loss_fn(logits) if loss_fn is not None else logits
+
+        return cls(layer_fns, pre_fn=pre if core.pre_process else None,
+                   post_fn=post if core.post_process else None)
+
+    def forward_node(self, k: int, x: torch.Tensor) -> torch.Tensor:
+        return self.nodes[k].forward(x)
+
+    def backward_node(self, k: int, g: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+        return self.nodes[k].backward(g)
+
+    def __len__(self):
+        return len(self.nodes)
+
+
+def combined_1f1b_step(fwd_plan: ModelChunkSchedulePlan, fwd_input: torch.Tensor,
+                       bwd_plan: Optional[ModelChunkSchedulePlan] = None,
+                       bwd_grad: Optional[torch.Tensor] = None):
+    """One steady-state slot: forward `fwd_plan` on `fwd_input` while
+    draining `bwd_plan`'s backward, node-interleaved.  Returns
+    (fwd_output, bwd_input_grad)."""
+    n_f = len(fwd_plan)
+    n_b = len(bwd_plan) if bwd_plan is not None else 0
+    steps = max(n_f, n_b)
+    x = fwd_input
+    g = bwd_grad
+    for k in range(steps):
+        if k < n_f:
+            x = fwd_plan.forward_node(k, x)
+        if bwd_plan is not None and k < n_b:
+            g = bwd_plan.backward_node(n_b - 1 - k, g)
+    return (x if n_f else None), (g if bwd_plan is not None else None)
