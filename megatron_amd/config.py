@@ -254,4 +254,6 @@ class DDPConfig:
     bucket_size: Optional[int] = 40_000_000  # elements per bucket target
     average_in_collective: bool = True
     check_for_nan_in_grad: bool = False
+    # allocate grad/param buffers in an RCCL-registered MemPool (N2)
+    use_rccl_registered_buffers: bool = False
     align_param_gather: bool = False

@@ -97,11 +97,24 @@ class ParamAndGradBuffer:
             self.buckets.append(_Bucket(cur_params, cur_start, offset, len(self.buckets)))
         self.total_elements = offset
 
-        # --- allocate storage ---
-        self.grad_data = torch.zeros(self.total_elements, dtype=grad_dtype, device=device)
-        self.param_data = None
-        if ddp_config.use_distributed_optimizer:
-            self.param_data = torch.empty(self.total_elements, dtype=param_dtype, device=device)
+        # --- allocate storage (optionally inside an RCCL-registered pool,
+        # reference nccl_allocator.py N2: zero-copy xGMI transports) ---
+        import contextlib
+
+        alloc_ctx = contextlib.nullcontext()
+        self._comm_pool = None
+        if getattr(ddp_config, "use_rccl_registered_buffers", False):
+            from megatron_amd.distributed.rccl_allocator import RcclRegisteredPool
+            from megatron_amd.parallel import grid as G
+
+            self._comm_pool = RcclRegisteredPool(
+                G.get_grid().group("dp_cp") if G.grid_initialized() else None)
+            alloc_ctx = self._comm_pool.use()
+        with alloc_ctx:
+            self.grad_data = torch.zeros(self.total_elements, dtype=grad_dtype, device=device)
+            self.param_data = None
+            if ddp_config.use_distributed_optimizer:
+                self.param_data = torch.empty(self.total_elements, dtype=param_dtype, device=device)
 
         for p in params:
             start, end, bidx = self.param_index[p]

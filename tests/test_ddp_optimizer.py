@@ -151,3 +151,30 @@ def test_grad_accumulation_equals_big_batch():
     g2 = {n: p.main_grad.clone() for n, p in chunks2[0].module.named_parameters()}
     for n in g1:
         assert_close(g1[n], g2[n], rtol=1e-4, atol=1e-5, msg=n)
+
+
+def test_rccl_registered_buffers_fallback_cpu():
+    """The registered-pool path must degrade to plain allocation without
+    CUDA (and DDP training still works)."""
+    from megatron_amd.distributed.ddp import DistributedDataParallel
+    from megatron_amd.distributed.rccl_allocator import RcclRegisteredPool, registered_comm_pool
+
+    from tests.utils import init_single
+
+    init_single()
+    pool = RcclRegisteredPool()
+    assert not pool.active  # no CUDA here
+    with registered_comm_pool() as p:
+        t = torch.zeros(16)
+    assert t.sum() == 0
+
+    cfg = _tiny_cfg()
+    ddp_cfg = DDPConfig(use_rccl_registered_buffers=True)
+    torch.manual_seed(0)
+    model = DistributedDataParallel(cfg, ddp_cfg, _provider(cfg))
+    x = torch.randint(0, 96, (2, 8))
+    out = model(x, position_ids=None, attention_mask=None)
+    out.float().square().mean().backward()
+    model.finish_grad_sync()
+    for p_ in model.parameters():
+        assert p_.main_grad is not None
