@@ -62,6 +62,12 @@ class MoELayer(nn.Module):
                 dispatched, tokens_per_expert = self.dispatcher.dispatch(tokens, probs, top_idx)
             shared_out = self.shared_expert(full.reshape(-1, h))
             torch.cuda.current_stream().wait_stream(self._comm_stream)
+            # comm-stream allocations consumed on the compute stream: pin
+            # their lifetime to it so the caching allocator doesn't reuse
+            # the blocks while the expert GEMMs still read them
+            dispatched.record_stream(torch.cuda.current_stream())
+            if tokens_per_expert.is_cuda:
+                tokens_per_expert.record_stream(torch.cuda.current_stream())
         else:
             dispatched, tokens_per_expert = self.dispatcher.dispatch(tokens, probs, top_idx)
             if self.shared_expert is not None:

@@ -122,6 +122,14 @@ def capture_block_hip_graphs(block, sample_hidden: torch.Tensor,
 
     gc.collect()
     torch.cuda.synchronize()
+    from megatron_amd.moe.moe_layer import MoELayer
+
+    for layer in block.layers:
+        for m in layer.modules():
+            if isinstance(m, MoELayer):
+                raise RuntimeError(
+                    "hipGraph capture requires static shapes; MoE token routing "
+                    "is data-dependent — capture dense blocks only")
     graphed = []
     for layer in block.layers:
         graphs = _LayerGraphs(layer, sample_hidden, rotary_freqs, attention_mask,
