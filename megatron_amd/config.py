@@ -72,6 +72,9 @@ class TransformerConfig(ParallelConfig):
     position_embedding_type: str = "rope"  # 'rope' | 'learned' | 'none'
     rotary_base: float = 500000.0  # llama-3 default
     rotary_percent: float = 1.0
+    # None or {'type': 'llama3'|'linear', 'factor': ..., 'low_freq_factor': ...,
+    # 'high_freq_factor': ..., 'original_max_position_embeddings': ...}
+    rope_scaling: Optional[dict] = None
     attention_dropout: float = 0.0
     hidden_dropout: float = 0.0
     # sliding-window attention: None or window size (causal look-back)

@@ -98,6 +98,7 @@ class GPTModel(nn.Module):
                 base=self.config.rotary_base,
                 device=device,
                 rotary_percent=self.config.rotary_percent,
+                rope_scaling=getattr(self.config, "rope_scaling", None),
             )
         return self._rope_cache[key]
 

@@ -26,7 +26,7 @@ class MoEStatsTracker:
         for i, r in enumerate(self.routers):
             self._counts[i] += r.local_tokens_per_expert
             for k, v in r.aux_losses.items():
-                self._aux_sums[i][k] = self._aux_sums[i].get(k, 0.0) + float(v)
+                self._aux_sums[i][k] = self._aux_sums[i].get(k, 0.0) + float(v.detach() if hasattr(v, 'detach') else v)
         self._steps += 1
 
     def report(self, reset: bool = True) -> Dict[str, dict]:

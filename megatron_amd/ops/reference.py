@@ -44,6 +44,24 @@ def squared_relu(x: torch.Tensor) -> torch.Tensor:
     return torch.pow(F.relu(x.float()), 2).to(x.dtype)
 
 
+def apply_llama3_rope_scaling(inv_freq: torch.Tensor, factor: float = 8.0,
+                              low_freq_factor: float = 1.0,
+                              high_freq_factor: float = 4.0,
+                              original_max_position: int = 8192) -> torch.Tensor:
+    """Llama-3.1 long-context rope scaling (reference rope_utils /
+    HF _compute_llama3_parameters): high-frequency channels untouched,
+    low-frequency channels divided by `factor`, smooth ramp between."""
+    low_wavelen = original_max_position / low_freq_factor
+    high_wavelen = original_max_position / high_freq_factor
+    wavelen = 2 * math.pi / inv_freq
+    scaled = torch.where(wavelen > low_wavelen, inv_freq / factor, inv_freq)
+    smooth = (original_max_position / wavelen - low_freq_factor) / (
+        high_freq_factor - low_freq_factor)
+    smoothed = (1 - smooth) / factor * inv_freq + smooth * inv_freq
+    mid = (wavelen <= low_wavelen) & (wavelen >= high_wavelen)
+    return torch.where(mid, smoothed, scaled)
+
+
 def rope_freqs(
     seq_len: int,
     dim: int,
@@ -51,11 +69,28 @@ def rope_freqs(
     device=None,
     dtype=torch.float32,
     rotary_percent: float = 1.0,
+    rope_scaling: dict = None,
 ) -> torch.Tensor:
     """Precomputed rotation angles [seq, dim_rot/2] (host-side table — G13/App-B:
-    on-device trig turns RoPE memory-bound into VALU-bound)."""
+    on-device trig turns RoPE memory-bound into VALU-bound).  rope_scaling:
+    None or {'type': 'llama3', 'factor': ..., ...} (llama-3.1 long context)."""
     rot_dim = int(dim * rotary_percent)
     inv_freq = 1.0 / (base ** (torch.arange(0, rot_dim, 2, device=device, dtype=torch.float32) / rot_dim))
+    if rope_scaling:
+        kind = rope_scaling.get("type", "llama3")
+        if kind == "llama3":
+            inv_freq = apply_llama3_rope_scaling(
+                inv_freq,
+                factor=rope_scaling.get("factor", 8.0),
+                low_freq_factor=rope_scaling.get("low_freq_factor", 1.0),
+                high_freq_factor=rope_scaling.get("high_freq_factor", 4.0),
+                original_max_position=rope_scaling.get(
+                    "original_max_position_embeddings", 8192),
+            )
+        elif kind == "linear":
+            inv_freq = inv_freq / rope_scaling.get("factor", 1.0)
+        else:
+            raise ValueError(f"unknown rope_scaling type {kind!r}")
     t = torch.arange(seq_len, device=device, dtype=torch.float32)
     freqs = torch.outer(t, inv_freq)  # [seq, rot_dim/2]
     return freqs.to(dtype)

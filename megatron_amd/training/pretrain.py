@@ -151,6 +151,11 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             enabled=True, fault_type=args.fault_injection_type,
             at_iteration=args.fault_injection_iteration,
             ranks=list(args.fault_injection_ranks)), rank=args.rank)
+    moe_stats = None
+    if args.num_experts:
+        from megatron_amd.moe.moe_logging import MoEStatsTracker
+
+        moe_stats = MoEStatsTracker(chunks[0] if isinstance(chunks, list) else chunks)
     straggler = StragglerDetector(enabled=args.log_straggler,
                                   control_port=getattr(args, 'straggler_ctrlr_port', None))
     energy = EnergyMonitor() if args.log_energy else None
@@ -236,6 +241,8 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             sys.exit(result["exit_code"])
         iteration += 1
         scheduler.step()
+        if moe_stats is not None:
+            moe_stats.collect()
         if (args.check_weight_hash_across_dp_replicas_interval
                 and iteration % args.check_weight_hash_across_dp_replicas_interval == 0):
             from megatron_amd.distributed.checks import check_param_hashes_across_dp_replicas
@@ -254,6 +261,10 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                    f"lr {optimizer.get_lr():.3e} | iter time {t*1000:.1f}ms | tokens/s {tokens_per_s:.0f}")
             if args.log_throughput:
                 msg += f" | TFLOP/s/GPU {tflops:.1f} | MFU {tflops/MI355X_BF16_DENSE_PEAK_TFLOPS*100:.1f}%"
+            if moe_stats is not None:
+                rep = moe_stats.report()
+                worst = max((v["max_violation"] for v in rep.values()), default=0.0)
+                msg += f" | moe max-violation {worst:.2f}"
             if result.get("grad_norm") is not None:
                 msg += f" | grad norm {result['grad_norm']:.3f}"
             if args.log_memory and torch.cuda.is_available():

@@ -64,3 +64,37 @@ def test_attention_sliding_window():
     scores = scores.masked_fill(~mask, float("-inf"))
     manual = torch.einsum("bhst,tbhd->sbhd", torch.softmax(scores, -1), v)
     assert_close(out, manual, rtol=1e-4, atol=1e-4)
+
+
+def test_llama3_rope_scaling_matches_hf_formula():
+    import math
+
+    import torch
+
+    from megatron_amd.ops.reference import apply_llama3_rope_scaling, rope_freqs
+
+    dim, base = 128, 500000.0
+    inv = 1.0 / (base ** (torch.arange(0, dim, 2, dtype=torch.float32) / dim))
+    factor, lo, hi, orig = 8.0, 1.0, 4.0, 8192
+    got = apply_llama3_rope_scaling(inv, factor, lo, hi, orig)
+    # HF transformers _compute_llama3_parameters reference computation
+    low_wl = orig / lo
+    high_wl = orig / hi
+    expect = []
+    for f in inv.tolist():
+        wl = 2 * math.pi / f
+        if wl < high_wl:
+            expect.append(f)
+        elif wl > low_wl:
+            expect.append(f / factor)
+        else:
+            smooth = (orig / wl - lo) / (hi - lo)
+            expect.append((1 - smooth) * f / factor + smooth * f)
+    assert torch.allclose(got, torch.tensor(expect), rtol=1e-6, atol=0)
+    # high-freq (small wavelength) channels unchanged; lowest-freq divided by 8
+    assert got[0] == inv[0]
+    assert abs(got[-1] - inv[-1] / factor) < 1e-9
+    # end-to-end: freqs table differs once scaling is on
+    f0 = rope_freqs(16, dim, base=base)
+    f1 = rope_freqs(16, dim, base=base, rope_scaling={"type": "llama3"})
+    assert not torch.allclose(f0, f1)

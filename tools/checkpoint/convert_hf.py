@@ -128,6 +128,7 @@ def config_from_hf(hf_cfg):
         rotary_base=getattr(hf_cfg, "rope_theta", 10000.0),
         layernorm_epsilon=hf_cfg.rms_norm_eps,
         untie_embeddings_and_output_weights=not getattr(hf_cfg, "tie_word_embeddings", False),
+        rope_scaling=(dict(getattr(hf_cfg, "rope_scaling")) if getattr(hf_cfg, "rope_scaling", None) else None),
         num_experts=getattr(hf_cfg, "num_local_experts", None),
         moe_router_topk=getattr(hf_cfg, "num_experts_per_tok", 2),
         moe_router_pre_softmax=False,
