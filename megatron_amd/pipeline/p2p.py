@@ -38,6 +38,37 @@ class P2PCommunicator:
     def _device(self):
         return torch.device("cuda", torch.cuda.current_device()) if torch.cuda.is_available() else torch.device("cpu")
 
+    def _exchange_shapes(self, tensor_send_next, tensor_send_prev,
+                         recv_next: bool, recv_prev: bool, dev):
+        """Variable-seq-len shape pre-exchange (reference
+        p2p_communication.py:186 _communicate_shapes): a tiny [3] int64
+        message per direction carries the payload shape so recv buffers can
+        be sized exactly."""
+        to_t = lambda t: torch.tensor(t.shape, dtype=torch.int64, device=dev)
+        recv_prev_shape = torch.empty(3, dtype=torch.int64, device=dev) if recv_prev else None
+        recv_next_shape = torch.empty(3, dtype=torch.int64, device=dev) if recv_next else None
+        ops = []
+        even = self.grid.pp_rank % 2 == 0
+
+        def sends():
+            if tensor_send_prev is not None:
+                ops.append(dist.P2POp(dist.isend, to_t(tensor_send_prev), self.prev_rank, group=self.group))
+            if tensor_send_next is not None:
+                ops.append(dist.P2POp(dist.isend, to_t(tensor_send_next), self.next_rank, group=self.group))
+
+        def recvs():
+            if recv_prev_shape is not None:
+                ops.append(dist.P2POp(dist.irecv, recv_prev_shape, self.prev_rank, group=self.group))
+            if recv_next_shape is not None:
+                ops.append(dist.P2POp(dist.irecv, recv_next_shape, self.next_rank, group=self.group))
+
+        (sends() or recvs()) if even else (recvs() or sends())
+        if ops:
+            for r in dist.batch_isend_irecv(ops):
+                r.wait()
+        return (tuple(recv_prev_shape.tolist()) if recv_prev_shape is not None else None,
+                tuple(recv_next_shape.tolist()) if recv_next_shape is not None else None)
+
     def communicate(
         self,
         tensor_send_next: Optional[torch.Tensor] = None,
@@ -47,8 +78,16 @@ class P2PCommunicator:
     ):
         """Fused bidirectional exchange; returns (tensor_recv_prev, tensor_recv_next)."""
         dev = self._device()
-        tensor_recv_prev = self._buf(dev) if recv_prev else None
-        tensor_recv_next = self._buf(dev) if recv_next else None
+        if self.config.variable_seq_lengths:
+            prev_shape, next_shape = self._exchange_shapes(
+                tensor_send_next, tensor_send_prev, recv_next, recv_prev, dev)
+            tensor_recv_prev = (torch.empty(prev_shape, dtype=self.dtype, device=dev,
+                                            requires_grad=True) if recv_prev else None)
+            tensor_recv_next = (torch.empty(next_shape, dtype=self.dtype, device=dev,
+                                            requires_grad=True) if recv_next else None)
+        else:
+            tensor_recv_prev = self._buf(dev) if recv_prev else None
+            tensor_recv_next = self._buf(dev) if recv_next else None
         ops = []
         # rank-parity ordering keeps pairwise matching deterministic even if
         # the backend serializes (reference p2p_communication.py:67)

@@ -139,3 +139,37 @@ def test_schedule_table():
 
     t = get_schedule_table(4, 2, 2)
     assert t == [(0, 0), (1, 0), (0, 1), (1, 1), (2, 0), (3, 0), (2, 1), (3, 1)]
+
+
+def _varlen_p2p_case(rank, world):
+    from megatron_amd.pipeline.p2p import P2PCommunicator
+    from megatron_amd.config import TransformerConfig
+
+    G.initialize_model_parallel(pipeline_parallel_size=2)
+    cfg = TransformerConfig(num_layers=2, hidden_size=8, num_attention_heads=2,
+                            vocab_size=32, pipeline_parallel_size=2,
+                            variable_seq_lengths=True, pipeline_dtype=torch.float32,
+                            gradient_accumulation_fusion=False)
+    comm = P2PCommunicator(cfg, seq_length=16, micro_batch_size=2)
+    # stage 0 sends two different-length activations; stage 1 receives with
+    # exact shapes via the pre-exchange
+    for s in (5, 11):
+        if rank == 0:
+            payload = torch.full((s, 2, 8), float(s))
+            comm.send_forward(payload, is_last_stage=False)
+        else:
+            t = comm.recv_forward(is_first_stage=False)
+            assert t.shape == (s, 2, 8), t.shape
+            assert torch.all(t == float(s))
+    # and backward direction
+    for s in (3, 7):
+        if rank == 1:
+            comm.send_backward(torch.full((s, 2, 8), -float(s)), is_first_stage=False)
+        else:
+            g = comm.recv_backward(is_last_stage=False)
+            assert g.shape == (s, 2, 8)
+            assert torch.all(g == -float(s))
+
+
+def test_varlen_p2p_shape_exchange():
+    spawn_dist(_varlen_p2p_case, 2)
