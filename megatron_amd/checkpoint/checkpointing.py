@@ -62,6 +62,54 @@ def save_checkpoint(root: str, models: List, optimizer, iteration: int,
     return writer
 
 
+def save_non_persistent_checkpoint(local_root: str, models: List, optimizer,
+                                   iteration: int, scheduler=None,
+                                   extra: Optional[dict] = None, retain: int = 1):
+    """Fast in-job restart checkpoints on node-local storage (SSD/ramdisk)
+    — reference checkpointing.py:1528 non-persistent local ckpts.  Same
+    format as regular checkpoints; only the newest `retain` iterations are
+    kept (older ones deleted after a successful save)."""
+    import re
+    import shutil
+
+    save_checkpoint(local_root, models, optimizer, iteration, scheduler, extra)
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    if rank == 0:
+        kept = sorted(
+            (int(m.group(1)) for d in os.listdir(local_root)
+             for m in [re.fullmatch(r"iter_(\d+)", d)] if m),
+            reverse=True)
+        for it in kept[retain:]:
+            shutil.rmtree(os.path.join(local_root, f"iter_{it:07d}"), ignore_errors=True)
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def latest_checkpoint_iteration(root: str) -> Optional[int]:
+    try:
+        with open(os.path.join(root, TRACKER)) as f:
+            return int(f.read().strip())
+    except (OSError, ValueError):
+        return None
+
+
+def resolve_resume_source(persistent_root: Optional[str],
+                          non_persistent_root: Optional[str]):
+    """(root, iteration) of the newest available checkpoint across the
+    persistent and node-local non-persistent trees (reference: prefer the
+    non-persistent copy when it is newer)."""
+    candidates = []
+    for root in (persistent_root, non_persistent_root):
+        if root:
+            it = latest_checkpoint_iteration(root)
+            if it is not None:
+                candidates.append((it, root))
+    if not candidates:
+        return None, None
+    it, root = max(candidates)
+    return root, it
+
+
 def load_checkpoint(root: str, models: List, optimizer, scheduler=None,
                     iteration: Optional[int] = None, load_rng: bool = True) -> int:
     if iteration is None:

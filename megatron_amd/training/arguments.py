@@ -139,6 +139,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--save", type=str, default=None)
     g.add_argument("--load", type=str, default=None)
     g.add_argument("--save-interval", type=int, default=None)
+    g.add_argument("--non-persistent-ckpt-dir", type=str, default=None,
+                   help="node-local (SSD/ramdisk) fast-restart checkpoint tree")
+    g.add_argument("--non-persistent-save-interval", type=int, default=None)
     g.add_argument("--async-save", action="store_true")
     g.add_argument("--no-load-rng", action="store_true")
 

@@ -175,7 +175,10 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
     sig_handler = DistSignalHandler() if args.exit_signal_handler else None
     if args.load:
         try:
-            iteration = load_checkpoint(args.load, chunks, optimizer, scheduler,
+            from megatron_amd.checkpoint.checkpointing import resolve_resume_source
+
+            _root, _it = resolve_resume_source(args.load, args.non_persistent_ckpt_dir)
+            iteration = load_checkpoint(_root or args.load, chunks, optimizer, scheduler,
                                         load_rng=not args.no_load_rng)
             _print_rank0(f"loaded checkpoint at iteration {iteration}")
         except FileNotFoundError:
@@ -243,6 +246,12 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
         scheduler.step()
         if moe_stats is not None:
             moe_stats.collect()
+        if (args.non_persistent_save_interval and args.non_persistent_ckpt_dir
+                and iteration % args.non_persistent_save_interval == 0):
+            from megatron_amd.checkpoint.checkpointing import save_non_persistent_checkpoint
+
+            save_non_persistent_checkpoint(args.non_persistent_ckpt_dir, chunks,
+                                           optimizer, iteration, scheduler)
         if (args.check_weight_hash_across_dp_replicas_interval
                 and iteration % args.check_weight_hash_across_dp_replicas_interval == 0):
             from megatron_amd.distributed.checks import check_param_hashes_across_dp_replicas

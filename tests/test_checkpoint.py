@@ -157,3 +157,34 @@ def test_reshard_tp1_to_tp2(tmp_path):
     spawn_dist(_tp2_load_case, 2, str(tmp_path), out_file)
     tp2_logits = torch.load(out_file)
     assert_close(ref_logits.detach(), tp2_logits, rtol=1e-4, atol=1e-5)
+
+
+def test_non_persistent_checkpoint_retention_and_resume(tmp_path):
+    from megatron_amd.checkpoint.checkpointing import (
+        load_checkpoint,
+        resolve_resume_source,
+        save_checkpoint,
+        save_non_persistent_checkpoint,
+    )
+
+    init_single(seed=11)
+    cfg = _cfg()
+    chunks, opt = setup_model_and_optimizer(_provider, cfg,
+                                            OptimizerConfig(lr=1e-3), DDPConfig())
+    persistent = str(tmp_path / "persist")
+    local = str(tmp_path / "local")
+    save_checkpoint(persistent, chunks, opt, iteration=2)
+    for it in (3, 4, 5):
+        save_non_persistent_checkpoint(local, chunks, opt, iteration=it, retain=2)
+    import os
+
+    kept = sorted(d for d in os.listdir(local) if d.startswith("iter_"))
+    assert kept == ["iter_0000004", "iter_0000005"]  # retention pruned 3
+    # resume picks the newer non-persistent copy
+    root, it = resolve_resume_source(persistent, local)
+    assert (root, it) == (local, 5)
+    it_loaded = load_checkpoint(root, chunks, opt, iteration=it)
+    assert it_loaded == 5
+    # without the local tree, the persistent one is used
+    root2, it2 = resolve_resume_source(persistent, str(tmp_path / "missing"))
+    assert (root2, it2) == (persistent, 2)
