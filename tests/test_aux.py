@@ -207,3 +207,37 @@ def test_straggler_control_port_toggle():
             assert det.enabled is expect
     finally:
         det.close()
+
+
+def test_debug_dumper(tmp_path):
+    import json
+
+    import torch
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.utils.debug_dumps import DebugDumper
+    from tests.utils import init_single
+
+    init_single()
+    torch.manual_seed(0)
+    model = GPTModel(TransformerConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+        vocab_size=64, ffn_hidden_size=48, gradient_accumulation_fusion=False))
+    path = str(tmp_path / "dumps.jsonl")
+    d = DebugDumper(path)
+    n = d.watch(model, ["decoder.layers.*.self_attention", "decoder.layers.*.mlp"], grads=True)
+    assert n == 4
+    toks = torch.randint(0, 64, (2, 8))
+    out = model(toks, position_ids=None, attention_mask=None)
+    out.float().square().mean().backward()
+    d.next_step()
+    acts = [r for r in d.records if r["kind"] == "activation"]
+    dgrads = [r for r in d.records if r["kind"] == "dgrad"]
+    assert len(acts) == 4 and len(dgrads) == 4
+    assert all(r["n_nonfinite"] == 0 for r in d.records)
+    assert d.nonfinite_modules() == []
+    d.close()
+    lines = [json.loads(l) for l in open(path)]
+    assert len(lines) == len(d.records)
+    assert {"norm", "absmax", "module"} <= set(lines[0])
