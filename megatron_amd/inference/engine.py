@@ -257,6 +257,12 @@ class DynamicInferenceEngine:
         req.block_table.extend(self.context.allocator.allocate(need))
         return True
 
+    def _hit_stop_string(self, req: _Request) -> bool:
+        if not req.params.stop_strings or self.tokenizer is None:
+            return False
+        text = self.tokenizer.detokenize(req.result.output_tokens)
+        return any(text.endswith(ss) for ss in req.params.stop_strings)
+
     def _finish(self, req: _Request):
         self.context.allocator.free(req.block_table)
         req.block_table = []
@@ -345,6 +351,8 @@ class DynamicInferenceEngine:
             req.result.output_tokens.append(tok)
             if req.params.stop_on_eod and tok == self.eod:
                 req.result.output_tokens.pop()
+                self._finish(req)
+            elif self._hit_stop_string(req):
                 self._finish(req)
             elif len(req.result.output_tokens) >= req.params.max_tokens:
                 self._finish(req)

@@ -19,6 +19,9 @@ class SamplingParams:
     stop_on_eod: bool = True
     return_log_probs: bool = False
     seed: Optional[int] = None
+    # generation stops (token excluded) when the detokenized output ends
+    # with any of these (requires a tokenizer on the engine)
+    stop_strings: tuple = ()
 
 
 def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0) -> torch.Tensor:

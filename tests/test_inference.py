@@ -242,3 +242,28 @@ def test_disaggregated_prefill_decode_matches_single(tiny_model):
         assert a.output_tokens == b.output_tokens
     # prefill pool fully drained back
     assert prefill_eng.context.allocator.num_free == 16
+
+
+def test_stop_strings(tiny_model):
+    class _Tok:
+        eod = VOCAB - 1
+
+        def tokenize(self, s):
+            return [ord(c) % VOCAB for c in s]
+
+        def detokenize(self, toks):
+            return "".join(chr(97 + (t % 26)) for t in toks)
+
+    tok = _Tok()
+    eng = DynamicInferenceEngine(tiny_model, tokenizer=tok, num_blocks=16, block_size=8)
+    free = eng.generate([[1, 2, 3]], SamplingParams(max_tokens=12, greedy=True, stop_on_eod=False))[0]
+    assert len(free.output_tokens) == 12
+    # pick a stop string that actually occurs in the free run's text
+    text = tok.detokenize(free.output_tokens)
+    stop = text[3:5]
+    eng2 = DynamicInferenceEngine(tiny_model, tokenizer=tok, num_blocks=16, block_size=8)
+    stopped = eng2.generate([[1, 2, 3]], SamplingParams(
+        max_tokens=12, greedy=True, stop_on_eod=False, stop_strings=(stop,)))[0]
+    out_text = tok.detokenize(stopped.output_tokens)
+    assert out_text.endswith(stop)
+    assert len(stopped.output_tokens) == 5
