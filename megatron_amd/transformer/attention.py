@@ -102,6 +102,17 @@ class SelfAttention(nn.Module):
                 core_out = ulysses_attention(q, k, v, scale=self.softmax_scale)
             else:
                 core_out = ring_attention(q, k, v, scale=self.softmax_scale)
+        elif self.config.recompute_granularity == "selective" and self.training:
+            # selective recompute: checkpoint only the core-attention region
+            # (reference transformer_config 'selective' — the s^2-shaped
+            # softmax state is recomputed in backward, everything else saved)
+            from megatron_amd.parallel.random import checkpoint as rng_checkpoint
+
+            def _core(q_, k_, v_):
+                return ops.flash_attention(q_, k_, v_, causal=self.config.causal_attention,
+                                           scale=self.softmax_scale, window=self.window)
+
+            core_out = rng_checkpoint(_core, False, q, k, v)
         else:
             core_out = ops.flash_attention(q, k, v, causal=self.config.causal_attention,
                                            scale=self.softmax_scale, window=self.window)

@@ -164,3 +164,25 @@ def _tp_vs_single_case(rank, world):
 
 def test_tp2_matches_dense_recompute():
     spawn_dist(_tp_vs_single_case, 2)
+
+
+def test_recompute_selective_matches_no_recompute():
+    init_single(seed=99)
+    torch.manual_seed(11)
+    cfg = _tiny_cfg()
+    m = GPTModel(cfg)
+    m.train()
+    ids = torch.randint(0, 96, (2, 16))
+    labels = torch.randint(0, 96, (2, 16))
+    loss1 = m(ids, labels=labels).mean()
+    loss1.backward()
+    g1 = {n: p.grad.clone() for n, p in m.named_parameters()}
+    m.zero_grad()
+    sel = cfg.replace(recompute_granularity="selective")
+    for layer in m.decoder.layers:
+        layer.self_attention.config = sel
+    loss2 = m(ids, labels=labels).mean()
+    loss2.backward()
+    assert_close(loss1, loss2, rtol=1e-6, atol=1e-6)
+    for n, p in m.named_parameters():
+        assert_close(p.grad, g1[n], rtol=1e-5, atol=1e-6, msg=n)
