@@ -71,7 +71,16 @@ class _ParallelLinearFn(torch.autograd.Function):
         ctx.async_grad_allreduce = async_grad_allreduce
         ctx.sequence_parallel = sequence_parallel
         if sequence_parallel:
-            total_input = _gather_along_first_dim(inp, G.get_tensor_model_parallel_group())
+            # fwd-only gather (backward re-gathers): land it in the reused
+            # global scratch (reference GlobalMemoryBuffer, utils.py:693)
+            from megatron_amd.parallel.memory_buffer import get_global_memory_buffer
+
+            group = G.get_tensor_model_parallel_group()
+            world = dist.get_world_size(group) if dist.is_initialized() else 1
+            shape = (inp.shape[0] * world,) + tuple(inp.shape[1:])
+            buf = get_global_memory_buffer().get_tensor(shape, inp.dtype, "sp-ag-fwd",
+                                                        device=inp.device)
+            total_input = _gather_along_first_dim(inp, group, output_buffer=buf)
         else:
             total_input = inp
         ctx.save_for_backward(inp, weight)

@@ -145,3 +145,19 @@ def test_vocab_parallel_cross_entropy_tp1():
     loss = vocab_parallel_cross_entropy(logits, target)
     ref = torch.nn.functional.cross_entropy(logits.detach().reshape(-1, 64), target.reshape(-1), reduction="none").view(6, 2)
     assert_close(loss, ref, rtol=1e-5, atol=1e-5)
+
+
+def test_global_memory_buffer_reuse():
+    import torch
+
+    from megatron_amd.parallel.memory_buffer import GlobalMemoryBuffer
+
+    gmb = GlobalMemoryBuffer()
+    a = gmb.get_tensor((4, 8), torch.float32, "x")
+    ptr = a.data_ptr()
+    b = gmb.get_tensor((2, 8), torch.float32, "x")  # smaller: same storage
+    assert b.data_ptr() == ptr
+    c = gmb.get_tensor((8, 8), torch.float32, "x")  # bigger: regrown
+    assert c.numel() == 64
+    d = gmb.get_tensor((4, 8), torch.float64, "x")  # dtype keyed separately
+    assert d.dtype == torch.float64
