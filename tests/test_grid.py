@@ -3,6 +3,7 @@
 import pytest
 
 from megatron_amd.parallel.grid import ParallelGrid, compose, decompose, orthogonal_rank_groups
+from megatron_amd.parallel import grid as G
 
 
 def test_decompose_compose_roundtrip():
@@ -62,3 +63,55 @@ def test_expert_groups():
     assert len(g.ranks("expert_dp")) == 2
     seen = set(g.ranks("ep")) | set(g.ranks("expert_dp")) | set(g.ranks("etp"))
     assert 0 in seen
+
+
+def test_hyper_comm_grid_coords():
+    from megatron_amd.parallel.hyper_grid import HyperCommGrid
+
+    g = HyperCommGrid(["tp", "dp", "pp"], [2, 3, 2])
+    assert g.total == 12
+    # round trip
+    for r in range(12):
+        assert g.rank_at(**g.coords_of(r)) == r
+    # tp fastest-varying
+    assert g.coords_of(1) == {"tp": 1, "dp": 0, "pp": 0}
+    assert g.coords_of(2) == {"tp": 0, "dp": 1, "pp": 0}
+    assert g.coords_of(6) == {"tp": 0, "dp": 0, "pp": 1}
+    # group spans
+    assert g.ranks_for(["tp"], 3) == [2, 3]
+    assert g.ranks_for(["dp"], 1) == [1, 3, 5]
+    assert g.ranks_for(["tp", "dp"], 7) == [6, 7, 8, 9, 10, 11]
+    groups = g.all_groups_for(["dp"])
+    assert len(groups) == 4  # 2 tp x 2 pp
+    flat = sorted(r for grp in groups for r in grp)
+    assert flat == list(range(12))
+
+
+def test_hyper_comm_grid_offset_subgrid():
+    from megatron_amd.parallel.hyper_grid import HyperCommGrid
+
+    # a vision-encoder sub-grid occupying ranks 8..11 next to an LLM grid
+    g = HyperCommGrid(["tp", "dp"], [2, 2], rank_offset=8)
+    assert g.ranks_for(["tp"], 10) == [10, 11]
+    assert g.coords_of(9) == {"tp": 1, "dp": 0}
+
+
+def _hyper_groups_case(rank, world):
+    import torch
+    import torch.distributed as dist
+
+    from megatron_amd.parallel.hyper_grid import HyperCommGrid
+
+    G.initialize_model_parallel()
+    g = HyperCommGrid(["a", "b"], [2, 1])
+    grp = g.group_for(["a"])
+    assert dist.get_world_size(grp) == 2
+    t = torch.tensor([float(rank + 1)])
+    dist.all_reduce(t, group=grp)
+    assert float(t) == 3.0
+
+
+def test_hyper_comm_grid_groups_gloo():
+    from tests.utils import spawn_dist
+
+    spawn_dist(_hyper_groups_case, 2)
