@@ -96,6 +96,32 @@ def rope_freqs(
     return freqs.to(dtype)
 
 
+def mrope_freqs(position_ids: torch.Tensor, dim: int, base: float = 10000.0,
+                mrope_section=(16, 24, 24), rotary_percent: float = 1.0) -> torch.Tensor:
+    """Multimodal rotary freqs (Qwen2-VL style mrope; reference
+    models/common/embeddings multimodal rotary :266 analog).
+
+    position_ids: [3, s] — (temporal, height, width) position per token
+    (text tokens carry the same value in all three rows).  The rotation
+    channels are split into `mrope_section` groups; group j's angles are
+    indexed by position row j.  Returns [s, dim_rot/2] angles usable by
+    rope_apply."""
+    assert position_ids.dim() == 2 and position_ids.shape[0] == len(mrope_section)
+    rot_dim = int(dim * rotary_percent)
+    n_half = rot_dim // 2
+    assert sum(mrope_section) == n_half, (mrope_section, n_half)
+    inv_freq = 1.0 / (base ** (torch.arange(0, rot_dim, 2, device=position_ids.device,
+                                            dtype=torch.float32) / rot_dim))
+    s = position_ids.shape[1]
+    out = torch.empty(s, n_half, dtype=torch.float32, device=position_ids.device)
+    start = 0
+    for row, width in enumerate(mrope_section):
+        t = position_ids[row].float()
+        out[:, start:start + width] = torch.outer(t, inv_freq[start:start + width])
+        start += width
+    return out
+
+
 def rope_apply(x: torch.Tensor, freqs: torch.Tensor) -> torch.Tensor:
     """Rotate-half RoPE (GPT-NeoX / llama convention).
 

@@ -98,3 +98,33 @@ def test_llama3_rope_scaling_matches_hf_formula():
     f0 = rope_freqs(16, dim, base=base)
     f1 = rope_freqs(16, dim, base=base, rope_scaling={"type": "llama3"})
     assert not torch.allclose(f0, f1)
+
+
+def test_mrope_reduces_to_rope_for_text():
+    import torch
+
+    from megatron_amd.ops.reference import mrope_freqs, rope_freqs
+
+    s, dim = 12, 128
+    # text-only: all three position rows identical -> plain RoPE
+    pos = torch.arange(s).unsqueeze(0).repeat(3, 1)
+    got = mrope_freqs(pos, dim, base=10000.0, mrope_section=(16, 24, 24))
+    expect = rope_freqs(s, dim, base=10000.0)
+    assert torch.allclose(got, expect, atol=1e-6)
+
+
+def test_mrope_sections_use_their_position_rows():
+    import torch
+
+    from megatron_amd.ops.reference import mrope_freqs
+
+    dim = 64  # n_half = 32
+    sec = (8, 12, 12)
+    pos = torch.stack([torch.zeros(4), torch.ones(4) * 2, torch.ones(4) * 5]).long()
+    f = mrope_freqs(pos, dim, mrope_section=sec)
+    # temporal section: position 0 -> zero angles
+    assert torch.all(f[:, :8] == 0)
+    # height/width sections scale with their own positions
+    assert torch.all(f[:, 8:20] > 0) and torch.all(f[:, 20:] > 0)
+    f2 = mrope_freqs(pos * 2, dim, mrope_section=sec)
+    assert torch.allclose(f2[:, 8:], f[:, 8:] * 2)
