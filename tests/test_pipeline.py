@@ -173,3 +173,52 @@ def _varlen_p2p_case(rank, world):
 
 def test_varlen_p2p_shape_exchange():
     spawn_dist(_varlen_p2p_case, 2)
+
+
+def _tp2pp2_case(rank, world, ckpt_dir):
+    from megatron_amd.checkpoint.checkpointing import load_checkpoint
+
+    G.initialize_model_parallel(tensor_parallel_size=2, pipeline_parallel_size=2)
+    model_parallel_seed(1234)
+    cfg = _cfg(pp=2)
+    cfg = cfg.replace(tensor_parallel_size=2)
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    load_checkpoint(ckpt_dir, chunks, opt, load_rng=False)
+    batches = _gen_batches(4)
+    its = [iter(batches) for _ in chunks]
+    r = train_step(forward_step, its, chunks, opt, cfg, 4, SEQ, 2)
+    grid = G.get_grid()
+    if grid.is_pipeline_last_stage(ignore_virtual=True) and grid.rank_in("tp") == 0:
+        with open(os.environ["PP_TEST_OUT"], "w") as f:
+            json.dump([r["lm_loss"]], f)
+
+
+def test_tp2_pp2_composition_matches_single(tmp_path, monkeypatch):
+    """4-rank composition (TP=2 x PP=2) resharded from a single-process
+    checkpoint reproduces the single-process first-step loss."""
+    from megatron_amd.checkpoint.checkpointing import save_checkpoint
+
+    out = tmp_path / "tp2pp2.json"
+    ckpt = str(tmp_path / "ckpt")
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+
+    init_single()
+    cfg = _cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    for c in chunks:
+        _fill_deterministic(c)
+    for o in opt.chained_optimizers:
+        if hasattr(o, "reload_model_params"):
+            o.reload_model_params()
+    save_checkpoint(ckpt, chunks, opt, iteration=0)
+    batches = _gen_batches(4)
+    r = train_step(forward_step, [iter(batches)], chunks, opt, cfg, 4, SEQ, 2)
+    ref_loss = r["lm_loss"]
+
+    spawn_dist(_tp2pp2_case, 4, ckpt)
+    got = json.load(open(out))[0]
+    assert abs(got - ref_loss) < 5e-4, (got, ref_loss)
