@@ -96,5 +96,6 @@ def test_tiny_mamba_loss_curve_matches_golden():
 
         return MambaModel(config, pre_process=pre_process, post_process=post_process)
 
-    losses = _run_config(["--hybrid-override-pattern", "M*"], provider=provider)
+    losses = _run_config(["--hybrid-override-pattern", "M*", "--mamba-num-groups", "2",
+                          "--mamba-head-dim", "32"], provider=provider)
     _check_or_record("tiny_mamba.json", losses)
