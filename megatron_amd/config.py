@@ -79,6 +79,10 @@ class TransformerConfig(ParallelConfig):
     hidden_dropout: float = 0.0
     # sliding-window attention: None or window size (causal look-back)
     window_size: Optional[int] = None
+    # every k-th layer (0-indexed: layers with number % k == k-1) uses FULL
+    # attention, others the sliding window (reference window_attn_skip_freq;
+    # gemma-2/llama-4 style interleaved local/global). None -> all windowed.
+    window_attn_skip_freq: Optional[int] = None
     qk_layernorm: bool = False
     softmax_scale: Optional[float] = None
 

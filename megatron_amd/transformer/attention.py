@@ -47,6 +47,9 @@ class SelfAttention(nn.Module):
         )
         self.softmax_scale = config.softmax_scale or (1.0 / (self.kv_channels**0.5))
         self.window = config.window_size
+        skip = getattr(config, "window_attn_skip_freq", None)
+        if self.window is not None and skip and (layer_number % skip == skip - 1):
+            self.window = None  # this layer is a global-attention layer
         if config.qk_layernorm:
             self.q_layernorm = nn.Parameter(torch.ones(self.kv_channels, dtype=config.params_dtype))
             self.k_layernorm = nn.Parameter(torch.ones(self.kv_channels, dtype=config.params_dtype))

@@ -186,3 +186,27 @@ def test_recompute_selective_matches_no_recompute():
     assert_close(loss1, loss2, rtol=1e-6, atol=1e-6)
     for n, p in m.named_parameters():
         assert_close(p.grad, g1[n], rtol=1e-5, atol=1e-6, msg=n)
+
+
+def test_window_attn_skip_freq_pattern():
+    init_single()
+    cfg = _tiny_cfg(window_size=8, window_attn_skip_freq=2)
+    m = GPTModel(cfg)
+    windows = [l.self_attention.window for l in m.decoder.layers]
+    assert windows == [8, None]  # layer 1 (number%2==1) is global
+    ids = torch.randint(0, 96, (2, 16))
+    out = m(ids)
+    assert out.shape == (16, 2, 96)
+    # all-windowed differs from interleaved for long-enough context
+    cfg2 = _tiny_cfg(window_size=8)
+    torch.manual_seed(0)
+    m2 = GPTModel(cfg2)
+    torch.manual_seed(0)
+    init_single()
+    cfg3 = _tiny_cfg(window_size=8, window_attn_skip_freq=2)
+    m3 = GPTModel(cfg3)
+    m3.load_state_dict(m2.state_dict())
+    with torch.no_grad():
+        a = m2(ids)
+        b = m3(ids)
+    assert not torch.allclose(a, b, atol=1e-5)
