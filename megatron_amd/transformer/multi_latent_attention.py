@@ -69,9 +69,69 @@ class MLASelfAttention(nn.Module):
                                                    device=device)
         return self._freq_cache[key]
 
+    # -- absorbed-matmul inference path (reference :1231 FusedMLASelfAttention)
+    #
+    # The KV cache stores the LATENT per token: [normed c_kv (r) | roped
+    # k_rope (rope)] — (r + rope) elements instead of nh*(dqk + dv).  At
+    # attend time W_uk is absorbed into the query (q̃ = q_nope @ W_uk, per
+    # head) so scores are taken directly against the latent, and W_uv is
+    # applied after the probability-weighted latent sum.  Algebraically
+    # identical to the training path: q_nope·(W_uk c) == (q_nope W_uk)·c and
+    # Σ p (W_uv c) == W_uv (Σ p c).
+
+    def _latent_cache(self, ctx, device, dtype):
+        store = getattr(ctx, "mla_latent_cache", None)
+        if store is None:
+            store = {}
+            ctx.mla_latent_cache = store
+        if self.layer_number not in store:
+            width = self.config.kv_lora_rank + self.rope
+            store[self.layer_number] = torch.zeros(ctx.max_seq, ctx.max_batch, width,
+                                                   dtype=dtype, device=device)
+        return store[self.layer_number]
+
+    def _uk_uv(self):
+        w = self.linear_kv_up.weight  # [nh*(nope+dv), r]
+        w = w.view(self.nh, self.nope + self.dv, -1)
+        return w[:, : self.nope], w[:, self.nope:]  # [nh, nope, r], [nh, dv, r]
+
+    def _absorbed_attend(self, ctx, q_nope, q_rope, c_kv, k_rope):
+        """q_nope [s,b,nh,nope], q_rope [s,b,nh,rope], c_kv (normed)
+        [s,b,r], k_rope (roped) [s,b,rope] -> core out [s,b,nh,dv]."""
+        s, b = q_nope.shape[0], q_nope.shape[1]
+        r = self.config.kv_lora_rank
+        W_uk, W_uv = self._uk_uv()
+        q_abs = torch.einsum("sbhn,hnr->sbhr", q_nope.float(), W_uk.float())
+        lat_q = torch.cat([q_abs, q_rope.float()], dim=-1)  # [s,b,nh,r+rope]
+
+        cache = self._latent_cache(ctx, q_nope.device, torch.float32)
+        lens = ctx.context_lens[:b]
+        new_latent = torch.cat([c_kv.float(), k_rope.float()], dim=-1)  # [s,b,r+rope]
+        if s > 1:  # prefill (uniform offset)
+            off = int(lens[0])
+            cache[off:off + s, :b] = new_latent
+            L = off + s
+            pos_q = off + torch.arange(s, device=q_nope.device)
+        else:  # ragged decode: scatter at each row's length
+            ar = torch.arange(b, device=q_nope.device)
+            cache[lens, ar] = new_latent[0]
+            L = int((lens + 1).max())
+            pos_q = lens.view(1, -1)  # [1, b]
+        kv_lat = cache[:L, :b]  # [L, b, r+rope]
+        logits = torch.einsum("sbhr,lbr->sbhl", lat_q, kv_lat) * self.softmax_scale
+        key_pos = torch.arange(L, device=q_nope.device)
+        if pos_q.dim() == 1:
+            mask = key_pos.view(1, 1, L) <= pos_q.view(s, 1, 1)  # [s,1,L]
+        else:
+            mask = key_pos.view(1, 1, L) <= pos_q.view(1, b, 1)  # [1,b,L]
+        logits = logits.masked_fill(~mask.unsqueeze(2), float("-inf"))
+        probs = torch.softmax(logits, dim=-1)
+        c_hat = torch.einsum("sbhl,lbr->sbhr", probs, kv_lat[..., :r])
+        out = torch.einsum("sbhr,hdr->sbhd", c_hat, W_uv.float())
+        return out.to(q_nope.dtype)
+
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None,
                 inference_context=None, packed_seq_params=None):
-        assert inference_context is None, "MLA KV-cache (absorbed) path: round 2"
         assert packed_seq_params is None, "MLA packed (THD) path: round 2"
         s, b = hidden_states.shape[0], hidden_states.shape[1]
         if self.linear_q_down is not None:
@@ -83,6 +143,24 @@ class MLASelfAttention(nn.Module):
 
         down = self.linear_kv_down(hidden_states)  # [s, b, lora + rope]
         c_kv, k_rope = torch.split(down, [self.config.kv_lora_rank, self.rope], dim=2)
+
+        if inference_context is not None:
+            table = self._rope_freqs(inference_context.max_seq, hidden_states.device)
+            pos = inference_context.rope_positions(s)  # [s] or [1, b]
+            fr = table[pos]
+            if fr.dim() == 2:  # [s, rope/2]
+                q_rope = ops.rope_apply(q_rope.contiguous(), fr)
+                k_rope = ops.rope_apply(k_rope.view(s, b, 1, self.rope), fr)
+            else:  # ragged decode: [1, b, rope/2]
+                q_rope = ref.rope_apply_per_row(q_rope.contiguous(), fr)
+                k_rope = ref.rope_apply_per_row(k_rope.view(s, b, 1, self.rope), fr)
+            # NOTE: c_kv is cached post-norm; k_rope cached post-rope
+            core = self._absorbed_attend(inference_context,
+                                         q_nope, q_rope,
+                                         self.kv_norm(c_kv), k_rope.view(s, b, self.rope))
+            out, _ = self.linear_proj(core.reshape(s, b, self.nh * self.dv))
+            return out
+
         kv, _ = self.linear_kv_up(self.kv_norm(c_kv))
         kv = kv.view(s, b, self.nh, self.nope + self.dv)
         k_nope, v = torch.split(kv, [self.nope, self.dv], dim=3)
