@@ -241,3 +241,42 @@ def test_debug_dumper(tmp_path):
     lines = [json.loads(l) for l in open(path)]
     assert len(lines) == len(d.records)
     assert {"norm", "absmax", "module"} <= set(lines[0])
+
+
+def test_evaluate_ppl_and_cloze():
+    import math
+
+    import torch
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from tests.utils import init_single
+    from tools.evaluate import evaluate_cloze, evaluate_perplexity
+
+    init_single()
+    torch.manual_seed(0)
+    model = GPTModel(TransformerConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+        vocab_size=64, ffn_hidden_size=48, gradient_accumulation_fusion=False)).eval()
+    g = torch.Generator().manual_seed(1)
+    batches = []
+    for _ in range(3):
+        t = torch.randint(0, 64, (2, 17), generator=g)
+        batches.append({"tokens": t[:, :-1], "labels": t[:, 1:]})
+    r = evaluate_perplexity(model, batches)
+    assert r["tokens"] == 3 * 2 * 16
+    # random-init model on random tokens: loss near ln(V)
+    assert abs(r["loss"] - math.log(64)) < 1.0
+    assert r["ppl"] > 1.0
+
+    # cloze: build samples where the target is the model's own greedy pick ->
+    # accuracy 1.0; then shift targets -> accuracy 0.0
+    samples = []
+    for _ in range(4):
+        ctx = torch.randint(0, 64, (9,), generator=g).tolist()
+        with torch.no_grad():
+            pred = int(model(torch.tensor([ctx]))[-1, 0].float().argmax())
+        samples.append(ctx + [pred])
+    assert evaluate_cloze(model, samples)["accuracy"] == 1.0
+    wrong = [s[:-1] + [(s[-1] + 1) % 64] for s in samples]
+    assert evaluate_cloze(model, wrong)["accuracy"] == 0.0
