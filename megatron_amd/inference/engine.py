@@ -275,7 +275,8 @@ class DynamicInferenceEngine:
         logits_row = logits_row.unsqueeze(0)
         if req.gen is not None:
             logits_row = logits_row.cpu()
-        tok = int(sample(logits_row, req.params, req.gen)[0])
+        tok = int(sample(logits_row, req.params, req.gen,
+                         prev_tokens=[req.result.output_tokens])[0])
         if req.params.return_log_probs:
             req.result.log_probs.append(float(log_prob_of(logits_row, torch.tensor([tok]))[0]))
         return tok
@@ -337,7 +338,8 @@ class DynamicInferenceEngine:
         batch_toks = None
         p0 = batch[0].params
         if all((r.params is p0) and r.gen is None for r in batch):
-            batch_toks = sample(logits, p0).tolist()
+            batch_toks = sample(logits, p0,
+                                prev_tokens=[r.result.output_tokens for r in batch]).tolist()
             if p0.return_log_probs:
                 lps = log_prob_of(logits, torch.as_tensor(batch_toks, device=logits.device)).tolist()
         still = []

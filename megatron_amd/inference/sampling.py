@@ -22,6 +22,9 @@ class SamplingParams:
     # generation stops (token excluded) when the detokenized output ends
     # with any of these (requires a tokenizer on the engine)
     stop_strings: tuple = ()
+    # multiplicative penalty on logits of already-generated tokens (>1
+    # discourages repeats; HF convention: divide positive / multiply negative)
+    repetition_penalty: float = 1.0
 
 
 def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0) -> torch.Tensor:
@@ -40,9 +43,29 @@ def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0) -> t
     return logits
 
 
+def apply_repetition_penalty(logits: torch.Tensor, prev_tokens,
+                             penalty: float) -> torch.Tensor:
+    """HF-convention repetition penalty: for each already-seen token id,
+    positive logits are divided by `penalty`, negative multiplied.
+    logits [b, V]; prev_tokens: per-row sequences of ids."""
+    if penalty == 1.0:
+        return logits
+    logits = logits.clone()
+    for i, toks in enumerate(prev_tokens):
+        if not len(toks):
+            continue
+        ids = torch.as_tensor(list(set(toks)), device=logits.device, dtype=torch.long)
+        row = logits[i, ids]
+        logits[i, ids] = torch.where(row > 0, row / penalty, row * penalty)
+    return logits
+
+
 def sample(logits: torch.Tensor, params: SamplingParams,
-           generator: Optional[torch.Generator] = None) -> torch.Tensor:
+           generator: Optional[torch.Generator] = None,
+           prev_tokens=None) -> torch.Tensor:
     """logits: [b, V] (full vocab, fp32) -> next token ids [b]."""
+    if params.repetition_penalty != 1.0 and prev_tokens is not None:
+        logits = apply_repetition_penalty(logits, prev_tokens, params.repetition_penalty)
     if params.greedy or params.temperature == 0.0:
         return logits.argmax(dim=-1)
     logits = logits / max(params.temperature, 1e-6)

@@ -267,3 +267,18 @@ def test_stop_strings(tiny_model):
     out_text = tok.detokenize(stopped.output_tokens)
     assert out_text.endswith(stop)
     assert len(stopped.output_tokens) == 5
+
+
+def test_repetition_penalty():
+    from megatron_amd.inference.sampling import apply_repetition_penalty
+
+    logits = torch.tensor([[2.0, -2.0, 1.0, 0.5]])
+    out = apply_repetition_penalty(logits, [[0, 1]], penalty=2.0)
+    assert float(out[0, 0]) == 1.0      # positive divided
+    assert float(out[0, 1]) == -4.0     # negative multiplied
+    assert float(out[0, 2]) == 1.0      # unseen untouched
+    # greedy with penalty breaks a repeat loop
+    params = SamplingParams(greedy=True, repetition_penalty=10.0)
+    rep = torch.tensor([[5.0, 4.9, 0.0]])
+    tok = sample(rep, params, prev_tokens=[[0]])
+    assert int(tok[0]) == 1
