@@ -148,6 +148,9 @@ class TransformerConfig(ParallelConfig):
     # ---- execution ----
     # overlap TP dgrad all-reduce with wgrad GEMM (reference layers.py:622)
     async_tensor_model_parallel_allreduce: bool = True
+    # SP forward AG as a chunked ring overlapped with the GEMM (TE
+    # userbuffers ub_overlap_ag analog, parallel/overlap.py)
+    tp_comm_overlap: bool = False
     gradient_accumulation_fusion: bool = True
     persist_layer_norm: bool = True
     deterministic_mode: bool = False

@@ -65,11 +65,27 @@ class _ParallelLinearFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, inp, weight, bias, grad_accum_fusion, async_grad_allreduce,
-                sequence_parallel, fp8_recipes=None):
+                sequence_parallel, fp8_recipes=None, tp_comm_overlap=False):
         ctx.use_bias = bias is not None
         ctx.grad_accum_fusion = grad_accum_fusion
         ctx.async_grad_allreduce = async_grad_allreduce
         ctx.sequence_parallel = sequence_parallel
+        if sequence_parallel and tp_comm_overlap and fp8_recipes is None:
+            # chunked ring AG+GEMM: each ring hop overlaps the previous
+            # chunk's GEMM (parallel/overlap.py; TE ub_overlap_ag analog).
+            # Backward is unchanged - it re-gathers with its own async AG.
+            from megatron_amd.parallel.overlap import ring_allgather_gemm
+
+            ctx.save_for_backward(inp, weight)
+            ctx.fp8 = False
+            group = G.get_tensor_model_parallel_group()
+            world = dist.get_world_size(group) if dist.is_initialized() else 1
+            x2 = inp.reshape(-1, inp.shape[-1])
+            y2 = ring_allgather_gemm(x2, weight, group=group)
+            output = y2.view(inp.shape[0] * world, *inp.shape[1:-1], weight.shape[0])
+            if bias is not None:
+                output = output + bias
+            return output
         if sequence_parallel:
             # fwd-only gather (backward re-gathers): land it in the reused
             # global scratch (reference GlobalMemoryBuffer, utils.py:693)
@@ -180,15 +196,16 @@ class _ParallelLinearFn(torch.autograd.Function):
         if allreduce_handle is not None:
             allreduce_handle.wait()
 
-        return grad_input, grad_weight, grad_bias, None, None, None, None
+        return grad_input, grad_weight, grad_bias, None, None, None, None, None
 
 
 def linear_with_grad_accumulation_and_async_allreduce(
     inp, weight, bias, grad_accum_fusion, async_grad_allreduce, sequence_parallel,
-    fp8_recipes=None,
+    fp8_recipes=None, tp_comm_overlap=False,
 ):
     return _ParallelLinearFn.apply(inp, weight, bias, grad_accum_fusion,
-                                   async_grad_allreduce, sequence_parallel, fp8_recipes)
+                                   async_grad_allreduce, sequence_parallel, fp8_recipes,
+                                   tp_comm_overlap)
 
 
 def _init_weight(weight: torch.Tensor, init_std: float, generator: Optional[torch.Generator] = None):
@@ -228,6 +245,7 @@ class ColumnParallelLinear(nn.Module):
             config.async_tensor_model_parallel_allreduce and tp > 1 and not self.sequence_parallel
         )
         self.grad_accum_fusion = config.gradient_accumulation_fusion
+        self.tp_comm_overlap = getattr(config, "tp_comm_overlap", False)
         self.fp8_recipes = None
         if getattr(config, "fp8", None):
             from megatron_amd.ops.fp8 import make_recipes
@@ -260,7 +278,8 @@ class ColumnParallelLinear(nn.Module):
             inp = x
         out = linear_with_grad_accumulation_and_async_allreduce(
             inp, self.weight, bias, self.grad_accum_fusion, self.async_tp_allreduce,
-            self.sequence_parallel, self.fp8_recipes if self.training else None
+            self.sequence_parallel, self.fp8_recipes if self.training else None,
+            self.tp_comm_overlap,
         )
         if self.gather_output:
             from megatron_amd.parallel.mappings import gather_from_tensor_model_parallel_region
@@ -298,6 +317,7 @@ class RowParallelLinear(nn.Module):
         self.skip_bias_add = skip_bias_add
         self.sequence_parallel = config.sequence_parallel and tp > 1
         self.grad_accum_fusion = config.gradient_accumulation_fusion
+        self.tp_comm_overlap = getattr(config, "tp_comm_overlap", False)
         self.fp8_recipes = None
         if getattr(config, "fp8", None):
             from megatron_amd.ops.fp8 import make_recipes

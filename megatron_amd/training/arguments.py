@@ -87,6 +87,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--expert-model-parallel-size", "--ep", type=int, default=1)
     g.add_argument("--expert-tensor-parallel-size", type=int, default=None)
     g.add_argument("--sequence-parallel", action="store_true")
+    g.add_argument("--tp-comm-overlap", action="store_true",
+                   help="SP forward AG as a chunked ring overlapped with the GEMM")
 
     g = p.add_argument_group("training")
     g.add_argument("--micro-batch-size", type=int, default=1)
@@ -268,6 +270,7 @@ def configs_from_args(args):
         expert_parallel_size=args.expert_model_parallel_size,
         expert_tensor_parallel_size=args.expert_tensor_parallel_size,
         sequence_parallel=args.sequence_parallel,
+        tp_comm_overlap=args.tp_comm_overlap,
         bf16=args.bf16,
         fp16=args.fp16,
         fp8=args.fp8_format,

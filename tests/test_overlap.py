@@ -40,3 +40,39 @@ def _tp2_case(rank, world):
 
 def test_tp2_overlap_equivalence():
     spawn_dist(_tp2_case, 2)
+
+
+def _tp2_linear_overlap_case(rank, world):
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.parallel.layers import ColumnParallelLinear
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    G.initialize_model_parallel(tensor_parallel_size=2)
+    model_parallel_seed(3)
+    base = dict(num_layers=1, hidden_size=16, num_attention_heads=2, vocab_size=32,
+                ffn_hidden_size=32, tensor_parallel_size=2, sequence_parallel=True,
+                gradient_accumulation_fusion=False)
+    cfg_plain = TransformerConfig(**base)
+    cfg_ovl = TransformerConfig(**base, tp_comm_overlap=True)
+
+    torch.manual_seed(7)
+    lin_a = ColumnParallelLinear(16, 24, config=cfg_plain, bias=False)
+    lin_b = ColumnParallelLinear(16, 24, config=cfg_ovl, bias=False)
+    with torch.no_grad():
+        lin_b.weight.copy_(lin_a.weight)
+
+    torch.manual_seed(11)  # same shard input on both paths
+    x = torch.randn(4, 2, 16)  # [s/tp, b, h]
+    xa = x.clone().requires_grad_(True)
+    xb = x.clone().requires_grad_(True)
+    ya, _ = lin_a(xa)
+    yb, _ = lin_b(xb)
+    assert torch.allclose(ya, yb, atol=1e-6), (ya - yb).abs().max()
+    ya.square().sum().backward()
+    yb.square().sum().backward()
+    assert torch.allclose(xa.grad, xb.grad, atol=1e-5)
+    assert torch.allclose(lin_a.weight.grad, lin_b.weight.grad, atol=1e-5)
+
+
+def test_tp2_column_linear_overlap_equivalence():
+    spawn_dist(_tp2_linear_overlap_case, 2)
