@@ -282,3 +282,21 @@ def test_repetition_penalty():
     rep = torch.tensor([[5.0, 4.9, 0.0]])
     tok = sample(rep, params, prev_tokens=[[0]])
     assert int(tok[0]) == 1
+
+
+def test_moe_model_serving_matches_oracle():
+    """MoE models serve through the engines (expert routing is per-token,
+    cache-independent): greedy engine output == no-cache oracle."""
+    G.destroy_model_parallel()
+    G.initialize_model_parallel(world_size=1, rank=0)
+    model_parallel_seed(77)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=128, vocab_size=VOCAB, num_experts=4, moe_router_topk=2,
+        moe_ffn_hidden_size=96, max_position_embeddings=256)
+    model = GPTModel(cfg).eval()
+    prompts = [[3, 7, 11, 2, 9], [5, 1]]
+    eng = DynamicInferenceEngine(model, num_blocks=16, block_size=8)
+    res = eng.generate(prompts, SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
+    for p, r in zip(prompts, res):
+        assert r.output_tokens == _oracle_greedy(model, p, 6)
