@@ -348,3 +348,50 @@ def test_capacity_dropping_layer():
     # backward flows
     out.sum().backward()
     assert layer.experts.weight1.grad is not None
+
+
+def test_router_trace_and_replay(tmp_path):
+    from megatron_amd.moe.router_replay import (
+        RouterReplayer,
+        RouterTraceRecorder,
+        routing_divergence,
+    )
+
+    init_single()
+    cfg = _cfg()
+    layer = MoELayer(cfg)
+    _fill(layer)
+    x1 = torch.randn(6, 2, cfg.hidden_size)
+    x2 = torch.randn(6, 2, cfg.hidden_size)
+
+    rec = RouterTraceRecorder(layer)
+    out_ref = []
+    out_ref.append(layer(x1).detach()); rec.step()
+    out_ref.append(layer(x2).detach()); rec.step()
+    rec.save(str(tmp_path / "trace.pt"))
+    rec.close()
+    trace = torch.load(str(tmp_path / "trace.pt"))
+    assert len(trace) == 2 and trace[0][0]["indices"].shape == (12, cfg.moe_router_topk)
+
+    # replay onto a DIFFERENT-router model: outputs use recorded routing
+    cfg2 = _cfg()
+    layer2 = MoELayer(cfg2)
+    _fill(layer2)
+    with torch.no_grad():
+        layer2.router.weight.add_(torch.randn_like(layer2.router.weight))  # diverged router
+    rep = RouterReplayer(layer2, trace)
+    out1 = layer2(x1).detach(); rep.step()
+    layer2(x2); rep.step()
+    rep.close()
+    # with identical experts and pinned routing+probs, outputs match the
+    # original layer's despite the perturbed router weights
+    assert torch.allclose(out1, out_ref[0], atol=1e-5)
+
+    # divergence metric
+    rec2 = RouterTraceRecorder(layer2)
+    layer2(x1); rec2.step()
+    rec2.close()
+    d = routing_divergence(trace, rec2.trace)
+    assert d["fraction_diverged"] > 0  # perturbed router routes differently
+    d0 = routing_divergence(trace, trace)
+    assert d0["fraction_diverged"] == 0 and d0["first_divergence"] is None
