@@ -161,3 +161,31 @@ def test_decoupled_lr_groups():
     assert ok
     assert not torch.equal(w_emb, model.embedding.weight)
     assert not torch.equal(w_main, model.decoder.layers[0].self_attention.linear_qkv.weight)
+
+
+def test_layer_wise_matches_fp32_without_clip():
+    from megatron_amd.optimizer.layer_wise import LayerWiseOptimizer
+
+    init_single()
+    model_a = _Tiny()
+    model_b = copy.deepcopy(model_a)
+    cfg = OptimizerConfig(lr=1e-3, weight_decay=0.01, clip_grad=0.0)
+    opt_a = FP32Optimizer(cfg, [model_a])
+    opt_b = LayerWiseOptimizer(cfg, [model_b])
+    torch.manual_seed(4)
+    for _ in range(4):
+        x, y = torch.randn(16, 16), torch.randn(16, 16)
+        # a: standard order
+        opt_a.zero_grad()
+        ((model_a(x) - y) ** 2).mean().backward()
+        opt_a.step()
+        # b: updates fire inside backward
+        opt_b.zero_grad()
+        ((model_b(x) - y) ** 2).mean().backward()
+        ok, norm, _ = opt_b.step()
+        assert ok and float(norm) > 0
+        # grads were freed by the in-backward updates
+        assert all(p.grad is None for p in model_b.parameters())
+    for (na, pa), (nb, pb) in zip(model_a.named_parameters(), model_b.named_parameters()):
+        assert_close(pa, pb, rtol=1e-5, atol=1e-6, msg=na)
+    opt_b.close()
