@@ -95,6 +95,71 @@ class MLASelfAttention(nn.Module):
         w = w.view(self.nh, self.nope + self.dv, -1)
         return w[:, : self.nope], w[:, self.nope:]  # [nh, nope, r], [nh, dv, r]
 
+    def _latent_pool(self, ctx, device):
+        """Paged latent pool mirroring the dynamic context's block structure
+        ([num_blocks+1, block_size, r+rope], slot-indexed by the same block
+        tables)."""
+        store = getattr(ctx, "mla_latent_pool", None)
+        if store is None:
+            store = {}
+            ctx.mla_latent_pool = store
+        if self.layer_number not in store:
+            width = self.config.kv_lora_rank + self.rope
+            n_slots = ctx.k_cache[0].shape[0]
+            store[self.layer_number] = torch.zeros(n_slots, ctx.block_size, width,
+                                                   dtype=torch.float32, device=device)
+        return store[self.layer_number]
+
+    def _absorbed_attend_paged(self, ctx, q_nope, q_rope, c_kv, k_rope):
+        """Dynamic (paged) twin of _absorbed_attend: latents live in a paged
+        pool addressed by the engine's block tables."""
+        s, b = q_nope.shape[0], q_nope.shape[1]
+        r = self.config.kv_lora_rank
+        bs = ctx.block_size
+        W_uk, W_uv = self._uk_uv()
+        q_abs = torch.einsum("sbhn,hnr->sbhr", q_nope.float(), W_uk.float())
+        lat_q = torch.cat([q_abs, q_rope.float()], dim=-1)
+        pool = self._latent_pool(ctx, q_nope.device).view(-1, r + self.rope)
+        new_latent = torch.cat([c_kv.float(), k_rope.float()], dim=-1)  # [s,b,r+rope]
+
+        if ctx._mode == "prefill":
+            assert b == 1
+            pos = torch.arange(ctx._prior_len, ctx._prior_len + s, device=q_nope.device)
+            pool[ctx._slot_index(pos, ctx._prefill_table)] = new_latent[:, 0]
+            L = ctx._prior_len + s
+            all_slots = ctx._slot_index(torch.arange(L, device=q_nope.device), ctx._prefill_table)
+            kv_lat = pool[all_slots].unsqueeze(1)  # [L, 1, r+rope]
+            logits = torch.einsum("sbhr,lbr->sbhl", lat_q, kv_lat) * self.softmax_scale
+            key_pos = torch.arange(L, device=q_nope.device)
+            mask = key_pos.view(1, 1, 1, L) <= pos.view(s, 1, 1, 1)
+            logits = logits.masked_fill(~mask, float("-inf"))
+            probs = torch.softmax(logits, dim=-1)
+            c_hat = torch.einsum("sbhl,lbr->sbhr", probs, kv_lat[..., :r])
+        else:  # ragged decode over block tables
+            lens = ctx._context_lens
+            rows = torch.arange(b, device=q_nope.device)
+            slots = (ctx._block_tables[rows, torch.div(lens, bs, rounding_mode="floor")] * bs
+                     + lens % bs)
+            pool[slots] = new_latent[0]
+            new_lens = lens + 1
+            if ctx._static:
+                tables = ctx._block_tables
+            else:
+                nb = int(torch.div(new_lens.max() + bs - 1, bs, rounding_mode="floor"))
+                tables = ctx._block_tables[:, :nb]
+            slot_grid = (tables.unsqueeze(-1) * bs +
+                         torch.arange(bs, device=q_nope.device).view(1, 1, bs)).reshape(b, -1)
+            kv_lat = pool[slot_grid]  # [b, L, r+rope]
+            L = kv_lat.shape[1]
+            logits = torch.einsum("bhr,blr->bhl", lat_q[0], kv_lat) * self.softmax_scale
+            key_pos = torch.arange(L, device=q_nope.device)
+            mask = key_pos.view(1, 1, L) < new_lens.view(b, 1, 1)
+            logits = logits.masked_fill(~mask, float("-inf"))
+            probs = torch.softmax(logits, dim=-1)
+            c_hat = torch.einsum("bhl,blr->bhr", probs, kv_lat[..., :r]).unsqueeze(0)
+        out = torch.einsum("sbhr,hdr->sbhd", c_hat, W_uv.float())
+        return out.to(q_nope.dtype)
+
     def _absorbed_attend(self, ctx, q_nope, q_rope, c_kv, k_rope):
         """q_nope [s,b,nh,nope], q_rope [s,b,nh,rope], c_kv (normed)
         [s,b,r], k_rope (roped) [s,b,rope] -> core out [s,b,nh,dv]."""
@@ -145,7 +210,8 @@ class MLASelfAttention(nn.Module):
         c_kv, k_rope = torch.split(down, [self.config.kv_lora_rank, self.rope], dim=2)
 
         if inference_context is not None:
-            table = self._rope_freqs(inference_context.max_seq, hidden_states.device)
+            max_pos = getattr(inference_context, "max_seq", self.config.max_position_embeddings)
+            table = self._rope_freqs(max_pos, hidden_states.device)
             pos = inference_context.rope_positions(s)  # [s] or [1, b]
             fr = table[pos]
             if fr.dim() == 2:  # [s, rope/2]
@@ -155,9 +221,11 @@ class MLASelfAttention(nn.Module):
                 q_rope = ref.rope_apply_per_row(q_rope.contiguous(), fr)
                 k_rope = ref.rope_apply_per_row(k_rope.view(s, b, 1, self.rope), fr)
             # NOTE: c_kv is cached post-norm; k_rope cached post-rope
-            core = self._absorbed_attend(inference_context,
-                                         q_nope, q_rope,
-                                         self.kv_norm(c_kv), k_rope.view(s, b, self.rope))
+            attend = (self._absorbed_attend_paged
+                      if hasattr(inference_context, "block_size")
+                      else self._absorbed_attend)
+            core = attend(inference_context, q_nope, q_rope,
+                          self.kv_norm(c_kv), k_rope.view(s, b, self.rope))
             out, _ = self.linear_proj(core.reshape(s, b, self.nh * self.dv))
             return out
 

@@ -108,3 +108,27 @@ def test_mla_absorbed_prefill_matches_training_forward():
         inf_logits = m(toks, inference_context=ctx)
     assert torch.allclose(train_logits, inf_logits, atol=1e-4), \
         (train_logits - inf_logits).abs().max()
+
+
+def test_mla_dynamic_engine_matches_oracle():
+    """MLA through the dynamic (paged latent pool) engine: continuous
+    batching + chunked prefill must equal no-cache greedy generation."""
+    from megatron_amd.inference import DynamicInferenceEngine, SamplingParams
+
+    init_single()
+    model_parallel_seed(31)
+    m = GPTModel(_cfg()).eval()
+    prompts = [[3, 7, 11, 2, 9], [5, 1], [8, 8, 4, 2, 1, 0, 9]]
+    params = SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False)
+    eng = DynamicInferenceEngine(m, num_blocks=16, block_size=4, max_prefill_tokens=4)
+    res = eng.generate(prompts, params)
+    for p, r in zip(prompts, res):
+        toks = list(p)
+        expect = []
+        for _ in range(6):
+            with torch.no_grad():
+                logits = m(torch.tensor([toks]))
+            tok = int(logits[-1, 0].float().argmax())
+            expect.append(tok)
+            toks.append(tok)
+        assert r.output_tokens == expect, (r.output_tokens, expect)
