@@ -25,8 +25,8 @@ class KVHostOffloader:
         self._ids = itertools.count()
 
     def swap_out(self, block_table: List[int]) -> int:
-        """Copy the blocks' K/V (all layers) to host and free them.
-        Returns a handle for swap_in."""
+        """Copy the blocks' K/V (all layers, plus any MLA latent pools) to
+        host and free them.  Returns a handle for swap_in."""
         ctx = self.context
         idx = torch.as_tensor(block_table, dtype=torch.long, device=ctx.device)
         k = torch.stack([ctx.k_cache[l][idx] for l in range(ctx.num_layers)])
@@ -34,8 +34,12 @@ class KVHostOffloader:
         k_host, v_host = k.cpu(), v.cpu()
         if torch.cuda.is_available() and k.is_cuda:
             k_host, v_host = k_host.pin_memory(), v_host.pin_memory()
+        latents = None
+        pools = getattr(ctx, "mla_latent_pool", None)
+        if pools:
+            latents = {layer: pool[idx].cpu() for layer, pool in pools.items()}
         handle = next(self._ids)
-        self._store[handle] = (k_host, v_host)
+        self._store[handle] = (k_host, v_host, latents)
         ctx.allocator.free(list(block_table))
         return handle
 
@@ -46,13 +50,18 @@ class KVHostOffloader:
         """Re-allocate device blocks and restore the saved K/V into them.
         Returns the new block table."""
         ctx = self.context
-        k_host, v_host = self._store.pop(handle)
+        k_host, v_host, latents = self._store.pop(handle)
         n = k_host.shape[1]
         blocks = ctx.allocator.allocate(n)
         idx = torch.as_tensor(blocks, dtype=torch.long, device=ctx.device)
         for l in range(ctx.num_layers):
             ctx.k_cache[l][idx] = k_host[l].to(ctx.device, non_blocking=True)
             ctx.v_cache[l][idx] = v_host[l].to(ctx.device, non_blocking=True)
+        if latents:
+            pools = getattr(ctx, "mla_latent_pool", {})
+            for layer, lat in latents.items():
+                if layer in pools:
+                    pools[layer][idx] = lat.to(ctx.device, non_blocking=True)
         if torch.cuda.is_available() and ctx.k_cache[0].is_cuda:
             torch.cuda.current_stream().synchronize()
         return blocks

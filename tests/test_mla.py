@@ -132,3 +132,22 @@ def test_mla_dynamic_engine_matches_oracle():
             expect.append(tok)
             toks.append(tok)
         assert r.output_tokens == expect, (r.output_tokens, expect)
+
+
+def test_mla_preemption_equivalence():
+    """MLA + tight pool: preemption must swap the latent pool too (greedy
+    outputs equal the big-pool run)."""
+    from megatron_amd.inference import DynamicInferenceEngine, SamplingParams
+
+    init_single()
+    model_parallel_seed(41)
+    m = GPTModel(_cfg()).eval()
+    prompts = [[3, 7, 11, 2, 9], [5, 1], [8, 8, 4, 2], [12, 13, 14]]
+    params = SamplingParams(max_tokens=8, greedy=True, stop_on_eod=False)
+    big = DynamicInferenceEngine(m, num_blocks=64, block_size=4)
+    expected = big.generate(prompts, params)
+    small = DynamicInferenceEngine(m, num_blocks=9, block_size=4)
+    got = small.generate(prompts, params)
+    for a, b in zip(expected, got):
+        assert a.output_tokens == b.output_tokens
+    assert small.offloader._ids.__reduce__()[1][0] > 0  # preemption occurred
