@@ -34,6 +34,7 @@ class KVPackage:
     k_host: torch.Tensor  # [L, n_blocks, bs, hkv, d]
     v_host: torch.Tensor
     block_size: int
+    latents: Optional[dict] = None  # MLA latent-pool blocks per layer
 
 
 class PrefillWorker:
@@ -66,9 +67,11 @@ class PrefillWorker:
         idx = torch.as_tensor(table, dtype=torch.long, device=ctx.device)
         k = torch.stack([ctx.k_cache[l][idx] for l in range(ctx.num_layers)]).cpu()
         v = torch.stack([ctx.v_cache[l][idx] for l in range(ctx.num_layers)]).cpu()
+        pools = getattr(ctx, "mla_latent_pool", None)
+        latents = {l: p[idx].cpu() for l, p in pools.items()} if pools else None
         ctx.allocator.free(table)
         return KVPackage(prompt=prompt, first_token=first, k_host=k, v_host=v,
-                         block_size=ctx.block_size)
+                         block_size=ctx.block_size, latents=latents)
 
 
 class DecodeWorker:
@@ -87,6 +90,17 @@ class DecodeWorker:
         for l in range(ctx.num_layers):
             ctx.k_cache[l][idx] = pkg.k_host[l].to(ctx.device)
             ctx.v_cache[l][idx] = pkg.v_host[l].to(ctx.device)
+        if pkg.latents:
+            pools = getattr(ctx, "mla_latent_pool", None)
+            if pools is None:
+                pools = {}
+                ctx.mla_latent_pool = pools
+            for l, lat in pkg.latents.items():
+                if l not in pools:  # seed the decode engine's pool lazily
+                    pools[l] = torch.zeros(ctx.k_cache[0].shape[0], ctx.block_size,
+                                           lat.shape[-1], dtype=torch.float32,
+                                           device=ctx.device)
+                pools[l][idx] = lat.to(ctx.device)
         # register as an in-flight request that already emitted first_token
         rid = next(eng._ids)
         req = _Request(rid, list(pkg.prompt), params,

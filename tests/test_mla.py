@@ -151,3 +151,21 @@ def test_mla_preemption_equivalence():
     for a, b in zip(expected, got):
         assert a.output_tokens == b.output_tokens
     assert small.offloader._ids.__reduce__()[1][0] > 0  # preemption occurred
+
+
+def test_mla_disaggregated_matches_single():
+    from megatron_amd.inference import DynamicInferenceEngine, SamplingParams
+    from megatron_amd.inference.disaggregation import disaggregated_generate
+
+    init_single()
+    model_parallel_seed(51)
+    m = GPTModel(_cfg()).eval()
+    prompts = [[3, 7, 11, 2, 9], [5, 1]]
+    params = SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False)
+    single = DynamicInferenceEngine(m, num_blocks=32, block_size=4)
+    expected = single.generate(prompts, params)
+    pre = DynamicInferenceEngine(m, num_blocks=16, block_size=4)
+    dec = DynamicInferenceEngine(m, num_blocks=32, block_size=4)
+    got = disaggregated_generate(pre, dec, prompts, params)
+    for a, b in zip(expected, got):
+        assert a.output_tokens == b.output_tokens
