@@ -49,6 +49,43 @@ def create_app(engine, tokenizer=None):
             ]
         }
 
+    @app.post("/api/stream")
+    async def stream(request: Request):
+        """SSE token streaming for a single prompt (dynamic engine only):
+        each generated token is emitted as a `data:` event as soon as its
+        decode step completes; the final event carries done+text."""
+        import json as _json
+
+        from fastapi.responses import StreamingResponse
+
+        req = await request.json()
+        prompt = req["prompt"]
+        params = SamplingParams(
+            max_tokens=int(req.get("max_tokens", 64)),
+            temperature=float(req.get("temperature", 1.0)),
+            top_k=int(req.get("top_k", 0)), top_p=float(req.get("top_p", 0.0)),
+            greedy=bool(req.get("greedy", False)), seed=req.get("seed"))
+
+        def gen():
+            with lock:
+                rid = engine.add_request(prompt, params)
+                sent = 0
+                while True:
+                    engine.step()
+                    if rid in engine.finished:
+                        r = engine.finished.pop(rid)
+                        for t in r.output_tokens[sent:]:
+                            yield f"data: {_json.dumps({'token': t})}\n\n"
+                        yield f"data: {_json.dumps({'done': True, 'text': r.text})}\n\n"
+                        return
+                    cur = next((q for q in engine.active if q.rid == rid), None)
+                    if cur is not None and len(cur.result.output_tokens) > sent:
+                        for t in cur.result.output_tokens[sent:]:
+                            yield f"data: {_json.dumps({'token': t})}\n\n"
+                        sent = len(cur.result.output_tokens)
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
     return app
 
 

@@ -300,3 +300,28 @@ def test_moe_model_serving_matches_oracle():
     res = eng.generate(prompts, SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
     for p, r in zip(prompts, res):
         assert r.output_tokens == _oracle_greedy(model, p, 6)
+
+
+def test_stream_endpoint(tiny_model):
+    import json as _json
+
+    from fastapi.testclient import TestClient
+
+    from megatron_amd.inference.server import create_app
+
+    eng = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    app = create_app(eng)
+    client = TestClient(app)
+    payload = {"prompt": [3, 7, 11], "max_tokens": 5, "greedy": True}
+    with client.stream("POST", "/api/stream", json=payload) as r:
+        assert r.status_code == 200
+        events = []
+        for line in r.iter_lines():
+            if line.startswith("data: "):
+                events.append(_json.loads(line[6:]))
+    toks = [e["token"] for e in events if "token" in e]
+    assert events[-1].get("done") is True
+    # streamed tokens equal a non-streamed greedy run
+    eng2 = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    expect = eng2.generate([[3, 7, 11]], SamplingParams(max_tokens=5, greedy=True))[0]
+    assert toks == expect.output_tokens
