@@ -395,3 +395,62 @@ def test_router_trace_and_replay(tmp_path):
     assert d["fraction_diverged"] > 0  # perturbed router routes differently
     d0 = routing_divergence(trace, trace)
     assert d0["fraction_diverged"] == 0 and d0["first_divergence"] is None
+
+
+def _tp2ep2_case(rank, world):
+    G.initialize_model_parallel(tensor_parallel_size=2, expert_parallel_size=2)
+    model_parallel_seed(1234)
+    cfg = _cfg(num_experts=4, ep=2, tensor_parallel_size=2)
+    layer = MoELayer(cfg)
+
+    def fullfill(shape, key):
+        g = torch.Generator().manual_seed(zlib.crc32(key.encode()) % (2**31))
+        return torch.randn(shape, generator=g) * 0.1
+
+    grid = G.get_grid()
+    tp_rank, ep_rank = grid.rank_in("etp"), grid.rank_in("ep")
+    ffn = cfg.moe_ffn_hidden_size  # 48
+    ffn_pp = ffn // 2
+    w1 = fullfill((4, 2 * ffn, 32), "w1f")   # [E, gate;up, h]
+    w2 = fullfill((4, 32, ffn), "w2f")       # [E, h, ffn]
+    with torch.no_grad():
+        layer.router.weight.copy_(fullfill((4, 32), "router"))
+        for i, e in enumerate(range(ep_rank * 2, ep_rank * 2 + 2)):
+            gate = w1[e, tp_rank * ffn_pp:(tp_rank + 1) * ffn_pp]
+            up = w1[e, ffn + tp_rank * ffn_pp: ffn + (tp_rank + 1) * ffn_pp]
+            layer.experts.weight1[i].copy_(torch.cat([gate, up], dim=0))
+            layer.experts.weight2[i].copy_(w2[e, :, tp_rank * ffn_pp:(tp_rank + 1) * ffn_pp])
+    torch.manual_seed(1)  # same batch on all ranks (dp replicas)
+    x = torch.randn(6, 2, 32)
+    out = layer(x)
+    if rank == 0:
+        torch.save(out.detach(), os.environ["MOE_TEST_OUT"])
+
+
+def test_tp2_ep2_matches_single(tmp_path, monkeypatch):
+    """World-4 composition TP=2 x EP=2 (ETP sharding of expert weights)
+    reproduces the single-process dense MoE layer output."""
+    out_path = tmp_path / "moe_tp_ep.pt"
+    monkeypatch.setenv("MOE_TEST_OUT", str(out_path))
+
+    init_single()
+    cfg = _cfg(num_experts=4)
+    layer = MoELayer(cfg)
+
+    def fullfill(shape, key):
+        g = torch.Generator().manual_seed(zlib.crc32(key.encode()) % (2**31))
+        return torch.randn(shape, generator=g) * 0.1
+
+    ffn = cfg.moe_ffn_hidden_size
+    w1 = fullfill((4, 2 * ffn, 32), "w1f")
+    with torch.no_grad():
+        layer.router.weight.copy_(fullfill((4, 32), "router"))
+        layer.experts.weight1.copy_(w1)
+        layer.experts.weight2.copy_(fullfill((4, 32, ffn), "w2f"))
+    torch.manual_seed(1)
+    x = torch.randn(6, 2, 32)
+    ref = layer(x)
+
+    spawn_dist(_tp2ep2_case, 4)
+    got = torch.load(out_path)
+    assert_close(got, ref.detach(), rtol=1e-4, atol=1e-5)
