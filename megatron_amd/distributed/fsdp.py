@@ -170,6 +170,13 @@ class FullyShardedDataParallel(nn.Module):
                  reshard_after_forward: bool = False):
         super().__init__()
         self.module = module
+        # expert-parallel params are already sharded over EP with their own
+        # (edp) replication group; flat-sharding them over the dense DP group
+        # would double-shard and mis-reduce — explicit unsupported combo
+        for n, p in module.named_parameters():
+            assert not getattr(p, "is_expert_parallel", False), (
+                f"FSDP over expert-parallel params ({n}) is unsupported; "
+                "use EP + the distributed optimizer instead")
         self.dp_group = dp_group if dp_group is not None else G.get_data_parallel_group()
         self.units: List[_FSDPUnit] = []
         self._param_to_unit: Dict[nn.Parameter, _FSDPUnit] = {}
