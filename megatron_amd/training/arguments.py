@@ -93,6 +93,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g = p.add_argument_group("training")
     g.add_argument("--micro-batch-size", type=int, default=1)
     g.add_argument("--global-batch-size", type=int, default=None)
+    g.add_argument("--rampup-batch-size", type=int, nargs=3, default=None,
+                   metavar=("START", "INCREMENT", "SAMPLES"),
+                   help="grow GBS from START by INCREMENT over SAMPLES consumed samples")
     g.add_argument("--seq-length", type=int, default=4096)
     g.add_argument("--train-iters", type=int, default=10)
     g.add_argument("--eval-interval", type=int, default=0)

@@ -185,6 +185,11 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             _print_rank0(f"no checkpoint found in {args.load}; starting fresh")
 
     n_chunks = len(chunks)
+    from megatron_amd.training.microbatches import MicrobatchCalculator
+
+    mb_calc = MicrobatchCalculator(args.global_batch_size, args.micro_batch_size,
+                                   args.data_parallel_size,
+                                   rampup=getattr(args, "rampup_batch_size", None))
     consumed = iteration * args.global_batch_size  # dataloader resume point
     data_iters = [RerunDataIterator(build_data_iterator(args, str(device), start_sample=consumed))
                   for _ in range(n_chunks)]
@@ -230,8 +235,10 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             prof.__enter__()
         timers("iteration").start()
         straggler.start()
+        cur_gbs, cur_nmb = mb_calc.get(consumed)
         result = train_step(forward_step, data_iters, chunks, optimizer, cfg,
-                            args.num_microbatches, args.seq_length, args.micro_batch_size)
+                            cur_nmb, args.seq_length, args.micro_batch_size)
+        consumed += cur_gbs
         straggler.stop()
         timers("iteration").stop()
         if fault_injector is not None:

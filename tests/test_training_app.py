@@ -85,3 +85,29 @@ def test_pretrain_muon_and_fault_injection(tmp_path):
         pretrain(model_provider, TINY + ["--train-iters", "5",
                                          "--fault-injection-type", "crash",
                                          "--fault-injection-iteration", "1"])
+
+
+def test_microbatch_rampup_calculator():
+    from megatron_amd.training.microbatches import MicrobatchCalculator
+
+    calc = MicrobatchCalculator(32, 2, 1, rampup=(8, 8, 96))
+    # 3 increments (8->16->24->32) over 96 samples: one every 32 samples
+    assert calc.get(0) == (8, 4)
+    assert calc.get(31) == (8, 4)
+    assert calc.get(32) == (16, 8)
+    assert calc.get(64) == (24, 12)
+    assert calc.get(96) == (32, 16)
+    assert calc.get(10_000) == (32, 16)
+    # constant mode
+    assert MicrobatchCalculator(32, 2, 4).get(999) == (32, 4)
+    # invalid: increment not divisible by mbs*dp
+    import pytest as _pytest
+
+    with _pytest.raises(AssertionError):
+        MicrobatchCalculator(32, 2, 2, rampup=(8, 6, 96))
+
+
+def test_pretrain_with_rampup():
+    it = pretrain(model_provider, TINY + ["--train-iters", "4",
+                                          "--rampup-batch-size", "2", "2", "8"])
+    assert it == 4
