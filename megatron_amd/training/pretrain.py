@@ -185,6 +185,19 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             _print_rank0(f"no checkpoint found in {args.load}; starting fresh")
 
     n_chunks = len(chunks)
+    # dump the resolved run configuration (reference core/config_logger.py)
+    if args.rank == 0 and args.save:
+        import dataclasses
+        import json as _json
+        import os as _os
+
+        _os.makedirs(args.save, exist_ok=True)
+        with open(_os.path.join(args.save, "run_config.json"), "w") as f:
+            _json.dump({"args": {k: str(v) for k, v in sorted(vars(args).items())},
+                        "transformer_config": {k: str(v) for k, v in
+                                               sorted(dataclasses.asdict(cfg).items())}},
+                       f, indent=1)
+
     from megatron_amd.training.microbatches import MicrobatchCalculator
 
     mb_calc = MicrobatchCalculator(args.global_batch_size, args.micro_batch_size,

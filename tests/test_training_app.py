@@ -111,3 +111,13 @@ def test_pretrain_with_rampup():
     it = pretrain(model_provider, TINY + ["--train-iters", "4",
                                           "--rampup-batch-size", "2", "2", "8"])
     assert it == 4
+
+
+def test_run_config_dumped(tmp_path):
+    ckpt = str(tmp_path / "ckpt")
+    pretrain(model_provider, TINY + ["--train-iters", "1", "--save", ckpt])
+    import json
+
+    cfgd = json.load(open(os.path.join(ckpt, "run_config.json")))
+    assert cfgd["args"]["num_layers"] == "2"
+    assert "hidden_size" in cfgd["transformer_config"]
