@@ -95,3 +95,39 @@ def test_mixtral_round_trip_and_logits():
         a = model(toks, position_ids=None, attention_mask=None)
         b = model2(toks, position_ids=None, attention_mask=None)
     assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_export_hf_dir_round_trip(tmp_path):
+    """Exported safetensors + config re-import to identical logits."""
+    import json
+
+    import torch
+    from safetensors.torch import load_file
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from tools.checkpoint.convert_hf import hf_to_mcore_state_dict
+    from tools.checkpoint.export_hf import export_hf_dir
+
+    from tests.utils import init_single
+
+    init_single()
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+        vocab_size=64, ffn_hidden_size=48, gradient_accumulation_fusion=False)
+    torch.manual_seed(1)
+    model = GPTModel(cfg)
+    out = str(tmp_path / "hf")
+    hf_cfg = export_hf_dir(model, cfg, out)
+    assert hf_cfg["architectures"] == ["LlamaForCausalLM"]
+    assert json.load(open(out + "/config.json"))["num_hidden_layers"] == 2
+
+    hf_sd = load_file(out + "/model.safetensors")
+    back = hf_to_mcore_state_dict(hf_sd, cfg)
+    model2 = GPTModel(cfg)
+    model2.load_state_dict(back, strict=False)
+    toks = torch.randint(0, 64, (2, 8))
+    with torch.no_grad():
+        a = model(toks, position_ids=None, attention_mask=None)
+        b = model2(toks, position_ids=None, attention_mask=None)
+    assert torch.allclose(a, b, atol=1e-6)
