@@ -65,6 +65,50 @@ def cp_rope_positions(seq_len_global: int, cp_rank: int, cp_size: int,
                       torch.arange(c1 * L, (c1 + 1) * L, device=device)])
 
 
+class _CPGatherSeq(torch.autograd.Function):
+    """All-gather CP sequence shards back into NATURAL order.
+
+    Forward: all-gather + chunk reorder (inverts slice_for_cp_rank's
+    load-balanced layout).  Backward: sum the full-sequence grads over the
+    CP group (every rank consumed the full sequence) and slice this rank's
+    shard back out — the gather/reduce-scatter adjoint pair, done as
+    all-reduce+slice so it also runs on gloo."""
+
+    @staticmethod
+    def forward(ctx, x, seq_dim, mode):
+        cp = G.get_context_parallel_world_size()
+        ctx.seq_dim, ctx.mode, ctx.cp = seq_dim, mode, cp
+        if cp == 1:
+            return x
+        group = G.get_grid().group("cp")
+        shards = [torch.empty_like(x) for _ in range(cp)]
+        dist.all_gather(shards, x.contiguous(), group=group)
+        if mode == "a2a":
+            return torch.cat(shards, dim=seq_dim)
+        L = x.size(seq_dim) // 2
+        slots = [None] * (2 * cp)
+        for r in range(cp):
+            c0, c1 = cp_chunk_ids(r, cp)
+            slots[c0] = shards[r].narrow(seq_dim, 0, L)
+            slots[c1] = shards[r].narrow(seq_dim, L, L)
+        return torch.cat(slots, dim=seq_dim)
+
+    @staticmethod
+    def backward(ctx, g):
+        if ctx.cp == 1:
+            return g, None, None
+        group = G.get_grid().group("cp")
+        g = g.contiguous()
+        dist.all_reduce(g, group=group)
+        r = G.get_context_parallel_rank()
+        return slice_for_cp_rank(g, r, ctx.cp, seq_dim=ctx.seq_dim, mode=ctx.mode), None, None
+
+
+def gather_cp_sequence(x: torch.Tensor, seq_dim: int = 0, mode: str = "p2p") -> torch.Tensor:
+    """Differentiable shard -> full-sequence gather (natural order)."""
+    return _CPGatherSeq.apply(x, seq_dim, mode)
+
+
 def get_batch_on_this_cp_rank(batch: dict, mode: str = "p2p") -> dict:
     cp = G.get_context_parallel_world_size()
     if cp == 1:

@@ -144,6 +144,16 @@ class MambaMixer(nn.Module):
 
     def forward(self, hidden_states: torch.Tensor, inference_state: Optional[dict] = None):
         """hidden_states: [s, b, h] (sbh, like the attention path)."""
+        # CP: the scan is sequential over the sequence, so CP shards are
+        # gathered to the full sequence, scanned, and re-sliced (reference
+        # mamba_context_parallel.py; this is the correct redundant-compute
+        # baseline — head-split CP is the round-2 optimization)
+        cp = G.get_context_parallel_world_size()
+        cp_mode = getattr(self.config, "cp_comm_type", "p2p")
+        if cp > 1 and inference_state is None:
+            from megatron_amd.parallel.context_parallel import gather_cp_sequence
+
+            hidden_states = gather_cp_sequence(hidden_states, seq_dim=0, mode=cp_mode)
         s, b, _ = hidden_states.shape
         zxbcdt, _ = self.in_proj(hidden_states)  # [s, b, d_proj/tp]
         z, x, B, C, dt = torch.split(zxbcdt, self.split_sizes, dim=-1)
@@ -180,6 +190,11 @@ class MambaMixer(nn.Module):
         y = y.reshape(b, s, self.d_inner_local).permute(1, 0, 2)  # [s, b, d_inner/tp]
         y = self.norm(y, z)
         out, _ = self.out_proj(y)
+        if cp > 1 and inference_state is None:
+            from megatron_amd.parallel.context_parallel import slice_for_cp_rank
+
+            out = slice_for_cp_rank(out, G.get_context_parallel_rank(), cp,
+                                    seq_dim=0, mode=cp_mode)
         return out
 
     def _step(self, z, x, B, C, dt, state: dict):
