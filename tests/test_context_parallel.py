@@ -174,3 +174,26 @@ def test_pick_cp_size_powers_of_two():
     assert pick_cp_size(8192, 8, 4096) == 2
     assert pick_cp_size(32768, 8, 4096) == 8
     assert pick_cp_size(10 ** 6, 4, 4096) == 4  # clamped at max_cp
+
+
+def test_balanced_cp_fuzz():
+    import random
+
+    from megatron_amd.parallel.balanced_cp import BalancedCPScheduler
+
+    for seed in range(5):
+        rng = random.Random(seed)
+        world = rng.choice([4, 8])
+        sched = BalancedCPScheduler(world_size=world, max_cp=world,
+                                    chunk_target=rng.choice([512, 2048]))
+        seq_lens = [rng.randint(64, 16384) for _ in range(rng.randint(1, 40))]
+        per_rank = sched.schedule(seq_lens)
+        seen = {}
+        for r, assigns in per_rank.items():
+            for a in assigns:
+                seen.setdefault(a.sample, set()).add(r)
+        assert set(seen) == set(range(len(seq_lens)))
+        for i, ranks in seen.items():
+            assert len(ranks) == next(a.cp_size for aa in per_rank.values()
+                                      for a in aa if a.sample == i)
+        assert sched.balance_ratio(seq_lens) < 2.5

@@ -73,3 +73,24 @@ def test_gpt_packed_forward_matches_separate():
     assert out_p.shape == (12, 1, 64)
     assert_close(out_p[:7], out_1, rtol=1e-5, atol=1e-5)
     assert_close(out_p[7:], out_2, rtol=1e-5, atol=1e-5)
+
+
+def test_pack_sequences_fuzz_conservation():
+    """Property: packing never loses or duplicates tokens and every row's
+    cu_seqlens is consistent, across random document mixes."""
+    import random
+
+    rng = random.Random(0)
+    for trial in range(10):
+        n_docs = rng.randint(1, 12)
+        docs = [torch.randint(1, 999, (rng.randint(1, 20),)) for _ in range(n_docs)]
+        L = rng.choice([16, 24, 32])
+        rows = pack_sequences(docs, seq_length=L, pad_id=0)
+        flat = torch.cat([r["tokens"][r["tokens"] != 0] for r in rows]).sort().values
+        expect = torch.cat([d[:L] for d in docs]).sort().values
+        assert torch.equal(flat, expect), trial
+        for r in rows:
+            cu = r["cu_seqlens"]
+            assert cu[0] == 0 and int(cu[-1]) <= L
+            assert (cu[1:] > cu[:-1]).all() or int(cu[-1]) == L
+            assert r["tokens"].shape == (L,)
