@@ -78,3 +78,38 @@ def log_prob_of(logits: torch.Tensor, tokens: torch.Tensor) -> torch.Tensor:
     """log p(token) under logits. logits [b, V], tokens [b] -> [b]."""
     logp = torch.log_softmax(logits.float(), dim=-1)
     return logp.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
+
+
+@torch.no_grad()
+def beam_search(model, prompt, beam_width: int = 4, max_new_tokens: int = 16,
+                eod: Optional[int] = None, length_penalty: float = 1.0):
+    """Beam search decode (reference legacy text-generation beam search).
+
+    Full-forward scoring (no KV reuse — correctness-first reference path;
+    the cached engines cover throughput).  Returns a list of
+    (tokens, score) sorted best-first, score = sum logp / len^length_penalty
+    over the generated part."""
+    device = next(model.parameters()).device
+    beams = [(list(prompt), 0.0, False)]
+    for _ in range(max_new_tokens):
+        alive = [b for b in beams if not b[2]]
+        if not alive:
+            break
+        batch = torch.tensor([b[0] for b in alive], device=device)
+        logits = model(batch)  # [s, b, V]
+        logp = torch.log_softmax(logits[-1].float(), dim=-1)  # [b, V]
+        cands = [b for b in beams if b[2]]
+        for (toks, score, _), row in zip(alive, logp):
+            topv, topi = row.topk(min(beam_width, row.numel()))
+            for v, i in zip(topv.tolist(), topi.tolist()):
+                done = eod is not None and i == eod
+                cands.append((toks + [i], score + v, done))
+        cands.sort(key=lambda b: b[1], reverse=True)
+        beams = cands[:beam_width]
+    n_prompt = len(prompt)
+
+    def final_score(b):
+        gen_len = max(len(b[0]) - n_prompt, 1)
+        return b[1] / (gen_len ** length_penalty)
+
+    return [(b[0], final_score(b)) for b in sorted(beams, key=final_score, reverse=True)]

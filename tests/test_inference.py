@@ -325,3 +325,31 @@ def test_stream_endpoint(tiny_model):
     eng2 = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
     expect = eng2.generate([[3, 7, 11]], SamplingParams(max_tokens=5, greedy=True))[0]
     assert toks == expect.output_tokens
+
+
+def test_beam_search_exhaustive_and_ordering(tiny_model):
+    import itertools
+
+    from megatron_amd.inference.sampling import beam_search
+
+    prompt = [3, 7]
+    # with beam_width >= V^1 per step expansion kept, width V makes the
+    # search exhaustive for short horizons: compare against brute force
+    n_new = 2
+    best = beam_search(tiny_model, prompt, beam_width=VOCAB, max_new_tokens=n_new,
+                       length_penalty=0.0)
+    # brute force over all 2-token continuations (vectorized: one forward
+    # for step 1, one batched forward over all V step-2 prefixes)
+    lp1 = torch.log_softmax(tiny_model(torch.tensor([prompt]))[-1, 0].float(), dim=-1)  # [V]
+    batch = torch.tensor([prompt + [t] for t in range(VOCAB)])
+    lp2 = torch.log_softmax(tiny_model(batch)[-1].float(), dim=-1)  # [V, V]
+    total = lp1.view(-1, 1) + lp2  # [t1, t2]
+    flat = int(total.argmax())
+    brute = [flat // VOCAB, flat % VOCAB]
+    assert best[0][0][len(prompt):] == brute
+    # small beam: results sorted best-first, right lengths
+    res = beam_search(tiny_model, prompt, beam_width=3, max_new_tokens=4)
+    assert len(res) == 3
+    scores = [s for _, s in res]
+    assert scores == sorted(scores, reverse=True)
+    assert all(len(t) == len(prompt) + 4 for t, _ in res)
