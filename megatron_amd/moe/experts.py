@@ -52,6 +52,8 @@ class GroupedMLP(nn.Module):
             return ops.swiglu(x)
         if self.config.activation == "geglu":
             return ops.geglu(x)
+        if self.config.activation == "squared_relu":
+            return ops.squared_relu(x)
         return torch.nn.functional.gelu(x)
 
     def forward(self, tokens: torch.Tensor, tokens_per_expert: torch.Tensor) -> torch.Tensor:
