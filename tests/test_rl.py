@@ -90,3 +90,39 @@ def test_rl_step_end_to_end():
     with torch.no_grad():
         lp_after = policy_logprobs(model.eval(), ids, chosen)
     assert torch.isfinite(lp_after[mask.bool()]).all()
+
+
+def test_refit_updates_rollout_model():
+    """RL refit flow: after a training step changes the policy, refit_model
+    pushes the new weights into a separate rollout/inference model."""
+    import torch
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.resharding import refit_model
+    from tests.utils import init_single
+
+    init_single()
+    cfg = TransformerConfig(num_layers=2, hidden_size=32, num_attention_heads=4,
+                            num_query_groups=2, vocab_size=64, ffn_hidden_size=48,
+                            gradient_accumulation_fusion=False)
+    torch.manual_seed(0)
+    train_model = GPTModel(cfg)
+    torch.manual_seed(0)
+    rollout_model = GPTModel(cfg).eval()
+    toks = torch.randint(0, 64, (2, 8))
+    with torch.no_grad():
+        before = rollout_model(toks, position_ids=None, attention_mask=None).clone()
+
+    # a "training step": perturb the policy
+    opt = torch.optim.SGD(train_model.parameters(), lr=0.5)
+    loss = train_model(toks, labels=toks).mean()
+    loss.backward()
+    opt.step()
+
+    refit_model(train_model, rollout_model)
+    with torch.no_grad():
+        after = rollout_model(toks, position_ids=None, attention_mask=None)
+        expect = train_model(toks, position_ids=None, attention_mask=None)
+    assert not torch.allclose(after, before, atol=1e-5)
+    assert torch.allclose(after, expect, atol=1e-5)
