@@ -139,7 +139,8 @@ class StaticInferenceEngine:
         results = [GenerationResult(i, p) for i, p in enumerate(prompts)]
         finished = torch.zeros(b, dtype=torch.bool, device=self.device)
         for _ in range(params.max_tokens):
-            next_tok = sample(logits.cpu() if gen is not None else logits, params, gen).to(self.device)
+            next_tok = sample(logits.cpu() if gen is not None else logits, params, gen,
+                              prev_tokens=[r.output_tokens for r in results]).to(self.device)
             if params.return_log_probs:
                 lp = log_prob_of(logits, next_tok)
             for i in range(b):
