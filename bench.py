@@ -67,6 +67,8 @@ def parse_args():
     p.add_argument("--vpp", type=int, default=None)
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--no-dist-opt", action="store_true")
+    p.add_argument("--hip-graphs", action="store_true",
+                   help="capture per-layer fwd/bwd hipGraphs after warmup (dense models, dp-only)")
     p.add_argument("--seed", type=int, default=1234)
     return p.parse_args()
 
@@ -162,6 +164,19 @@ def main():
     for _ in range(args.warmup):
         one_step()
     barrier_sync()
+    if args.hip_graphs:
+        # capture after warmup at a step boundary (no live autograd graphs)
+        from megatron_amd.transformer.hip_graphs import capture_block_hip_graphs
+
+        core = chunks[0].module if hasattr(chunks[0], "module") else chunks[0]
+        sample = torch.randn(args.seq_len, args.micro_batch_size, cfg.hidden_size,
+                             device=device, dtype=cfg.params_dtype)
+        freqs = core._rotary_freqs(args.seq_len, device)
+        n = capture_block_hip_graphs(core.decoder, sample, rotary_freqs=freqs)
+        if rank == 0:
+            print(f"# captured {n} layer hipGraphs", flush=True)
+        one_step()  # one replay-path warmup step
+        barrier_sync()
     t0 = time.perf_counter()
     last = None
     for _ in range(args.steps):
