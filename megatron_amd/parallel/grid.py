@@ -182,8 +182,14 @@ class ParallelGrid:
     def _register(self, name: str, groups: List[List[int]], gloo: bool = False):
         my_group, my_ranks, my_gloo = None, None, None
         use_dist = dist.is_initialized() and self.world_size > 1
+        # per-group RCCL tuning (reference get_nccl_options yaml):
+        # CTA counts / stream priority from the loaded comm config
+        from megatron_amd.parallel.comm_config import pg_options_for
+
+        pg_opts = pg_options_for(name) if self._backend == "nccl" else None
         for ranks in groups:
-            pg = dist.new_group(ranks=ranks, backend=self._backend) if use_dist else None
+            pg = (dist.new_group(ranks=ranks, backend=self._backend, pg_options=pg_opts)
+                  if use_dist else None)
             pg_gloo = (
                 dist.new_group(ranks=ranks, backend="gloo", timeout=timedelta(minutes=30))
                 if (use_dist and gloo and self._gloo_ok and self._backend != "gloo")

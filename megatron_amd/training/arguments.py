@@ -87,6 +87,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--expert-model-parallel-size", "--ep", type=int, default=1)
     g.add_argument("--expert-tensor-parallel-size", type=int, default=None)
     g.add_argument("--sequence-parallel", action="store_true")
+    g.add_argument("--nccl-communicator-config-path", type=str, default=None,
+                   help="yaml of per-group RCCL knobs (min/max CTAs, stream priority)")
     g.add_argument("--tp-comm-overlap", action="store_true",
                    help="SP forward AG as a chunked ring overlapped with the GEMM")
 

@@ -41,6 +41,11 @@ def _print_rank0(*a):
 
 
 def initialize(args):
+    if getattr(args, "nccl_communicator_config_path", None):
+        # must load BEFORE groups are created so the options apply
+        from megatron_amd.parallel.comm_config import load_comm_config
+
+        load_comm_config(args.nccl_communicator_config_path)
     if torch.cuda.is_available():
         local_rank = int(os.environ.get("LOCAL_RANK", args.rank))
         torch.cuda.set_device(local_rank)

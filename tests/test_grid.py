@@ -115,3 +115,21 @@ def test_hyper_comm_grid_groups_gloo():
     from tests.utils import spawn_dist
 
     spawn_dist(_hyper_groups_case, 2)
+
+
+def test_comm_config_options(tmp_path):
+    from megatron_amd.parallel.comm_config import comm_config, load_comm_config, pg_options_for
+
+    cfg = tmp_path / "nccl.yaml"
+    cfg.write_text("tp:\n  max_ctas: 8\n  is_high_priority_stream: true\ndp:\n  min_ctas: 2\n")
+    load_comm_config(str(cfg))
+    assert comm_config()["tp"]["max_ctas"] == 8
+    # CPU container: options resolve to None (gloo ignores them) but the
+    # config is validated
+    assert pg_options_for("tp") is None or pg_options_for("tp").is_high_priority_stream
+    assert pg_options_for("pp") is None  # unconfigured group
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="unknown comm-config"):
+        load_comm_config({"tp": {"bogus_knob": 1}})
+    load_comm_config({})  # reset
