@@ -83,6 +83,12 @@ class HuggingFaceTokenizer(MegatronTokenizer):
     def detokenize(self, ids) -> str:
         return self._tok.decode(ids)
 
+    def apply_chat_template(self, messages, add_generation_prompt: bool = True) -> List[int]:
+        """Chat-format token ids via the model's HF chat template
+        (serving-side convenience; messages = [{'role', 'content'}, ...])."""
+        return self._tok.apply_chat_template(
+            messages, add_generation_prompt=add_generation_prompt, tokenize=True)
+
     @property
     def vocab_size(self) -> int:
         return len(self._tok)
