@@ -49,6 +49,25 @@ def create_app(engine, tokenizer=None):
             ]
         }
 
+    @app.post("/api/chat")
+    async def chat(request: Request):
+        """Chat completion: messages are rendered with the tokenizer's chat
+        template and generated as one continuation."""
+        req = await request.json()
+        assert tokenizer is not None and hasattr(tokenizer, "apply_chat_template"), \
+            "chat endpoint needs a chat-template-capable tokenizer"
+        prompt_ids = tokenizer.apply_chat_template(req["messages"], add_generation_prompt=True)
+        params = SamplingParams(
+            max_tokens=int(req.get("max_tokens", 256)),
+            temperature=float(req.get("temperature", 1.0)),
+            top_k=int(req.get("top_k", 0)), top_p=float(req.get("top_p", 0.0)),
+            greedy=bool(req.get("greedy", False)), seed=req.get("seed"))
+        with lock:
+            r = engine.generate([prompt_ids], params)[0]
+        return {"message": {"role": "assistant",
+                            "content": tokenizer.detokenize(r.output_tokens)},
+                "tokens": r.output_tokens}
+
     @app.post("/api/stream")
     async def stream(request: Request):
         """SSE token streaming for a single prompt (dynamic engine only):

@@ -353,3 +353,34 @@ def test_beam_search_exhaustive_and_ordering(tiny_model):
     scores = [s for _, s in res]
     assert scores == sorted(scores, reverse=True)
     assert all(len(t) == len(prompt) + 4 for t, _ in res)
+
+
+def test_chat_endpoint(tiny_model):
+    from fastapi.testclient import TestClient
+
+    from megatron_amd.inference.server import create_app
+
+    class _ChatTok:
+        eod = VOCAB - 1
+
+        def apply_chat_template(self, messages, add_generation_prompt=True):
+            ids = []
+            for m in messages:
+                ids += [ord(c) % (VOCAB - 2) for c in (m["role"] + ":" + m["content"])]
+            return ids
+
+        def tokenize(self, s):
+            return [ord(c) % (VOCAB - 2) for c in s]
+
+        def detokenize(self, toks):
+            return "".join(chr(97 + (t % 26)) for t in toks)
+
+    tok = _ChatTok()
+    eng = DynamicInferenceEngine(tiny_model, tokenizer=tok, num_blocks=16, block_size=8)
+    client = TestClient(create_app(eng, tok))
+    r = client.post("/api/chat", json={"messages": [{"role": "user", "content": "hi"}],
+                                       "max_tokens": 4, "greedy": True})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["message"]["role"] == "assistant"
+    assert len(body["tokens"]) <= 4 and isinstance(body["message"]["content"], str)
