@@ -189,6 +189,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--profile-step-start", type=int, default=3)
     g.add_argument("--profile-step-end", type=int, default=5)
     g.add_argument("--profile-dir", type=str, default="./torchprof")
+    g.add_argument("--memory-snapshot-path", type=str, default=None,
+                   help="dump a torch.cuda memory-history snapshot here at exit")
 
     return p
 

@@ -147,6 +147,9 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
         from megatron_amd.utils.annotations import enable_annotations
 
         enable_annotations(True)
+    if getattr(args, "memory_snapshot_path", None) and torch.cuda.is_available():
+        # reference training.py:2859 --memory-snapshot-path
+        torch.cuda.memory._record_memory_history(max_entries=100_000)
     initialize_rerun_state_machine(args.rerun_mode)
     fault_injector = None
     if getattr(args, "fault_injection_type", None):
@@ -345,5 +348,8 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
         save_checkpoint(args.save, chunks, optimizer, iteration, scheduler)
         _print_rank0(f"saved final checkpoint at iteration {iteration}")
         append_progress_log(args.save, args.rank, f"finished at iteration {iteration}")
+    if getattr(args, "memory_snapshot_path", None) and torch.cuda.is_available():
+        torch.cuda.memory._dump_snapshot(args.memory_snapshot_path)
+        torch.cuda.memory._record_memory_history(enabled=None)
     metrics.close()
     return iteration
