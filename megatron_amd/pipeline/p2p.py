@@ -62,7 +62,12 @@ class P2PCommunicator:
             if recv_next_shape is not None:
                 ops.append(dist.P2POp(dist.irecv, recv_next_shape, self.next_rank, group=self.group))
 
-        (sends() or recvs()) if even else (recvs() or sends())
+        if even:
+            sends()
+            recvs()
+        else:
+            recvs()
+            sends()
         if ops:
             for r in dist.batch_isend_irecv(ops):
                 r.wait()
