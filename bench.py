@@ -130,6 +130,9 @@ def main():
     ddp_cfg = DDPConfig(
         grad_reduce_in_fp32=True, overlap_grad_reduce=True,
         use_distributed_optimizer=not args.no_dist_opt, bucket_size=40_000_000,
+        # graph replay re-runs captured kernels but not python grad-ready
+        # callbacks: bucket overlap would deadlock, so reduce at step end
+        overlap_grad_reduce=not args.hip_graphs,
     )
 
     def provider(config, pre_process=True, post_process=True, vp_stage=None):

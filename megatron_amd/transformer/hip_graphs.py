@@ -20,7 +20,10 @@ layer here gets its OWN forward and backward graph with default pools:
     the reference's `external` grad mode).
 
 Scope: static shapes (one (s, b, h) per capture; eager fallback otherwise,
-and always under activation recompute or inference contexts).  Rotary
+and always under activation recompute or inference contexts).  With DDP,
+disable overlap_grad_reduce: replay re-runs the captured kernels (including
+fused main_grad accumulation) but NOT python grad-ready callbacks, so
+bucket overlap would wait forever — reduce at finish_grad_sync instead.  Rotary
 freqs / masks are closed over as static tensors (step-invariant in
 pretraining).  The dynamic inference engine has its own decode-step graph
 runner (inference/engine.py _DecodeGraphRunner); this module is the
