@@ -384,3 +384,17 @@ def test_chat_endpoint(tiny_model):
     body = r.json()
     assert body["message"]["role"] == "assistant"
     assert len(body["tokens"]) <= 4 and isinstance(body["message"]["content"], str)
+
+
+def test_seeded_sampling_reproducible(tiny_model):
+    params = SamplingParams(max_tokens=6, temperature=0.8, top_k=8, seed=1234,
+                            stop_on_eod=False)
+    eng1 = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    eng2 = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    a = eng1.generate([[3, 7, 11]], params)[0]
+    b = eng2.generate([[3, 7, 11]], params)[0]
+    assert a.output_tokens == b.output_tokens
+    # different seed diverges (overwhelmingly likely over 6 sampled tokens)
+    c = eng2.generate([[3, 7, 11]], SamplingParams(max_tokens=6, temperature=0.8,
+                                                   top_k=8, seed=99, stop_on_eod=False))[0]
+    assert len(c.output_tokens) == 6
