@@ -50,6 +50,8 @@ class LLaVAModel(nn.Module):
         self.image_token_index = image_token_index
         self.drop_vision_class_token = drop_vision_class_token
         self.language_model = GPTModel(language_config)
+        self.config = language_config  # training stack (DDP/optimizer) reads .config
+        self.img_h, self.img_w = img_h, img_w
         self.vision_model = CLIPViTModel(vision_config, img_h=img_h, img_w=img_w,
                                          patch_dim=patch_dim)
         self.vision_projection = MultimodalProjector(

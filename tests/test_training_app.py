@@ -121,3 +121,13 @@ def test_run_config_dumped(tmp_path):
     cfgd = json.load(open(os.path.join(ckpt, "run_config.json")))
     assert cfgd["args"]["num_layers"] == "2"
     assert "hidden_size" in cfgd["transformer_config"]
+
+
+def test_pretrain_vlm_entry():
+    import pretrain_vlm
+
+    it = pretrain(pretrain_vlm.model_provider,
+                  TINY + ["--train-iters", "2", "--seed", "5",
+                          "--position-embedding-type", "rope"],
+                  forward_step_builder=pretrain_vlm.forward_step_builder)
+    assert it == 2
