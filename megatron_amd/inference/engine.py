@@ -84,6 +84,8 @@ class GenerationResult:
     prompt_tokens: List[int]
     output_tokens: List[int] = field(default_factory=list)
     log_probs: List[float] = field(default_factory=list)
+    # per position: [(token, logprob) x top_n] when params.top_n_logprobs > 0
+    top_logprobs: List[list] = field(default_factory=list)
     text: Optional[str] = None
     finished: bool = False
 
@@ -280,6 +282,10 @@ class DynamicInferenceEngine:
                          prev_tokens=[req.result.output_tokens])[0])
         if req.params.return_log_probs:
             req.result.log_probs.append(float(log_prob_of(logits_row, torch.tensor([tok]))[0]))
+        if req.params.top_n_logprobs > 0:
+            lp = torch.log_softmax(logits_row[0].float(), dim=-1)
+            v, i = lp.topk(req.params.top_n_logprobs)
+            req.result.top_logprobs.append(list(zip(i.tolist(), v.tolist())))
         return tok
 
     @torch.no_grad()
@@ -338,7 +344,7 @@ class DynamicInferenceEngine:
         # batched sampling fast path: all requests share one params object
         batch_toks = None
         p0 = batch[0].params
-        if all((r.params is p0) and r.gen is None for r in batch):
+        if all((r.params is p0) and r.gen is None for r in batch) and p0.top_n_logprobs == 0:
             batch_toks = sample(logits, p0,
                                 prev_tokens=[r.result.output_tokens for r in batch]).tolist()
             if p0.return_log_probs:

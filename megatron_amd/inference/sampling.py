@@ -25,6 +25,8 @@ class SamplingParams:
     # multiplicative penalty on logits of already-generated tokens (>1
     # discourages repeats; HF convention: divide positive / multiply negative)
     repetition_penalty: float = 1.0
+    # attach the top-N (token, logprob) alternatives per generated position
+    top_n_logprobs: int = 0
 
 
 def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0) -> torch.Tensor:

@@ -398,3 +398,19 @@ def test_seeded_sampling_reproducible(tiny_model):
     c = eng2.generate([[3, 7, 11]], SamplingParams(max_tokens=6, temperature=0.8,
                                                    top_k=8, seed=99, stop_on_eod=False))[0]
     assert len(c.output_tokens) == 6
+
+
+def test_top_n_logprobs(tiny_model):
+    params = SamplingParams(max_tokens=4, greedy=True, stop_on_eod=False,
+                            top_n_logprobs=3, return_log_probs=True)
+    eng = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    r = eng.generate([[3, 7, 11]], params)[0]
+    assert len(r.top_logprobs) == 4
+    for pos, (alts, chosen, tok) in enumerate(zip(r.top_logprobs, r.log_probs, r.output_tokens)):
+        assert len(alts) == 3
+        # greedy: the chosen token is the top-1 alternative
+        assert alts[0][0] == tok
+        assert abs(alts[0][1] - chosen) < 1e-4
+        # sorted descending
+        lps = [lp for _, lp in alts]
+        assert lps == sorted(lps, reverse=True)
