@@ -34,6 +34,9 @@ def create_app(engine, tokenizer=None):
             temperature=float(req.get("temperature", 1.0)),
             top_k=int(req.get("top_k", 0)), top_p=float(req.get("top_p", 0.0)),
             greedy=bool(req.get("greedy", False)), return_log_probs=logprobs,
+            top_n_logprobs=int(req.get("top_n_logprobs", 0)),
+            repetition_penalty=float(req.get("repetition_penalty", 1.0)),
+            stop_strings=tuple(req.get("stop_strings", ())),
             seed=req.get("seed"))
         with lock:
             results = engine.generate(prompts, params)
@@ -44,6 +47,7 @@ def create_app(engine, tokenizer=None):
                     "tokens": r.output_tokens,
                     "text": r.text,
                     "logprobs": r.log_probs if logprobs else None,
+                    "top_logprobs": r.top_logprobs or None,
                 }
                 for r in results
             ]
