@@ -414,3 +414,46 @@ def test_top_n_logprobs(tiny_model):
         # sorted descending
         lps = [lp for _, lp in alts]
         assert lps == sorted(lps, reverse=True)
+
+
+def _tp2_serving_case(rank, world, ckpt_dir):
+    import json
+    import os
+
+    from megatron_amd.checkpoint.checkpointing import load_checkpoint
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    G.initialize_model_parallel(tensor_parallel_size=2)
+    model_parallel_seed(123)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=128, vocab_size=VOCAB, max_position_embeddings=256,
+        tensor_parallel_size=2)
+    model = GPTModel(cfg).eval()
+    load_checkpoint(ckpt_dir, [model], None, load_rng=False)
+    eng = DynamicInferenceEngine(model, num_blocks=16, block_size=8, use_hip_graphs=False)
+    res = eng.generate([[3, 7, 11, 2, 9], [5, 1]],
+                       SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
+    if rank == 0:
+        with open(os.environ["TP_SERVE_OUT"], "w") as f:
+            json.dump([r.output_tokens for r in res], f)
+
+
+def test_tp2_serving_matches_single(tiny_model, tmp_path, monkeypatch):
+    """TP=2 continuous-batching serving (vocab-parallel logits gathered over
+    the TP group) equals single-rank generation, via checkpoint reshard."""
+    import json
+
+    from megatron_amd.checkpoint.checkpointing import save_checkpoint
+    from tests.utils import spawn_dist
+
+    out = tmp_path / "tp_serve.json"
+    ckpt = str(tmp_path / "ckpt")
+    monkeypatch.setenv("TP_SERVE_OUT", str(out))
+    save_checkpoint(ckpt, [tiny_model], None, iteration=0)
+    eng = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    expected = eng.generate([[3, 7, 11, 2, 9], [5, 1]],
+                            SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
+    spawn_dist(_tp2_serving_case, 2, ckpt)
+    got = json.load(open(out))
+    assert got == [r.output_tokens for r in expected]
