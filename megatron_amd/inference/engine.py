@@ -17,6 +17,7 @@ from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence
 
 import torch
+import torch.distributed as dist
 
 from megatron_amd.inference.contexts import DynamicInferenceContext, StaticInferenceContext
 from megatron_amd.inference.sampling import SamplingParams, log_prob_of, sample
@@ -113,6 +114,40 @@ class StaticInferenceEngine:
             cfg.kv_channels, dtype=cfg.params_dtype, device=device)
         self.max_seq = max_seq
 
+    def _pp_forward(self, tokens: torch.Tensor) -> Optional[torch.Tensor]:
+        """Pipeline-parallel forward for inference (reference inference PP
+        communication): stages relay hidden states over the PP group; logits
+        exist on the last stage only, which returns them (None elsewhere).
+        All ranks must call with the same tokens."""
+        grid = G.get_grid()
+        if grid.pp == 1:
+            return self.model(tokens, inference_context=self.context)
+        s = tokens.shape[1]
+        b = tokens.shape[0]
+        h = self.model.config.hidden_size
+        dtype = self.model.config.params_dtype
+        group = grid.group("pp")
+        if not grid.is_pipeline_first_stage(ignore_virtual=True):
+            buf = torch.empty(s, b, h, dtype=dtype, device=self.device)
+            dist.recv(buf, src=grid.pipeline_prev_rank(), group=group)
+            self.model.set_input_tensor(buf)
+        out = self.model(tokens, inference_context=self.context)
+        if not grid.is_pipeline_last_stage(ignore_virtual=True):
+            dist.send(out.contiguous(), dst=grid.pipeline_next_rank(), group=group)
+            return None
+        return out
+
+    def _pp_broadcast_tokens(self, toks: Optional[torch.Tensor], b: int) -> torch.Tensor:
+        """Sampled tokens live on the last stage; share them with the rest."""
+        grid = G.get_grid()
+        if grid.pp == 1:
+            return toks
+        if toks is None:
+            toks = torch.empty(b, dtype=torch.long, device=self.device)
+        src = grid.ranks("pp")[-1]
+        dist.broadcast(toks, src=src, group=grid.group("pp"))
+        return toks
+
     @torch.no_grad()
     def generate(self, prompts: Sequence, params: SamplingParams = SamplingParams()) -> List[GenerationResult]:
         if self.tokenizer is not None and isinstance(prompts[0], str):
@@ -132,31 +167,40 @@ class StaticInferenceEngine:
             tokens[i, : len(p)] = torch.as_tensor(p, device=self.device)
 
         self.context.reset(b)
-        logits_tp = self.model(tokens, inference_context=self.context)  # [Lmax, b, V/tp]
+        pp_last = G.get_grid().is_pipeline_last_stage(ignore_virtual=True) if G.grid_initialized() else True
+        logits_tp = self._pp_forward(tokens)  # [Lmax, b, V/tp] on the last stage
         self.context.set_prompt_lens(lens)
-        last_pos = torch.as_tensor(lens, device=self.device) - 1
-        last = logits_tp[last_pos, torch.arange(b, device=self.device)]  # [b, V/tp]
-        logits = _full_logits(last).float()
+        if pp_last:
+            last_pos = torch.as_tensor(lens, device=self.device) - 1
+            last = logits_tp[last_pos, torch.arange(b, device=self.device)]  # [b, V/tp]
+            logits = _full_logits(last).float()
+        else:
+            logits = None
 
         results = [GenerationResult(i, p) for i, p in enumerate(prompts)]
         finished = torch.zeros(b, dtype=torch.bool, device=self.device)
         for _ in range(params.max_tokens):
-            next_tok = sample(logits.cpu() if gen is not None else logits, params, gen,
-                              prev_tokens=[r.output_tokens for r in results]).to(self.device)
-            if params.return_log_probs:
-                lp = log_prob_of(logits, next_tok)
+            if pp_last:
+                next_tok = sample(logits.cpu() if gen is not None else logits, params, gen,
+                                  prev_tokens=[r.output_tokens for r in results]).to(self.device)
+                if params.return_log_probs:
+                    lp = log_prob_of(logits, next_tok)
+            else:
+                next_tok = None
+            next_tok = self._pp_broadcast_tokens(next_tok, b)
             for i in range(b):
                 if not bool(finished[i]):
                     results[i].output_tokens.append(int(next_tok[i]))
-                    if params.return_log_probs:
+                    if params.return_log_probs and pp_last:
                         results[i].log_probs.append(float(lp[i]))
             if params.stop_on_eod:
                 finished |= next_tok == self.eod
             if bool(finished.all()):
                 break
-            logits_tp = self.model(next_tok.view(b, 1), inference_context=self.context)
+            logits_tp = self._pp_forward(next_tok.view(b, 1))
             self.context.advance(1)
-            logits = _full_logits(logits_tp[0]).float()
+            if pp_last:
+                logits = _full_logits(logits_tp[0]).float()
 
         for r in results:
             if params.stop_on_eod and r.output_tokens and r.output_tokens[-1] == self.eod:

@@ -457,3 +457,48 @@ def test_tp2_serving_matches_single(tiny_model, tmp_path, monkeypatch):
     spawn_dist(_tp2_serving_case, 2, ckpt)
     got = json.load(open(out))
     assert got == [r.output_tokens for r in expected]
+
+
+def _pp2_serving_case(rank, world, ckpt_dir):
+    import json
+    import os
+
+    from megatron_amd.checkpoint.checkpointing import load_checkpoint
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    G.initialize_model_parallel(pipeline_parallel_size=2)
+    model_parallel_seed(123)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=128, vocab_size=VOCAB, max_position_embeddings=256,
+        pipeline_parallel_size=2)
+    grid = G.get_grid()
+    model = GPTModel(cfg, pre_process=grid.is_pipeline_first_stage(ignore_virtual=True),
+                     post_process=grid.is_pipeline_last_stage(ignore_virtual=True)).eval()
+    load_checkpoint(ckpt_dir, [model], None, load_rng=False)
+    eng = StaticInferenceEngine(model, max_batch=4, max_seq=64)
+    res = eng.generate([[3, 7, 11, 2, 9], [5, 1]],
+                       SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
+    if rank == 0:  # first stage also has the broadcast tokens
+        with open(os.environ["PP_SERVE_OUT"], "w") as f:
+            json.dump([r.output_tokens for r in res], f)
+
+
+def test_pp2_serving_matches_single(tiny_model, tmp_path, monkeypatch):
+    """PP=2 static-engine serving (hidden relay + token broadcast over the
+    PP group) equals single-rank generation."""
+    import json
+
+    from megatron_amd.checkpoint.checkpointing import save_checkpoint
+    from tests.utils import spawn_dist
+
+    out = tmp_path / "pp_serve.json"
+    ckpt = str(tmp_path / "ckpt")
+    monkeypatch.setenv("PP_SERVE_OUT", str(out))
+    save_checkpoint(ckpt, [tiny_model], None, iteration=0)
+    eng = StaticInferenceEngine(tiny_model, max_batch=4, max_seq=64)
+    expected = eng.generate([[3, 7, 11, 2, 9], [5, 1]],
+                            SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
+    spawn_dist(_pp2_serving_case, 2, ckpt)
+    got = json.load(open(out))
+    assert got == [r.output_tokens for r in expected]
