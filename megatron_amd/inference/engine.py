@@ -244,8 +244,9 @@ class DynamicInferenceEngine:
         self.max_batch = max_batch
         self.max_prefill_tokens = max_prefill_tokens
         self._graphs = None
+        pp = G.get_grid().pp if G.grid_initialized() else 1
         if (use_hip_graphs and torch.cuda.is_available() and device.type == "cuda"
-                and tp == 1):
+                and tp == 1 and pp == 1):
             self._graphs = _DecodeGraphRunner(self.model, self.context, device)
         self._ids = itertools.count()
         self.waiting: List[_Request] = []
@@ -290,6 +291,39 @@ class DynamicInferenceEngine:
             self.preempted.pop(0)
             req.block_table = self.offloader.swap_in(handle)
             self.active.append(req)
+
+    def _pp(self):
+        grid = G.get_grid() if G.grid_initialized() else None
+        return grid if (grid is not None and grid.pp > 1) else None
+
+    def _model_forward(self, toks: torch.Tensor):
+        """Model forward with pipeline-parallel hidden relay (logits on the
+        last stage only; None elsewhere).  Mirrors the static engine."""
+        grid = self._pp()
+        if grid is None:
+            return self.model(toks, inference_context=self.context)
+        s, b = toks.shape[1], toks.shape[0]
+        h = self.model.config.hidden_size
+        group = grid.group("pp")
+        if not grid.is_pipeline_first_stage(ignore_virtual=True):
+            buf = torch.empty(s, b, h, dtype=self.model.config.params_dtype,
+                              device=self.device)
+            dist.recv(buf, src=grid.pipeline_prev_rank(), group=group)
+            self.model.set_input_tensor(buf)
+        out = self.model(toks, inference_context=self.context)
+        if not grid.is_pipeline_last_stage(ignore_virtual=True):
+            dist.send(out.contiguous(), dst=grid.pipeline_next_rank(), group=group)
+            return None
+        return out
+
+    def _pp_share_tokens(self, toks: Optional[List[int]], n: int) -> List[int]:
+        grid = self._pp()
+        if grid is None:
+            return toks
+        t = (torch.as_tensor(toks, dtype=torch.long, device=self.device)
+             if toks is not None else torch.empty(n, dtype=torch.long, device=self.device))
+        dist.broadcast(t, src=grid.ranks("pp")[-1], group=grid.group("pp"))
+        return t.tolist()
 
     def _blocks_for(self, n_tokens: int) -> int:
         bs = self.context.block_size
@@ -347,11 +381,15 @@ class DynamicInferenceEngine:
                 self.context.begin_prefill(req.block_table, req.cached)
                 toks = torch.as_tensor(req.prompt[req.cached:req.cached + chunk],
                                        device=self.device).view(1, -1)
-                logits_tp = self.model(toks, inference_context=self.context)  # [chunk, 1, V/tp]
+                logits_tp = self._model_forward(toks)  # [chunk, 1, V/tp] on pp-last
                 req.cached += chunk
                 if req.cached == len(req.prompt):
-                    logits = _full_logits(logits_tp[-1, 0]).float()
-                    tok = self._sample_row(logits, req)
+                    if logits_tp is not None:
+                        logits = _full_logits(logits_tp[-1, 0]).float()
+                        tok = self._sample_row(logits, req)
+                    else:
+                        tok = None
+                    tok = self._pp_share_tokens([tok] if tok is not None else None, 1)[0]
                     req.result.output_tokens.append(tok)
                     self.waiting.pop(0)
                     if req.params.stop_on_eod and tok == self.eod:
@@ -383,21 +421,28 @@ class DynamicInferenceEngine:
         else:
             self.context.begin_decode(tables, lens)
             toks = torch.as_tensor([r.next_input for r in batch], device=self.device).view(-1, 1)
-            logits_tp = self.model(toks, inference_context=self.context)  # [1, b, V/tp]
-            logits = _full_logits(logits_tp[0]).float()
+            logits_tp = self._model_forward(toks)  # [1, b, V/tp] on pp-last
+            logits = _full_logits(logits_tp[0]).float() if logits_tp is not None else None
+        pp_last = logits is not None
         # batched sampling fast path: all requests share one params object
         batch_toks = None
         p0 = batch[0].params
-        if all((r.params is p0) and r.gen is None for r in batch) and p0.top_n_logprobs == 0:
+        if pp_last and all((r.params is p0) and r.gen is None for r in batch) and p0.top_n_logprobs == 0:
             batch_toks = sample(logits, p0,
                                 prev_tokens=[r.result.output_tokens for r in batch]).tolist()
             if p0.return_log_probs:
                 lps = log_prob_of(logits, torch.as_tensor(batch_toks, device=logits.device)).tolist()
+        if self._pp() is not None:
+            if batch_toks is None and pp_last:
+                batch_toks = [self._sample_row(logits[i], r) for i, r in enumerate(batch)]
+            batch_toks = self._pp_share_tokens(batch_toks, len(batch))
         still = []
         for i, req in enumerate(batch):
             if batch_toks is not None:
                 tok = int(batch_toks[i])
-                if p0.return_log_probs:
+                if pp_last and p0.return_log_probs and req.params is p0 and req.gen is None                         and self._pp() is not None:
+                    pass  # per-row _sample_row already recorded probs on pp-last
+                elif pp_last and batch_toks is not None and p0.return_log_probs                         and self._pp() is None:
                     req.result.log_probs.append(float(lps[i]))
             else:
                 tok = self._sample_row(logits[i], req)

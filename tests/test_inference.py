@@ -502,3 +502,49 @@ def test_pp2_serving_matches_single(tiny_model, tmp_path, monkeypatch):
     spawn_dist(_pp2_serving_case, 2, ckpt)
     got = json.load(open(out))
     assert got == [r.output_tokens for r in expected]
+
+
+def _pp2_dynamic_case(rank, world, ckpt_dir):
+    import json
+    import os
+
+    from megatron_amd.checkpoint.checkpointing import load_checkpoint
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    G.initialize_model_parallel(pipeline_parallel_size=2)
+    model_parallel_seed(123)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=128, vocab_size=VOCAB, max_position_embeddings=256,
+        pipeline_parallel_size=2)
+    grid = G.get_grid()
+    model = GPTModel(cfg, pre_process=grid.is_pipeline_first_stage(ignore_virtual=True),
+                     post_process=grid.is_pipeline_last_stage(ignore_virtual=True)).eval()
+    load_checkpoint(ckpt_dir, [model], None, load_rng=False)
+    eng = DynamicInferenceEngine(model, num_blocks=16, block_size=8,
+                                 max_prefill_tokens=4, use_hip_graphs=False)
+    res = eng.generate([[3, 7, 11, 2, 9], [5, 1]],
+                       SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
+    if rank == 0:
+        with open(os.environ["PP_DYN_OUT"], "w") as f:
+            json.dump([r.output_tokens for r in res], f)
+
+
+def test_pp2_dynamic_serving_matches_single(tiny_model, tmp_path, monkeypatch):
+    """PP=2 continuous batching (chunked prefill + decode with hidden relay)
+    equals single-rank generation."""
+    import json
+
+    from megatron_amd.checkpoint.checkpointing import save_checkpoint
+    from tests.utils import spawn_dist
+
+    out = tmp_path / "pp_dyn.json"
+    ckpt = str(tmp_path / "ckpt")
+    monkeypatch.setenv("PP_DYN_OUT", str(out))
+    save_checkpoint(ckpt, [tiny_model], None, iteration=0)
+    eng = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    expected = eng.generate([[3, 7, 11, 2, 9], [5, 1]],
+                            SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False))
+    spawn_dist(_pp2_dynamic_case, 2, ckpt)
+    got = json.load(open(out))
+    assert got == [r.output_tokens for r in expected]
