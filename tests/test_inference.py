@@ -548,3 +548,20 @@ def test_pp2_dynamic_serving_matches_single(tiny_model, tmp_path, monkeypatch):
     spawn_dist(_pp2_dynamic_case, 2, ckpt)
     got = json.load(open(out))
     assert got == [r.output_tokens for r in expected]
+
+
+def test_dp_coordinator_with_dynamic_engines(tiny_model):
+    from megatron_amd.inference.coordinator import DataParallelCoordinator
+
+    params = SamplingParams(max_tokens=5, greedy=True, stop_on_eod=False)
+    prompts = [[1, 2, 3], [4, 5], [7, 8, 9, 10], [11]]
+    single = DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8)
+    expected = single.generate(prompts, params)
+    engines = [DynamicInferenceEngine(tiny_model, num_blocks=16, block_size=8) for _ in range(2)]
+    coord = DataParallelCoordinator(engines, max_batch_per_engine=2)
+    try:
+        results = coord.generate(prompts, params)
+        for r, e in zip(results, expected):
+            assert r.output_tokens == e.output_tokens
+    finally:
+        coord.shutdown()
