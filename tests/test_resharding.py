@@ -89,3 +89,31 @@ def _assemble_case(rank, world):
 
 def test_assemble_global_tensors_tp2():
     spawn_dist(_assemble_case, 2)
+
+
+def _ep2_refit_case(rank, world):
+    G.initialize_model_parallel(expert_parallel_size=2)
+    model_parallel_seed(100)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+        vocab_size=64, ffn_hidden_size=48, num_experts=4, moe_router_topk=2,
+        moe_ffn_hidden_size=40, expert_parallel_size=2,
+        gradient_accumulation_fusion=False)
+    src = GPTModel(cfg)
+    model_parallel_seed(200)
+    dst = GPTModel(cfg)
+    refit_model(src, dst)
+    for (na, pa), (nb, pb) in zip(src.named_parameters(), dst.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), f"rank {rank}: {na}"
+    tokens = torch.randint(0, 64, (2, 8))
+    dist.broadcast(tokens, src=0)
+    with torch.no_grad():
+        a = src(tokens, position_ids=None, attention_mask=None)
+        b = dst(tokens, position_ids=None, attention_mask=None)
+    assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_refit_moe_ep2(tmp_path):
+    """Refit across EP=2 MoE models: expert flat-atlas shards (gated fc1
+    split, per-expert offsets) assemble and re-slice correctly."""
+    spawn_dist(_ep2_refit_case, 2)
