@@ -5,6 +5,7 @@ import os
 import zlib
 
 import torch
+import torch.distributed as dist
 
 from megatron_amd.checkpoint import load_checkpoint, save_checkpoint
 from megatron_amd.checkpoint.sharded import ShardedTensor, load as sh_load, save as sh_save
@@ -188,3 +189,49 @@ def test_non_persistent_checkpoint_retention_and_resume(tmp_path):
     # without the local tree, the persistent one is used
     root2, it2 = resolve_resume_source(persistent, str(tmp_path / "missing"))
     assert (root2, it2) == (persistent, 2)
+
+
+def _ep2_save_case(rank, world, ckpt_dir, out_file):
+    from megatron_amd.checkpoint.checkpointing import save_checkpoint
+    from megatron_amd.models.gpt import GPTModel
+
+    G.initialize_model_parallel(expert_parallel_size=2)
+    model_parallel_seed(77)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+        vocab_size=64, ffn_hidden_size=48, num_experts=4, moe_router_topk=2,
+        moe_ffn_hidden_size=40, expert_parallel_size=2,
+        gradient_accumulation_fusion=False)
+    model = GPTModel(cfg).eval()
+    save_checkpoint(ckpt_dir, [model], None, iteration=0)
+    torch.manual_seed(5)
+    tokens = torch.randint(0, 64, (2, 8))
+    dist.broadcast(tokens, src=0)
+    with torch.no_grad():  # EP forward is collective: all ranks participate
+        out = model(tokens, position_ids=None, attention_mask=None)
+    if rank == 0:
+        torch.save({"tokens": tokens, "out": out}, out_file)
+
+
+def test_moe_ep2_checkpoint_reshards_to_single(tmp_path):
+    """An EP=2 MoE checkpoint loads into a single-process (EP=1) model with
+    identical logits — expert shards reassemble through the atlas."""
+    from megatron_amd.checkpoint.checkpointing import load_checkpoint
+    from megatron_amd.models.gpt import GPTModel
+    from tests.utils import spawn_dist
+
+    ckpt = str(tmp_path / "ckpt")
+    ref_file = str(tmp_path / "ref.pt")
+    spawn_dist(_ep2_save_case, 2, ckpt, ref_file)
+
+    init_single(seed=99)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4, num_query_groups=2,
+        vocab_size=64, ffn_hidden_size=48, num_experts=4, moe_router_topk=2,
+        moe_ffn_hidden_size=40, gradient_accumulation_fusion=False)
+    model = GPTModel(cfg).eval()
+    load_checkpoint(ckpt, [model], None, load_rng=False)
+    ref = torch.load(ref_file)
+    with torch.no_grad():
+        out = model(ref["tokens"], position_ids=None, attention_mask=None)
+    assert_close(out, ref["out"], rtol=1e-5, atol=1e-5)
