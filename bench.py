@@ -128,7 +128,7 @@ def main():
         use_distributed_optimizer=not args.no_dist_opt,
     )
     ddp_cfg = DDPConfig(
-        grad_reduce_in_fp32=True, overlap_grad_reduce=True,
+        grad_reduce_in_fp32=True,
         use_distributed_optimizer=not args.no_dist_opt, bucket_size=40_000_000,
         # graph replay re-runs captured kernels but not python grad-ready
         # callbacks: bucket overlap would deadlock, so reduce at step end
