@@ -64,10 +64,17 @@ class ParamAndGradBuffer:
         param_dtype: torch.dtype,
         grad_dtype: torch.dtype,
         device: torch.device,
+        grad_scale_denom: int = 0,
     ):
         self.ddp_config = ddp_config
         self.dp_group = dp_group
         self.dp_size = dist.get_world_size(group=dp_group) if (dp_group is not None and dist.is_initialized()) else 1
+        # Grads are averaged over this many ranks.  For the dense buffer this
+        # equals the dp_cp group size (== dp_size); for the expert buffer the
+        # reduce group is edp but the average must still be over dp_cp so
+        # expert grads get the same effective scale as dense grads (reference
+        # expert_gradient_scaling_factor = edp/dp_cp pre-scale + AVG over edp).
+        self.grad_scale_denom = grad_scale_denom or self.dp_size
         self.grad_dtype = grad_dtype
         self.param_dtype = param_dtype
         self.device = device
@@ -136,13 +143,20 @@ class ParamAndGradBuffer:
     # -- per-bucket collectives ---------------------------------------------
 
     def _launch_grad_reduce(self, bucket: _Bucket, async_op: bool):
+        denom = self.grad_scale_denom
         if self.dp_size == 1:
+            # no collective, but the average denominator may still exceed 1
+            # (expert buffer with edp=1 while dp_cp>1)
+            if denom > 1:
+                bucket.grad_view.mul_(1.0 / denom)
             return
         # AVG happens in-collective on RCCL; gloo has no AVG -> pre-scale + SUM
         if self.ddp_config.average_in_collective and dist.get_backend(self.dp_group) == "nccl":
+            if denom != self.dp_size:
+                bucket.grad_view.mul_(self.dp_size / denom)
             op = dist.ReduceOp.AVG
         else:
-            bucket.grad_view.mul_(1.0 / self.dp_size)
+            bucket.grad_view.mul_(1.0 / denom)
             op = dist.ReduceOp.SUM
         if self.ddp_config.use_distributed_optimizer:
             bucket.comm_handle = dist.reduce_scatter_tensor(
@@ -233,9 +247,14 @@ class DistributedDataParallel(nn.Module):
                 self.param_to_buffer[p] = buf
         if expert_params:
             edp_group = G.get_grid().group("expert_dp") if G.grid_initialized() else None
+            dp_cp_size = (
+                dist.get_world_size(group=self.dp_group)
+                if (self.dp_group is not None and dist.is_initialized()) else 1
+            )
             ebuf = ParamAndGradBuffer(
                 list(reversed(expert_params)), ddp_config, edp_group,
                 expert_params[0].dtype, grad_dtype, device,
+                grad_scale_denom=dp_cp_size,
             )
             self.buffers.append(ebuf)
             for p in ebuf.param_index:

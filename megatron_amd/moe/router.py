@@ -19,20 +19,42 @@ import torch.nn.functional as F
 
 class AuxLossScaler(torch.autograd.Function):
     """Pass activations through; inject d(aux_loss) with the main-loss scale
-    in backward so router gradients flow without touching the loss plumbing."""
+    in backward so router gradients flow without touching the loss plumbing.
+
+    The scale (1/(ntok*num_microbatches)) is only known after the microbatch's
+    loss_func has run, i.e. after this Function's forward.  Under 1F1B the
+    forward of microbatch i+k runs before the backward of microbatch i, so a
+    plain class attribute would hand backward a *later* microbatch's scale
+    (wrong with variable per-microbatch token counts).  Instead forward parks
+    its ctx on a pending list and ``bind_scale`` — called by the scheduler
+    right after loss_func — stamps the current scale onto exactly the ctxs of
+    the microbatch that just finished its forward."""
 
     main_loss_backward_scale: float = 1.0
+    _pending: list = []
 
     @staticmethod
     def forward(ctx, output, aux_loss):
         ctx.save_for_backward(aux_loss)
+        ctx.scale = None
+        if torch.is_grad_enabled():
+            AuxLossScaler._pending.append(ctx)
         return output
 
     @staticmethod
     def backward(ctx, grad_output):
         (aux,) = ctx.saved_tensors
-        scale = AuxLossScaler.main_loss_backward_scale
+        scale = ctx.scale if ctx.scale is not None else AuxLossScaler.main_loss_backward_scale
         return grad_output, torch.full_like(aux, scale)
+
+    @staticmethod
+    def bind_scale(scale: float):
+        """Bind `scale` to every ctx created since the last call (i.e. this
+        microbatch's MoE layers) and remember it as the fallback."""
+        AuxLossScaler.main_loss_backward_scale = scale
+        for ctx in AuxLossScaler._pending:
+            ctx.scale = scale
+        AuxLossScaler._pending.clear()
 
 
 def group_limited_topk(

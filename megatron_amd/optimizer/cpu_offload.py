@@ -18,6 +18,7 @@ import torch
 from megatron_amd import ops
 from megatron_amd.config import OptimizerConfig
 from megatron_amd.optimizer.clip import (
+    split_grads_for_norm,
     clip_grads_by_total_norm,
     get_grad_norm,
     param_is_not_tensor_parallel_duplicate,
@@ -57,8 +58,8 @@ class CPUOffloadOptimizer(_BaseOptimizer):
             if g is None:
                 g = p.grad if p.grad is not None else torch.zeros_like(p)
             dev_grads.append(g.float())
-        norm_grads = [g for p, g in zip(self.params, dev_grads) if param_is_not_tensor_parallel_duplicate(p)]
-        total_norm = get_grad_norm(norm_grads)
+        dense_g, expert_g = split_grads_for_norm(self.params, dev_grads)
+        total_norm = get_grad_norm(dense_g, expert_grads=expert_g)
         if self.config.clip_grad > 0:
             clip_grads_by_total_norm(dev_grads, self.config.clip_grad, total_norm)
 

@@ -94,11 +94,20 @@ class DistributedOptimizer(_BaseOptimizer):
     def step(self):
         self.finish_grad_sync()
         grads = [seg.grad_view.float() for seg in self.segments]
-        norm_grads = [g for seg, g in zip(self.segments, grads) if seg.norm_ok]
-        extra = []
+        dense_g, expert_g = [], []
+        for seg, g in zip(self.segments, grads):
+            if not seg.norm_ok:
+                continue
+            is_exp = getattr(seg.param, "is_expert_parallel", False)
+            (expert_g if is_exp else dense_g).append(g)
+        extra, eextra = [], []
         if G.grid_initialized():
+            # shards are disjoint over dp_cp (dense) / edp (expert)
             extra.append(G.get_grid().group("dp_cp"))
-        total_norm = get_grad_norm(norm_grads, extra_groups=extra)
+            eextra.append(G.get_grid().group("expert_dp"))
+        total_norm = get_grad_norm(
+            dense_g, extra_groups=extra, expert_grads=expert_g, expert_extra_groups=eextra
+        )
         if self.config.clip_grad > 0:
             clip_grads_by_total_norm(grads, self.config.clip_grad, total_norm)
         self.step_count += 1

@@ -16,6 +16,7 @@ import torch.distributed as dist
 from megatron_amd import ops
 from megatron_amd.config import OptimizerConfig
 from megatron_amd.optimizer.clip import (
+    split_grads_for_norm,
     clip_grads_by_total_norm,
     get_grad_norm,
     param_is_not_tensor_parallel_duplicate,
@@ -105,8 +106,8 @@ class FP32Optimizer(_BaseOptimizer):
     def step(self):
         self.finish_grad_sync()
         grads = self._grads()
-        norm_grads = [g for p, g in zip(self.params, grads) if param_is_not_tensor_parallel_duplicate(p)]
-        total_norm = get_grad_norm(norm_grads)
+        dense_g, expert_g = split_grads_for_norm(self.params, grads)
+        total_norm = get_grad_norm(dense_g, expert_grads=expert_g)
         if self.config.clip_grad > 0:
             clip_grads_by_total_norm(grads, self.config.clip_grad, total_norm)
         self.step_count += 1
@@ -190,8 +191,8 @@ class MixedPrecisionOptimizer(_BaseOptimizer):
             self.grad_scaler.update(found_inf.item() > 0)
             if found_inf.item() > 0:
                 return False, None, None
-        norm_grads = [g for p, g in zip(self.params, grads) if param_is_not_tensor_parallel_duplicate(p)]
-        total_norm = get_grad_norm(norm_grads)
+        dense_g, expert_g = split_grads_for_norm(self.params, grads)
+        total_norm = get_grad_norm(dense_g, expert_grads=expert_g)
         if self.config.clip_grad > 0:
             clip_grads_by_total_norm(grads, self.config.clip_grad, total_norm)
         self.step_count += 1

@@ -158,6 +158,11 @@ class ParallelGrid:
         make("dp_cp", [False, True, True, False], gloo=True)
         make("mp", [True, True, False, True])  # model-parallel: tp x cp x pp
         make("tp_dp_cp", [True, True, True, False])
+        # grad-stats group: tp x pp ONLY.  Grads are already reduced (or
+        # reduce-scatter-sharded) over dp_cp, which spans cp, so including cp
+        # in the norm reduction would count every element cp times
+        # (reference uses tp x pp for grad stats too).
+        make("tp_pp", [True, False, False, True])
 
         # expert groups: factor the (tp*cp*dp) span as (etp, ep, edp), pp slowest
         espan_shape = [etp, ep, self.edp, pp]
@@ -165,6 +170,10 @@ class ParallelGrid:
         make("ep", [False, True, False, False], espan_shape)
         make("expert_dp", [False, False, True, False], espan_shape, gloo=True)
         make("etp_ep", [True, True, False, False], espan_shape)
+        # expert grad-stats group: etp x ep x pp — expert params are sharded
+        # over (etp, ep) and replicated over edp, so this span counts each
+        # expert-grad element exactly once.
+        make("etp_ep_pp", [True, True, False, True], espan_shape)
 
         # embedding group: first and last pp stage within each (dp, cp, tp) column
         emb_groups = []

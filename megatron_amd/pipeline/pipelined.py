@@ -33,11 +33,19 @@ def _fwd(forward_step_func, data_iterator, model, input_tensor, losses, num_toke
         scale = 1.0 / (max(int(num_tokens), 1) * num_microbatches)
         from megatron_amd.moe.router import AuxLossScaler
 
-        AuxLossScaler.main_loss_backward_scale = scale
+        AuxLossScaler.bind_scale(scale)
         out = loss * scale
         if config.grad_scale_func is not None:
             out = config.grad_scale_func(out)
         return out
+    # Non-last stages never see loss_func's token count, but their MoE aux
+    # losses still need the main-loss scale.  Estimate tokens from the stage
+    # output shape [s, b, h] (exact when the loss mask is all-ones).
+    if output.dim() >= 2:
+        est_tokens = output.shape[0] * output.shape[1]
+        from megatron_amd.moe.router import AuxLossScaler
+
+        AuxLossScaler.bind_scale(1.0 / (max(est_tokens, 1) * num_microbatches))
     return output
 
 

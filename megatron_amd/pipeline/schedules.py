@@ -48,7 +48,7 @@ def _forward_step(forward_step_func, data_iterator, model, losses, num_tokens_ac
     scale = 1.0 / (max(int(num_tokens), 1) * num_microbatches)
     from megatron_amd.moe.router import AuxLossScaler
 
-    AuxLossScaler.main_loss_backward_scale = scale
+    AuxLossScaler.bind_scale(scale)
     if config.grad_scale_func is not None:
         return config.grad_scale_func(loss * scale)
     return loss * scale

@@ -178,3 +178,73 @@ def test_rccl_registered_buffers_fallback_cpu():
     model.finish_grad_sync()
     for p_ in model.parameters():
         assert p_.main_grad is not None
+
+
+# --- expert-grad scaling + grad-norm group fixes (round-2 ADVICE items) -----
+
+
+def _expert_scale_case(rank, world):
+    """ep=world -> edp=1, dp_cp=world: expert grads must still be averaged
+    over dp_cp (reference expert_gradient_scaling_factor), not left unscaled."""
+    import torch.nn as nn
+
+    G.initialize_model_parallel(expert_parallel_size=world)
+
+    class Tiny(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.dense = nn.Parameter(torch.ones(8))
+            self.expert = nn.Parameter(torch.ones(8))
+            self.expert.is_expert_parallel = True
+
+        def forward(self, x):
+            return (x * self.dense).sum() + (x * self.expert).sum()
+
+    from megatron_amd.distributed.ddp import DistributedDataParallel
+
+    model = Tiny()
+    ddp = DistributedDataParallel(None, DDPConfig(overlap_grad_reduce=False), model)
+    x = torch.ones(8)
+    ddp(x).backward()
+    ddp.finish_grad_sync()
+    # dense: grad 1.0 on each of `world` dp ranks, averaged -> 1.0
+    assert_close(model.dense.main_grad, torch.ones(8), rtol=0, atol=1e-6)
+    # expert: edp group is size 1 (no collective) but the average denominator
+    # must be dp_cp = world -> 1/world
+    assert_close(model.expert.main_grad, torch.full((8,), 1.0 / world), rtol=0, atol=1e-6)
+
+
+def test_expert_grad_scaled_by_dp_cp():
+    spawn_dist(_expert_scale_case, 2)
+
+
+def _cp_grad_norm_case(rank, world):
+    """cp=world: grads (already dp_cp-reduced) are replicated over cp; the
+    norm reduction must NOT count them cp times."""
+    from megatron_amd.optimizer.clip import get_grad_norm
+
+    G.initialize_model_parallel(context_parallel_size=world)
+    g = torch.full((4,), 2.0)  # identical on every cp rank, ||g|| = 4.0
+    total = get_grad_norm([g])
+    assert abs(float(total) - 4.0) < 1e-5, f"cp-inflated norm: {float(total)}"
+
+
+def test_grad_norm_not_inflated_by_cp():
+    spawn_dist(_cp_grad_norm_case, 2)
+
+
+def _ep_grad_norm_case(rank, world):
+    """ep=world: each rank's local experts contribute; the norm must include
+    every rank's expert span (reduced over etp x ep x pp)."""
+    from megatron_amd.optimizer.clip import get_grad_norm
+
+    G.initialize_model_parallel(expert_parallel_size=world)
+    dense = torch.full((4,), 2.0)       # replicated -> ||.|| = 4
+    expert = torch.full((4,), float(rank + 1))  # rank0: 2.0, rank1: 4.0 norms
+    total = get_grad_norm([dense], expert_grads=[expert])
+    # total^2 = 16 + (4*1 + 4*4) = 36 -> 6.0
+    assert abs(float(total) - 6.0) < 1e-5, f"expert norm wrong: {float(total)}"
+
+
+def test_grad_norm_includes_all_ep_ranks():
+    spawn_dist(_ep_grad_norm_case, 2)
