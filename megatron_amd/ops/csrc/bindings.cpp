@@ -18,6 +18,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor out, torch::Tensor lse,
                                     bool causal, double scale, long window);
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor target, long vocab_start);
+void ce_bwd(torch::Tensor logits, torch::Tensor target, torch::Tensor row_max,
+            torch::Tensor row_inv_sumexp, torch::Tensor grad_out, long vocab_start);
 std::vector<int64_t> symm_ipc_handle(torch::Tensor buf);
 int64_t symm_open_handle(std::vector<int64_t> bytes);
 void symm_close_handle(int64_t ptr);
@@ -37,6 +40,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_gemm_accum", &wgrad_gemm_accum);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
   m.def("symm_ipc_handle", &symm_ipc_handle);
   m.def("symm_open_handle", &symm_open_handle);
   m.def("symm_close_handle", &symm_close_handle);

@@ -2,13 +2,19 @@
 
 Capability analog of reference megatron/core/tensor_parallel/cross_entropy.py
 (:13 _VocabParallelCrossEntropy, :213 vocab_parallel_cross_entropy) and the
-fused variant fusions/fused_cross_entropy.py.
+fused variant fusions/fused_cross_entropy.py:13-65.
 
-Stages (each a single HIP kernel when the native extension is built;
-torch fallback otherwise), with two small TP all-reduces between:
-  1. per-row max over the local vocab shard              -> all-reduce(MAX)
-  2. exp-sum + target-logit pick over the local shard    -> all-reduce(SUM)
-  3. loss + in-place softmax-grad preparation
+Native path (GPU, label_smoothing=0): ONE online HIP kernel pass over the
+bf16 logits in forward (running (max, sumexp) merge + target pick —
+`ops/csrc/cross_entropy.hip`), two tiny [T] all-reduces over TP (row max,
+then rebased sumexp + target logit packed into one collective), and ONE
+backward kernel pass that writes the bf16 softmax-grad IN-PLACE over the
+logits buffer.  Nothing of shape [T, V] is ever materialized beyond the
+logits themselves (the previous torch-composed path kept an fp32 softmax
+copy alive from forward to backward: 8.4 GB at T=16k, V=128k).
+
+Fallback (CPU tensors, or label_smoothing > 0): composed torch ops with
+the same two-all-reduce algorithm.
 """
 
 from __future__ import annotations
@@ -19,7 +25,51 @@ import torch.distributed as dist
 from megatron_amd.parallel import grid as G
 
 
+class _VocabParallelCrossEntropyNative(torch.autograd.Function):
+    """Fused kernel path: loss [T] fp32 from logits [T, V/tp] bf16."""
+
+    @staticmethod
+    def forward(ctx, logits, target):
+        from megatron_amd import ops
+
+        group = G.get_tensor_model_parallel_group()
+        tp = G.get_tensor_model_parallel_world_size()
+        rank = G.get_tensor_model_parallel_rank()
+        part_vocab = logits.size(-1)
+        vocab_start = rank * part_vocab
+
+        logits = logits.contiguous()
+        m_loc, s_loc, pred_raw = ops._C.ce_fwd(logits, target, vocab_start)
+        if tp > 1:
+            m = m_loc.clone()
+            dist.all_reduce(m, op=dist.ReduceOp.MAX, group=group)
+            # rebase local sumexp from the local max to the global max,
+            # then one packed SUM collective for (sumexp, target_logit)
+            sp = torch.stack((s_loc * torch.exp(m_loc - m), pred_raw))
+            dist.all_reduce(sp, op=dist.ReduceOp.SUM, group=group)
+            s, pred_raw = sp[0], sp[1]
+        else:
+            m, s = m_loc, s_loc
+        loss = torch.log(s) - (pred_raw - m)
+        ctx.save_for_backward(logits, target, m, s)
+        ctx.vocab_start = vocab_start
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        from megatron_amd import ops
+
+        logits, target, m, s = ctx.saved_tensors
+        # write (softmax - onehot) * grad_output into the logits buffer —
+        # the logits are dead after this point (the producing GEMM's
+        # backward needs its own inputs, not its output)
+        ops._C.ce_bwd(logits, target, m, s.reciprocal(), grad_output.float(), ctx.vocab_start)
+        return logits, None
+
+
 class _VocabParallelCrossEntropy(torch.autograd.Function):
+    """Composed-torch fallback (CPU / label smoothing)."""
+
     @staticmethod
     def forward(ctx, logits, target, label_smoothing=0.0):
         # logits: [tokens, V/tp] (any leading dims flattened by caller), target: [tokens]
@@ -85,6 +135,19 @@ class _VocabParallelCrossEntropy(torch.autograd.Function):
 
 def vocab_parallel_cross_entropy(logits: torch.Tensor, target: torch.Tensor, label_smoothing: float = 0.0):
     """logits [s, b, V/tp], target [s, b] -> loss [s, b] (fp32)."""
+    from megatron_amd import ops
+
     s, b, v = logits.shape
-    loss = _VocabParallelCrossEntropy.apply(logits.reshape(s * b, v), target.reshape(-1), label_smoothing)
+    flat = logits.reshape(s * b, v)
+    tgt = target.reshape(-1)
+    if (
+        label_smoothing == 0.0
+        and flat.is_cuda
+        and flat.dtype == torch.bfloat16
+        and v % 8 == 0
+        and ops.has_native()
+    ):
+        loss = _VocabParallelCrossEntropyNative.apply(flat, tgt)
+    else:
+        loss = _VocabParallelCrossEntropy.apply(flat, tgt, label_smoothing)
     return loss.view(s, b)

@@ -20,6 +20,7 @@ SRC = [
     "megatron_amd/ops/csrc/rope.hip",
     "megatron_amd/ops/csrc/adamw.hip",
     "megatron_amd/ops/csrc/wgrad.hip",
+    "megatron_amd/ops/csrc/cross_entropy.hip",
     "megatron_amd/ops/csrc/attention_fwd.hip",
     "megatron_amd/ops/csrc/attention_bwd.hip",
     "megatron_amd/ops/csrc/symm_allreduce.hip",

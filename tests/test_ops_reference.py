@@ -128,3 +128,33 @@ def test_mrope_sections_use_their_position_rows():
     assert torch.all(f[:, 8:20] > 0) and torch.all(f[:, 20:] > 0)
     f2 = mrope_freqs(pos * 2, dim, mrope_section=sec)
     assert torch.allclose(f2[:, 8:], f[:, 8:] * 2)
+
+
+def test_ce_native_tp_merge_math():
+    """CPU check of the fused-CE TP merge algebra (cross_entropy.py native
+    path): per-shard online (max, sumexp, raw-target-logit) stats merged via
+    MAX + rebased SUM must reproduce full-softmax cross entropy."""
+    import torch
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    T, V = 64, 256
+    logits = torch.randn(T, V) * 4
+    target = torch.randint(0, V, (T,))
+    ms, ss, ps = [], [], []
+    for r, lg in enumerate(logits.chunk(2, dim=-1)):
+        vp = lg.shape[-1]
+        m_loc = lg.max(-1).values
+        s_loc = torch.exp(lg - m_loc[:, None]).sum(-1)
+        mt = target - r * vp
+        in_sh = (mt >= 0) & (mt < vp)
+        pred = torch.where(
+            in_sh, lg.gather(-1, mt.clamp(0, vp - 1)[:, None]).squeeze(-1), torch.zeros(T)
+        )
+        ms.append(m_loc); ss.append(s_loc); ps.append(pred)
+    m = torch.maximum(ms[0], ms[1])                       # all-reduce MAX
+    s = ss[0] * torch.exp(ms[0] - m) + ss[1] * torch.exp(ms[1] - m)  # rebased SUM
+    pred_raw = ps[0] + ps[1]                              # SUM
+    loss = torch.log(s) - (pred_raw - m)
+    ref = F.cross_entropy(logits, target, reduction="none")
+    assert torch.allclose(loss, ref, rtol=1e-5, atol=1e-5)
