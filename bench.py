@@ -130,9 +130,9 @@ def main():
     ddp_cfg = DDPConfig(
         grad_reduce_in_fp32=True,
         use_distributed_optimizer=not args.no_dist_opt, bucket_size=40_000_000,
-        # graph replay re-runs captured kernels but not python grad-ready
-        # callbacks: bucket overlap would deadlock, so reduce at step end
-        overlap_grad_reduce=not args.hip_graphs,
+        # graphs + overlap coexist: graphed backward fires the DDP grad-ready
+        # callbacks after each layer replay (hip_graphs.py)
+        overlap_grad_reduce=True,
     )
 
     def provider(config, pre_process=True, post_process=True, vp_stage=None):
@@ -168,14 +168,21 @@ def main():
         one_step()
     barrier_sync()
     if args.hip_graphs:
-        # capture after warmup at a step boundary (no live autograd graphs)
+        # capture after warmup at a step boundary (no live autograd graphs);
+        # under no_sync so capture-time ready-callbacks cannot launch reduces
+        import contextlib as _ctx
+
         from megatron_amd.transformer.hip_graphs import capture_block_hip_graphs
 
         core = chunks[0].module if hasattr(chunks[0], "module") else chunks[0]
         sample = torch.randn(args.seq_len, args.micro_batch_size, cfg.hidden_size,
                              device=device, dtype=cfg.params_dtype)
         freqs = core._rotary_freqs(args.seq_len, device)
-        n = capture_block_hip_graphs(core.decoder, sample, rotary_freqs=freqs)
+        with _ctx.ExitStack() as stack:
+            for ch in chunks:
+                if hasattr(ch, "no_sync"):
+                    stack.enter_context(ch.no_sync())
+            n = capture_block_hip_graphs(core.decoder, sample, rotary_freqs=freqs)
         if rank == 0:
             print(f"# captured {n} layer hipGraphs", flush=True)
         one_step()  # one replay-path warmup step

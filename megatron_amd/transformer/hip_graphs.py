@@ -95,6 +95,17 @@ class _GraphedLayerFn(torch.autograd.Function):
         g.g_bwd.replay()
         dx = g.static_grad_input.clone() if g.static_grad_input is not None else None
         dps = tuple(p.clone() if p is not None else None for p in g.static_param_grads)
+        # overlap_grad_reduce coexistence: params whose wgrad is fused into
+        # the graph (static grad None -> main_grad accumulated in-replay)
+        # never reach autograd's post-accumulate hooks, so fire their DDP
+        # grad-ready callbacks here.  The replay is enqueued on the current
+        # stream and torch.distributed orders its RCCL stream after it, so
+        # the bucket reduce launched by the callback sees the final grads.
+        for p, sg in zip(g.params, g.static_param_grads):
+            if sg is None:
+                cb = getattr(p, "_ddp_grad_ready_cb", None)
+                if cb is not None:
+                    cb()
         return (None, dx) + dps
 
 
