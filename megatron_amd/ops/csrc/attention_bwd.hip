@@ -469,6 +469,613 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkv_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// v3 kernels: 2x register blocking (32 output rows per wave, 256 per block).
+// Same technique stack as v2 plus the fwd-v2 lesson: every LDS A/B-operand
+// read now feeds 2 MFMAs instead of 1 (the kernels were LDS-read-bound, so
+// halving bytes-per-FLOP is the first-order lever).  S/dS tiles are
+// processed per 16-row q sub-tile immediately (backward has no online
+// softmax - lse/drow are known), keeping the transient register footprint
+// flat; packed P/dS live as 2x bf16 dwords until the phase-2 MFMAs.
+// ---------------------------------------------------------------------------
+
+// dq v3: 8 waves x 32 q rows (qt in {0,1}), kv tiles of 64.
+template <int D>
+__global__ __launch_bounds__(512) void attn_bwd_dq_kernel_v3(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    const bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ drow,
+    bf16* __restrict__ dq,
+    int sq, int skv, int b, int hq, int hkv, float scale, int causal, int window) {
+  constexpr int KVBLK = 64;
+  constexpr int ND = D / 16, NKD = D / 32;
+  constexpr int KCH = D / 8;
+  constexpr int SREG = KVBLK * D / 8 / 512;
+
+  __shared__ __align__(16) short K_lds[KVBLK][D];
+  __shared__ __align__(16) short V_lds[KVBLK][D];
+  __shared__ __align__(16) short Kt_lds[D][KVBLK];
+
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int g = lane >> 4, c = lane & 15;
+  const int q0 = blockIdx.x * 256;
+  const int batch = blockIdx.y / hq, head = blockIdx.y % hq;
+  const int kv_head = head / (hq / hkv);
+  const long q_srow = (long)b * hq * D, k_srow = (long)b * hkv * D;
+  const bf16* qbase = q + ((long)batch * hq + head) * D;
+  const bf16* dobase = dout + ((long)batch * hq + head) * D;
+  const bf16* kbase = k + ((long)batch * hkv + kv_head) * D;
+  const bf16* vbase = v + ((long)batch * hkv + kv_head) * D;
+  bf16* dqbase = dq + ((long)batch * hq + head) * D;
+  const float* lse_row = lse + ((long)batch * hq + head) * sq;
+  const float* dr_row = drow + ((long)batch * hq + head) * sq;
+
+  const int row0 = q0 + wid * 32;
+  const int off = skv - sq;
+
+  bf16x8 qa[2][NKD], doa[2][NKD];
+  float my_lse_l2[2], my_dr_s[2];
+#pragma unroll
+  for (int qt = 0; qt < 2; ++qt) {
+    int qrow = min(row0 + qt * 16 + c, sq - 1);
+    const bf16* qr = qbase + (long)qrow * q_srow;
+    const bf16* dor = dobase + (long)qrow * q_srow;
+#pragma unroll
+    for (int kd = 0; kd < NKD; ++kd) {
+      short8 t1 = *reinterpret_cast<const short8*>(qr + kd * 32 + g * 8);
+      qa[qt][kd] = *reinterpret_cast<bf16x8*>(&t1);
+      short8 t2 = *reinterpret_cast<const short8*>(dor + kd * 32 + g * 8);
+      doa[qt][kd] = *reinterpret_cast<bf16x8*>(&t2);
+    }
+    my_lse_l2[qt] = lse_row[qrow] * LOG2E;
+    my_dr_s[qt] = dr_row[qrow] * scale;
+  }
+  const float c1 = scale * LOG2E;
+
+  f32x4 dq_acc[2][ND];
+#pragma unroll
+  for (int qt = 0; qt < 2; ++qt)
+#pragma unroll
+    for (int n = 0; n < ND; ++n) dq_acc[qt][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  int t_end = causal ? min((q0 + 255 + off) / KVBLK, (skv - 1) / KVBLK) : (skv - 1) / KVBLK;
+  int t_start = (window > 0) ? max(0, (q0 + off - window + 1) / KVBLK) : 0;
+
+  int st_row[SREG], st_c8[SREG], sv_row[SREG], sv_c8[SREG];
+#pragma unroll
+  for (int i = 0; i < SREG; ++i) {
+    int idx = (int)threadIdx.x + i * 512;
+    st_row[i] = idx / KCH;
+    st_c8[i] = idx % KCH;
+    sv_row[i] = idx & (KVBLK - 1);
+    sv_c8[i] = idx / KVBLK;
+  }
+
+  short8 kreg[SREG], vreg[SREG], ktreg[SREG];
+  auto stage_loads = [&](int k0) {
+#pragma unroll
+    for (int i = 0; i < SREG; ++i) {
+      int krow = k0 + st_row[i];
+      kreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      ktreg[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (krow < skv) {
+        kreg[i] = *reinterpret_cast<const short8*>(kbase + (long)krow * k_srow + st_c8[i] * 8);
+        vreg[i] = *reinterpret_cast<const short8*>(vbase + (long)krow * k_srow + st_c8[i] * 8);
+      }
+      int trow = k0 + sv_row[i];
+      if (trow < skv)
+        ktreg[i] = *reinterpret_cast<const short8*>(kbase + (long)trow * k_srow + sv_c8[i] * 8);
+    }
+  };
+  auto stage_writes = [&]() {
+#pragma unroll
+    for (int i = 0; i < SREG; ++i) {
+      *reinterpret_cast<short8*>(&K_lds[st_row[i]][swzb(st_row[i], st_c8[i]) * 8]) = kreg[i];
+      *reinterpret_cast<short8*>(&V_lds[st_row[i]][swzb(st_row[i], st_c8[i]) * 8]) = vreg[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = sv_c8[i] * 8 + j;
+        Kt_lds[d][swzb(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = ktreg[i][j];
+      }
+    }
+  };
+
+  stage_loads(t_start * KVBLK);
+  stage_writes();
+  __syncthreads();
+
+  for (int t = t_start; t <= t_end; ++t) {
+    const int k0 = t * KVBLK;
+    const bool have_next = (t + 1 <= t_end);
+    if (have_next) stage_loads(k0 + KVBLK);
+
+    bool wave_skip = (causal && k0 > row0 + 31 + off) ||
+                     (window > 0 && k0 + KVBLK - 1 < row0 + off - window + 1);
+    if (!wave_skip) {
+      // per 16-row kv sub-tile: S^T/dP^T MFMAs -> dS -> packed pairs
+      int pk[KVBLK / 16][2][2];  // [n][qt][pair]
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int n = 0; n < KVBLK / 16; ++n) {
+        f32x4 st[2], dpt[2];
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) {
+          st[qt] = f32x4{0.f, 0.f, 0.f, 0.f};
+          dpt[qt] = f32x4{0.f, 0.f, 0.f, 0.f};
+        }
+        int krow = n * 16 + c;
+#pragma unroll
+        for (int kd = 0; kd < NKD; ++kd) {
+          short8 t1 = *reinterpret_cast<const short8*>(&K_lds[krow][swzb(krow, kd * 4 + g) * 8]);
+          bf16x8 ak = *reinterpret_cast<bf16x8*>(&t1);
+          short8 t2 = *reinterpret_cast<const short8*>(&V_lds[krow][swzb(krow, kd * 4 + g) * 8]);
+          bf16x8 av = *reinterpret_cast<bf16x8*>(&t2);
+#pragma unroll
+          for (int qt = 0; qt < 2; ++qt) {
+            st[qt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak, qa[qt][kd], st[qt], 0, 0, 0);
+            dpt[qt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, doa[qt][kd], dpt[qt], 0, 0, 0);
+          }
+        }
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) {
+          const int qrow = row0 + qt * 16 + c;
+          float ds[4];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int kvrow = k0 + n * 16 + g * 4 + r;
+            bool ok = (kvrow < skv) && (qrow < sq);
+            if (causal) ok &= (kvrow <= qrow + off);
+            if (window > 0) ok &= (kvrow > qrow + off - window);
+            float pt = ok ? exp2f(__builtin_fmaf(st[qt][r], c1, -my_lse_l2[qt])) : 0.f;
+            ds[r] = pt * __builtin_fmaf(dpt[qt][r], scale, -my_dr_s[qt]);
+          }
+          pk[n][qt][0] = pack_bf16x2(ds[0], ds[1]);
+          pk[n][qt][1] = pack_bf16x2(ds[2], ds[3]);
+        }
+      }
+
+      // dq += dS K : one Kt load feeds both qt MFMAs
+#pragma unroll
+      for (int ks2 = 0; ks2 < KVBLK / 32; ++ks2) {
+        bf16x8 a[2];
+#pragma unroll
+        for (int qt = 0; qt < 2; ++qt) {
+          int av4[4];
+          dfrag_to_at(pk[2 * ks2][qt], pk[2 * ks2 + 1][qt], av4);
+          int4 av = make_int4(av4[0], av4[1], av4[2], av4[3]);
+          a[qt] = *reinterpret_cast<bf16x8*>(&av);
+        }
+#pragma unroll
+        for (int n = 0; n < ND; ++n) {
+          int vrow = n * 16 + c;
+          short8 t3 = *reinterpret_cast<const short8*>(&Kt_lds[vrow][swzb(vrow, ks2 * 4 + g) * 8]);
+          bf16x8 bK = *reinterpret_cast<bf16x8*>(&t3);
+#pragma unroll
+          for (int qt = 0; qt < 2; ++qt)
+            dq_acc[qt][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[qt], bK, dq_acc[qt][n], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+
+    if (have_next) {
+      __syncthreads();
+      __builtin_amdgcn_s_waitcnt(WAIT_VM0);
+      __builtin_amdgcn_sched_barrier(0);
+      stage_writes();
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int qt = 0; qt < 2; ++qt)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = row0 + qt * 16 + g * 4 + r;
+      if (row >= sq) continue;
+      bf16* dqr = dqbase + (long)row * q_srow;
+#pragma unroll
+      for (int n = 0; n < ND; ++n) dqr[n * 16 + c] = f2bf(dq_acc[qt][n][r]);
+    }
+}
+
+// dv v3: 8 waves x 32 kv rows (kvt in {0,1}), q tiles of 64, GQA heads looped.
+// dv = P^T dO only (dk split into its own kernel: the combined 32-row
+// version needs 2x f32 accumulators and spills 261 VGPRs; the split costs a
+// recomputed S GEMM but keeps both kernels at 2 waves/SIMD with 0 spills).
+template <int D>
+__global__ __launch_bounds__(512) void attn_bwd_dv_kernel_v3(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    const bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ drow,
+    bf16* __restrict__ dv,
+    int sq, int skv, int b, int hq, int hkv, float scale, int causal, int window) {
+  constexpr int QBLK = 64;
+  constexpr int ND = D / 16, NKD = D / 32;
+  constexpr int KCH = D / 8;
+  constexpr int SREG = QBLK * D / 8 / 512;
+
+  __shared__ __align__(16) short Q_lds[QBLK][D];
+  __shared__ __align__(16) short dOt_lds[D][QBLK];
+  __shared__ float lse_s[QBLK];
+
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int g = lane >> 4, c = lane & 15;
+  const int k0 = blockIdx.x * 256;
+  const int batch = blockIdx.y / hkv, kv_head = blockIdx.y % hkv;
+  const int rep = hq / hkv;
+  const long q_srow = (long)b * hq * D, k_srow = (long)b * hkv * D;
+  const bf16* kbase = k + ((long)batch * hkv + kv_head) * D;
+  bf16* dvbase = dv + ((long)batch * hkv + kv_head) * D;
+
+  const int krow0 = k0 + wid * 32;
+  const int off = skv - sq;
+  const float c1 = scale * LOG2E;
+
+  bf16x8 ka[2][NKD];
+#pragma unroll
+  for (int kvt = 0; kvt < 2; ++kvt) {
+    int krow = min(krow0 + kvt * 16 + c, skv - 1);
+    const bf16* kr = kbase + (long)krow * k_srow;
+#pragma unroll
+    for (int kd = 0; kd < NKD; ++kd) {
+      short8 t1 = *reinterpret_cast<const short8*>(kr + kd * 32 + g * 8);
+      ka[kvt][kd] = *reinterpret_cast<bf16x8*>(&t1);
+    }
+  }
+
+  f32x4 dv_acc[2][ND];
+#pragma unroll
+  for (int kvt = 0; kvt < 2; ++kvt)
+#pragma unroll
+    for (int n = 0; n < ND; ++n) dv_acc[kvt][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  int tq_start = causal ? max(0, (k0 - off) / QBLK) : 0;
+  int tq_end = (sq - 1) / QBLK;
+  if (window > 0) tq_end = min(tq_end, (k0 + 255 - off + window - 1) / QBLK);
+
+  int st_row[SREG], st_c8[SREG], sv_row[SREG], sv_c8[SREG];
+#pragma unroll
+  for (int i = 0; i < SREG; ++i) {
+    int idx = (int)threadIdx.x + i * 512;
+    st_row[i] = idx / KCH;
+    st_c8[i] = idx % KCH;
+    sv_row[i] = idx & (QBLK - 1);
+    sv_c8[i] = idx / QBLK;
+  }
+
+  for (int hg = 0; hg < rep; ++hg) {
+    const int head = kv_head * rep + hg;
+    const bf16* qbase = q + ((long)batch * hq + head) * D;
+    const bf16* dobase = dout + ((long)batch * hq + head) * D;
+    const float* lse_row = lse + ((long)batch * hq + head) * sq;
+
+    short8 qrm[SREG], dotr[SREG];
+    auto stage_loads = [&](int qt0) {
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+        int r1 = qt0 + st_row[i];
+        qrm[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        dotr[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (r1 < sq)
+          qrm[i] = *reinterpret_cast<const short8*>(qbase + (long)r1 * q_srow + st_c8[i] * 8);
+        int r2 = qt0 + sv_row[i];
+        if (r2 < sq)
+          dotr[i] = *reinterpret_cast<const short8*>(dobase + (long)r2 * q_srow + sv_c8[i] * 8);
+      }
+    };
+    auto stage_writes = [&](int qt0) {
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+        *reinterpret_cast<short8*>(&Q_lds[st_row[i]][swzb(st_row[i], st_c8[i]) * 8]) = qrm[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = sv_c8[i] * 8 + j;
+          dOt_lds[d][swzb(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = dotr[i][j];
+        }
+      }
+      if (threadIdx.x < QBLK) {
+        int qi = min(qt0 + (int)threadIdx.x, sq - 1);
+        lse_s[threadIdx.x] = lse_row[qi] * LOG2E;
+      }
+    };
+
+    stage_loads(tq_start * QBLK);
+    stage_writes(tq_start * QBLK);
+    __syncthreads();
+
+    for (int t = tq_start; t <= tq_end; ++t) {
+      const int qt0 = t * QBLK;
+      const bool have_next = (t + 1 <= tq_end);
+      if (have_next) stage_loads(qt0 + QBLK);
+
+      bool wave_skip = (causal && qt0 + QBLK - 1 < krow0 - off) ||
+                       (window > 0 && qt0 >= krow0 + 31 - off + window);
+      if (!wave_skip) {
+        int pkP[QBLK / 16][2][2];  // [n][kvt][pair]
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int n = 0; n < QBLK / 16; ++n) {
+          f32x4 st[2];
+          st[0] = f32x4{0.f, 0.f, 0.f, 0.f};
+          st[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+          int qrow = n * 16 + c;
+#pragma unroll
+          for (int kd = 0; kd < NKD; ++kd) {
+            short8 t1 = *reinterpret_cast<const short8*>(&Q_lds[qrow][swzb(qrow, kd * 4 + g) * 8]);
+            bf16x8 aq = *reinterpret_cast<bf16x8*>(&t1);
+#pragma unroll
+            for (int kvt = 0; kvt < 2; ++kvt)
+              st[kvt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq, ka[kvt][kd], st[kvt], 0, 0, 0);
+          }
+#pragma unroll
+          for (int kvt = 0; kvt < 2; ++kvt) {
+            const int kvcol = krow0 + kvt * 16 + c;
+            float pv[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              int qi = n * 16 + g * 4 + r;
+              int qrow_ = qt0 + qi;
+              bool ok = (kvcol < skv) && (qrow_ < sq);
+              if (causal) ok &= (kvcol <= qrow_ + off);
+              if (window > 0) ok &= (kvcol > qrow_ + off - window);
+              pv[r] = ok ? exp2f(__builtin_fmaf(st[kvt][r], c1, -lse_s[qi])) : 0.f;
+            }
+            pkP[n][kvt][0] = pack_bf16x2(pv[0], pv[1]);
+            pkP[n][kvt][1] = pack_bf16x2(pv[2], pv[3]);
+          }
+        }
+
+        // dV += P^T dO : one dOt load feeds both kvt MFMAs
+#pragma unroll
+        for (int ks2 = 0; ks2 < QBLK / 32; ++ks2) {
+          bf16x8 aP[2];
+#pragma unroll
+          for (int kvt = 0; kvt < 2; ++kvt) {
+            int avP[4];
+            dfrag_to_at(pkP[2 * ks2][kvt], pkP[2 * ks2 + 1][kvt], avP);
+            int4 aP4 = make_int4(avP[0], avP[1], avP[2], avP[3]);
+            aP[kvt] = *reinterpret_cast<bf16x8*>(&aP4);
+          }
+#pragma unroll
+          for (int n = 0; n < ND; ++n) {
+            int drow_ = n * 16 + c;
+            short8 t3 = *reinterpret_cast<const short8*>(&dOt_lds[drow_][swzb(drow_, ks2 * 4 + g) * 8]);
+            bf16x8 bdo = *reinterpret_cast<bf16x8*>(&t3);
+#pragma unroll
+            for (int kvt = 0; kvt < 2; ++kvt)
+              dv_acc[kvt][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aP[kvt], bdo, dv_acc[kvt][n], 0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+      }
+
+      if (have_next) {
+        __syncthreads();
+        __builtin_amdgcn_s_waitcnt(WAIT_VM0);
+        __builtin_amdgcn_sched_barrier(0);
+        stage_writes(qt0 + QBLK);
+        __syncthreads();
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int kvt = 0; kvt < 2; ++kvt)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = krow0 + kvt * 16 + g * 4 + r;
+      if (row >= skv) continue;
+      bf16* dvr = dvbase + (long)row * k_srow;
+#pragma unroll
+      for (int n = 0; n < ND; ++n) dvr[n * 16 + c] = f2bf(dv_acc[kvt][n][r]);
+    }
+}
+
+// dk v3: 8 waves x 32 kv rows, dk = dS^T Q with dS = P*(scale*dP - drow).
+// A-fragments (Q, dO rows) are read DIRECTLY FROM GLOBAL in the S/dP phase:
+// all 8 waves read the same 32 KB tile, so after the first touch the reads
+// are L1/L2 hits, and dropping the Q_lds/dO_lds tiles + their staging
+// registers is what keeps this kernel at 2 waves/SIMD without spills (the
+// LDS-staged variant spilled 126 VGPRs).  Only the transposed Qt tile (the
+// phase-2 B operand) stays in LDS.
+template <int D>
+__global__ __launch_bounds__(512) void attn_bwd_dk_kernel_v3(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    const bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ drow,
+    bf16* __restrict__ dk,
+    int sq, int skv, int b, int hq, int hkv, float scale, int causal, int window) {
+  constexpr int QBLK = 64;
+  constexpr int ND = D / 16, NKD = D / 32;
+  constexpr int SREG = QBLK * D / 8 / 512;
+
+  __shared__ __align__(16) short Qt_lds[D][QBLK];
+  __shared__ float lse_s[QBLK];
+  __shared__ float dr_s[QBLK];
+
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int g = lane >> 4, c = lane & 15;
+  const int k0 = blockIdx.x * 256;
+  const int batch = blockIdx.y / hkv, kv_head = blockIdx.y % hkv;
+  const int rep = hq / hkv;
+  const long q_srow = (long)b * hq * D, k_srow = (long)b * hkv * D;
+  const bf16* kbase = k + ((long)batch * hkv + kv_head) * D;
+  const bf16* vbase = v + ((long)batch * hkv + kv_head) * D;
+  bf16* dkbase = dk + ((long)batch * hkv + kv_head) * D;
+
+  const int krow0 = k0 + wid * 32;
+  const int off = skv - sq;
+  const float c1 = scale * LOG2E;
+
+  bf16x8 ka[2][NKD], va[2][NKD];
+#pragma unroll
+  for (int kvt = 0; kvt < 2; ++kvt) {
+    int krow = min(krow0 + kvt * 16 + c, skv - 1);
+    const bf16* kr = kbase + (long)krow * k_srow;
+    const bf16* vr = vbase + (long)krow * k_srow;
+#pragma unroll
+    for (int kd = 0; kd < NKD; ++kd) {
+      short8 t1 = *reinterpret_cast<const short8*>(kr + kd * 32 + g * 8);
+      ka[kvt][kd] = *reinterpret_cast<bf16x8*>(&t1);
+      short8 t2 = *reinterpret_cast<const short8*>(vr + kd * 32 + g * 8);
+      va[kvt][kd] = *reinterpret_cast<bf16x8*>(&t2);
+    }
+  }
+
+  f32x4 dk_acc[2][ND];
+#pragma unroll
+  for (int kvt = 0; kvt < 2; ++kvt)
+#pragma unroll
+    for (int n = 0; n < ND; ++n) dk_acc[kvt][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  int tq_start = causal ? max(0, (k0 - off) / QBLK) : 0;
+  int tq_end = (sq - 1) / QBLK;
+  if (window > 0) tq_end = min(tq_end, (k0 + 255 - off + window - 1) / QBLK);
+
+  int sv_row[SREG], sv_c8[SREG];
+#pragma unroll
+  for (int i = 0; i < SREG; ++i) {
+    int idx = (int)threadIdx.x + i * 512;
+    sv_row[i] = idx & (QBLK - 1);
+    sv_c8[i] = idx / QBLK;
+  }
+
+  for (int hg = 0; hg < rep; ++hg) {
+    const int head = kv_head * rep + hg;
+    const bf16* qbase = q + ((long)batch * hq + head) * D;
+    const bf16* dobase = dout + ((long)batch * hq + head) * D;
+    const float* lse_row = lse + ((long)batch * hq + head) * sq;
+    const float* dr_row = drow + ((long)batch * hq + head) * sq;
+
+    short8 qtr[SREG];
+    auto stage_loads = [&](int qt0) {
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+        qtr[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        int r2 = qt0 + sv_row[i];
+        if (r2 < sq)
+          qtr[i] = *reinterpret_cast<const short8*>(qbase + (long)r2 * q_srow + sv_c8[i] * 8);
+      }
+    };
+    auto stage_writes = [&](int qt0) {
+#pragma unroll
+      for (int i = 0; i < SREG; ++i) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = sv_c8[i] * 8 + j;
+          Qt_lds[d][swzb(d, sv_row[i] >> 3) * 8 + (sv_row[i] & 7)] = qtr[i][j];
+        }
+      }
+      if (threadIdx.x < QBLK) {
+        int qi = min(qt0 + (int)threadIdx.x, sq - 1);
+        lse_s[threadIdx.x] = lse_row[qi] * LOG2E;
+        dr_s[threadIdx.x] = dr_row[qi] * scale;
+      }
+    };
+
+    stage_loads(tq_start * QBLK);
+    stage_writes(tq_start * QBLK);
+    __syncthreads();
+
+    for (int t = tq_start; t <= tq_end; ++t) {
+      const int qt0 = t * QBLK;
+      const bool have_next = (t + 1 <= tq_end);
+      if (have_next) stage_loads(qt0 + QBLK);
+
+      bool wave_skip = (causal && qt0 + QBLK - 1 < krow0 - off) ||
+                       (window > 0 && qt0 >= krow0 + 31 - off + window);
+      if (!wave_skip) {
+        // per n-pair: S/dP MFMAs -> dS packs for n and n+1, then immediately
+        // the ks2 = n/2 slice of dK += dS^T Q (keeps only 2 pkD slots live)
+        int pkD[2][2][2];  // [n&1][kvt][pair]
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int n = 0; n < QBLK / 16; ++n) {
+          f32x4 st[2], dpt[2];
+#pragma unroll
+          for (int kvt = 0; kvt < 2; ++kvt) {
+            st[kvt] = f32x4{0.f, 0.f, 0.f, 0.f};
+            dpt[kvt] = f32x4{0.f, 0.f, 0.f, 0.f};
+          }
+          int qrow = min(qt0 + n * 16 + c, sq - 1);
+          const bf16* qr = qbase + (long)qrow * q_srow;
+          const bf16* dor = dobase + (long)qrow * q_srow;
+#pragma unroll
+          for (int kd = 0; kd < NKD; ++kd) {
+            short8 t1 = *reinterpret_cast<const short8*>(qr + kd * 32 + g * 8);
+            bf16x8 aq = *reinterpret_cast<bf16x8*>(&t1);
+            short8 t2 = *reinterpret_cast<const short8*>(dor + kd * 32 + g * 8);
+            bf16x8 ado = *reinterpret_cast<bf16x8*>(&t2);
+#pragma unroll
+            for (int kvt = 0; kvt < 2; ++kvt) {
+              st[kvt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq, ka[kvt][kd], st[kvt], 0, 0, 0);
+              dpt[kvt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado, va[kvt][kd], dpt[kvt], 0, 0, 0);
+            }
+          }
+#pragma unroll
+          for (int kvt = 0; kvt < 2; ++kvt) {
+            const int kvcol = krow0 + kvt * 16 + c;
+            float dsv[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              int qi = n * 16 + g * 4 + r;
+              int qrow_ = qt0 + qi;
+              bool ok = (kvcol < skv) && (qrow_ < sq);
+              if (causal) ok &= (kvcol <= qrow_ + off);
+              if (window > 0) ok &= (kvcol > qrow_ + off - window);
+              float pt = ok ? exp2f(__builtin_fmaf(st[kvt][r], c1, -lse_s[qi])) : 0.f;
+              dsv[r] = pt * __builtin_fmaf(dpt[kvt][r], scale, -dr_s[qi]);
+            }
+            pkD[n & 1][kvt][0] = pack_bf16x2(dsv[0], dsv[1]);
+            pkD[n & 1][kvt][1] = pack_bf16x2(dsv[2], dsv[3]);
+          }
+          if (n & 1) {
+            const int ks2 = n >> 1;
+            bf16x8 aD[2];
+#pragma unroll
+            for (int kvt = 0; kvt < 2; ++kvt) {
+              int avD[4];
+              dfrag_to_at(pkD[0][kvt], pkD[1][kvt], avD);
+              int4 aD4 = make_int4(avD[0], avD[1], avD[2], avD[3]);
+              aD[kvt] = *reinterpret_cast<bf16x8*>(&aD4);
+            }
+#pragma unroll
+            for (int nd = 0; nd < ND; ++nd) {
+              int drow_ = nd * 16 + c;
+              short8 t4 = *reinterpret_cast<const short8*>(&Qt_lds[drow_][swzb(drow_, ks2 * 4 + g) * 8]);
+              bf16x8 bq = *reinterpret_cast<bf16x8*>(&t4);
+#pragma unroll
+              for (int kvt = 0; kvt < 2; ++kvt)
+                dk_acc[kvt][nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aD[kvt], bq, dk_acc[kvt][nd], 0, 0, 0);
+            }
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+      }
+
+      if (have_next) {
+        __syncthreads();
+        __builtin_amdgcn_s_waitcnt(WAIT_VM0);
+        __builtin_amdgcn_sched_barrier(0);
+        stage_writes(qt0 + QBLK);
+        __syncthreads();
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int kvt = 0; kvt < 2; ++kvt)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = krow0 + kvt * 16 + g * 4 + r;
+      if (row >= skv) continue;
+      bf16* dkr = dkbase + (long)row * k_srow;
+#pragma unroll
+      for (int n = 0; n < ND; ++n) dkr[n * 16 + c] = f2bf(dk_acc[kvt][n][r]);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // host wrapper
 // ---------------------------------------------------------------------------
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
@@ -490,18 +1097,41 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                      (const short8*)doc.data_ptr(), (const short8*)oc.data_ptr(),
                      drow.data_ptr<float>(), rows, b * hq, sq, d8);
 
-  dim3 grid_dq((sq + 127) / 128, b * hq);
-  dim3 grid_dkv((skv + 127) / 128, b * hkv);
+  // v3 (32 rows/wave register blocking) for long sequences; v2 fallback via
+  // env or when the sequence is too short to fill 256-row blocks.
+  static const bool force_v2 = getenv("MEGATRON_AMD_ATTN_BWD_V2") != nullptr;
+  const bool use_v3_dq = !force_v2 && sq >= 256;
+  const bool use_v3_dkv = !force_v2 && skv >= 256;
+  dim3 grid_dq((sq + (use_v3_dq ? 255 : 127)) / (use_v3_dq ? 256 : 128), b * hq);
+  dim3 grid_dkv((skv + (use_v3_dkv ? 255 : 127)) / (use_v3_dkv ? 256 : 128), b * hkv);
   if (d == 128) {
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid_dq, dim3(512), 0, stream,
-                       (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
-                       (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
-                       (bf16*)dq.data_ptr(), sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), grid_dkv, dim3(512), 0, stream,
-                       (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
-                       (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
-                       (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
-                       sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
+    if (use_v3_dq)
+      hipLaunchKernelGGL((attn_bwd_dq_kernel_v3<128>), grid_dq, dim3(512), 0, stream,
+                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+                         (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                         (bf16*)dq.data_ptr(), sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
+    else
+      hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid_dq, dim3(512), 0, stream,
+                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+                         (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                         (bf16*)dq.data_ptr(), sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
+    if (use_v3_dkv) {
+      hipLaunchKernelGGL((attn_bwd_dv_kernel_v3<128>), grid_dkv, dim3(512), 0, stream,
+                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+                         (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                         (bf16*)dv.data_ptr(),
+                         sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
+      hipLaunchKernelGGL((attn_bwd_dk_kernel_v3<128>), grid_dkv, dim3(512), 0, stream,
+                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+                         (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                         (bf16*)dk.data_ptr(),
+                         sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
+    } else
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), grid_dkv, dim3(512), 0, stream,
+                         (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+                         (const bf16*)doc.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                         (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
+                         sq, skv, b, hq, hkv, (float)scale, causal ? 1 : 0, (int)window);
   } else {
     hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), grid_dq, dim3(512), 0, stream,
                        (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),

@@ -34,7 +34,8 @@ setup(
             sources=SRC,
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"]
+                + os.environ.get("MEGATRON_AMD_HIPCC_EXTRA", "").split(),
             },
             libraries=["rocblas"],
         ),
