@@ -70,7 +70,38 @@ def parse_args():
     p.add_argument("--hip-graphs", action="store_true",
                    help="capture per-layer fwd/bwd hipGraphs after warmup (dense models, dp-only)")
     p.add_argument("--seed", type=int, default=1234)
+    p.add_argument("--tune-gemm", action="store_true",
+                   help="run hipBLASLt TunableOp algo search and save profiles/tunableop_gfx950.csv")
+    p.add_argument("--no-tunableop", action="store_true",
+                   help="skip loading the committed TunableOp GEMM selections")
     return p.parse_args()
+
+
+def setup_tunableop(args, rank):
+    """hipBLASLt GEMM algorithm selection (PyTorch TunableOp).
+
+    The committed profiles/tunableop_gfx950.csv holds the best hipBLASLt algo
+    per GEMM shape, found once on MI355X with --tune-gemm; every later run
+    (including the driver's) just loads it -- no tuning cost in the timed
+    region.  Reference analog: Tensile solution selection inside TE.
+    """
+    if not torch.cuda.is_available() or args.no_tunableop:
+        return None
+    import torch.cuda.tunable as tunable
+
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "profiles", "tunableop_gfx950.csv")
+    if args.tune_gemm:
+        tunable.enable(True)
+        tunable.tuning_enable(True)
+        tunable.set_filename(path, insert_device_ordinal=False)
+        return path if rank == 0 else None
+    if os.path.exists(path):
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        tunable.set_filename(path, insert_device_ordinal=False)
+        tunable.read_file(path)
+    return None
 
 
 def main():
@@ -88,6 +119,8 @@ def main():
 
     if use_dist and not torch.distributed.is_initialized():
         G.init_distributed()
+
+    tune_out = setup_tunableop(args, rank)
 
     tp = args.tp
     if tp is None:
@@ -201,6 +234,12 @@ def main():
             t = t.to(device)
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
+
+    if tune_out is not None:
+        import torch.cuda.tunable as tunable
+
+        tunable.write_file(tune_out)
+        print(f"# wrote TunableOp results to {tune_out}", flush=True)
 
     ms_per_step = elapsed / args.steps * 1000.0
     global_batch = args.micro_batch_size * args.grad_accum * dp

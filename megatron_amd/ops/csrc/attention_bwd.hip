@@ -1101,7 +1101,11 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   // env or when the sequence is too short to fill 256-row blocks.
   static const bool force_v2 = getenv("MEGATRON_AMD_ATTN_BWD_V2") != nullptr;
   const bool use_v3_dq = !force_v2 && sq >= 256;
-  const bool use_v3_dkv = !force_v2 && skv >= 256;
+  // dv/dk split (v3) measured slower than the combined v2 dkv at s=4096
+  // (dk's global A-fragment reads thrash L1; see profiles/README.md round-2
+  // notes) - combined v2 stays the default, split kept for future work.
+  static const bool force_dkv_v3 = getenv("MEGATRON_AMD_ATTN_DKV_V3") != nullptr;
+  const bool use_v3_dkv = force_dkv_v3 && !force_v2 && skv >= 256;
   dim3 grid_dq((sq + (use_v3_dq ? 255 : 127)) / (use_v3_dq ? 256 : 128), b * hq);
   dim3 grid_dkv((skv + (use_v3_dkv ? 255 : 127)) / (use_v3_dkv ? 256 : 128), b * hkv);
   if (d == 128) {
