@@ -57,22 +57,35 @@ class GroupedMLP(nn.Module):
         return torch.nn.functional.gelu(x)
 
     def forward(self, tokens: torch.Tensor, tokens_per_expert: torch.Tensor) -> torch.Tensor:
-        """tokens [Tlocal, h] sorted by local expert; per-expert GEMM loop
-        (grouped-GEMM kernel slot, K11)."""
-        splits = tokens_per_expert.tolist()
-        outs = []
-        start = 0
-        for e, n in enumerate(splits):
-            if n == 0:
-                outs.append(tokens.new_zeros(0, self.config.hidden_size))
-                continue
-            x = tokens[start : start + n]
-            start += n
-            h = torch.matmul(x, self.weight1[e].t())
+        """tokens [Tlocal, h] sorted by local expert.
+
+        GPU: ONE hipBLASLt grouped GEMM per linear (K11, ops.grouped_linear)
+        — fc1 for all experts in one launch, fused activation, fc2 in one
+        launch; the per-expert sizes sync below is the layer's single chosen
+        host sync point (reference token_dispatcher.py:453-460).  CPU /
+        no-native: per-expert matmul loop."""
+        if tokens_per_expert.is_cuda:
+            splits = tokens_per_expert.cpu().tolist()  # single sync per layer
+        else:
+            splits = [int(x) for x in tokens_per_expert.tolist()]
+        if tokens.is_cuda and tokens.dtype == torch.bfloat16 and ops.has_native():
+            h = ops.grouped_linear(tokens, self.weight1, splits)
             h = self._act(h)
-            y = torch.matmul(h, self.weight2[e].t())
-            outs.append(y)
-        out = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+            out = ops.grouped_linear(h, self.weight2, splits)
+        else:
+            outs = []
+            start = 0
+            for e, n in enumerate(splits):
+                if n == 0:
+                    outs.append(tokens.new_zeros(0, self.config.hidden_size))
+                    continue
+                x = tokens[start : start + n]
+                start += n
+                h = torch.matmul(x, self.weight1[e].t())
+                h = self._act(h)
+                y = torch.matmul(h, self.weight2[e].t())
+                outs.append(y)
+            out = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
         tp_group = G.get_tensor_model_parallel_group() if G.grid_initialized() else None
         if tp_group is not None and G.get_tensor_model_parallel_world_size() > 1:
             out = _ReduceExpertOutput.apply(out)

@@ -111,34 +111,35 @@ class MoEAlltoAllTokenDispatcher:
             return permuted, tokens_per_expert
 
         # per-EP-peer split sizes (each peer owns num_local_experts experts)
-        input_splits = tokens_per_expert.view(self.ep, self.num_local_experts).sum(dim=1)
         counts_matrix = torch.empty(self.ep * self.num_experts, dtype=tokens_per_expert.dtype,
                                     device=tokens_per_expert.device)
         torch.distributed.all_gather_into_tensor(
             counts_matrix, tokens_per_expert.contiguous(), group=self.group
         )
-        counts_matrix = counts_matrix.view(self.ep, self.num_experts)  # [src_rank, expert]
+        # ONE DtoH copy of the [ep, E] counts matrix is the layer's single
+        # host sync point; every split list below is derived host-side
+        # (reference cuda_sync_point discipline, token_dispatcher.py:453-460)
+        counts_host = counts_matrix.view(self.ep, self.num_experts).cpu()  # [src_rank, expert]
         rank = torch.distributed.get_rank(group=self.group)
-        my_slice = counts_matrix[:, rank * self.num_local_experts : (rank + 1) * self.num_local_experts]
-        output_splits = my_slice.sum(dim=1)  # tokens arriving from each peer
-        self._input_splits = input_splits.tolist()
-        self._output_splits = output_splits.tolist()
+        my_slice = counts_host[:, rank * self.num_local_experts : (rank + 1) * self.num_local_experts]
+        self._input_splits = counts_host[rank].view(self.ep, self.num_local_experts).sum(dim=1).tolist()
+        self._output_splits = my_slice.sum(dim=1).tolist()  # tokens arriving from each peer
 
         recv = all_to_all(self.group, permuted, self._output_splits, self._input_splits)
 
         # received tokens are grouped by (src_rank, local_expert); resort to
         # (local_expert, src_rank) so each expert's tokens are contiguous
         # (reference sort_chunks_by_idxs moe_utils.py:628)
-        chunk_sizes = my_slice.reshape(-1)  # [ep * n_local] in (rank, expert) order
-        idx = (
-            torch.arange(self.ep * self.num_local_experts, device=tokens.device)
-            .view(self.ep, self.num_local_experts).t().reshape(-1)
-        )
-        self._chunk_sizes = chunk_sizes.tolist()
-        self._chunk_perm = idx.tolist()
+        self._chunk_sizes = my_slice.reshape(-1).tolist()  # [ep * n_local] in (rank, expert) order
+        # _chunk_perm[i] indexes (rank-major) chunks in (expert, rank) order
+        self._chunk_perm = [
+            r * self.num_local_experts + e
+            for e in range(self.num_local_experts)
+            for r in range(self.ep)
+        ]
         chunks = torch.split(recv, self._chunk_sizes)
         reordered = torch.cat([chunks[i] for i in self._chunk_perm], dim=0) if len(chunks) > 1 else recv
-        tokens_per_local_expert = my_slice.sum(dim=0)
+        tokens_per_local_expert = my_slice.sum(dim=0)  # host tensor: experts sync-free downstream
         self._restore = True
         return reordered, tokens_per_local_expert
 

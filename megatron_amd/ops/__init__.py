@@ -209,6 +209,58 @@ def wgrad_gemm_accum(main_grad: torch.Tensor, grad_output_2d: torch.Tensor, inpu
 
 
 # ---------------------------------------------------------------------------
+# grouped GEMM for MoE experts  (K11)
+# ---------------------------------------------------------------------------
+
+
+class _GroupedLinearFn(torch.autograd.Function):
+    """y_e = x_e @ w_e^T over variable-size expert batches (one hipBLASLt
+    grouped-gemm launch).  Backward: grouped dgrad + grouped wgrad with fp32
+    accumulation directly into weight.main_grad when the DDP fused path is
+    active (mirrors the dense _ParallelLinearFn contract)."""
+
+    @staticmethod
+    def forward(ctx, a, weight, sizes):
+        out = _C.grouped_gemm(a.contiguous(), weight, list(sizes), True)
+        ctx.save_for_backward(a, weight)
+        ctx.sizes = sizes
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        sizes = list(ctx.sizes)
+        da = _C.grouped_gemm(dy, weight, sizes, False)
+        main_grad = getattr(weight, "main_grad", None)
+        if main_grad is not None and main_grad.is_cuda and main_grad.dtype == torch.float32:
+            _C.grouped_gemm_wgrad(dy, a.contiguous(), sizes, main_grad.view(weight.shape))
+            weight.grad_added_to_main_grad = True
+            cb = getattr(weight, "_ddp_grad_ready_cb", None)
+            if cb is not None:
+                cb()
+            dw = None
+        else:
+            dw32 = torch.zeros(weight.shape, dtype=torch.float32, device=weight.device)
+            _C.grouped_gemm_wgrad(dy, a.contiguous(), sizes, dw32)
+            dw = dw32.to(weight.dtype)
+        return da, dw, None
+
+
+def grouped_linear(a: torch.Tensor, weight: torch.Tensor, sizes) -> torch.Tensor:
+    """a [M, k] bf16 rows grouped by expert; weight [E, n, k]; sizes: host
+    ints per expert summing to M.  Returns [M, n]."""
+    if _use_native(a):
+        return _GroupedLinearFn.apply(a, weight, tuple(int(s) for s in sizes))
+    outs, start = [], 0
+    for e, n_e in enumerate(sizes):
+        n_e = int(n_e)
+        outs.append(a[start : start + n_e] @ weight[e].t())
+        start += n_e
+    return torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+
+
+# ---------------------------------------------------------------------------
 # multi-tensor optimizer kernels  (K10)
 # ---------------------------------------------------------------------------
 

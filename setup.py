@@ -21,6 +21,7 @@ SRC = [
     "megatron_amd/ops/csrc/adamw.hip",
     "megatron_amd/ops/csrc/wgrad.hip",
     "megatron_amd/ops/csrc/cross_entropy.hip",
+    "megatron_amd/ops/csrc/grouped_gemm.cpp",
     "megatron_amd/ops/csrc/attention_fwd.hip",
     "megatron_amd/ops/csrc/attention_bwd.hip",
     "megatron_amd/ops/csrc/symm_allreduce.hip",
@@ -37,7 +38,7 @@ setup(
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"]
                 + os.environ.get("MEGATRON_AMD_HIPCC_EXTRA", "").split(),
             },
-            libraries=["rocblas"],
+            libraries=["rocblas", "hipblaslt"],
         ),
         CppExtension(
             name="megatron_amd.datasets._data_helpers",
