@@ -163,6 +163,12 @@ class ParallelGrid:
         # in the norm reduction would count every element cp times
         # (reference uses tp x pp for grad stats too).
         make("tp_pp", [True, False, False, True])
+        # second PP communicator for the backward (grad) direction: forward
+        # activations and backward grads between the same rank pair must not
+        # share a channel, or staggered schedules (ragged interleaving,
+        # pp=2 ring wrap) can cross-match a grad send with an activation
+        # recv (messages match by order within one communicator).
+        make("pp_bwd", [False, False, False, True])
 
         # expert groups: factor the (tp*cp*dp) span as (etp, ep, edp), pp slowest
         espan_shape = [etp, ep, self.edp, pp]

@@ -24,7 +24,11 @@ class P2PCommunicator:
         self.grid = grid
         self.prev_rank = grid.pipeline_prev_rank()
         self.next_rank = grid.pipeline_next_rank()
-        self.group = grid.group("pp")
+        self.group = grid.group("pp")          # forward direction (activations)
+        try:
+            self.group_bwd = grid.group("pp_bwd")  # backward direction (grads)
+        except KeyError:
+            self.group_bwd = self.group
         s = seq_length
         if config.sequence_parallel:
             s //= max(1, config.tensor_parallel_size)
@@ -52,7 +56,7 @@ class P2PCommunicator:
 
         def sends():
             if tensor_send_prev is not None:
-                ops.append(dist.P2POp(dist.isend, to_t(tensor_send_prev), self.prev_rank, group=self.group))
+                ops.append(dist.P2POp(dist.isend, to_t(tensor_send_prev), self.prev_rank, group=self.group_bwd))
             if tensor_send_next is not None:
                 ops.append(dist.P2POp(dist.isend, to_t(tensor_send_next), self.next_rank, group=self.group))
 
@@ -60,7 +64,7 @@ class P2PCommunicator:
             if recv_prev_shape is not None:
                 ops.append(dist.P2POp(dist.irecv, recv_prev_shape, self.prev_rank, group=self.group))
             if recv_next_shape is not None:
-                ops.append(dist.P2POp(dist.irecv, recv_next_shape, self.next_rank, group=self.group))
+                ops.append(dist.P2POp(dist.irecv, recv_next_shape, self.next_rank, group=self.group_bwd))
 
         if even:
             sends()
@@ -68,9 +72,14 @@ class P2PCommunicator:
         else:
             recvs()
             sends()
-        if ops:
-            for r in dist.batch_isend_irecv(ops):
-                r.wait()
+        reqs = []
+        groups = [self.group] + ([self.group_bwd] if self.group_bwd is not self.group else [])
+        for grp in groups:
+            grp_ops = [o for o in ops if o.group is grp]
+            if grp_ops:
+                reqs.extend(dist.batch_isend_irecv(grp_ops))
+        for r in reqs:
+            r.wait()
         return (tuple(recv_prev_shape.tolist()) if recv_prev_shape is not None else None,
                 tuple(recv_next_shape.tolist()) if recv_next_shape is not None else None)
 
@@ -99,24 +108,30 @@ class P2PCommunicator:
         even = self.grid.pp_rank % 2 == 0
         def add_sends():
             if tensor_send_prev is not None:
-                ops.append(dist.P2POp(dist.isend, tensor_send_prev.contiguous(), self.prev_rank, group=self.group))
+                ops.append(dist.P2POp(dist.isend, tensor_send_prev.contiguous(), self.prev_rank, group=self.group_bwd))
             if tensor_send_next is not None:
                 ops.append(dist.P2POp(dist.isend, tensor_send_next.contiguous(), self.next_rank, group=self.group))
         def add_recvs():
             if tensor_recv_prev is not None:
                 ops.append(dist.P2POp(dist.irecv, tensor_recv_prev, self.prev_rank, group=self.group))
             if tensor_recv_next is not None:
-                ops.append(dist.P2POp(dist.irecv, tensor_recv_next, self.next_rank, group=self.group))
+                ops.append(dist.P2POp(dist.irecv, tensor_recv_next, self.next_rank, group=self.group_bwd))
         if even:
             add_sends()
             add_recvs()
         else:
             add_recvs()
             add_sends()
-        if ops:
-            reqs = dist.batch_isend_irecv(ops)
-            for r in reqs:
-                r.wait()
+        # batch_isend_irecv requires a single group per batch: issue one
+        # batch per direction-communicator, wait on both
+        reqs = []
+        groups = [self.group] + ([self.group_bwd] if self.group_bwd is not self.group else [])
+        for grp in groups:
+            grp_ops = [o for o in ops if o.group is grp]
+            if grp_ops:
+                reqs.extend(dist.batch_isend_irecv(grp_ops))
+        for r in reqs:
+            r.wait()
         return tensor_recv_prev, tensor_recv_next
 
     # -- convenience wrappers (reference API shape) --------------------------

@@ -192,7 +192,9 @@ def forward_backward_pipelining_with_interleaving(
     pp, pp_rank = grid.pp, grid.pp_rank
     num_chunks = len(model)
     group_size = config.microbatch_group_size_per_vp_stage or pp
-    assert num_microbatches % pp == 0, "interleaved schedule requires num_microbatches % pp == 0"
+    # num_microbatches need not divide pp: the schedule table's last group is
+    # simply smaller and the warmup count is clamped to the table length
+    # (reference schedules.py:959 handles the general case the same way).
 
     no_syncs = [m.no_sync() for m in model if hasattr(m, "no_sync")]
     for c in no_syncs:

@@ -222,3 +222,36 @@ def test_tp2_pp2_composition_matches_single(tmp_path, monkeypatch):
     spawn_dist(_tp2pp2_case, 4, ckpt)
     got = json.load(open(out))[0]
     assert abs(got - ref_loss) < 5e-4, (got, ref_loss)
+
+
+def _pp2_ragged_case(rank, world, vpp):
+    G.initialize_model_parallel(pipeline_parallel_size=world, virtual_pipeline_parallel_size=vpp)
+    model_parallel_seed(1234)
+    cfg = _cfg(pp=world, vpp=vpp)
+    losses = _run(cfg, 2, 3, _gen_batches(6))  # 3 microbatches % pp=2 != 0
+    if G.get_grid().is_pipeline_last_stage(ignore_virtual=True):
+        with open(os.environ["PP_TEST_OUT"], "w") as f:
+            json.dump(losses, f)
+
+
+def test_pp2_interleaved_ragged_microbatches(tmp_path, monkeypatch):
+    """num_microbatches % pp != 0 (reference schedules.py:959 clamping)."""
+    out = tmp_path / "ppr.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+    init_single()
+    ref = _run(_cfg(), 2, 3, _gen_batches(6))
+    spawn_dist(_pp2_ragged_case, 2, 2)
+    pp_losses = json.load(open(out))
+    for a, b in zip(ref, pp_losses):
+        assert abs(a - b) < 2e-4, (ref, pp_losses)
+
+
+def test_pp2_noninterleaved_ragged_microbatches(tmp_path, monkeypatch):
+    out = tmp_path / "ppr2.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+    init_single()
+    ref = _run(_cfg(), 2, 3, _gen_batches(6))
+    spawn_dist(_pp2_ragged_case, 2, None)
+    pp_losses = json.load(open(out))
+    for a, b in zip(ref, pp_losses):
+        assert abs(a - b) < 2e-4, (ref, pp_losses)
