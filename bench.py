@@ -70,6 +70,10 @@ def parse_args():
     p.add_argument("--hip-graphs", action="store_true",
                    help="capture per-layer fwd/bwd hipGraphs after warmup (dense models, dp-only)")
     p.add_argument("--seed", type=int, default=1234)
+    p.add_argument("--num-layers", type=int, default=None,
+                   help="override layer count (reduced-depth evidence runs; the JSON "
+                        "config records the override so the line is never mistaken "
+                        "for the full model)")
     p.add_argument("--tune-gemm", action="store_true",
                    help="run hipBLASLt TunableOp algo search and save profiles/tunableop_gfx950.csv")
     p.add_argument("--no-tunableop", action="store_true",
@@ -143,6 +147,10 @@ def main():
     dp = grid.dp
 
     mdl = dict(MODELS[args.model])
+    layers_overridden = False
+    if args.num_layers is not None and args.num_layers != mdl["num_layers"]:
+        mdl["num_layers"] = args.num_layers
+        layers_overridden = True
     # 70B at 1 GPU cannot fit; scale layer count for sub-node smoke unless full node
     cfg = TransformerConfig(
         **mdl,
@@ -270,7 +278,8 @@ def main():
             "mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 1) if device.type == "cuda" else None,
             "loss": None if last is None else round(last["lm_loss"], 4),
             "config": {
-                "model": args.model,
+                "model": args.model if not layers_overridden else f"{args.model}@{mdl['num_layers']}L",
+                "num_layers": mdl["num_layers"],
                 "global_batch": global_batch,
                 "seq_len": args.seq_len,
                 "parallelism": par,

@@ -102,25 +102,29 @@ void run_grouped(hipblasOperation_t opA, hipblasOperation_t opB,
   hipblaslt_ext::GemmProblemType ptype(opA, opB, tAB, tAB, tCD, tCD, comp);
   HIPBLASLT_CHECK(ca.gg.setProblem(m, n, k, batch, lda, ldb, ldc, ldc,
                                    strideA, strideB, strideC, strideC, epi, inputs, ptype));
+  // algo discovery per the hipBLASLt grouped-gemm sample flow: enumerate
+  // all grouped-gemm algos once, pick the first that supports the problem
+  auto find_algo = [&]() {
+    std::vector<hipblasLtMatmulHeuristicResult_t> all;
+    HIPBLASLT_CHECK(hipblaslt_ext::getAllAlgos(
+        lt_handle(), hipblaslt_ext::GemmType::HIPBLASLT_GROUPED_GEMM,
+        opA, opB, tAB, tAB, tCD, tCD, comp, all));
+    for (auto& r : all) {
+      size_t ws = kMaxWorkspace;
+      if (ca.gg.isAlgoSupported(r.algo, ws) == HIPBLAS_STATUS_SUCCESS && ws <= kMaxWorkspace) {
+        ca.algo = r.algo;
+        ca.has_algo = true;
+        return;
+      }
+    }
+    TORCH_CHECK(false, "grouped_gemm: no hipblaslt grouped algo supports this problem");
+  };
   if (!ca.has_algo) {
-    hipblaslt_ext::GemmPreference pref;
-    pref.setMaxWorkspaceBytes(kMaxWorkspace);
-    std::vector<hipblasLtMatmulHeuristicResult_t> results;
-    HIPBLASLT_CHECK(ca.gg.algoGetHeuristic(8, pref, results));
-    TORCH_CHECK(!results.empty(), "grouped_gemm: no hipblaslt algo found");
-    ca.algo = results[0].algo;
-    ca.has_algo = true;
-  }
-  size_t ws_bytes = kMaxWorkspace;
-  if (ca.gg.isAlgoSupported(ca.algo, ws_bytes) != HIPBLAS_STATUS_SUCCESS) {
-    // m-vector changed enough that the cached algo no longer applies
-    hipblaslt_ext::GemmPreference pref;
-    pref.setMaxWorkspaceBytes(kMaxWorkspace);
-    std::vector<hipblasLtMatmulHeuristicResult_t> results;
-    HIPBLASLT_CHECK(ca.gg.algoGetHeuristic(8, pref, results));
-    TORCH_CHECK(!results.empty(), "grouped_gemm: no hipblaslt algo found (resize)");
-    ca.algo = results[0].algo;
-    ws_bytes = kMaxWorkspace;
+    find_algo();
+  } else {
+    size_t ws_bytes = kMaxWorkspace;
+    if (ca.gg.isAlgoSupported(ca.algo, ws_bytes) != HIPBLAS_STATUS_SUCCESS)
+      find_algo();  // m-vector changed enough that the cached algo no longer applies
   }
   auto stream = at::cuda::getCurrentHIPStream();
   HIPBLASLT_CHECK(ca.gg.initialize(ca.algo, lt_workspace(kMaxWorkspace), false, stream));
