@@ -113,7 +113,8 @@ class TopKRouter(nn.Module):
             # multiplicative input jitter (reference router.py apply_input_jitter)
             noise = torch.empty_like(hidden).uniform_(1.0 - self.jitter_eps, 1.0 + self.jitter_eps)
             hidden = hidden * noise
-        logits = F.linear(hidden.float(), self.weight)  # [T, E]
+        # gate math in fp32 even if a module-wide .bfloat16() cast hit the weight
+        logits = F.linear(hidden.float(), self.weight.float())  # [T, E]
         self.aux_losses = {}
 
         if self.score_function == "sigmoid":

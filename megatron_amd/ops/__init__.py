@@ -213,11 +213,50 @@ def wgrad_gemm_accum(main_grad: torch.Tensor, grad_output_2d: torch.Tensor, inpu
 # ---------------------------------------------------------------------------
 
 
+_USE_TORCH_GROUPED = (
+    hasattr(torch, "_grouped_mm")
+    and os.environ.get("MEGATRON_AMD_NO_TORCH_GROUPED", "0") != "1"
+)
+
+
 class _GroupedLinearFn(torch.autograd.Function):
-    """y_e = x_e @ w_e^T over variable-size expert batches (one hipBLASLt
-    grouped-gemm launch).  Backward: grouped dgrad + grouped wgrad with fp32
-    accumulation directly into weight.main_grad when the DDP fused path is
-    active (mirrors the dense _ParallelLinearFn contract)."""
+    """y_e = x_e @ w_e^T over variable-size expert batches: one grouped-GEMM
+    launch (torch._grouped_mm -> CK/hipBLASLt kernels; this image's
+    hipblaslt-ext C++ grouped path is broken — every algo reports internal
+    error — so the torch op is the native route).  Backward: grouped dgrad +
+    grouped wgrad; wgrad accumulates fp32 into weight.main_grad when the DDP
+    fused path is active (mirrors the dense _ParallelLinearFn contract)."""
+
+    @staticmethod
+    def forward(ctx, a, weight, sizes):
+        a = a.contiguous()
+        offs = torch.tensor(sizes, device=a.device, dtype=torch.int32).cumsum(0, dtype=torch.int32)
+        out = torch._grouped_mm(a, weight.transpose(1, 2), offs=offs)
+        ctx.save_for_backward(a, weight, offs)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, weight, offs = ctx.saved_tensors
+        dy = dy.contiguous()
+        da = torch._grouped_mm(dy, weight, offs=offs)
+        main_grad = getattr(weight, "main_grad", None)
+        if main_grad is not None and main_grad.is_cuda and main_grad.dtype == torch.float32:
+            dw32 = torch._grouped_mm(dy.t(), a, offs=offs, out_dtype=torch.float32)
+            main_grad.view(weight.shape).add_(dw32)
+            weight.grad_added_to_main_grad = True
+            cb = getattr(weight, "_ddp_grad_ready_cb", None)
+            if cb is not None:
+                cb()
+            dw = None
+        else:
+            dw = torch._grouped_mm(dy.t(), a, offs=offs).to(weight.dtype)
+        return da, dw, None
+
+
+class _GroupedLinearFnHipblaslt(torch.autograd.Function):
+    """hipBLASLt-ext variant (kept for environments where the ext grouped
+    path works; MEGATRON_AMD_NO_TORCH_GROUPED=1 selects it)."""
 
     @staticmethod
     def forward(ctx, a, weight, sizes):
@@ -251,7 +290,8 @@ def grouped_linear(a: torch.Tensor, weight: torch.Tensor, sizes) -> torch.Tensor
     """a [M, k] bf16 rows grouped by expert; weight [E, n, k]; sizes: host
     ints per expert summing to M.  Returns [M, n]."""
     if _use_native(a):
-        return _GroupedLinearFn.apply(a, weight, tuple(int(s) for s in sizes))
+        fn = _GroupedLinearFn if _USE_TORCH_GROUPED else _GroupedLinearFnHipblaslt
+        return fn.apply(a, weight, tuple(int(s) for s in sizes))
     outs, start = [], 0
     for e, n_e in enumerate(sizes):
         n_e = int(n_e)

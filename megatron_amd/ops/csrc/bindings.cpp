@@ -21,6 +21,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor target, long vocab_start);
 void ce_bwd(torch::Tensor logits, torch::Tensor target, torch::Tensor row_max,
             torch::Tensor row_inv_sumexp, torch::Tensor grad_out, long vocab_start);
+std::string grouped_gemm_probe();
 torch::Tensor grouped_gemm(torch::Tensor a, torch::Tensor b, std::vector<int64_t> sizes, bool trans_b);
 void grouped_gemm_wgrad(torch::Tensor dy, torch::Tensor x, std::vector<int64_t> sizes, torch::Tensor dw);
 std::vector<int64_t> symm_ipc_handle(torch::Tensor buf);
@@ -45,6 +46,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("grouped_gemm", &grouped_gemm);
+  m.def("grouped_gemm_probe", &grouped_gemm_probe);
   m.def("grouped_gemm_wgrad", &grouped_gemm_wgrad);
   m.def("symm_ipc_handle", &symm_ipc_handle);
   m.def("symm_open_handle", &symm_open_handle);

@@ -88,7 +88,13 @@ void run_grouped(hipblasOperation_t opA, hipblasOperation_t opB,
   static float one = 1.0f;
   static float zero = 0.0f;
   std::vector<int64_t> batch(ng, 1);
-  std::vector<int64_t> strideA(ng, 0), strideB(ng, 0), strideC(ng, 0);
+  // valid batch strides even at batch_count 1 (Tensile validates them)
+  std::vector<int64_t> strideA(ng), strideB(ng), strideC(ng);
+  for (size_t i = 0; i < ng; ++i) {
+    strideA[i] = lda[i] * (opA == HIPBLAS_OP_N ? k[i] : m[i]);
+    strideB[i] = ldb[i] * (opB == HIPBLAS_OP_N ? n[i] : k[i]);
+    strideC[i] = ldc[i] * n[i];
+  }
   std::vector<hipblaslt_ext::GemmEpilogue> epi(ng);
   std::vector<hipblaslt_ext::GemmInputs> inputs(ng);
   for (size_t i = 0; i < ng; ++i) {
@@ -177,6 +183,62 @@ torch::Tensor grouped_gemm(torch::Tensor a, torch::Tensor b,
               m, n, k, lda, ldb, ldc, aptr, bptr, cptr, 0.f,
               trans_b ? 0 : 1);
   return c;
+}
+
+std::string grouped_gemm_probe() {
+  // diagnostic: how many grouped algos exist, and why isAlgoSupported fails
+  std::string out;
+  std::vector<hipblasLtMatmulHeuristicResult_t> all;
+  hipblasStatus_t st = hipblaslt_ext::getAllAlgos(
+      lt_handle(), hipblaslt_ext::GemmType::HIPBLASLT_GROUPED_GEMM,
+      HIPBLAS_OP_T, HIPBLAS_OP_N, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF,
+      HIPBLAS_COMPUTE_32F, all);
+  out += "getAllAlgos status=" + std::to_string((int)st) + " n=" + std::to_string(all.size());
+  hipblaslt_ext::GroupedGemm gg(lt_handle(), HIPBLAS_OP_T, HIPBLAS_OP_N,
+                                HIP_R_16BF, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF,
+                                HIPBLAS_COMPUTE_32F);
+  // tiny fixed problem: 2 groups of C[64,128] = A[64,256] @ B[128,256]^T
+  auto opt = torch::TensorOptions().dtype(torch::kBFloat16).device(torch::kCUDA);
+  auto A = torch::randn({128, 256}, opt);
+  auto B = torch::randn({2, 128, 256}, opt);
+  auto C = torch::empty({128, 128}, opt);
+  std::vector<int64_t> m(2, 128), n(2, 64), k(2, 256), batch(2, 1);
+  std::vector<int64_t> lda(2, 256), ldb(2, 256), ldc(2, 128);
+  std::vector<int64_t> sA(2, 256 * 128), sB(2, 256 * 64), sC(2, 128 * 64);
+  std::vector<hipblaslt_ext::GemmEpilogue> epi(2);
+  std::vector<hipblaslt_ext::GemmInputs> inputs(2);
+  static float onef = 1.f, zerof = 0.f;
+  for (int i = 0; i < 2; ++i) {
+    inputs[i].setA((char*)B.data_ptr() + i * 128 * 256 * 2);
+    inputs[i].setB((char*)A.data_ptr() + i * 64 * 256 * 2);
+    inputs[i].setC((char*)C.data_ptr() + i * 64 * 128 * 2);
+    inputs[i].setD((char*)C.data_ptr() + i * 64 * 128 * 2);
+    inputs[i].setAlpha(&onef);
+    inputs[i].setBeta(&zerof);
+  }
+  hipblaslt_ext::GemmProblemType ptype(HIPBLAS_OP_T, HIPBLAS_OP_N, HIP_R_16BF, HIP_R_16BF,
+                                       HIP_R_16BF, HIP_R_16BF, HIPBLAS_COMPUTE_32F);
+  st = gg.setProblem(m, n, k, batch, lda, ldb, ldc, ldc, sA, sB, sC, sC, epi, inputs, ptype);
+  out += " setProblem=" + std::to_string((int)st);
+  int ok = 0, first_fail = -999;
+  for (size_t i = 0; i < all.size() && i < 200; ++i) {
+    size_t ws = kMaxWorkspace;
+    hipblasStatus_t s2 = gg.isAlgoSupported(all[i].algo, ws);
+    if (s2 == HIPBLAS_STATUS_SUCCESS) ok++;
+    else if (first_fail == -999) first_fail = (int)s2;
+  }
+  out += " supported=" + std::to_string(ok) + " first_fail_status=" + std::to_string(first_fail);
+  // also try heuristic on the set problem
+  hipblaslt_ext::GemmPreference pref;
+  pref.setMaxWorkspaceBytes(kMaxWorkspace);
+  std::vector<hipblasLtMatmulHeuristicResult_t> res;
+  try {
+    st = gg.algoGetHeuristic(4, pref, res);
+    out += " heuristic_status=" + std::to_string((int)st) + " nres=" + std::to_string(res.size());
+  } catch (const std::exception& e) {
+    out += std::string(" heuristic_threw=") + e.what();
+  }
+  return out;
 }
 
 void grouped_gemm_wgrad(torch::Tensor dy, torch::Tensor x,
