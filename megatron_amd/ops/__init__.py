@@ -242,8 +242,11 @@ class _GroupedLinearFn(torch.autograd.Function):
         da = torch._grouped_mm(dy, weight, offs=offs)
         main_grad = getattr(weight, "main_grad", None)
         if main_grad is not None and main_grad.is_cuda and main_grad.dtype == torch.float32:
-            dw32 = torch._grouped_mm(dy.t(), a, offs=offs, out_dtype=torch.float32)
-            main_grad.view(weight.shape).add_(dw32)
+            # this build's _grouped_mm requires out dtype == input dtype; the
+            # in-GEMM accumulation is still fp32, only the output rounds to
+            # bf16 before the fp32 main_grad add
+            dw = torch._grouped_mm(dy.t(), a, offs=offs)
+            main_grad.view(weight.shape).add_(dw)
             weight.grad_added_to_main_grad = True
             cb = getattr(weight, "_ddp_grad_ready_cb", None)
             if cb is not None:
