@@ -36,6 +36,17 @@ class DelayedScaling:
 
     def scale_for(self, t: torch.Tensor) -> torch.Tensor:
         amax = t.detach().abs().max().float().clamp(min=1e-12)
+        # DP-consistent scales (reference fp8 amax group reduction,
+        # parallel_state.py:1877): replicas must quantize identically or
+        # their updates diverge
+        import torch.distributed as dist
+
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            from megatron_amd.parallel import grid as G
+
+            if G.grid_initialized() and G.get_data_parallel_world_size(with_context_parallel=True) > 1:
+                dist.all_reduce(amax, op=dist.ReduceOp.MAX,
+                                group=G.get_grid().group("dp_cp"))
         if self._history is None:
             self._history = torch.zeros(self.history_len, device=t.device)
         if self._history.device != t.device:

@@ -131,3 +131,57 @@ def test_mxfp4_pack_unpack():
     assert packed.dtype == torch.uint8 and packed.numel() == q.numel() // 2
     back = unpack_fp4_codes(packed, q.numel()).reshape(q.shape)
     assert torch.equal(back, q)
+
+
+@pytest.mark.gpu
+def test_fp8_training_tracks_bf16():
+    """Short training run: fp8-GEMM loss trace stays within tolerance of the
+    bf16 trace (reference core/fp8_utils.py recipes)."""
+    import torch
+
+    from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from megatron_amd.training.training import setup_model_and_optimizer, train_step
+    from tests.utils import init_single
+
+    def run(fp8):
+        init_single()
+        model_parallel_seed(77)
+        cfg = TransformerConfig(
+            num_layers=4, hidden_size=256, num_attention_heads=8, num_query_groups=4,
+            ffn_hidden_size=512, vocab_size=512, bf16=True,
+            fp8="hybrid" if fp8 else None,
+            gradient_accumulation_fusion=True, max_position_embeddings=256,
+        )
+        opt_cfg = OptimizerConfig(lr=3e-4, bf16=True, use_distributed_optimizer=False)
+
+        def provider(config, pre_process=True, post_process=True, vp_stage=None):
+            return GPTModel(config, pre_process=pre_process, post_process=post_process)
+
+        chunks, opt = setup_model_and_optimizer(provider, cfg, opt_cfg, device=torch.device("cuda"))
+        g = torch.Generator(device="cpu").manual_seed(5)
+        losses = []
+        for _ in range(20):
+            toks = torch.randint(0, 512, (4, 128), generator=g).cuda()
+            labels = torch.randint(0, 512, (4, 128), generator=g).cuda()
+
+            def fwd(it, model):
+                out = model(toks, labels=labels)
+
+                def loss_func(loss_sb):
+                    s = loss_sb.sum()
+                    return s, torch.tensor(loss_sb.numel(), device="cuda"), {"loss_sum": s.detach()}
+
+                return out, loss_func
+
+            r = train_step(fwd, None, chunks, opt, cfg, 1, 128, 4)
+            losses.append(r["lm_loss"])
+        return losses
+
+    bf16_losses = run(False)
+    fp8_losses = run(True)
+    # same trajectory within fp8 quantization noise
+    for a, b in zip(bf16_losses[5:], fp8_losses[5:]):
+        assert abs(a - b) / max(abs(b), 1e-6) < 0.08, (bf16_losses, fp8_losses)
+    assert fp8_losses[-1] < fp8_losses[0], "fp8 run is not learning"
