@@ -235,3 +235,85 @@ def test_moe_ep2_checkpoint_reshards_to_single(tmp_path):
     with torch.no_grad():
         out = model(ref["tokens"], position_ids=None, attention_mask=None)
     assert_close(out, ref["out"], rtol=1e-5, atol=1e-5)
+
+
+# --- round-2: windowed load + background-process async save -----------------
+
+
+def test_windowed_load_does_not_assemble_full_tensors(tmp_path, monkeypatch):
+    """Model-param pieces overlapping the request are copied directly; the
+    full-global-tensor assembly path must not run (70B-scalability contract,
+    reference fully_parallel.py:522)."""
+    import json as _json
+
+    from megatron_amd.checkpoint import sharded as S
+
+    path = str(tmp_path / "ck")
+    os.makedirs(path)
+    full = torch.arange(8 * 6, dtype=torch.float32).view(8, 6)
+    # two "ranks", each holding a 4-row slab
+    meta = {}
+    for r in range(2):
+        piece = full[r * 4 : (r + 1) * 4].clone()
+        torch.save({"w": [piece]}, os.path.join(path, f"shards_r{r}.pt"))
+        meta.setdefault("w", {"global_shape": [8, 6], "dtype": "float32", "pieces": []})
+        meta["w"]["pieces"].append({
+            "global_offset": [r * 4, 0], "local_shape": [4, 6],
+            "flattened_range": None, "file": f"shards_r{r}.pt",
+        })
+    _json.dump(meta, open(os.path.join(path, "metadata.json"), "w"))
+    torch.save({"iteration": 3}, os.path.join(path, "common.pt"))
+
+    def boom(self, key):
+        raise AssertionError("full-tensor assembly ran for a rectangular key")
+
+    monkeypatch.setattr(S._ShardReader, "assemble", boom)
+    # request rows 2..6 (straddles both source pieces)
+    st = S.ShardedTensor(key="w", data=torch.zeros(4, 6), global_shape=(8, 6),
+                         global_offset=(2, 0))
+    common = S.load({"w": st}, path)
+    assert common["iteration"] == 3
+    assert torch.equal(st.data, full[2:6])
+
+
+def test_windowed_load_flat_range_same_box(tmp_path):
+    """Distributed-optimizer flat shards resume via flat-segment copies when
+    the box layout is unchanged."""
+    import json as _json
+
+    from megatron_amd.checkpoint import sharded as S
+
+    path = str(tmp_path / "ck")
+    os.makedirs(path)
+    box = torch.arange(24, dtype=torch.float32)
+    # two flat pieces over the same [4,6] box: [0,10) and [10,24)
+    torch.save({"opt": [box[0:10].clone()]}, os.path.join(path, "shards_r0.pt"))
+    torch.save({"opt": [box[10:24].clone()]}, os.path.join(path, "shards_r1.pt"))
+    meta = {"opt": {"global_shape": [4, 6], "dtype": "float32", "pieces": [
+        {"global_offset": [0, 0], "local_shape": [4, 6], "flattened_range": [0, 10], "file": "shards_r0.pt"},
+        {"global_offset": [0, 0], "local_shape": [4, 6], "flattened_range": [10, 24], "file": "shards_r1.pt"},
+    ]}}
+    _json.dump(meta, open(os.path.join(path, "metadata.json"), "w"))
+    torch.save({}, os.path.join(path, "common.pt"))
+    # request flat [6, 18) of the same box
+    st = S.ShardedTensor(key="opt", data=torch.zeros(12), global_shape=(4, 6),
+                         global_offset=(0, 0), local_shape=(4, 6), flattened_range=(6, 18))
+    S.load({"opt": st}, path)
+    assert torch.equal(st.data, box[6:18])
+
+
+def test_async_save_uses_background_process(tmp_path):
+    from megatron_amd.checkpoint import sharded as S
+
+    st = S.ShardedTensor(key="w", data=torch.randn(16, 8), global_shape=(16, 8),
+                         global_offset=(0, 0))
+    writer = S.save({"w": st}, {"iteration": 1}, str(tmp_path / "ck"), async_save=True)
+    import multiprocessing
+
+    assert isinstance(writer, multiprocessing.process.BaseProcess), type(writer)
+    writer.join(timeout=120)
+    assert writer.exitcode == 0
+    st2 = S.ShardedTensor(key="w", data=torch.zeros(16, 8), global_shape=(16, 8),
+                          global_offset=(0, 0))
+    S.load({"w": st2}, str(tmp_path / "ck"))
+    assert torch.equal(st2.data, st.data)
