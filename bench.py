@@ -100,7 +100,12 @@ def setup_tunableop(args, rank):
     if args.tune_gemm:
         tunable.enable(True)
         tunable.tuning_enable(True)
+        # cap per-candidate search cost so one pass covers fwd+dgrad shapes
+        tunable.set_max_tuning_duration(10)
+        tunable.set_max_tuning_iterations(10)
         tunable.set_filename(path, insert_device_ordinal=False)
+        if os.path.exists(path):
+            tunable.read_file(path)  # extend prior results instead of redoing
         return path if rank == 0 else None
     if os.path.exists(path):
         tunable.enable(True)
