@@ -154,17 +154,19 @@ def test_fp8_training_tracks_bf16():
             fp8="hybrid" if fp8 else None,
             gradient_accumulation_fusion=True, max_position_embeddings=256,
         )
-        opt_cfg = OptimizerConfig(lr=3e-4, bf16=True, use_distributed_optimizer=False)
+        opt_cfg = OptimizerConfig(lr=1e-3, bf16=True, use_distributed_optimizer=False)
 
         def provider(config, pre_process=True, post_process=True, vp_stage=None):
             return GPTModel(config, pre_process=pre_process, post_process=post_process)
 
         chunks, opt = setup_model_and_optimizer(provider, cfg, opt_cfg, device=torch.device("cuda"))
         g = torch.Generator(device="cpu").manual_seed(5)
+        # one fixed batch repeated: the loss must drop (memorization) and the
+        # fp8 trajectory must track bf16
+        toks = torch.randint(0, 512, (4, 128), generator=g).cuda()
+        labels = torch.randint(0, 512, (4, 128), generator=g).cuda()
         losses = []
         for _ in range(20):
-            toks = torch.randint(0, 512, (4, 128), generator=g).cuda()
-            labels = torch.randint(0, 512, (4, 128), generator=g).cuda()
 
             def fwd(it, model):
                 out = model(toks, labels=labels)
