@@ -48,14 +48,32 @@ def _allreduce_word_embedding_grads(models: List[torch.nn.Module], config):
         return
     if not (grid.is_pipeline_first_stage(ignore_virtual=True) or grid.is_pipeline_last_stage(ignore_virtual=True)):
         return
+    group = grid.group("embd")
+    if group is None or len(grid.ranks("embd")) < 2:
+        return
     for model in models:
         core = model.module if hasattr(model, "module") else model
         if getattr(core, "share_embeddings_and_output_weights", False):
             w = core.shared_embedding_or_output_weight()
             if w is not None and w.requires_grad:
                 g = _grad_of(w)
-                if g is not None and grid.group("embd") is not None and len(grid.ranks("embd")) > 1:
-                    dist.all_reduce(g.data, group=grid.group("embd"))
+                if g is not None:
+                    dist.all_reduce(g.data, group=group)
+        # MTP with untied embeddings: the first stage's input embedding and
+        # the last stage's mtp_embedding replica also form a tied pair
+        # (tied models already covered above: mtp_embedding IS output weight)
+        cfg = getattr(core, "config", None)
+        if cfg is not None and getattr(cfg, "mtp_num_layers", 0) and not getattr(
+                core, "share_embeddings_and_output_weights", False):
+            w = None
+            if getattr(core, "pre_process", False) and getattr(core, "embedding", None) is not None:
+                w = core.embedding.weight
+            elif getattr(core, "mtp_embedding", None) is not None:
+                w = core.mtp_embedding.weight
+            if w is not None and w.requires_grad:
+                g = _grad_of(w)
+                if g is not None:
+                    dist.all_reduce(g.data, group=group)
 
 
 def update_router_expert_bias(models: List[torch.nn.Module], config):

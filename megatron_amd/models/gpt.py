@@ -61,13 +61,25 @@ class GPTModel(nn.Module):
             if self.share_embeddings_and_output_weights and pre_process:
                 self.output_layer.weight = self.embedding.weight
             self.mtp = None
+            self.mtp_embedding = None
             if config.mtp_num_layers:
-                assert pre_process, "MTP v1 needs the embedding on this stage (pp=1)"
                 from megatron_amd.transformer.multi_token_prediction import (
                     MultiTokenPredictionBlock,
                 )
 
                 self.mtp = MultiTokenPredictionBlock(config)
+                if not pre_process:
+                    # PP>1: the last stage needs a word-embedding replica for
+                    # the MTP token re-embed; its grads are summed over the
+                    # embd group and its value is broadcast-synced at setup
+                    # (reference multi_token_prediction.py + finalize:164)
+                    self.mtp_embedding = VocabParallelEmbedding(
+                        config.vocab_size, config.hidden_size, config=config)
+                    self.mtp_embedding.weight.is_embedding_or_output_parameter = True
+                    self.mtp_embedding.weight.muon_exclude = True
+                    if self.share_embeddings_and_output_weights:
+                        # tied: the output weight on this stage IS the replica
+                        self.mtp_embedding.weight = self.output_layer.weight
         self._rope_cache = {}
         # set by pipeline runner between stages
         self.input_tensor: Optional[torch.Tensor] = None
@@ -172,7 +184,8 @@ class GPTModel(nn.Module):
         labels_sb = labels.transpose(0, 1).contiguous()  # [s, b]
         loss = vocab_parallel_cross_entropy(logits, labels_sb)
         if self.mtp is not None and input_ids is not None:
+            emb = self.embedding if self.pre_process else self.mtp_embedding
             loss = loss + self.mtp(
-                hidden, input_ids, labels, self.embedding, self.output_layer,
+                hidden, input_ids, labels, emb, self.output_layer,
                 rotary, vocab_parallel_cross_entropy)
         return loss

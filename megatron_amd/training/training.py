@@ -55,11 +55,48 @@ def setup_model_and_optimizer(
     ddp_chunks = [DistributedDataParallel(config, ddp_config, m) for m in chunks]
     for c in ddp_chunks:
         c.broadcast_params()
+    _sync_embd_replicas(ddp_chunks, config)
     optimizer = get_optimizer(opt_config, ddp_chunks)
 
     # wire the schedule hooks (reference three-hook contract)
     config.finalize_model_grads_func = finalize_model_grads
     return ddp_chunks, optimizer
+
+
+def _sync_embd_replicas(chunks, config):
+    """First/last-PP-stage embedding replicas (tied output weight, MTP
+    embedding) start from the first stage's values (reference
+    setup_embeddings_and_output_layer).  The broadcast count is derived from
+    config so every embd-group member issues matching collectives."""
+    import torch.distributed as dist
+
+    if not (G.grid_initialized() and dist.is_initialized()):
+        return
+    grid = G.get_grid()
+    if grid.pp == 1:
+        return
+    group = grid.group("embd")
+    ranks = grid.ranks("embd")
+    if group is None or len(ranks) < 2:
+        return
+    tied = not config.untie_embeddings_and_output_weights
+    mtp_untied = bool(getattr(config, "mtp_num_layers", 0)) and not tied
+    for c in chunks:
+        core = c.module if hasattr(c, "module") else c
+        if not (getattr(core, "pre_process", False) or getattr(core, "post_process", False)):
+            continue
+        if tied:
+            w = core.shared_embedding_or_output_weight()
+            if w is not None:
+                dist.broadcast(w.data, src=ranks[0], group=group)
+        if mtp_untied:
+            w = None
+            if getattr(core, "pre_process", False) and getattr(core, "embedding", None) is not None:
+                w = core.embedding.weight
+            elif getattr(core, "mtp_embedding", None) is not None:
+                w = core.mtp_embedding.weight
+            if w is not None:
+                dist.broadcast(w.data, src=ranks[0], group=group)
 
 
 def train_step(
