@@ -81,8 +81,9 @@ def _fill_mtp_deterministic(model):
         fill(core.decoder.final_layernorm.weight, "final_ln")
         core.decoder.final_layernorm.weight.data.add_(1.0)
         fill(core.output_layer.weight, "output")
-        if core.mtp_embedding is not None:
+        if core.mtp_embedding is not None and not core.share_embeddings_and_output_weights:
             fill(core.mtp_embedding.weight, "embedding")  # replica of stage-0 weight
+            # (tied: mtp_embedding IS the output weight, already filled)
         for k, head in enumerate(core.mtp.heads):
             for name, p in head.named_parameters():
                 fill(p, f"mtp{k}.{name}")
