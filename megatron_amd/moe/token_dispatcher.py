@@ -156,17 +156,56 @@ class MoEAlltoAllTokenDispatcher:
         return unpermute(expert_out, self._sort_order, self._probs, self._T)
 
 
+class _GatherTokensEP(torch.autograd.Function):
+    """all-gather along dim 0 over the EP group; backward reduce-scatters."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        world = torch.distributed.get_world_size(group=group)
+        out = torch.empty(world * x.shape[0], *x.shape[1:], dtype=x.dtype, device=x.device)
+        torch.distributed.all_gather_into_tensor(out, x.contiguous(), group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        world = torch.distributed.get_world_size(group=ctx.group)
+        out = torch.empty(g.shape[0] // world, *g.shape[1:], dtype=g.dtype, device=g.device)
+        torch.distributed.reduce_scatter_tensor(out, g.contiguous(), group=ctx.group)
+        return out, None
+
+
+class _ReduceScatterTokensEP(torch.autograd.Function):
+    """reduce-scatter along dim 0 over the EP group; backward all-gathers."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        world = torch.distributed.get_world_size(group=group)
+        out = torch.empty(x.shape[0] // world, *x.shape[1:], dtype=x.dtype, device=x.device)
+        torch.distributed.reduce_scatter_tensor(out, x.contiguous(), group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        world = torch.distributed.get_world_size(group=ctx.group)
+        out = torch.empty(world * g.shape[0], *g.shape[1:], dtype=g.dtype, device=g.device)
+        torch.distributed.all_gather_into_tensor(out, g.contiguous(), group=ctx.group)
+        return out, None
+
+
 class MoEAllGatherTokenDispatcher:
-    """Simpler dispatcher: every rank computes its local experts on the full
-    (DP-local) token set using the routing map; no A2A.  Used as the
-    equivalence oracle for the alltoall dispatcher (reference :230)."""
+    """Every rank all-gathers the EP group's tokens, computes its LOCAL
+    experts on the full set, and reduce-scatters the prob-weighted combine
+    (reference MoEAllGatherTokenDispatcher :230).  At ep=1 it degrades to a
+    pure local permute and doubles as the alltoall equivalence oracle."""
 
     def __init__(self, config):
         self.config = config
         self.num_experts = config.num_experts
         self.ep = G.get_expert_model_parallel_world_size() if G.grid_initialized() else 1
         self.num_local_experts = self.num_experts // max(self.ep, 1)
-        assert self.ep == 1, "allgather dispatcher here supports ep=1 (oracle use)"
+        self.group = G.get_grid().group("ep") if (G.grid_initialized() and self.ep > 1) else None
 
     def _keep_mask(self, tokens, probs, top_idx):
         cf = getattr(self.config, "moe_expert_capacity_factor", None)
@@ -180,10 +219,36 @@ class MoEAllGatherTokenDispatcher:
                                     getattr(self.config, "moe_token_drop_policy", "probs"))
 
     def dispatch(self, tokens, probs, top_idx):
+        if self.group is None:
+            permuted, sort_order, tokens_per_expert = permute(
+                tokens, top_idx, self.num_experts, self._keep_mask(tokens, probs, top_idx))
+            self._sort_order, self._T, self._probs = sort_order, tokens.shape[0], probs
+            return permuted, tokens_per_expert
+
+        # EP>1: every rank sees the group's full token set and runs only its
+        # local experts; tokens/probs gather differentiably (router grads
+        # flow back to the owning rank via the backward reduce-scatter)
+        rank = torch.distributed.get_rank(group=self.group)
+        full_tokens = _GatherTokensEP.apply(tokens, self.group)          # [ep*T, h]
+        full_probs = _GatherTokensEP.apply(probs, self.group)            # [ep*T, k]
+        full_idx = torch.empty(self.ep * top_idx.shape[0], top_idx.shape[1],
+                               dtype=top_idx.dtype, device=top_idx.device)
+        torch.distributed.all_gather_into_tensor(full_idx, top_idx.contiguous(), group=self.group)
+
+        keep = self._keep_mask(full_tokens, full_probs, full_idx)
+        lo = rank * self.num_local_experts
+        hi = lo + self.num_local_experts
+        mine = ((full_idx >= lo) & (full_idx < hi)).reshape(-1)
+        keep = mine if keep is None else (keep & mine)
         permuted, sort_order, tokens_per_expert = permute(
-            tokens, top_idx, self.num_experts, self._keep_mask(tokens, probs, top_idx))
-        self._sort_order, self._T, self._probs = sort_order, tokens.shape[0], probs
-        return permuted, tokens_per_expert
+            full_tokens, full_idx, self.num_experts, keep)
+        self._sort_order = sort_order
+        self._T = full_tokens.shape[0]
+        self._probs = full_probs
+        return permuted, tokens_per_expert[lo:hi]
 
     def combine(self, expert_out):
-        return unpermute(expert_out, self._sort_order, self._probs, self._T)
+        out = unpermute(expert_out, self._sort_order, self._probs, self._T)
+        if self.group is None:
+            return out
+        return _ReduceScatterTokensEP.apply(out, self.group)

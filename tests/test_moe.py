@@ -454,3 +454,57 @@ def test_tp2_ep2_matches_single(tmp_path, monkeypatch):
     spawn_dist(_tp2ep2_case, 4)
     got = torch.load(out_path)
     assert_close(got, ref.detach(), rtol=1e-4, atol=1e-5)
+
+
+def _ag_ep_case(rank, world):
+    """allgather dispatcher at EP>1 (reference token_dispatcher.py:230)."""
+    G.initialize_model_parallel(expert_parallel_size=world)
+    model_parallel_seed(1234)
+    cfg = _cfg(ep=world, moe_token_dispatcher_type="allgather")
+    layer = MoELayer(cfg)
+
+    def fullfill(shape, key):
+        g = torch.Generator().manual_seed(zlib.crc32(key.encode()) % (2**31))
+        return torch.randn(shape, generator=g) * 0.1
+
+    n_local = cfg.num_experts // world
+    with torch.no_grad():
+        layer.router.weight.copy_(fullfill((4, 32), "router"))
+        w1 = fullfill((4, 96, 32), "w1")
+        w2 = fullfill((4, 32, 48), "w2")
+        layer.experts.weight1.copy_(w1[rank * n_local : (rank + 1) * n_local])
+        layer.experts.weight2.copy_(w2[rank * n_local : (rank + 1) * n_local])
+    torch.manual_seed(1)
+    x = torch.randn(6, 2, 32, requires_grad=True)
+    out = layer(x)
+    out.sum().backward()
+    assert layer.router.weight.grad is not None
+    assert layer.experts.weight1.grad is not None and layer.experts.weight1.grad.abs().sum() > 0
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+    if rank == 0:
+        torch.save(out.detach(), os.environ["MOE_TEST_OUT"])
+
+
+def test_allgather_dispatcher_ep2_matches_ep1(tmp_path, monkeypatch):
+    out_path = tmp_path / "moe_ag.pt"
+    monkeypatch.setenv("MOE_TEST_OUT", str(out_path))
+
+    init_single()
+    cfg = _cfg()
+    layer = MoELayer(cfg)
+
+    def fullfill(shape, key):
+        g = torch.Generator().manual_seed(zlib.crc32(key.encode()) % (2**31))
+        return torch.randn(shape, generator=g) * 0.1
+
+    with torch.no_grad():
+        layer.router.weight.copy_(fullfill((4, 32), "router"))
+        layer.experts.weight1.copy_(fullfill((4, 96, 32), "w1"))
+        layer.experts.weight2.copy_(fullfill((4, 32, 48), "w2"))
+    torch.manual_seed(1)
+    x = torch.randn(6, 2, 32)
+    ref = layer(x)
+
+    spawn_dist(_ag_ep_case, 2)
+    ag_out = torch.load(out_path)
+    assert_close(ref.detach(), ag_out, rtol=1e-5, atol=1e-6)
