@@ -133,10 +133,16 @@ class _ParallelLinearFn(torch.autograd.Function):
             # async all-gather of the (seq-sharded) input for the wgrad GEMM,
             # overlapped with the dgrad GEMM below (reference layers.py:609-616)
             from megatron_amd.parallel.mappings import _world
+            from megatron_amd.parallel.memory_buffer import get_global_memory_buffer
+
             world = _world(group)
             shape = list(inp.shape)
             shape[0] *= world
-            total_input = torch.empty(shape, dtype=inp.dtype, device=inp.device)
+            # reuse the global scratch (a fresh torch.empty per microbatch
+            # churned the allocator on the hot path; distinct tag from the
+            # forward AG buffer, which may still be alive in this step)
+            total_input = get_global_memory_buffer().get_tensor(
+                shape, inp.dtype, "sp-ag-bwd", device=inp.device)
             gather_handle = dist.all_gather_into_tensor(
                 total_input, inp.contiguous(), group=group, async_op=True
             )
