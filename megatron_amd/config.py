@@ -75,6 +75,7 @@ class TransformerConfig(ParallelConfig):
     # None or {'type': 'llama3'|'linear', 'factor': ..., 'low_freq_factor': ...,
     # 'high_freq_factor': ..., 'original_max_position_embeddings': ...}
     rope_scaling: Optional[dict] = None
+    label_smoothing: float = 0.0
     attention_dropout: float = 0.0
     hidden_dropout: float = 0.0
     # sliding-window attention: None or window size (causal look-back)

@@ -182,7 +182,8 @@ class GPTModel(nn.Module):
         if labels is None:
             return logits
         labels_sb = labels.transpose(0, 1).contiguous()  # [s, b]
-        loss = vocab_parallel_cross_entropy(logits, labels_sb)
+        loss = vocab_parallel_cross_entropy(logits, labels_sb,
+                                            label_smoothing=self.config.label_smoothing)
         if self.mtp is not None and input_ids is not None:
             emb = self.embedding if self.pre_process else self.mtp_embedding
             loss = loss + self.mtp(

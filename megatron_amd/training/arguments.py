@@ -43,6 +43,10 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--hidden-dropout", type=float, default=0.0)
     g.add_argument("--window-size", type=int, default=None)
     g.add_argument("--qk-layernorm", action="store_true")
+    g.add_argument("--qk-clip-threshold", type=float, default=None,
+                   help="per-head qk-logit clip tracking threshold (muon/qk-clip)")
+    g.add_argument("--rope-scaling", action="store_true",
+                   help="llama-3.1-style rope frequency scaling")
     g.add_argument("--init-method-std", type=float, default=0.02)
     g.add_argument("--mtp-num-layers", type=int, default=0)
     g.add_argument("--multi-latent-attention", action="store_true")
@@ -77,6 +81,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--moe-aux-loss-type", choices=["aux", "seq_aux"], default="aux")
     g.add_argument("--moe-router-enable-expert-bias", action="store_true", default=False)
     g.add_argument("--moe-router-bias-update-rate", type=float, default=1e-3)
+    g.add_argument("--moe-expert-capacity-factor", type=float, default=None,
+                   help="None = dropless; otherwise capacity = ceil(T*topk/E * factor)")
+    g.add_argument("--moe-token-drop-policy", choices=["probs", "position"], default="probs")
 
     g = p.add_argument_group("parallelism")
     g.add_argument("--tensor-model-parallel-size", "--tp", type=int, default=1)
@@ -91,6 +98,12 @@ def build_arg_parser() -> argparse.ArgumentParser:
                    help="yaml of per-group RCCL knobs (min/max CTAs, stream priority)")
     g.add_argument("--tp-comm-overlap", action="store_true",
                    help="SP forward AG as a chunked ring overlapped with the GEMM")
+    g.add_argument("--use-rccl-registered-buffers", action="store_true",
+                   help="allocate DDP buffers in an RCCL-registered pool (zero-copy xGMI)")
+    g.add_argument("--hip-graphs", action="store_true",
+                   help="capture per-layer fwd/bwd hipGraphs (static shapes, dense layers)")
+    g.add_argument("--use-fsdp", action="store_true",
+                   help="Megatron-FSDP-style per-layer param/grad sharding instead of DDP+ZeRO-1")
 
     g = p.add_argument_group("training")
     g.add_argument("--micro-batch-size", type=int, default=1)
@@ -112,6 +125,7 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--fp8-amax-history-len", type=int, default=16)
     g.add_argument("--fp8-margin", type=int, default=0)
     g.add_argument("--fp16", action="store_true")
+    g.add_argument("--label-smoothing", type=float, default=0.0)
     g.add_argument("--seed", type=int, default=1234)
     g.add_argument("--deterministic-mode", action="store_true")
 
@@ -269,6 +283,12 @@ def configs_from_args(args):
         moe_aux_loss_type=args.moe_aux_loss_type,
         moe_router_enable_expert_bias=args.moe_router_enable_expert_bias,
         moe_router_bias_update_rate=args.moe_router_bias_update_rate,
+        moe_expert_capacity_factor=args.moe_expert_capacity_factor,
+        moe_token_drop_policy=args.moe_token_drop_policy,
+        qk_clip_threshold=args.qk_clip_threshold,
+        label_smoothing=args.label_smoothing,
+        rope_scaling={"factor": 8.0, "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192} if args.rope_scaling else None,
         tensor_parallel_size=args.tensor_model_parallel_size,
         pipeline_parallel_size=args.pipeline_model_parallel_size,
         virtual_pipeline_parallel_size=args.virtual_pipeline_model_parallel_size,
@@ -307,6 +327,7 @@ def configs_from_args(args):
         overlap_grad_reduce=args.overlap_grad_reduce,
         use_distributed_optimizer=args.use_distributed_optimizer,
         bucket_size=args.bucket_size,
+        use_rccl_registered_buffers=args.use_rccl_registered_buffers,
     )
     return cfg, opt_cfg, ddp_cfg
 
@@ -344,7 +365,12 @@ def parse_and_validate_args(argv=None):
     args = parser.parse_args(argv)
     if getattr(args, "yaml_cfg", None):
         apply_yaml_config(args, args.yaml_cfg, parser)
-    args.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    import torch.distributed as _dist
+
+    if _dist.is_initialized():
+        args.world_size = _dist.get_world_size()
+    else:
+        args.world_size = int(os.environ.get("WORLD_SIZE", "1"))
     args.rank = int(os.environ.get("RANK", "0"))
     validate_args(args)
     return args
