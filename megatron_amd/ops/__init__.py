@@ -209,6 +209,43 @@ def wgrad_gemm_accum(main_grad: torch.Tensor, grad_output_2d: torch.Tensor, inpu
 
 
 # ---------------------------------------------------------------------------
+# Mamba causal conv1d + SiLU  (K14)
+# ---------------------------------------------------------------------------
+
+
+class _CausalConv1dSiluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        out, pre = _C.causal_conv1d_fwd(x, weight, bias)
+        ctx.save_for_backward(x, pre, weight)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, pre, weight = ctx.saved_tensors
+        dx, dw, db = _C.causal_conv1d_bwd(dy, x, pre, weight)
+        return dx, dw.view(weight.shape).to(weight.dtype), db.to(weight.dtype)
+
+
+def causal_conv1d_silu(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
+    """silu(depthwise causal conv) over channel-last x [b, l, C];
+    weight [C, 1, K], bias [C].  HIP kernel for K=4 bf16; torch fallback
+    (F.conv1d + silu) otherwise."""
+    if (
+        _use_native(x)
+        and x.dtype == torch.bfloat16
+        and weight.shape[-1] == 4
+        and x.shape[-1] % 8 == 0
+    ):
+        return _CausalConv1dSiluFn.apply(x.contiguous(), weight, bias)
+    import torch.nn.functional as F
+
+    y = F.conv1d(x.transpose(1, 2), weight, bias, groups=x.shape[-1],
+                 padding=weight.shape[-1] - 1)[..., : x.shape[1]]
+    return F.silu(y.transpose(1, 2))
+
+
+# ---------------------------------------------------------------------------
 # grouped GEMM for MoE experts  (K11)
 # ---------------------------------------------------------------------------
 

@@ -24,6 +24,9 @@ void ce_bwd(torch::Tensor logits, torch::Tensor target, torch::Tensor row_max,
 std::string grouped_gemm_probe();
 torch::Tensor grouped_gemm(torch::Tensor a, torch::Tensor b, std::vector<int64_t> sizes, bool trans_b);
 void grouped_gemm_wgrad(torch::Tensor dy, torch::Tensor x, std::vector<int64_t> sizes, torch::Tensor dw);
+std::vector<torch::Tensor> causal_conv1d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
+std::vector<torch::Tensor> causal_conv1d_bwd(torch::Tensor dy, torch::Tensor x,
+                                             torch::Tensor pre, torch::Tensor w);
 std::vector<int64_t> symm_ipc_handle(torch::Tensor buf);
 int64_t symm_open_handle(std::vector<int64_t> bytes);
 void symm_close_handle(int64_t ptr);
@@ -48,6 +51,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grouped_gemm", &grouped_gemm);
   m.def("grouped_gemm_probe", &grouped_gemm_probe);
   m.def("grouped_gemm_wgrad", &grouped_gemm_wgrad);
+  m.def("causal_conv1d_fwd", &causal_conv1d_fwd);
+  m.def("causal_conv1d_bwd", &causal_conv1d_bwd);
   m.def("symm_ipc_handle", &symm_ipc_handle);
   m.def("symm_open_handle", &symm_open_handle);
   m.def("symm_close_handle", &symm_close_handle);

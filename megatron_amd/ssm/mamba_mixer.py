@@ -126,12 +126,11 @@ class MambaMixer(nn.Module):
     # ---- helpers -------------------------------------------------------
 
     def _conv(self, xBC: torch.Tensor) -> torch.Tensor:
-        """Depthwise causal conv over the sequence. xBC: [b, l, conv_dim]."""
-        y = F.conv1d(
-            xBC.transpose(1, 2), self.conv_weight, self.conv_bias,
-            groups=self.conv_dim_local, padding=self.d_conv - 1,
-        )[..., : xBC.shape[1]]
-        return F.silu(y.transpose(1, 2))
+        """Depthwise causal conv + SiLU over the sequence (fused HIP kernel
+        K14 on GPU; MIOpen F.conv1d fallback). xBC: [b, l, conv_dim]."""
+        from megatron_amd import ops
+
+        return ops.causal_conv1d_silu(xBC, self.conv_weight, self.conv_bias)
 
     def allocate_inference_state(self, batch: int, device, dtype) -> dict:
         return {

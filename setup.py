@@ -22,6 +22,7 @@ SRC = [
     "megatron_amd/ops/csrc/wgrad.hip",
     "megatron_amd/ops/csrc/cross_entropy.hip",
     "megatron_amd/ops/csrc/grouped_gemm.cpp",
+    "megatron_amd/ops/csrc/causal_conv1d.hip",
     "megatron_amd/ops/csrc/attention_fwd.hip",
     "megatron_amd/ops/csrc/attention_bwd.hip",
     "megatron_amd/ops/csrc/symm_allreduce.hip",

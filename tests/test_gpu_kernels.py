@@ -301,3 +301,32 @@ def test_grouped_gemm_wgrad_accumulates_into_main_grad():
         expect[e] = dy[start : start + m].t().float() @ a[start : start + m].detach().float()
         start += m
     assert _rel_err(w.main_grad, expect) < 3e-2
+
+
+# --- Mamba causal conv1d + SiLU (K14) ---------------------------------------
+
+
+@pytest.mark.parametrize("b,l,C", [(2, 128, 256), (1, 333, 64), (4, 64, 1024)])
+def test_causal_conv1d_silu_fwd_bwd(b, l, C):
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    K = 4
+    x = torch.randn(b, l, C, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(C, 1, K, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    bias = torch.randn(C, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = ops.causal_conv1d_silu(x, w, bias)
+
+    x2 = x.detach().clone().float().requires_grad_(True)
+    w2 = w.detach().clone().float().requires_grad_(True)
+    b2 = bias.detach().clone().float().requires_grad_(True)
+    y = F.conv1d(x2.transpose(1, 2), w2, b2, groups=C, padding=K - 1)[..., :l]
+    out_ref = F.silu(y.transpose(1, 2))
+    assert _rel_err(out, out_ref) < 2e-2
+
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    out_ref.backward(dy.float())
+    assert _rel_err(x.grad, x2.grad) < 3e-2
+    assert _rel_err(w.grad, w2.grad) < 3e-2
+    assert _rel_err(bias.grad, b2.grad) < 3e-2
