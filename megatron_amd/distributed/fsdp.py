@@ -107,6 +107,8 @@ class _FSDPUnit:
 
     def unshard(self, async_op: bool = False):
         if self._flat is not None:
+            if self._ag_handle is not None and not async_op:
+                self.finish_unshard()  # a prefetch is in flight: just wait
             return
         self._flat = torch.empty(self.flat_size, dtype=self.dtype, device=self.model_shard.device)
         h = dist.all_gather_into_tensor(self._flat, self.model_shard, group=self.dp_group,
@@ -238,8 +240,14 @@ class FullyShardedDataParallel(nn.Module):
                     p.register_post_accumulate_grad_hook(self._grad_hook(u))
 
     def _fwd_pre(self, u):
+        idx = self.units.index(u)
+
         def hook(mod, args):
             u.unshard()
+            # AG-prefetch pipeline (reference megatron_fsdp overlap): kick
+            # the NEXT unit's all-gather so it overlaps this unit's compute
+            if idx + 1 < len(self.units):
+                self.units[idx + 1].unshard(async_op=True)
         return hook
 
     def _fwd_post(self, u):
@@ -251,8 +259,12 @@ class FullyShardedDataParallel(nn.Module):
         return hook
 
     def _bwd_pre(self, u):
+        idx = self.units.index(u)
+
         def hook(mod, grad_out):
             u.unshard()
+            if idx - 1 >= 0:
+                self.units[idx - 1].unshard(async_op=True)  # bwd runs in reverse
         return hook
 
     def _grad_hook(self, u):

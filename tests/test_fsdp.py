@@ -132,3 +132,36 @@ def _sd_worker(rank, world):
 
 def test_fsdp_state_dict():
     spawn_dist(_sd_worker, world_size=2)
+
+
+def _prefetch_case(rank, world):
+    """AG-prefetch: forward of unit i must launch unit i+1's async gather."""
+    import torch.nn as nn
+
+    from megatron_amd.distributed.fsdp import FullyShardedDataParallel, _FSDPUnit
+    from megatron_amd.parallel import grid as G
+
+    G.initialize_model_parallel()
+    torch.manual_seed(7)
+    model = nn.Sequential(nn.Linear(32, 32), nn.Linear(32, 32), nn.Linear(32, 32))
+    async_calls = []
+    orig = _FSDPUnit.unshard
+
+    def spy(self, async_op=False):
+        if async_op and self._flat is None:
+            async_calls.append(self.name)
+        return orig(self, async_op)
+
+    _FSDPUnit.unshard = spy
+    try:
+        fsdp = FullyShardedDataParallel(model, unit_classes=(nn.Linear,),
+                                        reshard_after_forward=True)
+        x = torch.randn(8, 32)
+        fsdp(x).sum().backward()
+    finally:
+        _FSDPUnit.unshard = orig
+    assert len(async_calls) >= 2, async_calls  # fwd prefetches 1,2; bwd prefetches back
+
+
+def test_fsdp_ag_prefetch_pipeline():
+    spawn_dist(_prefetch_case, 2)
