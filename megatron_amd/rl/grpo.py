@@ -37,14 +37,17 @@ def grpo_loss(
     clip_ratio: float = 0.2,
     ref_logprobs: Optional[torch.Tensor] = None,
     kl_coeff: float = 0.0,
+    advantages_per_token: bool = False,
 ) -> torch.Tensor:
     """Clipped surrogate averaged over response tokens.
 
     logprobs/behavior_logprobs/ref_logprobs: [N, T] per-token logprob of the
-    chosen token; advantages [N]; response_mask [N, T] (1 on response tokens).
+    chosen token; advantages [N] (or [N, T] with advantages_per_token, the
+    THD-packed layout where every rollout's advantage is pre-spread along
+    the pack); response_mask [N, T] (1 on response tokens).
     """
     ratio = torch.exp(logprobs - behavior_logprobs)
-    adv = advantages.unsqueeze(-1)
+    adv = advantages if advantages_per_token else advantages.unsqueeze(-1)
     unclipped = ratio * adv
     clipped = torch.clamp(ratio, 1.0 - clip_ratio, 1.0 + clip_ratio) * adv
     obj = torch.minimum(unclipped, clipped)

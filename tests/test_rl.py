@@ -11,7 +11,12 @@ from megatron_amd.rl.grpo import group_relative_advantages, grpo_loss
 from megatron_amd.rl.loop import pack_rollouts, policy_logprobs, rl_step
 from megatron_amd.rl.rollout import Rollout
 
+from megatron_amd.parallel.random import model_parallel_seed
 from tests.utils import assert_close, init_single
+
+MODEL_KW = dict(num_layers=2, hidden_size=64, num_attention_heads=4,
+                num_query_groups=2, vocab_size=64, ffn_hidden_size=96,
+                max_position_embeddings=128, gradient_accumulation_fusion=False)
 
 
 def test_group_relative_advantages():
@@ -126,3 +131,62 @@ def test_refit_updates_rollout_model():
         expect = train_model(toks, position_ids=None, attention_mask=None)
     assert not torch.allclose(after, before, atol=1e-5)
     assert torch.allclose(after, expect, atol=1e-5)
+
+
+def test_packed_rollout_logprobs_match_padded():
+    """THD-packed policy recompute == right-padded recompute (the packing
+    must not change any rollout's logprobs)."""
+    import torch
+
+    from megatron_amd.rl.loop import (
+        pack_rollouts,
+        pack_rollouts_thd,
+        policy_logprobs,
+        policy_logprobs_packed,
+    )
+    from megatron_amd.rl.rollout import Rollout
+
+    init_single()
+    model_parallel_seed(11)
+    cfg = TransformerConfig(**MODEL_KW)
+    model = GPTModel(cfg)
+    model.eval()
+    rollouts = [
+        Rollout(prompt_tokens=[5, 6, 7], response_tokens=[8, 9], behavior_logprobs=[-1.0, -1.1]),
+        Rollout(prompt_tokens=[3, 4], response_tokens=[1, 2, 10], behavior_logprobs=[-0.5, -0.6, -0.7]),
+    ]
+    dev = torch.device("cpu")
+    ids, chosen, mask, _ = pack_rollouts(rollouts, dev)
+    with torch.no_grad():
+        lp_pad = policy_logprobs(model, ids, chosen)
+    pids, pchosen, pmask, _, psp = pack_rollouts_thd(rollouts, dev)
+    with torch.no_grad():
+        lp_pack = policy_logprobs_packed(model, pids, pchosen, psp)
+    # compare masked positions rollout by rollout
+    off = 0
+    for i, r in enumerate(rollouts):
+        n = len(r.prompt_tokens) + len(r.response_tokens)
+        sel_pack = lp_pack[0, off : off + n][pmask[0, off : off + n] > 0]
+        sel_pad = lp_pad[i][mask[i] > 0]
+        torch.testing.assert_close(sel_pack, sel_pad, rtol=1e-4, atol=1e-5)
+        off += n
+
+
+def test_rl_step_packed_runs():
+    import torch
+
+    from megatron_amd.rl.loop import rl_step
+
+    init_single()
+    model_parallel_seed(12)
+    cfg = TransformerConfig(**MODEL_KW)
+    model = GPTModel(cfg)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-4)
+
+    def env(prompt, resp):
+        return float(len(set(resp)))
+
+    loss, reward, rollouts = rl_step(model, opt, [[3, 4, 5], [6, 7]], env,
+                                     group_size=2, max_tokens=4, packed=True)
+    assert torch.isfinite(torch.tensor(loss))
+    assert len(rollouts) == 4
