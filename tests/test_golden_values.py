@@ -173,3 +173,40 @@ def test_tiny_gpt_pp2_golden(tmp_path, monkeypatch):
     spawn_dist(_dist_golden_case, 2,
                ["--pipeline-model-parallel-size", "2", "--train-iters", "8"], "pp2")
     _check_or_record("tiny_pp2.json", json.load(open(out))["lm_loss"])
+
+
+def test_tiny_bert_loss_curve_matches_golden():
+    def provider(config, pre_process=True, post_process=True, vp_stage=None):
+        from megatron_amd.models.bert import BertModel
+
+        return BertModel(config, pre_process=pre_process, post_process=post_process)
+
+    losses = _run_config(["--position-embedding-type", "learned",
+                          "--normalization", "layernorm", "--activation", "gelu"],
+                         provider=provider)
+    _check_or_record("tiny_bert.json", losses)
+
+
+def test_tiny_t5_loss_curve_matches_golden():
+    import pretrain_t5 as T5E
+
+    losses = []
+    orig = P.train_step
+
+    def wrapped(*a, **k):
+        r = orig(*a, **k)
+        losses.append(r["lm_loss"])
+        return r
+
+    P.train_step = wrapped
+    try:
+        P.pretrain(T5E.model_provider, [
+            "--num-layers", "2", "--hidden-size", "64", "--num-attention-heads", "4",
+            "--num-query-groups", "2", "--ffn-hidden-size", "128", "--seq-length", "64",
+            "--micro-batch-size", "2", "--global-batch-size", "4", "--vocab-size", "256",
+            "--mock-data", "--train-iters", "8", "--log-interval", "0", "--seed", "42",
+            "--deterministic-mode",
+        ], forward_step_builder=T5E.forward_step_builder)
+    finally:
+        P.train_step = orig
+    _check_or_record("tiny_t5.json", losses)
