@@ -317,3 +317,45 @@ def test_async_save_uses_background_process(tmp_path):
                           global_offset=(0, 0))
     S.load({"w": st2}, str(tmp_path / "ck"))
     assert torch.equal(st2.data, st.data)
+
+
+def _dp2_overlap_resume_case(rank, world, ckdir):
+    """DP=2 x dist-opt x overlap_param_gather: resume must be exact
+    (the composition never had a multi-rank test)."""
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    G.initialize_model_parallel()
+    model_parallel_seed(77)
+    cfg = _cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, weight_decay=0.01, clip_grad=1.0,
+                              use_distributed_optimizer=True, overlap_param_gather=True)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, use_distributed_optimizer=True,
+                        overlap_grad_reduce=True, bucket_size=10_000)
+    # per-rank distinct data (dp sharding)
+    batches = _gen_batches(8, seed=7 + rank)
+
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    ref = _train(chunks, opt, cfg, batches, 4, 2)
+
+    G.destroy_model_parallel()
+    G.initialize_model_parallel()
+    model_parallel_seed(77)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    first = _train(chunks, opt, cfg, batches, 2, 2)
+    save_checkpoint(ckdir, chunks, opt, 2)
+
+    G.destroy_model_parallel()
+    G.initialize_model_parallel()
+    model_parallel_seed(123)
+    chunks2, opt2 = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    it = load_checkpoint(ckdir, chunks2, opt2)
+    assert it == 2
+    second = _train(chunks2, opt2, cfg, batches[4:], 2, 2)
+    for a, b in zip(ref, first + second):
+        assert abs(a - b) < 1e-5, (rank, ref, first + second)
+
+
+def test_dp2_dist_opt_overlap_param_gather_resume(tmp_path):
+    from tests.utils import spawn_dist
+
+    spawn_dist(_dp2_overlap_resume_case, 2, str(tmp_path / "ck"))
