@@ -42,10 +42,14 @@ __global__ void symm_barrier_kernel(PeerPtrs p, int world, int rank,
     __atomic_store_n(p.flags[threadIdx.x] + rank * FLAG_STRIDE, seq,
                      __ATOMIC_RELEASE);
   }
-  // wait: everyone announced to me
+  // wait: everyone announced to me.  The spin is bounded (~seconds) so a
+  // protocol bug degrades to a wrong answer the tests catch instead of a
+  // wedged GPU; production stalls this long are real failures anyway.
   if (threadIdx.x < world) {
+    long guard = 0;
     while (__atomic_load_n(p.flags[rank] + threadIdx.x * FLAG_STRIDE,
                            __ATOMIC_ACQUIRE) < seq) {
+      if (++guard > (1L << 33)) break;
     }
   }
   __syncthreads();
