@@ -76,6 +76,7 @@ class TransformerConfig(ParallelConfig):
     # 'high_freq_factor': ..., 'original_max_position_embeddings': ...}
     rope_scaling: Optional[dict] = None
     label_smoothing: float = 0.0
+    overlap_moe_expert_parallel_comm: bool = False  # combined-1F1B co-schedule
     attention_dropout: float = 0.0
     hidden_dropout: float = 0.0
     # sliding-window attention: None or window size (causal look-back)

@@ -122,3 +122,97 @@ def combined_1f1b_step(fwd_plan: ModelChunkSchedulePlan, fwd_input: torch.Tensor
         if bwd_plan is not None and k < n_b:
             g = bwd_plan.backward_node(n_b - 1 - k, g)
     return (x if n_f else None), (g if bwd_plan is not None else None)
+
+
+def forward_backward_no_pipelining_combined(
+    *,
+    forward_step_func,  # unused: the combined path owns the batch->loss plumbing
+    data_iterator,
+    model,
+    num_microbatches: int,
+    seq_length: int = None,
+    micro_batch_size: int = None,
+    forward_only: bool = False,
+    **kw,
+):
+    """pp=1 schedule with layer-granular fwd/bwd co-scheduling
+    (config.overlap_moe_expert_parallel_comm): the forward of microbatch i
+    runs node-interleaved with the backward of microbatch i-1, so each MoE
+    layer's EP all-to-all (on the comm stream) overlaps the neighbor node's
+    GEMMs (reference combined_1f1b.py:35).
+
+    Contract: GPT-family model; data_iterator yields {"tokens", "labels"}
+    [b, s]; per-token loss averaged over the whole batch (the bench/pretrain
+    protocol).  CP/SP stay on the standard path.
+    """
+    import contextlib
+
+    from megatron_amd.moe.router import AuxLossScaler
+    from megatron_amd.parallel.cross_entropy import vocab_parallel_cross_entropy
+
+    if isinstance(model, list):
+        assert len(model) == 1
+        model = model[0]
+    if isinstance(data_iterator, list):
+        data_iterator = data_iterator[0]
+    core = model.module if hasattr(model, "module") else model
+    config = core.config
+    assert config.context_parallel_size == 1 and not config.sequence_parallel, (
+        "combined 1F1B co-schedule supports the dense/MoE pp=1 path; CP/SP use the standard schedule")
+
+    no_sync = config.no_sync_func
+    if no_sync is None and hasattr(model, "no_sync"):
+        no_sync = model.no_sync
+    if no_sync is None:
+        no_sync = contextlib.nullcontext
+
+    losses: List[dict] = []
+    device = next(core.parameters()).device
+    num_tokens_acc = torch.zeros((), dtype=torch.long, device=device)
+
+    def make_plan(batch):
+        tokens, labels = batch["tokens"], batch["labels"]
+        seq = tokens.shape[1]
+        freqs = core._rotary_freqs(seq, tokens.device)
+        labels_sb = labels.transpose(0, 1).contiguous()
+
+        def loss_fn(logits):
+            loss_sb = vocab_parallel_cross_entropy(
+                logits, labels_sb, label_smoothing=config.label_smoothing)
+            ntok = loss_sb.numel()
+            ssum = loss_sb.sum()
+            losses.append({"loss_sum": ssum.detach()})
+            num_tokens_acc.add_(ntok)
+            scale = 1.0 / (max(int(ntok), 1) * num_microbatches)
+            AuxLossScaler.bind_scale(scale)
+            out = ssum * scale
+            if config.grad_scale_func is not None:
+                out = config.grad_scale_func(out)
+            return out
+
+        plan = ModelChunkSchedulePlan.from_gpt(core, rotary_freqs=freqs, loss_fn=loss_fn)
+        return plan, tokens
+
+    with no_sync():
+        prev_plan = None
+        for _ in range(num_microbatches):
+            plan, tokens = make_plan(next(data_iterator))
+            if forward_only:
+                x = tokens
+                for k in range(len(plan)):
+                    x = plan.forward_node(k, x)
+                continue
+            combined_1f1b_step(plan, tokens, prev_plan, None)
+            prev_plan = plan
+        if prev_plan is not None:
+            n = len(prev_plan)
+            g = None
+            for k in range(n):
+                g = prev_plan.backward_node(n - 1 - k, g)
+
+    if not forward_only:
+        if hasattr(model, "start_grad_sync"):
+            model.start_grad_sync()
+        if config.finalize_model_grads_func is not None:
+            config.finalize_model_grads_func([model], config)
+    return losses, num_tokens_acc

@@ -24,7 +24,12 @@ import torch
 from megatron_amd.parallel import grid as G
 
 
-def get_forward_backward_func():
+def get_forward_backward_func(config=None):
+    if (config is not None and getattr(config, "overlap_moe_expert_parallel_comm", False)
+            and not (G.grid_initialized() and G.get_pipeline_model_parallel_world_size() > 1)):
+        from megatron_amd.pipeline.combined_1f1b import forward_backward_no_pipelining_combined
+
+        return forward_backward_no_pipelining_combined
     if G.grid_initialized() and G.get_pipeline_model_parallel_world_size() > 1:
         grid = G.get_grid()
         if grid.vpp is not None and grid.vpp > 1:

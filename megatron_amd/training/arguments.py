@@ -102,6 +102,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
                    help="allocate DDP buffers in an RCCL-registered pool (zero-copy xGMI)")
     g.add_argument("--hip-graphs", action="store_true",
                    help="capture per-layer fwd/bwd hipGraphs (static shapes, dense layers)")
+    g.add_argument("--overlap-moe-expert-parallel-comm", action="store_true",
+                   help="combined-1F1B: layer-granular fwd/bwd co-schedule so MoE EP a2a overlaps compute")
     g.add_argument("--use-fsdp", action="store_true",
                    help="Megatron-FSDP-style per-layer param/grad sharding instead of DDP+ZeRO-1")
 
@@ -287,6 +289,7 @@ def configs_from_args(args):
         moe_token_drop_policy=args.moe_token_drop_policy,
         qk_clip_threshold=args.qk_clip_threshold,
         label_smoothing=args.label_smoothing,
+        overlap_moe_expert_parallel_comm=args.overlap_moe_expert_parallel_comm,
         rope_scaling={"factor": 8.0, "low_freq_factor": 1.0, "high_freq_factor": 4.0,
                       "original_max_position_embeddings": 8192} if args.rope_scaling else None,
         tensor_parallel_size=args.tensor_model_parallel_size,

@@ -114,7 +114,7 @@ def train_step(
     a replay, reference training.py:2395)."""
     rsm = get_rerun_state_machine()
     iters = data_iterator if isinstance(data_iterator, list) else [data_iterator]
-    fb_func = get_forward_backward_func()
+    fb_func = get_forward_backward_func(config)
     losses = num_tokens = None
     while rsm.should_run_forward_backward(iters):
         for chunk in model_chunks:

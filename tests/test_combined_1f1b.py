@@ -78,3 +78,63 @@ def test_combined_1f1b_matches_plain_two_microbatches():
     for name, p in model.named_parameters():
         if name in ref_grads:
             assert torch.allclose(p.grad, ref_grads[name], atol=1e-5), name
+
+
+def test_combined_schedule_matches_standard_training():
+    """The combined pp=1 schedule (overlap_moe_expert_parallel_comm) must
+    produce the same losses and updated weights as the standard schedule."""
+    import copy
+
+    from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from megatron_amd.training.training import setup_model_and_optimizer, train_step
+    from tests.utils import init_single
+
+    def run(combined):
+        init_single()
+        model_parallel_seed(21)
+        cfg = TransformerConfig(
+            num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+            vocab_size=96, ffn_hidden_size=96, num_experts=4, moe_router_topk=2,
+            moe_ffn_hidden_size=64, moe_aux_loss_coeff=0.01,
+            overlap_moe_expert_parallel_comm=combined,
+            gradient_accumulation_fusion=False,
+        )
+        opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+        chunks, opt = setup_model_and_optimizer(_seeded_provider, cfg, opt_cfg,
+                                                DDPConfig(grad_reduce_in_fp32=True))
+        g = torch.Generator().manual_seed(9)
+        batches = []
+        for _ in range(4):
+            t = torch.randint(0, 96, (2, 17), generator=g)
+            batches.append({"tokens": t[:, :-1], "labels": t[:, 1:]})
+
+        def fwd(it, model):
+            batch = next(it)
+
+            def loss_func(loss_sb):
+                s = loss_sb.sum()
+                return s, torch.tensor(loss_sb.numel()), {"loss_sum": s.detach()}
+
+            return model(batch["tokens"], labels=batch["labels"]), loss_func
+
+        losses = []
+        for s in range(2):
+            it = iter(batches[s * 2 : (s + 1) * 2])
+            r = train_step(fwd, [it], chunks, opt, cfg, 2, 16, 2)
+            losses.append(r["lm_loss"])
+        params = {n: p.detach().clone() for n, p in chunks[0].module.named_parameters()}
+        return losses, params
+
+    def _seeded_provider(config, pre_process=True, post_process=True, vp_stage=None):
+        torch.manual_seed(42)
+        return GPTModel(config, pre_process=pre_process, post_process=post_process)
+
+    globals()["_seeded_provider"] = _seeded_provider
+    l_std, p_std = run(False)
+    l_cmb, p_cmb = run(True)
+    for a, b in zip(l_std, l_cmb):
+        assert abs(a - b) < 1e-5, (l_std, l_cmb)
+    for n in p_std:
+        torch.testing.assert_close(p_cmb[n], p_std[n], rtol=1e-5, atol=1e-6)
