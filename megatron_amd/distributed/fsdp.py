@@ -127,6 +127,10 @@ class _FSDPUnit:
     def reshard(self):
         if self._flat is None:
             return
+        if self._ag_handle is not None:
+            # never free a buffer an async gather may still be writing
+            self._ag_handle.wait()
+            self._ag_handle = None
         self._set_stub_views()
         self._flat = None
 
@@ -246,7 +250,8 @@ class FullyShardedDataParallel(nn.Module):
             u.unshard()
             # AG-prefetch pipeline (reference megatron_fsdp overlap): kick
             # the NEXT unit's all-gather so it overlaps this unit's compute
-            if idx + 1 < len(self.units):
+            # (no-op at dp=1: nothing to gather, skip the async churn)
+            if idx + 1 < len(self.units) and u.dp_size > 1:
                 self.units[idx + 1].unshard(async_op=True)
         return hook
 
@@ -263,7 +268,7 @@ class FullyShardedDataParallel(nn.Module):
 
         def hook(mod, grad_out):
             u.unshard()
-            if idx - 1 >= 0:
+            if idx - 1 >= 0 and u.dp_size > 1:
                 self.units[idx - 1].unshard(async_op=True)  # bwd runs in reverse
         return hook
 

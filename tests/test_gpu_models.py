@@ -101,6 +101,31 @@ def test_mamba_gpu_bf16_train_and_decode():
     assert err < 0.05, f"decode/prefill mismatch: {err}"
 
 
+def test_packed_seq_gpu():
+    """Packed (THD) forward on GPU equals the two separate forwards."""
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.transformer.packed_seq import PackedSeqParams
+
+    init_single()
+    dev = torch.device("cuda:0")
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=128, num_attention_heads=4, num_query_groups=2,
+        vocab_size=96, ffn_hidden_size=256, params_dtype=torch.bfloat16, bf16=True,
+        gradient_accumulation_fusion=False, max_position_embeddings=128)
+    torch.manual_seed(1)
+    model = GPTModel(cfg).to(dev).eval()
+    d1 = torch.randint(0, 96, (1, 40), device=dev)
+    d2 = torch.randint(0, 96, (1, 24), device=dev)
+    packed = torch.cat([d1, d2], dim=1)
+    p = PackedSeqParams.from_lengths([40, 24], device=dev)
+    with torch.no_grad():
+        out_p = model(packed, position_ids=None, attention_mask=None, packed_seq_params=p)
+        out_1 = model(d1, position_ids=None, attention_mask=None)
+        out_2 = model(d2, position_ids=None, attention_mask=None)
+    assert torch.allclose(out_p[:40].float(), out_1.float(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(out_p[40:].float(), out_2.float(), atol=5e-2, rtol=5e-2)
+
+
 def test_fsdp_gpu_single_rank():
     """FSDP mechanics on one GPU rank (world 1 NCCL/RCCL)."""
     import os
@@ -176,28 +201,3 @@ def test_moe_shared_expert_stream_overlap_gpu():
         seq = seq + layer.shared_expert(tokens)
         seq = seq.view(x.shape).to(x.dtype)
     assert torch.allclose(out.detach(), seq, atol=3e-2, rtol=3e-2)
-
-
-def test_packed_seq_gpu():
-    """Packed (THD) forward on GPU equals the two separate forwards."""
-    from megatron_amd.models.gpt import GPTModel
-    from megatron_amd.transformer.packed_seq import PackedSeqParams
-
-    init_single()
-    dev = torch.device("cuda:0")
-    cfg = TransformerConfig(
-        num_layers=2, hidden_size=128, num_attention_heads=4, num_query_groups=2,
-        vocab_size=96, ffn_hidden_size=256, params_dtype=torch.bfloat16, bf16=True,
-        gradient_accumulation_fusion=False, max_position_embeddings=128)
-    torch.manual_seed(1)
-    model = GPTModel(cfg).to(dev).eval()
-    d1 = torch.randint(0, 96, (1, 40), device=dev)
-    d2 = torch.randint(0, 96, (1, 24), device=dev)
-    packed = torch.cat([d1, d2], dim=1)
-    p = PackedSeqParams.from_lengths([40, 24], device=dev)
-    with torch.no_grad():
-        out_p = model(packed, position_ids=None, attention_mask=None, packed_seq_params=p)
-        out_1 = model(d1, position_ids=None, attention_mask=None)
-        out_2 = model(d2, position_ids=None, attention_mask=None)
-    assert torch.allclose(out_p[:40].float(), out_1.float(), atol=5e-2, rtol=5e-2)
-    assert torch.allclose(out_p[40:].float(), out_2.float(), atol=5e-2, rtol=5e-2)
