@@ -359,3 +359,58 @@ def test_dp2_dist_opt_overlap_param_gather_resume(tmp_path):
     from tests.utils import spawn_dist
 
     spawn_dist(_dp2_overlap_resume_case, 2, str(tmp_path / "ck"))
+
+
+def test_windowed_load_memory_bounded(tmp_path):
+    """Scalability contract: loading one rank's shard
+    from a multi-rank checkpoint must NOT materialize full global tensors.
+    A 1 GB-scale synthetic checkpoint (8 source ranks) is loaded in a fresh
+    subprocess requesting 1/8 of each tensor; peak RSS above the
+    post-import baseline must stay near one shard file, far under the
+    full-assembly cost."""
+    import subprocess
+    import sys
+
+    path = str(tmp_path / "big")
+    os.makedirs(path)
+    # 2 keys x [8192, 16384] fp32 = 512 MB each, 1 GB total, 8 row-slabs
+    import json as _json
+
+    meta = {}
+    rows, cols, nsrc = 8192, 16384, 8
+    for key in ("a", "b"):
+        meta[key] = {"global_shape": [rows, cols], "dtype": "float32", "pieces": []}
+    for r in range(nsrc):
+        payload = {}
+        for key in ("a", "b"):
+            payload[key] = [torch.full((rows // nsrc, cols), float(r))]
+            meta[key]["pieces"].append({
+                "global_offset": [r * rows // nsrc, 0],
+                "local_shape": [rows // nsrc, cols],
+                "flattened_range": None, "file": f"shards_r{r}.pt",
+            })
+        torch.save(payload, os.path.join(path, f"shards_r{r}.pt"))
+    _json.dump(meta, open(os.path.join(path, "metadata.json"), "w"))
+    torch.save({}, os.path.join(path, "common.pt"))
+
+    prog = f"""
+import resource, torch, sys
+sys.path.insert(0, {repr(os.getcwd())})
+base = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+from megatron_amd.checkpoint.sharded import ShardedTensor, load
+req = {{}}
+for key in ("a", "b"):
+    st = ShardedTensor(key, torch.zeros({rows // nsrc}, {cols}), ({rows}, {cols}), ({rows // nsrc}, 0))
+    req[key] = st
+load(req, {repr(path)})
+assert float(req["a"].data.mean()) == 1.0  # slab 1 of key a
+peak = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+print("DELTA_MB", (peak - base) / 1024)
+"""
+    out = subprocess.run([sys.executable, "-c", prog], capture_output=True, text=True,
+                         timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    delta_mb = float(out.stdout.split("DELTA_MB")[1].strip())
+    # requester shards: 2 x 64MB dst (pre-allocated before baseline? no: after)
+    # windowed budget: ~1 shard file (128MB) + 2 x 64MB dst + slack << 1GB
+    assert delta_mb < 450, f"load peaked {delta_mb} MB above baseline (full assembly?)"
