@@ -104,6 +104,118 @@ class _CPGatherSeq(torch.autograd.Function):
         return slice_for_cp_rank(g, r, ctx.cp, seq_dim=ctx.seq_dim, mode=ctx.mode), None, None
 
 
+class _CPSeqToChannel(torch.autograd.Function):
+    """CP all-to-all turning a sequence-sharded full-channel tensor into a
+    full-sequence channel-sharded one (Ulysses-style, for the Mamba scan:
+    each rank scans the WHOLE sequence for 1/cp of the channels instead of
+    redundantly scanning everything).
+
+    forward:  [s/cp, b, C] -> [s, b, C/cp] in NATURAL sequence order
+              (inverts the zigzag load-balanced layout for mode "p2p")
+    backward: exact inverse all-to-all.
+    """
+
+    @staticmethod
+    def forward(ctx, x, mode):
+        cp = G.get_context_parallel_world_size()
+        ctx.mode, ctx.cp = mode, cp
+        if cp == 1:
+            return x
+        group = G.get_grid().group("cp")
+        s_loc, b, C = x.shape
+        ctx.shape_in = (s_loc, b, C)
+        send = x.view(s_loc, b, cp, C // cp).permute(2, 0, 1, 3).contiguous()
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send, group=group)
+        # recv[r] = rank r's seq shard of my channel chunk (rank order)
+        if mode == "a2a":
+            return recv.reshape(cp * s_loc, b, C // cp)
+        L = s_loc // 2
+        slots = [None] * (2 * cp)
+        for r in range(cp):
+            c0, c1 = cp_chunk_ids(r, cp)
+            slots[c0] = recv[r].narrow(0, 0, L)
+            slots[c1] = recv[r].narrow(0, L, L)
+        return torch.cat(slots, dim=0)
+
+    @staticmethod
+    def backward(ctx, g):
+        if ctx.cp == 1:
+            return g, None
+        group = G.get_grid().group("cp")
+        s_loc, b, C = ctx.shape_in
+        cp = ctx.cp
+        g = g.contiguous()
+        if ctx.mode == "a2a":
+            send = g.view(cp, s_loc, b, C // cp)
+        else:
+            L = s_loc // 2
+            parts = []
+            for r in range(cp):
+                c0, c1 = cp_chunk_ids(r, cp)
+                parts.append(torch.cat([g.narrow(0, c0 * L, L), g.narrow(0, c1 * L, L)], dim=0))
+            send = torch.stack(parts, dim=0)
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send.contiguous(), group=group)
+        out = recv.permute(1, 2, 0, 3).reshape(s_loc, b, C)
+        return out, None
+
+
+class _CPChannelToSeq(torch.autograd.Function):
+    """Inverse of _CPSeqToChannel: [s, b, C/cp] -> [s/cp, b, C]."""
+
+    @staticmethod
+    def forward(ctx, y, mode):
+        cp = G.get_context_parallel_world_size()
+        ctx.mode, ctx.cp = mode, cp
+        if cp == 1:
+            return y
+        group = G.get_grid().group("cp")
+        s_full, b, Cc = y.shape
+        s_loc = s_full // cp
+        ctx.shape_out = (s_loc, b, Cc)
+        if mode == "a2a":
+            send = y.contiguous().view(cp, s_loc, b, Cc)
+        else:
+            L = s_loc // 2
+            parts = []
+            for r in range(cp):
+                c0, c1 = cp_chunk_ids(r, cp)
+                parts.append(torch.cat([y.narrow(0, c0 * L, L), y.narrow(0, c1 * L, L)], dim=0))
+            send = torch.stack(parts, dim=0)
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send.contiguous(), group=group)
+        return recv.permute(1, 2, 0, 3).reshape(s_loc, b, cp * Cc)
+
+    @staticmethod
+    def backward(ctx, g):
+        if ctx.cp == 1:
+            return g, None
+        group = G.get_grid().group("cp")
+        s_loc, b, Cc = ctx.shape_out
+        cp = ctx.cp
+        send = g.contiguous().view(s_loc, b, cp, Cc).permute(2, 0, 1, 3).contiguous()
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send, group=group)
+        if ctx.mode == "a2a":
+            return recv.reshape(cp * s_loc, b, Cc), None
+        L = s_loc // 2
+        slots = [None] * (2 * cp)
+        for r in range(cp):
+            c0, c1 = cp_chunk_ids(r, cp)
+            slots[c0] = recv[r].narrow(0, 0, L)
+            slots[c1] = recv[r].narrow(0, L, L)
+        return torch.cat(slots, dim=0), None
+
+
+def cp_seq_to_channel(x: torch.Tensor, mode: str = "p2p") -> torch.Tensor:
+    return _CPSeqToChannel.apply(x, mode)
+
+
+def cp_channel_to_seq(y: torch.Tensor, mode: str = "p2p") -> torch.Tensor:
+    return _CPChannelToSeq.apply(y, mode)
+
+
 def gather_cp_sequence(x: torch.Tensor, seq_dim: int = 0, mode: str = "p2p") -> torch.Tensor:
     """Differentiable shard -> full-sequence gather (natural order)."""
     return _CPGatherSeq.apply(x, seq_dim, mode)

@@ -149,7 +149,13 @@ class MambaMixer(nn.Module):
         # baseline — head-split CP is the round-2 optimization)
         cp = G.get_context_parallel_world_size()
         cp_mode = getattr(self.config, "cp_comm_type", "p2p")
-        if cp > 1 and inference_state is None:
+        # head-split CP (reference MambaContextParallel intent, Ulysses-style):
+        # a2a the (x, B, C, dt) streams so each rank scans the FULL sequence
+        # for 1/cp of the heads/groups — no redundant compute; z stays
+        # sequence-sharded (it only gates the output after the inverse a2a).
+        headsplit = (cp > 1 and inference_state is None
+                     and self.nheads_local % cp == 0 and self.ngroups_local % cp == 0)
+        if cp > 1 and inference_state is None and not headsplit:
             from megatron_amd.parallel.context_parallel import gather_cp_sequence
 
             hidden_states = gather_cp_sequence(hidden_states, seq_dim=0, mode=cp_mode)
@@ -159,6 +165,9 @@ class MambaMixer(nn.Module):
 
         if inference_state is not None and s == 1:
             return self._step(z, x, B, C, dt, inference_state)
+
+        if headsplit:
+            return self._forward_headsplit_cp(z, x, B, C, dt, cp, cp_mode)
 
         xBC = torch.cat([x, B, C], dim=-1).permute(1, 0, 2)  # [b, l, conv_dim]
         if inference_state is not None:
@@ -194,6 +203,58 @@ class MambaMixer(nn.Module):
 
             out = slice_for_cp_rank(out, G.get_context_parallel_rank(), cp,
                                     seq_dim=0, mode=cp_mode)
+        return out
+
+    def _headsplit_indices(self, cp: int, r: int, device):
+        """Channel indices of CP chunk r inside the concatenated [x|B|C]
+        conv layout (each component is chunked contiguously)."""
+        dlc = self.d_inner_local // cp
+        gsc = self.ngroups_local * self.d_state // cp
+        xs = torch.arange(r * dlc, (r + 1) * dlc, device=device)
+        bs = self.d_inner_local + torch.arange(r * gsc, (r + 1) * gsc, device=device)
+        cs = self.d_inner_local + self.ngroups_local * self.d_state + torch.arange(
+            r * gsc, (r + 1) * gsc, device=device)
+        return torch.cat([xs, bs, cs])
+
+    def _forward_headsplit_cp(self, z, x, B, C, dt, cp: int, cp_mode: str):
+        from megatron_amd.parallel.context_parallel import (
+            cp_channel_to_seq,
+            cp_seq_to_channel,
+        )
+
+        r = G.get_context_parallel_rank()
+        device = x.device
+        # channel permutation so equal a2a chunks carry (x_r | B_r | C_r)
+        perm = torch.cat([self._headsplit_indices(cp, rr, device) for rr in range(cp)])
+        xBC = torch.cat([x, B, C], dim=-1).index_select(-1, perm)
+        xBC = cp_seq_to_channel(xBC, mode=cp_mode)          # [s_full, b, conv_dim/cp]
+        dtf = cp_seq_to_channel(dt, mode=cp_mode)           # [s_full, b, nheads/cp]
+        s, b = xBC.shape[0], xBC.shape[1]
+
+        # sliced per-channel weights for this rank's chunk
+        my_idx = self._headsplit_indices(cp, r, device)
+        conv_w = self.conv_weight.index_select(0, my_idx)
+        conv_b = self.conv_bias.index_select(0, my_idx)
+        from megatron_amd import ops as _ops
+
+        xBCb = xBC.permute(1, 0, 2)                         # [b, l, conv_dim/cp]
+        xBCb = _ops.causal_conv1d_silu(xBCb, conv_w, conv_b)
+
+        nh, ng = self.nheads_local // cp, self.ngroups_local // cp
+        xs, Bs, Cs = torch.split(
+            xBCb, [nh * self.headdim, ng * self.d_state, ng * self.d_state], dim=-1)
+        xs = xs.view(b, s, nh, self.headdim)
+        Bs = Bs.view(b, s, ng, self.d_state)
+        Cs = Cs.view(b, s, ng, self.d_state)
+        hsel = torch.arange(r * nh, (r + 1) * nh, device=device)
+        dtv = F.softplus(dtf.permute(1, 0, 2).float() + self.dt_bias.index_select(0, hsel))
+        A = -torch.exp(self.A_log.index_select(0, hsel))
+        y = ssd_chunked_scan(xs, dtv.to(xs.dtype), A, Bs, Cs,
+                             D=self.D.index_select(0, hsel), chunk_size=self.chunk_size)
+        y = y.reshape(b, s, nh * self.headdim).permute(1, 0, 2)  # [s_full, b, d_inner/(tp*cp)]
+        y = cp_channel_to_seq(y, mode=cp_mode)              # [s_loc, b, d_inner/tp]
+        y = self.norm(y, z)
+        out, _ = self.out_proj(y)
         return out
 
     def _step(self, z, x, B, C, dt, state: dict):

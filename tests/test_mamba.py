@@ -253,3 +253,47 @@ def test_mamba_cp2_matches_single(tmp_path, monkeypatch):
     spawn_dist(_mamba_cp2_case, 2)
     got = json.load(open(out))
     assert abs(got - ref) < 2e-3 * max(abs(ref), 1.0), (got, ref)
+
+
+def _mamba_cp2_headsplit_case(rank, world):
+    """Asserts the head-split path (no redundant gathered scan) is the one
+    running under CP=2, with grads flowing."""
+    import torch.distributed as dist
+
+    from megatron_amd.parallel import grid as G
+    from megatron_amd.parallel.context_parallel import get_batch_on_this_cp_rank
+    from megatron_amd.parallel.random import model_parallel_seed
+    from megatron_amd.ssm.mamba_mixer import MambaMixer
+
+    G.initialize_model_parallel(context_parallel_size=2)
+    model_parallel_seed(1234)
+    cfg = _mixer_config(context_parallel_size=2)
+    torch.manual_seed(42)
+    m = MambaModel(cfg)
+    calls = []
+    orig = MambaMixer._forward_headsplit_cp
+
+    def spy(self, *a, **k):
+        calls.append(1)
+        return orig(self, *a, **k)
+
+    MambaMixer._forward_headsplit_cp = spy
+    try:
+        torch.manual_seed(9)
+        tokens = torch.randint(0, cfg.vocab_size, (2, 16))
+        labels = torch.randint(0, cfg.vocab_size, (2, 16))
+        dist.broadcast(tokens, src=0)
+        dist.broadcast(labels, src=0)
+        local = get_batch_on_this_cp_rank({"tokens": tokens, "labels": labels})
+        loss = m(local["tokens"], labels=local["labels"]).sum()
+        loss.backward()
+    finally:
+        MambaMixer._forward_headsplit_cp = orig
+    assert len(calls) >= 1, "head-split CP path did not run"
+    for n, p in m.named_parameters():
+        if "conv_weight" in n or "A_log" in n:
+            assert p.grad is not None and torch.isfinite(p.grad).all(), n
+
+
+def test_mamba_cp2_headsplit_active_and_trains():
+    spawn_dist(_mamba_cp2_headsplit_case, 2)
