@@ -61,6 +61,17 @@ class RerunDataIterator:
         self._history.append(batch)
         return batch
 
+    def state_dict(self) -> dict:
+        """Checkpointable replay state (reference RerunDataIterator): lets a
+        restarted job replay the in-flight iteration's batches."""
+        return {"history": self._history, "replaying": self._replaying,
+                "replay_pos": self._replay_pos}
+
+    def load_state_dict(self, sd: dict):
+        self._history = list(sd.get("history", []))
+        self._replaying = bool(sd.get("replaying", False))
+        self._replay_pos = int(sd.get("replay_pos", 0))
+
     def start_iteration(self):
         self._history.clear()
         self._replaying = False
@@ -75,6 +86,7 @@ class RerunStateMachine:
 
     def __init__(self, mode: RerunMode = RerunMode.DISABLED,
                  error_injection_rate: float = 0.0):
+        self._local_request = False
         self.mode = RerunMode(mode)
         self.state = RerunState.NOT_RUNNING_YET
         self._first_values: List[float] = []
@@ -97,12 +109,22 @@ class RerunStateMachine:
             self._first_values.clear()
             self._failed_msg = None
             self._request_rerun = False
+            self._local_request = False
             self._value_index = 0
             self._step += 1
             for it in its:
                 if isinstance(it, RerunDataIterator):
                     it.start_iteration()
             return True
+        if self.state == RerunState.FIRST_RUN:
+            # COLLECTIVE decision: if any rank flagged a result, every rank
+            # must replay, or the replayed iteration's collectives desync
+            # and the job hangs (reference rerun_state_machine rank sync)
+            self._local_request = self._request_rerun
+            if torch.distributed.is_initialized() and torch.distributed.get_world_size() > 1:
+                flag = torch.tensor([1 if self._request_rerun else 0])
+                torch.distributed.all_reduce(flag, op=torch.distributed.ReduceOp.MAX)
+                self._request_rerun = bool(int(flag))
         if self.state == RerunState.FIRST_RUN and self._request_rerun:
             self.state = RerunState.RERUNNING
             self._rerun_values.clear()
@@ -116,24 +138,49 @@ class RerunStateMachine:
 
     def should_checkpoint_and_exit(self) -> Optional[int]:
         """After the while-loop: non-None exit code if the machine concluded
-        this iteration hit a fault."""
+        this iteration hit a fault.  The verdict comes from the rank(s) that
+        DETECTED the bad value (non-flagging ranks trivially reproduce their
+        own healthy values) and is agreed collectively so every rank exits
+        with the same code."""
         if self.state == RerunState.RERUNNING:
-            # compare replay values against the first run
-            same = len(self._first_values) == len(self._rerun_values) and all(
-                (a == b) or (a != a and b != b)  # NaN == NaN for this purpose
-                for a, b in zip(self._first_values, self._rerun_values))
+            local_transient = 0
+            local_persistent = 0
+            if self._local_request:
+                same = len(self._first_values) == len(self._rerun_values) and all(
+                    (a == b) or (a != a and b != b)  # NaN == NaN for this purpose
+                    for a, b in zip(self._first_values, self._rerun_values))
+                if same:
+                    local_persistent = 1
+                    logger.error("rerun reproduced the invalid result: deterministic error (%s)",
+                                 self._failed_msg)
+                else:
+                    local_transient = 1
+                    logger.error("rerun produced a different result: transient fault (%s)",
+                                 self._failed_msg)
+            if torch.distributed.is_initialized() and torch.distributed.get_world_size() > 1:
+                flags = torch.tensor([local_transient, local_persistent])
+                torch.distributed.all_reduce(flags, op=torch.distributed.ReduceOp.MAX)
+                local_transient, local_persistent = int(flags[0]), int(flags[1])
             self.state = RerunState.NOT_RUNNING_YET
-            if same:
+            if local_persistent:
                 self.stats["persistent"] += 1
-                logger.error("rerun reproduced the invalid result: deterministic error (%s)",
-                             self._failed_msg)
                 return EXIT_CODE_FAILED_ON_RESULT_VALIDATION
-            self.stats["transient"] += 1
-            logger.error("rerun produced a different result: transient fault (%s)",
-                         self._failed_msg)
-            return EXIT_CODE_SUCCESS_ON_RESULT_VALIDATION
+            if local_transient:
+                self.stats["transient"] += 1
+                return EXIT_CODE_SUCCESS_ON_RESULT_VALIDATION
+            return None
         self.state = RerunState.NOT_RUNNING_YET
         return None
+
+    # -- checkpointing (reference RerunStateMachine state_dict) -------------
+
+    def state_dict(self) -> dict:
+        return {"mode": self.mode.value, "stats": dict(self.stats), "step": self._step}
+
+    def load_state_dict(self, sd: dict):
+        self.mode = RerunMode(sd.get("mode", self.mode.value))
+        self.stats.update(sd.get("stats", {}))
+        self._step = sd.get("step", self._step)
 
     # -- called from loss functions ---------------------------------------
 

@@ -89,3 +89,68 @@ def test_non_restartable_exception_propagates():
 
     with pytest.raises(ValueError):
         run_with_inprocess_restart(train_fn, RestartConfig())
+
+
+def _rerun_dp2_case(rank, world):
+    """Only rank 1 flags a bad value: BOTH ranks must replay (collective
+    decision) and both must receive the same transient exit code."""
+    import torch.distributed as dist
+
+    from megatron_amd.utils.rerun_state_machine import (
+        RerunDataIterator,
+        RerunMode,
+        RerunStateMachine,
+    )
+
+    sm = RerunStateMachine(RerunMode.VALIDATE_RESULTS)
+    it = RerunDataIterator(iter([{"x": torch.tensor([float(i + rank)])} for i in range(4)]))
+    runs = []
+    call_count = {"n": 0}
+
+    def flaky_value():
+        # rank 1's first evaluation is bad; its replay is clean -> transient
+        call_count["n"] += 1
+        if rank == 1 and call_count["n"] == 1:
+            return torch.tensor(float("nan"))
+        return torch.tensor(1.0)
+
+    while sm.should_run_forward_backward(it):
+        batch = next(it)
+        runs.append(float(batch["x"]))
+        v = flaky_value()
+        sm.validate_result(v, lambda t: bool(torch.isnan(t).any()), "nan loss")
+    code = sm.should_checkpoint_and_exit()
+    assert len(runs) == 2, runs            # every rank replayed
+    assert runs[0] == runs[1]              # identical data on replay
+    assert code == 17, (rank, code)        # transient verdict agreed by all
+
+
+def test_rerun_collective_decision_dp2():
+    from tests.utils import spawn_dist
+
+    spawn_dist(_rerun_dp2_case, 2)
+
+
+def test_rerun_state_dict_roundtrip():
+    from megatron_amd.utils.rerun_state_machine import (
+        RerunDataIterator,
+        RerunMode,
+        RerunStateMachine,
+    )
+
+    sm = RerunStateMachine(RerunMode.VALIDATE_RESULTS)
+    sm.stats["transient"] = 3
+    sm._step = 7
+    sm2 = RerunStateMachine(RerunMode.DISABLED)
+    sm2.load_state_dict(sm.state_dict())
+    assert sm2.mode == RerunMode.VALIDATE_RESULTS
+    assert sm2.stats["transient"] == 3 and sm2._step == 7
+
+    it = RerunDataIterator(iter([1, 2, 3]))
+    it.start_iteration()
+    assert next(it) == 1
+    sd = it.state_dict()
+    it2 = RerunDataIterator(iter([4, 5]))
+    it2.load_state_dict(sd)
+    it2.rewind()
+    assert next(it2) == 1  # replays the checkpointed in-flight batch
