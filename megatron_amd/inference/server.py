@@ -58,9 +58,10 @@ def create_app(engine, tokenizer=None):
         """Chat completion: messages are rendered with the tokenizer's chat
         template and generated as one continuation."""
         req = await request.json()
-        assert tokenizer is not None and hasattr(tokenizer, "apply_chat_template"), \
-            "chat endpoint needs a chat-template-capable tokenizer"
-        prompt_ids = tokenizer.apply_chat_template(req["messages"], add_generation_prompt=True)
+        assert tokenizer is not None, "chat endpoint needs a tokenizer"
+        from megatron_amd.tokenizers import apply_chat_template
+
+        prompt_ids = apply_chat_template(tokenizer, req["messages"], add_generation_prompt=True)
         params = SamplingParams(
             max_tokens=int(req.get("max_tokens", 256)),
             temperature=float(req.get("temperature", 1.0)),
@@ -89,6 +90,20 @@ def create_app(engine, tokenizer=None):
             top_k=int(req.get("top_k", 0)), top_p=float(req.get("top_p", 0.0)),
             greedy=bool(req.get("greedy", False)), seed=req.get("seed"))
 
+        detok = None
+        if tokenizer is not None:
+            from megatron_amd.tokenizers import IncrementalDetokenizer
+
+            detok = IncrementalDetokenizer(tokenizer)
+
+        def emit(t):
+            ev = {"token": t}
+            if detok is not None:
+                delta = detok.put(t)
+                if delta:
+                    ev["text"] = delta  # stable utf-8 prefix delta
+            return f"data: {_json.dumps(ev)}\n\n"
+
         def gen():
             with lock:
                 rid = engine.add_request(prompt, params)
@@ -98,13 +113,14 @@ def create_app(engine, tokenizer=None):
                     if rid in engine.finished:
                         r = engine.finished.pop(rid)
                         for t in r.output_tokens[sent:]:
-                            yield f"data: {_json.dumps({'token': t})}\n\n"
-                        yield f"data: {_json.dumps({'done': True, 'text': r.text})}\n\n"
+                            yield emit(t)
+                        tail = detok.flush() if detok is not None else None
+                        yield f"data: {_json.dumps({'done': True, 'text': r.text, 'tail': tail})}\n\n"
                         return
                     cur = next((q for q in engine.active if q.rid == rid), None)
                     if cur is not None and len(cur.result.output_tokens) > sent:
                         for t in cur.result.output_tokens[sent:]:
-                            yield f"data: {_json.dumps({'token': t})}\n\n"
+                            yield emit(t)
                         sent = len(cur.result.output_tokens)
 
         return StreamingResponse(gen(), media_type="text/event-stream")

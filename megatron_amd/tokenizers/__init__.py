@@ -167,3 +167,73 @@ def pad_vocab_size(vocab_size: int, multiple: int, tp_size: int) -> int:
     (reference training/tokenizer _vocab_size_with_padding)."""
     m = multiple * tp_size
     return ((vocab_size + m - 1) // m) * m
+
+
+# ---------------------------------------------------------------------------
+# serving / training utilities (reference tokenizers/ chat templates,
+# vocab padding, and the streaming detokenizer the SSE server needs)
+# ---------------------------------------------------------------------------
+
+
+def pad_vocab_size(vocab_size: int, tensor_parallel_size: int = 1,
+                   make_divisible_by: int = 128) -> int:
+    """Pad the vocabulary so each TP shard is a multiple of
+    ``make_divisible_by`` (reference _vocab_size_with_padding): keeps the
+    output-layer GEMM and vocab-parallel CE on aligned shard sizes."""
+    mult = make_divisible_by * tensor_parallel_size
+    return ((vocab_size + mult - 1) // mult) * mult
+
+
+class IncrementalDetokenizer:
+    """Streaming detokenization for SSE serving: UTF-8 multi-byte sequences
+    (and tokenizers whose pieces merge across boundaries) cannot be decoded
+    token-by-token; this holds back the undecodable tail and emits only the
+    stable prefix delta per step (reference inference text-gen controller
+    detokenize-incrementally behavior)."""
+
+    REPLACEMENT = "�"
+
+    def __init__(self, tokenizer: MegatronTokenizer):
+        self._tok = tokenizer
+        self._ids: List[int] = []
+        self._emitted = 0  # chars already streamed out
+
+    def put(self, token_id: int) -> str:
+        """Add one token; returns the newly stable text delta ('' if the
+        tail is still an incomplete sequence)."""
+        self._ids.append(int(token_id))
+        text = self._tok.detokenize(self._ids)
+        # hold back a trailing replacement char (incomplete utf-8 etc.)
+        stable_end = len(text)
+        while stable_end > 0 and text[stable_end - 1] == self.REPLACEMENT:
+            stable_end -= 1
+        delta = text[self._emitted:stable_end]
+        self._emitted = stable_end
+        return delta
+
+    def flush(self) -> str:
+        """Final flush: emit whatever remains (incl. replacement chars)."""
+        text = self._tok.detokenize(self._ids)
+        delta = text[self._emitted:]
+        self._emitted = len(text)
+        return delta
+
+
+def apply_chat_template(tokenizer: MegatronTokenizer, messages: List[dict],
+                        add_generation_prompt: bool = True) -> List[int]:
+    """Render a chat conversation to token ids.
+
+    HF tokenizers with a built-in chat template use it; every other
+    tokenizer gets a simple generic template (role-tagged lines), so the
+    /v1/chat endpoint works with any backend (reference tokenizers/text
+    model-specific parsers)."""
+    hf = getattr(tokenizer, "_tok", None)
+    if hf is not None and getattr(hf, "chat_template", None):
+        return list(hf.apply_chat_template(
+            messages, add_generation_prompt=add_generation_prompt, tokenize=True))
+    parts = []
+    for m in messages:
+        parts.append(f"<|{m.get('role', 'user')}|>\n{m.get('content', '')}\n")
+    if add_generation_prompt:
+        parts.append("<|assistant|>\n")
+    return tokenizer.tokenize("".join(parts))

@@ -233,3 +233,36 @@ def test_t5_pad_batch_shapes():
     assert batch["decoder_tokens"].shape == (3, 32)
     assert batch["loss_mask"].sum() > 0
     assert batch["encoder_mask"][0].sum() == ds[0]["encoder_tokens"].numel()
+
+
+def test_incremental_detokenizer_utf8_boundaries():
+    from megatron_amd.tokenizers import ByteLevelTokenizer, IncrementalDetokenizer
+
+    tok = ByteLevelTokenizer()
+    text = "héllo 🌍!"
+    ids = tok.tokenize(text)
+    detok = IncrementalDetokenizer(tok)
+    out = ""
+    for i in ids:
+        delta = detok.put(i)
+        assert "�" not in delta  # never emits an incomplete sequence
+        out += delta
+    out += detok.flush()
+    assert out == text
+
+
+def test_pad_vocab_size():
+    from megatron_amd.tokenizers import pad_vocab_size
+
+    assert pad_vocab_size(128000, 8) == 128000  # already 1024-aligned
+    assert pad_vocab_size(32000, 8) == 32768
+    assert pad_vocab_size(100, 1) == 128
+
+
+def test_apply_chat_template_generic():
+    from megatron_amd.tokenizers import ByteLevelTokenizer, apply_chat_template
+
+    tok = ByteLevelTokenizer()
+    ids = apply_chat_template(tok, [{"role": "user", "content": "hi"}])
+    text = tok.detokenize(ids)
+    assert "<|user|>" in text and "<|assistant|>" in text and "hi" in text
