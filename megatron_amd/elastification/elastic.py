@@ -91,3 +91,158 @@ def elastic_memory_profile(model: nn.Module, fracs: List[float]) -> Dict[float, 
     for m, oa, ia in saved:
         m.out_active, m.in_active = oa, ia
     return out
+
+
+# ---------------------------------------------------------------------------
+# Flextron-style elastic transformer pieces: elastic MLP widths inside a real
+# GPT, an input-adaptive width router, sandwich-rule training, and export
+# (reference megatron/elastification/ core loop).
+# ---------------------------------------------------------------------------
+
+
+class ElasticMLP(nn.Module):
+    """Wraps a dense MLP so its ffn width is a runtime-selectable prefix.
+
+    The full weights stay shared; fraction f uses gate rows [0 : f*ffn],
+    up rows [ffn : ffn + f*ffn] of fc1 and columns [0 : f*ffn] of fc2 —
+    Flextron's nested sub-network semantics.  TP=1 scope (elastic widths
+    and TP sharding compose in the reference via per-shard fractions; out
+    of scope here)."""
+
+    def __init__(self, mlp, widths=(0.25, 0.5, 1.0)):
+        super().__init__()
+        assert getattr(mlp, "gated", False), "elastic MLP expects a gated (swiglu/geglu) MLP"
+        self.mlp = mlp
+        self.widths = tuple(sorted(widths))
+        self.frac = 1.0
+        self.ffn = mlp.linear_fc2.weight.shape[1]
+
+    def set_width(self, frac: float):
+        self.frac = float(frac)
+
+    def forward(self, hidden_states: torch.Tensor) -> torch.Tensor:
+        import torch.nn.functional as F
+
+        k = max(1, int(round(self.ffn * self.frac)))
+        w1 = self.mlp.linear_fc1.weight
+        w2 = self.mlp.linear_fc2.weight
+        w1s = torch.cat([w1[:k], w1[self.ffn : self.ffn + k]], dim=0)
+        x = F.linear(hidden_states, w1s)
+        x = self.mlp._act(x)
+        return F.linear(x, w2[:, :k])
+
+
+class WidthRouter(nn.Module):
+    """Input-adaptive width selection (Flextron router): a tiny classifier
+    over the width set from the mean-pooled hidden state; straight-through
+    hard choice in training, argmax at eval.  Returns (frac, aux) where aux
+    is a differentiable surrogate that lets the router learn."""
+
+    def __init__(self, hidden_size: int, widths=(0.25, 0.5, 1.0), latency_penalty: float = 0.0):
+        super().__init__()
+        self.widths = tuple(sorted(widths))
+        self.proj = nn.Linear(hidden_size, len(self.widths))
+        self.latency_penalty = latency_penalty
+
+    def forward(self, hidden_states: torch.Tensor):
+        import torch.nn.functional as F
+
+        pooled = hidden_states.float().mean(dim=tuple(range(hidden_states.dim() - 1)))
+        logits = self.proj(pooled)
+        probs = F.softmax(logits, dim=-1)
+        idx = int(probs.argmax())
+        # straight-through scale: multiplying the MLP output by
+        # (1 - p_sel.detach() + p_sel) routes gradient into the router
+        p_sel = probs[idx]
+        st_scale = 1.0 - p_sel.detach() + p_sel
+        # latency-aware regularizer: expected width fraction
+        aux = self.latency_penalty * (probs * torch.tensor(self.widths, dtype=probs.dtype)).sum()
+        return self.widths[idx], st_scale, aux
+
+
+def elastify_gpt(model, widths=(0.25, 0.5, 1.0), with_router: bool = False,
+                 latency_penalty: float = 0.0) -> int:
+    """Replace every decoder layer's MLP with an ElasticMLP (optionally
+    routed).  Returns the number of layers elastified."""
+    core = model.module if hasattr(model, "module") else model
+    n = 0
+    for layer in core.decoder.layers:
+        if hasattr(layer, "mlp") and getattr(layer.mlp, "gated", False):
+            em = ElasticMLP(layer.mlp, widths)
+            if with_router:
+                em.router = WidthRouter(core.config.hidden_size, widths, latency_penalty)
+
+                orig_forward = em.forward
+
+                def routed_forward(h, _em=em, _orig=orig_forward):
+                    frac, st, aux = _em.router(h)
+                    _em.set_width(frac)
+                    out = _orig(h) * st.to(h.dtype)
+                    _em.router_aux = aux
+                    return out
+
+                em.forward = routed_forward
+            layer.mlp = em
+            n += 1
+    return n
+
+
+def set_gpt_width(model, frac: float) -> int:
+    core = model.module if hasattr(model, "module") else model
+    n = 0
+    for m in core.modules():
+        if isinstance(m, ElasticMLP):
+            m.set_width(frac)
+            n += 1
+    return n
+
+
+def sandwich_step(model, loss_fn, widths=(0.25, 0.5, 1.0), num_random: int = 1, rng=None):
+    """Flextron/slimmable sandwich rule: accumulate grads at the largest
+    width, the smallest, and `num_random` middle widths.  `loss_fn(model)`
+    runs one forward and returns a scalar loss.  Returns {frac: loss}."""
+    import random as _random
+
+    rng = rng or _random.Random(0)
+    fracs = [max(widths), min(widths)]
+    middle = [w for w in widths if min(widths) < w < max(widths)]
+    for _ in range(num_random):
+        if middle:
+            fracs.append(rng.choice(middle))
+    out = {}
+    for f in fracs:
+        set_gpt_width(model, f)
+        loss = loss_fn(model)
+        loss.backward()
+        out[f] = float(loss.detach())
+    set_gpt_width(model, max(widths))
+    return out
+
+
+def materialize_gpt(model, frac: float):
+    """Export the active sub-network as a standalone dense GPTModel with
+    ffn_hidden_size = frac * ffn (weights sliced, not shared)."""
+    import dataclasses
+
+    from megatron_amd.models.gpt import GPTModel
+
+    core = model.module if hasattr(model, "module") else model
+    cfg = dataclasses.replace(core.config)
+    k = max(1, int(round(core.config.ffn_hidden_size * frac)))
+    cfg.ffn_hidden_size = k
+    new = GPTModel(cfg, pre_process=core.pre_process, post_process=core.post_process)
+    # elastified layers nest the dense MLP one level deeper (mlp.mlp.*)
+    src = {n.replace(".mlp.mlp.", ".mlp."): p for n, p in core.named_parameters()}
+    with torch.no_grad():
+        for name, p in new.named_parameters():
+            if name not in src:
+                continue
+            s = src[name]
+            if "linear_fc1.weight" in name:
+                ffn = core.config.ffn_hidden_size
+                p.copy_(torch.cat([s[:k], s[ffn : ffn + k]], dim=0))
+            elif "linear_fc2.weight" in name:
+                p.copy_(s[:, :k])
+            elif p.shape == s.shape:
+                p.copy_(s)
+    return new
