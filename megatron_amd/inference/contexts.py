@@ -132,6 +132,96 @@ class KVBlockAllocator:
         self._free.extend(blocks)
 
 
+class PrefixCachingAllocator(KVBlockAllocator):
+    """Refcounted block allocator with content-hash prefix reuse (reference
+    analog: vLLM-style automatic prefix caching adapted to the paged
+    context; the reference's engines gained this via its kv reuse work).
+
+    Only FULL prompt blocks are registered (keyed by the running hash of the
+    token chain), so shared blocks are immutable and no copy-on-write is
+    needed: decode always appends into an unshared tail block.  Freed
+    cached blocks keep their KV and move to an LRU; allocation evicts LRU
+    entries only when the free list is empty."""
+
+    def __init__(self, num_blocks: int, first_id: int = 0):
+        super().__init__(num_blocks, first_id)
+        self._hash_to_block: dict = {}
+        self._block_to_hash: dict = {}
+        self._ref: dict = {}
+        self._lru: dict = {}   # block -> tick (ref==0 cached blocks)
+        self._tick = 0
+        self.hits = 0
+        self.misses = 0
+
+    @property
+    def num_free(self) -> int:
+        return len(self._free) + len(self._lru)
+
+    def allocate(self, n: int) -> List[int]:
+        out = []
+        for _ in range(n):
+            if self._free:
+                out.append(self._free.pop())
+            elif self._lru:
+                blk = min(self._lru, key=self._lru.get)  # evict oldest
+                del self._lru[blk]
+                h = self._block_to_hash.pop(blk, None)
+                if h is not None:
+                    self._hash_to_block.pop(h, None)
+                self._ref.pop(blk, None)
+                out.append(blk)
+            else:
+                self._free.extend(out)
+                raise RuntimeError(f"out of KV blocks: need {n}")
+        for b in out:
+            self._ref[b] = 1
+        return out
+
+    def free(self, blocks: List[int]):
+        for b in blocks:
+            r = self._ref.get(b, 1) - 1
+            if r > 0:
+                self._ref[b] = r
+                continue
+            self._ref.pop(b, None)
+            if b in self._block_to_hash:
+                self._tick += 1
+                self._lru[b] = self._tick   # keep KV for reuse
+            else:
+                self._free.append(b)
+
+    # -- prefix reuse -----------------------------------------------------
+
+    def lookup(self, chain_hash) -> Optional[int]:
+        blk = self._hash_to_block.get(chain_hash)
+        if blk is None:
+            self.misses += 1
+            return None
+        self.hits += 1
+        if blk in self._lru:        # revive from the evictable pool
+            del self._lru[blk]
+            self._ref[blk] = 1
+        else:
+            self._ref[blk] = self._ref.get(blk, 0) + 1
+        return blk
+
+    def register(self, chain_hash, block: int):
+        if chain_hash in self._hash_to_block:
+            return
+        self._hash_to_block[chain_hash] = block
+        self._block_to_hash[block] = chain_hash
+
+
+def prompt_block_hashes(prompt: List[int], block_size: int) -> List[int]:
+    """Running content hash per FULL block of the prompt."""
+    out = []
+    h = 0
+    for i in range(len(prompt) // block_size):
+        h = hash((h, tuple(prompt[i * block_size : (i + 1) * block_size])))
+        out.append(h)
+    return out
+
+
 class DynamicInferenceContext:
     """Paged KV cache for continuous batching (reference dynamic_context.py:297).
 

@@ -228,7 +228,7 @@ class DynamicInferenceEngine:
 
     def __init__(self, model, tokenizer=None, num_blocks: int = 512, block_size: int = 256,
                  max_batch: int = 64, max_prefill_tokens: int = 8192, device=None,
-                 use_hip_graphs: bool = True):
+                 use_hip_graphs: bool = True, enable_prefix_caching: bool = True):
         self.model = model.eval()
         self.tokenizer = tokenizer
         cfg = model.config
@@ -257,6 +257,12 @@ class DynamicInferenceEngine:
 
         self.offloader = KVHostOffloader(self.context)
         self.preempted: List[tuple] = []  # (req, handle)
+        self.prefix_caching = enable_prefix_caching
+        if enable_prefix_caching:
+            from megatron_amd.inference.contexts import PrefixCachingAllocator
+
+            # swap in the refcounted reuse allocator (same id space)
+            self.context.allocator = PrefixCachingAllocator(num_blocks, first_id=1)
 
     def add_request(self, prompt, params: SamplingParams = SamplingParams()) -> int:
         if self.tokenizer is not None and isinstance(prompt, str):
@@ -338,6 +344,40 @@ class DynamicInferenceEngine:
         req.block_table.extend(self.context.allocator.allocate(need))
         return True
 
+    def _reuse_prefix(self, req: _Request):
+        """Adopt cached KV blocks for the longest full-block prompt prefix.
+        The reused tokens are marked cached so chunked prefill starts after
+        them (they are never recomputed).  At least one token is always
+        prefilled so the first sampled logits exist."""
+        from megatron_amd.inference.contexts import prompt_block_hashes
+
+        bs = self.context.block_size
+        hashes = prompt_block_hashes(req.prompt, bs)
+        req._hashes = hashes
+        usable = len(hashes)
+        if usable * bs == len(req.prompt):
+            usable -= 1  # keep >= 1 token to prefill
+        for h in hashes[:usable]:
+            blk = self.context.allocator.lookup(h)
+            if blk is None:
+                break
+            req.block_table.append(blk)
+            req.cached += bs
+
+    def _register_prefix_blocks(self, req: _Request):
+        """After prefill progress: publish newly completed full prompt
+        blocks for reuse by later requests."""
+        from megatron_amd.inference.contexts import prompt_block_hashes
+
+        bs = self.context.block_size
+        hashes = getattr(req, "_hashes", None)
+        if hashes is None:
+            hashes = prompt_block_hashes(req.prompt, bs)
+            req._hashes = hashes
+        full = min(req.cached // bs, len(hashes))
+        for i in range(full):
+            self.context.allocator.register(hashes[i], req.block_table[i])
+
     def _hit_stop_string(self, req: _Request) -> bool:
         if not req.params.stop_strings or self.tokenizer is None:
             return False
@@ -373,6 +413,8 @@ class DynamicInferenceEngine:
         self._restore_preempted()
         if self.waiting and len(self.active) < self.max_batch:
             req = self.waiting[0]
+            if self.prefix_caching and req.cached == 0 and not req.block_table:
+                self._reuse_prefix(req)
             chunk = min(self.max_prefill_tokens, len(req.prompt) - req.cached)
             if not self._ensure_blocks(req, req.cached + chunk):
                 if not self.active:
@@ -383,6 +425,8 @@ class DynamicInferenceEngine:
                                        device=self.device).view(1, -1)
                 logits_tp = self._model_forward(toks)  # [chunk, 1, V/tp] on pp-last
                 req.cached += chunk
+                if self.prefix_caching:
+                    self._register_prefix_blocks(req)
                 if req.cached == len(req.prompt):
                     if logits_tp is not None:
                         logits = _full_logits(logits_tp[-1, 0]).float()
