@@ -42,12 +42,28 @@ def _gather_weight(shard: torch.Tensor, group) -> torch.Tensor:
     return torch.cat(parts, dim=0)
 
 
+def _gather_weight_async(shard: torch.Tensor, group):
+    """Launch the shard all-gather without blocking; returns (buf, handle)
+    (the GTP prefetch chain's primitive; reference gtp streams :352)."""
+    world = dist.get_world_size(group) if (dist.is_initialized() and group is not None) else 1
+    if world == 1:
+        return shard, None
+    buf = torch.empty(world * shard.shape[0], *shard.shape[1:],
+                      dtype=shard.dtype, device=shard.device)
+    with torch.no_grad():
+        h = dist.all_gather_into_tensor(buf, shard.contiguous(), group=group, async_op=True)
+    return buf, h
+
+
 class _GTPLinearFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight_shard, bias, group):
-        full_w = _gather_weight(weight_shard, group)
+    def forward(ctx, x, weight_shard, bias, group, module):
+        full_w = module._take_prefetched() if module is not None else None
+        if full_w is None:
+            full_w = _gather_weight(weight_shard, group)
         ctx.save_for_backward(x, weight_shard)
         ctx.group = group
+        ctx.module = module
         out = torch.matmul(x, full_w.t())
         if bias is not None:
             out = out + bias
@@ -58,7 +74,9 @@ class _GTPLinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight_shard = ctx.saved_tensors
         group = ctx.group
-        full_w = _gather_weight(weight_shard, group)  # rematerialize
+        full_w = ctx.module._take_prefetched() if ctx.module is not None else None
+        if full_w is None:
+            full_w = _gather_weight(weight_shard, group)  # rematerialize
         dx = torch.matmul(dy, full_w)
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
@@ -77,7 +95,7 @@ class _GTPLinearFn(torch.autograd.Function):
         else:
             dw_shard = dw_full
         db = dy2.sum(dim=0) if ctx.has_bias else None
-        return dx, dw_shard, db, None
+        return dx, dw_shard, db, None, None
 
 
 class GTPLinear(nn.Module):
@@ -110,5 +128,49 @@ class GTPLinear(nn.Module):
                 nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
         self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
 
+        self._prefetched = None  # (buf, handle) set by a GTPChain
+
+    def prefetch(self):
+        """Launch this layer's weight all-gather ahead of time (chain)."""
+        if self._prefetched is None:
+            self._prefetched = _gather_weight_async(self.weight, self.group)
+
+    def _take_prefetched(self):
+        if self._prefetched is None:
+            return None
+        buf, h = self._prefetched
+        self._prefetched = None
+        if h is not None:
+            with torch.no_grad():
+                h.wait()  # gloo completes chunk-view copies inside wait
+        return buf
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return _GTPLinearFn.apply(x, self.weight, self.bias, self.group)
+        return _GTPLinearFn.apply(x, self.weight, self.bias, self.group, self)
+
+
+class GTPChain:
+    """Weight-gather prefetch chain over GTP layers in execution order
+    (reference generalized_tensor_parallelism chains :103): when layer i
+    starts its forward, layer i+1's all-gather is already in flight (and
+    layer i-1's during backward), hiding gather latency under the GEMMs."""
+
+    def __init__(self, layers):
+        self.layers = list(layers)
+        for i, m in enumerate(self.layers):
+            m.register_forward_pre_hook(self._fwd_pre(i))
+            m.register_full_backward_pre_hook(self._bwd_pre(i))
+
+    def _fwd_pre(self, i):
+        def hook(mod, args):
+            mod.prefetch()
+            if i + 1 < len(self.layers):
+                self.layers[i + 1].prefetch()
+        return hook
+
+    def _bwd_pre(self, i):
+        def hook(mod, gout):
+            mod.prefetch()
+            if i - 1 >= 0:
+                self.layers[i - 1].prefetch()
+        return hook

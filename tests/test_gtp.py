@@ -72,3 +72,50 @@ def _gtp2_case(rank, world):
 
 def test_gtp2_matches_linear():
     spawn_dist(_gtp2_case, 2)
+
+
+def _gtp_chain_case(rank, world):
+    """Chained GTP layers prefetch each other's gathers and match the
+    unchained computation exactly."""
+    import torch.distributed as dist
+    import torch.nn as nn
+
+    from megatron_amd.parallel import grid as G
+    from megatron_amd.parallel.gtp import GTPChain, GTPLinear
+
+    G.initialize_model_parallel()
+    torch.manual_seed(0)
+
+    def build():
+        torch.manual_seed(0)
+        ls = [GTPLinear(16, 16, group=dist.group.WORLD,
+                        init_method=lambda t: nn.init.normal_(t, 0, 0.1))
+              for _ in range(3)]
+        return ls
+
+    plain = build()
+    chained = build()
+    GTPChain(chained)
+    x = torch.randn(4, 16, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+
+    def run(layers, inp):
+        h = inp
+        for l in layers:
+            h = torch.relu(l(h))
+        return h
+
+    out_p = run(plain, x)
+    out_c = run(chained, x2)
+    torch.testing.assert_close(out_c, out_p, rtol=1e-6, atol=1e-7)
+    out_p.sum().backward()
+    out_c.sum().backward()
+    torch.testing.assert_close(x2.grad, x.grad, rtol=1e-6, atol=1e-7)
+    for a, b in zip(plain, chained):
+        torch.testing.assert_close(b.weight.grad, a.weight.grad, rtol=1e-6, atol=1e-7)
+
+
+def test_gtp_chain_prefetch_matches():
+    from tests.utils import spawn_dist
+
+    spawn_dist(_gtp_chain_case, 2)
