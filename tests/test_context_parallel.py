@@ -197,3 +197,81 @@ def test_balanced_cp_fuzz():
             assert len(ranks) == next(a.cp_size for aa in per_rank.values()
                                       for a in aa if a.sample == i)
         assert sched.balance_ratio(seq_lens) < 2.5
+
+
+# --- hybrid (per-sample) CP wired through the scheduler ---------------------
+
+
+def _hybrid_cp_case(rank, world):
+    """Two samples of different lengths: the short one runs on ONE rank
+    (cp=1), the long one ring-attends across both (cp=2); every output
+    slice must equal the single-process full attention of its sample."""
+    import torch.distributed as dist
+
+    from megatron_amd.parallel import grid as G
+    from megatron_amd.parallel.balanced_cp import BalancedCPScheduler
+    from megatron_amd.parallel.context_parallel import cp_chunk_ids, slice_for_cp_rank
+    from megatron_amd.parallel.hybrid_cp import (
+        HybridCPGroups,
+        build_hybrid_cp_batch,
+        hybrid_cp_attention,
+    )
+    from megatron_amd.ops import reference as ref
+
+    G.initialize_model_parallel()
+    groups = HybridCPGroups(world)
+
+    torch.manual_seed(7)
+    b, h, d = 1, 2, 16
+    lens = [8, 16]
+    qkv = [tuple(torch.randn(L, b, h, d) for _ in range(3)) for L in lens]
+    for tens in qkv:
+        for t in tens:
+            dist.broadcast(t, src=0)
+
+    sched = BalancedCPScheduler(world_size=world, max_cp=world, chunk_target=8)
+    per_rank = sched.schedule(lens)
+    # slice q/k/v per assignment
+    my = sorted(per_rank[rank], key=lambda a: a.sample)
+    batch_qkv = []
+    for a in my:
+        q, k, v = qkv[a.sample]
+        if a.cp_size == 1:
+            batch_qkv.append((q, k, v))
+        else:
+            batch_qkv.append(tuple(
+                slice_for_cp_rank(t, a.cp_rank, a.cp_size, seq_dim=0, mode="p2p")
+                for t in (q, k, v)))
+    outs = hybrid_cp_attention(batch_qkv, my, groups)
+
+    for a, out in zip(my, outs):
+        q, k, v = qkv[a.sample]
+        full = ref.attention(q.float(), k.float(), v.float(), causal=True)
+        if a.cp_size == 1:
+            expect = full
+        else:
+            expect = slice_for_cp_rank(full, a.cp_rank, a.cp_size, seq_dim=0, mode="p2p")
+        err = (out.float() - expect).abs().max()
+        assert float(err) < 1e-4, (a, float(err))
+
+
+def test_hybrid_cp_per_sample_groups():
+    spawn_dist(_hybrid_cp_case, 2)
+
+
+def test_build_hybrid_cp_batch_packs_slices():
+    from megatron_amd.parallel.balanced_cp import BalancedCPScheduler
+    from megatron_amd.parallel.hybrid_cp import build_hybrid_cp_batch
+
+    sched = BalancedCPScheduler(world_size=2, max_cp=2, chunk_target=8)
+    lens = [8, 16, 8]
+    per_rank = sched.schedule(lens)
+    samples = [torch.arange(L).float().unsqueeze(-1) for L in lens]
+    total = 0
+    for r in (0, 1):
+        batch = build_hybrid_cp_batch(samples, per_rank, r)
+        packed = batch.packed()
+        assert packed.shape[0] == int(batch.cu_seqlens[-1])
+        total += packed.shape[0]
+    # every token placed exactly once across ranks
+    assert total == sum(lens)
