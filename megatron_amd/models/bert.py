@@ -57,6 +57,8 @@ class BertModel(nn.Module):
                 gather_output=False, skip_bias_add=True)
             if pre_process:
                 self.output_layer.weight = self.embedding.weight
+            self.pooler = (nn.Linear(config.hidden_size, config.hidden_size,
+                                     dtype=config.params_dtype) if add_binary_head else None)
             self.binary_head = (nn.Linear(config.hidden_size, 2, dtype=config.params_dtype)
                                 if add_binary_head else None)
         self.input_tensor: Optional[torch.Tensor] = None
@@ -78,13 +80,17 @@ class BertModel(nn.Module):
                 hidden = hidden + self.tokentype_embedding(tokentype_ids).transpose(0, 1).to(hidden.dtype)
         else:
             hidden = self.input_tensor
-        hidden = self.encoder(hidden)
+        hidden = self.encoder(hidden, attention_mask=attention_mask)
         if not self.post_process:
             return hidden
         logits, _ = self.output_layer(self.lm_head(hidden))  # [s, b, V/tp]
-        binary = self.binary_head(hidden[0]) if self.binary_head is not None else None
+        self.binary_logits = None
+        if self.binary_head is not None:
+            # reference Pooler: tanh(dense(first-token hidden)) -> NSP head
+            pooled = torch.tanh(self.pooler(hidden[0]))
+            self.binary_logits = self.binary_head(pooled)
         if labels is None:
-            return (logits, binary) if binary is not None else logits
+            return logits
         loss = vocab_parallel_cross_entropy(logits, labels.transpose(0, 1).contiguous())
         if loss_mask is not None:
             loss = loss * loss_mask.transpose(0, 1).to(loss.dtype)

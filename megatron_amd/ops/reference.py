@@ -237,3 +237,32 @@ def attention(
     probs = torch.softmax(scores, dim=-1)
     out = torch.matmul(probs, vf)  # [b, hq, s, d]
     return out.permute(2, 0, 1, 3).to(q.dtype)
+
+
+def attention_padded(q, k, v, key_valid_mask, causal: bool = False, scale=None):
+    """Attention with a [b, s_k] key-padding mask (True = valid token).
+
+    Capability analog of the reference's arbitrary-mask (non-flash) path
+    (fused_softmax scaled_masked variant): used by BERT-style bidirectional
+    batches with right padding.  fp32 softmax; [s,b,h,d] layouts."""
+    import math as _math
+
+    if scale is None:
+        scale = 1.0 / _math.sqrt(q.shape[-1])
+    s_q, b, hq, d = q.shape
+    hkv = k.shape[2]
+    rep = hq // hkv
+    qf = q.permute(1, 2, 0, 3).float()
+    kf = k.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
+    vf = v.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b, hq, s_q, s_k]
+    bad = ~key_valid_mask.to(torch.bool)                     # [b, s_k]
+    scores = scores.masked_fill(bad[:, None, None, :], float("-inf"))
+    if causal:
+        s_k = k.shape[0]
+        cm = torch.ones(s_q, s_k, dtype=torch.bool, device=q.device).tril_(s_k - s_q)
+        scores = scores.masked_fill(~cm, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    probs = torch.nan_to_num(probs)  # fully-masked rows (pad queries) -> 0
+    out = torch.matmul(probs, vf)
+    return out.permute(2, 0, 1, 3).to(q.dtype)

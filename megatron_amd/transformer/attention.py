@@ -97,6 +97,15 @@ class SelfAttention(nn.Module):
             core_out = inference_context.attend(
                 self.layer_number, q, k, v, self.softmax_scale, self.window
             )
+        elif attention_mask is not None:
+            # key-padding mask ([b, s] bool, True = valid): the arbitrary-
+            # mask path (BERT-style padded batches) — torch composition,
+            # not the flash kernel (reference's unfused masked-softmax path)
+            from megatron_amd.ops import reference as _ref
+
+            core_out = _ref.attention_padded(
+                q, k, v, attention_mask, causal=self.config.causal_attention,
+                scale=self.softmax_scale)
         elif G.get_context_parallel_world_size() > 1:
             from megatron_amd.parallel.context_parallel import ring_attention, ulysses_attention
 

@@ -79,3 +79,40 @@ def test_pretrain_bert_end_to_end():
         "--activation", "gelu", "--position-embedding-type", "learned",
     ], forward_step_builder=pretrain_bert.forward_step_builder)
     assert it == 2
+
+
+def test_bert_padding_mask_isolates_pad_tokens():
+    """A padded short sample must produce the same valid-position logits as
+    the unpadded sample (the key-padding mask path, reference masked-
+    softmax behavior)."""
+    init_single()
+    cfg = _cfg()
+    torch.manual_seed(0)
+    m = BertModel(cfg)
+    m.eval()
+    toks = torch.randint(0, cfg.vocab_size, (1, 12))
+    with torch.no_grad():
+        full = m(toks, attention_mask=torch.ones(1, 12, dtype=torch.bool))
+        padded_toks = torch.cat([toks, torch.zeros(1, 4, dtype=torch.long)], dim=1)
+        mask = torch.cat([torch.ones(1, 12, dtype=torch.bool),
+                          torch.zeros(1, 4, dtype=torch.bool)], dim=1)
+        padded = m(padded_toks, attention_mask=mask)
+    torch.testing.assert_close(padded[:12], full, rtol=1e-4, atol=1e-5)
+
+
+def test_bert_binary_head_and_pooler():
+    init_single()
+    cfg = _cfg()
+    torch.manual_seed(0)
+    m = BertModel(cfg, add_binary_head=True)
+    toks = torch.randint(0, cfg.vocab_size, (2, 8))
+    labels = torch.randint(0, cfg.vocab_size, (2, 8))
+    loss = m(toks, labels=labels)
+    binary = m.binary_logits
+    assert binary.shape == (2, 2)
+    # NSP + MLM joint training step
+    nsp_tgt = torch.tensor([0, 1])
+    total = loss.sum() + torch.nn.functional.cross_entropy(binary, nsp_tgt)
+    total.backward()
+    assert m.pooler.weight.grad is not None
+    assert m.binary_head.weight.grad is not None
