@@ -209,12 +209,14 @@ def attention(
     causal: bool = True,
     scale: Optional[float] = None,
     window: Optional[int] = None,
+    dropout_p: float = 0.0,
+    training: bool = False,
 ) -> torch.Tensor:
     """Unfused reference attention.
 
     q: [s, b, hq, d], k/v: [s, b, hkv, d] -> out [s, b, hq, d].
-    GQA: hq is a multiple of hkv.  fp32 softmax.
-    """
+    GQA: hq is a multiple of hkv.  fp32 softmax; optional attention-prob
+    dropout (the reference's attention_dropout — applied post-softmax)."""
     s, b, hq, d = q.shape
     hkv = k.shape[2]
     rep = hq // hkv
@@ -235,6 +237,8 @@ def attention(
             mask &= torch.ones(s, sk, dtype=torch.bool, device=q.device).triu_(off - window + 1)
         scores = scores.masked_fill(~mask, float("-inf"))
     probs = torch.softmax(scores, dim=-1)
+    if dropout_p > 0.0 and training:
+        probs = torch.nn.functional.dropout(probs, p=dropout_p)
     out = torch.matmul(probs, vf)  # [b, hq, s, d]
     return out.permute(2, 0, 1, 3).to(q.dtype)
 

@@ -125,6 +125,18 @@ class SelfAttention(nn.Module):
                                            scale=self.softmax_scale, window=self.window)
 
             core_out = rng_checkpoint(_core, False, q, k, v)
+        elif self.config.attention_dropout > 0.0 and self.training:
+            # attention-prob dropout is not in the flash kernel; the torch
+            # composition applies it exactly (RNG-tracker forked so TP ranks
+            # draw independent masks for their own heads)
+            from megatron_amd.ops import reference as _ref
+            from megatron_amd.parallel.random import get_rng_tracker
+
+            with get_rng_tracker().fork():
+                core_out = _ref.attention(
+                    q, k, v, causal=self.config.causal_attention,
+                    scale=self.softmax_scale, window=self.window,
+                    dropout_p=self.config.attention_dropout, training=True)
         else:
             core_out = ops.flash_attention(q, k, v, causal=self.config.causal_attention,
                                            scale=self.softmax_scale, window=self.window)

@@ -201,3 +201,29 @@ def test_moe_shared_expert_stream_overlap_gpu():
         seq = seq + layer.shared_expert(tokens)
         seq = seq.view(x.shape).to(x.dtype)
     assert torch.allclose(out.detach(), seq, atol=3e-2, rtol=3e-2)
+
+
+def test_gpu_dropout_model_step():
+    """GPU model step with dropout > 0 (r1 gap: no GPU test ran dropout)."""
+    init_single()
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    model_parallel_seed(3)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=128, num_attention_heads=4, num_query_groups=2,
+        vocab_size=128, ffn_hidden_size=256, hidden_dropout=0.1, attention_dropout=0.1,
+        params_dtype=torch.bfloat16, bf16=True, gradient_accumulation_fusion=False,
+        max_position_embeddings=64)
+    from megatron_amd.models.gpt import GPTModel
+
+    m = GPTModel(cfg).cuda()
+    toks = torch.randint(0, 128, (2, 32), device="cuda")
+    labels = torch.randint(0, 128, (2, 32), device="cuda")
+    m.train()
+    l1 = m(toks, labels=labels)
+    l2 = m(toks, labels=labels)
+    assert not torch.allclose(l1, l2)
+    l1.float().sum().backward()
+    for p in m.parameters():
+        if p.grad is not None:
+            assert torch.isfinite(p.grad.float()).all()
