@@ -57,8 +57,11 @@ class _FSDPUnit:
         self.module = module
         self.params = params
         self.dp_group = dp_group
-        self.dp_size = dist.get_world_size(dp_group)
-        self.dp_rank = dist.get_rank(dp_group)
+        if dist.is_initialized():
+            self.dp_size = dist.get_world_size(dp_group)
+            self.dp_rank = dist.get_rank(dp_group)
+        else:
+            self.dp_size, self.dp_rank = 1, 0
         self.reshard_after_forward = reshard_after_forward
 
         self.dtype = params[0].dtype
@@ -111,6 +114,14 @@ class _FSDPUnit:
                 self.finish_unshard()  # a prefetch is in flight: just wait
             return
         self._flat = torch.empty(self.flat_size, dtype=self.dtype, device=self.model_shard.device)
+        if self.dp_size == 1:
+            self._flat.copy_(self.model_shard)
+            if not async_op:
+                self._set_flat_views()
+            else:
+                self._ag_handle = None
+                self._set_flat_views()
+            return
         h = dist.all_gather_into_tensor(self._flat, self.model_shard, group=self.dp_group,
                                         async_op=async_op)
         if async_op:
@@ -149,9 +160,12 @@ class _FSDPUnit:
             if p.grad is not None:
                 flat_grad[o : o + n].copy_(p.grad.reshape(-1)).mul_(inv)
                 p.grad = None
-        tmp = torch.empty_like(self.grad_shard)
-        dist.reduce_scatter_tensor(tmp, flat_grad, group=self.dp_group)
-        self.grad_shard += tmp  # accumulates across microbatches
+        if self.dp_size == 1:
+            self.grad_shard += flat_grad
+        else:
+            tmp = torch.empty_like(self.grad_shard)
+            dist.reduce_scatter_tensor(tmp, flat_grad, group=self.dp_group)
+            self.grad_shard += tmp  # accumulates across microbatches
         if self.is_last_microbatch:
             self.master_shard.grad = self.grad_shard
         self.reshard()
@@ -183,7 +197,8 @@ class FullyShardedDataParallel(nn.Module):
             assert not getattr(p, "is_expert_parallel", False), (
                 f"FSDP over expert-parallel params ({n}) is unsupported; "
                 "use EP + the distributed optimizer instead")
-        self.dp_group = dp_group if dp_group is not None else G.get_data_parallel_group()
+        self.dp_group = (dp_group if dp_group is not None
+                         else (G.get_data_parallel_group() if dist.is_initialized() else None))
         self.units: List[_FSDPUnit] = []
         self._param_to_unit: Dict[nn.Parameter, _FSDPUnit] = {}
 
@@ -222,6 +237,8 @@ class FullyShardedDataParallel(nn.Module):
     def _sync_shards_from_rank0(self):
         """Ranks may have different random init; shard content must agree on
         the weights rank 0 holds.  Broadcast each rank's shard source."""
+        if not dist.is_initialized():
+            return
         for u in self.units:
             # rebuild rank-0's flat buffer on every rank
             flat = torch.empty(u.flat_size, dtype=u.dtype, device=u.model_shard.device)
@@ -313,7 +330,8 @@ class FullyShardedDataParallel(nn.Module):
                          device=self.units[0].model_shard.device)
         for u in self.units:
             sq += u.grad_shard.float().pow(2).sum()
-        dist.all_reduce(sq, group=self.dp_group)
+        if dist.is_initialized():
+            dist.all_reduce(sq, group=self.dp_group)
         norm = sq.sqrt()
         scale = max_norm / (norm + 1e-6)
         if scale < 1.0:

@@ -135,7 +135,14 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
     device = torch.device("cuda", torch.cuda.current_device()) if torch.cuda.is_available() else torch.device("cpu")
     timers = Timers()
 
-    chunks, optimizer = setup_model_and_optimizer(model_provider, cfg, opt_cfg, ddp_cfg, device=device)
+    if getattr(args, "use_fsdp", False):
+        from megatron_amd.training.training import setup_fsdp_model_and_optimizer
+
+        assert cfg.pipeline_parallel_size == 1 and cfg.tensor_parallel_size == 1, (
+            "--use-fsdp composes with DP only (TP/PP use DDP + ZeRO-1)")
+        chunks, optimizer = setup_fsdp_model_and_optimizer(model_provider, cfg, opt_cfg, device=device)
+    else:
+        chunks, optimizer = setup_model_and_optimizer(model_provider, cfg, opt_cfg, ddp_cfg, device=device)
     scheduler = OptimizerParamScheduler(optimizer, opt_cfg, args.train_iters)
     if getattr(args, "gpu_sniff_test", False):
         from megatron_amd.utils.gpu_health import gpu_sniff_test

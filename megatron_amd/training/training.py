@@ -20,6 +20,85 @@ from megatron_amd.pipeline.schedules import get_forward_backward_func
 from megatron_amd.utils.rerun_state_machine import get_rerun_state_machine
 
 
+class FSDPTrainAdapter:
+    """Adapts FullyShardedDataParallel + a plain AdamW over its fp32 master
+    shards to the train_step optimizer/chunk contract (reference
+    torch_fully_sharded_data_parallel.py wrapper role)."""
+
+    def __init__(self, fsdp, opt_config: OptimizerConfig):
+        self.fsdp = fsdp
+        self.config = opt_config
+        shards = fsdp.shard_parameters()
+        self.inner = torch.optim.AdamW(
+            shards, lr=opt_config.lr, weight_decay=opt_config.weight_decay,
+            betas=(opt_config.adam_beta1, opt_config.adam_beta2), eps=opt_config.adam_eps)
+        self.chained_optimizers = [self]
+        self.step_count = 0
+
+    # -- optimizer contract used by train_step / scheduler ------------------
+    def zero_grad(self):
+        self.inner.zero_grad(set_to_none=True)
+
+    def finish_grad_sync(self):
+        pass  # per-unit reduce-scatter completes at grad-ready hooks
+
+    def step(self):
+        norm = None
+        if self.config.clip_grad and self.config.clip_grad > 0:
+            norm = self.fsdp.clip_grad_norm(self.config.clip_grad)
+        self.inner.step()
+        self.fsdp.update_model_shards()
+        self.step_count += 1
+        return True, norm, None
+
+    def reload_model_params(self):
+        pass
+
+    def state_dict(self):
+        return {"inner": self.inner.state_dict(), "step": self.step_count}
+
+    def load_state_dict(self, sd):
+        self.inner.load_state_dict(sd["inner"])
+        self.step_count = sd.get("step", 0)
+
+    def set_lr(self, lr: float):
+        for g in self.inner.param_groups:
+            g["lr"] = lr
+
+    def set_wd(self, wd: float):
+        for g in self.inner.param_groups:
+            g["weight_decay"] = wd
+
+    def get_lr(self) -> float:
+        return self.inner.param_groups[0]["lr"]
+
+    @property
+    def param_groups(self):
+        return self.inner.param_groups
+
+
+def setup_fsdp_model_and_optimizer(model_provider, config, opt_config,
+                                   device: Optional[torch.device] = None):
+    """--use-fsdp path: ZeRO-3 sharding instead of DDP + ZeRO-1."""
+    from megatron_amd.distributed.fsdp import FullyShardedDataParallel
+
+    if device is not None:
+        with torch.device(device):
+            m = model_provider(config, pre_process=True, post_process=True, vp_stage=None)
+        m = m.to(device)
+    else:
+        m = model_provider(config, pre_process=True, post_process=True, vp_stage=None)
+    fsdp = FullyShardedDataParallel(m)
+    # train_step chunk surface
+    fsdp.no_sync = fsdp.no_last_microbatch
+    fsdp.start_grad_sync = lambda: None
+    fsdp.finish_grad_sync = lambda: None
+    fsdp.broadcast_params = lambda: None
+    opt = FSDPTrainAdapter(fsdp, opt_config)
+    config.finalize_model_grads_func = None
+    return [fsdp], opt
+
+
 def setup_model_and_optimizer(
     model_provider: Callable[..., torch.nn.Module],
     config: TransformerConfig,

@@ -210,3 +210,12 @@ def test_tiny_t5_loss_curve_matches_golden():
     finally:
         P.train_step = orig
     _check_or_record("tiny_t5.json", losses)
+
+
+def test_tiny_gpt_fsdp_tracks_ddp_golden():
+    """--use-fsdp (ZeRO-3 path) must follow the DDP golden trajectory
+    (same math; torch-AdamW vs fused-AdamW drift only)."""
+    golden = json.load(open(GOLDEN))
+    losses = _run_config(["--use-fsdp", "--train-iters", "8"])
+    for i, (got, want) in enumerate(zip(losses, golden["lm_loss"])):
+        assert abs(got - want) < 5e-2, (i, losses, golden["lm_loss"])
