@@ -253,8 +253,38 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
     _print_rank0(f"training: {args.train_iters} iters, GBS {args.global_batch_size}, "
                  f"microbatches {args.num_microbatches}, dp {args.data_parallel_size}")
 
+    graphs_pending = bool(getattr(args, "hip_graphs", False)) and torch.cuda.is_available()
+
+    def _capture_graphs():
+        """--hip-graphs: per-layer fwd/bwd capture after the first iteration
+        (a clean step boundary), under no_sync so capture-time grad-ready
+        callbacks cannot launch reduces (see transformer/hip_graphs.py)."""
+        import contextlib as _ctx
+
+        from megatron_amd.transformer.hip_graphs import capture_block_hip_graphs
+
+        core = chunks[0].module if hasattr(chunks[0], "module") else chunks[0]
+        if args.num_experts or cfg.pipeline_parallel_size > 1 or n_chunks > 1:
+            _print_rank0("--hip-graphs: skipped (MoE routing / PP chunks are dynamic)")
+            return
+        s_local = args.seq_length
+        if cfg.sequence_parallel and cfg.tensor_parallel_size > 1:
+            s_local //= cfg.tensor_parallel_size
+        sample = torch.randn(s_local, args.micro_batch_size, cfg.hidden_size,
+                             device=device, dtype=cfg.params_dtype)
+        freqs = core._rotary_freqs(args.seq_length, device)
+        with _ctx.ExitStack() as stack:
+            for ch in chunks:
+                if hasattr(ch, "no_sync"):
+                    stack.enter_context(ch.no_sync())
+            n = capture_block_hip_graphs(core.decoder, sample, rotary_freqs=freqs)
+        _print_rank0(f"--hip-graphs: captured {n} layer graphs")
+
     prof = None
     while iteration < args.train_iters:
+        if graphs_pending and iteration >= 1:
+            _capture_graphs()
+            graphs_pending = False
         if args.profile and iteration == args.profile_step_start and args.rank == 0:
             prof = torch.profiler.profile(
                 activities=[torch.profiler.ProfilerActivity.CPU, torch.profiler.ProfilerActivity.CUDA],
