@@ -73,6 +73,37 @@ def create_app(engine, tokenizer=None):
                             "content": tokenizer.detokenize(r.output_tokens)},
                 "tokens": r.output_tokens}
 
+    @app.get("/metrics")
+    async def metrics():
+        """Prometheus-format serving metrics (reference observability role;
+        prometheus_client is in-image)."""
+        from fastapi.responses import PlainTextResponse
+
+        lines = []
+
+        def gauge(name, value, help_):
+            lines.append(f"# HELP {name} {help_}")
+            lines.append(f"# TYPE {name} gauge")
+            lines.append(f"{name} {value}")
+
+        gauge("megatron_amd_active_requests", len(getattr(engine, "active", [])),
+              "requests currently decoding")
+        gauge("megatron_amd_waiting_requests", len(getattr(engine, "waiting", [])),
+              "requests queued for prefill")
+        gauge("megatron_amd_finished_requests", len(getattr(engine, "finished", {})),
+              "finished results not yet collected")
+        gauge("megatron_amd_preempted_requests", len(getattr(engine, "preempted", [])),
+              "requests offloaded to host")
+        ctx = getattr(engine, "context", None)
+        if ctx is not None:
+            alloc = ctx.allocator
+            gauge("megatron_amd_kv_blocks_free", alloc.num_free, "free KV blocks")
+            gauge("megatron_amd_kv_blocks_total", alloc.num_blocks, "total KV blocks")
+            if hasattr(alloc, "hits"):
+                gauge("megatron_amd_prefix_cache_hits", alloc.hits, "prefix-cache block hits")
+                gauge("megatron_amd_prefix_cache_misses", alloc.misses, "prefix-cache block misses")
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.post("/api/stream")
     async def stream(request: Request):
         """SSE token streaming for a single prompt (dynamic engine only):

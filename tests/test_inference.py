@@ -621,3 +621,19 @@ def test_prefix_caching_concurrent_share_and_free(tiny_model):
     assert eng.finished[r2].output_tokens
     # allocator accounting intact: everything eventually reusable
     assert eng.context.allocator.num_free == eng.context.allocator.num_blocks
+
+
+def test_metrics_endpoint(tiny_model):
+    from fastapi.testclient import TestClient
+
+    from megatron_amd.inference.server import create_app
+
+    eng = _prefix_engine(tiny_model, enable_prefix_caching=True)
+    app = create_app(eng, None)
+    c = TestClient(app)
+    r = c.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    for key in ("megatron_amd_kv_blocks_free", "megatron_amd_prefix_cache_hits",
+                "megatron_amd_active_requests"):
+        assert key in body, body
