@@ -197,7 +197,6 @@ class MLASelfAttention(nn.Module):
 
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None,
                 inference_context=None, packed_seq_params=None):
-        assert packed_seq_params is None, "MLA packed (THD) path: round 2"
         s, b = hidden_states.shape[0], hidden_states.shape[1]
         if self.linear_q_down is not None:
             q, _ = self.linear_q_up(self.q_norm(self.linear_q_down(hidden_states)))
@@ -233,13 +232,23 @@ class MLASelfAttention(nn.Module):
         kv = kv.view(s, b, self.nh, self.nope + self.dv)
         k_nope, v = torch.split(kv, [self.nope, self.dv], dim=3)
 
-        freqs = self._rope_freqs(s, hidden_states.device)
+        if packed_seq_params is not None:
+            # THD pack: per-document RoPE positions + block-diagonal attention
+            table = self._rope_freqs(packed_seq_params.max_seqlen, hidden_states.device)
+            freqs = table[packed_seq_params.positions().to(hidden_states.device)]
+        else:
+            freqs = self._rope_freqs(s, hidden_states.device)
         q_rope = ops.rope_apply(q_rope.contiguous(), freqs)
         k_rope = ops.rope_apply(k_rope.view(s, b, 1, self.rope), freqs)
 
         qf = torch.cat([q_nope, q_rope], dim=3)
         kf = torch.cat([k_nope, k_rope.expand(s, b, self.nh, self.rope)], dim=3)
-        if self.dqk in (64, 128) and self.dv == self.dqk:
+        if packed_seq_params is not None:
+            core = ref.attention_varlen(
+                qf, kf.contiguous(), v.contiguous(),
+                packed_seq_params.cu_seqlens.to(qf.device),
+                causal=self.config.causal_attention, scale=self.softmax_scale)
+        elif self.dqk in (64, 128) and self.dv == self.dqk:
             core = ops.flash_attention(qf, kf.contiguous(), v.contiguous(),
                                        causal=self.config.causal_attention,
                                        scale=self.softmax_scale)

@@ -169,3 +169,28 @@ def test_mla_disaggregated_matches_single():
     got = disaggregated_generate(pre, dec, prompts, params)
     for a, b in zip(expected, got):
         assert a.output_tokens == b.output_tokens
+
+
+def test_mla_packed_sequences_match_separate():
+    """Two documents packed into one [T, 1] stream attend block-diagonally
+    with per-document RoPE restart: each document's output equals running it
+    alone (closes the round-1 'MLA packed: round 2' limitation)."""
+    from megatron_amd.transformer.packed_seq import PackedSeqParams
+
+    from megatron_amd.transformer.multi_latent_attention import MLASelfAttention
+
+    init_single()
+    torch.manual_seed(0)
+    cfg = _cfg()
+    att = MLASelfAttention(cfg, layer_number=1).eval()
+    l1, l2 = 12, 8
+    h1 = torch.randn(l1, 1, cfg.hidden_size)
+    h2 = torch.randn(l2, 1, cfg.hidden_size)
+    with torch.no_grad():
+        o1 = att(h1)
+        o2 = att(h2)
+        packed = torch.cat([h1, h2], dim=0)
+        psp = PackedSeqParams.from_lengths([l1, l2])
+        op = att(packed, packed_seq_params=psp)
+    torch.testing.assert_close(op[:l1], o1, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(op[l1:], o2, rtol=1e-4, atol=1e-5)
