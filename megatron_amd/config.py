@@ -101,6 +101,7 @@ class TransformerConfig(ParallelConfig):
     moe_token_dispatcher_type: str = "alltoall"  # 'alltoall' | 'allgather'
     moe_expert_capacity_factor: Optional[float] = None  # None -> dropless
     moe_token_drop_policy: str = "probs"  # 'probs' | 'position'
+    moe_pad_expert_input_to_capacity: bool = False  # static expert shapes
     moe_router_dtype: str = "fp32"
     # DeepSeek node-limited routing (reference moe_utils.py:673): experts split
     # into num_groups; each token routes only within its best group_topk groups.

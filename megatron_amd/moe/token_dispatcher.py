@@ -78,6 +78,30 @@ def unpermute(permuted: torch.Tensor, sort_order: torch.Tensor, probs: torch.Ten
     return out_flat.sum(dim=1)
 
 
+def pad_to_capacity(permuted: torch.Tensor, tokens_per_expert, capacity: int):
+    """Pad/position each expert's rows into a fixed [E*capacity, h] buffer
+    (reference moe_pad_expert_input_to_capacity): downstream grouped GEMMs
+    see STATIC shapes (graph-capturable).  Returns (padded, row_index) where
+    row_index[i] is the padded position of permuted row i (for unpadding)."""
+    E = len(tokens_per_expert)
+    counts = [int(x) for x in tokens_per_expert]
+    total = permuted.shape[0]
+    padded = permuted.new_zeros(E * capacity, permuted.shape[1])
+    idx = torch.empty(total, dtype=torch.long, device=permuted.device)
+    start = 0
+    for e, n in enumerate(counts):
+        n = min(n, capacity)
+        idx[start : start + n] = torch.arange(
+            e * capacity, e * capacity + n, device=permuted.device)
+        start += n
+    padded.index_copy_(0, idx, permuted)
+    return padded, idx
+
+
+def unpad_from_capacity(padded_out: torch.Tensor, row_index: torch.Tensor) -> torch.Tensor:
+    return padded_out.index_select(0, row_index)
+
+
 class MoEAlltoAllTokenDispatcher:
     """permute -> A2A(EP) -> sort-by-local-expert -> experts -> A2A -> unpermute."""
 
@@ -106,9 +130,10 @@ class MoEAlltoAllTokenDispatcher:
         self._sort_order = sort_order
         self._T = T
         self._probs = probs
+        self._pad_index = None
         if self.ep == 1:
             self._restore = None
-            return permuted, tokens_per_expert
+            return self._maybe_pad(permuted, tokens_per_expert)
 
         # per-EP-peer split sizes (each peer owns num_local_experts experts)
         counts_matrix = torch.empty(self.ep * self.num_experts, dtype=tokens_per_expert.dtype,
@@ -141,9 +166,29 @@ class MoEAlltoAllTokenDispatcher:
         reordered = torch.cat([chunks[i] for i in self._chunk_perm], dim=0) if len(chunks) > 1 else recv
         tokens_per_local_expert = my_slice.sum(dim=0)  # host tensor: experts sync-free downstream
         self._restore = True
-        return reordered, tokens_per_local_expert
+        return self._maybe_pad(reordered, tokens_per_local_expert)
+
+    def _maybe_pad(self, permuted, tokens_per_expert):
+        if not getattr(self.config, "moe_pad_expert_input_to_capacity", False):
+            return permuted, tokens_per_expert
+        import math
+
+        cf = getattr(self.config, "moe_expert_capacity_factor", None)
+        assert cf, "moe_pad_expert_input_to_capacity requires moe_expert_capacity_factor"
+        T_total = self._T * self._probs.shape[1]
+        cap = max(1, math.ceil(T_total / self.num_experts * cf))
+        if self.ep > 1:
+            # after the a2a each LOCAL expert holds up to `cap` tokens from
+            # EACH of the ep source ranks
+            cap *= self.ep
+        padded, self._pad_index = pad_to_capacity(permuted, tokens_per_expert, cap)
+        static_counts = torch.full((len(tokens_per_expert),), cap, dtype=torch.long)
+        return padded, static_counts
 
     def combine(self, expert_out: torch.Tensor) -> torch.Tensor:
+        if self._pad_index is not None:
+            expert_out = unpad_from_capacity(expert_out, self._pad_index)
+            self._pad_index = None
         if self.ep > 1:
             # invert the (expert, rank) reorder, then A2A back
             sizes = [self._chunk_sizes[i] for i in self._chunk_perm]

@@ -84,6 +84,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--moe-expert-capacity-factor", type=float, default=None,
                    help="None = dropless; otherwise capacity = ceil(T*topk/E * factor)")
     g.add_argument("--moe-token-drop-policy", choices=["probs", "position"], default="probs")
+    g.add_argument("--moe-pad-expert-input-to-capacity", action="store_true",
+                   help="pad each expert's tokens to exactly capacity (static grouped-GEMM shapes)")
 
     g = p.add_argument_group("parallelism")
     g.add_argument("--tensor-model-parallel-size", "--tp", type=int, default=1)
@@ -287,6 +289,7 @@ def configs_from_args(args):
         moe_router_bias_update_rate=args.moe_router_bias_update_rate,
         moe_expert_capacity_factor=args.moe_expert_capacity_factor,
         moe_token_drop_policy=args.moe_token_drop_policy,
+        moe_pad_expert_input_to_capacity=args.moe_pad_expert_input_to_capacity,
         qk_clip_threshold=args.qk_clip_threshold,
         label_smoothing=args.label_smoothing,
         overlap_moe_expert_parallel_comm=args.overlap_moe_expert_parallel_comm,

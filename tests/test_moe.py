@@ -508,3 +508,38 @@ def test_allgather_dispatcher_ep2_matches_ep1(tmp_path, monkeypatch):
     spawn_dist(_ag_ep_case, 2)
     ag_out = torch.load(out_path)
     assert_close(ref.detach(), ag_out, rtol=1e-5, atol=1e-6)
+
+
+def test_pad_to_capacity_static_shapes_match_unpadded():
+    """moe_pad_expert_input_to_capacity: experts see a fixed [E*cap, h]
+    buffer and the layer output equals the unpadded run (reference flag)."""
+    init_single()
+    torch.manual_seed(2)
+    x = torch.randn(6, 2, 32)
+
+    def run(pad):
+        init_single()
+        cfg = _cfg(moe_expert_capacity_factor=2.0,   # large: nothing dropped
+                   moe_pad_expert_input_to_capacity=pad)
+        layer = MoELayer(cfg)
+        _fill(layer)
+        seen = {}
+        orig = layer.experts.forward
+
+        def spy(tokens, tpe):
+            seen["shape"] = tuple(tokens.shape)
+            seen["counts"] = [int(c) for c in tpe]
+            return orig(tokens, tpe)
+
+        layer.experts.forward = spy
+        return layer(x), seen
+
+    out_pad, seen_pad = run(True)
+    out_ref, seen_ref = run(False)
+    torch.testing.assert_close(out_pad, out_ref, rtol=1e-5, atol=1e-6)
+    import math
+
+    cap = math.ceil(6 * 2 * 2 / 4 * 2.0)
+    assert seen_pad["shape"][0] == 4 * cap          # static buffer
+    assert all(c == cap for c in seen_pad["counts"])  # static per-expert size
+    assert seen_ref["shape"][0] != 4 * cap or sum(seen_ref["counts"]) != 4 * cap
