@@ -109,3 +109,73 @@ def test_quant_recipe_resolution():
     mods = dict(model.named_modules())
     first = [m for n, m in mods.items() if n.startswith("decoder.layers.1") and hasattr(m, "quant_recipe")]
     assert first and all(m.quant_recipe.name == "fp8" for m in first)
+
+
+# --- PTQ calibration / fake-quant / export flow -----------------------------
+
+
+def _ptq_model_and_cfg():
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.post_training.quant_config import QuantRecipeConfig
+    from tests.utils import init_single
+
+    init_single()
+    torch.manual_seed(0)
+    cfg = TransformerConfig(num_layers=2, hidden_size=32, num_attention_heads=4,
+                            vocab_size=64, ffn_hidden_size=48,
+                            gradient_accumulation_fusion=False)
+    m = GPTModel(cfg)
+    qc = QuantRecipeConfig.from_dict({
+        "rules": [{"match": "decoder.layers.*.mlp.*", "name": "fp8"},
+                  {"match": "decoder.layers.*.self_attention.*", "name": "fp8"}],
+        "default": {"name": "bf16"},
+    })
+    return m, qc
+
+
+def test_ptq_calibration_collects_amax():
+    from megatron_amd.post_training.ptq import CalibrationCollector
+
+    m, qc = _ptq_model_and_cfg()
+    toks = torch.randint(0, 64, (2, 8))
+    with CalibrationCollector(m, qc) as cal:
+        for _ in range(3):
+            m(toks)
+    assert cal.act_amax, "no activations collected"
+    for name, amax in cal.act_amax.items():
+        assert "mlp" in name or "self_attention" in name
+        assert float(amax) > 0 and cal.samples[name] == 3
+    # embeddings / output excluded by the pattern rules
+    assert not any("output_layer" in n for n in cal.act_amax)
+
+
+def test_ptq_fake_quant_bounds_error_and_keeps_quality():
+    from megatron_amd.post_training.ptq import quantize_model_weights
+
+    m, qc = _ptq_model_and_cfg()
+    toks = torch.randint(0, 64, (2, 8))
+    labels = torch.randint(0, 64, (2, 8))
+    with torch.no_grad():
+        base = m(toks, labels=labels).sum()
+        w_before = m.decoder.layers[0].mlp.linear_fc1.weight.detach().clone()
+        scales = quantize_model_weights(m, qc, mode="int8")
+        w_after = m.decoder.layers[0].mlp.linear_fc1.weight.detach()
+        quant = m(toks, labels=labels).sum()
+    assert scales
+    # weights changed but by bounded quantization error
+    assert not torch.equal(w_before, w_after)
+    rel = (w_after - w_before).abs().max() / w_before.abs().max()
+    assert float(rel) < 0.02
+    assert abs(float(quant - base)) / abs(float(base)) < 0.05
+
+
+def test_ptq_export_int8_payloads():
+    from megatron_amd.post_training.ptq import export_quantized_state_dict
+
+    m, qc = _ptq_model_and_cfg()
+    sd = export_quantized_state_dict(m, qc, mode="int8")
+    qw = [k for k, v in sd.items() if v.dtype == torch.int8]
+    assert qw and all(k + "_scale" in sd for k in qw)
+    # unmatched tensors exported untouched
+    assert any("embedding" in k and sd[k].dtype != torch.int8 for k in sd)
