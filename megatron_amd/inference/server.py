@@ -37,6 +37,7 @@ def create_app(engine, tokenizer=None):
             top_n_logprobs=int(req.get("top_n_logprobs", 0)),
             repetition_penalty=float(req.get("repetition_penalty", 1.0)),
             stop_strings=tuple(req.get("stop_strings", ())),
+            stop_on_eod=bool(req.get("stop_on_eod", True)),
             seed=req.get("seed"))
         with lock:
             results = engine.generate(prompts, params)

@@ -190,3 +190,42 @@ def test_rl_step_packed_runs():
                                      group_size=2, max_tokens=4, packed=True)
     assert torch.isfinite(torch.tensor(loss))
     assert len(rollouts) == 4
+
+
+def test_server_agent_rollouts_through_rest():
+    """ServerAgent: rollouts generated via the REST /api/generate endpoint
+    (server-integrated rollouts, reference megatron/rl agent API)."""
+    from fastapi.testclient import TestClient
+
+    from megatron_amd.inference.engine import StaticInferenceEngine
+    from megatron_amd.inference.server import create_app
+    from megatron_amd.rl.agent import InProcessAgent, ServerAgent
+
+    init_single()
+    model_parallel_seed(13)
+    cfg = TransformerConfig(**MODEL_KW)
+    model = GPTModel(cfg).eval()
+    engine = StaticInferenceEngine(model, max_batch=8, max_seq=128)
+    app = create_app(engine, None)
+    client = TestClient(app)
+
+    def post(url, payload):
+        r = client.post(url, json=payload)
+        assert r.status_code == 200, r.text
+        return r.json()
+
+    def env(prompt, resp):
+        return float(len(resp))
+
+    agent = ServerAgent(env, post)
+    rollouts = agent.get_rollouts([[3, 4, 5], [6, 7]], group_size=2, max_tokens=4, seed=1)
+    assert len(rollouts) == 4
+    for r in rollouts:
+        assert len(r.response_tokens) > 0
+        assert len(r.behavior_logprobs) == len(r.response_tokens)
+        assert r.reward == len(r.response_tokens)
+
+    # the in-process agent with the same seed produces the same tokens
+    local = InProcessAgent(model, env, max_seq=128).get_rollouts(
+        [[3, 4, 5], [6, 7]], group_size=2, max_tokens=4, seed=1)
+    assert [r.response_tokens for r in local] == [r.response_tokens for r in rollouts]
