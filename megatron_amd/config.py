@@ -67,7 +67,8 @@ class TransformerConfig(ParallelConfig):
     layernorm_epsilon: float = 1e-5
     # 'swiglu' (llama), 'gelu', 'squared_relu'
     activation: str = "swiglu"
-    add_linear_bias: bool = False  # llama-style: no bias anywhere
+    add_linear_bias: bool = False
+    add_qkv_bias: bool = False  # bias only on the fused QKV projection (Qwen2-style)  # llama-style: no bias anywhere
     untie_embeddings_and_output_weights: bool = True
     position_embedding_type: str = "rope"  # 'rope' | 'learned' | 'none'
     rotary_base: float = 500000.0  # llama-3 default

@@ -40,7 +40,8 @@ class SelfAttention(nn.Module):
         kv_size = self.num_query_groups * self.kv_channels
         self.qkv_size = q_size + 2 * kv_size
         self.linear_qkv = ColumnParallelLinear(
-            self.hidden_size, self.qkv_size, config=config, bias=config.add_linear_bias
+            self.hidden_size, self.qkv_size, config=config,
+            bias=config.add_linear_bias or getattr(config, "add_qkv_bias", False)
         )
         self.linear_proj = RowParallelLinear(
             q_size, self.hidden_size, config=config, bias=config.add_linear_bias

@@ -131,3 +131,46 @@ def test_export_hf_dir_round_trip(tmp_path):
         a = model(toks, position_ids=None, attention_mask=None)
         b = model2(toks, position_ids=None, attention_mask=None)
     assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_qwen2_style_qkv_bias_roundtrip():
+    """Qwen2-family: QKV biases fuse into the grouped layout, load into a
+    model with add_qkv_bias, and split back exactly."""
+    import tools.checkpoint.convert_hf as C
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+
+    init_single()
+    cfg = TransformerConfig(num_layers=2, hidden_size=32, num_attention_heads=4,
+                            num_query_groups=2, ffn_hidden_size=48, vocab_size=64,
+                            add_qkv_bias=True, untie_embeddings_and_output_weights=True,
+                            gradient_accumulation_fusion=False)
+    d = cfg.kv_channels
+    torch.manual_seed(0)
+    hf = {"model.embed_tokens.weight": torch.randn(64, 32),
+          "model.norm.weight": torch.randn(32), "lm_head.weight": torch.randn(64, 32)}
+    for i in range(2):
+        p = f"model.layers.{i}."
+        hf[p + "self_attn.q_proj.weight"] = torch.randn(4 * d, 32)
+        hf[p + "self_attn.k_proj.weight"] = torch.randn(2 * d, 32)
+        hf[p + "self_attn.v_proj.weight"] = torch.randn(2 * d, 32)
+        hf[p + "self_attn.q_proj.bias"] = torch.randn(4 * d)
+        hf[p + "self_attn.k_proj.bias"] = torch.randn(2 * d)
+        hf[p + "self_attn.v_proj.bias"] = torch.randn(2 * d)
+        hf[p + "self_attn.o_proj.weight"] = torch.randn(32, 4 * d)
+        hf[p + "mlp.gate_proj.weight"] = torch.randn(48, 32)
+        hf[p + "mlp.up_proj.weight"] = torch.randn(48, 32)
+        hf[p + "mlp.down_proj.weight"] = torch.randn(32, 48)
+        hf[p + "input_layernorm.weight"] = torch.randn(32)
+        hf[p + "post_attention_layernorm.weight"] = torch.randn(32)
+    sd = C.hf_to_mcore_state_dict(hf, cfg)
+    assert "decoder.layers.0.self_attention.linear_qkv.bias" in sd
+    model = GPTModel(cfg)
+    missing, unexpected = model.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rotary" in m or "freqs" in m for m in missing), missing
+    back = C.mcore_to_hf_state_dict(sd, cfg)
+    for k in hf:
+        if "bias" in k:
+            torch.testing.assert_close(back[k], hf[k])

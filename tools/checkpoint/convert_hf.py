@@ -44,6 +44,16 @@ def hf_to_mcore_state_dict(hf_sd: dict, cfg) -> dict:
             groups.append(k[g * d:(g + 1) * d])
             groups.append(v[g * d:(g + 1) * d])
         out[us + "self_attention.linear_qkv.weight"] = torch.cat(groups, dim=0)
+        if hf + "self_attn.q_proj.bias" in hf_sd:  # Qwen2-style QKV bias
+            qb = hf_sd[hf + "self_attn.q_proj.bias"]
+            kb = hf_sd[hf + "self_attn.k_proj.bias"]
+            vb = hf_sd[hf + "self_attn.v_proj.bias"]
+            bg = []
+            for g in range(ng):
+                bg.append(qb[g * rep * d:(g + 1) * rep * d])
+                bg.append(kb[g * d:(g + 1) * d])
+                bg.append(vb[g * d:(g + 1) * d])
+            out[us + "self_attention.linear_qkv.bias"] = torch.cat(bg, dim=0)
         out[us + "self_attention.linear_proj.weight"] = hf_sd[hf + "self_attn.o_proj.weight"]
         if hf + "block_sparse_moe.gate.weight" in hf_sd:
             # Mixtral MoE block: gate -> router, experts w1/w3 -> fused
@@ -92,6 +102,17 @@ def mcore_to_hf_state_dict(sd: dict, cfg) -> dict:
         out[hf + "self_attn.k_proj.weight"] = torch.cat(ks, 0)
         out[hf + "self_attn.v_proj.weight"] = torch.cat(vs, 0)
         out[hf + "self_attn.o_proj.weight"] = sd[us + "self_attention.linear_proj.weight"]
+        if us + "self_attention.linear_qkv.bias" in sd:
+            b = sd[us + "self_attention.linear_qkv.bias"]
+            qbs, kbs, vbs = [], [], []
+            for g in range(ng):
+                blk = b[g * gsz:(g + 1) * gsz]
+                qbs.append(blk[: rep * d])
+                kbs.append(blk[rep * d: rep * d + d])
+                vbs.append(blk[rep * d + d:])
+            out[hf + "self_attn.q_proj.bias"] = torch.cat(qbs, 0)
+            out[hf + "self_attn.k_proj.bias"] = torch.cat(kbs, 0)
+            out[hf + "self_attn.v_proj.bias"] = torch.cat(vbs, 0)
         if us + "mlp.router.weight" in sd:
             out[hf + "block_sparse_moe.gate.weight"] = sd[us + "mlp.router.weight"]
             w1 = sd[us + "mlp.experts.weight1"]  # [E, 2*moe_ffn, h]
@@ -132,6 +153,8 @@ def config_from_hf(hf_cfg):
         num_experts=getattr(hf_cfg, "num_local_experts", None),
         moe_router_topk=getattr(hf_cfg, "num_experts_per_tok", 2),
         moe_router_pre_softmax=False,
+        add_qkv_bias=bool(getattr(hf_cfg, "attention_bias", False)
+                          or "qwen2" in str(getattr(hf_cfg, "model_type", ""))),
     )
 
 
