@@ -116,6 +116,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
                    metavar=("START", "INCREMENT", "SAMPLES"),
                    help="grow GBS from START by INCREMENT over SAMPLES consumed samples")
     g.add_argument("--seq-length", type=int, default=4096)
+    g.add_argument("--packed-sequences", action="store_true",
+                   help="THD training: microbatch rows flattened into one packed stream "
+                        "(block-diagonal attention, per-document RoPE restart)")
     g.add_argument("--train-iters", type=int, default=10)
     g.add_argument("--eval-interval", type=int, default=0)
     g.add_argument("--eval-iters", type=int, default=2)

@@ -228,6 +228,28 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             from megatron_amd.parallel.context_parallel import get_batch_on_this_cp_rank
 
             batch = get_batch_on_this_cp_rank(batch, mode=cfg.cp_comm_type)
+        psp = None
+        if getattr(args, "packed_sequences", False):
+            # THD training: flatten the microbatch into one packed stream;
+            # per-row cu_seqlens (or one doc per row) shift to global offsets
+            from megatron_amd.transformer.packed_seq import PackedSeqParams
+
+            b, s = batch["tokens"].shape
+            if "cu_seqlens" in batch:
+                cu = [0]
+                for r in range(b):
+                    row_cu = [int(x) for x in batch["cu_seqlens"][r] if int(x) > 0]
+                    base = r * s
+                    cu.extend(base + x for x in row_cu if base + x > cu[-1])
+                lengths = [b2 - a2 for a2, b2 in zip(cu, cu[1:])]
+            else:
+                lengths = [s] * b
+            psp = PackedSeqParams.from_lengths(lengths, device=batch["tokens"].device)
+            batch = dict(batch)
+            batch["tokens"] = batch["tokens"].reshape(1, b * s)
+            batch["labels"] = batch["labels"].reshape(1, b * s)
+            if "loss_mask" in batch:
+                batch["loss_mask"] = batch["loss_mask"].reshape(1, b * s)
 
         def loss_func(loss_sb):
             if "loss_mask" in batch:
@@ -242,7 +264,7 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                 s, lambda t: not bool(torch.isfinite(t).all()), "nan/inf loss")
             return s, ntok, {"loss_sum": s.detach()}
 
-        out = model(batch["tokens"], labels=batch["labels"])
+        out = model(batch["tokens"], labels=batch["labels"], packed_seq_params=psp)
         return out, loss_func
 
     if forward_step_builder is not None:

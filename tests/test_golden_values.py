@@ -219,3 +219,13 @@ def test_tiny_gpt_fsdp_tracks_ddp_golden():
     losses = _run_config(["--use-fsdp", "--train-iters", "8"])
     for i, (got, want) in enumerate(zip(losses, golden["lm_loss"])):
         assert abs(got - want) < 5e-2, (i, losses, golden["lm_loss"])
+
+
+def test_tiny_gpt_packed_sequences_matches_golden():
+    """--packed-sequences with one document per row is mathematically the
+    unpacked run (block-diagonal attention == per-row attention, RoPE
+    restarts per row) — must land on the plain golden curve."""
+    golden = json.load(open(GOLDEN))
+    losses = _run_config(["--packed-sequences", "--train-iters", "10"])
+    for i, (got, want) in enumerate(zip(losses, golden["lm_loss"])):
+        assert abs(got - want) < 2e-3, (i, losses, golden["lm_loss"])
