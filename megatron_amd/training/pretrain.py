@@ -264,7 +264,10 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                 s, lambda t: not bool(torch.isfinite(t).all()), "nan/inf loss")
             return s, ntok, {"loss_sum": s.detach()}
 
-        out = model(batch["tokens"], labels=batch["labels"], packed_seq_params=psp)
+        if psp is not None:
+            out = model(batch["tokens"], labels=batch["labels"], packed_seq_params=psp)
+        else:
+            out = model(batch["tokens"], labels=batch["labels"])
         return out, loss_func
 
     if forward_step_builder is not None:
