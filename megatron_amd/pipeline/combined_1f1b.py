@@ -216,3 +216,153 @@ def forward_backward_no_pipelining_combined(
         if config.finalize_model_grads_func is not None:
             config.finalize_model_grads_func([model], config)
     return losses, num_tokens_acc
+
+
+def forward_backward_pipelining_without_interleaving_combined(
+    *,
+    forward_step_func,  # unused: the combined path owns batch->loss plumbing
+    data_iterator,
+    model,
+    num_microbatches: int,
+    seq_length: int,
+    micro_batch_size: int,
+    forward_only: bool = False,
+    **kw,
+):
+    """Non-interleaved 1F1B with layer-granular fwd/bwd co-scheduling in the
+    steady state (reference combined_1f1b.py:35): the backward of microbatch
+    j is deferred one slot and interleaved node-by-node with the forward of
+    microbatch j+warmup+1, so MoE a2a on either side overlaps the other's
+    compute.  One extra microbatch of activations stays live vs standard
+    1F1B.  Contract: GPT-family chunks; data_iterator yields
+    {"tokens", "labels"}.
+    """
+    import contextlib
+    from collections import deque
+
+    from megatron_amd.moe.router import AuxLossScaler
+    from megatron_amd.parallel import grid as G
+    from megatron_amd.parallel.cross_entropy import vocab_parallel_cross_entropy
+    from megatron_amd.pipeline.p2p import P2PCommunicator
+
+    if isinstance(model, list):
+        assert len(model) == 1
+        model = model[0]
+    if isinstance(data_iterator, list):
+        data_iterator = data_iterator[0]
+    core = model.module if hasattr(model, "module") else model
+    config = core.config
+    grid = G.get_grid()
+    comm = P2PCommunicator(config, seq_length, micro_batch_size)
+    is_first = grid.is_pipeline_first_stage(ignore_virtual=True)
+    is_last = grid.is_pipeline_last_stage(ignore_virtual=True)
+    pp, pp_rank = grid.pp, grid.pp_rank
+
+    no_sync = config.no_sync_func
+    if no_sync is None and hasattr(model, "no_sync"):
+        no_sync = model.no_sync
+    if no_sync is None:
+        no_sync = contextlib.nullcontext
+
+    losses = []
+    device = next(core.parameters()).device
+    num_tokens_acc = torch.zeros((), dtype=torch.long, device=device)
+
+    def make_plan(batch):
+        tokens = batch["tokens"]
+        seq = tokens.shape[1]
+        freqs = core._rotary_freqs(seq, tokens.device)
+        loss_fn = None
+        if is_last:
+            labels_sb = batch["labels"].transpose(0, 1).contiguous()
+
+            def loss_fn(logits):
+                loss_sb = vocab_parallel_cross_entropy(
+                    logits, labels_sb, label_smoothing=config.label_smoothing)
+                ntok = loss_sb.numel()
+                ssum = loss_sb.sum()
+                losses.append({"loss_sum": ssum.detach()})
+                num_tokens_acc.add_(ntok)
+                scale = 1.0 / (max(int(ntok), 1) * num_microbatches)
+                AuxLossScaler.bind_scale(scale)
+                out = ssum * scale
+                if config.grad_scale_func is not None:
+                    out = config.grad_scale_func(out)
+                return out
+
+        plan = ModelChunkSchedulePlan.from_gpt(core, rotary_freqs=freqs, loss_fn=loss_fn)
+        return plan, tokens
+
+    def _bind_est_scale(out):
+        # Non-last stages never see the token count; estimate from the stage
+        # output shape [s, b, h] (same rule as pipelined._fwd).
+        if not is_last and out is not None and out.dim() >= 2:
+            AuxLossScaler.bind_scale(
+                1.0 / (max(out.shape[0] * out.shape[1], 1) * num_microbatches))
+
+    def fwd_whole(plan, x):
+        for k in range(len(plan)):
+            x = plan.forward_node(k, x)
+        _bind_est_scale(x)
+        return x
+
+    def bwd_whole(plan, g):
+        n = len(plan)
+        for k in range(n):
+            g = plan.backward_node(n - 1 - k, g)
+        return g
+
+    num_warmup = min(pp - pp_rank - 1, num_microbatches)
+    num_steady = num_microbatches - num_warmup
+    plans = deque()
+
+    with no_sync():
+        # ---- warmup forwards ----
+        for _ in range(num_warmup):
+            input_tensor = comm.recv_forward(is_first)
+            plan, tokens = make_plan(next(data_iterator))
+            out = fwd_whole(plan, tokens if is_first else input_tensor)
+            comm.send_forward(out, is_last)
+            if not forward_only:
+                plans.append(plan)
+
+        pending = None  # (plan, grad) whose backward interleaves the next fwd
+        if num_steady > 0:
+            input_tensor = comm.recv_forward(is_first)
+        for i in range(num_steady):
+            plan, tokens = make_plan(next(data_iterator))
+            x = tokens if is_first else input_tensor
+            if forward_only:
+                fwd_whole(plan, x)
+                if i < num_steady - 1:
+                    input_tensor = comm.recv_forward(is_first)
+                continue
+            if pending is None:
+                out = fwd_whole(plan, x)
+            else:
+                bplan, bgrad = pending
+                out, in_grad = combined_1f1b_step(plan, x, bplan, bgrad)
+                _bind_est_scale(out)
+                comm.send_backward(in_grad, is_first)
+            grad = comm.send_forward_recv_backward(out, is_last)
+            plans.append(plan)
+            pending = (plans.popleft(), grad)
+            if i < num_steady - 1:
+                input_tensor = comm.recv_forward(is_first)
+
+        if not forward_only:
+            # ---- cooldown: drain the deferred backward, then the warmup ones
+            if pending is not None:
+                in_grad = bwd_whole(*pending)
+                comm.send_backward(in_grad, is_first)
+            while plans:
+                grad = comm.recv_backward(is_last)
+                in_grad = bwd_whole(plans.popleft(), grad)
+                comm.send_backward(in_grad, is_first)
+
+    if not forward_only:
+        if hasattr(model, "start_grad_sync"):
+            model.start_grad_sync()
+        if config.finalize_model_grads_func is not None:
+            config.finalize_model_grads_func([model], config)
+    return losses, num_tokens_acc

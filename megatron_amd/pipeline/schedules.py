@@ -25,11 +25,18 @@ from megatron_amd.parallel import grid as G
 
 
 def get_forward_backward_func(config=None):
-    if (config is not None and getattr(config, "overlap_moe_expert_parallel_comm", False)
-            and not (G.grid_initialized() and G.get_pipeline_model_parallel_world_size() > 1)):
+    combined = config is not None and getattr(config, "overlap_moe_expert_parallel_comm", False)
+    if combined and not (G.grid_initialized() and G.get_pipeline_model_parallel_world_size() > 1):
         from megatron_amd.pipeline.combined_1f1b import forward_backward_no_pipelining_combined
 
         return forward_backward_no_pipelining_combined
+    if (combined and G.grid_initialized() and G.get_pipeline_model_parallel_world_size() > 1
+            and (G.get_grid().vpp is None or G.get_grid().vpp <= 1)):
+        from megatron_amd.pipeline.combined_1f1b import (
+            forward_backward_pipelining_without_interleaving_combined,
+        )
+
+        return forward_backward_pipelining_without_interleaving_combined
     if G.grid_initialized() and G.get_pipeline_model_parallel_world_size() > 1:
         grid = G.get_grid()
         if grid.vpp is not None and grid.vpp > 1:

@@ -255,3 +255,35 @@ def test_pp2_noninterleaved_ragged_microbatches(tmp_path, monkeypatch):
     pp_losses = json.load(open(out))
     for a, b in zip(ref, pp_losses):
         assert abs(a - b) < 2e-4, (ref, pp_losses)
+
+
+def _cfg_moe(pp=1, combined=False):
+    return TransformerConfig(
+        num_layers=N_LAYERS, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        vocab_size=VOCAB, ffn_hidden_size=128, num_experts=4, moe_router_topk=2,
+        moe_ffn_hidden_size=64, moe_aux_loss_coeff=0.01, pipeline_parallel_size=pp,
+        overlap_moe_expert_parallel_comm=combined, gradient_accumulation_fusion=True,
+    )
+
+
+def _pp2_combined_case(rank, world):
+    G.initialize_model_parallel(pipeline_parallel_size=world)
+    model_parallel_seed(1234)
+    losses = _run(_cfg_moe(pp=world, combined=True), 2, 4, _gen_batches(8))
+    if G.get_grid().is_pipeline_last_stage(ignore_virtual=True):
+        with open(os.environ["PP_TEST_OUT"], "w") as f:
+            json.dump(losses, f)
+
+
+def test_pp2_combined_1f1b_matches_single(tmp_path, monkeypatch):
+    """MoE model under the combined (layer-interleaved fwd/bwd) pp=2 schedule
+    trains identically to a single-rank standard run (reference
+    combined_1f1b.py steady-state co-schedule)."""
+    out = tmp_path / "ppc.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+    init_single()
+    ref = _run(_cfg_moe(), 2, 4, _gen_batches(8))
+    spawn_dist(_pp2_combined_case, 2)
+    pp_losses = json.load(open(out))
+    for a, b in zip(ref, pp_losses):
+        assert abs(a - b) < 2e-4, (ref, pp_losses)
