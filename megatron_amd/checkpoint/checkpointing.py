@@ -29,7 +29,10 @@ def save_checkpoint(root: str, models: List, optimizer, iteration: int,
     sharded = {}
     param_maps = []
     for i, m in enumerate(models):
-        sd, pm = model_sharded_state_dict(m, prefix=f"model{i}." if len(models) > 1 else "model.")
+        # One canonical namespace regardless of VPP chunk count: layer keys are
+        # globally numbered, so chunks never collide and a checkpoint written
+        # at any (pp, vpp) layout loads at any other.
+        sd, pm = model_sharded_state_dict(m, prefix="model.")
         sharded.update(sd)
         param_maps.append(pm)
     if optimizer is not None:
@@ -119,7 +122,10 @@ def load_checkpoint(root: str, models: List, optimizer, scheduler=None,
     sharded = {}
     param_maps = []
     for i, m in enumerate(models):
-        sd, pm = model_sharded_state_dict(m, prefix=f"model{i}." if len(models) > 1 else "model.")
+        # One canonical namespace regardless of VPP chunk count: layer keys are
+        # globally numbered, so chunks never collide and a checkpoint written
+        # at any (pp, vpp) layout loads at any other.
+        sd, pm = model_sharded_state_dict(m, prefix="model.")
         sharded.update(sd)
         param_maps.append(pm)
     if optimizer is not None:

@@ -76,9 +76,11 @@ class GroupedMLP(nn.Module):
             outs = []
             start = 0
             for e, n in enumerate(splits):
-                if n == 0:
-                    outs.append(tokens.new_zeros(0, self.config.hidden_size))
-                    continue
+                # n == 0 still goes through the matmuls: an empty [0, h] slice
+                # keeps the autograd graph connected, so the EP a2a backward
+                # runs on EVERY rank even when this rank's experts got no
+                # tokens this microbatch (skipping it desyncs the per-pair
+                # message order of the EP communicator).
                 x = tokens[start : start + n]
                 start += n
                 h = torch.matmul(x, self.weight1[e].t())
