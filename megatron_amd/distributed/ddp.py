@@ -75,6 +75,7 @@ class ParamAndGradBuffer:
         # expert grads get the same effective scale as dense grads (reference
         # expert_gradient_scaling_factor = edp/dp_cp pre-scale + AVG over edp).
         self.grad_scale_denom = grad_scale_denom or self.dp_size
+        self.param_names: List[tuple] = []  # [(name, param)], set by the DDP wrapper
         self.grad_dtype = grad_dtype
         self.param_dtype = param_dtype
         self.device = device
@@ -143,6 +144,16 @@ class ParamAndGradBuffer:
     # -- per-bucket collectives ---------------------------------------------
 
     def _launch_grad_reduce(self, bucket: _Bucket, async_op: bool):
+        if self.ddp_config.check_for_nan_in_grad:
+            # reference check_for_nan_in_grad: fail fast, per bucket, BEFORE
+            # the collective so the faulty rank is identifiable
+            if not torch.isfinite(bucket.grad_view.sum()):
+                bad = [n for n, p in self.param_names
+                       if any(p is q for q in bucket.params)]
+                raise RuntimeError(
+                    f"NaN/Inf grad in bucket {bucket.index} before reduce "
+                    f"(rank {dist.get_rank() if dist.is_initialized() else 0}); "
+                    f"params: {bad[:8]}")
         denom = self.grad_scale_denom
         if self.dp_size == 1:
             # no collective, but the average denominator may still exceed 1
@@ -259,6 +270,10 @@ class DistributedDataParallel(nn.Module):
             self.buffers.append(ebuf)
             for p in ebuf.param_index:
                 self.param_to_buffer[p] = ebuf
+
+        name_of = {p: n for n, p in self.module.named_parameters()}
+        for buf in self.buffers:
+            buf.param_names = [(name_of.get(p, "?"), p) for p in buf.param_index]
 
         self._grad_hook_handles = []
         for p in self.param_to_buffer:

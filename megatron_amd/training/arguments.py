@@ -162,6 +162,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--accumulate-allreduce-grads-in-fp32", dest="grad_reduce_in_fp32",
                    action="store_true", default=True)
     g.add_argument("--bucket-size", type=int, default=40_000_000)
+    g.add_argument("--check-for-nan-in-loss-and-grad", dest="check_for_nan_in_grad",
+                   action="store_true", default=False)
 
     g = p.add_argument_group("checkpointing")
     g.add_argument("--save", type=str, default=None)
@@ -336,6 +338,7 @@ def configs_from_args(args):
         overlap_grad_reduce=args.overlap_grad_reduce,
         use_distributed_optimizer=args.use_distributed_optimizer,
         bucket_size=args.bucket_size,
+        check_for_nan_in_grad=args.check_for_nan_in_grad,
         use_rccl_registered_buffers=args.use_rccl_registered_buffers,
     )
     return cfg, opt_cfg, ddp_cfg

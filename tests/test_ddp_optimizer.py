@@ -248,3 +248,36 @@ def _ep_grad_norm_case(rank, world):
 
 def test_grad_norm_includes_all_ep_ranks():
     spawn_dist(_ep_grad_norm_case, 2)
+
+
+def test_check_for_nan_in_grad_names_the_bucket():
+    """check_for_nan_in_grad fails fast before the reduce and names params
+    in the offending bucket (reference DistributedDataParallelConfig)."""
+    import pytest
+
+    from megatron_amd.config import DDPConfig, TransformerConfig
+    from megatron_amd.distributed.ddp import DistributedDataParallel
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from tests.utils import init_single
+
+    init_single()
+    model_parallel_seed(3)
+    cfg = TransformerConfig(num_layers=2, hidden_size=32, num_attention_heads=4,
+                            num_query_groups=4, vocab_size=64, ffn_hidden_size=48,
+                            gradient_accumulation_fusion=True)
+    m = DistributedDataParallel(cfg,
+                                DDPConfig(check_for_nan_in_grad=True,
+                                          overlap_grad_reduce=False,
+                                          bucket_size=5_000), GPTModel(cfg))
+    tokens = torch.randint(0, 64, (2, 8))
+    m.zero_grad_buffer()
+    m(tokens, labels=tokens).sum().backward()
+    m.finish_grad_sync()  # clean grads pass
+
+    m.zero_grad_buffer()
+    m(tokens, labels=tokens).sum().backward()
+    p = m.module.output_layer.weight
+    p.main_grad.view(-1)[0] = float("nan")
+    with pytest.raises(RuntimeError, match="NaN/Inf grad .*bucket"):
+        m.finish_grad_sync()
