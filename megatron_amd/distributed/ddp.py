@@ -337,16 +337,21 @@ class DistributedDataParallel(nn.Module):
             buf.zero_grad()
 
     def broadcast_params(self):
-        """Sync initial params across DP (rank0 -> all)."""
-        if self.dp_group is None or dist.get_world_size(group=self.dp_group) == 1:
+        """Sync initial params across each buffer's OWN replica group: dense
+        over dp, expert over edp.  Broadcasting experts over the dense dp
+        group would overwrite different EP ranks' distinct experts."""
+        if self.dp_group is None or not dist.is_initialized():
             return
-        src = dist.get_process_group_ranks(self.dp_group)[0]
         for buf in self.buffers:
+            group = buf.dp_group
+            if group is None or dist.get_world_size(group=group) == 1:
+                continue
+            src = dist.get_process_group_ranks(group)[0]
             if buf.param_data is not None:
-                dist.broadcast(buf.param_data, src=src, group=self.dp_group)
+                dist.broadcast(buf.param_data, src=src, group=group)
             else:
                 for p in buf.param_index:
-                    dist.broadcast(p.data, src=src, group=self.dp_group)
+                    dist.broadcast(p.data, src=src, group=group)
 
     def state_dict(self, *args, **kwargs):
         return self.module.state_dict(*args, **kwargs)

@@ -52,7 +52,7 @@ class _FSDPUnit:
     """One sharding unit: a module whose params share a flat buffer."""
 
     def __init__(self, name: str, module: nn.Module, params: List[nn.Parameter],
-                 dp_group, reshard_after_forward: bool):
+                 dp_group, reshard_after_forward: bool, grad_scale_denom: int = 0):
         self.name = name
         self.module = module
         self.params = params
@@ -62,6 +62,10 @@ class _FSDPUnit:
             self.dp_rank = dist.get_rank(dp_group)
         else:
             self.dp_size, self.dp_rank = 1, 0
+        # expert units reduce-scatter over edp but must average over dp_cp so
+        # expert grads carry the same per-token scale as dense grads (same
+        # rule as ParamAndGradBuffer.grad_scale_denom)
+        self.grad_scale_denom = grad_scale_denom or self.dp_size
         self.reshard_after_forward = reshard_after_forward
 
         self.dtype = params[0].dtype
@@ -79,10 +83,14 @@ class _FSDPUnit:
             self.offsets.append(off)
             off += n
 
-        # build the flat buffer once, copy weights in, keep only our shard
+        # build the flat buffer once, sync to the group's rank-0 weights
+        # (per-rank random init must agree before sharding), keep our shard
         flat = torch.zeros(self.flat_size, dtype=self.dtype, device=device)
         for p, o in zip(params, self.offsets):
             flat[o : o + p.numel()].copy_(p.detach().reshape(-1))
+        if dist.is_initialized() and self.dp_size > 1:
+            src = dist.get_process_group_ranks(dp_group)[0] if dp_group is not None else 0
+            dist.broadcast(flat, src=src, group=dp_group)
         shard_view = flat[self.dp_rank * self.shard_size : (self.dp_rank + 1) * self.shard_size]
         self.model_shard = shard_view.clone()  # model-dtype shard (AG source)
         # fp32 master shard exposed to the optimizer
@@ -155,7 +163,7 @@ class _FSDPUnit:
         # pack grads into a flat fp32 buffer, pre-scaled for DP averaging
         flat_grad = torch.zeros(self.flat_size, dtype=torch.float32,
                                 device=self.model_shard.device)
-        inv = 1.0 / self.dp_size
+        inv = 1.0 / self.grad_scale_denom
         for p, o, n in zip(self.params, self.offsets, self.numels):
             if p.grad is not None:
                 flat_grad[o : o + n].copy_(p.grad.reshape(-1)).mul_(inv)
@@ -190,15 +198,15 @@ class FullyShardedDataParallel(nn.Module):
                  reshard_after_forward: bool = False):
         super().__init__()
         self.module = module
-        # expert-parallel params are already sharded over EP with their own
-        # (edp) replication group; flat-sharding them over the dense DP group
-        # would double-shard and mis-reduce — explicit unsupported combo
-        for n, p in module.named_parameters():
-            assert not getattr(p, "is_expert_parallel", False), (
-                f"FSDP over expert-parallel params ({n}) is unsupported; "
-                "use EP + the distributed optimizer instead")
         self.dp_group = (dp_group if dp_group is not None
                          else (G.get_data_parallel_group() if dist.is_initialized() else None))
+        # expert-parallel params are sharded over EP already; their FSDP flat
+        # shard lives on the expert-data-parallel (edp) replica group, with
+        # grads averaged over dp_cp like the dense ones
+        grid_ok = dist.is_initialized() and G.grid_initialized()
+        self.edp_group = G.get_grid().group("expert_dp") if grid_ok else None
+        dp_cp = (dist.get_world_size(self.dp_group)
+                 if (self.dp_group is not None and dist.is_initialized()) else 1)
         self.units: List[_FSDPUnit] = []
         self._param_to_unit: Dict[nn.Parameter, _FSDPUnit] = {}
 
@@ -210,10 +218,19 @@ class FullyShardedDataParallel(nn.Module):
             if not params:
                 return
             seen_params.update(id(p) for p in params)
-            u = _FSDPUnit(name, mod, params, self.dp_group, reshard_after_forward)
-            self.units.append(u)
-            for p in params:
-                self._param_to_unit[p] = u
+            dense = [p for p in params if not getattr(p, "is_expert_parallel", False)]
+            expert = [p for p in params if getattr(p, "is_expert_parallel", False)]
+            if dense:
+                u = _FSDPUnit(name, mod, dense, self.dp_group, reshard_after_forward)
+                self.units.append(u)
+                for p in dense:
+                    self._param_to_unit[p] = u
+            if expert:
+                u = _FSDPUnit(name + ".experts", mod, expert, self.edp_group,
+                              reshard_after_forward, grad_scale_denom=dp_cp)
+                self.units.append(u)
+                for p in expert:
+                    self._param_to_unit[p] = u
 
         for name, mod in module.named_modules():
             is_unit = isinstance(mod, unit_classes) if unit_classes else False
@@ -230,26 +247,8 @@ class FullyShardedDataParallel(nn.Module):
         make_unit("root", module, [p for p in module.parameters() if id(p) not in seen_params])
 
         self._register_hooks()
-        self._sync_shards_from_rank0()
 
     # ---- setup ----------------------------------------------------------
-
-    def _sync_shards_from_rank0(self):
-        """Ranks may have different random init; shard content must agree on
-        the weights rank 0 holds.  Broadcast each rank's shard source."""
-        if not dist.is_initialized():
-            return
-        for u in self.units:
-            # rebuild rank-0's flat buffer on every rank
-            flat = torch.empty(u.flat_size, dtype=u.dtype, device=u.model_shard.device)
-            dist.all_gather_into_tensor(flat, u.model_shard, group=self.dp_group)
-            if self.dp_group is None or self.dp_group is dist.group.WORLD:
-                src = 0
-            else:
-                src = dist.get_global_rank(self.dp_group, 0)
-            dist.broadcast(flat, src=src, group=self.dp_group)
-            u.model_shard.copy_(flat[u.dp_rank * u.shard_size : (u.dp_rank + 1) * u.shard_size])
-            u.master_shard.data.copy_(u.model_shard.float())
 
     def _register_hooks(self):
         for u in self.units:

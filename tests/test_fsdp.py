@@ -165,3 +165,75 @@ def _prefetch_case(rank, world):
 
 def test_fsdp_ag_prefetch_pipeline():
     spawn_dist(_prefetch_case, 2)
+
+
+def _ep_worker(rank, world):
+    """FSDP over a MoE model at EP=2 x world 4 (dp=4, edp=2): dense params
+    flat-shard over dp, expert params over edp with dp_cp grad averaging.
+    Reference twin: plain per-rank replicas with manual group all-reduces."""
+    from megatron_amd.parallel import grid as G
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    G.destroy_model_parallel()
+    G.initialize_model_parallel(expert_parallel_size=2)
+
+    def build():
+        model_parallel_seed(1234)
+        torch.manual_seed(50 + G.get_grid().rank_in("ep"))  # per-EP-slice init
+        return GPTModel(_cfg(num_experts=4, moe_router_topk=2, moe_ffn_hidden_size=32,
+                             expert_parallel_size=2, gradient_accumulation_fusion=False))
+
+    model = build()
+    ref = build()
+
+    fsdp = FullyShardedDataParallel(model)
+    assert any(u.name.endswith(".experts") for u in fsdp.units)
+    opt = torch.optim.AdamW(fsdp.shard_parameters(), lr=1e-2)
+
+    # the FSDP wrapper synced params from each unit-group's rank 0; mirror
+    # that on the reference replicas
+    grid = G.get_grid()
+    dp_group, edp_group = grid.group("dp"), grid.group("expert_dp")
+    for n, p in ref.named_parameters():
+        if getattr(p, "is_expert_parallel", False):
+            dist.broadcast(p.data, src=dist.get_process_group_ranks(edp_group)[0],
+                           group=edp_group)
+        else:
+            dist.broadcast(p.data, src=dist.get_process_group_ranks(dp_group)[0],
+                           group=dp_group)
+    ref_opt = torch.optim.AdamW(ref.parameters(), lr=1e-2)
+
+    dp_cp = dist.get_world_size(dp_group)
+    for step in range(3):
+        g = torch.Generator().manual_seed(11 * step)
+        ids = torch.randint(0, 64, (world * 2, 16), generator=g)[rank * 2 : rank * 2 + 2]
+
+        fsdp.zero_grad_buffer()
+        loss = fsdp(input_ids=ids, labels=ids).mean()
+        loss.backward()
+        opt.step()
+        fsdp.update_model_shards()
+
+        ref_opt.zero_grad()
+        ref_loss = ref(input_ids=ids, labels=ids).mean()
+        ref_loss.backward()
+        for p in ref.parameters():
+            if p.grad is None:
+                p.grad = torch.zeros_like(p)
+            if getattr(p, "is_expert_parallel", False):
+                dist.all_reduce(p.grad, group=edp_group)
+                p.grad /= dp_cp
+            else:
+                dist.all_reduce(p.grad, group=dp_group)
+                p.grad /= dp_cp
+        ref_opt.step()
+        assert_close(loss.detach(), ref_loss.detach(), rtol=1e-5, atol=1e-6,
+                     msg=f"step {step}")
+
+    full = fsdp.state_dict()
+    for n, p in ref.named_parameters():
+        assert_close(full[n], p.detach(), rtol=1e-4, atol=1e-5, msg=n)
+
+
+def test_fsdp_expert_parallel_ep2():
+    spawn_dist(_ep_worker, world_size=4)
