@@ -139,7 +139,7 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
         from megatron_amd.training.training import setup_fsdp_model_and_optimizer
 
         assert cfg.pipeline_parallel_size == 1 and cfg.tensor_parallel_size == 1, (
-            "--use-fsdp composes with DP only (TP/PP use DDP + ZeRO-1)")
+            "--use-fsdp composes with DP and EP (TP/PP use DDP + ZeRO-1)")
         chunks, optimizer = setup_fsdp_model_and_optimizer(model_provider, cfg, opt_cfg, device=device)
     else:
         chunks, optimizer = setup_model_and_optimizer(model_provider, cfg, opt_cfg, ddp_cfg, device=device)
