@@ -70,7 +70,9 @@ class TransformerConfig(ParallelConfig):
     add_linear_bias: bool = False
     add_qkv_bias: bool = False  # bias only on the fused QKV projection (Qwen2-style)  # llama-style: no bias anywhere
     untie_embeddings_and_output_weights: bool = True
-    position_embedding_type: str = "rope"  # 'rope' | 'learned' | 'none'
+    position_embedding_type: str = "rope"  # 'rope' | 'learned' | 'relative' | 'none'
+    relative_attention_num_buckets: int = 32   # T5 relative-position bias
+    relative_attention_max_distance: int = 128
     rotary_base: float = 500000.0  # llama-3 default
     rotary_percent: float = 1.0
     # None or {'type': 'llama3'|'linear', 'factor': ..., 'low_freq_factor': ...,

@@ -211,6 +211,7 @@ def attention(
     window: Optional[int] = None,
     dropout_p: float = 0.0,
     training: bool = False,
+    bias: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Unfused reference attention.
 
@@ -229,6 +230,8 @@ def attention(
         kf = kf.repeat_interleave(rep, dim=1)
         vf = vf.repeat_interleave(rep, dim=1)
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b, hq, s, s]
+    if bias is not None:
+        scores = scores + bias.float()  # additive [hq, s, sk] (T5 relative bias)
     sk = k.shape[0]
     off = sk - s  # decode: query row i attends kv <= i + off
     if causal:

@@ -59,7 +59,8 @@ class SelfAttention(nn.Module):
             self.k_layernorm = None
 
     def forward(self, hidden_states: torch.Tensor, rotary_freqs: Optional[torch.Tensor] = None,
-                attention_mask=None, inference_context=None, packed_seq_params=None) -> torch.Tensor:
+                attention_mask=None, inference_context=None, packed_seq_params=None,
+                attention_bias=None) -> torch.Tensor:
         # hidden_states: [s(/tp if SP), b, h]
         qkv, _ = self.linear_qkv(hidden_states)  # [s, b, qkv_size/tp]
         s, b = qkv.shape[0], qkv.shape[1]
@@ -98,6 +99,15 @@ class SelfAttention(nn.Module):
             core_out = inference_context.attend(
                 self.layer_number, q, k, v, self.softmax_scale, self.window
             )
+        elif attention_bias is not None:
+            # additive score bias [hq_local, sq, sk] (T5 relative position
+            # bias): unfused path — the bias re-materializes per step, so
+            # the flash kernel's fused softmax does not apply
+            from megatron_amd.ops import reference as _ref
+
+            core_out = _ref.attention(
+                q, k, v, causal=self.config.causal_attention,
+                scale=self.softmax_scale, bias=attention_bias)
         elif attention_mask is not None:
             # key-padding mask ([b, s] bool, True = valid): the arbitrary-
             # mask path (BERT-style padded batches) — torch composition,

@@ -92,11 +92,12 @@ class TransformerLayer(nn.Module):
             self.mlp = _make_mixer(config, layer_number)
         self.hidden_dropout = config.hidden_dropout
 
-    def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None, packed_seq_params=None):
+    def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None, packed_seq_params=None, attention_bias=None):
         residual = hidden_states
         x = self.input_layernorm(hidden_states)
         x = self.self_attention(x, rotary_freqs=rotary_freqs, attention_mask=attention_mask,
-                                inference_context=inference_context, packed_seq_params=packed_seq_params)
+                                inference_context=inference_context, packed_seq_params=packed_seq_params,
+                                attention_bias=attention_bias)
         x = ops.bias_dropout_add(x, None, residual, self.hidden_dropout, self.training)
         residual = x
         y = self.pre_mlp_layernorm(x)
@@ -162,7 +163,7 @@ class TransformerBlock(nn.Module):
 
         return rng_checkpoint(run, self.config.distribute_saved_activations, *args)
 
-    def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None, packed_seq_params=None):
+    def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None, packed_seq_params=None, attention_bias=None):
         recompute = (
             self.config.recompute_granularity == "full" and self.training and inference_context is None
         )
@@ -182,9 +183,10 @@ class TransformerBlock(nn.Module):
                     if g is not None:
                         hidden_states = g(hidden_states)
                         continue
+                kw = {"attention_bias": attention_bias} if attention_bias is not None else {}
                 hidden_states = layer(
                     hidden_states, rotary_freqs=rotary_freqs, attention_mask=attention_mask,
-                    inference_context=inference_context, packed_seq_params=packed_seq_params,
+                    inference_context=inference_context, packed_seq_params=packed_seq_params, **kw,
                 )
         if self.final_layernorm is not None:
             hidden_states = self.final_layernorm(hidden_states)
