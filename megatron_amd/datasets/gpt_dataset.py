@@ -24,14 +24,37 @@ from megatron_amd.datasets.helpers import build_blending_indices, build_sample_i
 from megatron_amd.datasets.indexed import IndexedDataset
 
 
+def eod_boundaries(tokens: torch.Tensor, eod: int, seq_length: int,
+                   max_docs: int = 64) -> torch.Tensor:
+    """Document end offsets within one sample (cumulative, ending at
+    seq_length), zero-padded to a fixed width so samples collate.  The
+    training loop turns these into varlen cu_seqlens -> block-diagonal
+    attention + per-document position restart (reference
+    get_ltor_masks_and_position_ids, gpt_dataset.py) — mapped onto the
+    flash-varlen path instead of a materialized [s, s] mask."""
+    ends = (tokens == eod).nonzero(as_tuple=True)[0] + 1
+    cu = ends[ends < seq_length].tolist() + [seq_length]
+    cu = cu[: max_docs]
+    if cu[-1] != seq_length:
+        cu[-1] = seq_length
+    out = torch.zeros(max_docs, dtype=torch.long)
+    out[: len(cu)] = torch.tensor(cu, dtype=torch.long)
+    return out
+
+
 class GPTDataset(torch.utils.data.Dataset):
     def __init__(self, indexed: IndexedDataset, num_samples: Optional[int],
                  seq_length: int, seed: int = 1234, cache_dir: Optional[str] = None,
-                 document_subset: Optional[np.ndarray] = None, name: str = "train"):
+                 document_subset: Optional[np.ndarray] = None, name: str = "train",
+                 eod: Optional[int] = None, reset_attention_mask: bool = False,
+                 eod_mask_loss: bool = False):
         self.indexed = indexed
         self.seq_length = seq_length
         self.seed = seed
         self.name = name
+        self.eod = eod
+        self.reset_attention_mask = reset_attention_mask
+        self.eod_mask_loss = eod_mask_loss
         if document_subset is None:
             document_subset = np.arange(len(indexed.document_indices) - 1, dtype=np.int64)
         self.documents = document_subset
@@ -98,11 +121,17 @@ class GPTDataset(torch.utils.data.Dataset):
         tokens = self._sample_tokens(int(self.shuffle_idx[idx % len(self.shuffle_idx)]))
         tokens = torch.from_numpy(tokens.astype(np.int64))
         assert tokens.numel() == self.seq_length + 1
-        return {
+        out = {
             "tokens": tokens[:-1],
             "labels": tokens[1:],
             "loss_mask": torch.ones(self.seq_length, dtype=torch.float32),
         }
+        if self.eod is not None:
+            if self.eod_mask_loss:
+                out["loss_mask"][out["labels"] == self.eod] = 0.0
+            if self.reset_attention_mask:
+                out["cu_seqlens"] = eod_boundaries(out["tokens"], self.eod, self.seq_length)
+        return out
 
 
 class BlendedDataset(torch.utils.data.Dataset):
@@ -134,7 +163,8 @@ def _parse_split(split: str) -> np.ndarray:
 
 def build_gpt_datasets(data_paths: Sequence, seq_length: int, seed: int,
                        train_samples: int, split: str = "969,30,1",
-                       cache_dir: Optional[str] = None):
+                       cache_dir: Optional[str] = None, eod: Optional[int] = None,
+                       reset_attention_mask: bool = False, eod_mask_loss: bool = False):
     """data_paths: [prefix] or [w1, prefix1, w2, prefix2, ...]. Returns
     (train, valid, test) datasets; splits partition each corpus by document."""
     if len(data_paths) == 1:
@@ -165,7 +195,9 @@ def build_gpt_datasets(data_paths: Sequence, seq_length: int, seed: int,
             # blended builder uses the same 0.5% pad)
             want = count if len(prefixes) == 1 else int(wnorm[pi] * count * 1.005) + 1
             subsets.append(GPTDataset(indexed, want, seq_length, seed,
-                                      cache_dir, docs, name=names[si]))
+                                      cache_dir, docs, name=names[si], eod=eod,
+                                      reset_attention_mask=reset_attention_mask,
+                                      eod_mask_loss=eod_mask_loss))
         if not subsets:
             out.append(None)
         elif len(subsets) == 1:
@@ -181,7 +213,10 @@ def build_gpt_train_iterator(args, device, dp_rank: int, dp_size: int,
     ``start_sample`` resumes mid-epoch after a checkpoint load."""
     train_samples = args.train_iters * args.global_batch_size
     ds = build_gpt_datasets(args.data_path, args.seq_length, args.seed,
-                            train_samples, args.split)
+                            train_samples, args.split,
+                            eod=getattr(args, "eod_id", None),
+                            reset_attention_mask=getattr(args, "reset_attention_mask", False),
+                            eod_mask_loss=getattr(args, "eod_mask_loss", False))
     train = {"train": ds[0], "valid": ds[1] or ds[0], "test": ds[2] or ds[0]}[split]
     sampler = _ShardedSequentialSampler(len(train), args.micro_batch_size, dp_rank, dp_size,
                                         start_sample=start_sample)

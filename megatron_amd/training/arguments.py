@@ -116,6 +116,15 @@ def build_arg_parser() -> argparse.ArgumentParser:
                    metavar=("START", "INCREMENT", "SAMPLES"),
                    help="grow GBS from START by INCREMENT over SAMPLES consumed samples")
     g.add_argument("--seq-length", type=int, default=4096)
+    g.add_argument("--reset-attention-mask", action="store_true", default=False,
+                   help="document-boundary (EOD) attention reset: samples carry "
+                        "cu_seqlens and train block-diagonal via the varlen path")
+    g.add_argument("--reset-position-ids", action="store_true", default=False,
+                   help="restart positions at each EOD (implied by "
+                        "--reset-attention-mask on the varlen path)")
+    g.add_argument("--eod-mask-loss", action="store_true", default=False)
+    g.add_argument("--eod-id", type=int, default=None,
+                   help="EOD token id for the reset/mask flags (default: tokenizer eod)")
     g.add_argument("--packed-sequences", action="store_true",
                    help="THD training: microbatch rows flattened into one packed stream "
                         "(block-diagonal attention, per-document RoPE restart)")

@@ -229,7 +229,7 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
 
             batch = get_batch_on_this_cp_rank(batch, mode=cfg.cp_comm_type)
         psp = None
-        if getattr(args, "packed_sequences", False):
+        if getattr(args, "packed_sequences", False) or getattr(args, "reset_attention_mask", False):
             # THD training: flatten the microbatch into one packed stream;
             # per-row cu_seqlens (or one doc per row) shift to global offsets
             from megatron_amd.transformer.packed_seq import PackedSeqParams

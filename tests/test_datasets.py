@@ -266,3 +266,38 @@ def test_apply_chat_template_generic():
     ids = apply_chat_template(tok, [{"role": "user", "content": "hi"}])
     text = tok.detokenize(ids)
     assert "<|user|>" in text and "<|assistant|>" in text and "hi" in text
+
+
+def test_gpt_dataset_eod_boundaries_and_loss_mask(tmp_path):
+    """--reset-attention-mask / --eod-mask-loss: samples carry cu_seqlens at
+    EOD boundaries and zero loss on EOD labels (reference
+    get_ltor_masks_and_position_ids mapped to the varlen path)."""
+    import numpy as np
+
+    from megatron_amd.datasets.gpt_dataset import GPTDataset, eod_boundaries
+    from megatron_amd.datasets.indexed import IndexedDatasetBuilder, IndexedDataset
+
+    eod = 0
+    prefix = str(tmp_path / "corpus")
+    b = IndexedDatasetBuilder(prefix, dtype=np.int32)
+    rng = np.random.RandomState(3)
+    for _ in range(32):
+        doc = rng.randint(1, 50, size=rng.randint(4, 12)).tolist() + [eod]
+        b.add_document(doc)
+    b.finalize()
+
+    ds = GPTDataset(IndexedDataset(prefix), 8, 16, seed=5, eod=eod,
+                    reset_attention_mask=True, eod_mask_loss=True)
+    s = ds[0]
+    assert "cu_seqlens" in s
+    cu = [int(x) for x in s["cu_seqlens"] if int(x) > 0]
+    assert cu[-1] == 16 and cu == sorted(cu)
+    # every non-final boundary sits right after an EOD token
+    for c in cu[:-1]:
+        assert int(s["tokens"][c - 1]) == eod
+    # loss masked exactly where the label is EOD
+    assert torch.equal(s["loss_mask"] == 0.0, s["labels"] == eod)
+
+    t = torch.tensor([5, 0, 7, 8, 0, 9, 9, 9])
+    cu = eod_boundaries(t, 0, 8, max_docs=4)
+    assert [int(x) for x in cu if x > 0] == [2, 5, 8]
