@@ -144,6 +144,97 @@ class TiktokenTokenizer(MegatronTokenizer):
         return self._enc.eot_token
 
 
+class GPT2BPETokenizer(MegatronTokenizer):
+    """Self-contained byte-level BPE over local vocab.json + merges.txt
+    (reference megatron/training/tokenizer GPT2BPETokenizer role): no
+    network, no HF dependency.  Byte-level means any input round-trips:
+    raw bytes map to 256 printable unicode marks, BPE merges apply on top."""
+
+    def __init__(self, vocab_file: str, merges_file: str,
+                 special_tokens: Optional[List[str]] = None):
+        import json
+
+        import regex
+
+        with open(vocab_file, encoding="utf-8") as f:
+            self.encoder = json.load(f)
+        self.decoder = {v: k for k, v in self.encoder.items()}
+        with open(merges_file, encoding="utf-8") as f:
+            merges = [tuple(line.split()) for line in f.read().split("\n")
+                      if line and not line.startswith("#") and len(line.split()) == 2]
+        self.bpe_ranks = {m: i for i, m in enumerate(merges)}
+        self.byte_encoder = self._bytes_to_unicode()
+        self.byte_decoder = {c: b for b, c in self.byte_encoder.items()}
+        # the GPT-2 pre-tokenization pattern (contractions, letter/number
+        # runs, punctuation runs, whitespace)
+        self.pat = regex.compile(
+            r"'s|'t|'re|'ve|'m|'ll|'d| ?\p{L}+| ?\p{N}+| ?[^\s\p{L}\p{N}]+|\s+(?!\S)|\s+")
+        self._cache: dict = {}
+        self._eod = self.encoder.get("<|endoftext|>", len(self.encoder) - 1)
+        for tok in special_tokens or []:
+            if tok not in self.encoder:
+                self.encoder[tok] = len(self.encoder)
+                self.decoder[self.encoder[tok]] = tok
+
+    @staticmethod
+    def _bytes_to_unicode():
+        # printable ranges map to themselves; everything else is shifted into
+        # the 256+ private block so merges files stay printable
+        bs = (list(range(ord("!"), ord("~") + 1)) + list(range(0xA1, 0xAD))
+              + list(range(0xAE, 0x100)))
+        cs = bs[:]
+        n = 0
+        for b in range(256):
+            if b not in bs:
+                bs.append(b)
+                cs.append(256 + n)
+                n += 1
+        return dict(zip(bs, [chr(c) for c in cs]))
+
+    def _bpe(self, token: str) -> List[str]:
+        if token in self._cache:
+            return self._cache[token]
+        word = list(token)
+        while len(word) > 1:
+            pairs = {(word[i], word[i + 1]) for i in range(len(word) - 1)}
+            best = min(pairs, key=lambda p: self.bpe_ranks.get(p, float("inf")))
+            if best not in self.bpe_ranks:
+                break
+            a, b = best
+            out = []
+            i = 0
+            while i < len(word):
+                if i < len(word) - 1 and word[i] == a and word[i + 1] == b:
+                    out.append(a + b)
+                    i += 2
+                else:
+                    out.append(word[i])
+                    i += 1
+            word = out
+        self._cache[token] = word
+        return word
+
+    def tokenize(self, text: str) -> List[int]:
+        ids: List[int] = []
+        for chunk in self.pat.findall(text):
+            mapped = "".join(self.byte_encoder[b] for b in chunk.encode("utf-8"))
+            ids.extend(self.encoder[p] for p in self._bpe(mapped))
+        return ids
+
+    def detokenize(self, ids) -> str:
+        text = "".join(self.decoder[int(i)] for i in ids)
+        raw = bytes(self.byte_decoder[c] for c in text)
+        return raw.decode("utf-8", errors="replace")
+
+    @property
+    def vocab_size(self) -> int:
+        return len(self.encoder)
+
+    @property
+    def eod(self) -> int:
+        return self._eod
+
+
 def build_tokenizer(tokenizer_type: str, tokenizer_model: Optional[str] = None,
                     vocab_size: Optional[int] = None) -> MegatronTokenizer:
     t = tokenizer_type.lower()
@@ -159,14 +250,11 @@ def build_tokenizer(tokenizer_type: str, tokenizer_model: Optional[str] = None,
         return SentencePieceTokenizer(tokenizer_model)
     if t in ("tiktokentokenizer", "tiktoken"):
         return TiktokenTokenizer(tokenizer_model or "cl100k_base")
+    if t in ("gpt2bpetokenizer", "gpt2", "gpt2-bpe"):
+        assert tokenizer_model, "--tokenizer-model VOCAB,MERGES required for GPT2 BPE"
+        vocab, merges = tokenizer_model.split(",")
+        return GPT2BPETokenizer(vocab, merges)
     raise ValueError(f"unknown tokenizer type {tokenizer_type}")
-
-
-def pad_vocab_size(vocab_size: int, multiple: int, tp_size: int) -> int:
-    """Pad to a multiple of (multiple * tp) for clean TP sharding
-    (reference training/tokenizer _vocab_size_with_padding)."""
-    m = multiple * tp_size
-    return ((vocab_size + m - 1) // m) * m
 
 
 # ---------------------------------------------------------------------------

@@ -301,3 +301,42 @@ def test_gpt_dataset_eod_boundaries_and_loss_mask(tmp_path):
     t = torch.tensor([5, 0, 7, 8, 0, 9, 9, 9])
     cu = eod_boundaries(t, 0, 8, max_docs=4)
     assert [int(x) for x in cu if x > 0] == [2, 5, 8]
+
+
+def test_gpt2_bpe_tokenizer(tmp_path):
+    """Self-contained byte-level BPE: merge order follows ranks, any unicode
+    round-trips through the byte alphabet, special tokens append."""
+    import json
+
+    from megatron_amd.tokenizers import GPT2BPETokenizer, build_tokenizer
+
+    byte_enc = GPT2BPETokenizer._bytes_to_unicode()
+    vocab = {c: i for i, c in enumerate(byte_enc.values())}
+    for tok in ["he", "ll", "hell", "hello", "<|endoftext|>"]:
+        vocab[tok] = len(vocab)
+    vf, mf = str(tmp_path / "vocab.json"), str(tmp_path / "merges.txt")
+    json.dump(vocab, open(vf, "w"))
+    open(mf, "w").write("#version\nh e\nl l\nhe ll\nhell o\n")
+
+    tok = GPT2BPETokenizer(vf, mf)
+    ids = tok.tokenize("hello")
+    assert [tok.decoder[i] for i in ids] == ["hello"]
+    ids = tok.tokenize("hell")
+    assert [tok.decoder[i] for i in ids] == ["hell"]
+    # unmergeable text falls back to byte tokens and still round-trips
+    for text in ["hello world", "héllo ☃ snow", "  spaces\tand\nnewlines", "日本語"]:
+        assert tok.detokenize(tok.tokenize(text)) == text, text
+    assert tok.eod == vocab["<|endoftext|>"]
+    assert tok.vocab_size == len(vocab)
+
+    t2 = build_tokenizer("gpt2", f"{vf},{mf}")
+    assert t2.tokenize("hello") == tok.tokenize("hello")
+
+    # streaming detok holds back incomplete utf-8
+    from megatron_amd.tokenizers import IncrementalDetokenizer
+
+    snow_ids = tok.tokenize("☃")
+    d = IncrementalDetokenizer(tok)
+    parts = [d.put(i) for i in snow_ids]
+    assert "".join(parts) + d.flush() == "☃"
+    assert all(p == "" for p in parts[:-1]) or parts[-1] == "☃" or "☃" in "".join(parts)
