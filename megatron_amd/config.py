@@ -249,6 +249,10 @@ class OptimizerConfig:
     hysteresis: int = 2
     # ZeRO-1 distributed optimizer
     use_distributed_optimizer: bool = False
+    # >1: shard optimizer state over dp_cp/N-rank sub-groups and replicate
+    # across the N instances (reference num_distributed_optimizer_instances):
+    # smaller all-gather domains at large DP
+    num_distributed_optimizer_instances: int = 1
     overlap_param_gather: bool = False
     # lr schedule
     lr_decay_style: str = "cosine"  # 'constant' | 'linear' | 'cosine' | 'wsd'
@@ -267,6 +271,10 @@ class DDPConfig:
     grad_reduce_in_fp32: bool = False
     overlap_grad_reduce: bool = True
     use_distributed_optimizer: bool = False
+    # >1: shard optimizer state over dp_cp/N-rank sub-groups and replicate
+    # across the N instances (reference num_distributed_optimizer_instances):
+    # smaller all-gather domains at large DP
+    num_distributed_optimizer_instances: int = 1
     bucket_size: Optional[int] = 40_000_000  # elements per bucket target
     average_in_collective: bool = True
     check_for_nan_in_grad: bool = False

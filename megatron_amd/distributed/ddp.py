@@ -40,6 +40,39 @@ def _pad_to(x: int, align: int) -> int:
     return int(math.ceil(x / align) * align) if align > 1 else x
 
 
+_GROUP_CACHE: Dict[tuple, object] = {}
+
+
+def _cached_group(ranks: tuple):
+    if ranks not in _GROUP_CACHE:
+        _GROUP_CACHE[ranks] = dist.new_group(ranks=list(ranks))
+    return _GROUP_CACHE[ranks]
+
+
+def instance_groups(dp_group, n_instances: int):
+    """Split the dp(_cp) group into n contiguous optimizer instances.
+    Returns (intra, inter) for this rank: ``intra`` = this rank's instance
+    (shard/all-gather domain), ``inter`` = same shard position across
+    instances (grad replication all-reduce).  Every rank must call this with
+    the same arguments (dist.new_group is collective); groups are cached."""
+    ranks = dist.get_process_group_ranks(dp_group)
+    n = len(ranks)
+    assert n % n_instances == 0, (n, n_instances)
+    per = n // n_instances
+    me = dist.get_rank()
+    intra = inter = None
+    for i in range(n_instances):
+        g = _cached_group(tuple(ranks[i * per : (i + 1) * per]))
+        if me in ranks[i * per : (i + 1) * per]:
+            intra = g
+    for k in range(per):
+        col = [ranks[i * per + k] for i in range(n_instances)]
+        g = _cached_group(tuple(col))
+        if me in col:
+            inter = g
+    return intra, inter
+
+
 class _Bucket:
     def __init__(self, params: List[torch.nn.Parameter], start: int, end: int, index: int):
         self.params = params
@@ -65,6 +98,7 @@ class ParamAndGradBuffer:
         grad_dtype: torch.dtype,
         device: torch.device,
         grad_scale_denom: int = 0,
+        inter_group=None,
     ):
         self.ddp_config = ddp_config
         self.dp_group = dp_group
@@ -76,6 +110,7 @@ class ParamAndGradBuffer:
         # expert_gradient_scaling_factor = edp/dp_cp pre-scale + AVG over edp).
         self.grad_scale_denom = grad_scale_denom or self.dp_size
         self.param_names: List[tuple] = []  # [(name, param)], set by the DDP wrapper
+        self.inter_group = inter_group  # cross-instance grad all-reduce (multi-instance dist-opt)
         self.grad_dtype = grad_dtype
         self.param_dtype = param_dtype
         self.device = device
@@ -186,6 +221,11 @@ class ParamAndGradBuffer:
         for b in self.buckets:
             if b.comm_handle is not None and not isinstance(b.comm_handle, bool):
                 b.comm_handle.wait()
+            if self.inter_group is not None and b.comm_handle is not None:
+                # multi-instance dist-opt: the intra-instance reduce-scatter
+                # left each instance's partial sum in the shard; sum the
+                # replicas across instances (reference two-level grad reduce)
+                dist.all_reduce(b.local_grad_shard, group=self.inter_group)
             b.comm_handle = None
             b.params_with_grad = set()
 
@@ -248,10 +288,18 @@ class DistributedDataParallel(nn.Module):
         )
         self.buffers: List[ParamAndGradBuffer] = []
         self.param_to_buffer: Dict[torch.nn.Parameter, ParamAndGradBuffer] = {}
+        n_inst = getattr(ddp_config, "num_distributed_optimizer_instances", 1)
+        dense_group, inter_group, dense_denom = self.dp_group, None, 0
+        if (n_inst > 1 and ddp_config.use_distributed_optimizer
+                and self.dp_group is not None and dist.is_initialized()):
+            dense_group, inter_group = instance_groups(self.dp_group, n_inst)
+            dense_denom = dist.get_world_size(group=self.dp_group)
+        self.dense_shard_group = dense_group
         if dense_params:
             buf = ParamAndGradBuffer(
-                list(reversed(dense_params)), ddp_config, self.dp_group,
+                list(reversed(dense_params)), ddp_config, dense_group,
                 dense_params[0].dtype, grad_dtype, device,
+                grad_scale_denom=dense_denom, inter_group=inter_group,
             )
             self.buffers.append(buf)
             for p in buf.param_index:
@@ -343,7 +391,9 @@ class DistributedDataParallel(nn.Module):
         if self.dp_group is None or not dist.is_initialized():
             return
         for buf in self.buffers:
-            group = buf.dp_group
+            # multi-instance dense buffer shards over the intra-instance
+            # group, but initial params must agree across ALL dp replicas
+            group = self.dp_group if buf.inter_group is not None else buf.dp_group
             if group is None or dist.get_world_size(group=group) == 1:
                 continue
             src = dist.get_process_group_ranks(group)[0]

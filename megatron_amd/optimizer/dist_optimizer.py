@@ -102,8 +102,12 @@ class DistributedOptimizer(_BaseOptimizer):
             (expert_g if is_exp else dense_g).append(g)
         extra, eextra = [], []
         if G.grid_initialized():
-            # shards are disjoint over dp_cp (dense) / edp (expert)
-            extra.append(G.get_grid().group("dp_cp"))
+            # shards are disjoint over the dense shard group (dp_cp, or the
+            # intra-instance sub-group under multi-instance) / edp (expert);
+            # the cross-instance replicas must NOT be re-counted
+            dense_group = getattr(self.model_chunks[0], "dense_shard_group", None)
+            extra.append(dense_group if dense_group is not None
+                         else G.get_grid().group("dp_cp"))
             eextra.append(G.get_grid().group("expert_dp"))
         total_norm = get_grad_norm(
             dense_g, extra_groups=extra, expert_grads=expert_g, expert_extra_groups=eextra

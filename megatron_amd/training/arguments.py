@@ -173,6 +173,7 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--bucket-size", type=int, default=40_000_000)
     g.add_argument("--check-for-nan-in-loss-and-grad", dest="check_for_nan_in_grad",
                    action="store_true", default=False)
+    g.add_argument("--num-distributed-optimizer-instances", type=int, default=1)
 
     g = p.add_argument_group("checkpointing")
     g.add_argument("--save", type=str, default=None)
@@ -348,6 +349,7 @@ def configs_from_args(args):
         use_distributed_optimizer=args.use_distributed_optimizer,
         bucket_size=args.bucket_size,
         check_for_nan_in_grad=args.check_for_nan_in_grad,
+        num_distributed_optimizer_instances=args.num_distributed_optimizer_instances,
         use_rccl_registered_buffers=args.use_rccl_registered_buffers,
     )
     return cfg, opt_cfg, ddp_cfg

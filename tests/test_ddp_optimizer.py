@@ -281,3 +281,48 @@ def test_check_for_nan_in_grad_names_the_bucket():
     p.main_grad.view(-1)[0] = float("nan")
     with pytest.raises(RuntimeError, match="NaN/Inf grad .*bucket"):
         m.finish_grad_sync()
+
+
+def _dp4_two_instance_case(rank, world):
+    """dist-opt with 2 optimizer instances over dp=4: shards live on 2-rank
+    sub-groups, grads two-level reduced (RS intra + AR across instances)."""
+    import json, os
+
+    G.initialize_model_parallel()
+    model_parallel_seed(1234)
+    cfg = _tiny_cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, weight_decay=0.01, clip_grad=1.0,
+                              use_distributed_optimizer=True)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, use_distributed_optimizer=True,
+                        bucket_size=10_000, num_distributed_optimizer_instances=2)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    buf = chunks[0].buffers[0]
+    assert buf.inter_group is not None
+    assert dist.get_world_size(group=buf.dp_group) == 2  # intra-instance
+
+    all_batches = _gen_batches(8)
+    mine = [all_batches[i] for i in range(len(all_batches)) if i % world == rank]
+    losses = _run_steps(chunks, opt, cfg, mine, 2, 1)
+    t = torch.tensor(losses)
+    dist.broadcast(t, src=0)
+    assert torch.allclose(t, torch.tensor(losses), atol=1e-6), "ranks disagree"
+    if rank == 0:
+        with open(os.environ["DP_TEST_OUT"], "w") as f:
+            json.dump(losses, f)
+
+
+def test_dp4_dist_opt_two_instances_matches_single(tmp_path, monkeypatch):
+    import json
+
+    out = tmp_path / "dp4i2.json"
+    monkeypatch.setenv("DP_TEST_OUT", str(out))
+    init_single()
+    cfg = _tiny_cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, weight_decay=0.01, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    ref = _run_steps(chunks, opt, cfg, _gen_batches(8), 2, 4)
+    spawn_dist(_dp4_two_instance_case, 4)
+    got = json.load(open(out))
+    for a, b in zip(ref, got):
+        assert abs(a - b) < 1e-4, (ref, got)
