@@ -148,6 +148,11 @@ class TransformerConfig(ParallelConfig):
     # activation recompute: None | 'full' | 'selective'
     recompute_granularity: Optional[str] = None
     recompute_num_layers: Optional[int] = None
+    # activation CPU offloading (reference cpu_offloading): saved activations
+    # of the first N layers live in pinned host memory between fwd and bwd
+    # (288 GB HBM3E rarely needs it — this is for >seq-len-stretch cases)
+    activation_cpu_offload: bool = False
+    activation_offload_layers: Optional[int] = None
     # distribute saved activations over TP group when recomputing
     distribute_saved_activations: bool = False
 

@@ -174,6 +174,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--check-for-nan-in-loss-and-grad", dest="check_for_nan_in_grad",
                    action="store_true", default=False)
     g.add_argument("--num-distributed-optimizer-instances", type=int, default=1)
+    g.add_argument("--activation-cpu-offload", action="store_true", default=False)
+    g.add_argument("--activation-offload-layers", type=int, default=None)
 
     g = p.add_argument_group("checkpointing")
     g.add_argument("--save", type=str, default=None)
@@ -326,6 +328,8 @@ def configs_from_args(args):
         fp8_margin=args.fp8_margin,
         recompute_granularity=args.recompute_granularity,
         recompute_num_layers=args.recompute_num_layers,
+        activation_cpu_offload=args.activation_cpu_offload,
+        activation_offload_layers=args.activation_offload_layers,
         deterministic_mode=args.deterministic_mode,
         gradient_accumulation_fusion=torch.cuda.is_available(),
     )

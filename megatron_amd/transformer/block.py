@@ -11,6 +11,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import contextlib
+
 import torch
 import torch.nn as nn
 
@@ -172,10 +174,23 @@ class TransformerBlock(nn.Module):
             if self.config.recompute_num_layers is not None
             else len(self.layers)
         )
+        offload_n = 0
+        if self.config.activation_cpu_offload and self.training and inference_context is None:
+            offload_n = (self.config.activation_offload_layers
+                         if self.config.activation_offload_layers is not None
+                         else len(self.layers))
         for i, layer in enumerate(self.layers):
-            if recompute and i < num_ckpt:
-                hidden_states = self._checkpointed(layer, hidden_states, rotary_freqs)
+            if offload_n and i < offload_n and not (recompute and i < num_ckpt):
+                # saved tensors of this layer go to pinned host memory and
+                # come back at backward (reference cpu_offloading); pinning
+                # only matters on GPU, CPU runs degenerate to plain saves
+                ctx = torch.autograd.graph.save_on_cpu(pin_memory=hidden_states.is_cuda)
             else:
+                ctx = contextlib.nullcontext()
+            with ctx:
+                if recompute and i < num_ckpt:
+                    hidden_states = self._checkpointed(layer, hidden_states, rotary_freqs)
+                    continue
                 if getattr(self, "_graphed_layers", None) is not None and not recompute:
                     from megatron_amd.transformer.hip_graphs import graphed_layer_or_none
 

@@ -236,3 +236,46 @@ def test_dropout_paths_train():
     e1 = m(toks, labels=labels)
     e2 = m(toks, labels=labels)
     torch.testing.assert_close(e1, e2)  # deterministic in eval
+
+
+def test_activation_cpu_offload_grads_match():
+    """--activation-cpu-offload (reference cpu_offloading): saved activations
+    round-trip through save_on_cpu; grads must equal the plain run, and the
+    hook must actually engage (spy on save_on_cpu)."""
+    import unittest.mock as mock
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from tests.utils import init_single
+
+    def run(offload):
+        init_single()
+        model_parallel_seed(77)
+        torch.manual_seed(3)
+        cfg = TransformerConfig(num_layers=3, hidden_size=32, num_attention_heads=4,
+                                num_query_groups=4, vocab_size=64, ffn_hidden_size=48,
+                                activation_cpu_offload=offload,
+                                activation_offload_layers=2 if offload else None,
+                                gradient_accumulation_fusion=False)
+        m = GPTModel(cfg)
+        tokens = torch.randint(0, 64, (2, 12), generator=torch.Generator().manual_seed(5))
+        calls = []
+        orig = torch.autograd.graph.save_on_cpu
+
+        def spy(*a, **k):
+            calls.append(1)
+            return orig(*a, **k)
+
+        with mock.patch.object(torch.autograd.graph, "save_on_cpu", spy):
+            loss = m(tokens, labels=tokens).sum()
+            loss.backward()
+        grads = {n: p.grad.clone() for n, p in m.named_parameters() if p.grad is not None}
+        return float(loss), grads, len(calls)
+
+    l0, g0, c0 = run(False)
+    l1, g1, c1 = run(True)
+    assert c0 == 0 and c1 == 2  # exactly the first two layers offloaded
+    assert abs(l0 - l1) < 1e-6
+    for n in g0:
+        torch.testing.assert_close(g1[n], g0[n], rtol=1e-6, atol=1e-7, msg=n)
