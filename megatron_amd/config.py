@@ -227,6 +227,7 @@ class OptimizerConfig:
     """Analog of the reference OptimizerConfig (optimizer/optimizer_config.py)."""
 
     optimizer: str = "adam"  # 'adam' | 'muon' | 'sgd'
+    sgd_momentum: float = 0.9
     # Muon (reference optimizer/muon.py): orthogonalized momentum for 2-D weights
     muon_momentum: float = 0.95
     muon_ns_steps: int = 5

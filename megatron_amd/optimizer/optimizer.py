@@ -38,6 +38,27 @@ def _wd_group(param, name_hint: str = "") -> bool:
     return param.dim() > 1
 
 
+def apply_param_update(config, mains, grads, exp_avg, exp_avg_sq, lr, wd, step,
+                       model_params_bf16=None):
+    """One param-group update: fused AdamW (K10) or SGD-with-momentum,
+    chosen by config.optimizer (reference supports both)."""
+    if config.optimizer == "sgd":
+        if wd:
+            torch._foreach_add_(grads, mains, alpha=wd)
+        torch._foreach_mul_(exp_avg, config.sgd_momentum)
+        torch._foreach_add_(exp_avg, grads)
+        torch._foreach_add_(mains, exp_avg, alpha=-lr)
+        if model_params_bf16 is not None:
+            for mp, p in zip(mains, model_params_bf16):
+                p.copy_(mp.to(p.dtype))
+        return
+    ops.fused_adamw(
+        mains, grads, exp_avg, exp_avg_sq,
+        lr, config.adam_beta1, config.adam_beta2, config.adam_eps, wd, step,
+        **({"model_params_bf16": model_params_bf16} if model_params_bf16 is not None else {}),
+    )
+
+
 class _BaseOptimizer:
     # scheduler lr is multiplied by this (decoupled-lr groups, reference
     # optimizer/__init__.py lr_mult for embedding/output params)
@@ -116,15 +137,13 @@ class FP32Optimizer(_BaseOptimizer):
             idx = [i for i, m in enumerate(decay_mask) if m == apply_wd]
             if not idx:
                 continue
-            ops.fused_adamw(
+            apply_param_update(
+                self.config,
                 [self.params[i].data for i in idx],
                 [grads[i] for i in idx],
                 [self.exp_avg[i] for i in idx],
                 [self.exp_avg_sq[i] for i in idx],
                 self._lr,
-                self.config.adam_beta1,
-                self.config.adam_beta2,
-                self.config.adam_eps,
                 self._wd if apply_wd else 0.0,
                 self.step_count,
             )
@@ -201,15 +220,13 @@ class MixedPrecisionOptimizer(_BaseOptimizer):
             idx = [i for i, m in enumerate(decay_mask) if m == apply_wd]
             if not idx:
                 continue
-            ops.fused_adamw(
+            apply_param_update(
+                self.config,
                 [self.main_params[i] for i in idx],
                 [grads[i] for i in idx],
                 [self.exp_avg[i] for i in idx],
                 [self.exp_avg_sq[i] for i in idx],
                 self._lr,
-                self.config.adam_beta1,
-                self.config.adam_beta2,
-                self.config.adam_eps,
                 self._wd if apply_wd else 0.0,
                 self.step_count,
                 model_params_bf16=[self.params[i].data for i in idx],

@@ -146,7 +146,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--deterministic-mode", action="store_true")
 
     g = p.add_argument_group("optimizer")
-    g.add_argument("--optimizer", choices=["adam", "muon"], default="adam")
+    g.add_argument("--optimizer", choices=["adam", "muon", "sgd"], default="adam")
+    g.add_argument("--sgd-momentum", type=float, default=0.9)
     g.add_argument("--optimizer-cpu-offload", action="store_true", default=False)
     g.add_argument("--muon-momentum", type=float, default=0.95)
     g.add_argument("--lr", type=float, default=3e-4)
@@ -335,6 +336,7 @@ def configs_from_args(args):
     )
     opt_cfg = OptimizerConfig(
         optimizer=args.optimizer,
+        sgd_momentum=args.sgd_momentum,
         optimizer_cpu_offload=args.optimizer_cpu_offload,
         muon_momentum=args.muon_momentum,
         lr=args.lr, min_lr=args.min_lr,

@@ -189,3 +189,51 @@ def test_layer_wise_matches_fp32_without_clip():
     for (na, pa), (nb, pb) in zip(model_a.named_parameters(), model_b.named_parameters()):
         assert_close(pa, pb, rtol=1e-5, atol=1e-6, msg=na)
     opt_b.close()
+
+
+def test_sgd_optimizer_matches_torch():
+    """--optimizer sgd: momentum-SGD through the mixed-precision wrapper
+    equals torch.optim.SGD on an unwrapped twin."""
+    import copy
+
+    from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from megatron_amd.training.training import setup_model_and_optimizer
+    from tests.utils import init_single
+
+    init_single()
+    model_parallel_seed(17)
+
+    def provider(config, pre_process=True, post_process=True, vp_stage=None):
+        torch.manual_seed(21)
+        return GPTModel(config)
+
+    cfg = TransformerConfig(num_layers=2, hidden_size=32, num_attention_heads=4,
+                            num_query_groups=4, vocab_size=64, ffn_hidden_size=48,
+                            gradient_accumulation_fusion=False)
+    opt_cfg = OptimizerConfig(lr=1e-2, weight_decay=0.0, clip_grad=0.0,
+                              optimizer="sgd", sgd_momentum=0.9)
+    chunks, opt = setup_model_and_optimizer(provider, cfg, opt_cfg,
+                                            DDPConfig(grad_reduce_in_fp32=True))
+    torch.manual_seed(21)
+    twin = GPTModel(cfg)
+    twin.load_state_dict({k: v for k, v in chunks[0].module.state_dict().items()})
+    topt = torch.optim.SGD(twin.parameters(), lr=1e-2, momentum=0.9)
+
+    tokens = torch.randint(0, 64, (2, 10), generator=torch.Generator().manual_seed(4))
+    for _ in range(3):
+        chunks[0].zero_grad_buffer()
+        chunks[0](tokens, labels=tokens).sum().backward()
+        chunks[0].start_grad_sync()
+        ok, _, _ = opt.step()
+        assert ok
+
+        topt.zero_grad()
+        twin(tokens, labels=tokens).sum().backward()
+        topt.step()
+
+    for (n, p), (n2, q) in zip(chunks[0].module.named_parameters(),
+                               twin.named_parameters()):
+        assert n == n2
+        torch.testing.assert_close(p.detach(), q.detach(), rtol=1e-5, atol=1e-6, msg=n)
