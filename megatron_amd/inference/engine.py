@@ -210,6 +210,81 @@ class StaticInferenceEngine:
                 r.text = self.tokenizer.detokenize(r.output_tokens)
         return results
 
+    @torch.no_grad()
+    def generate_speculative(self, prompts: Sequence, params: SamplingParams = SamplingParams(),
+                             draft_fn=None, num_draft: int = 2) -> List[GenerationResult]:
+        """Draft-verify speculative decoding (greedy, pp=1): each step feeds
+        the committed token plus ``num_draft`` drafted tokens as one chunk;
+        the model's own argmax verifies the drafts, accepting the longest
+        matching prefix, so outputs are token-identical to plain greedy
+        decode while accepted drafts cost one forward for several tokens.
+
+        ``draft_fn(tokens) -> List[int]`` proposes continuations of the full
+        context; default is prompt-lookup (n-gram match against the context
+        — training-free, strongest on repetitive text).  Rejected draft KV
+        entries are left in place and overwritten by the next chunk at the
+        same offsets (the static cache masks by tracked length)."""
+        assert params.greedy, "speculative decoding is exact for greedy only"
+        if not G.grid_initialized() or G.get_grid().pp == 1:
+            pass
+        else:
+            raise NotImplementedError("speculative decode: pp=1 only")
+        if self.tokenizer is not None and isinstance(prompts[0], str):
+            prompts = [self.tokenizer.tokenize(p) for p in prompts]
+        if draft_fn is None:
+            draft_fn = lambda toks: _prompt_lookup_draft(toks, num_draft)
+
+        results = []
+        for rid, prompt in enumerate(prompts):
+            prompt = list(prompt)
+            assert len(prompt) + params.max_tokens + num_draft <= self.max_seq
+            self.context.reset(1)
+            toks = torch.as_tensor([prompt], device=self.device)
+            logits_tp = self.model(toks, inference_context=self.context)
+            self.context.set_prompt_lens([len(prompt)])
+            cur = int(_full_logits(logits_tp[-1]).float().argmax(dim=-1)[0])
+            out = [cur]
+            while len(out) < params.max_tokens and not (params.stop_on_eod and cur == self.eod):
+                drafts = [int(d) for d in (draft_fn(prompt + out) or [])][:num_draft]
+                chunk = torch.as_tensor([[cur] + drafts], device=self.device)
+                logits_tp = self.model(chunk, inference_context=self.context)
+                preds = _full_logits(logits_tp[:, 0]).float().argmax(dim=-1)  # [1+D]
+                accepted = 0
+                for j, d in enumerate(drafts):
+                    if int(preds[j]) == d:
+                        accepted += 1
+                    else:
+                        break
+                new = drafts[:accepted] + [int(preds[accepted])]
+                if params.stop_on_eod:
+                    cut = next((i + 1 for i, t in enumerate(new) if t == self.eod), len(new))
+                    new = new[:cut]
+                out.extend(new)
+                cur = out[-1]
+                self.context.advance(len(new))
+            out = out[: params.max_tokens]
+            r = GenerationResult(rid, prompt)
+            r.output_tokens = out
+            if params.stop_on_eod and r.output_tokens and r.output_tokens[-1] == self.eod:
+                r.output_tokens = r.output_tokens[:-1]
+            r.finished = True
+            if self.tokenizer is not None:
+                r.text = self.tokenizer.detokenize(r.output_tokens)
+            results.append(r)
+        return results
+
+
+def _prompt_lookup_draft(tokens: List[int], num_draft: int) -> List[int]:
+    """Prompt-lookup drafting: find the most recent earlier occurrence of the
+    trailing 2-gram and propose the tokens that followed it."""
+    if len(tokens) < 3:
+        return []
+    key = tokens[-2:]
+    for i in range(len(tokens) - 3, -1, -1):
+        if tokens[i : i + 2] == key:
+            return tokens[i + 2 : i + 2 + num_draft]
+    return []
+
 
 @dataclass
 class _Request:

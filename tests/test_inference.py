@@ -637,3 +637,67 @@ def test_metrics_endpoint(tiny_model):
     for key in ("megatron_amd_kv_blocks_free", "megatron_amd_prefix_cache_hits",
                 "megatron_amd_active_requests"):
         assert key in body, body
+
+
+def test_speculative_decode_matches_plain_greedy():
+    """Draft-verify speculative decoding is token-identical to plain greedy
+    for ANY drafter: oracle drafts (always accepted), adversarial wrong
+    drafts (always rejected), and the default prompt-lookup."""
+    G.destroy_model_parallel()
+    G.initialize_model_parallel(world_size=1, rank=0)
+    model_parallel_seed(61)
+    m = GPTModel(TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=128, vocab_size=VOCAB, max_position_embeddings=256)).eval()
+    eng = StaticInferenceEngine(m, max_batch=4, max_seq=128)
+    prompts = [[3, 7, 11, 2, 9], [5, 1, 5, 1, 5, 1, 5], [8]]
+    params = SamplingParams(max_tokens=10, greedy=True, stop_on_eod=False)
+    plain = eng.generate(prompts, params)
+
+    # oracle drafter: proposes exactly what the model will emit
+    oracle = {tuple(p): r.output_tokens for p, r in zip(prompts, plain)}
+
+    def oracle_draft(toks):
+        for p, out in oracle.items():
+            if tuple(toks[: len(p)]) == p and toks[len(p):] == out[: len(toks) - len(p)]:
+                done = len(toks) - len(p)
+                return out[done:done + 3]
+        return []
+
+    for draft_fn, label in [(oracle_draft, "oracle"),
+                            (lambda t: [0, 0], "wrong"),
+                            (None, "prompt-lookup")]:
+        got = eng.generate_speculative(prompts, params, draft_fn=draft_fn, num_draft=3)
+        for a, b in zip(plain, got):
+            assert a.output_tokens == b.output_tokens, (label, a.output_tokens, b.output_tokens)
+
+
+def test_speculative_decode_accepts_oracle_drafts():
+    """With an oracle drafter the engine must commit >1 token per forward
+    (the acceptance path really runs)."""
+    G.destroy_model_parallel()
+    G.initialize_model_parallel(world_size=1, rank=0)
+    model_parallel_seed(61)
+    m = GPTModel(TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=128, vocab_size=VOCAB, max_position_embeddings=256)).eval()
+    eng = StaticInferenceEngine(m, max_batch=2, max_seq=128)
+    prompts = [[3, 7, 11, 2, 9]]
+    params = SamplingParams(max_tokens=9, greedy=True, stop_on_eod=False)
+    expect = eng.generate(prompts, params)[0].output_tokens
+
+    calls = []
+    orig_forward = m.forward
+
+    def counting_forward(*a, **k):
+        calls.append(1)
+        return orig_forward(*a, **k)
+
+    m.forward = counting_forward
+    got = eng.generate_speculative(
+        prompts, params,
+        draft_fn=lambda t: expect[len(t) - len(prompts[0]):][:3], num_draft=3)
+    m.forward = orig_forward
+    assert got[0].output_tokens == expect
+    # 1 prefill + ceil((9-1)/4) verify chunks of 4 = 3 forwards total
+    assert len(calls) <= 1 + 3, len(calls)
