@@ -97,9 +97,10 @@ class TransformerLayer(nn.Module):
     def forward(self, hidden_states, rotary_freqs=None, attention_mask=None, inference_context=None, packed_seq_params=None, attention_bias=None):
         residual = hidden_states
         x = self.input_layernorm(hidden_states)
+        kw = {"attention_bias": attention_bias} if attention_bias is not None else {}
         x = self.self_attention(x, rotary_freqs=rotary_freqs, attention_mask=attention_mask,
                                 inference_context=inference_context, packed_seq_params=packed_seq_params,
-                                attention_bias=attention_bias)
+                                **kw)
         x = ops.bias_dropout_add(x, None, residual, self.hidden_dropout, self.training)
         residual = x
         y = self.pre_mlp_layernorm(x)
