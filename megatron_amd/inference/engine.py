@@ -231,7 +231,11 @@ class StaticInferenceEngine:
             raise NotImplementedError("speculative decode: pp=1 only")
         if self.tokenizer is not None and isinstance(prompts[0], str):
             prompts = [self.tokenizer.tokenize(p) for p in prompts]
-        if draft_fn is None:
+        mtp_drafter = None
+        if draft_fn == "mtp":
+            mtp_drafter = MTPDrafter(self.model)
+            draft_fn = None
+        if draft_fn is None and mtp_drafter is None:
             draft_fn = lambda toks: _prompt_lookup_draft(toks, num_draft)
 
         results = []
@@ -244,8 +248,14 @@ class StaticInferenceEngine:
             self.context.set_prompt_lens([len(prompt)])
             cur = int(_full_logits(logits_tp[-1]).float().argmax(dim=-1)[0])
             out = [cur]
+            cur_hidden_idx = len(prompt) - 1
             while len(out) < params.max_tokens and not (params.stop_on_eod and cur == self.eod):
-                drafts = [int(d) for d in (draft_fn(prompt + out) or [])][:num_draft]
+                if mtp_drafter is not None:
+                    pos = len(prompt) + len(out) - 1
+                    drafts = mtp_drafter(mtp_drafter.hidden[cur_hidden_idx, 0], cur, pos)
+                else:
+                    drafts = [int(d) for d in (draft_fn(prompt + out) or [])]
+                drafts = [int(d) for d in drafts][:num_draft]
                 chunk = torch.as_tensor([[cur] + drafts], device=self.device)
                 logits_tp = self.model(chunk, inference_context=self.context)
                 preds = _full_logits(logits_tp[:, 0]).float().argmax(dim=-1)  # [1+D]
@@ -261,8 +271,11 @@ class StaticInferenceEngine:
                     new = new[:cut]
                 out.extend(new)
                 cur = out[-1]
+                cur_hidden_idx = accepted  # chunk position whose pred we committed
                 self.context.advance(len(new))
             out = out[: params.max_tokens]
+            if mtp_drafter is not None:
+                mtp_drafter.hidden = None
             r = GenerationResult(rid, prompt)
             r.output_tokens = out
             if params.stop_on_eod and r.output_tokens and r.output_tokens[-1] == self.eod:
@@ -271,7 +284,46 @@ class StaticInferenceEngine:
             if self.tokenizer is not None:
                 r.text = self.tokenizer.detokenize(r.output_tokens)
             results.append(r)
+        if mtp_drafter is not None:
+            mtp_drafter.remove()
         return results
+
+
+class MTPDrafter:
+    """Model-based drafter: runs the model's MTP head(s) statelessly on the
+    last committed position (DeepSeek-V3-style self-speculation).  The head
+    chain proposes one token per depth; the verify pass keeps outputs exact,
+    so the single-position approximation (no cross-token attention inside
+    the head) only affects the acceptance rate, never correctness."""
+
+    def __init__(self, model):
+        core = model.module if hasattr(model, "module") else model
+        assert core.mtp is not None, "MTPDrafter needs a model built with mtp_num_layers"
+        self.core = core
+        self.hidden = None  # [s, 1, h] from the latest forward
+        self._hook = core.decoder.register_forward_hook(
+            lambda mod, args, out: setattr(self, "hidden", out.detach()))
+
+    def __call__(self, position_hidden: torch.Tensor, next_token: int,
+                 position: int) -> List[int]:
+        core = self.core
+        table = core._rotary_freqs(core.config.max_position_embeddings,
+                                   position_hidden.device)
+        drafts = []
+        h_k = position_hidden.view(1, 1, -1)
+        tok = next_token
+        for head in core.mtp.heads:
+            emb = core.embedding(torch.tensor([[tok]], device=h_k.device))
+            fused = torch.cat([head.norm_hidden(h_k), head.norm_embed(emb)], dim=-1)
+            h_k = head.layer(head.proj(fused),
+                             rotary_freqs=table[position : position + 1])
+            logits, _ = core.output_layer(h_k)
+            tok = int(_full_logits(logits[0, 0]).float().argmax())
+            drafts.append(tok)
+        return drafts
+
+    def remove(self):
+        self._hook.remove()
 
 
 def _prompt_lookup_draft(tokens: List[int], num_draft: int) -> List[int]:

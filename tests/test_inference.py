@@ -701,3 +701,25 @@ def test_speculative_decode_accepts_oracle_drafts():
     assert got[0].output_tokens == expect
     # 1 prefill + ceil((9-1)/4) verify chunks of 4 = 3 forwards total
     assert len(calls) <= 1 + 3, len(calls)
+
+
+def test_speculative_decode_mtp_drafter_matches_plain():
+    """Self-speculation through the model's own MTP head: outputs stay
+    token-identical to plain greedy (verify pass is exact regardless of
+    draft quality)."""
+    from megatron_amd.inference import SamplingParams, StaticInferenceEngine
+
+    G.destroy_model_parallel()
+    G.initialize_model_parallel(world_size=1, rank=0)
+    model_parallel_seed(71)
+    m = GPTModel(TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+        ffn_hidden_size=128, vocab_size=VOCAB, max_position_embeddings=256,
+        mtp_num_layers=1)).eval()
+    eng = StaticInferenceEngine(m, max_batch=2, max_seq=128)
+    prompts = [[3, 7, 11, 2, 9], [5, 1, 4]]
+    params = SamplingParams(max_tokens=8, greedy=True, stop_on_eod=False)
+    plain = eng.generate(prompts, params)
+    got = eng.generate_speculative(prompts, params, draft_fn="mtp", num_draft=1)
+    for a, b in zip(plain, got):
+        assert a.output_tokens == b.output_tokens, (a.output_tokens, b.output_tokens)
