@@ -174,3 +174,43 @@ def test_qwen2_style_qkv_bias_roundtrip():
     for k in hf:
         if "bias" in k:
             torch.testing.assert_close(back[k], hf[k])
+
+
+def test_deepseek_mla_moe_round_trip():
+    """MLA + routed-MoE + shared-expert model exports to DeepSeek-style HF
+    names and converts back bit-identically; converted dict loads strict."""
+    from tools.checkpoint.convert_hf import (
+        deepseek_hf_to_mcore_state_dict,
+        mcore_to_deepseek_hf_state_dict,
+    )
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    init_single()
+    model_parallel_seed(19)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=4,
+        ffn_hidden_size=128, vocab_size=128, max_position_embeddings=64,
+        multi_latent_attention=True, q_lora_rank=48, kv_lora_rank=32,
+        qk_nope_head_dim=16, qk_rope_head_dim=16, v_head_dim=16,
+        num_experts=4, moe_router_topk=2, moe_ffn_hidden_size=32,
+        moe_shared_expert_intermediate_size=32,
+        moe_router_enable_expert_bias=True,
+        untie_embeddings_and_output_weights=True)
+    m = GPTModel(cfg)
+    sd = {k: v for k, v in m.state_dict().items()}
+    hf = mcore_to_deepseek_hf_state_dict(sd, cfg)
+    assert "model.layers.0.self_attn.kv_a_proj_with_mqa.weight" in hf
+    assert "model.layers.1.mlp.experts.3.down_proj.weight" in hf
+    assert "model.layers.0.mlp.shared_experts.up_proj.weight" in hf
+    assert "model.layers.0.mlp.gate.e_score_correction_bias" in hf
+    back = deepseek_hf_to_mcore_state_dict(hf, cfg)
+    for k, v in back.items():
+        assert k in sd, k
+        torch.testing.assert_close(v.float(), sd[k].float(), rtol=0, atol=0, msg=k)
+    missing, unexpected = m.load_state_dict(back, strict=False)
+    assert not unexpected, unexpected[:5]
+    # only buffers/aux state may be missing, no weight tensors
+    assert all("weight" not in k or "router" in k for k in missing), missing

@@ -135,6 +135,106 @@ def mcore_to_hf_state_dict(sd: dict, cfg) -> dict:
     return out
 
 
+def deepseek_hf_to_mcore_state_dict(hf_sd: dict, cfg) -> dict:
+    """DeepSeek-V2/V3-style HF names -> our GPTModel-with-MLA names.
+
+    Attention: q_a/q_b (or q_proj), kv_a_proj_with_mqa ([kv_lora | k_rope]
+    row order matches linear_kv_down), kv_b_proj (per-head [nope | v] matches
+    linear_kv_up).  MoE layers: mlp.gate -> router, experts'
+    gate/up/down -> fused [gate;up] weight1 / weight2 stacks, shared_experts
+    -> mlp.shared_expert.  Structural mapping only — numerical parity
+    additionally requires the same RoPE convention at runtime."""
+    out = {"embedding.weight": hf_sd["model.embed_tokens.weight"]}
+    for i in range(cfg.num_layers):
+        hf = f"model.layers.{i}."
+        us = f"decoder.layers.{i}."
+        a = hf + "self_attn."
+        ua = us + "self_attention."
+        if a + "q_a_proj.weight" in hf_sd:  # low-rank Q
+            out[ua + "linear_q_down.weight"] = hf_sd[a + "q_a_proj.weight"]
+            out[ua + "q_norm.weight"] = hf_sd[a + "q_a_layernorm.weight"]
+            out[ua + "linear_q_up.weight"] = hf_sd[a + "q_b_proj.weight"]
+        else:
+            out[ua + "linear_q_up.weight"] = hf_sd[a + "q_proj.weight"]
+        out[ua + "linear_kv_down.weight"] = hf_sd[a + "kv_a_proj_with_mqa.weight"]
+        out[ua + "kv_norm.weight"] = hf_sd[a + "kv_a_layernorm.weight"]
+        out[ua + "linear_kv_up.weight"] = hf_sd[a + "kv_b_proj.weight"]
+        out[ua + "linear_proj.weight"] = hf_sd[a + "o_proj.weight"]
+        if hf + "mlp.gate.weight" in hf_sd:  # routed MoE layer
+            out[us + "mlp.router.weight"] = hf_sd[hf + "mlp.gate.weight"].float()
+            if hf + "mlp.gate.e_score_correction_bias" in hf_sd:
+                out[us + "mlp.router.expert_bias"] = hf_sd[hf + "mlp.gate.e_score_correction_bias"].float()
+            w1s, w2s = [], []
+            for e in range(cfg.num_experts):
+                ex = hf + f"mlp.experts.{e}."
+                w1s.append(torch.cat([hf_sd[ex + "gate_proj.weight"],
+                                      hf_sd[ex + "up_proj.weight"]], dim=0))
+                w2s.append(hf_sd[ex + "down_proj.weight"])
+            out[us + "mlp.experts.weight1"] = torch.stack(w1s)
+            out[us + "mlp.experts.weight2"] = torch.stack(w2s)
+            if hf + "mlp.shared_experts.gate_proj.weight" in hf_sd:
+                sh = hf + "mlp.shared_experts."
+                out[us + "mlp.shared_expert.linear_fc1.weight"] = torch.cat(
+                    [hf_sd[sh + "gate_proj.weight"], hf_sd[sh + "up_proj.weight"]], dim=0)
+                out[us + "mlp.shared_expert.linear_fc2.weight"] = hf_sd[sh + "down_proj.weight"]
+        else:
+            out[us + "mlp.linear_fc1.weight"] = torch.cat(
+                [hf_sd[hf + "mlp.gate_proj.weight"], hf_sd[hf + "mlp.up_proj.weight"]], dim=0)
+            out[us + "mlp.linear_fc2.weight"] = hf_sd[hf + "mlp.down_proj.weight"]
+        out[us + "input_layernorm.weight"] = hf_sd[hf + "input_layernorm.weight"]
+        out[us + "pre_mlp_layernorm.weight"] = hf_sd[hf + "post_attention_layernorm.weight"]
+    out["decoder.final_layernorm.weight"] = hf_sd["model.norm.weight"]
+    out["output_layer.weight"] = hf_sd.get("lm_head.weight", hf_sd["model.embed_tokens.weight"])
+    return out
+
+
+def mcore_to_deepseek_hf_state_dict(sd: dict, cfg) -> dict:
+    """Inverse of deepseek_hf_to_mcore_state_dict (dense + routed layers)."""
+    out = {"model.embed_tokens.weight": sd["embedding.weight"]}
+    for i in range(cfg.num_layers):
+        hf = f"model.layers.{i}."
+        us = f"decoder.layers.{i}."
+        a, ua = hf + "self_attn.", us + "self_attention."
+        if ua + "linear_q_down.weight" in sd:
+            out[a + "q_a_proj.weight"] = sd[ua + "linear_q_down.weight"]
+            out[a + "q_a_layernorm.weight"] = sd[ua + "q_norm.weight"]
+            out[a + "q_b_proj.weight"] = sd[ua + "linear_q_up.weight"]
+        else:
+            out[a + "q_proj.weight"] = sd[ua + "linear_q_up.weight"]
+        out[a + "kv_a_proj_with_mqa.weight"] = sd[ua + "linear_kv_down.weight"]
+        out[a + "kv_a_layernorm.weight"] = sd[ua + "kv_norm.weight"]
+        out[a + "kv_b_proj.weight"] = sd[ua + "linear_kv_up.weight"]
+        out[a + "o_proj.weight"] = sd[ua + "linear_proj.weight"]
+        if us + "mlp.router.weight" in sd:
+            out[hf + "mlp.gate.weight"] = sd[us + "mlp.router.weight"]
+            if us + "mlp.router.expert_bias" in sd:
+                out[hf + "mlp.gate.e_score_correction_bias"] = sd[us + "mlp.router.expert_bias"]
+            w1 = sd[us + "mlp.experts.weight1"]
+            w2 = sd[us + "mlp.experts.weight2"]
+            half = w1.shape[1] // 2
+            for e in range(w1.shape[0]):
+                ex = hf + f"mlp.experts.{e}."
+                out[ex + "gate_proj.weight"] = w1[e, :half]
+                out[ex + "up_proj.weight"] = w1[e, half:]
+                out[ex + "down_proj.weight"] = w2[e]
+            if us + "mlp.shared_expert.linear_fc1.weight" in sd:
+                fc1 = sd[us + "mlp.shared_expert.linear_fc1.weight"]
+                sh = hf + "mlp.shared_experts."
+                out[sh + "gate_proj.weight"] = fc1[: fc1.shape[0] // 2]
+                out[sh + "up_proj.weight"] = fc1[fc1.shape[0] // 2 :]
+                out[sh + "down_proj.weight"] = sd[us + "mlp.shared_expert.linear_fc2.weight"]
+        else:
+            fc1 = sd[us + "mlp.linear_fc1.weight"]
+            out[hf + "mlp.gate_proj.weight"] = fc1[: fc1.shape[0] // 2]
+            out[hf + "mlp.up_proj.weight"] = fc1[fc1.shape[0] // 2 :]
+            out[hf + "mlp.down_proj.weight"] = sd[us + "mlp.linear_fc2.weight"]
+        out[hf + "input_layernorm.weight"] = sd[us + "input_layernorm.weight"]
+        out[hf + "post_attention_layernorm.weight"] = sd[us + "pre_mlp_layernorm.weight"]
+    out["model.norm.weight"] = sd["decoder.final_layernorm.weight"]
+    out["lm_head.weight"] = sd["output_layer.weight"]
+    return out
+
+
 def config_from_hf(hf_cfg):
     from megatron_amd.config import TransformerConfig
 
@@ -150,11 +250,23 @@ def config_from_hf(hf_cfg):
         layernorm_epsilon=hf_cfg.rms_norm_eps,
         untie_embeddings_and_output_weights=not getattr(hf_cfg, "tie_word_embeddings", False),
         rope_scaling=(dict(getattr(hf_cfg, "rope_scaling")) if getattr(hf_cfg, "rope_scaling", None) else None),
-        num_experts=getattr(hf_cfg, "num_local_experts", None),
+        num_experts=(getattr(hf_cfg, "num_local_experts", None)
+                     or getattr(hf_cfg, "n_routed_experts", None)),
         moe_router_topk=getattr(hf_cfg, "num_experts_per_tok", 2),
         moe_router_pre_softmax=False,
         add_qkv_bias=bool(getattr(hf_cfg, "attention_bias", False)
                           or "qwen2" in str(getattr(hf_cfg, "model_type", ""))),
+        # DeepSeek-style MLA dims (None/0 elsewhere)
+        multi_latent_attention=bool(getattr(hf_cfg, "kv_lora_rank", None)),
+        q_lora_rank=getattr(hf_cfg, "q_lora_rank", None),
+        kv_lora_rank=getattr(hf_cfg, "kv_lora_rank", None) or 0,
+        qk_nope_head_dim=getattr(hf_cfg, "qk_nope_head_dim", None) or 0,
+        qk_rope_head_dim=getattr(hf_cfg, "qk_rope_head_dim", None) or 0,
+        v_head_dim=getattr(hf_cfg, "v_head_dim", None) or 0,
+        moe_ffn_hidden_size=getattr(hf_cfg, "moe_intermediate_size", None),
+        moe_shared_expert_intermediate_size=(
+            (getattr(hf_cfg, "n_shared_experts", 0) or 0)
+            * (getattr(hf_cfg, "moe_intermediate_size", 0) or 0) or None),
     )
 
 
