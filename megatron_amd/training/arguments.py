@@ -73,7 +73,17 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--moe-router-score-function", choices=["softmax", "sigmoid"], default="softmax")
     g.add_argument("--moe-shared-expert-intermediate-size", type=int, default=None)
     g.add_argument("--moe-token-dispatcher-type", choices=["alltoall", "allgather"], default="alltoall")
-    g.add_argument("--moe-layer-freq", type=int, default=1)
+    def _moe_freq(v):
+        try:
+            return int(v)
+        except ValueError:
+            import ast
+
+            pat = ast.literal_eval(v)  # e.g. "[0]*1+[1]*3" forms are not allowed; pass a plain list
+            assert isinstance(pat, (list, tuple)) and all(x in (0, 1) for x in pat)
+            return list(pat)
+    g.add_argument("--moe-layer-freq", type=_moe_freq, default=1,
+                   help="int N (every Nth layer is MoE) or a 0/1 list like [0,0,1,1]")
     g.add_argument("--moe-router-pre-softmax", action="store_true", default=False)
     g.add_argument("--moe-router-num-groups", type=int, default=None)
     g.add_argument("--moe-router-group-topk", type=int, default=None)

@@ -35,7 +35,11 @@ def num_floating_point_operations(config, batch_size: int, seq_len: int) -> floa
         shared = 0
         if config.moe_shared_expert_intermediate_size:
             shared = 2 * h * config.moe_shared_expert_intermediate_size * 3
-        n_moe = L // config.moe_layer_freq
+        freq = config.moe_layer_freq
+        if isinstance(freq, (list, tuple)):
+            n_moe = sum(bool(freq[i % len(freq)]) for i in range(L))
+        else:
+            n_moe = L // freq
         per_token_layer = per_token_layer - 2 * h * f * (3 if gated else 2) * (n_moe / L)
         per_token_layer += (moe_mlp + shared) * (n_moe / L)
     fwd = tokens * (L * per_token_layer + 2 * h * V)

@@ -46,7 +46,9 @@ class Norm(nn.Module):
 
 
 def _make_mixer(config, layer_number: int):
-    if config.num_experts is not None and (layer_number % config.moe_layer_freq == config.moe_layer_freq - 1):
+    from megatron_amd.transformer.layer_specs import moe_layer_pattern
+
+    if moe_layer_pattern(config, layer_number):
         from megatron_amd.moe.moe_layer import MoELayer
 
         return MoELayer(config, layer_number=layer_number)

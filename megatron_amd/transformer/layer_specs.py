@@ -44,14 +44,23 @@ def get_gpt_layer_spec(moe: bool = False, mla: bool = False) -> ModuleSpec:
     )
 
 
+def moe_layer_pattern(config, layer_index: int) -> bool:
+    """Whether layer ``layer_index`` (0-based global) is a MoE layer.
+    config.moe_layer_freq: int N = every Nth layer (reference), or a 0/1
+    list per layer (DeepSeek "first k dense" patterns)."""
+    if config.num_experts is None:
+        return False
+    freq = config.moe_layer_freq
+    if isinstance(freq, (list, tuple)):
+        return bool(freq[layer_index % len(freq)])
+    return layer_index % freq == freq - 1
+
+
 def get_gpt_decoder_block_spec(config) -> List[ModuleSpec]:
     """Per-layer spec list mixing dense and MoE layers according to
     config.moe_layer_freq (reference gpt_layer_specs.py:676)."""
     specs = []
     for i in range(config.num_layers):
-        is_moe = (
-            config.num_experts is not None
-            and (i % config.moe_layer_freq == config.moe_layer_freq - 1)
-        )
-        specs.append(get_gpt_layer_spec(moe=is_moe, mla=config.multi_latent_attention))
+        specs.append(get_gpt_layer_spec(moe=moe_layer_pattern(config, i),
+                                        mla=config.multi_latent_attention))
     return specs

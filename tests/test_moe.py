@@ -543,3 +543,29 @@ def test_pad_to_capacity_static_shapes_match_unpadded():
     assert seen_pad["shape"][0] == 4 * cap          # static buffer
     assert all(c == cap for c in seen_pad["counts"])  # static per-expert size
     assert seen_ref["shape"][0] != 4 * cap or sum(seen_ref["counts"]) != 4 * cap
+
+
+def test_moe_layer_pattern_list():
+    """moe_layer_freq as a 0/1 list (DeepSeek first-k-dense): dense layers
+    get plain MLPs, listed layers get MoE; the model trains."""
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.moe.moe_layer import MoELayer
+    from megatron_amd.parallel.random import model_parallel_seed
+    from megatron_amd.transformer.mlp import MLP
+    from tests.utils import init_single
+
+    init_single()
+    model_parallel_seed(7)
+    cfg = TransformerConfig(num_layers=4, hidden_size=64, num_attention_heads=4,
+                            num_query_groups=2, vocab_size=96, ffn_hidden_size=96,
+                            num_experts=4, moe_router_topk=2, moe_ffn_hidden_size=48,
+                            moe_layer_freq=[0, 0, 1, 1])
+    m = GPTModel(cfg)
+    kinds = [type(l.mlp) for l in m.decoder.layers]
+    assert kinds[0] is MLP and kinds[1] is MLP
+    assert kinds[2] is MoELayer and kinds[3] is MoELayer
+    tokens = torch.randint(0, 96, (2, 12))
+    loss = m(tokens, labels=tokens)
+    loss.sum().backward()
+    assert m.decoder.layers[2].mlp.router.weight.grad is not None
