@@ -114,7 +114,14 @@ def resolve_resume_source(persistent_root: Optional[str],
 
 
 def load_checkpoint(root: str, models: List, optimizer, scheduler=None,
-                    iteration: Optional[int] = None, load_rng: bool = True) -> int:
+                    iteration: Optional[int] = None, load_rng: bool = True,
+                    load_optim: bool = True) -> int:
+    """load_optim=False (--finetune / --no-load-optim): restore model weights
+    only — optimizer state, scheduler and step counts start fresh."""
+    if not load_optim:
+        optimizer_arg = optimizer
+        optimizer = None
+        scheduler = None
     if iteration is None:
         with open(os.path.join(root, TRACKER)) as f:
             iteration = int(f.read().strip())
@@ -154,6 +161,11 @@ def load_checkpoint(root: str, models: List, optimizer, scheduler=None,
                     p.data.copy_(mp.to(p.dtype))
     if scheduler is not None and common.get("scheduler") is not None:
         scheduler.load_state_dict(common["scheduler"])
+    if not load_optim and optimizer_arg is not None:
+        # weights changed under the optimizer: refresh fp32 main copies
+        for sub in optimizer_arg.chained_optimizers:
+            if hasattr(sub, "reload_model_params"):
+                sub.reload_model_params()
     rank = dist.get_rank() if dist.is_initialized() else 0
     rng_file = os.path.join(path, f"rng_r{rank}.pt")
     if load_rng and os.path.exists(rng_file):

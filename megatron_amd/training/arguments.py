@@ -197,6 +197,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--non-persistent-save-interval", type=int, default=None)
     g.add_argument("--async-save", action="store_true")
     g.add_argument("--no-load-rng", action="store_true")
+    g.add_argument("--no-load-optim", action="store_true")
+    g.add_argument("--finetune", action="store_true",
+                   help="load model weights only; reset iteration, optimizer and schedule")
 
     g = p.add_argument_group("data")
     g.add_argument("--data-path", type=str, nargs="*", default=None)

@@ -193,8 +193,12 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
             from megatron_amd.checkpoint.checkpointing import resolve_resume_source
 
             _root, _it = resolve_resume_source(args.load, args.non_persistent_ckpt_dir)
+            load_optim = not (args.finetune or args.no_load_optim)
             iteration = load_checkpoint(_root or args.load, chunks, optimizer, scheduler,
-                                        load_rng=not args.no_load_rng)
+                                        load_rng=not args.no_load_rng and not args.finetune,
+                                        load_optim=load_optim)
+            if args.finetune:
+                iteration = 0  # new run: fresh schedule over the loaded weights
             _print_rank0(f"loaded checkpoint at iteration {iteration}")
         except FileNotFoundError:
             _print_rank0(f"no checkpoint found in {args.load}; starting fresh")

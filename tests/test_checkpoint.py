@@ -414,3 +414,33 @@ print("DELTA_MB", (peak - base) / 1024)
     # requester shards: 2 x 64MB dst (pre-allocated before baseline? no: after)
     # windowed budget: ~1 shard file (128MB) + 2 x 64MB dst + slack << 1GB
     assert delta_mb < 450, f"load peaked {delta_mb} MB above baseline (full assembly?)"
+
+
+def test_finetune_load_weights_only(tmp_path):
+    """--finetune / --no-load-optim: model weights restore, optimizer moments
+    and step counts start fresh."""
+    from megatron_amd.checkpoint.checkpointing import load_checkpoint, save_checkpoint
+
+    init_single()
+    model_parallel_seed(1234)
+    cfg = _cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    batches = _gen_batches(4)
+    _train(chunks, opt, cfg, batches, 2, 2)
+    save_checkpoint(str(tmp_path / "ck"), chunks, opt, iteration=2)
+    want = {n: p.detach().clone() for n, p in chunks[0].module.named_parameters()}
+
+    model_parallel_seed(999)
+    chunks2, opt2 = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    it = load_checkpoint(str(tmp_path / "ck"), chunks2, opt2, load_rng=False,
+                         load_optim=False)
+    for n, p in chunks2[0].module.named_parameters():
+        torch.testing.assert_close(p.detach(), want[n], rtol=0, atol=0, msg=n)
+    for sub in opt2.chained_optimizers:
+        assert sub.step_count == 0
+        moments = getattr(sub, "exp_avg", None)
+        if moments is None and hasattr(sub, "segments"):
+            moments = [seg.exp_avg for seg in sub.segments]
+        assert moments and all(float(m.abs().sum()) == 0.0 for m in moments)
