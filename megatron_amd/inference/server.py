@@ -74,6 +74,70 @@ def create_app(engine, tokenizer=None):
                             "content": tokenizer.detokenize(r.output_tokens)},
                 "tokens": r.output_tokens}
 
+    def _sp(req, defaults_max=256):
+        return SamplingParams(
+            max_tokens=int(req.get("max_tokens", defaults_max)),
+            temperature=float(req.get("temperature", 1.0)),
+            top_k=int(req.get("top_k", 0)), top_p=float(req.get("top_p", 0.0)),
+            greedy=float(req.get("temperature", 1.0)) == 0.0,
+            return_log_probs=bool(req.get("logprobs", False)),
+            stop_strings=tuple(req.get("stop", []) if isinstance(req.get("stop", []), list)
+                               else [req["stop"]]),
+            seed=req.get("seed"))
+
+    @app.post("/v1/completions")
+    async def v1_completions(request: Request):
+        """OpenAI-compatible completions endpoint (prompt string or token
+        list; temperature 0 = greedy; `stop` strings honored)."""
+        req = await request.json()
+        assert tokenizer is not None, "/v1 endpoints need a tokenizer"
+        prompt = req.get("prompt", "")
+        prompts = prompt if isinstance(prompt, list) and prompt and isinstance(prompt[0], (list, str)) \
+            else [prompt]
+        n = int(req.get("n", 1))
+        prompts = [p for p in prompts for _ in range(n)]
+        with lock:
+            results = engine.generate(prompts, _sp(req, defaults_max=16))
+        return {
+            "object": "text_completion",
+            "model": req.get("model", "megatron_amd"),
+            "choices": [
+                {"index": i, "text": r.text,
+                 "finish_reason": ("length" if len(r.output_tokens) >= int(req.get("max_tokens", 16))
+                                   else "stop"),
+                 "logprobs": ({"token_logprobs": r.log_probs}
+                              if req.get("logprobs") else None)}
+                for i, r in enumerate(results)
+            ],
+            "usage": {
+                "prompt_tokens": sum(len(r.prompt_tokens) for r in results),
+                "completion_tokens": sum(len(r.output_tokens) for r in results),
+            },
+        }
+
+    @app.post("/v1/chat/completions")
+    async def v1_chat_completions(request: Request):
+        """OpenAI-compatible chat endpoint over the tokenizer's template."""
+        req = await request.json()
+        assert tokenizer is not None, "/v1 endpoints need a tokenizer"
+        from megatron_amd.tokenizers import apply_chat_template
+
+        prompt_ids = apply_chat_template(tokenizer, req["messages"],
+                                         add_generation_prompt=True)
+        with lock:
+            r = engine.generate([prompt_ids], _sp(req))[0]
+        return {
+            "object": "chat.completion",
+            "model": req.get("model", "megatron_amd"),
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant",
+                                     "content": tokenizer.detokenize(r.output_tokens)},
+                         "finish_reason": ("length" if len(r.output_tokens) >= int(req.get("max_tokens", 256))
+                                           else "stop")}],
+            "usage": {"prompt_tokens": len(prompt_ids),
+                      "completion_tokens": len(r.output_tokens)},
+        }
+
     @app.get("/metrics")
     async def metrics():
         """Prometheus-format serving metrics (reference observability role;

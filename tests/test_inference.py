@@ -723,3 +723,38 @@ def test_speculative_decode_mtp_drafter_matches_plain():
     got = eng.generate_speculative(prompts, params, draft_fn="mtp", num_draft=1)
     for a, b in zip(plain, got):
         assert a.output_tokens == b.output_tokens, (a.output_tokens, b.output_tokens)
+
+
+def test_openai_v1_endpoints(tiny_model):
+    """/v1/completions and /v1/chat/completions: OpenAI-shaped responses,
+    temperature-0 greedy equals /api/generate greedy."""
+    from fastapi.testclient import TestClient
+
+    from megatron_amd.inference.server import create_app
+    from megatron_amd.tokenizers import ByteLevelTokenizer
+
+    tok = ByteLevelTokenizer()  # any string tokenizes (chat template tags too)
+    eng = DynamicInferenceEngine(tiny_model, tokenizer=tok, num_blocks=32, block_size=8)
+    client = TestClient(create_app(eng, tok))
+
+    base = client.post("/api/generate", json={
+        "prompts": ["3 7 11"], "max_tokens": 5, "greedy": True,
+        "stop_on_eod": False}).json()["generations"][0]
+
+    r = client.post("/v1/completions", json={
+        "prompt": "3 7 11", "max_tokens": 5, "temperature": 0.0, "n": 1})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "text_completion"
+    assert body["choices"][0]["finish_reason"] in ("stop", "length")
+    assert body["usage"]["completion_tokens"] > 0
+    assert body["choices"][0]["text"] == base["text"]
+
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "3 7 11"}],
+        "max_tokens": 4, "temperature": 0.0})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert isinstance(body["choices"][0]["message"]["content"], str)
+    assert body["usage"]["prompt_tokens"] > 0
