@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import dataclasses
 from dataclasses import dataclass, field
-from typing import Callable, Optional
+from typing import Callable, Optional, List
 
 import torch
 
@@ -151,6 +151,9 @@ class TransformerConfig(ParallelConfig):
     # activation CPU offloading (reference cpu_offloading): saved activations
     # of the first N layers live in pinned host memory between fwd and bwd
     # (288 GB HBM3E rarely needs it — this is for >seq-len-stretch cases)
+    # selective-recompute module list (reference --recompute-modules):
+    # "core_attn" (default), "mlp", "moe" — module-level rng-replay recompute
+    recompute_modules: Optional[List[str]] = None
     activation_cpu_offload: bool = False
     activation_offload_layers: Optional[int] = None
     # distribute saved activations over TP group when recomputing

@@ -186,6 +186,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
                    action="store_true", default=False)
     g.add_argument("--num-distributed-optimizer-instances", type=int, default=1)
     g.add_argument("--activation-cpu-offload", action="store_true", default=False)
+    g.add_argument("--recompute-modules", nargs="+", default=None,
+                   choices=["core_attn", "mlp", "moe"])
     g.add_argument("--activation-offload-layers", type=int, default=None)
 
     g = p.add_argument_group("checkpointing")
@@ -343,6 +345,7 @@ def configs_from_args(args):
         recompute_granularity=args.recompute_granularity,
         recompute_num_layers=args.recompute_num_layers,
         activation_cpu_offload=args.activation_cpu_offload,
+        recompute_modules=args.recompute_modules,
         activation_offload_layers=args.activation_offload_layers,
         deterministic_mode=args.deterministic_mode,
         gradient_accumulation_fusion=torch.cuda.is_available(),

@@ -125,7 +125,8 @@ class SelfAttention(nn.Module):
                 core_out = ulysses_attention(q, k, v, scale=self.softmax_scale)
             else:
                 core_out = ring_attention(q, k, v, scale=self.softmax_scale)
-        elif self.config.recompute_granularity == "selective" and self.training:
+        elif (self.config.recompute_granularity == "selective" and self.training
+              and "core_attn" in (self.config.recompute_modules or ["core_attn"])):
             # selective recompute: checkpoint only the core-attention region
             # (reference transformer_config 'selective' — the s^2-shaped
             # softmax state is recomputed in backward, everything else saved)

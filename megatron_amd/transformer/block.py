@@ -106,7 +106,15 @@ class TransformerLayer(nn.Module):
         x = ops.bias_dropout_add(x, None, residual, self.hidden_dropout, self.training)
         residual = x
         y = self.pre_mlp_layernorm(x)
-        y = self.mlp(y)
+        if (self.config.recompute_granularity == "selective" and self.training
+                and any(m in ("mlp", "moe") for m in (self.config.recompute_modules or []))):
+            # module-level recompute (reference --recompute-modules): the
+            # MLP/MoE block (incl. EP a2a) reruns in backward under RNG replay
+            from megatron_amd.parallel.random import checkpoint as rng_checkpoint
+
+            y = rng_checkpoint(self.mlp, False, y)
+        else:
+            y = self.mlp(y)
         y = ops.bias_dropout_add(y, None, residual, self.hidden_dropout, self.training)
         return y
 

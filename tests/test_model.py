@@ -279,3 +279,47 @@ def test_activation_cpu_offload_grads_match():
     assert abs(l0 - l1) < 1e-6
     for n in g0:
         torch.testing.assert_close(g1[n], g0[n], rtol=1e-6, atol=1e-7, msg=n)
+
+
+def test_recompute_modules_moe_grads_match():
+    """--recompute-modules moe: the MoE block reruns in backward (router
+    forward fires twice per layer) and grads equal the no-recompute run."""
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from tests.utils import init_single
+
+    def run(modules):
+        init_single()
+        model_parallel_seed(23)
+        torch.manual_seed(4)
+        cfg = TransformerConfig(num_layers=2, hidden_size=32, num_attention_heads=4,
+                                num_query_groups=4, vocab_size=64, ffn_hidden_size=48,
+                                num_experts=4, moe_router_topk=2, moe_ffn_hidden_size=32,
+                                moe_aux_loss_coeff=0.01,
+                                recompute_granularity="selective" if modules else None,
+                                recompute_modules=modules,
+                                gradient_accumulation_fusion=False)
+        m = GPTModel(cfg)
+        calls = []
+        for layer in m.decoder.layers:
+            router = layer.mlp.router
+            orig = router.forward
+            def make(orig):
+                def f(*a, **k):
+                    calls.append(1)
+                    return orig(*a, **k)
+                return f
+            router.forward = make(orig)
+        tokens = torch.randint(0, 64, (2, 10), generator=torch.Generator().manual_seed(8))
+        loss = m(tokens, labels=tokens).sum()
+        loss.backward()
+        return loss, {n: p.grad.clone() for n, p in m.named_parameters()
+                      if p.grad is not None}, len(calls)
+
+    l0, g0, c0 = run(None)
+    l1, g1, c1 = run(["moe"])
+    assert c0 == 2 and c1 == 4  # recompute reruns each layer's router once
+    torch.testing.assert_close(l1, l0, rtol=1e-6, atol=1e-7)
+    for n in g0:
+        torch.testing.assert_close(g1[n], g0[n], rtol=1e-5, atol=1e-6, msg=n)
