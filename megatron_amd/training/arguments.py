@@ -217,6 +217,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--log-timers", action="store_true")
     g.add_argument("--tensorboard-dir", type=str, default=None)
     g.add_argument("--log-memory", action="store_true")
+    g.add_argument("--log-params-norm", action="store_true")
+    g.add_argument("--log-num-zeros-in-grad", action="store_true")
     g.add_argument("--log-straggler", action="store_true")
     g.add_argument("--straggler-report-interval", type=int, default=10)
     g.add_argument("--straggler-ctrlr-port", type=int, default=None,

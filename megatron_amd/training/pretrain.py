@@ -370,6 +370,20 @@ def pretrain(model_provider: Callable, argv=None, forward_step_builder=None):
                 msg += f" | moe max-violation {worst:.2f}"
             if result.get("grad_norm") is not None:
                 msg += f" | grad norm {result['grad_norm']:.3f}"
+            if args.log_params_norm:
+                from megatron_amd.optimizer.clip import get_grad_norm, split_grads_for_norm
+
+                ps = [p for c in chunks
+                      for p in (c.module if hasattr(c, "module") else c).parameters()]
+                dense, expert = split_grads_for_norm(ps, [p.data.float() for p in ps])
+                msg += f" | params norm {float(get_grad_norm(dense, expert_grads=expert)):.3f}"
+            if args.log_num_zeros_in_grad:
+                nz = sum(int((getattr(p, "main_grad", None) if getattr(p, "main_grad", None)
+                              is not None else (p.grad if p.grad is not None else p.new_zeros(1)))
+                             .eq(0).sum())
+                         for c in chunks
+                         for p in (c.module if hasattr(c, "module") else c).parameters())
+                msg += f" | zeros in grad {nz}"
             if args.log_memory and torch.cuda.is_available():
                 msg += f" | mem {torch.cuda.max_memory_allocated()/2**30:.1f}GB"
             if energy is not None:

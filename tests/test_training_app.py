@@ -131,3 +131,20 @@ def test_pretrain_vlm_entry():
                           "--position-embedding-type", "rope"],
                   forward_step_builder=pretrain_vlm.forward_step_builder)
     assert it == 2
+
+
+def test_pretrain_extra_log_flags_and_finetune(tmp_path, capsys):
+    """--log-params-norm / --log-num-zeros-in-grad lines render; --finetune
+    restarts iteration 0 from saved weights."""
+    ckpt = str(tmp_path / "ck")
+    it = pretrain(model_provider, TINY + ["--train-iters", "2", "--save", ckpt,
+                                          "--seed", "3", "--log-interval", "1",
+                                          "--log-params-norm", "--log-num-zeros-in-grad"])
+    assert it == 2
+    out = capsys.readouterr().out
+    assert "params norm" in out and "zeros in grad" in out
+    it2 = pretrain(model_provider, TINY + ["--train-iters", "2", "--load", ckpt,
+                                           "--seed", "3", "--finetune"])
+    assert it2 == 2  # ran 2 fresh iterations, not resumed-at-2-no-op
+    out = capsys.readouterr().out
+    assert "loaded checkpoint at iteration 0" in out  # weights-only, reset
