@@ -341,11 +341,12 @@ class _RingAttention(torch.autograd.Function):
     def forward(ctx, q, k, v, scale, group):
         cp, r, *_ = _ring_peers(group)
         L = q.shape[0] // 2
+        dk_dim, dv_dim = k.shape[-1], v.shape[-1]  # may differ (MLA)
         my_chunks = cp_chunk_ids(r, cp)
-        o = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        o = torch.zeros((*q.shape[:3], dv_dim), dtype=torch.float32, device=q.device)
         lse = torch.full((q.shape[1], q.shape[2], q.shape[0]), float("-inf"),
                          dtype=torch.float32, device=q.device)
-        kv = torch.stack([k, v])
+        kv = torch.cat([k, v], dim=-1)  # one travel tensor even when dk != dv
         for step in range(cp):
             src = (r - step) % cp
             src_chunks = cp_chunk_ids(src, cp)
@@ -359,8 +360,8 @@ class _RingAttention(torch.autograd.Function):
                         continue
                     causal = my_chunks[qi] == src_chunks[ki]
                     qc = q[qi * L:(qi + 1) * L]
-                    kc = kv[0, ki * L:(ki + 1) * L]
-                    vc = kv[1, ki * L:(ki + 1) * L]
+                    kc = kv[ki * L:(ki + 1) * L, ..., :dk_dim]
+                    vc = kv[ki * L:(ki + 1) * L, ..., dk_dim:]
                     oc, lsec = _fwd_partial(qc, kc, vc, causal, scale)
                     sl = slice(qi * L, (qi + 1) * L)
                     o_m, lse_m = _merge(o[sl], lse[..., sl], oc, lsec)
@@ -380,11 +381,12 @@ class _RingAttention(torch.autograd.Function):
         scale, group = ctx.scale, ctx.group
         cp, r, *_ = _ring_peers(group)
         L = q.shape[0] // 2
+        dk_dim = k.shape[-1]
         my_chunks = cp_chunk_ids(r, cp)
         dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
         # kv and its gradient accumulator travel together around the ring;
         # after cp steps the accumulator is back at the kv owner.
-        kv = torch.stack([k, v])
+        kv = torch.cat([k, v], dim=-1)
         dkv = torch.zeros(kv.shape, dtype=torch.float32, device=q.device)
         for step in range(cp):
             src = (r - step) % cp
@@ -397,18 +399,19 @@ class _RingAttention(torch.autograd.Function):
                     sl = slice(qi * L, (qi + 1) * L)
                     kl = slice(ki * L, (ki + 1) * L)
                     dqc, dkc, dvc = _bwd_partial(
-                        dout[sl], q[sl], kv[0, kl], kv[1, kl], out[sl], lse[..., sl],
-                        causal, scale)
+                        dout[sl], q[sl], kv[kl, ..., :dk_dim], kv[kl, ..., dk_dim:],
+                        out[sl], lse[..., sl], causal, scale)
                     dq[sl] += dqc
-                    dkv[0, kl] += dkc
-                    dkv[1, kl] += dvc
+                    dkv[kl, ..., :dk_dim] += dkc
+                    dkv[kl, ..., dk_dim:] += dvc
             if step < cp - 1:
                 kv = _ring_sendrecv(kv, group)
                 dkv = _ring_sendrecv(dkv, group)
         # final rotation returns dkv to its owner (cp-1 hops done, one more)
         if cp > 1:
             dkv = _ring_sendrecv(dkv, group)
-        return dq.to(q.dtype), dkv[0].to(q.dtype), dkv[1].to(q.dtype), None, None
+        return (dq.to(q.dtype), dkv[..., :dk_dim].to(q.dtype),
+                dkv[..., dk_dim:].to(q.dtype), None, None)
 
 
 def ring_attention(q, k, v, scale: Optional[float] = None, group=None):
@@ -444,11 +447,16 @@ def ulysses_attention(q, k, v, scale: Optional[float] = None, group=None):
         return out  # [s_full, b, h/cp, d]
 
     def hp2sp(x):
-        s_full = x.shape[0]
+        s_full, d_ = x.shape[0], x.shape[-1]
         out = all_to_all(group, x.contiguous())
         s_ = s_full // cp
-        return out.view(cp, s_, b, x.shape[2], d).permute(1, 2, 0, 3, 4).reshape(s_, b, x.shape[2] * cp, d)
+        return out.view(cp, s_, b, x.shape[2], d_).permute(1, 2, 0, 3, 4).reshape(s_, b, x.shape[2] * cp, d_)
 
     qh, kh, vh = sp2hp(q), sp2hp(k), sp2hp(v)
-    oh = ops.flash_attention(qh, kh, vh, causal=True, scale=scale)
+    if v.shape[-1] == d and d in (64, 128) and q.is_cuda:
+        oh = ops.flash_attention(qh, kh, vh, causal=True, scale=scale)
+    else:
+        from megatron_amd.ops import reference as _ref
+
+        oh = _ref.attention(qh, kh, vh, causal=True, scale=scale)
     return hp2sp(oh)

@@ -232,10 +232,21 @@ class MLASelfAttention(nn.Module):
         kv = kv.view(s, b, self.nh, self.nope + self.dv)
         k_nope, v = torch.split(kv, [self.nope, self.dv], dim=3)
 
+        cp = G.get_context_parallel_world_size() if G.grid_initialized() else 1
         if packed_seq_params is not None:
             # THD pack: per-document RoPE positions + block-diagonal attention
             table = self._rope_freqs(packed_seq_params.max_seqlen, hidden_states.device)
             freqs = table[packed_seq_params.positions().to(hidden_states.device)]
+        elif cp > 1:
+            # CP: this rank holds a zigzag (p2p) or contiguous (a2a) shard of
+            # the global sequence; RoPE must use GLOBAL positions
+            from megatron_amd.parallel.context_parallel import cp_rope_positions
+
+            s_global = s * cp
+            table = self._rope_freqs(s_global, hidden_states.device)
+            pos = cp_rope_positions(s_global, G.get_context_parallel_rank(), cp,
+                                    hidden_states.device, mode=self.config.cp_comm_type)
+            freqs = table[pos]
         else:
             freqs = self._rope_freqs(s, hidden_states.device)
         q_rope = ops.rope_apply(q_rope.contiguous(), freqs)
@@ -243,6 +254,17 @@ class MLASelfAttention(nn.Module):
 
         qf = torch.cat([q_nope, q_rope], dim=3)
         kf = torch.cat([k_nope, k_rope.expand(s, b, self.nh, self.rope)], dim=3)
+        if cp > 1 and packed_seq_params is None:
+            from megatron_amd.parallel.context_parallel import ring_attention, ulysses_attention
+
+            if self.config.cp_comm_type == "a2a":
+                core = ulysses_attention(qf, kf.contiguous(), v.contiguous(),
+                                         scale=self.softmax_scale)
+            else:
+                core = ring_attention(qf, kf.contiguous(), v.contiguous(),
+                                      scale=self.softmax_scale)
+            out, _ = self.linear_proj(core.reshape(s, b, self.nh * self.dv))
+            return out
         if packed_seq_params is not None:
             core = ref.attention_varlen(
                 qf, kf.contiguous(), v.contiguous(),
