@@ -36,7 +36,7 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--activation", choices=["swiglu", "geglu", "gelu", "squared_relu"], default="swiglu")
     g.add_argument("--add-linear-bias", action="store_true")
     g.add_argument("--untie-embeddings-and-output-weights", action="store_true", default=True)
-    g.add_argument("--position-embedding-type", choices=["rope", "learned", "none"], default="rope")
+    g.add_argument("--position-embedding-type", choices=["rope", "learned", "relative", "none"], default="rope")
     g.add_argument("--rotary-base", type=float, default=500000.0)
     g.add_argument("--rotary-percent", type=float, default=1.0)
     g.add_argument("--attention-dropout", type=float, default=0.0)

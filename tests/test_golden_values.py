@@ -229,3 +229,49 @@ def test_tiny_gpt_packed_sequences_matches_golden():
     losses = _run_config(["--packed-sequences", "--train-iters", "10"])
     for i, (got, want) in enumerate(zip(losses, golden["lm_loss"])):
         assert abs(got - want) < 2e-3, (i, losses, golden["lm_loss"])
+
+
+def test_tiny_gpt_sgd_golden():
+    losses = _run_config(["--optimizer", "sgd", "--lr", "0.05"])
+    _check_or_record("tiny_sgd.json", losses)
+
+
+def test_tiny_moe_pattern_golden():
+    losses = _run_config(["--num-experts", "4", "--moe-router-topk", "2",
+                          "--moe-aux-loss-coeff", "0.01",
+                          "--moe-layer-freq", "[0,1]"])
+    _check_or_record("tiny_moe_pattern.json", losses)
+
+
+def test_tiny_moe_recompute_modules_golden():
+    """MoE module recompute must track the plain MoE golden exactly."""
+    losses = _run_config(["--num-experts", "4", "--moe-router-topk", "2",
+                          "--moe-aux-loss-coeff", "0.01",
+                          "--recompute-granularity", "selective",
+                          "--recompute-modules", "core_attn", "moe"])
+    _check_or_record("tiny_moe.json", losses)
+
+
+def test_tiny_t5_relative_bias_golden():
+    import pretrain_t5 as T5E
+
+    losses = []
+    orig = P.train_step
+
+    def wrapped(*a, **k):
+        r = orig(*a, **k)
+        losses.append(r["lm_loss"])
+        return r
+
+    P.train_step = wrapped
+    try:
+        P.pretrain(T5E.model_provider, [
+            "--num-layers", "2", "--hidden-size", "64", "--num-attention-heads", "4",
+            "--num-query-groups", "2", "--ffn-hidden-size", "128", "--seq-length", "64",
+            "--micro-batch-size", "2", "--global-batch-size", "4", "--vocab-size", "256",
+            "--mock-data", "--train-iters", "8", "--log-interval", "0", "--seed", "42",
+            "--deterministic-mode", "--position-embedding-type", "relative",
+        ], forward_step_builder=T5E.forward_step_builder)
+    finally:
+        P.train_step = orig
+    _check_or_record("tiny_t5_relative.json", losses)
