@@ -231,15 +231,25 @@ class DynamicInferenceContext:
 
     def __init__(self, num_layers: int, num_kv_heads: int, head_dim: int,
                  num_blocks: int = 512, block_size: int = 256,
-                 dtype=torch.bfloat16, device="cuda"):
+                 dtype=torch.bfloat16, device="cuda", kv_cache_dtype=None):
         self.block_size = block_size
         self.num_layers = num_layers
         # physical block 0 is a scratch target for padded rows in hipGraph
         # decode replays; the allocator only hands out ids 1..num_blocks
         self.allocator = KVBlockAllocator(num_blocks, first_id=1)
+        # kv_cache_dtype="fp8": halve cache memory by storing e4m3 values with
+        # one fp32 scale per (slot, head) — quantize on write, dequantize on
+        # the paged gather (reference kv-cache quantization role)
+        self.kv_fp8 = kv_cache_dtype in ("fp8", torch.float8_e4m3fn)
+        store_dtype = torch.float8_e4m3fn if self.kv_fp8 else dtype
+        self.compute_dtype = dtype
         self.k_cache = [torch.zeros(num_blocks + 1, block_size, num_kv_heads, head_dim,
-                                    dtype=dtype, device=device) for _ in range(num_layers)]
+                                    dtype=store_dtype, device=device) for _ in range(num_layers)]
         self.v_cache = [torch.zeros_like(self.k_cache[0]) for _ in range(num_layers)]
+        if self.kv_fp8:
+            self.k_scale = [torch.ones((num_blocks + 1) * block_size, num_kv_heads,
+                                       dtype=torch.float32, device=device) for _ in range(num_layers)]
+            self.v_scale = [torch.ones_like(self.k_scale[0]) for _ in range(num_layers)]
         self.device = device
         self._static = False
         # step state, set by the engine before each forward
@@ -287,6 +297,31 @@ class DynamicInferenceContext:
         blk = table[torch.div(positions, self.block_size, rounding_mode="floor")]
         return blk * self.block_size + positions % self.block_size
 
+    def _quantize(self, x):
+        """x [..., hkv, d] -> (fp8 payload, fp32 scale [..., hkv])."""
+        amax = x.detach().float().abs().amax(dim=-1).clamp(min=1e-8)
+        sc = amax / 448.0
+        qx = (x.float() / sc.unsqueeze(-1)).clamp(-448, 448).to(torch.float8_e4m3fn)
+        return qx, sc
+
+    def _store(self, layer, flat_k, flat_v, slots, k, v):
+        if not self.kv_fp8:
+            flat_k[slots] = k
+            flat_v[slots] = v
+            return
+        qk, sk = self._quantize(k)
+        qv, sv = self._quantize(v)
+        flat_k[slots] = qk
+        flat_v[slots] = qv
+        self.k_scale[layer][slots] = sk
+        self.v_scale[layer][slots] = sv
+
+    def _load(self, layer, flat, scales, slots, kind):
+        x = flat[slots]
+        if not self.kv_fp8:
+            return x
+        return (x.float() * scales[slots].unsqueeze(-1)).to(self.compute_dtype)
+
     def attend(self, layer: int, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                scale: float, window=None) -> torch.Tensor:
         kc, vc = self.k_cache[layer], self.v_cache[layer]
@@ -298,15 +333,18 @@ class DynamicInferenceContext:
             assert q.shape[1] == 1, "prefill is one request at a time"
             pos = torch.arange(self._prior_len, self._prior_len + s, device=self.device)
             slots = self._slot_index(pos, self._prefill_table)
-            flat_k[slots] = k[:, 0]
-            flat_v[slots] = v[:, 0]
-            if self._prior_len == 0:
+            self._store(layer, flat_k, flat_v, slots, k[:, 0], v[:, 0])
+            if self._prior_len == 0 and not self.kv_fp8:
                 return ops.flash_attention(q, k, v, causal=True, scale=scale, window=window)
-            # chunked prefill: gather prior + current contiguous KV
+            # chunked prefill: gather prior + current contiguous KV.  (fp8
+            # mode reads back even fresh writes so prefill and decode see the
+            # SAME quantized values — keeps the two paths consistent.)
             all_pos = torch.arange(0, self._prior_len + s, device=self.device)
             all_slots = self._slot_index(all_pos, self._prefill_table)
-            k_full = flat_k[all_slots].unsqueeze(1)  # [L, 1, hkv, d]
-            v_full = flat_v[all_slots].unsqueeze(1)
+            k_full = self._load(layer, flat_k, self.k_scale[layer] if self.kv_fp8 else None,
+                                all_slots, "k").unsqueeze(1)  # [L, 1, hkv, d]
+            v_full = self._load(layer, flat_v, self.v_scale[layer] if self.kv_fp8 else None,
+                                all_slots, "v").unsqueeze(1)
             return ops.flash_attention(q, k_full, v_full, causal=True, scale=scale, window=window)
         # decode: scatter the new token, then paged gather + masked attention
         b = q.shape[1]
@@ -314,8 +352,7 @@ class DynamicInferenceContext:
         rows = torch.arange(b, device=self.device)
         slots = (self._block_tables[rows, torch.div(lens, bs, rounding_mode="floor")] * bs
                  + lens % bs)
-        flat_k[slots] = k[0]
-        flat_v[slots] = v[0]
+        self._store(layer, flat_k, flat_v, slots, k[0], v[0])
         new_lens = lens + 1
         if self._static:
             tables = self._block_tables  # fixed width under graph capture
@@ -325,7 +362,9 @@ class DynamicInferenceContext:
         # gather [b, nb*bs, hkv, d] via block-id expansion
         slot_grid = (tables.unsqueeze(-1) * bs +
                      torch.arange(bs, device=self.device).view(1, 1, bs)).reshape(b, -1)
-        k_gath = flat_k[slot_grid]  # [b, L, hkv, d]
-        v_gath = flat_v[slot_grid]
+        k_gath = self._load(layer, flat_k, self.k_scale[layer] if self.kv_fp8 else None,
+                            slot_grid, "k")  # [b, L, hkv, d]
+        v_gath = self._load(layer, flat_v, self.v_scale[layer] if self.kv_fp8 else None,
+                            slot_grid, "v")
         out = _masked_decode_attention(q[0], k_gath, v_gath, new_lens, scale)
         return out.unsqueeze(0)

@@ -355,7 +355,8 @@ class DynamicInferenceEngine:
 
     def __init__(self, model, tokenizer=None, num_blocks: int = 512, block_size: int = 256,
                  max_batch: int = 64, max_prefill_tokens: int = 8192, device=None,
-                 use_hip_graphs: bool = True, enable_prefix_caching: bool = True):
+                 use_hip_graphs: bool = True, enable_prefix_caching: bool = True,
+                 kv_cache_dtype=None):
         self.model = model.eval()
         self.tokenizer = tokenizer
         cfg = model.config
@@ -367,7 +368,7 @@ class DynamicInferenceEngine:
             cfg.num_layers,
             (cfg.num_query_groups or cfg.num_attention_heads) // tp,
             cfg.kv_channels, num_blocks=num_blocks, block_size=block_size,
-            dtype=cfg.params_dtype, device=device)
+            dtype=cfg.params_dtype, device=device, kv_cache_dtype=kv_cache_dtype)
         self.max_batch = max_batch
         self.max_prefill_tokens = max_prefill_tokens
         self._graphs = None

@@ -38,8 +38,16 @@ class KVHostOffloader:
         pools = getattr(ctx, "mla_latent_pool", None)
         if pools:
             latents = {layer: pool[idx].cpu() for layer, pool in pools.items()}
+        scales = None
+        if getattr(ctx, "kv_fp8", False):
+            # fp8 cache: the per-slot scales must travel with the payload
+            bs = ctx.block_size
+            slot = (idx.unsqueeze(-1) * bs
+                    + torch.arange(bs, device=idx.device)).reshape(-1)
+            scales = (torch.stack([ctx.k_scale[l][slot] for l in range(ctx.num_layers)]).cpu(),
+                      torch.stack([ctx.v_scale[l][slot] for l in range(ctx.num_layers)]).cpu())
         handle = next(self._ids)
-        self._store[handle] = (k_host, v_host, latents)
+        self._store[handle] = (k_host, v_host, latents, scales)
         ctx.allocator.free(list(block_table))
         return handle
 
@@ -50,7 +58,7 @@ class KVHostOffloader:
         """Re-allocate device blocks and restore the saved K/V into them.
         Returns the new block table."""
         ctx = self.context
-        k_host, v_host, latents = self._store.pop(handle)
+        k_host, v_host, latents, scales = self._store.pop(handle)
         n = k_host.shape[1]
         blocks = ctx.allocator.allocate(n)
         idx = torch.as_tensor(blocks, dtype=torch.long, device=ctx.device)
@@ -62,6 +70,13 @@ class KVHostOffloader:
             for layer, lat in latents.items():
                 if layer in pools:
                     pools[layer][idx] = lat.to(ctx.device, non_blocking=True)
+        if scales is not None:
+            bs = ctx.block_size
+            slot = (idx.unsqueeze(-1) * bs
+                    + torch.arange(bs, device=idx.device)).reshape(-1)
+            for l in range(ctx.num_layers):
+                ctx.k_scale[l][slot] = scales[0][l].to(ctx.device, non_blocking=True)
+                ctx.v_scale[l][slot] = scales[1][l].to(ctx.device, non_blocking=True)
         if torch.cuda.is_available() and ctx.k_cache[0].is_cuda:
             torch.cuda.current_stream().synchronize()
         return blocks

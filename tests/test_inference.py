@@ -758,3 +758,49 @@ def test_openai_v1_endpoints(tiny_model):
     assert body["object"] == "chat.completion"
     assert isinstance(body["choices"][0]["message"]["content"], str)
     assert body["usage"]["prompt_tokens"] > 0
+
+
+def test_fp8_kv_cache_attend_close_to_exact():
+    """Paged context with kv_cache_dtype='fp8': attention output tracks the
+    bf16-cache output within e4m3 quantization error."""
+    from megatron_amd.inference.contexts import DynamicInferenceContext
+
+    torch.manual_seed(0)
+    kw = dict(num_layers=1, num_kv_heads=2, head_dim=32, num_blocks=8,
+              block_size=4, dtype=torch.float32, device="cpu")
+    exact = DynamicInferenceContext(**kw)
+    quant = DynamicInferenceContext(**kw, kv_cache_dtype="fp8")
+    assert quant.k_cache[0].dtype == torch.float8_e4m3fn
+
+    table = exact.allocator.allocate(4)
+    quant.allocator.allocate(4)  # same ids
+    L = 9
+    k = torch.randn(L, 1, 2, 32)
+    v = torch.randn(L, 1, 2, 32)
+    q = torch.randn(L, 1, 4, 32)
+    outs = []
+    for ctx in (exact, quant):
+        ctx.begin_prefill(table, prior_len=0)
+        outs.append(ctx.attend(0, q, k, v, scale=0.18))
+    torch.testing.assert_close(outs[1], outs[0], rtol=0.2, atol=0.08)
+    # decode step consistency
+    outs2 = []
+    for ctx in (exact, quant):
+        ctx.begin_decode([table], [L])
+        outs2.append(ctx.attend(0, q[:1], k[:1], v[:1], scale=0.18))
+    torch.testing.assert_close(outs2[1], outs2[0], rtol=0.2, atol=0.08)
+
+
+def test_fp8_kv_cache_preemption_equivalence(tiny_model):
+    """fp8 KV + tight pool: swapped-out blocks carry their scales, so the
+    preempted run equals the big-pool fp8 run token-for-token."""
+    prompts = [[3, 7, 11, 2, 9], [5, 1], [8, 8, 4, 2], [12, 13, 14]]
+    params = SamplingParams(max_tokens=8, greedy=True, stop_on_eod=False)
+    big = DynamicInferenceEngine(tiny_model, num_blocks=64, block_size=4,
+                                 kv_cache_dtype="fp8")
+    expected = big.generate(prompts, params)
+    small = DynamicInferenceEngine(tiny_model, num_blocks=9, block_size=4,
+                                   kv_cache_dtype="fp8")
+    got = small.generate(prompts, params)
+    for a, b in zip(expected, got):
+        assert a.output_tokens == b.output_tokens
