@@ -26,6 +26,9 @@ def main(argv=None):
     parser.add_argument("--host", type=str, default="127.0.0.1")
     parser.add_argument("--kv-blocks", type=int, default=4096)
     parser.add_argument("--kv-block-size", type=int, default=256)
+    parser.add_argument("--kv-cache-dtype", choices=["fp8"], default=None,
+                        help="fp8: e4m3 KV payload + per-slot scales (half the cache memory)")
+    parser.add_argument("--disable-prefix-caching", action="store_true")
     args = parser.parse_args(argv)
     import os
 
@@ -44,7 +47,9 @@ def main(argv=None):
         load_checkpoint(args.load, [model], None, None, load_rng=False)
     tokenizer = build_tokenizer(args.tokenizer_type, args.tokenizer_model, args.vocab_size)
     engine = DynamicInferenceEngine(model, tokenizer, num_blocks=args.kv_blocks,
-                                    block_size=args.kv_block_size, device=device)
+                                    block_size=args.kv_block_size, device=device,
+                                    kv_cache_dtype=args.kv_cache_dtype,
+                                    enable_prefix_caching=not args.disable_prefix_caching)
     print(f"serving on {args.host}:{args.port}", flush=True)
     run_server(engine, tokenizer, host=args.host, port=args.port)
 
