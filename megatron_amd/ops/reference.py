@@ -89,11 +89,51 @@ def rope_freqs(
             )
         elif kind == "linear":
             inv_freq = inv_freq / rope_scaling.get("factor", 1.0)
+        elif kind == "yarn":
+            inv_freq = apply_yarn_rope_scaling(
+                inv_freq,
+                factor=rope_scaling.get("factor", 8.0),
+                beta_fast=rope_scaling.get("beta_fast", 32.0),
+                beta_slow=rope_scaling.get("beta_slow", 1.0),
+                original_max_position=rope_scaling.get(
+                    "original_max_position_embeddings", 4096),
+                base=base, rot_dim=rot_dim,
+            )
         else:
             raise ValueError(f"unknown rope_scaling type {kind!r}")
     t = torch.arange(seq_len, device=device, dtype=torch.float32)
     freqs = torch.outer(t, inv_freq)  # [seq, rot_dim/2]
     return freqs.to(dtype)
+
+
+def yarn_mscale(factor: float) -> float:
+    """YaRN attention-temperature: each of q/k cos/sin is scaled by
+    0.1*ln(factor)+1, i.e. scores by its square."""
+    import math as _m
+
+    return 0.1 * _m.log(factor) + 1.0 if factor > 1.0 else 1.0
+
+
+def apply_yarn_rope_scaling(inv_freq, factor: float, beta_fast: float,
+                            beta_slow: float, original_max_position: int,
+                            base: float, rot_dim: int):
+    """NTK-by-parts interpolation (YaRN): high-frequency dims (short
+    wavelengths, fully inside the original window beta_fast times) keep
+    their frequency; low-frequency dims interpolate by 1/factor; the band
+    between ramps linearly in dimension index."""
+    import math as _m
+
+    def find_dim(num_rotations):
+        # dim index whose wavelength fits `num_rotations` times in the window
+        return (rot_dim * _m.log(original_max_position / (num_rotations * 2 * _m.pi))
+                / (2 * _m.log(base)))
+
+    lo = max(_m.floor(find_dim(beta_fast)), 0)
+    hi = min(_m.ceil(find_dim(beta_slow)), rot_dim // 2 - 1)
+    ramp = ((torch.arange(rot_dim // 2, dtype=torch.float32, device=inv_freq.device) - lo)
+            / max(hi - lo, 1)).clamp(0.0, 1.0)
+    keep = 1.0 - ramp  # 1 for high-freq dims, 0 for low-freq
+    return inv_freq * keep + (inv_freq / factor) * ramp
 
 
 def mrope_freqs(position_ids: torch.Tensor, dim: int, base: float = 10000.0,

@@ -47,6 +47,13 @@ class SelfAttention(nn.Module):
             q_size, self.hidden_size, config=config, bias=config.add_linear_bias
         )
         self.softmax_scale = config.softmax_scale or (1.0 / (self.kv_channels**0.5))
+        rs = getattr(config, "rope_scaling", None)
+        if rs and rs.get("type") == "yarn":
+            from megatron_amd.ops.reference import yarn_mscale
+
+            # YaRN temperature: q and k tables each carry mscale -> scores
+            # carry its square; equivalent to scaling softmax_scale here
+            self.softmax_scale *= yarn_mscale(float(rs.get("factor", 8.0))) ** 2
         self.window = config.window_size
         skip = getattr(config, "window_attn_skip_freq", None)
         if self.window is not None and skip and (layer_number % skip == skip - 1):

@@ -158,3 +158,36 @@ def test_ce_native_tp_merge_math():
     loss = torch.log(s) - (pred_raw - m)
     ref = F.cross_entropy(logits, target, reduction="none")
     assert torch.allclose(loss, ref, rtol=1e-5, atol=1e-5)
+
+
+def test_yarn_rope_scaling():
+    """YaRN NTK-by-parts: high-frequency dims keep their rotation rate,
+    low-frequency dims interpolate by 1/factor, the band ramps; the
+    attention temperature squares into softmax_scale."""
+    import math
+
+    from megatron_amd.ops.reference import rope_freqs, yarn_mscale
+
+    base = rope_freqs(8, 64, base=10000.0)
+    yarn = rope_freqs(8, 64, base=10000.0,
+                      rope_scaling={"type": "yarn", "factor": 8.0,
+                                    "original_max_position_embeddings": 512})
+    # angles at position 1 = inv_freq directly
+    f0, f1 = base[1], yarn[1]
+    assert torch.allclose(f1[0], f0[0])            # fastest dim untouched
+    assert torch.allclose(f1[-1], f0[-1] / 8.0)    # slowest dim fully scaled
+    ratio = (f1 / f0)
+    assert bool((ratio[1:] <= ratio[:-1] + 1e-6).all())  # monotone ramp
+    assert abs(yarn_mscale(8.0) - (0.1 * math.log(8.0) + 1.0)) < 1e-9
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.transformer.attention import SelfAttention
+    from tests.utils import init_single
+
+    init_single()
+    cfg = TransformerConfig(num_layers=1, hidden_size=64, num_attention_heads=4,
+                            num_query_groups=4, vocab_size=64, ffn_hidden_size=64,
+                            rope_scaling={"type": "yarn", "factor": 8.0})
+    att = SelfAttention(cfg, layer_number=1)
+    expect = (1.0 / math.sqrt(att.kv_channels)) * yarn_mscale(8.0) ** 2
+    assert abs(att.softmax_scale - expect) < 1e-9
