@@ -89,6 +89,7 @@ class GenerationResult:
     top_logprobs: List[list] = field(default_factory=list)
     text: Optional[str] = None
     finished: bool = False
+    aborted: bool = False
 
 
 def _full_logits(logits_tp: torch.Tensor) -> torch.Tensor:
@@ -403,6 +404,30 @@ class DynamicInferenceEngine:
 
     def has_work(self) -> bool:
         return bool(self.waiting or self.active or self.preempted)
+
+    def abort(self, rid: int) -> bool:
+        """Cancel a request in any state (waiting / active / preempted):
+        its blocks (or host swap) are released and a finished-with-abort
+        result is published.  Returns False for unknown/finished ids."""
+        for q in (self.waiting, self.active):
+            for req in q:
+                if req.rid == rid:
+                    q.remove(req)
+                    self.context.allocator.free(req.block_table)
+                    req.block_table = []
+                    req.result.finished = True
+                    req.result.aborted = True
+                    self.finished[rid] = req.result
+                    return True
+        for i, (req, handle) in enumerate(self.preempted):
+            if req.rid == rid:
+                self.preempted.pop(i)
+                self.offloader.drop(handle)
+                req.result.finished = True
+                req.result.aborted = True
+                self.finished[rid] = req.result
+                return True
+        return False
 
     def _preempt_one(self, exclude=None) -> bool:
         """Swap the youngest active request's KV to host, freeing its blocks."""

@@ -51,6 +51,10 @@ class KVHostOffloader:
         ctx.allocator.free(list(block_table))
         return handle
 
+    def drop(self, handle: int) -> None:
+        """Discard a swapped-out request's host copy (aborted requests)."""
+        self._store.pop(handle, None)
+
     def num_blocks_of(self, handle: int) -> int:
         return self._store[handle][0].shape[1]
 

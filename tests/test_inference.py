@@ -804,3 +804,31 @@ def test_fp8_kv_cache_preemption_equivalence(tiny_model):
     got = small.generate(prompts, params)
     for a, b in zip(expected, got):
         assert a.output_tokens == b.output_tokens
+
+
+def test_engine_abort_releases_blocks(tiny_model):
+    """abort() cancels waiting/active/preempted requests, frees their KV
+    (allocator returns to full), and survivors complete normally."""
+    eng = DynamicInferenceEngine(tiny_model, num_blocks=32, block_size=4,
+                                 enable_prefix_caching=False)
+    free0 = eng.context.allocator.num_free
+    params = SamplingParams(max_tokens=6, greedy=True, stop_on_eod=False)
+    r1 = eng.add_request([3, 7, 11, 2], params)
+    r2 = eng.add_request([5, 1, 9], params)
+    r3 = eng.add_request([8, 8, 4], params)
+    # start r1 (prefill) so it is ACTIVE, leave r2/r3 waiting
+    eng.step()
+    assert eng.abort(r2)          # waiting
+    assert eng.abort(r1)          # active (blocks allocated)
+    assert not eng.abort(12345)   # unknown
+    while eng.has_work():
+        eng.step()
+    assert eng.finished[r1].aborted and eng.finished[r2].aborted
+    done = eng.finished[r3]
+    assert not done.aborted and len(done.output_tokens) == 6
+    assert eng.context.allocator.num_free == free0
+    # aborted output must equal an untouched run's for the survivor
+    solo = DynamicInferenceEngine(tiny_model, num_blocks=32, block_size=4,
+                                  enable_prefix_caching=False)
+    expect = solo.generate([[8, 8, 4]], params)[0]
+    assert done.output_tokens == expect.output_tokens
