@@ -214,3 +214,32 @@ def test_deepseek_mla_moe_round_trip():
     assert not unexpected, unexpected[:5]
     # only buffers/aux state may be missing, no weight tensors
     assert all("weight" not in k or "router" in k for k in missing), missing
+
+
+def test_export_hf_dir_deepseek(tmp_path):
+    """MLA models export as deepseek_v2-style HF dirs (weights + config)."""
+    import json as _json
+
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from tools.checkpoint.export_hf import export_hf_dir
+
+    init_single()
+    model_parallel_seed(5)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=4,
+        ffn_hidden_size=128, vocab_size=128, max_position_embeddings=64,
+        multi_latent_attention=True, q_lora_rank=48, kv_lora_rank=32,
+        qk_nope_head_dim=16, qk_rope_head_dim=16, v_head_dim=16,
+        untie_embeddings_and_output_weights=True)
+    m = GPTModel(cfg)
+    out = str(tmp_path / "hf")
+    hf_cfg = export_hf_dir(m, cfg, out)
+    assert hf_cfg["model_type"] == "deepseek_v2"
+    assert hf_cfg["kv_lora_rank"] == 32
+    from safetensors.torch import load_file
+
+    sd = load_file(out + "/model.safetensors")
+    assert "model.layers.0.self_attn.kv_a_proj_with_mqa.weight" in sd
+    assert _json.load(open(out + "/config.json"))["q_lora_rank"] == 48

@@ -27,21 +27,30 @@ def export_hf_dir(model, cfg, out_dir: str) -> dict:
     """Materialize the HF state dict + config for `model` into out_dir."""
     from safetensors.torch import save_file
 
-    from tools.checkpoint.convert_hf import mcore_to_hf_state_dict
+    from tools.checkpoint.convert_hf import (
+        mcore_to_deepseek_hf_state_dict,
+        mcore_to_hf_state_dict,
+    )
 
     core = model.module if hasattr(model, "module") else model
     sd = {k: v for k, v in core.state_dict().items()
-          if "expert_bias" not in k and "local_tokens" not in k}
-    hf_sd = mcore_to_hf_state_dict(sd, cfg)
+          if "local_tokens" not in k}
+    if getattr(cfg, "multi_latent_attention", False):
+        hf_sd = mcore_to_deepseek_hf_state_dict(sd, cfg)
+    else:
+        sd = {k: v for k, v in sd.items() if "expert_bias" not in k}
+        hf_sd = mcore_to_hf_state_dict(sd, cfg)
     hf_sd = {k: v.detach().to(torch.bfloat16 if cfg.bf16 else v.dtype).contiguous()
              for k, v in hf_sd.items()}
     os.makedirs(out_dir, exist_ok=True)
     save_file(hf_sd, os.path.join(out_dir, "model.safetensors"),
               metadata={"format": "pt"})
     is_moe = cfg.num_experts is not None
+    is_mla = getattr(cfg, "multi_latent_attention", False)
     hf_cfg = {
-        "architectures": ["MixtralForCausalLM" if is_moe else "LlamaForCausalLM"],
-        "model_type": "mixtral" if is_moe else "llama",
+        "architectures": (["DeepseekV2ForCausalLM"] if is_mla
+                          else ["MixtralForCausalLM" if is_moe else "LlamaForCausalLM"]),
+        "model_type": "deepseek_v2" if is_mla else ("mixtral" if is_moe else "llama"),
         "hidden_size": cfg.hidden_size,
         "intermediate_size": cfg.moe_ffn_hidden_size if is_moe else cfg.ffn_hidden_size,
         "num_hidden_layers": cfg.num_layers,
@@ -58,6 +67,15 @@ def export_hf_dir(model, cfg, out_dir: str) -> dict:
     if is_moe:
         hf_cfg["num_local_experts"] = cfg.num_experts
         hf_cfg["num_experts_per_tok"] = cfg.moe_router_topk
+    if is_mla:
+        hf_cfg.update({
+            "q_lora_rank": cfg.q_lora_rank, "kv_lora_rank": cfg.kv_lora_rank,
+            "qk_nope_head_dim": cfg.qk_nope_head_dim,
+            "qk_rope_head_dim": cfg.qk_rope_head_dim, "v_head_dim": cfg.v_head_dim,
+        })
+        if is_moe:
+            hf_cfg["n_routed_experts"] = cfg.num_experts
+            hf_cfg["moe_intermediate_size"] = cfg.moe_ffn_hidden_size
     if cfg.rope_scaling:
         hf_cfg["rope_scaling"] = cfg.rope_scaling
     with open(os.path.join(out_dir, "config.json"), "w") as f:
