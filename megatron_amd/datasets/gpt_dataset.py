@@ -214,6 +214,7 @@ def build_gpt_train_iterator(args, device, dp_rank: int, dp_size: int,
     train_samples = args.train_iters * args.global_batch_size
     ds = build_gpt_datasets(args.data_path, args.seq_length, args.seed,
                             train_samples, args.split,
+                            cache_dir=getattr(args, "data_cache_path", None),
                             eod=getattr(args, "eod_id", None),
                             reset_attention_mask=getattr(args, "reset_attention_mask", False),
                             eod_mask_loss=getattr(args, "eod_mask_loss", False))

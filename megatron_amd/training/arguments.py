@@ -133,6 +133,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
                    help="restart positions at each EOD (implied by "
                         "--reset-attention-mask on the varlen path)")
     g.add_argument("--eod-mask-loss", action="store_true", default=False)
+    g.add_argument("--data-cache-path", type=str, default=None,
+                   help="directory for dataset index caches (default: next to the data)")
     g.add_argument("--eod-id", type=int, default=None,
                    help="EOD token id for the reset/mask flags (default: tokenizer eod)")
     g.add_argument("--packed-sequences", action="store_true",
