@@ -106,6 +106,26 @@ def update_router_expert_bias(models: List[torch.nn.Module], config):
         r.local_tokens_per_expert.zero_()
 
 
+def _allreduce_tp_replicated_grads(models):
+    """Params tagged ``average_gradients_across_tp_domain`` (HF-interop
+    modules: replicated over TP, not sharded) get their grads averaged over
+    the TP group so replicas never drift (reference huggingface/module.py)."""
+    group = G.get_tensor_model_parallel_group()
+    if group is None or G.get_tensor_model_parallel_world_size() <= 1:
+        return
+    for model in models:
+        core = model.module if hasattr(model, "module") else model
+        for p in core.parameters():
+            if not getattr(p, "average_gradients_across_tp_domain", False):
+                continue
+            g = getattr(p, "main_grad", None)
+            if g is None:
+                g = p.grad
+            if g is not None:
+                dist.all_reduce(g, group=group)
+                g /= G.get_tensor_model_parallel_world_size()
+
+
 def finalize_model_grads(models: List[torch.nn.Module], config=None):
     if config is None:
         core = models[0].module if hasattr(models[0], "module") else models[0]
@@ -114,4 +134,5 @@ def finalize_model_grads(models: List[torch.nn.Module], config=None):
         return
     _allreduce_word_embedding_grads(models, config)
     _allreduce_layernorm_grads(models, config)
+    _allreduce_tp_replicated_grads(models)
     update_router_expert_bias(models, config)
