@@ -34,6 +34,9 @@ def main(argv=None):
     p.add_argument("--isl", type=int, default=512)
     p.add_argument("--osl", type=int, default=128)
     p.add_argument("--block-size", type=int, default=256)
+    p.add_argument("--kv-cache-dtype", choices=["fp8"], default=None)
+    p.add_argument("--speculative", action="store_true",
+                   help="static-engine draft-verify decode (prompt-lookup drafts)")
     args = p.parse_args(argv)
 
     assert torch.cuda.is_available()
@@ -50,17 +53,25 @@ def main(argv=None):
     max_tokens_total = args.batch * (args.isl + args.osl + args.block_size)
     num_blocks = (max_tokens_total + args.block_size - 1) // args.block_size
     eng = DynamicInferenceEngine(model, num_blocks=num_blocks, block_size=args.block_size,
-                                 max_batch=args.batch)
+                                 max_batch=args.batch, kv_cache_dtype=args.kv_cache_dtype)
     rng = torch.Generator().manual_seed(0)
     prompts = [torch.randint(0, args.vocab, (args.isl,), generator=rng).tolist()
                for _ in range(args.batch)]
     params = SamplingParams(max_tokens=args.osl, greedy=True, stop_on_eod=False)
 
+    if args.speculative:
+        from megatron_amd.inference import StaticInferenceEngine
+
+        eng = StaticInferenceEngine(model, max_batch=args.batch,
+                                    max_seq=args.isl + args.osl + 16)
+        run = lambda ps, pr: eng.generate_speculative(ps, pr, num_draft=3)
+    else:
+        run = eng.generate
     # warmup: one tiny round
-    eng.generate(prompts[:2], SamplingParams(max_tokens=4, greedy=True, stop_on_eod=False))
+    run(prompts[:2], SamplingParams(max_tokens=4, greedy=True, stop_on_eod=False))
     torch.cuda.synchronize()
     t0 = time.time()
-    results = eng.generate(prompts, params)
+    results = run(prompts, params)
     torch.cuda.synchronize()
     dt = time.time() - t0
     out_tokens = sum(len(r.output_tokens) for r in results)
@@ -69,6 +80,7 @@ def main(argv=None):
         "unit": "output tok/s", "tpot_ms": round(1000 * dt / args.osl, 2),
         "elapsed_s": round(dt, 2), "batch": args.batch, "isl": args.isl, "osl": args.osl,
         "params_m": round(n_params / 1e6, 1), "dtype": "bf16", "data": "synthetic",
+        "kv_cache_dtype": args.kv_cache_dtype or "bf16", "speculative": args.speculative,
     }))
 
 
