@@ -287,3 +287,30 @@ def test_pp2_combined_1f1b_matches_single(tmp_path, monkeypatch):
     pp_losses = json.load(open(out))
     for a, b in zip(ref, pp_losses):
         assert abs(a - b) < 2e-4, (ref, pp_losses)
+
+
+def _pp2ep2_combined_case(rank, world, combined):
+    G.initialize_model_parallel(pipeline_parallel_size=2, expert_parallel_size=2)
+    model_parallel_seed(1234)
+    cfg = _cfg_moe(pp=2, combined=combined)
+    cfg = cfg.replace(expert_parallel_size=2)
+    losses = _run(cfg, 2, 4, _gen_batches(8))
+    grid = G.get_grid()
+    if grid.is_pipeline_last_stage(ignore_virtual=True) and grid.rank_in("dp") == 0:
+        with open(os.environ["PP_TEST_OUT"], "w") as f:
+            json.dump(losses, f)
+
+
+def test_pp2_ep2_combined_matches_standard(tmp_path, monkeypatch):
+    """The Mixtral-style layout (PP=2 x EP=2, world 4): the combined
+    co-schedule must train identically to the standard 1F1B schedule."""
+    out_s = tmp_path / "std.json"
+    out_c = tmp_path / "cmb.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out_s))
+    spawn_dist(_pp2ep2_combined_case, 4, False)
+    monkeypatch.setenv("PP_TEST_OUT", str(out_c))
+    spawn_dist(_pp2ep2_combined_case, 4, True)
+    std = json.load(open(out_s))
+    cmb = json.load(open(out_c))
+    for a, b in zip(std, cmb):
+        assert abs(a - b) < 2e-4, (std, cmb)
