@@ -314,3 +314,38 @@ def test_pp2_ep2_combined_matches_standard(tmp_path, monkeypatch):
     cmb = json.load(open(out_c))
     for a, b in zip(std, cmb):
         assert abs(a - b) < 2e-4, (std, cmb)
+
+
+def _pp4_case(rank, world, vpp, combined=False):
+    G.initialize_model_parallel(pipeline_parallel_size=world,
+                                virtual_pipeline_parallel_size=vpp)
+    model_parallel_seed(1234)
+    cfg = _cfg(pp=world, vpp=vpp)
+    if combined:
+        cfg = cfg.replace(overlap_moe_expert_parallel_comm=True)
+    losses = _run(cfg, 2, 4, _gen_batches(8))
+    if G.get_grid().is_pipeline_last_stage(ignore_virtual=True):
+        with open(os.environ["PP_TEST_OUT"], "w") as f:
+            json.dump(losses, f)
+
+
+def test_pp4_matches_single(tmp_path, monkeypatch):
+    """Four-deep pipeline (warmup 3 on stage 0) reproduces the single run —
+    the depth the 70B TP4xPP2 and deeper-PP configs rely on."""
+    out = tmp_path / "pp4.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+    ref = _single_reference()
+    spawn_dist(_pp4_case, 4, None)
+    got = json.load(open(out))
+    for a, b in zip(ref, got):
+        assert abs(a - b) < 2e-4, (ref, got)
+
+
+def test_pp4_combined_matches_single(tmp_path, monkeypatch):
+    out = tmp_path / "pp4c.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+    ref = _single_reference()
+    spawn_dist(_pp4_case, 4, None, True)
+    got = json.load(open(out))
+    for a, b in zip(ref, got):
+        assert abs(a - b) < 2e-4, (ref, got)
