@@ -269,6 +269,18 @@ def validate_args(args) -> None:
         args.sequence_parallel = False
     if args.num_experts is not None:
         assert args.num_experts % args.expert_model_parallel_size == 0
+        if isinstance(args.moe_layer_freq, (list, tuple)):
+            assert len(args.moe_layer_freq) <= args.num_layers, (
+                "--moe-layer-freq pattern longer than the layer count")
+    if getattr(args, "reset_attention_mask", False) or getattr(args, "eod_mask_loss", False):
+        assert args.mock_data or args.eod_id is not None or args.tokenizer_type, (
+            "--reset-attention-mask/--eod-mask-loss need --eod-id or a tokenizer")
+    inst = getattr(args, "num_distributed_optimizer_instances", 1)
+    if inst > 1:
+        assert args.use_distributed_optimizer, (
+            "--num-distributed-optimizer-instances needs --use-distributed-optimizer")
+        assert (args.data_parallel_size * args.context_parallel_size) % inst == 0, (
+            "dp_cp must divide evenly into optimizer instances")
 
 
 def configs_from_args(args):
