@@ -27,10 +27,19 @@ class SamplingParams:
     repetition_penalty: float = 1.0
     # attach the top-N (token, logprob) alternatives per generated position
     top_n_logprobs: int = 0
+    # keep only tokens with prob >= min_p * max_prob (0 = off)
+    min_p: float = 0.0
+    # additive per-token-id logit bias {token_id: bias} (OpenAI logit_bias)
+    logit_bias: Optional[dict] = None
 
 
-def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0) -> torch.Tensor:
-    """Mask logits outside top-k / nucleus top-p to -inf. logits: [b, V]."""
+def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0,
+                  min_p: float = 0.0) -> torch.Tensor:
+    """Mask logits outside top-k / nucleus top-p / min-p to -inf. [b, V]."""
+    if min_p > 0.0:
+        probs = torch.softmax(logits.float(), dim=-1)
+        keep = probs >= min_p * probs.amax(dim=-1, keepdim=True)
+        logits = logits.masked_fill(~keep, float("-inf"))
     if top_k > 0:
         kth = torch.topk(logits, min(top_k, logits.size(-1)), dim=-1).values[..., -1, None]
         logits = logits.masked_fill(logits < kth, float("-inf"))
@@ -68,10 +77,14 @@ def sample(logits: torch.Tensor, params: SamplingParams,
     """logits: [b, V] (full vocab, fp32) -> next token ids [b]."""
     if params.repetition_penalty != 1.0 and prev_tokens is not None:
         logits = apply_repetition_penalty(logits, prev_tokens, params.repetition_penalty)
+    if params.logit_bias:
+        logits = logits.clone()
+        for tid, bias in params.logit_bias.items():
+            logits[:, int(tid)] += float(bias)
     if params.greedy or params.temperature == 0.0:
         return logits.argmax(dim=-1)
     logits = logits / max(params.temperature, 1e-6)
-    logits = filter_logits(logits, params.top_k, params.top_p)
+    logits = filter_logits(logits, params.top_k, params.top_p, params.min_p)
     probs = torch.softmax(logits.float(), dim=-1)
     return torch.multinomial(probs, 1, generator=generator).squeeze(-1)
 

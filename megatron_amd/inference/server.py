@@ -36,6 +36,8 @@ def create_app(engine, tokenizer=None):
             greedy=bool(req.get("greedy", False)), return_log_probs=logprobs,
             top_n_logprobs=int(req.get("top_n_logprobs", 0)),
             repetition_penalty=float(req.get("repetition_penalty", 1.0)),
+            min_p=float(req.get("min_p", 0.0)),
+            logit_bias={int(k): float(v) for k, v in (req.get("logit_bias") or {}).items()} or None,
             stop_strings=tuple(req.get("stop_strings", ())),
             stop_on_eod=bool(req.get("stop_on_eod", True)),
             seed=req.get("seed"))
@@ -81,6 +83,8 @@ def create_app(engine, tokenizer=None):
             top_k=int(req.get("top_k", 0)), top_p=float(req.get("top_p", 0.0)),
             greedy=float(req.get("temperature", 1.0)) == 0.0,
             return_log_probs=bool(req.get("logprobs", False)),
+            min_p=float(req.get("min_p", 0.0)),
+            logit_bias={int(k): float(v) for k, v in (req.get("logit_bias") or {}).items()} or None,
             stop_strings=tuple(req.get("stop", []) if isinstance(req.get("stop", []), list)
                                else [req["stop"]]),
             seed=req.get("seed"))

@@ -832,3 +832,25 @@ def test_engine_abort_releases_blocks(tiny_model):
                                   enable_prefix_caching=False)
     expect = solo.generate([[8, 8, 4]], params)[0]
     assert done.output_tokens == expect.output_tokens
+
+
+def test_min_p_and_logit_bias_sampling():
+    """min_p masks tokens below the threshold fraction of the max prob;
+    logit_bias shifts chosen ids (OpenAI semantics)."""
+    from megatron_amd.inference.sampling import SamplingParams, filter_logits, sample
+
+    logits = torch.tensor([[4.0, 3.9, 0.0, -2.0]])
+    out = filter_logits(logits.clone(), min_p=0.5)
+    assert torch.isfinite(out[0, 0]) and torch.isfinite(out[0, 1])
+    assert not torch.isfinite(out[0, 2]) and not torch.isfinite(out[0, 3])
+
+    # a huge positive bias forces an otherwise-unlikely token even greedily
+    p = SamplingParams(greedy=True, logit_bias={3: 100.0})
+    tok = sample(logits.clone(), p)
+    assert int(tok[0]) == 3
+    # min_p + temperature sampling never picks a masked token
+    g = torch.Generator().manual_seed(0)
+    p = SamplingParams(temperature=1.0, min_p=0.5, seed=0)
+    for _ in range(20):
+        t = int(sample(logits.clone(), p, generator=g)[0])
+        assert t in (0, 1)
