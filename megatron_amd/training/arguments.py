@@ -29,6 +29,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--num-query-groups", type=int, default=None)
     g.add_argument("--kv-channels", type=int, default=None)
     g.add_argument("--vocab-size", type=int, default=128256)
+    g.add_argument("--make-vocab-size-divisible-by", type=int, default=None,
+                   help="pad the vocab so each TP shard is a multiple of this")
     g.add_argument("--max-position-embeddings", type=int, default=4096)
     g.add_argument("--normalization", choices=["rmsnorm", "layernorm"], default="rmsnorm")
     g.add_argument("--norm-epsilon", type=float, default=1e-5)
@@ -263,6 +265,11 @@ def validate_args(args) -> None:
     )
     args.num_microbatches = args.global_batch_size // gbs_div
     args.data_parallel_size = dp
+    if args.make_vocab_size_divisible_by:
+        from megatron_amd.tokenizers import pad_vocab_size
+
+        args.vocab_size = pad_vocab_size(args.vocab_size, args.tensor_model_parallel_size,
+                                         args.make_vocab_size_divisible_by)
     if args.fp16 and args.bf16:
         raise ValueError("choose one of --fp16 / --bf16")
     if args.sequence_parallel and args.tensor_model_parallel_size == 1:
