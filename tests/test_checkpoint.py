@@ -444,3 +444,47 @@ def test_finetune_load_weights_only(tmp_path):
         if moments is None and hasattr(sub, "segments"):
             moments = [seg.exp_avg for seg in sub.segments]
         assert moments and all(float(m.abs().sum()) == 0.0 for m in moments)
+
+
+def _dp2_async_save_case(rank, world, ckdir):
+    """World-2 async save: each rank's background writer process completes,
+    and the checkpoint resumes exactly."""
+    from megatron_amd.parallel.random import model_parallel_seed
+
+    G.initialize_model_parallel()
+    model_parallel_seed(77)
+    cfg = _cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000)
+    batches = _gen_batches(8, seed=7 + rank)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    first = _train(chunks, opt, cfg, batches, 2, 2)
+    writer = save_checkpoint(ckdir, chunks, opt, 2, async_save=True)
+    if writer is not None:
+        writer.join(timeout=120)
+        assert writer.exitcode == 0
+    import torch.distributed as dist
+
+    dist.barrier()
+
+    G.destroy_model_parallel()
+    G.initialize_model_parallel()
+    model_parallel_seed(123)
+    chunks2, opt2 = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    it = load_checkpoint(ckdir, chunks2, opt2)
+    assert it == 2
+    second = _train(chunks2, opt2, cfg, batches[4:], 2, 2)
+
+    G.destroy_model_parallel()
+    G.initialize_model_parallel()
+    model_parallel_seed(77)
+    chunks3, opt3 = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    ref = _train(chunks3, opt3, cfg, batches, 4, 2)
+    for a, b in zip(ref, first + second):
+        assert abs(a - b) < 1e-5, (rank, ref, first + second)
+
+
+def test_dp2_async_save_resume(tmp_path):
+    from tests.utils import spawn_dist
+
+    spawn_dist(_dp2_async_save_case, 2, str(tmp_path / "ck"))
