@@ -234,9 +234,26 @@ def get_batch_on_this_cp_rank(batch: dict, mode: str = "p2p") -> dict:
 # attention partials (native flash kernel on GPU, fp32 torch on CPU)
 # ---------------------------------------------------------------------------
 
-def _fwd_partial(q, k, v, causal: bool, scale: float):
+def _pair_mask(s, skv, device, causal, pos_off, window):
+    """[s, skv] bool valid-mask for one chunk pair: query i attends key j iff
+    0 <= (pos_off + i - j) (causal) and < window (if set).  pos_off is the
+    global q-minus-kv position offset of the chunk pair (skv - s for the
+    plain single-chunk case)."""
+    m = None
+    if causal:
+        m = torch.ones(s, skv, dtype=torch.bool, device=device).tril_(pos_off)
+    if window is not None and window > 0:
+        w = torch.ones(s, skv, dtype=torch.bool, device=device).triu_(pos_off - window + 1)
+        m = w if m is None else (m & w)
+    return m
+
+
+def _fwd_partial(q, k, v, causal: bool, scale: float, window=None, pos_off=None):
     """returns (out [s,b,hq,d] same dtype, lse [b,hq,s] fp32)."""
-    if ops.has_native() and q.is_cuda:
+    if pos_off is None:
+        pos_off = k.shape[0] - q.shape[0]
+    if (ops.has_native() and q.is_cuda and window is None
+            and pos_off == k.shape[0] - q.shape[0]):
         return ops._C.attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
                                causal, scale, 0)
     s, b, hq, d = q.shape
@@ -246,19 +263,23 @@ def _fwd_partial(q, k, v, causal: bool, scale: float):
     kf = k.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
     vf = v.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b,hq,s,skv]
-    if causal:
-        skv = k.shape[0]
-        mask = torch.ones(s, skv, dtype=torch.bool, device=q.device).tril_(skv - s)
+    mask = _pair_mask(s, k.shape[0], q.device, causal, pos_off, window)
+    if mask is not None:
         scores = scores.masked_fill(~mask, float("-inf"))
     lse = torch.logsumexp(scores, dim=-1)  # [b,hq,s]
     out = torch.matmul(torch.softmax(scores, dim=-1), vf)
+    out = torch.nan_to_num(out)  # fully-masked rows (out-of-window) -> 0
     return out.permute(2, 0, 1, 3).to(q.dtype), lse
 
 
-def _bwd_partial(dout, q, k, v, out, lse, causal: bool, scale: float):
+def _bwd_partial(dout, q, k, v, out, lse, causal: bool, scale: float,
+                 window=None, pos_off=None):
     """FA2-style manual backward of one partial (global out/lse): returns
     (dq, dk, dv) in fp32."""
-    if ops.has_native() and q.is_cuda:
+    if pos_off is None:
+        pos_off = k.shape[0] - q.shape[0]
+    if (ops.has_native() and q.is_cuda and window is None
+            and pos_off == k.shape[0] - q.shape[0]):
         dq, dk, dv = ops._C.attn_bwd(dout.contiguous(), q.contiguous(), k.contiguous(),
                                      v.contiguous(), out.contiguous(), lse.contiguous(),
                                      causal, scale, 0)
@@ -273,9 +294,8 @@ def _bwd_partial(dout, q, k, v, out, lse, causal: bool, scale: float):
     of = out.permute(1, 2, 0, 3).float()
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
     p = torch.exp(scores - lse.unsqueeze(-1))  # uses the GLOBAL lse
-    if causal:
-        skv = k.shape[0]
-        mask = torch.ones(s, skv, dtype=torch.bool, device=q.device).tril_(skv - s)
+    mask = _pair_mask(s, k.shape[0], q.device, causal, pos_off, window)
+    if mask is not None:
         p = p * mask
     dvf = torch.matmul(p.transpose(-1, -2), dof)
     dp = torch.matmul(dof, vf.transpose(-1, -2))
@@ -338,7 +358,7 @@ class _RingAttention(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, q, k, v, scale, group):
+    def forward(ctx, q, k, v, scale, group, window=None):
         cp, r, *_ = _ring_peers(group)
         L = q.shape[0] // 2
         dk_dim, dv_dim = k.shape[-1], v.shape[-1]  # may differ (MLA)
@@ -358,11 +378,15 @@ class _RingAttention(torch.autograd.Function):
                 for ki in range(2):
                     if my_chunks[qi] < src_chunks[ki]:
                         continue
+                    delta = (my_chunks[qi] - src_chunks[ki]) * L
+                    if window is not None and window > 0 and delta - (L - 1) >= window:
+                        continue  # whole pair outside the window
                     causal = my_chunks[qi] == src_chunks[ki]
                     qc = q[qi * L:(qi + 1) * L]
                     kc = kv[ki * L:(ki + 1) * L, ..., :dk_dim]
                     vc = kv[ki * L:(ki + 1) * L, ..., dk_dim:]
-                    oc, lsec = _fwd_partial(qc, kc, vc, causal, scale)
+                    oc, lsec = _fwd_partial(qc, kc, vc, True, scale,
+                                            window=window, pos_off=delta)
                     sl = slice(qi * L, (qi + 1) * L)
                     o_m, lse_m = _merge(o[sl], lse[..., sl], oc, lsec)
                     o[sl] = o_m
@@ -372,13 +396,13 @@ class _RingAttention(torch.autograd.Function):
                 kv = kv_next
         out = o.to(q.dtype)
         ctx.save_for_backward(q, k, v, out, lse)
-        ctx.scale, ctx.group = scale, group
+        ctx.scale, ctx.group, ctx.window = scale, group, window
         return out
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        scale, group = ctx.scale, ctx.group
+        scale, group, window = ctx.scale, ctx.group, ctx.window
         cp, r, *_ = _ring_peers(group)
         L = q.shape[0] // 2
         dk_dim = k.shape[-1]
@@ -395,12 +419,15 @@ class _RingAttention(torch.autograd.Function):
                 for ki in range(2):
                     if my_chunks[qi] < src_chunks[ki]:
                         continue
-                    causal = my_chunks[qi] == src_chunks[ki]
+                    delta = (my_chunks[qi] - src_chunks[ki]) * L
+                    if window is not None and window > 0 and delta - (L - 1) >= window:
+                        continue
                     sl = slice(qi * L, (qi + 1) * L)
                     kl = slice(ki * L, (ki + 1) * L)
                     dqc, dkc, dvc = _bwd_partial(
                         dout[sl], q[sl], kv[kl, ..., :dk_dim], kv[kl, ..., dk_dim:],
-                        out[sl], lse[..., sl], causal, scale)
+                        out[sl], lse[..., sl], True, scale,
+                        window=window, pos_off=delta)
                     dq[sl] += dqc
                     dkv[kl, ..., :dk_dim] += dkc
                     dkv[kl, ..., dk_dim:] += dvc
@@ -411,20 +438,22 @@ class _RingAttention(torch.autograd.Function):
         if cp > 1:
             dkv = _ring_sendrecv(dkv, group)
         return (dq.to(q.dtype), dkv[..., :dk_dim].to(q.dtype),
-                dkv[..., dk_dim:].to(q.dtype), None, None)
+                dkv[..., dk_dim:].to(q.dtype), None, None, None)
 
 
-def ring_attention(q, k, v, scale: Optional[float] = None, group=None):
-    """q [2L,b,hq,d], k/v [2L,b,hkv,d] in CP 2-chunk layout -> out [2L,b,hq,d]."""
+def ring_attention(q, k, v, scale: Optional[float] = None, group=None, window=None):
+    """q [2L,b,hq,d], k/v [2L,b,hkv,d] in CP 2-chunk layout -> out [2L,b,hq,d].
+    `window`: causal sliding window in GLOBAL positions (masking is applied
+    per chunk pair with the pair's position offset)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     group = group if group is not None else G.get_context_parallel_group()
     if group is None or dist.get_world_size(group) == 1:
-        return ops.flash_attention(q, k, v, causal=True, scale=scale)
-    return _RingAttention.apply(q, k, v, scale, group)
+        return ops.flash_attention(q, k, v, causal=True, scale=scale, window=window)
+    return _RingAttention.apply(q, k, v, scale, group, window)
 
 
-def ulysses_attention(q, k, v, scale: Optional[float] = None, group=None):
+def ulysses_attention(q, k, v, scale: Optional[float] = None, group=None, window=None):
     """Ulysses a2a CP: scatter heads / gather sequence around full attention.
 
     q [s/cp, b, hq, d] (contiguous slicing) -> out [s/cp, b, hq, d]. Needs
@@ -435,7 +464,7 @@ def ulysses_attention(q, k, v, scale: Optional[float] = None, group=None):
     group = group if group is not None else G.get_context_parallel_group()
     cp = dist.get_world_size(group) if group is not None else 1
     if cp == 1:
-        return ops.flash_attention(q, k, v, causal=True, scale=scale)
+        return ops.flash_attention(q, k, v, causal=True, scale=scale, window=window)
     sl, b, hq, d = q.shape
     hkv = k.shape[2]
     assert hq % cp == 0 and hkv % cp == 0, "Ulysses needs heads divisible by cp"
@@ -454,9 +483,9 @@ def ulysses_attention(q, k, v, scale: Optional[float] = None, group=None):
 
     qh, kh, vh = sp2hp(q), sp2hp(k), sp2hp(v)
     if v.shape[-1] == d and d in (64, 128) and q.is_cuda:
-        oh = ops.flash_attention(qh, kh, vh, causal=True, scale=scale)
+        oh = ops.flash_attention(qh, kh, vh, causal=True, scale=scale, window=window)
     else:
         from megatron_amd.ops import reference as _ref
 
-        oh = _ref.attention(qh, kh, vh, causal=True, scale=scale)
+        oh = _ref.attention(qh, kh, vh, causal=True, scale=scale, window=window)
     return hp2sp(oh)

@@ -127,11 +127,12 @@ class SelfAttention(nn.Module):
         elif G.get_context_parallel_world_size() > 1:
             from megatron_amd.parallel.context_parallel import ring_attention, ulysses_attention
 
-            assert self.window is None, "sliding window not supported under CP yet"
             if self.config.cp_comm_type == "a2a":
-                core_out = ulysses_attention(q, k, v, scale=self.softmax_scale)
+                core_out = ulysses_attention(q, k, v, scale=self.softmax_scale,
+                                             window=self.window)
             else:
-                core_out = ring_attention(q, k, v, scale=self.softmax_scale)
+                core_out = ring_attention(q, k, v, scale=self.softmax_scale,
+                                          window=self.window)
         elif (self.config.recompute_granularity == "selective" and self.training
               and "core_attn" in (self.config.recompute_modules or ["core_attn"])):
             # selective recompute: checkpoint only the core-attention region
