@@ -41,6 +41,10 @@ class ParallelConfig:
     mtp_loss_scaling_factor: float = 0.1
     expert_parallel_size: int = 1
     virtual_pipeline_parallel_size: Optional[int] = None
+    # uneven PP splits (reference --decoder-first/last-pipeline-num-layers):
+    # give the embedding/loss stages fewer transformer layers
+    num_layers_in_first_pipeline_stage: Optional[int] = None
+    num_layers_in_last_pipeline_stage: Optional[int] = None
     sequence_parallel: bool = False
     fp8: object = None  # None | 'hybrid' | 'e4m3' (K15)
     fp8_amax_history_len: int = 16

@@ -105,6 +105,10 @@ def build_arg_parser() -> argparse.ArgumentParser:
     g.add_argument("--virtual-pipeline-model-parallel-size", "--vpp", type=int, default=None)
     g.add_argument("--context-parallel-size", "--cp", type=int, default=1)
     g.add_argument("--cp-comm-type", choices=["p2p", "a2a"], default="p2p")
+    g.add_argument("--decoder-first-pipeline-num-layers", type=int, default=None,
+                   dest="num_layers_in_first_pipeline_stage")
+    g.add_argument("--decoder-last-pipeline-num-layers", type=int, default=None,
+                   dest="num_layers_in_last_pipeline_stage")
     g.add_argument("--expert-model-parallel-size", "--ep", type=int, default=1)
     g.add_argument("--expert-tensor-parallel-size", type=int, default=None)
     g.add_argument("--sequence-parallel", action="store_true")
@@ -356,6 +360,8 @@ def configs_from_args(args):
         virtual_pipeline_parallel_size=args.virtual_pipeline_model_parallel_size,
         context_parallel_size=args.context_parallel_size,
         cp_comm_type=args.cp_comm_type,
+        num_layers_in_first_pipeline_stage=args.num_layers_in_first_pipeline_stage,
+        num_layers_in_last_pipeline_stage=args.num_layers_in_last_pipeline_stage,
         expert_parallel_size=args.expert_model_parallel_size,
         expert_tensor_parallel_size=args.expert_tensor_parallel_size,
         sequence_parallel=args.sequence_parallel,

@@ -119,10 +119,43 @@ class TransformerLayer(nn.Module):
         return y
 
 
+def _uneven_split(config):
+    """Per-stage layer counts for uneven first/last splits, or None."""
+    first = config.num_layers_in_first_pipeline_stage
+    last = config.num_layers_in_last_pipeline_stage
+    if first is None and last is None:
+        return None
+    pp = config.pipeline_parallel_size
+    assert config.virtual_pipeline_parallel_size in (None, 1), (
+        "uneven first/last stage splits are non-interleaved (vpp=1) in v1")
+    L = config.num_layers
+    mid_stages = pp - (first is not None) - (last is not None)
+    mid_layers = L - (first or 0) - (last or 0)
+    assert mid_stages >= 0 and mid_layers >= 0
+    assert mid_stages == 0 or mid_layers % mid_stages == 0, (
+        f"{mid_layers} middle layers must divide {mid_stages} middle stages")
+    per_mid = mid_layers // mid_stages if mid_stages else 0
+    counts = []
+    for r in range(pp):
+        if r == 0 and first is not None:
+            counts.append(first)
+        elif r == pp - 1 and last is not None:
+            counts.append(last)
+        else:
+            counts.append(per_mid)
+    assert sum(counts) == L, (counts, L)
+    return counts
+
+
 def get_num_layers_to_build(config) -> int:
-    """Per-pipeline-stage layer count (reference transformer_block.py:71)."""
+    """Per-pipeline-stage layer count (reference transformer_block.py:71;
+    uneven first/last splits via num_layers_in_first/last_pipeline_stage)."""
     pp = config.pipeline_parallel_size
     vpp = config.virtual_pipeline_parallel_size
+    counts = _uneven_split(config)
+    if counts is not None:
+        r = G.get_pipeline_model_parallel_rank() if G.grid_initialized() else 0
+        return counts[r]
     chunks = pp * (vpp or 1)
     assert config.num_layers % chunks == 0, (
         f"num_layers {config.num_layers} must divide pp*vpp = {chunks}"
@@ -136,6 +169,9 @@ def get_layer_offset(config, vp_stage: Optional[int] = None) -> int:
     pp_rank = G.get_pipeline_model_parallel_rank() if G.grid_initialized() else 0
     pp = config.pipeline_parallel_size
     vpp = config.virtual_pipeline_parallel_size
+    counts = _uneven_split(config)
+    if counts is not None:
+        return sum(counts[:pp_rank])
     per_chunk = get_num_layers_to_build(config)
     if vpp is not None and vp_stage is not None:
         # interleaved: chunk c on pp rank r holds layers [ (c*pp + r) * per_chunk , ... )

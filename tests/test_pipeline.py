@@ -349,3 +349,27 @@ def test_pp4_combined_matches_single(tmp_path, monkeypatch):
     got = json.load(open(out))
     for a, b in zip(ref, got):
         assert abs(a - b) < 2e-4, (ref, got)
+
+
+def _pp2_uneven_case(rank, world):
+    G.initialize_model_parallel(pipeline_parallel_size=world)
+    model_parallel_seed(1234)
+    cfg = _cfg(pp=world).replace(num_layers_in_first_pipeline_stage=1,
+                                 num_layers_in_last_pipeline_stage=3)
+    losses = _run(cfg, 2, 4, _gen_batches(8))
+    if G.get_grid().is_pipeline_last_stage(ignore_virtual=True):
+        with open(os.environ["PP_TEST_OUT"], "w") as f:
+            json.dump(losses, f)
+
+
+def test_pp2_uneven_stage_split_matches_single(tmp_path, monkeypatch):
+    """Uneven stage layout (1 layer on the embedding stage, 3 on the loss
+    stage — reference --decoder-first/last-pipeline-num-layers) trains
+    identically to the single-rank run."""
+    out = tmp_path / "ppu.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+    ref = _single_reference()
+    spawn_dist(_pp2_uneven_case, 2)
+    got = json.load(open(out))
+    for a, b in zip(ref, got):
+        assert abs(a - b) < 2e-4, (ref, got)
