@@ -469,6 +469,22 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkv_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// v4 dkv design (round-3 target; see profiles/README.md measurements):
+//   The v3 combined 32-row attempt spills (2 accumulator sets = 128 VGPRs
+//   minimum is fine, but the A/B operand sets on top are not).  Plan:
+//   phase 1 unchanged (8 waves x 16 kv rows compute S^T/dP^T per 64-q tile);
+//   then write packed bf16 P/dS tiles to LDS (64q x 128kv x 2 = 32 KB) and
+//   re-block PHASE 2 by (32-kv-row group, 64-d half): each wave keeps the
+//   same 64 accumulator VGPRs but every Qt/dOt B-read now feeds 2 MFMAs
+//   (two kv groups) and every P/dS A-read feeds 4 (ND=4 d-columns), cutting
+//   LDS bytes/MFMA from 16 to ~12 and deleting the permlane relayout chain.
+//   Cost: LDS rises 64->96 KB so occupancy drops 2->1 block/CU; whether the
+//   25% traffic cut beats the lost latency hiding must be MEASURED — if it
+//   loses, try folding Qt/dOt into the P/dS store instead (transpose during
+//   the phase-1->2 handoff) to stay at 2 blocks.
+// ---------------------------------------------------------------------------
+
+// ---------------------------------------------------------------------------
 // v3 kernels: 2x register blocking (32 output rows per wave, 256 per block).
 // Same technique stack as v2 plus the fwd-v2 lesson: every LDS A/B-operand
 // read now feeds 2 MFMAs instead of 1 (the kernels were LDS-read-bound, so
