@@ -280,3 +280,15 @@ def test_evaluate_ppl_and_cloze():
     assert evaluate_cloze(model, samples)["accuracy"] == 1.0
     wrong = [s[:-1] + [(s[-1] + 1) % 64] for s in samples]
     assert evaluate_cloze(model, wrong)["accuracy"] == 0.0
+
+
+def test_tools_import_surface():
+    """Every CLI tool parses/imports (no syntax or import rot)."""
+    import importlib
+    import sys
+
+    sys.path.insert(0, ".")
+    for mod in ["tools.evaluate", "tools.preprocess_data", "tools.profile_model",
+                "tools.bench_inference", "tools.run_text_generation_server",
+                "tools.checkpoint.convert_hf", "tools.checkpoint.export_hf"]:
+        importlib.import_module(mod)
