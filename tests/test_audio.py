@@ -52,3 +52,16 @@ def test_audio_model_text_only_path():
     ids = torch.randint(1, 128, (2, 10))
     loss = m(None, ids, labels=ids)
     assert loss.shape == (10, 2)
+
+
+def test_pretrain_audio_runs():
+    import pretrain_audio as A
+    from megatron_amd.training.pretrain import pretrain
+
+    it = pretrain(A.model_provider, [
+        "--num-layers", "2", "--hidden-size", "64", "--num-attention-heads", "4",
+        "--num-query-groups", "2", "--ffn-hidden-size", "128", "--seq-length", "32",
+        "--micro-batch-size", "2", "--global-batch-size", "4", "--vocab-size", "128",
+        "--mock-data", "--train-iters", "2", "--log-interval", "0", "--seed", "11",
+    ], forward_step_builder=A.forward_step_builder)
+    assert it == 2
