@@ -366,3 +366,17 @@ def test_gpt_cp2_sliding_window_matches_single():
     for mode in ("p2p", "a2a"):
         spawn_dist(_run_gpt_cp_window, 2, cfg_kwargs, tokens, labels, loss_ref,
                    grad_ref, mode)
+
+
+def test_gpt_cp4_matches_single():
+    """CP=4 ring (3 rotation hops, 8 zigzag chunks) and Ulysses at world 4
+    equal the single run — the multi-hop path CP=2 never exercises."""
+    cfg_kwargs = dict(num_layers=2, hidden_size=64, num_attention_heads=4,
+                      num_query_groups=4, ffn_hidden_size=128, vocab_size=128,
+                      max_position_embeddings=128)
+    g = torch.Generator().manual_seed(23)
+    tokens = torch.randint(0, 128, (2, 64), generator=g)
+    labels = torch.randint(0, 128, (2, 64), generator=g)
+    loss_ref, grad_ref = _loss_single(cfg_kwargs, tokens, labels)
+    for mode in ("p2p", "a2a"):
+        spawn_dist(_run_gpt_cp, 4, cfg_kwargs, tokens, labels, loss_ref, grad_ref, mode)
