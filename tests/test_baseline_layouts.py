@@ -115,3 +115,47 @@ def test_baseline_layout_matches_single(tmp_path, monkeypatch, name, ref_kw):
     spawn_dist(_layout_case, 8, name, ckpt)
     got = json.load(open(out))[0]
     assert abs(got - ref_loss) < 5e-4, (name, got, ref_loss)
+
+
+def _tp2cp2_case(rank, world, ckpt_dir):
+    from megatron_amd.parallel.context_parallel import slice_for_cp_rank
+
+    G.initialize_model_parallel(tensor_parallel_size=2, context_parallel_size=2)
+    model_parallel_seed(1234)
+    cfg = _cfg(tensor_parallel_size=2, context_parallel_size=2)
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    load_checkpoint(ckpt_dir, chunks, opt, load_rng=False)
+    grid = G.get_grid()
+    cp_rank = grid.rank_in("cp")
+
+    def fwd(it, model):
+        batch = next(it)
+        t = slice_for_cp_rank(batch["tokens"], cp_rank, 2, seq_dim=1, mode="p2p")
+        l = slice_for_cp_rank(batch["labels"], cp_rank, 2, seq_dim=1, mode="p2p")
+
+        def loss_func(loss_sb):
+            s = loss_sb.sum()
+            return s, torch.tensor(loss_sb.numel()), {"loss_sum": s.detach()}
+
+        return model(t, labels=l), loss_func
+
+    batches = _gen_batches(4)
+    r = train_step(fwd, [iter(batches)], chunks, opt, cfg, 4, SEQ, 2)
+    if grid.rank_in("tp") == 0 and grid.rank_in("cp") == 0:
+        with open(os.environ["LAYOUT_TEST_OUT"], "w") as f:
+            json.dump([r["lm_loss"]], f)
+
+
+def test_tp2_cp2_composition_matches_single(tmp_path, monkeypatch):
+    """TP=2 x CP=2 (world 4): head-sharded ring attention over sequence
+    shards, resharded from a single-process checkpoint, reproduces the
+    single-process first-step loss."""
+    out = tmp_path / "tc.json"
+    ckpt = str(tmp_path / "ckpt")
+    monkeypatch.setenv("LAYOUT_TEST_OUT", str(out))
+    ref_loss = _single_ref({}, ckpt)
+    spawn_dist(_tp2cp2_case, 4, ckpt)
+    got = json.load(open(out))[0]
+    assert abs(got - ref_loss) < 2e-3, (got, ref_loss)
