@@ -326,3 +326,47 @@ def test_dp4_dist_opt_two_instances_matches_single(tmp_path, monkeypatch):
     got = json.load(open(out))
     for a, b in zip(ref, got):
         assert abs(a - b) < 1e-4, (ref, got)
+
+
+def _dp4_instance_resume_case(rank, world, ckdir):
+    """Save at 1 optimizer instance, resume at 2: the sharded optimizer
+    atlas reshards across instance layouts."""
+    import json, os
+
+    from megatron_amd.checkpoint.checkpointing import load_checkpoint, save_checkpoint
+
+    G.initialize_model_parallel()
+    model_parallel_seed(77)
+    cfg = _tiny_cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, weight_decay=0.01, clip_grad=1.0,
+                              use_distributed_optimizer=True)
+    mk = lambda inst: DDPConfig(grad_reduce_in_fp32=True, use_distributed_optimizer=True,
+                                bucket_size=10_000,
+                                num_distributed_optimizer_instances=inst)
+    all_batches = _gen_batches(8)
+    mine = [all_batches[i] for i in range(len(all_batches)) if i % world == rank]
+
+    # reference: 1 instance straight through
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, mk(1))
+    ref = _run_steps(chunks, opt, cfg, mine, 2, 1)
+
+    # save at 1 instance after step 1
+    G.destroy_model_parallel(); G.initialize_model_parallel()
+    model_parallel_seed(77)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, mk(1))
+    first = _run_steps(chunks, opt, cfg, mine[:1], 1, 1)
+    save_checkpoint(ckdir, chunks, opt, 1)
+
+    # resume at 2 instances
+    G.destroy_model_parallel(); G.initialize_model_parallel()
+    model_parallel_seed(123)
+    chunks2, opt2 = setup_model_and_optimizer(_provider, cfg, opt_cfg, mk(2))
+    it = load_checkpoint(ckdir, chunks2, opt2)
+    assert it == 1
+    second = _run_steps(chunks2, opt2, cfg, mine[1:], 1, 1)
+    for a, b in zip(ref, first + second):
+        assert abs(a - b) < 1e-5, (rank, ref, first + second)
+
+
+def test_dp4_dist_opt_instance_reshard_resume(tmp_path):
+    spawn_dist(_dp4_instance_resume_case, 4, str(tmp_path / "ck"))
