@@ -119,6 +119,8 @@ class TransformerConfig(ParallelConfig):
     # aux-loss-free balancing (reference finalize_model_grads.py:334): after
     # each step, expert_bias += rate * sign(mean_load - expert_load).
     moe_router_enable_expert_bias: bool = False
+    # benchmark/debug: force a perfectly balanced round-robin routing
+    moe_router_force_load_balancing: bool = False
     moe_router_bias_update_rate: float = 1e-3
     # MuonClip: clip attention-logit growth by rescaling q/k weights after
     # each step (reference optimizer/qk_clip.py); None disables tracking.

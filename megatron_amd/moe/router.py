@@ -133,6 +133,22 @@ class TopKRouter(nn.Module):
             probs = torch.softmax(top_logits, dim=-1)
             full_probs = torch.softmax(logits, dim=-1)
 
+        if getattr(self.config, "moe_router_force_load_balancing", False):
+            # benchmark/debug mode (reference force_load_balancing): override
+            # the selection with a perfectly balanced round-robin assignment;
+            # probs still come from the real gate so the numerics pipeline
+            # (weighting, grads through probs) stays representative.
+            T = logits.shape[0]
+            k = self.topk
+            base = torch.arange(T * k, device=logits.device) % self.num_experts
+            top_idx = base.view(T, k)
+            dup = top_idx[:, :1] == top_idx[:, 1:]  # avoid same expert twice
+            if dup.any():
+                top_idx = top_idx.clone()
+                top_idx[:, 1:][dup] = (top_idx[:, 1:][dup] + 1) % self.num_experts
+            probs = full_probs.gather(-1, top_idx)
+            probs = probs / probs.sum(dim=-1, keepdim=True).clamp(min=1e-20)
+
         routing_map = torch.zeros_like(logits).scatter_(1, top_idx, 1.0)
         with torch.no_grad():
             self.local_tokens_per_expert = routing_map.sum(dim=0)

@@ -84,6 +84,7 @@ def build_arg_parser() -> argparse.ArgumentParser:
             pat = ast.literal_eval(v)  # e.g. "[0]*1+[1]*3" forms are not allowed; pass a plain list
             assert isinstance(pat, (list, tuple)) and all(x in (0, 1) for x in pat)
             return list(pat)
+    g.add_argument("--moe-router-force-load-balancing", action="store_true", default=False)
     g.add_argument("--moe-layer-freq", type=_moe_freq, default=1,
                    help="int N (every Nth layer is MoE) or a 0/1 list like [0,0,1,1]")
     g.add_argument("--moe-router-pre-softmax", action="store_true", default=False)
@@ -340,6 +341,7 @@ def configs_from_args(args):
         moe_shared_expert_intermediate_size=args.moe_shared_expert_intermediate_size,
         moe_token_dispatcher_type=args.moe_token_dispatcher_type,
         moe_layer_freq=args.moe_layer_freq,
+        moe_router_force_load_balancing=args.moe_router_force_load_balancing,
         moe_router_pre_softmax=args.moe_router_pre_softmax,
         moe_router_num_groups=args.moe_router_num_groups,
         moe_router_group_topk=args.moe_router_group_topk,

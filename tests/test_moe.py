@@ -569,3 +569,26 @@ def test_moe_layer_pattern_list():
     loss = m(tokens, labels=tokens)
     loss.sum().backward()
     assert m.decoder.layers[2].mlp.router.weight.grad is not None
+
+
+def test_router_force_load_balancing_uniform_counts():
+    """Force-balanced routing: every expert receives T*topk/E tokens exactly;
+    probs still flow from the real gate (grads reach the router weight)."""
+    from megatron_amd.config import TransformerConfig
+    from megatron_amd.moe.router import TopKRouter
+    from tests.utils import init_single
+
+    init_single()
+    torch.manual_seed(0)
+    cfg = TransformerConfig(num_layers=1, hidden_size=32, num_attention_heads=4,
+                            num_query_groups=4, vocab_size=64, ffn_hidden_size=32,
+                            num_experts=4, moe_router_topk=2, moe_ffn_hidden_size=16,
+                            moe_router_force_load_balancing=True)
+    r = TopKRouter(cfg)
+    hidden = torch.randn(16, 32, requires_grad=True)
+    probs, idx = r(hidden)
+    counts = torch.bincount(idx.reshape(-1), minlength=4)
+    assert torch.all(counts == 8), counts  # 16*2/4
+    assert torch.all(idx[:, 0] != idx[:, 1])
+    probs.sum().backward()
+    assert r.weight.grad is not None
