@@ -64,6 +64,8 @@ def parse_args():
     p.add_argument("--tp", type=int, default=None, help="tensor parallel size (default: model-dependent)")
     p.add_argument("--pp", type=int, default=1)
     p.add_argument("--ep", type=int, default=None, help="expert parallel size (default: world for MoE)")
+    p.add_argument("--balanced-routing", action="store_true",
+                   help="force round-robin balanced MoE routing (stable tok/s, uniform a2a)")
     p.add_argument("--vpp", type=int, default=None)
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--no-dist-opt", action="store_true")
@@ -169,6 +171,7 @@ def main():
         sequence_parallel=sp,
         expert_parallel_size=ep,
         recompute_granularity="full" if args.recompute else None,
+        moe_router_force_load_balancing=args.balanced_routing,
         gradient_accumulation_fusion=device.type == "cuda",
         fp8="hybrid" if args.fp8 else None,
     )
