@@ -292,3 +292,20 @@ def test_tools_import_surface():
                 "tools.bench_inference", "tools.run_text_generation_server",
                 "tools.checkpoint.convert_hf", "tools.checkpoint.export_hf"]:
         importlib.import_module(mod)
+
+
+def test_bench_model_catalog_shapes():
+    """bench.py model catalog matches the BASELINE architectures (llama3-8b
+    / llama3-70b dims, mixtral expert count) and parses its flags."""
+    import importlib
+    import sys
+
+    sys.path.insert(0, ".")
+    bench = importlib.import_module("bench")
+    m8 = bench.MODELS["llama3-8b"]
+    assert (m8["num_layers"], m8["hidden_size"], m8["ffn_hidden_size"]) == (32, 4096, 14336)
+    assert m8["vocab_size"] == 128256
+    m70 = bench.MODELS["llama3-70b"]
+    assert (m70["num_layers"], m70["hidden_size"]) == (80, 8192)
+    mx = bench.MODELS["mixtral-8x7b"]
+    assert mx["num_experts"] == 8 and mx["moe_router_topk"] == 2
