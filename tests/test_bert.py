@@ -116,3 +116,36 @@ def test_bert_binary_head_and_pooler():
     total.backward()
     assert m.pooler.weight.grad is not None
     assert m.binary_head.weight.grad is not None
+
+
+def test_bert_dataset_sentence_pairs_nsp():
+    """sentence_pairs mode: tokentype marks segment B, is_next=0 iff the
+    second half was swapped in from another sample."""
+    import numpy as np
+
+    from megatron_amd.datasets.bert_dataset import BertMaskedDataset
+
+    class Base:
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            return {"tokens": torch.full((16,), i + 1, dtype=torch.long)}
+
+    ds = BertMaskedDataset(Base(), vocab_size=64, mask_id=63, seed=3,
+                           sentence_pairs=True)
+    seen = {0: 0, 1: 0}
+    for i in range(8):
+        s = ds[i]
+        assert torch.all(s["tokentype_ids"][:8] == 0)
+        assert torch.all(s["tokentype_ids"][8:] == 1)
+        label = int(s["is_next"])
+        seen[label] += 1
+        # unmasked positions of segment B reveal the source sample
+        tail = s["tokens"][8:]
+        clean = tail[(tail != 63) & (tail < 9)]
+        if label == 1 and clean.numel():
+            assert torch.all(clean == i + 1)
+        if label == 0 and clean.numel():
+            assert torch.all(clean != i + 1)
+    assert seen[0] > 0 and seen[1] > 0
