@@ -275,3 +275,30 @@ def test_tiny_t5_relative_bias_golden():
     finally:
         P.train_step = orig
     _check_or_record("tiny_t5_relative.json", losses)
+
+
+def test_tiny_bert_nsp_golden():
+    """pretrain_bert's joint MLM+NSP recipe (binary head active) pinned."""
+    import pretrain_bert as B
+
+    losses = []
+    orig = P.train_step
+
+    def wrapped(*a, **k):
+        r = orig(*a, **k)
+        losses.append(r["lm_loss"])
+        return r
+
+    P.train_step = wrapped
+    try:
+        P.pretrain(B.model_provider, [
+            "--num-layers", "2", "--hidden-size", "64", "--num-attention-heads", "4",
+            "--num-query-groups", "2", "--ffn-hidden-size", "128", "--seq-length", "64",
+            "--micro-batch-size", "2", "--global-batch-size", "4", "--vocab-size", "256",
+            "--mock-data", "--train-iters", "8", "--log-interval", "0", "--seed", "42",
+            "--deterministic-mode", "--position-embedding-type", "learned",
+            "--normalization", "layernorm", "--activation", "gelu",
+        ], forward_step_builder=B.forward_step_builder)
+    finally:
+        P.train_step = orig
+    _check_or_record("tiny_bert_nsp.json", losses)
