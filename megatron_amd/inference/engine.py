@@ -90,6 +90,8 @@ class GenerationResult:
     text: Optional[str] = None
     finished: bool = False
     aborted: bool = False
+    # log p(prompt[i] | prompt[:i]) for i >= 1 (prompt_logprobs=True)
+    prompt_log_probs: Optional[List[float]] = None
 
 
 def _full_logits(logits_tp: torch.Tensor) -> torch.Tensor:
@@ -179,6 +181,14 @@ class StaticInferenceEngine:
             logits = None
 
         results = [GenerationResult(i, p) for i, p in enumerate(prompts)]
+        if params.prompt_logprobs and pp_last:
+            # log p(prompt[i] | prompt[:i]): logits row i-1 scores token i
+            full = _full_logits(logits_tp).float().log_softmax(dim=-1)  # [L, b, V]
+            for i, p in enumerate(prompts):
+                ids = torch.as_tensor(p[1:], device=self.device)
+                rows = full[: len(p) - 1, i]
+                results[i].prompt_log_probs = rows.gather(
+                    -1, ids.unsqueeze(-1)).squeeze(-1).tolist()
         finished = torch.zeros(b, dtype=torch.bool, device=self.device)
         for _ in range(params.max_tokens):
             if pp_last:

@@ -31,6 +31,8 @@ class SamplingParams:
     min_p: float = 0.0
     # additive per-token-id logit bias {token_id: bias} (OpenAI logit_bias)
     logit_bias: Optional[dict] = None
+    # also return log p(prompt[i] | prompt[:i]) from prefill (RL / eval scoring)
+    prompt_logprobs: bool = False
 
 
 def filter_logits(logits: torch.Tensor, top_k: int = 0, top_p: float = 0.0,

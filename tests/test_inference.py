@@ -854,3 +854,21 @@ def test_min_p_and_logit_bias_sampling():
     for _ in range(20):
         t = int(sample(logits.clone(), p, generator=g)[0])
         assert t in (0, 1)
+
+
+def test_prompt_logprobs_match_manual(tiny_model):
+    """prompt_logprobs: per-position log p(prompt[i] | prefix) equals a
+    manual forward's log-softmax."""
+    eng = StaticInferenceEngine(tiny_model, max_batch=2, max_seq=64)
+    prompts = [[3, 7, 11, 2, 9], [5, 1, 4]]
+    params = SamplingParams(max_tokens=2, greedy=True, stop_on_eod=False,
+                            prompt_logprobs=True)
+    res = eng.generate(prompts, params)
+    for p, r in zip(prompts, res):
+        with torch.no_grad():
+            logits = tiny_model(torch.tensor([p]))  # [s, 1, V]
+        lp = logits[:, 0].float().log_softmax(-1)
+        want = [float(lp[i - 1, p[i]]) for i in range(1, len(p))]
+        assert len(r.prompt_log_probs) == len(p) - 1
+        for a, b in zip(r.prompt_log_probs, want):
+            assert abs(a - b) < 1e-4, (a, b)
