@@ -333,7 +333,8 @@ def forward_backward_pipelining_without_interleaving_combined(
             plan, tokens = make_plan(next(data_iterator))
             x = tokens if is_first else input_tensor
             if forward_only:
-                fwd_whole(plan, x)
+                out = fwd_whole(plan, x)
+                comm.send_forward(out, is_last)
                 if i < num_steady - 1:
                     input_tensor = comm.recv_forward(is_first)
                 continue

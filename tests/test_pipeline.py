@@ -373,3 +373,49 @@ def test_pp2_uneven_stage_split_matches_single(tmp_path, monkeypatch):
     got = json.load(open(out))
     for a, b in zip(ref, got):
         assert abs(a - b) < 2e-4, (ref, got)
+
+
+def _pp2_combined_fwd_only_case(rank, world):
+    """forward_only through the combined pp schedule (eval path) must relay
+    activations downstream and produce the same losses as training-forward."""
+    from megatron_amd.pipeline.schedules import get_forward_backward_func
+
+    G.initialize_model_parallel(pipeline_parallel_size=world)
+    model_parallel_seed(1234)
+    cfg = _cfg_moe(pp=world, combined=True)
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    for c in chunks:
+        _fill_deterministic(c)
+    fb = get_forward_backward_func(cfg)
+    batches = _gen_batches(4)
+    losses, ntok = fb(forward_step_func=forward_step, data_iterator=[iter(batches)],
+                      model=chunks, num_microbatches=4, seq_length=SEQ,
+                      micro_batch_size=2, forward_only=True)
+    if G.get_grid().is_pipeline_last_stage(ignore_virtual=True):
+        total = sum(float(x["loss_sum"]) for x in losses) / max(int(ntok), 1)
+        with open(os.environ["PP_TEST_OUT"], "w") as f:
+            json.dump([total], f)
+
+
+def test_pp2_combined_forward_only(tmp_path, monkeypatch):
+    out = tmp_path / "ppfo.json"
+    monkeypatch.setenv("PP_TEST_OUT", str(out))
+    init_single()
+    cfg = _cfg_moe()
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    for c in chunks:
+        _fill_deterministic(c)
+    batches = _gen_batches(4)
+    total = n = 0.0
+    for mb in batches:
+        with torch.no_grad():
+            loss = chunks[0](mb["tokens"], labels=mb["labels"])
+        total += float(loss.sum()); n += loss.numel()
+    ref = total / n
+    spawn_dist(_pp2_combined_fwd_only_case, 2)
+    got = json.load(open(out))[0]
+    assert abs(got - ref) < 2e-4, (got, ref)
