@@ -236,6 +236,8 @@ class StaticInferenceEngine:
         entries are left in place and overwritten by the next chunk at the
         same offsets (the static cache masks by tracked length)."""
         assert params.greedy, "speculative decoding is exact for greedy only"
+        assert not params.stop_strings and not params.return_log_probs, (
+            "speculative decode: stop_strings/logprobs unsupported — use generate()")
         if not G.grid_initialized() or G.get_grid().pp == 1:
             pass
         else:
