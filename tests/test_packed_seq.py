@@ -94,3 +94,30 @@ def test_pack_sequences_fuzz_conservation():
             assert cu[0] == 0 and int(cu[-1]) <= L
             assert (cu[1:] > cu[:-1]).all() or int(cu[-1]) == L
             assert r["tokens"].shape == (L,)
+
+
+def test_eod_boundaries_fuzz_roundtrip():
+    """Fuzz: random EOD layouts -> cu boundaries are sorted, end at seq_len,
+    split only after EODs, and PackedSeqParams positions restart per doc."""
+    import numpy as np
+
+    from megatron_amd.datasets.gpt_dataset import eod_boundaries
+    from megatron_amd.transformer.packed_seq import PackedSeqParams
+
+    rng = np.random.RandomState(0)
+    for trial in range(50):
+        s = int(rng.randint(4, 64))
+        eod = 0
+        toks = torch.from_numpy(rng.randint(0, 5, size=s)).long()
+        cu = eod_boundaries(toks, eod, s, max_docs=16)
+        vals = [int(x) for x in cu if x > 0]
+        assert vals == sorted(set(vals)) and vals[-1] == s
+        for c in vals[:-1]:
+            assert int(toks[c - 1]) == eod
+        lengths = [b - a for a, b in zip([0] + vals, vals)]
+        psp = PackedSeqParams.from_lengths(lengths)
+        pos = psp.positions()
+        off = 0
+        for L in lengths:
+            assert pos[off] == 0 and int(pos[off + L - 1]) == L - 1
+            off += L
