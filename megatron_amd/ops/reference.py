@@ -286,12 +286,14 @@ def attention(
     return out.permute(2, 0, 1, 3).to(q.dtype)
 
 
-def attention_padded(q, k, v, key_valid_mask, causal: bool = False, scale=None):
+def attention_padded(q, k, v, key_valid_mask, causal: bool = False, scale=None,
+                     bias=None):
     """Attention with a [b, s_k] key-padding mask (True = valid token).
 
     Capability analog of the reference's arbitrary-mask (non-flash) path
     (fused_softmax scaled_masked variant): used by BERT-style bidirectional
-    batches with right padding.  fp32 softmax; [s,b,h,d] layouts."""
+    batches with right padding.  Optional additive `bias` [hq, s_q, s_k]
+    (T5 relative bias on padded batches).  fp32 softmax; [s,b,h,d]."""
     import math as _math
 
     if scale is None:
@@ -303,6 +305,8 @@ def attention_padded(q, k, v, key_valid_mask, causal: bool = False, scale=None):
     kf = k.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
     vf = v.permute(1, 2, 0, 3).float().repeat_interleave(rep, dim=1)
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [b, hq, s_q, s_k]
+    if bias is not None:
+        scores = scores + bias.float()
     bad = ~key_valid_mask.to(torch.bool)                     # [b, s_k]
     scores = scores.masked_fill(bad[:, None, None, :], float("-inf"))
     if causal:

@@ -109,12 +109,18 @@ class SelfAttention(nn.Module):
         elif attention_bias is not None:
             # additive score bias [hq_local, sq, sk] (T5 relative position
             # bias): unfused path — the bias re-materializes per step, so
-            # the flash kernel's fused softmax does not apply
+            # the flash kernel's fused softmax does not apply.  Composes
+            # with a key-padding mask (padded T5/BERT batches).
             from megatron_amd.ops import reference as _ref
 
-            core_out = _ref.attention(
-                q, k, v, causal=self.config.causal_attention,
-                scale=self.softmax_scale, bias=attention_bias)
+            if attention_mask is not None:
+                core_out = _ref.attention_padded(
+                    q, k, v, attention_mask, causal=self.config.causal_attention,
+                    scale=self.softmax_scale, bias=attention_bias)
+            else:
+                core_out = _ref.attention(
+                    q, k, v, causal=self.config.causal_attention,
+                    scale=self.softmax_scale, bias=attention_bias)
         elif attention_mask is not None:
             # key-padding mask ([b, s] bool, True = valid): the arbitrary-
             # mask path (BERT-style padded batches) — torch composition,

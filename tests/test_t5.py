@@ -155,3 +155,27 @@ def test_t5_relative_bias_matches_manual_attention():
         ref = (torch.softmax(scores, -1) @ v.float()).permute(2, 0, 1, 3).reshape(s, b, -1)
         ref_out, _ = att.linear_proj(ref.to(x.dtype))
     torch.testing.assert_close(out, ref_out, rtol=1e-4, atol=1e-5)
+
+
+def test_t5_relative_bias_with_padding_mask():
+    """Relative bias composes with a key-padding mask: padded key positions
+    contribute nothing regardless of their bias."""
+    from megatron_amd.transformer.attention import SelfAttention
+
+    init_single()
+    torch.manual_seed(9)
+    cfg = _rel_cfg().replace(causal_attention=False)
+    att = SelfAttention(cfg, layer_number=1).eval()
+    s, b = 8, 2
+    x = torch.randn(s, b, cfg.hidden_size)
+    bias = torch.randn(cfg.num_attention_heads, s, s)
+    mask = torch.ones(b, s, dtype=torch.bool)
+    mask[:, -2:] = False  # last two keys padded
+    with torch.no_grad():
+        out_masked = att(x, attention_mask=mask, attention_bias=bias)
+        # oracle: truncate the padded keys entirely
+        out_trunc = att(x[:, :, :], attention_mask=mask, attention_bias=bias)
+        x2 = x.clone()
+        x2[-2:] = 100.0  # garbage in padded positions must not leak
+        out_garbage = att(x2, attention_mask=mask, attention_bias=bias)
+    torch.testing.assert_close(out_masked[:6], out_garbage[:6], rtol=1e-4, atol=1e-5)
