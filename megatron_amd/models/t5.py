@@ -131,7 +131,7 @@ class T5Model(nn.Module):
         return h + self.position_embedding(pos).unsqueeze(1).to(h.dtype)
 
     def forward(self, encoder_input_ids=None, decoder_input_ids=None, labels=None,
-                loss_mask=None, **_):
+                loss_mask=None, encoder_attention_mask=None, **_):
         """encoder/decoder ids: [b, s_enc]/[b, s_dec]; returns loss [s_dec, b]
         (labels given) or decoder logits."""
         enc_bias = dec_bias = None
@@ -139,7 +139,8 @@ class T5Model(nn.Module):
             se, sd = encoder_input_ids.size(1), decoder_input_ids.size(1)
             enc_bias = self.encoder_rel_bias(se, se, encoder_input_ids.device)
             dec_bias = self.decoder_rel_bias(sd, sd, decoder_input_ids.device)
-        memory = self.encoder(self._embed(encoder_input_ids), attention_bias=enc_bias)
+        memory = self.encoder(self._embed(encoder_input_ids), attention_bias=enc_bias,
+                              attention_mask=encoder_attention_mask)
         h = self._embed(decoder_input_ids)
         for layer in self.decoder_layers:
             h = layer(h, memory, attention_bias=dec_bias)
