@@ -266,8 +266,8 @@ def test_activation_cpu_offload_gpu():
                                 activation_cpu_offload=offload,
                                 gradient_accumulation_fusion=False)
         m = GPTModel(cfg).cuda()
-        ids = torch.randint(0, 512, (2, 64), device="cuda",
-                            generator=torch.Generator("cuda").manual_seed(3))
+        ids = torch.randint(0, 512, (2, 64),
+                            generator=torch.Generator().manual_seed(3)).cuda()
         loss = m(ids, labels=ids).float().sum()
         loss.backward()
         torch.cuda.synchronize()
