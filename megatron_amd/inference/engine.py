@@ -250,6 +250,15 @@ class StaticInferenceEngine:
             draft_fn = None
         if draft_fn is None and mtp_drafter is None:
             draft_fn = lambda toks: _prompt_lookup_draft(toks, num_draft)
+        try:
+            return self._generate_speculative_impl(prompts, params, draft_fn,
+                                                   mtp_drafter, num_draft)
+        finally:
+            if mtp_drafter is not None:
+                mtp_drafter.remove()
+
+    def _generate_speculative_impl(self, prompts, params, draft_fn, mtp_drafter,
+                                   num_draft):
 
         results = []
         for rid, prompt in enumerate(prompts):
@@ -297,8 +306,6 @@ class StaticInferenceEngine:
             if self.tokenizer is not None:
                 r.text = self.tokenizer.detokenize(r.output_tokens)
             results.append(r)
-        if mtp_drafter is not None:
-            mtp_drafter.remove()
         return results
 
 
