@@ -376,7 +376,7 @@ class DynamicInferenceEngine:
     def __init__(self, model, tokenizer=None, num_blocks: int = 512, block_size: int = 256,
                  max_batch: int = 64, max_prefill_tokens: int = 8192, device=None,
                  use_hip_graphs: bool = True, enable_prefix_caching: bool = True,
-                 kv_cache_dtype=None):
+                 kv_cache_dtype=None, scheduling_policy: str = "fcfs"):
         self.model = model.eval()
         self.tokenizer = tokenizer
         cfg = model.config
@@ -391,6 +391,8 @@ class DynamicInferenceEngine:
             dtype=cfg.params_dtype, device=device, kv_cache_dtype=kv_cache_dtype)
         self.max_batch = max_batch
         self.max_prefill_tokens = max_prefill_tokens
+        assert scheduling_policy in ("fcfs", "priority", "sjf")
+        self.scheduling_policy = scheduling_policy
         self._graphs = None
         pp = G.get_grid().pp if G.grid_initialized() else 1
         if (use_hip_graphs and torch.cuda.is_available() and device.type == "cuda"
@@ -412,14 +414,28 @@ class DynamicInferenceEngine:
             # swap in the refcounted reuse allocator (same id space)
             self.context.allocator = PrefixCachingAllocator(num_blocks, first_id=1)
 
-    def add_request(self, prompt, params: SamplingParams = SamplingParams()) -> int:
+    def add_request(self, prompt, params: SamplingParams = SamplingParams(),
+                    priority: int = 0) -> int:
+        """Lower `priority` values schedule first (ties: FCFS).  With
+        scheduling_policy="sjf", shorter prompts are prefilled first inside
+        a priority class."""
         if self.tokenizer is not None and isinstance(prompt, str):
             prompt = self.tokenizer.tokenize(prompt)
         rid = next(self._ids)
         gen = torch.Generator(device="cpu").manual_seed(params.seed) if params.seed is not None else None
         req = _Request(rid, list(prompt), params, GenerationResult(rid, list(prompt)), gen=gen)
+        req.priority = priority
         self.waiting.append(req)
         return rid
+
+    def _next_waiting(self):
+        """Pick the next prefill candidate per the scheduling policy."""
+        if self.scheduling_policy == "fcfs":
+            return self.waiting[0]
+        if self.scheduling_policy == "sjf":
+            return min(self.waiting,
+                       key=lambda r: (getattr(r, "priority", 0), len(r.prompt), r.rid))
+        return min(self.waiting, key=lambda r: (getattr(r, "priority", 0), r.rid))
 
     def has_work(self) -> bool:
         return bool(self.waiting or self.active or self.preempted)
@@ -584,7 +600,7 @@ class DynamicInferenceEngine:
         batched decode step over all active requests."""
         self._restore_preempted()
         if self.waiting and len(self.active) < self.max_batch:
-            req = self.waiting[0]
+            req = self._next_waiting()
             if self.prefix_caching and req.cached == 0 and not req.block_table:
                 self._reuse_prefix(req)
             chunk = min(self.max_prefill_tokens, len(req.prompt) - req.cached)
@@ -607,7 +623,7 @@ class DynamicInferenceEngine:
                         tok = None
                     tok = self._pp_share_tokens([tok] if tok is not None else None, 1)[0]
                     req.result.output_tokens.append(tok)
-                    self.waiting.pop(0)
+                    self.waiting.remove(req)
                     if req.params.stop_on_eod and tok == self.eod:
                         req.result.output_tokens.pop()
                         self._finish(req)

@@ -872,3 +872,37 @@ def test_prompt_logprobs_match_manual(tiny_model):
         assert len(r.prompt_log_probs) == len(p) - 1
         for a, b in zip(r.prompt_log_probs, want):
             assert abs(a - b) < 1e-4, (a, b)
+
+
+def test_scheduling_policies(tiny_model):
+    """priority/SJF scheduling: lower priority value (or shorter prompt)
+    prefills first; outputs are unchanged vs FCFS for identical requests."""
+    params = SamplingParams(max_tokens=4, greedy=True, stop_on_eod=False)
+    long_p = list(range(1, 13))
+    short_p = [5, 1]
+
+    eng = DynamicInferenceEngine(tiny_model, num_blocks=32, block_size=4,
+                                 scheduling_policy="sjf",
+                                 max_prefill_tokens=64)
+    r_long = eng.add_request(long_p, params)
+    r_short = eng.add_request(short_p, params)
+    eng.step()  # SJF: the SHORT prompt must enter active first
+    active_rids = [r.rid for r in eng.active]
+    assert active_rids == [r_short]
+    while eng.has_work():
+        eng.step()
+
+    pri = DynamicInferenceEngine(tiny_model, num_blocks=32, block_size=4,
+                                 scheduling_policy="priority",
+                                 max_prefill_tokens=64)
+    a = pri.add_request(long_p, params, priority=5)
+    b = pri.add_request(short_p, params, priority=0)
+    pri.step()
+    assert [r.rid for r in pri.active] == [b]
+    while pri.has_work():
+        pri.step()
+
+    fcfs = DynamicInferenceEngine(tiny_model, num_blocks=32, block_size=4)
+    expected = fcfs.generate([long_p, short_p], params)
+    assert eng.finished[r_long].output_tokens == expected[0].output_tokens
+    assert eng.finished[r_short].output_tokens == expected[1].output_tokens
