@@ -376,21 +376,28 @@ def test_rccl_registered_pool_world1():
 
     from megatron_amd.distributed.rccl_allocator import RcclRegisteredPool
 
+    created = False
     if not dist.is_initialized():
         import os
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29771")
         dist.init_process_group("nccl", rank=0, world_size=1)
+        created = True
     try:
-        pool = RcclRegisteredPool(None)
-    except RuntimeError as e:
-        pytest.skip(f"mem-pool registration unavailable: {e}")
-    if not pool.active:
-        pytest.skip("no nccl backend resolved for the default group")
-    with pool.use():
-        t = torch.ones(1 << 20, device="cuda")
-    dist.all_reduce(t)
-    torch.cuda.synchronize()
-    assert float(t.sum()) == float(1 << 20)
-    pool.close()
+        try:
+            pool = RcclRegisteredPool(None)
+        except RuntimeError as e:
+            pytest.skip(f"mem-pool registration unavailable: {e}")
+        if not pool.active:
+            pytest.skip("no nccl backend resolved for the default group")
+        with pool.use():
+            t = torch.ones(1 << 20, device="cuda")
+        dist.all_reduce(t)
+        torch.cuda.synchronize()
+        assert float(t.sum()) == float(1 << 20)
+        pool.close()
+    finally:
+        if created:
+            # leave no global pg behind: later tests expect a fresh state
+            dist.destroy_process_group()
