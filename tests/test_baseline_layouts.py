@@ -159,3 +159,41 @@ def test_tp2_cp2_composition_matches_single(tmp_path, monkeypatch):
     spawn_dist(_tp2cp2_case, 4, ckpt)
     got = json.load(open(out))[0]
     assert abs(got - ref_loss) < 2e-3, (got, ref_loss)
+
+
+def _dp8_case(rank, world, ckpt_dir):
+    """DP=8 (the Llama-3-8B headline layout): per-rank data shards, dist-opt
+    ZeRO-1, loss equals the single-process run over the union batch."""
+    G.initialize_model_parallel()
+    model_parallel_seed(1234)
+    cfg = _cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0, use_distributed_optimizer=True)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000,
+                        use_distributed_optimizer=True)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    load_checkpoint(ckpt_dir, chunks, opt, load_rng=False)
+    batches = _gen_batches(8)
+    mine = [batches[rank]]
+    r = train_step(forward_step, [iter(mine)], chunks, opt, cfg, 1, SEQ, 2)
+    if rank == 0:
+        with open(os.environ["LAYOUT_TEST_OUT"], "w") as f:
+            json.dump([r["lm_loss"]], f)
+
+
+def test_dp8_matches_single(tmp_path, monkeypatch):
+    out = tmp_path / "dp8.json"
+    ckpt = str(tmp_path / "ckpt")
+    monkeypatch.setenv("LAYOUT_TEST_OUT", str(out))
+    init_single()
+    model_parallel_seed(1234)
+    cfg = _cfg()
+    opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0, use_distributed_optimizer=True)
+    ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, bucket_size=10_000,
+                        use_distributed_optimizer=True)
+    chunks, opt = setup_model_and_optimizer(_provider, cfg, opt_cfg, ddp_cfg)
+    save_checkpoint(ckpt, chunks, opt, iteration=0)
+    r = train_step(forward_step, [iter(_gen_batches(8))], chunks, opt, cfg, 8, SEQ, 2)
+    ref_loss = r["lm_loss"]
+    spawn_dist(_dp8_case, 8, ckpt)
+    got = json.load(open(out))[0]
+    assert abs(got - ref_loss) < 5e-4, (got, ref_loss)
