@@ -204,3 +204,60 @@ def test_tp2_combined_matches_standard(tmp_path, monkeypatch):
     std, cmb = json.load(open(out_s)), json.load(open(out_c))
     for a, b in zip(std, cmb):
         assert abs(a - b) < 1e-5, (std, cmb)
+
+
+def test_combined_schedule_with_distributed_optimizer():
+    """Combined co-schedule x dist-opt (the Mixtral bench configuration):
+    losses and updated weights equal the standard schedule under ZeRO-1."""
+    from megatron_amd.config import DDPConfig, OptimizerConfig, TransformerConfig
+    from megatron_amd.models.gpt import GPTModel
+    from megatron_amd.parallel.random import model_parallel_seed
+    from megatron_amd.training.training import setup_model_and_optimizer, train_step
+    from tests.utils import init_single
+
+    def provider(config, pre_process=True, post_process=True, vp_stage=None):
+        torch.manual_seed(42)
+        return GPTModel(config)
+
+    def run(combined):
+        init_single()
+        model_parallel_seed(21)
+        cfg = TransformerConfig(
+            num_layers=2, hidden_size=64, num_attention_heads=4, num_query_groups=2,
+            vocab_size=96, ffn_hidden_size=96, num_experts=4, moe_router_topk=2,
+            moe_ffn_hidden_size=64, moe_aux_loss_coeff=0.01,
+            overlap_moe_expert_parallel_comm=combined,
+            gradient_accumulation_fusion=True)
+        opt_cfg = OptimizerConfig(lr=1e-3, clip_grad=1.0, use_distributed_optimizer=True)
+        ddp_cfg = DDPConfig(grad_reduce_in_fp32=True, use_distributed_optimizer=True,
+                            bucket_size=10_000)
+        chunks, opt = setup_model_and_optimizer(provider, cfg, opt_cfg, ddp_cfg)
+        g = torch.Generator().manual_seed(9)
+        batches = []
+        for _ in range(4):
+            t = torch.randint(0, 96, (2, 17), generator=g)
+            batches.append({"tokens": t[:, :-1], "labels": t[:, 1:]})
+
+        def fwd(it, model):
+            batch = next(it)
+
+            def loss_func(loss_sb):
+                s = loss_sb.sum()
+                return s, torch.tensor(loss_sb.numel()), {"loss_sum": s.detach()}
+
+            return model(batch["tokens"], labels=batch["labels"]), loss_func
+
+        losses = []
+        for s in range(2):
+            r = train_step(fwd, [iter(batches[s * 2:(s + 1) * 2])], chunks, opt,
+                           cfg, 2, 16, 2)
+            losses.append(r["lm_loss"])
+        params = {n: p.detach().clone() for n, p in chunks[0].module.named_parameters()}
+        return losses, params
+
+    l_std, p_std = run(False)
+    l_cmb, p_cmb = run(True)
+    for a, b in zip(l_std, l_cmb):
+        assert abs(a - b) < 1e-5, (l_std, l_cmb)
+    for n in p_std:
+        torch.testing.assert_close(p_cmb[n], p_std[n], rtol=1e-5, atol=1e-6, msg=n)
